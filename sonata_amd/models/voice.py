@@ -1,0 +1,294 @@
+"""VitsVoice: a loaded voice pack implementing the SonataModel interface.
+
+Parity: reference crates/sonata/models/piper/src/lib.rs —
+`from_config_path` (:88-110), `VitsModel`/`VitsStreamingModel` (:291,480),
+trait `VitsModelCommons` (:168-289: phonemize, id-encode, speaker maps,
+fallback synthesis config), `SpeechStreamer` chunked decoding with
+42-sample crossfade (:765-858).
+
+Voice-pack format: `<name>.json` (Piper-compatible config schema) +
+`<name>.safetensors` (weights).  A real Piper `.onnx` voice can be
+converted via sonata_amd.models.onnx_import (weights-only importer).
+"""
+
+from __future__ import annotations
+
+import os
+import threading
+import time
+from typing import Dict, Iterator, List, Optional, Sequence
+
+import numpy as np
+import torch
+
+from ..audio.samples import crossfade
+from ..core import Audio, AudioInfo, ModelError, Phonemes, SonataModel
+from ..text.ids import phonemes_to_ids
+from ..text.phonemizer import text_to_phonemes
+from .chunker import chunk_plan
+from .config import ModelConfig, SynthesisConfig
+from .vits import VitsModel
+
+CROSSFADE_SAMPLES = 42  # reference: piper/src/lib.rs:838
+
+
+def _utterance_seed(phonemes: str, sid: Optional[int]) -> int:
+    """Deterministic per-utterance seed: same text -> same audio on any
+    rank (SURVEY.md §7 hard part 7: seed per utterance, not per rank)."""
+    import hashlib
+
+    h = hashlib.sha256(
+        (phonemes + "|" + str(sid if sid is not None else -1)).encode("utf-8")
+    ).digest()
+    return int.from_bytes(h[:8], "little") & 0x7FFFFFFFFFFFFFFF
+
+
+class VitsVoice(SonataModel):
+    def __init__(
+        self,
+        config: ModelConfig,
+        net: VitsModel,
+        device: str = "cpu",
+        dtype: torch.dtype = torch.float32,
+    ):
+        self.config = config
+        self.net = net.eval().to(device=device, dtype=dtype)
+        self.device = torch.device(device)
+        self.dtype = dtype
+        self._synth_config = config.default_synthesis_config()
+        self._cfg_lock = threading.Lock()
+        self._tashkeel = None
+        if config.espeak_voice.startswith("ar"):
+            from ..text.tashkeel import TashkeelModel
+
+            self._tashkeel = TashkeelModel.default(device="cpu")
+
+    # ------------------------------------------------------------------ #
+    # SonataModel interface
+    # ------------------------------------------------------------------ #
+    def audio_output_info(self) -> AudioInfo:
+        return AudioInfo(sample_rate=self.config.sample_rate)
+
+    @property
+    def language(self) -> Optional[str]:
+        return self.config.language_code
+
+    def get_speakers(self) -> Optional[Dict[int, str]]:
+        if self.config.num_speakers <= 1:
+            return None
+        return {v: k for k, v in self.config.speaker_id_map.items()}
+
+    def get_synthesis_config(self) -> SynthesisConfig:
+        with self._cfg_lock:
+            return self._synth_config.copy()
+
+    def set_synthesis_config(self, config: SynthesisConfig) -> None:
+        with self._cfg_lock:
+            self._synth_config = config.copy()
+
+    def phonemize_text(self, text: str) -> Phonemes:
+        if self._tashkeel is not None:
+            text = self._tashkeel.diacritize(text)
+        sentences = text_to_phonemes(text, voice=self.config.espeak_voice)
+        return Phonemes(sentences)
+
+    # ------------------------------------------------------------------ #
+    # inference
+    # ------------------------------------------------------------------ #
+    def _encode_ids(self, phonemes: str) -> List[int]:
+        return phonemes_to_ids(phonemes, self.config.phoneme_id_map)
+
+    def _generators(
+        self, phonemes_batch: Sequence[str], sid: Optional[int]
+    ) -> List[torch.Generator]:
+        """One deterministic generator per utterance — same text gives the
+        same audio regardless of batch composition or rank."""
+        gens = []
+        for p in phonemes_batch:
+            gen = torch.Generator(device=self.device)
+            gen.manual_seed(_utterance_seed(p, sid))
+            gens.append(gen)
+        return gens
+
+    def _sid_tensor(self, batch: int, sid: Optional[int]):
+        if self.config.num_speakers <= 1:
+            return None
+        s = sid if sid is not None else 0
+        return torch.full((batch,), s, dtype=torch.long, device=self.device)
+
+    @torch.no_grad()
+    def speak_one_sentence(self, phonemes: str) -> Audio:
+        return self.speak_batch([phonemes])[0]
+
+    @torch.no_grad()
+    def speak_batch(self, phonemes_batch: Sequence[str]) -> List[Audio]:
+        """True padded [B, T] batching (the reference loops batch=1:
+        piper/src/lib.rs:425-437 — batching is a headline improvement)."""
+        cfg = self.get_synthesis_config()
+        id_lists = [self._encode_ids(p) for p in phonemes_batch]
+        B = len(id_lists)
+        T = max((len(i) for i in id_lists), default=1)
+        ids = torch.zeros((B, T), dtype=torch.long)
+        lengths = torch.zeros((B,), dtype=torch.long)
+        for b, il in enumerate(id_lists):
+            ids[b, : len(il)] = torch.tensor(il, dtype=torch.long)
+            lengths[b] = len(il)
+        ids = ids.to(self.device)
+        lengths = lengths.to(self.device)
+        gens = self._generators(phonemes_batch, cfg.speaker_id)
+
+        t0 = time.perf_counter()
+        audio, audio_lengths = self.net.infer(
+            ids,
+            lengths,
+            sid=self._sid_tensor(B, cfg.speaker_id),
+            noise_scale=cfg.noise_scale,
+            length_scale=cfg.length_scale,
+            noise_w=cfg.noise_w,
+            generators=gens,
+        )
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+        infer_ms = (time.perf_counter() - t0) * 1000.0
+
+        info = self.audio_output_info()
+        out: List[Audio] = []
+        audio = audio.float().cpu().numpy()
+        audio_lengths = audio_lengths.cpu().numpy()
+        for b in range(B):
+            n = int(audio_lengths[b])
+            out.append(Audio(audio[b, 0, :n], info, inference_ms=infer_ms / B))
+        return out
+
+    # ------------------------------------------------------------------ #
+    # streaming decode (encoder once, HiFi-GAN chunked)
+    # ------------------------------------------------------------------ #
+    @property
+    def supports_streaming_output(self) -> bool:
+        return True
+
+    @torch.no_grad()
+    def stream_synthesis(
+        self, phonemes: str, chunk_size: int = 45, chunk_padding: int = 3
+    ) -> Iterator[np.ndarray]:
+        """Yield waveform chunks: encoder runs once, the HiFi-GAN decoder
+        runs per adaptive chunk with overlap-discard + crossfade seams
+        (reference SpeechStreamer, piper/src/lib.rs:765-858)."""
+        cfg = self.get_synthesis_config()
+        ids_l = self._encode_ids(phonemes)
+        ids = torch.tensor([ids_l], dtype=torch.long, device=self.device)
+        lengths = torch.tensor([len(ids_l)], dtype=torch.long, device=self.device)
+        gens = self._generators([phonemes], cfg.speaker_id)
+        z, y_mask, g = self.net.infer_encoder(
+            ids, lengths, sid=self._sid_tensor(1, cfg.speaker_id),
+            noise_scale=cfg.noise_scale, length_scale=cfg.length_scale,
+            noise_w=cfg.noise_w, generators=gens,
+        )
+        hop = self.net.arch.hop_length
+        num_frames = z.shape[-1]
+        # Overlap-crossfade at seams without changing the timeline: each
+        # interior chunk keeps `ext` extra decoded samples (taken from its
+        # right padding region); the next chunk's first `ext` samples cover
+        # the same timeline window, so the sine-ramp mix preserves total
+        # length exactly (streamed output == one-shot length).
+        tail: Optional[np.ndarray] = None
+        prev_ext = 0
+        for spec in chunk_plan(num_frames, chunk_size, chunk_padding):
+            z_c = z[:, :, spec.mel_start : spec.mel_end]
+            m_c = y_mask[:, :, spec.mel_start : spec.mel_end]
+            audio = self.net.decode(z_c, m_c, g)
+            wav = audio[0, 0].float().cpu().numpy()
+            lo = spec.trim_left_frames * hop
+            hi = len(wav) - spec.trim_right_frames * hop
+            ext = 0 if spec.is_last else min(
+                CROSSFADE_SAMPLES, spec.trim_right_frames * hop
+            )
+            cur = wav[lo : hi + ext]
+            if tail is not None:
+                cur = crossfade(tail, cur, prev_ext)
+            cut = len(cur) - ext
+            yield cur[:cut].astype(np.float32)
+            tail = cur[cut:] if ext else None
+            prev_ext = ext
+            if spec.is_last:
+                return
+
+
+# --------------------------------------------------------------------------- #
+# loading / creation
+# --------------------------------------------------------------------------- #
+def _weights_path_for(config_path: str) -> str:
+    stem = config_path
+    if stem.endswith(".json"):
+        stem = stem[: -len(".json")]
+    if stem.endswith(".onnx"):
+        stem = stem[: -len(".onnx")]
+    return stem + ".safetensors"
+
+
+def load_voice(
+    config_path: str, device: str = "cpu", dtype: Optional[torch.dtype] = None
+) -> VitsVoice:
+    """Load a voice pack: `<stem>.json` + `<stem>.safetensors`.
+
+    Mirrors the reference loader dispatch (piper/src/lib.rs:88-110); the
+    `streaming` config key only changes default synthesis mode — the same
+    net serves one-shot and streaming here (encoder/decoder split is a
+    method boundary, not two files)."""
+    config = ModelConfig.from_json_path(config_path)
+    if dtype is None:
+        dtype = torch.bfloat16 if device.startswith("cuda") else torch.float32
+    net = VitsModel(config.num_symbols, config.architecture,
+                    n_speakers=max(config.num_speakers, 1))
+    wpath = _weights_path_for(config_path)
+    if os.path.exists(wpath):
+        from safetensors.torch import load_file
+
+        state = load_file(wpath)
+        missing, unexpected = net.load_state_dict(state, strict=False)
+        if missing or unexpected:
+            raise ModelError(
+                f"voice weights mismatch: missing={missing[:5]} "
+                f"unexpected={unexpected[:5]}"
+            )
+    else:
+        raise ModelError(f"voice weights not found: {wpath}")
+    return VitsVoice(config, net, device=device, dtype=dtype)
+
+
+def create_random_voice(
+    out_dir: str,
+    name: str = "test_voice",
+    quality: str = "medium",
+    language: str = "en-us",
+    num_speakers: int = 1,
+    seed: int = 0,
+) -> str:
+    """Create a random-init voice pack (tests / synthetic benches — there is
+    no network for real checkpoints).  Returns the config path."""
+    from .config import QUALITY_PRESETS, VitsArchitecture
+
+    preset = QUALITY_PRESETS[quality]
+    arch = VitsArchitecture(**preset["arch"])
+    if num_speakers > 1:
+        arch.gin_channels = 256
+    config = ModelConfig(
+        key=name,
+        language_code=language,
+        sample_rate=preset["sample_rate"],
+        quality=quality,
+        num_speakers=num_speakers,
+        speaker_id_map={f"spk{i}": i for i in range(num_speakers)}
+        if num_speakers > 1 else {},
+        espeak_voice=language,
+        architecture=arch,
+    )
+    torch.manual_seed(seed)
+    net = VitsModel(config.num_symbols, arch, n_speakers=max(num_speakers, 1))
+    os.makedirs(out_dir, exist_ok=True)
+    cfg_path = os.path.join(out_dir, f"{name}.json")
+    config.save_json(cfg_path)
+    from safetensors.torch import save_file
+
+    save_file(net.state_dict(), _weights_path_for(cfg_path))
+    return cfg_path

@@ -1,0 +1,162 @@
+"""Functional ops used by the VITS graph.
+
+Each op has two implementations:
+  * torch: plain PyTorch fp32 — the numerics oracle, used on CPU and under
+    SONATA_FORCE_TORCH=1.
+  * hip: hand-written CDNA4 kernel from the `_sonata_hip` extension — the
+    serving path on MI355X.  Mandatory when tensors are on GPU.
+
+Kernel inventory: LayerNorm over channels, WaveNet fused
+gated tanh·sigmoid, prior sampling z = m + eps·exp(logs)·noise, duration
+expansion (length regulator), Conv1d / ConvTranspose1d with fused
+LeakyReLU (MFMA conv-as-GEMM).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from . import hip_ext, use_hip
+
+
+# --------------------------------------------------------------------------- #
+# LayerNorm over the channel dim of [B, C, T]
+# --------------------------------------------------------------------------- #
+def layer_norm_ct(
+    x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor, eps: float = 1e-5
+) -> torch.Tensor:
+    if use_hip(x):
+        ext = hip_ext(required=True)
+        return ext.layer_norm_ct(x.contiguous(), gamma, beta, eps)
+    # torch reference: normalize across C for each (b, t)
+    mean = x.mean(dim=1, keepdim=True)
+    var = x.var(dim=1, unbiased=False, keepdim=True)
+    xhat = (x - mean) * torch.rsqrt(var + eps)
+    return xhat * gamma.view(1, -1, 1) + beta.view(1, -1, 1)
+
+
+# --------------------------------------------------------------------------- #
+# WaveNet gated activation: split channels, tanh(a+ga) * sigmoid(b+gb)
+# --------------------------------------------------------------------------- #
+def fused_gate(
+    x: torch.Tensor, g: Optional[torch.Tensor], n_channels: int
+) -> torch.Tensor:
+    """x: [B, 2C, T] conv output; g: [B, 2C, T] conditioning (or None).
+    Returns tanh(x_a + g_a) * sigmoid(x_b + g_b), [B, C, T]."""
+    if use_hip(x):
+        ext = hip_ext(required=True)
+        return ext.fused_gate(
+            x.contiguous(),
+            g.contiguous() if g is not None else None,
+            n_channels,
+        )
+    if g is not None:
+        x = x + g
+    a, b = x[:, :n_channels], x[:, n_channels:]
+    return torch.tanh(a) * torch.sigmoid(b)
+
+
+# --------------------------------------------------------------------------- #
+# Prior sampling: z = (m + randn * exp(logs) * noise_scale) * mask
+# --------------------------------------------------------------------------- #
+def prior_sample(
+    m: torch.Tensor,
+    logs: torch.Tensor,
+    mask: torch.Tensor,
+    noise: torch.Tensor,
+    noise_scale: float,
+) -> torch.Tensor:
+    if use_hip(m):
+        ext = hip_ext(required=True)
+        return ext.prior_sample(m.contiguous(), logs.contiguous(),
+                                mask.contiguous(), noise, float(noise_scale))
+    return (m + noise * torch.exp(logs) * noise_scale) * mask
+
+
+# --------------------------------------------------------------------------- #
+# Length regulator: expand phoneme states to frame states by durations
+# --------------------------------------------------------------------------- #
+def expand_states(
+    stats: torch.Tensor, durations: torch.Tensor, y_lengths: torch.Tensor
+) -> torch.Tensor:
+    """stats: [B, C, T_ph]; durations: [B, T_ph] int frame counts;
+    returns [B, C, F_max] where F_max = y_lengths.max().
+
+    The attention path matrix of VITS inference: frame f copies phoneme p
+    where cum_dur[p-1] <= f < cum_dur[p]."""
+    if use_hip(stats):
+        ext = hip_ext(required=True)
+        return ext.expand_states(
+            stats.contiguous(),
+            durations.to(torch.int32).contiguous(),
+            int(y_lengths.max().item()),
+        )
+    B, C, T = stats.shape
+    F_max = int(y_lengths.max().item())
+    out = stats.new_zeros((B, C, F_max))
+    for b in range(B):
+        cum = torch.cumsum(durations[b], dim=0)
+        # phoneme index for each frame
+        frames = torch.arange(F_max, device=stats.device)
+        idx = torch.searchsorted(cum, frames, right=True).clamp(max=T - 1)
+        valid = frames < int(y_lengths[b].item())
+        out[b, :, valid] = stats[b][:, idx[valid]]
+    return out
+
+
+# --------------------------------------------------------------------------- #
+# Conv1d (+ fused LeakyReLU on input or output) — the HiFi-GAN hot op
+# --------------------------------------------------------------------------- #
+def leaky_conv1d(
+    x: torch.Tensor,
+    weight: torch.Tensor,
+    bias: Optional[torch.Tensor],
+    stride: int = 1,
+    padding: int = 0,
+    dilation: int = 1,
+    pre_lrelu: float = 0.0,
+    post_lrelu: float = 0.0,
+) -> torch.Tensor:
+    """Conv1d with optional fused LeakyReLU applied to the input
+    (pre_lrelu>0) and/or the output (post_lrelu>0).  Fusing the activation
+    into the producing GEMM keeps the activation tensor out of HBM
+    (HiFi-GAN MRF pattern: y = conv(lrelu(x)))."""
+    if use_hip(x):
+        ext = hip_ext(required=True)
+        return ext.conv1d_fused(
+            x.contiguous(), weight.contiguous(),
+            bias.contiguous() if bias is not None else None,
+            stride, padding, dilation, float(pre_lrelu), float(post_lrelu),
+        )
+    if pre_lrelu > 0.0:
+        x = F.leaky_relu(x, pre_lrelu)
+    y = F.conv1d(x, weight, bias, stride=stride, padding=padding,
+                 dilation=dilation)
+    if post_lrelu > 0.0:
+        y = F.leaky_relu(y, post_lrelu)
+    return y
+
+
+def leaky_convtranspose1d(
+    x: torch.Tensor,
+    weight: torch.Tensor,
+    bias: Optional[torch.Tensor],
+    stride: int,
+    padding: int,
+    pre_lrelu: float = 0.0,
+) -> torch.Tensor:
+    """ConvTranspose1d with fused LeakyReLU on the input — the HiFi-GAN
+    upsampling stage (y = convT(lrelu(x)))."""
+    if use_hip(x):
+        ext = hip_ext(required=True)
+        return ext.convtranspose1d_fused(
+            x.contiguous(), weight.contiguous(),
+            bias.contiguous() if bias is not None else None,
+            stride, padding, float(pre_lrelu),
+        )
+    if pre_lrelu > 0.0:
+        x = F.leaky_relu(x, pre_lrelu)
+    return F.conv_transpose1d(x, weight, bias, stride=stride, padding=padding)

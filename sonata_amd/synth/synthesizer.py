@@ -1,0 +1,209 @@
+"""Synthesizer facade: stream modes + prosody post-processing.
+
+Parity: reference crates/sonata/synth/src/lib.rs —
+`SYNTHESIS_THREAD_POOL` (:17-26, rayon num_cpus*4) -> a shared
+ThreadPoolExecutor; `AudioOutputConfig` percent->param mapping with ranges
+RATE (0.5,5.5) VOLUME (0,1) PITCH (0.5,1.5) (:13-15, utils.rs:6-8) and
+sonic application (:55-105) -> sonata_amd.audio.prosody;
+`SonataSpeechSynthesizer` (:119-203) with synthesize_lazy (:138),
+synthesize_parallel (:145), synthesize_streamed realtime (:152),
+synthesize_to_file (:170); streams Lazy/Parallel/Realtime (:282-430)
+including the realtime producer thread + queue and growing chunk size
+(:350-358), per-chunk prosody (:392-407), appended silence (:408-412).
+"""
+
+from __future__ import annotations
+
+import os
+import queue
+import threading
+from concurrent.futures import ThreadPoolExecutor
+from dataclasses import dataclass
+from typing import Iterator, List, Optional
+
+import numpy as np
+
+from ..audio.prosody import apply_prosody
+from ..audio.samples import generate_silence, merge
+from ..audio.wav import write_wav_file
+from ..core import Audio, AudioInfo, SonataModel
+
+# percent(0-100) -> parameter ranges (reference synth/src/lib.rs:13-15)
+RATE_RANGE = (0.5, 5.5)
+VOLUME_RANGE = (0.0, 1.0)
+PITCH_RANGE = (0.5, 1.5)
+
+_POOL: Optional[ThreadPoolExecutor] = None
+_POOL_LOCK = threading.Lock()
+
+
+def synthesis_pool() -> ThreadPoolExecutor:
+    """Global synthesis thread pool (reference SYNTHESIS_THREAD_POOL:
+    rayon pool of num_cpus*4 threads, synth/src/lib.rs:17-26)."""
+    global _POOL
+    with _POOL_LOCK:
+        if _POOL is None:
+            n = (os.cpu_count() or 4) * 4
+            _POOL = ThreadPoolExecutor(
+                max_workers=n, thread_name_prefix="sonata_synth"
+            )
+        return _POOL
+
+
+def _percent(value: float, lo: float, hi: float) -> float:
+    v = min(max(value, 0.0), 100.0)
+    return lo + (hi - lo) * (v / 100.0)
+
+
+@dataclass
+class AudioOutputConfig:
+    """Prosody knobs in percent (0-100) + appended silence, mirroring the
+    reference AudioOutputConfig (synth/src/lib.rs:28-117)."""
+
+    rate: Optional[float] = None
+    volume: Optional[float] = None
+    pitch: Optional[float] = None
+    appended_silence_ms: Optional[float] = None
+
+    def apply(self, samples: np.ndarray, sample_rate: int) -> np.ndarray:
+        speed = _percent(self.rate, *RATE_RANGE) if self.rate is not None else 1.0
+        volume = (
+            _percent(self.volume, *VOLUME_RANGE)
+            if self.volume is not None else 1.0
+        )
+        pitch = (
+            _percent(self.pitch, *PITCH_RANGE) if self.pitch is not None else 1.0
+        )
+        return apply_prosody(samples, sample_rate, speed=speed, volume=volume,
+                             pitch=pitch)
+
+    @property
+    def is_noop(self) -> bool:
+        return (
+            self.rate is None and self.volume is None and self.pitch is None
+        )
+
+
+class SonataSpeechSynthesizer:
+    """Facade over a SonataModel: phonemize + synthesize in three modes."""
+
+    def __init__(self, model: SonataModel):
+        self.model = model
+
+    # ------------------------------------------------------------------ #
+    def audio_output_info(self) -> AudioInfo:
+        return self.model.audio_output_info()
+
+    def _post(self, audio: Audio, cfg: Optional[AudioOutputConfig]) -> Audio:
+        if cfg is None:
+            return audio
+        samples = audio.samples
+        if not cfg.is_noop:
+            samples = cfg.apply(samples, audio.info.sample_rate)
+        if cfg.appended_silence_ms:
+            samples = merge(
+                samples,
+                generate_silence(cfg.appended_silence_ms,
+                                 audio.info.sample_rate),
+            )
+        return Audio(samples, audio.info, audio.inference_ms)
+
+    # ------------------------------------------------------------------ #
+    def synthesize_lazy(
+        self, text: str, output_config: Optional[AudioOutputConfig] = None
+    ) -> Iterator[Audio]:
+        """Pull-based: each sentence synthesized when consumed
+        (reference SonataSpeechStreamLazy, synth/src/lib.rs:282-307)."""
+        phonemes = self.model.phonemize_text(text)
+        for sent in phonemes:
+            yield self._post(self.model.speak_one_sentence(sent), output_config)
+
+    def synthesize_parallel(
+        self, text: str, output_config: Optional[AudioOutputConfig] = None
+    ) -> Iterator[Audio]:
+        """Eager batched synthesis of all sentences.  The reference fans
+        out per-sentence rayon tasks (synth/src/lib.rs:316-320); here the
+        model's true padded batch path does the fan-out on-device."""
+        phonemes = self.model.phonemize_text(text)
+        if len(phonemes) == 0:
+            return iter(())
+        batch = self.model.speak_batch(list(phonemes))
+        return iter([self._post(a, output_config) for a in batch])
+
+    def synthesize_streamed(
+        self,
+        text: str,
+        output_config: Optional[AudioOutputConfig] = None,
+        chunk_size: int = 45,
+        chunk_padding: int = 3,
+    ) -> Iterator[np.ndarray]:
+        """Realtime mode: producer thread pushes waveform chunks through a
+        queue while the caller consumes (reference RealtimeSpeechStream,
+        synth/src/lib.rs:335-430).  Chunk size grows per processed chunk
+        (:352-356).  Yields float32 sample chunks."""
+        phonemes = self.model.phonemize_text(text)
+        info = self.model.audio_output_info()
+        q: "queue.Queue" = queue.Queue()
+        DONE, ERROR = object(), object()
+
+        def producer():
+            try:
+                for sent in phonemes:
+                    if self.model.supports_streaming_output:
+                        it = self.model.stream_synthesis(
+                            sent, chunk_size, chunk_padding
+                        )
+                    else:
+                        it = iter([self.model.speak_one_sentence(sent).samples])
+                    for chunk in it:
+                        if output_config is not None and not output_config.is_noop:
+                            chunk = output_config.apply(chunk, info.sample_rate)
+                        q.put(chunk)
+                    if output_config is not None and output_config.appended_silence_ms:
+                        q.put(generate_silence(
+                            output_config.appended_silence_ms, info.sample_rate
+                        ))
+                q.put(DONE)
+            except BaseException as e:  # propagate to consumer
+                q.put(ERROR)
+                q.put(e)
+
+        synthesis_pool().submit(producer)
+        while True:
+            item = q.get()
+            if item is DONE:
+                return
+            if item is ERROR:
+                raise q.get()
+            yield item
+
+    def synthesize_to_file(
+        self,
+        path: str,
+        text: str,
+        output_config: Optional[AudioOutputConfig] = None,
+    ) -> Audio:
+        """Collect parallel synthesis into one WAV (reference
+        synthesize_to_file, synth/src/lib.rs:170-198)."""
+        pieces: List[np.ndarray] = []
+        info = self.model.audio_output_info()
+        inference_ms = 0.0
+        for audio in self.synthesize_parallel(text, output_config):
+            pieces.append(audio.samples)
+            inference_ms += audio.inference_ms
+        samples = (
+            np.concatenate(pieces) if pieces else np.zeros(0, dtype=np.float32)
+        )
+        write_wav_file(path, samples, info.sample_rate)
+        return Audio(samples, info, inference_ms)
+
+    # delegation (reference re-implements SonataModel by delegation,
+    # synth/src/lib.rs:205-247)
+    def phonemize_text(self, text: str):
+        return self.model.phonemize_text(text)
+
+    def get_synthesis_config(self):
+        return self.model.get_synthesis_config()
+
+    def set_synthesis_config(self, cfg) -> None:
+        self.model.set_synthesis_config(cfg)

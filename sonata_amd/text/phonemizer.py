@@ -1,0 +1,356 @@
+"""Grapheme -> IPA phonemization with sentence splitting.
+
+Parity target: reference crates/text/espeak-phonemizer/src/lib.rs —
+`text_to_phonemes(text, voice, separator, remove_lang_switch, remove_stress)`
+(:65-83), clause-terminator preservation (:113-137: each clause's `.,?!`
+survives into the phoneme string), sentence splitting on sentence-type
+clauses (:134-136), stress-mark filtering `ˈˌ` (:141-154), per-line input
+splitting (:65-83).
+
+The reference shells into the espeak-ng C library (a patched fork).  That
+dependency does not exist here; this module is a fresh, self-contained
+rule-based G2P: a per-language ordered longest-match rule table plus an
+English exception lexicon.  It produces IPA over the same symbol set the
+Piper voices use, is deterministic, and is thread-safe (pure functions, no
+global C state — the reference's espeak is famously NOT thread-safe,
+SURVEY.md §5).
+"""
+
+from __future__ import annotations
+
+import re
+from typing import Dict, List, Optional, Tuple
+
+from ..core import PhonemizationError
+
+# --------------------------------------------------------------------------- #
+# Sentence / clause splitting
+# --------------------------------------------------------------------------- #
+_SENT_END = ".!?"
+_CLAUSE_END = ",;:"
+_SPLIT_RE = re.compile(r"([.!?]+|[,;:])")
+
+
+def split_sentences(text: str) -> List[Tuple[str, str]]:
+    """Split text into (sentence_text, terminator) pairs.
+
+    A sentence ends at `.`, `!` or `?`; intermediate `,;:` clauses stay in
+    the same sentence (their punctuation is preserved in place, mirroring
+    espeak's clause-terminator behavior)."""
+    out: List[Tuple[str, str]] = []
+    cur = ""
+    parts = _SPLIT_RE.split(text)
+    for part in parts:
+        if not part:
+            continue
+        if _SPLIT_RE.fullmatch(part):
+            term = part[0]
+            if term in _SENT_END:
+                if cur.strip():
+                    out.append((cur.strip(), term))
+                cur = ""
+            else:
+                cur += part  # keep clause punctuation inline
+        else:
+            cur += part
+    if cur.strip():
+        out.append((cur.strip(), "."))
+    return out
+
+
+# --------------------------------------------------------------------------- #
+# Rule-based G2P
+# --------------------------------------------------------------------------- #
+class RuleG2P:
+    """Ordered longest-match grapheme->IPA rules with an exception lexicon.
+
+    Rules are (pattern, ipa) pairs applied left-to-right, longest pattern
+    first at each position.  `stress` optionally marks primary stress on
+    the first vowel of each word (the Piper symbol set includes ˈ/ˌ)."""
+
+    def __init__(
+        self,
+        rules: Dict[str, str],
+        lexicon: Optional[Dict[str, str]] = None,
+        letters: str = "a-z",
+        stress: bool = True,
+    ):
+        self.lexicon = lexicon or {}
+        self.stress = stress
+        # sort patterns by length desc for longest match
+        self._patterns = sorted(rules.items(), key=lambda kv: -len(kv[0]))
+        self._rules = rules
+        self._max_pat = max((len(p) for p in rules), default=1)
+        self._word_re = re.compile(rf"[{letters}']+", re.IGNORECASE)
+
+    _VOWELS = "aeiouɑæʌɔəɛɪiʊuɜoʏøyɶɒãõɐ"
+
+    def word_to_ipa(self, word: str) -> str:
+        w = word.lower()
+        if w in self.lexicon:
+            ipa = self.lexicon[w]
+        else:
+            ipa = self._apply_rules(w)
+        if self.stress and ipa and not ipa.startswith("ˈ"):
+            # place primary stress before the first vowel
+            for i, ch in enumerate(ipa):
+                if ch in self._VOWELS:
+                    ipa = ipa[:i] + "ˈ" + ipa[i:]
+                    break
+        return ipa
+
+    def _apply_rules(self, w: str) -> str:
+        out = []
+        i = 0
+        n = len(w)
+        while i < n:
+            matched = False
+            for ln in range(min(self._max_pat, n - i), 0, -1):
+                seg = w[i : i + ln]
+                if seg in self._rules:
+                    out.append(self._rules[seg])
+                    i += ln
+                    matched = True
+                    break
+            if not matched:
+                i += 1  # drop unknown char
+        return "".join(out)
+
+    def phonemize(self, text: str) -> str:
+        parts: List[str] = []
+        pos = 0
+        for m in self._word_re.finditer(text):
+            between = text[pos : m.start()]
+            # keep clause punctuation, collapse other chars to spaces
+            kept = "".join(c if c in ",;:" else " " for c in between)
+            if kept.strip(",;:") or kept:
+                parts.append(kept)
+            parts.append(self.word_to_ipa(m.group(0)))
+            pos = m.end()
+        tail = text[pos:]
+        parts.append("".join(c if c in ",;:" else " " for c in tail))
+        s = "".join(parts)
+        s = re.sub(r"[ \t]+", " ", s).strip()
+        return s
+
+
+# --------------------------------------------------------------------------- #
+# English (en-US)
+# --------------------------------------------------------------------------- #
+_EN_LEXICON = {
+    "a": "ə", "an": "ən", "the": "ðə", "of": "əv", "to": "tu", "and": "ænd",
+    "in": "ɪn", "is": "ɪz", "it": "ɪt", "you": "ju", "that": "ðæt",
+    "he": "hi", "she": "ʃi", "was": "wəz", "for": "fɔɹ", "on": "ɑn",
+    "are": "ɑɹ", "as": "æz", "with": "wɪð", "his": "hɪz", "her": "hɝ",
+    "they": "ðeɪ", "i": "aɪ", "at": "æt", "be": "bi", "this": "ðɪs",
+    "have": "hæv", "from": "fɹʌm", "or": "ɔɹ", "one": "wʌn", "had": "hæd",
+    "by": "baɪ", "word": "wɝd", "but": "bʌt", "not": "nɑt", "what": "wʌt",
+    "all": "ɔl", "were": "wɝ", "we": "wi", "when": "wɛn", "your": "jʊɹ",
+    "can": "kæn", "said": "sɛd", "there": "ðɛɹ", "use": "juz", "each": "itʃ",
+    "which": "wɪtʃ", "do": "du", "how": "haʊ", "their": "ðɛɹ", "if": "ɪf",
+    "will": "wɪl", "up": "ʌp", "other": "ʌðɚ", "about": "əbaʊt",
+    "out": "aʊt", "many": "mɛni", "then": "ðɛn", "them": "ðɛm",
+    "these": "ðiz", "so": "soʊ", "some": "sʌm", "would": "wʊd",
+    "make": "meɪk", "like": "laɪk", "him": "hɪm", "into": "ɪntu",
+    "time": "taɪm", "has": "hæz", "look": "lʊk", "two": "tu",
+    "more": "mɔɹ", "write": "ɹaɪt", "go": "ɡoʊ", "see": "si",
+    "number": "nʌmbɚ", "no": "noʊ", "way": "weɪ", "could": "kʊd",
+    "people": "pipəl", "my": "maɪ", "than": "ðæn", "first": "fɝst",
+    "water": "wɔtɚ", "been": "bɪn", "call": "kɔl", "who": "hu",
+    "its": "ɪts", "now": "naʊ", "find": "faɪnd", "long": "lɔŋ",
+    "down": "daʊn", "day": "deɪ", "did": "dɪd", "get": "ɡɛt",
+    "come": "kʌm", "made": "meɪd", "may": "meɪ", "part": "pɑɹt",
+    "over": "oʊvɚ", "new": "nu", "sound": "saʊnd", "take": "teɪk",
+    "only": "oʊnli", "little": "lɪtəl", "work": "wɝk", "know": "noʊ",
+    "place": "pleɪs", "year": "jɪɹ", "live": "lɪv", "me": "mi",
+    "back": "bæk", "give": "ɡɪv", "most": "moʊst", "very": "vɛɹi",
+    "after": "æftɚ", "thing": "θɪŋ", "our": "aʊɚ", "just": "dʒʌst",
+    "name": "neɪm", "good": "ɡʊd", "sentence": "sɛntəns", "man": "mæn",
+    "think": "θɪŋk", "say": "seɪ", "great": "ɡɹeɪt", "where": "wɛɹ",
+    "help": "hɛlp", "through": "θɹu", "much": "mʌtʃ", "before": "bɪfɔɹ",
+    "line": "laɪn", "right": "ɹaɪt", "too": "tu", "mean": "min",
+    "old": "oʊld", "any": "ɛni", "same": "seɪm", "tell": "tɛl",
+    "boy": "bɔɪ", "follow": "fɑloʊ", "came": "keɪm", "want": "wɑnt",
+    "show": "ʃoʊ", "also": "ɔlsoʊ", "around": "əɹaʊnd", "form": "fɔɹm",
+    "three": "θɹi", "small": "smɔl", "set": "sɛt", "put": "pʊt",
+    "end": "ɛnd", "does": "dʌz", "another": "ənʌðɚ", "well": "wɛl",
+    "large": "lɑɹdʒ", "must": "mʌst", "big": "bɪɡ", "even": "ivən",
+    "such": "sʌtʃ", "because": "bɪkɔz", "turn": "tɝn", "here": "hɪɹ",
+    "why": "waɪ", "ask": "æsk", "went": "wɛnt", "men": "mɛn",
+    "read": "ɹid", "need": "nid", "land": "lænd", "different": "dɪfɹənt",
+    "home": "hoʊm", "us": "ʌs", "move": "muv", "try": "tɹaɪ",
+    "kind": "kaɪnd", "hand": "hænd", "picture": "pɪktʃɚ", "again": "əɡɛn",
+    "change": "tʃeɪndʒ", "off": "ɔf", "play": "pleɪ", "spell": "spɛl",
+    "air": "ɛɹ", "away": "əweɪ", "animal": "ænəməl", "house": "haʊs",
+    "point": "pɔɪnt", "page": "peɪdʒ", "letter": "lɛtɚ", "mother": "mʌðɚ",
+    "answer": "ænsɚ", "found": "faʊnd", "study": "stʌdi", "still": "stɪl",
+    "learn": "lɝn", "should": "ʃʊd", "world": "wɝld", "high": "haɪ",
+    "every": "ɛvɹi", "near": "nɪɹ", "add": "æd", "food": "fud",
+    "between": "bɪtwin", "own": "oʊn", "below": "bɪloʊ", "country": "kʌntɹi",
+    "plant": "plænt", "last": "læst", "school": "skul", "father": "fɑðɚ",
+    "keep": "kip", "tree": "tɹi", "never": "nɛvɚ", "start": "stɑɹt",
+    "city": "sɪti", "earth": "ɝθ", "eye": "aɪ", "light": "laɪt",
+    "thought": "θɔt", "head": "hɛd", "under": "ʌndɚ", "story": "stɔɹi",
+    "saw": "sɔ", "left": "lɛft", "don't": "doʊnt", "few": "fju",
+    "while": "waɪl", "along": "əlɔŋ", "might": "maɪt", "close": "kloʊs",
+    "something": "sʌmθɪŋ", "seem": "sim", "next": "nɛkst", "hard": "hɑɹd",
+    "open": "oʊpən", "example": "ɪɡzæmpəl", "begin": "bɪɡɪn",
+    "life": "laɪf", "always": "ɔlweɪz", "those": "ðoʊz", "both": "boʊθ",
+    "paper": "peɪpɚ", "together": "təɡɛðɚ", "got": "ɡɑt", "group": "ɡɹup",
+    "often": "ɔfən", "run": "ɹʌn", "important": "ɪmpɔɹtənt",
+    "until": "ʌntɪl", "children": "tʃɪldɹən", "side": "saɪd",
+    "feet": "fit", "car": "kɑɹ", "mile": "maɪl", "night": "naɪt",
+    "walk": "wɔk", "white": "waɪt", "sea": "si", "began": "bɪɡæn",
+    "grow": "ɡɹoʊ", "took": "tʊk", "river": "ɹɪvɚ", "four": "fɔɹ",
+    "carry": "kæɹi", "state": "steɪt", "once": "wʌns", "book": "bʊk",
+    "hear": "hɪɹ", "stop": "stɑp", "without": "wɪðaʊt", "second": "sɛkənd",
+    "later": "leɪtɚ", "miss": "mɪs", "idea": "aɪdiə", "enough": "ɪnʌf",
+    "eat": "it", "face": "feɪs", "watch": "wɑtʃ", "far": "fɑɹ",
+    "really": "ɹɪli", "almost": "ɔlmoʊst", "let": "lɛt", "above": "əbʌv",
+    "girl": "ɡɝl", "sometimes": "sʌmtaɪmz", "mountain": "maʊntən",
+    "cut": "kʌt", "young": "jʌŋ", "talk": "tɔk", "soon": "sun",
+    "list": "lɪst", "song": "sɔŋ", "being": "biɪŋ", "leave": "liv",
+    "family": "fæməli", "hello": "hɛloʊ", "speech": "spitʃ",
+    "voice": "vɔɪs", "synthesis": "sɪnθəsɪs", "test": "tɛst",
+    "testing": "tɛstɪŋ", "quick": "kwɪk", "brown": "bɹaʊn",
+    "fox": "fɑks", "jumps": "dʒʌmps", "lazy": "leɪzi", "dog": "dɔɡ",
+    "today": "tədeɪ", "weather": "wɛðɚ", "nice": "naɪs",
+}
+
+# ordered longest-match English letter-to-sound rules
+_EN_RULES = {
+    "tion": "ʃən", "sion": "ʒən", "ought": "ɔt", "ight": "aɪt",
+    "tious": "ʃəs", "cious": "ʃəs", "ture": "tʃɚ", "sure": "ʒɚ",
+    "augh": "ɔ", "ough": "ʌf", "eigh": "eɪ",
+    "sch": "sk", "tch": "tʃ", "dge": "dʒ",
+    "ai": "eɪ", "ay": "eɪ", "ea": "i", "ee": "i", "ie": "i",
+    "oa": "oʊ", "oo": "u", "ou": "aʊ", "ow": "oʊ", "oy": "ɔɪ",
+    "oi": "ɔɪ", "au": "ɔ", "aw": "ɔ", "ew": "u", "ue": "u",
+    "ei": "eɪ", "ey": "eɪ", "ar": "ɑɹ", "er": "ɚ", "ir": "ɝ",
+    "or": "ɔɹ", "ur": "ɝ", "ck": "k", "ch": "tʃ", "sh": "ʃ",
+    "th": "θ", "ph": "f", "wh": "w", "ng": "ŋ", "qu": "kw",
+    "gh": "ɡ", "kn": "n", "wr": "ɹ", "mb": "m", "ce": "s",
+    "ci": "sɪ", "cy": "si", "ge": "dʒ", "gi": "dʒɪ", "gy": "dʒi",
+    "a": "æ", "b": "b", "c": "k", "d": "d", "e": "ɛ", "f": "f",
+    "g": "ɡ", "h": "h", "i": "ɪ", "j": "dʒ", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "ɑ", "p": "p", "r": "ɹ", "s": "s",
+    "t": "t", "u": "ʌ", "v": "v", "w": "w", "x": "ks", "y": "j",
+    "z": "z", "'": "",
+}
+
+# --------------------------------------------------------------------------- #
+# German
+# --------------------------------------------------------------------------- #
+_DE_RULES = {
+    "sch": "ʃ", "tsch": "tʃ", "chs": "ks", "ung": "ʊŋ",
+    "ei": "aɪ", "ai": "aɪ", "au": "aʊ", "eu": "ɔʏ", "äu": "ɔʏ",
+    "ie": "iː", "ch": "ç", "ck": "k", "sp": "ʃp", "st": "ʃt",
+    "th": "t", "ph": "f", "qu": "kv", "ß": "s",
+    "ä": "ɛ", "ö": "ø", "ü": "y",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "ɛ", "f": "f",
+    "g": "ɡ", "h": "h", "i": "ɪ", "j": "j", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "ɔ", "p": "p", "r": "ʁ", "s": "z",
+    "t": "t", "u": "ʊ", "v": "f", "w": "v", "x": "ks", "y": "y",
+    "z": "ts",
+}
+
+# --------------------------------------------------------------------------- #
+# Spanish
+# --------------------------------------------------------------------------- #
+_ES_RULES = {
+    "ch": "tʃ", "ll": "ʝ", "rr": "r", "qu": "k", "gu": "ɡ",
+    "ñ": "ɲ", "j": "x", "v": "b", "z": "θ", "ce": "θe", "ci": "θi",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "", "i": "i", "k": "k", "l": "l", "m": "m",
+    "n": "n", "o": "o", "p": "p", "r": "ɾ", "s": "s", "t": "t",
+    "u": "u", "w": "w", "x": "ks", "y": "ʝ",
+}
+
+# --------------------------------------------------------------------------- #
+# Arabic (expects diacritized input; tashkeel pre-pass adds vowel marks)
+# --------------------------------------------------------------------------- #
+_AR_RULES = {
+    # letters
+    "ا": "aː", "ب": "b", "ت": "t", "ث": "θ", "ج": "dʒ", "ح": "ħ",
+    "خ": "x", "د": "d", "ذ": "ð", "ر": "r", "ز": "z", "س": "s",
+    "ش": "ʃ", "ص": "sˤ", "ض": "dˤ", "ط": "tˤ", "ظ": "ðˤ", "ع": "ʕ",
+    "غ": "ɣ", "ف": "f", "ق": "q", "ك": "k", "ل": "l", "م": "m",
+    "ن": "n", "ه": "h", "و": "w", "ي": "j", "ء": "ʔ", "آ": "ʔaː",
+    "أ": "ʔ", "إ": "ʔɪ", "ؤ": "ʔ", "ئ": "ʔ", "ة": "h", "ى": "aː",
+    # diacritics (harakat)
+    "َ": "a",   # fatha
+    "ُ": "u",   # damma
+    "ِ": "ɪ",   # kasra
+    "ّ": "ː",   # shadda (gemination)
+    "ْ": "",    # sukun
+    "ً": "an",  # fathatan
+    "ٌ": "un",  # dammatan
+    "ٍ": "ɪn",  # kasratan
+}
+
+
+_G2P_REGISTRY: Dict[str, RuleG2P] = {}
+
+
+def _get_g2p(voice: str) -> RuleG2P:
+    key = voice.lower().replace("_", "-")
+    base = key.split("-")[0]
+    if key in _G2P_REGISTRY:
+        return _G2P_REGISTRY[key]
+    if base == "en":
+        g = RuleG2P(_EN_RULES, _EN_LEXICON, letters="a-zA-Z")
+    elif base == "de":
+        g = RuleG2P(_DE_RULES, letters="a-zA-Zäöüß")
+    elif base == "es":
+        g = RuleG2P(_ES_RULES, letters="a-zA-Zñáéíóúü")
+    elif base == "ar":
+        g = RuleG2P(
+            _AR_RULES,
+            letters="؀-ۿ",
+            stress=False,
+        )
+    else:
+        raise PhonemizationError(f"unsupported phonemizer language: {voice!r}")
+    _G2P_REGISTRY[key] = g
+    return g
+
+
+def available_languages() -> List[str]:
+    return ["en-us", "en", "de", "es", "ar"]
+
+
+_LANG_SWITCH_RE = re.compile(r"\([a-z-]{2,10}\)")
+_STRESS_RE = re.compile("[ˈˌ]")
+
+
+def text_to_phonemes(
+    text: str,
+    voice: str = "en-us",
+    separator: Optional[str] = None,
+    remove_lang_switch: bool = True,
+    remove_stress: bool = False,
+) -> List[str]:
+    """Phonemize `text` into per-sentence IPA strings.
+
+    Mirrors the reference API (espeak-phonemizer/src/lib.rs:65-83): input is
+    split per line, each line into sentences; each sentence's terminator
+    (./,/?/!) is preserved at its clause position; optional separator is
+    inserted between phonemes; stress marks optionally stripped."""
+    g2p = _get_g2p(voice)
+    out: List[str] = []
+    for line in text.splitlines() or [text]:
+        if not line.strip():
+            continue
+        for sent, term in split_sentences(line):
+            ipa = g2p.phonemize(sent)
+            if not ipa:
+                continue
+            ipa = ipa + term
+            if remove_lang_switch:
+                ipa = _LANG_SWITCH_RE.sub("", ipa)
+            if remove_stress:
+                ipa = _STRESS_RE.sub("", ipa)
+            if separator:
+                ipa = separator.join(ipa)
+            out.append(ipa)
+    return out

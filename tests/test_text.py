@@ -1,0 +1,128 @@
+"""Text front-end tests: phonemizer, sentence split, id encoding, tashkeel.
+
+Models the reference's espeak-phonemizer tests (src/lib.rs:160-252:
+en/ar phonemization, sentence splitting, separators, clause breakers,
+stress removal, line splitting).
+"""
+
+import pytest
+
+from sonata_amd.core import PhonemizationError
+from sonata_amd.text import (
+    BOS,
+    EOS,
+    PAD,
+    default_phoneme_id_map,
+    phonemes_to_ids,
+    split_sentences,
+    text_to_phonemes,
+)
+from sonata_amd.text.tashkeel import TashkeelModel
+
+
+def test_split_sentences_terminators():
+    s = split_sentences("Hello world. How are you? Fine!")
+    assert [t for _, t in s] == [".", "?", "!"]
+    assert s[0][0] == "Hello world"
+
+
+def test_split_sentences_clause_kept_inline():
+    s = split_sentences("One, two; three. Four.")
+    assert len(s) == 2
+    assert "," in s[0][0] and ";" in s[0][0]
+
+
+def test_split_sentences_no_terminator():
+    s = split_sentences("no punctuation here")
+    assert s == [("no punctuation here", ".")]
+
+
+def test_en_phonemize_basic():
+    out = text_to_phonemes("Hello world.", voice="en-us")
+    assert len(out) == 1
+    assert out[0].endswith(".")
+    assert "ɛ" in out[0] or "h" in out[0]
+
+
+def test_en_multiple_sentences():
+    out = text_to_phonemes("One. Two! Three?", voice="en")
+    assert len(out) == 3
+    assert out[0].endswith(".") and out[1].endswith("!") and out[2].endswith("?")
+
+
+def test_line_splitting():
+    out = text_to_phonemes("line one\nline two", voice="en-us")
+    assert len(out) == 2
+
+
+def test_stress_removal():
+    keep = text_to_phonemes("testing", voice="en-us")[0]
+    nostress = text_to_phonemes("testing", voice="en-us", remove_stress=True)[0]
+    assert "ˈ" in keep
+    assert "ˈ" not in nostress and "ˌ" not in nostress
+
+
+def test_separator():
+    out = text_to_phonemes("hi", voice="en-us", separator="|")[0]
+    assert "|" in out
+
+
+def test_clause_terminator_preserved_mid_sentence():
+    out = text_to_phonemes("One, two.", voice="en-us")[0]
+    assert "," in out and out.endswith(".")
+
+
+def test_german_and_spanish():
+    de = text_to_phonemes("Schön gut.", voice="de")[0]
+    assert "ʃ" in de
+    es = text_to_phonemes("mucho gusto.", voice="es")[0]
+    assert "tʃ" in es
+
+
+def test_arabic_phonemize():
+    out = text_to_phonemes("السلام عليكم.", voice="ar")
+    assert len(out) == 1
+    assert "s" in out[0] and "l" in out[0]
+
+
+def test_unknown_language_raises():
+    with pytest.raises(PhonemizationError):
+        text_to_phonemes("hi", voice="zz")
+
+
+def test_phoneme_id_map_stable():
+    m1 = default_phoneme_id_map()
+    m2 = default_phoneme_id_map()
+    assert m1 == m2
+    assert m1[PAD] == [0] and m1[BOS] == [1] and m1[EOS] == [2]
+
+
+def test_phonemes_to_ids_interleaves_pad():
+    m = default_phoneme_id_map()
+    ids = phonemes_to_ids("ab", m)
+    # BOS, a, PAD, b, PAD, EOS
+    assert ids[0] == 1 and ids[-1] == 2
+    assert ids[2] == 0 and ids[4] == 0
+    assert len(ids) == 6
+
+
+def test_phonemes_to_ids_skips_unknown():
+    m = default_phoneme_id_map()
+    ids = phonemes_to_ids("a☃b", m)  # snowman not in map
+    assert len(ids) == 6
+
+
+def test_tashkeel_inserts_diacritics():
+    model = TashkeelModel.default()
+    out = model.diacritize("سلام")
+    assert len(out) >= 4
+    # deterministic
+    assert out == model.diacritize("سلام")
+
+
+def test_tashkeel_save_load(tmp_path):
+    m = TashkeelModel.default()
+    p = str(tmp_path / "tashkeel.safetensors")
+    m.save(p)
+    m2 = TashkeelModel.load(p)
+    assert m.diacritize("سلام") == m2.diacritize("سلام")
