@@ -1,0 +1,376 @@
+// MFMA-tiled 1-D convolution family for the VITS graph (gfx950).
+//
+// The HiFi-GAN generator is ~all the FLOPs of Piper synthesis
+// (SURVEY.md §2.2): Conv1d (k=3/7/11, dilated) and ConvTranspose1d
+// (k=16/4, stride 8/2) plus many 1x1 projections.  All are expressed as
+// one MFMA GEMM kernel over "taps":
+//
+//   Out[co][n] = act( bias[co] + sum_tap sum_ci
+//                     W[tap][co][ci] * pre_act(x[ci][in_off(tap) + n]) )
+//
+//   Conv1d          : tap = kernel position j, in_off = n0 - pad + j*dil,
+//                     out index = n, ntaps = k.
+//   ConvTranspose1d : phase decomposition — for output phase r (t = s*v +
+//                     r - pad), tap m uses weight W[ci][co][r + s*m] and
+//                     input x[ci][v - m]; out index = s*v + r - pad.
+//                     ntaps = ceil((k - r)/s).  This turns the transposed
+//                     conv into s dense GEMMs with shifted input windows
+//                     (no zero-stuffing, no atomics).
+//
+// Tiling: block = 4 waves (256 thr), BM x BN output tile, K staged in
+// 32-deep slices of Cin through LDS.  A (weights) is read as contiguous
+// bf16x8 fragments (ds_read_b128); X is staged [ci][t] row-major with
+// +8 element row padding.  Weights are pre-permuted host-side (cached) to
+// [tap][Cout][Cin] so every global load is coalesced.
+//
+// mfma_f32_16x16x32_bf16 operand maps (verified against rocm CK headers,
+// ck_tile warp_gemm_attribute_mfma_impl.hpp: kABKLane=4, kABKPerLane=8):
+//   a_vec[q] = A[lane&15][(lane>>4)*8 + q]
+//   b_vec[q] = B[(lane>>4)*8 + q][lane&15]
+//   d[reg]   = D[(lane>>4)*4 + reg][lane&15]
+#include "common.h"
+
+#define BK 32
+
+// activation modes
+#define ACT_NONE 0
+#define ACT_LRELU 1
+#define ACT_TANH 2
+
+template <int BM, int BN, int WGM, int WGN>
+__global__ __launch_bounds__(256) void conv1d_mfma_kernel(
+    const bf16* __restrict__ x,     // [B][Cin][Tin]
+    const bf16* __restrict__ w,     // [ntaps][CoutP][CinP] pre-permuted
+    const float* __restrict__ bias, // [Cout] or null
+    bf16* __restrict__ out,         // [B][Cout][Tout]
+    int Cin, int CinP, int Cout, int CoutP, long Tin, long Tout,
+    int ntaps, int tap_in_stride,  // input offset step per tap (dil or -1)
+    int in_off0,                   // input offset of n=0, tap=0
+    long Nvirt,                    // GEMM N size (Tout or V)
+    int out_stride, int out_off,   // out t = n*out_stride + out_off
+    float pre_slope,               // <0: no pre-act; else LeakyReLU slope
+    int act_mode, float post_slope) {
+  // grid: (ceil(Nvirt/BN), ceil(Cout/BM) [, *s phases folded by caller],
+  //        B)
+  const int n_tile = blockIdx.x;
+  const int m_tile = blockIdx.y;
+  const int b = blockIdx.z;
+
+  const long n0 = (long)n_tile * BN;
+  const int m0 = m_tile * BM;
+
+  __shared__ bf16 Ws[BM][BK + 8];
+  __shared__ bf16 Xs[BK][BN + 8];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid / WGN;  // wave row (M)
+  const int wc = wid % WGN;  // wave col (N)
+  constexpr int WM = BM / WGM;  // per-wave M (32)
+  constexpr int WN = BN / WGN;  // per-wave N (32)
+  constexpr int MT = WM / 16;   // m fragments per wave
+  constexpr int NT = WN / 16;   // n fragments per wave
+
+  f32x4 acc[MT][NT];
+#pragma unroll
+  for (int i = 0; i < MT; ++i)
+#pragma unroll
+    for (int j = 0; j < NT; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const bf16* xb = x + (long)b * Cin * Tin;
+
+  for (int tap = 0; tap < ntaps; ++tap) {
+    const long w0 = n0 + in_off0 + (long)tap * tap_in_stride;
+    for (int c0 = 0; c0 < CinP; c0 += BK) {
+      // ---- stage W tile: Ws[m][kk] = w[tap][m0+m][c0+kk] -------------
+      // 256 threads fill BM*BK bf16; 8 contiguous per thread.
+      {
+        const long wbase = ((long)tap * CoutP + m0) * CinP + c0;
+        for (int e = tid * 8; e < BM * BK; e += 256 * 8) {
+          int m = e / BK, kk = e % BK;
+          *(ulonglong2*)&Ws[m][kk] =
+              *(const ulonglong2*)&w[wbase + (long)m * CinP + kk];
+        }
+      }
+      // ---- stage X tile: Xs[r][c] = pre(x[c0+r][w0+c]) ---------------
+      {
+        for (int e = tid * 8; e < BK * BN; e += 256 * 8) {
+          int r = e / BN, c = e % BN;
+          int ci = c0 + r;
+          bf16 vals[8];
+          long p = w0 + c;
+          if (ci < Cin && p >= 0 && p + 7 < Tin) {
+            *(ulonglong2*)vals = *(const ulonglong2*)&xb[(long)ci * Tin + p];
+            if (pre_slope >= 0.f) {
+#pragma unroll
+              for (int q = 0; q < 8; ++q)
+                vals[q] = f2bf(lrelu_(bf2f(vals[q]), pre_slope));
+            }
+          } else {
+#pragma unroll
+            for (int q = 0; q < 8; ++q) {
+              long pq = p + q;
+              float v = (ci < Cin && pq >= 0 && pq < Tin)
+                            ? bf2f(xb[(long)ci * Tin + pq])
+                            : 0.f;
+              if (pre_slope >= 0.f) v = lrelu_(v, pre_slope);
+              vals[q] = f2bf(v);
+            }
+          }
+          // c is a multiple of 8 and rows are 16B-aligned -> one b128 write
+          *(ulonglong2*)&Xs[r][c] = *(ulonglong2*)vals;
+        }
+      }
+      __syncthreads();
+
+      // ---- MFMA ------------------------------------------------------
+      const int kl = lane >> 4;      // k-lane group (0..3)
+      const int il = lane & 15;      // row/col within fragment
+#pragma unroll
+      for (int mi = 0; mi < MT; ++mi) {
+        bf16x8 a_frag =
+            *(const bf16x8*)&Ws[wr * WM + mi * 16 + il][kl * 8];
+#pragma unroll
+        for (int nj = 0; nj < NT; ++nj) {
+          bf16x8 b_frag;
+          const int ncol = wc * WN + nj * 16 + il;
+#pragma unroll
+          for (int q = 0; q < 8; ++q)
+            b_frag[q] = *(__bf16*)&Xs[kl * 8 + q][ncol];
+          acc[mi][nj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag, b_frag, acc[mi][nj], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // ---- epilogue: bias + activation + store --------------------------
+  bf16* ob = out + (long)b * Cout * Tout;
+  const int kl = lane >> 4;
+  const int il = lane & 15;
+#pragma unroll
+  for (int mi = 0; mi < MT; ++mi) {
+    const int row = m0 + wr * WM + mi * 16 + kl * 4;
+#pragma unroll
+    for (int nj = 0; nj < NT; ++nj) {
+      const long col = n0 + wc * WN + nj * 16 + il;
+      if (col >= Nvirt) continue;
+      const long t = col * out_stride + out_off;
+      if (t < 0 || t >= Tout) continue;
+#pragma unroll
+      for (int rg = 0; rg < 4; ++rg) {
+        const int co = row + rg;
+        if (co >= Cout) continue;
+        float v = acc[mi][nj][rg];
+        if (bias) v += bias[co];
+        if (act_mode == ACT_LRELU) v = lrelu_(v, post_slope);
+        else if (act_mode == ACT_TANH) v = tanhf(v);
+        ob[(long)co * Tout + t] = f2bf(v);
+      }
+    }
+  }
+}
+
+// --------------------------------------------------------------------------
+// Naive direct conv (grouped / f32 / stride>1 fallback + oracle).
+// One thread per (b, co, t_out).
+// --------------------------------------------------------------------------
+template <typename T>
+__global__ void conv1d_naive_kernel(
+    const T* __restrict__ x, const T* __restrict__ w,
+    const float* __restrict__ bias, T* __restrict__ out, int Cin, int Cout,
+    long Tin, long Tout, int k, int stride, int pad, int dil, int groups,
+    float pre_slope, int act_mode, float post_slope, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const long t = i % Tout;
+  const int co = (i / Tout) % Cout;
+  const long b = i / ((long)Tout * Cout);
+  const int cpg_in = Cin / groups;
+  const int cpg_out = Cout / groups;
+  const int g = co / cpg_out;
+  const T* xb = x + (b * Cin + (long)g * cpg_in) * Tin;
+  const T* wc = w + (long)co * cpg_in * k;
+  float accv = bias ? bias[co] : 0.f;
+  for (int ci = 0; ci < cpg_in; ++ci) {
+    for (int j = 0; j < k; ++j) {
+      long p = t * stride + (long)j * dil - pad;
+      if (p < 0 || p >= Tin) continue;
+      float xv = ld_f(xb + (long)ci * Tin + p);
+      if (pre_slope >= 0.f) xv = lrelu_(xv, pre_slope);
+      accv += xv * ld_f(wc + ci * k + j);
+    }
+  }
+  if (act_mode == ACT_LRELU) accv = lrelu_(accv, post_slope);
+  else if (act_mode == ACT_TANH) accv = tanhf(accv);
+  st_f(out + i, accv);
+}
+
+// naive transposed conv fallback: thread per (b, co, t_out)
+template <typename T>
+__global__ void convt1d_naive_kernel(
+    const T* __restrict__ x, const T* __restrict__ w,  // [Cin][Cout][k]
+    const float* __restrict__ bias, T* __restrict__ out, int Cin, int Cout,
+    long Tin, long Tout, int k, int stride, int pad, float pre_slope,
+    long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const long t = i % Tout;
+  const int co = (i / Tout) % Cout;
+  const long b = i / ((long)Tout * Cout);
+  const T* xb = x + b * (long)Cin * Tin;
+  float accv = bias ? bias[co] : 0.f;
+  for (int j = 0; j < k; ++j) {
+    long num = t + pad - j;
+    if (num < 0 || num % stride) continue;
+    long u = num / stride;
+    if (u >= Tin) continue;
+    for (int ci = 0; ci < Cin; ++ci) {
+      float xv = ld_f(xb + (long)ci * Tin + u);
+      if (pre_slope >= 0.f) xv = lrelu_(xv, pre_slope);
+      accv += xv * ld_f(w + ((long)ci * Cout + co) * k + j);
+    }
+  }
+  st_f(out + i, accv);
+}
+
+// ========================================================================
+// host wrappers
+// ========================================================================
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+static inline hipStream_t cur_stream2() {
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+static inline long conv_out_len(long Tin, int k, int stride, int pad,
+                                int dil) {
+  return (Tin + 2L * pad - (long)dil * (k - 1) - 1) / stride + 1;
+}
+
+// Launch helper: pick tile config by Cout and launch the MFMA kernel.
+static void launch_conv_mfma(const bf16* x, const bf16* w, const float* bias,
+                             bf16* out, int B, int Cin, int CinP, int Cout,
+                             int CoutP, long Tin, long Tout, int ntaps,
+                             int tap_in_stride, int in_off0, long Nvirt,
+                             int out_stride, int out_off, float pre_slope,
+                             int act_mode, float post_slope,
+                             hipStream_t stream) {
+#define LAUNCH(BM, BN, WGM, WGN)                                            \
+  do {                                                                      \
+    dim3 grid(ceil_div(Nvirt, BN), ceil_div(Cout, BM), B);                  \
+    hipLaunchKernelGGL((conv1d_mfma_kernel<BM, BN, WGM, WGN>), grid,        \
+                       dim3(256), 0, stream, x, w, bias, out, Cin, CinP,    \
+                       Cout, CoutP, Tin, Tout, ntaps, tap_in_stride,        \
+                       in_off0, Nvirt, out_stride, out_off, pre_slope,      \
+                       act_mode, post_slope);                               \
+  } while (0)
+  if (Cout >= 64) LAUNCH(64, 64, 2, 2);
+  else LAUNCH(32, 128, 1, 4);
+#undef LAUNCH
+}
+
+torch::Tensor conv1d_fused(torch::Tensor x, torch::Tensor w_perm,
+                           c10::optional<torch::Tensor> bias, long Cout,
+                           long k, long stride, long padding, long dilation,
+                           long groups, double pre_lrelu, long act_mode,
+                           double post_slope) {
+  // w_perm: [k][CoutP][CinP] bf16 (pre-permuted+padded) for the MFMA path
+  //         or the raw [Cout][Cin/g][k] tensor for the fallback path.
+  TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
+  const long B = x.size(0), Cin = x.size(1), Tin = x.size(2);
+  const long Tout = conv_out_len(Tin, k, stride, padding, dilation);
+  auto out = torch::empty({B, Cout, Tout}, x.options());
+  if (out.numel() == 0) return out;
+  torch::Tensor bias_f;
+  const float* bias_p = nullptr;
+  if (bias.has_value()) {
+    bias_f = bias->to(at::kFloat).contiguous();
+    bias_p = bias_f.data_ptr<float>();
+  }
+  const bool mfma_ok = x.scalar_type() == at::kBFloat16 && groups == 1 &&
+                       stride == 1 && w_perm.dim() == 3 &&
+                       w_perm.size(0) == k;
+  if (mfma_ok) {
+    const int CoutP = w_perm.size(1), CinP = w_perm.size(2);
+    launch_conv_mfma((const bf16*)x.data_ptr(),
+                     (const bf16*)w_perm.data_ptr(), bias_p,
+                     (bf16*)out.data_ptr(), B, Cin, CinP, Cout, CoutP, Tin,
+                     Tout, k, dilation, -(int)padding, Tout, 1, 0,
+                     (float)pre_lrelu, (int)act_mode, (float)post_slope,
+                     cur_stream2());
+  } else {
+    // fallback: w_perm is the raw [Cout][Cin/g][k] weight
+    const long n = B * Cout * Tout;
+    const int threads = 256;
+    DISPATCH_FT_CONV(x, {
+      hipLaunchKernelGGL(conv1d_naive_kernel<scalar_t>,
+                         dim3((n + threads - 1) / threads), dim3(threads), 0,
+                         cur_stream2(), (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)w_perm.data_ptr(), bias_p,
+                         (scalar_t*)out.data_ptr(), Cin, Cout, Tin, Tout, k,
+                         stride, padding, dilation, groups, (float)pre_lrelu,
+                         (int)act_mode, (float)post_slope, n);
+    });
+  }
+  return out;
+}
+
+torch::Tensor convtranspose1d_fused(torch::Tensor x, torch::Tensor w_perm,
+                                    c10::optional<torch::Tensor> bias,
+                                    long Cout, long k, long stride,
+                                    long padding, double pre_lrelu) {
+  // w_perm (MFMA): [s][kr_max][CoutP][CinP]; fallback: raw [Cin][Cout][k].
+  TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
+  const long B = x.size(0), Cin = x.size(1), Tin = x.size(2);
+  const long Tout = (Tin - 1) * stride - 2 * padding + k;
+  auto out = torch::empty({B, Cout, Tout}, x.options());
+  if (out.numel() == 0) return out;
+  torch::Tensor bias_f;
+  const float* bias_p = nullptr;
+  if (bias.has_value()) {
+    bias_f = bias->to(at::kFloat).contiguous();
+    bias_p = bias_f.data_ptr<float>();
+  }
+  const bool mfma_ok =
+      x.scalar_type() == at::kBFloat16 && w_perm.dim() == 4;
+  if (mfma_ok) {
+    const int kr_max = w_perm.size(1);
+    const int CoutP = w_perm.size(2), CinP = w_perm.size(3);
+    // one GEMM per phase r: out[s*v + r - pad] over v in [0, Vmax)
+    for (int r = 0; r < stride; ++r) {
+      const int ntaps = (k - r + stride - 1) / stride;
+      // v range for valid t: 0 <= s*v + r - pad < Tout
+      // v_lo = ceil((pad - r)/s); fold into out_off by shifting v start.
+      long v_lo = (padding - r + stride - 1) / stride;
+      if (v_lo < 0) v_lo = 0;
+      long v_hi = (Tout - 1 + padding - r) / stride;  // inclusive
+      if (v_hi >= Tin + kr_max) v_hi = Tin + kr_max;
+      long Nvirt = v_hi - v_lo + 1;
+      if (Nvirt <= 0) continue;
+      const bf16* wr =
+          (const bf16*)w_perm.data_ptr() + (long)r * kr_max * CoutP * CinP;
+      // tap m reads x[v - m] -> in_off0 = v_lo, tap stride -1
+      launch_conv_mfma((const bf16*)x.data_ptr(), wr, bias_p,
+                       (bf16*)out.data_ptr(), B, Cin, CinP, Cout, CoutP,
+                       Tin, Tout, ntaps, -1, (int)v_lo, Nvirt, stride,
+                       (int)(v_lo * stride + r - padding), (float)pre_lrelu,
+                       ACT_NONE, 0.f, cur_stream2());
+    }
+  } else {
+    const long n = B * Cout * Tout;
+    const int threads = 256;
+    DISPATCH_FT_CONV(x, {
+      hipLaunchKernelGGL(convt1d_naive_kernel<scalar_t>,
+                         dim3((n + threads - 1) / threads), dim3(threads), 0,
+                         cur_stream2(), (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)w_perm.data_ptr(), bias_p,
+                         (scalar_t*)out.data_ptr(), Cin, Cout, Tin, Tout, k,
+                         stride, padding, (float)pre_lrelu, n);
+    });
+  }
+  return out;
+}

@@ -1,0 +1,229 @@
+// Fused elementwise / normalization kernels for the VITS graph.
+//
+// Ops (SURVEY.md §2.2 kernel inventory):
+//   - layer_norm_ct: LayerNorm across channels of [B, C, T]
+//   - fused_gate:    WaveNet tanh(a+ga)·sigmoid(b+gb) gate
+//   - prior_sample:  z = (m + eps·exp(logs)·noise_scale)·mask
+//   - expand_states: duration length-regulator gather
+//
+// All are HBM-bandwidth-bound: one coalesced read per input element, one
+// write per output, activations fused so no intermediate tensors hit HBM.
+#include "common.h"
+
+// --------------------------------------------------------------------------
+// layer_norm_ct: x [B, C, T] -> per-(b,t) normalize across C.
+// Thread t-major: lane i handles time position t0+i so every c-iteration
+// reads a contiguous [blockDim.x] segment (fully coalesced along T).
+// --------------------------------------------------------------------------
+template <typename T>
+__global__ void layer_norm_ct_kernel(const T* __restrict__ x,
+                                     const float* __restrict__ gamma,
+                                     const float* __restrict__ beta,
+                                     T* __restrict__ out, int C, long T_len,
+                                     float eps, long n_bt) {
+  const long bt = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (bt >= n_bt) return;
+  const long b = bt / T_len;
+  const long t = bt % T_len;
+  const T* xp = x + (b * C) * T_len + t;
+  float mean = 0.f;
+  for (int c = 0; c < C; ++c) mean += ld_f(xp + (long)c * T_len);
+  mean /= C;
+  float var = 0.f;
+  for (int c = 0; c < C; ++c) {
+    float d = ld_f(xp + (long)c * T_len) - mean;
+    var += d * d;
+  }
+  float rstd = rsqrtf(var / C + eps);
+  T* op = out + (b * C) * T_len + t;
+  for (int c = 0; c < C; ++c) {
+    float v = (ld_f(xp + (long)c * T_len) - mean) * rstd;
+    st_f(op + (long)c * T_len, v * gamma[c] + beta[c]);
+  }
+}
+
+// --------------------------------------------------------------------------
+// fused_gate: x [B, 2C, T] (+ optional g) -> tanh·sigmoid gate [B, C, T]
+// --------------------------------------------------------------------------
+template <typename T, bool HAS_G>
+__global__ void fused_gate_kernel(const T* __restrict__ x,
+                                  const T* __restrict__ g,
+                                  T* __restrict__ out, long C_T, long CT2,
+                                  long n) {
+  // n = B*C*T ; C_T = C*T (half-channel offset); CT2 = 2*C*T (batch stride)
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  long b = i / C_T;
+  long r = i % C_T;
+  long ia = b * CT2 + r;
+  long ib = ia + C_T;
+  float va = ld_f(x + ia), vb = ld_f(x + ib);
+  if (HAS_G) {
+    va += ld_f(g + ia);
+    vb += ld_f(g + ib);
+  }
+  st_f(out + i, tanhf(va) * sigmoidf_(vb));
+}
+
+// --------------------------------------------------------------------------
+// prior_sample
+// --------------------------------------------------------------------------
+template <typename T>
+__global__ void prior_sample_kernel(const T* __restrict__ m,
+                                    const T* __restrict__ logs,
+                                    const T* __restrict__ mask,  // [B,1,T]
+                                    const T* __restrict__ noise,
+                                    T* __restrict__ out, long C_T, long T_len,
+                                    float ns, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  long b = i / C_T;
+  long t = i % T_len;
+  float msk = ld_f(mask + b * T_len + t);
+  float v = ld_f(m + i) + ld_f(noise + i) * __expf(ld_f(logs + i)) * ns;
+  st_f(out + i, v * msk);
+}
+
+// --------------------------------------------------------------------------
+// expand_states: stats [B, C, T] + durations [B, T] -> out [B, C, F]
+// Per block: batch row b.  Thread 0 builds the cumulative-duration ->
+// phoneme-index table in LDS (T is a few hundred), then all threads gather
+// coalesced along F.
+// --------------------------------------------------------------------------
+template <typename T>
+__global__ void expand_states_kernel(const T* __restrict__ stats,
+                                     const int* __restrict__ durs,
+                                     T* __restrict__ out, int C, int T_ph,
+                                     int F_max) {
+  extern __shared__ int idx_lds[];  // [F_max] phoneme index per frame
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  // serial cumsum + fill (T_ph and F_max are small; ~µs)
+  if (tid == 0) {
+    int f = 0;
+    for (int p = 0; p < T_ph && f < F_max; ++p) {
+      int d = durs[b * T_ph + p];
+      for (int k = 0; k < d && f < F_max; ++k) idx_lds[f++] = p;
+    }
+    // pad remaining frames with last phoneme (masked out later anyway)
+    int last = T_ph - 1;
+    while (f < F_max) idx_lds[f++] = last;
+  }
+  __syncthreads();
+  const T* sp = stats + (long)b * C * T_ph;
+  T* op = out + (long)b * C * F_max;
+  for (long cf = tid; cf < (long)C * F_max; cf += blockDim.x) {
+    int c = cf / F_max;
+    int f = cf % F_max;
+    op[cf] = sp[(long)c * T_ph + idx_lds[f]];
+  }
+}
+
+// ========================================================================
+// host wrappers
+// ========================================================================
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+static inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+#define DISPATCH_FT(TENSOR, NAME, ...)                                     \
+  do {                                                                     \
+    if ((TENSOR).scalar_type() == at::kFloat) {                            \
+      using scalar_t = float;                                              \
+      __VA_ARGS__;                                                         \
+    } else if ((TENSOR).scalar_type() == at::kBFloat16) {                  \
+      using scalar_t = bf16;                                               \
+      __VA_ARGS__;                                                         \
+    } else {                                                               \
+      TORCH_CHECK(false, NAME ": unsupported dtype");                      \
+    }                                                                      \
+  } while (0)
+
+torch::Tensor layer_norm_ct(torch::Tensor x, torch::Tensor gamma,
+                            torch::Tensor beta, double eps) {
+  TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
+  const long B = x.size(0), C = x.size(1), T = x.size(2);
+  auto out = torch::empty_like(x);
+  auto gamma_f = gamma.to(at::kFloat).contiguous();
+  auto beta_f = beta.to(at::kFloat).contiguous();
+  const int threads = 256;
+  dim3 grid(ceil_div(B * T, threads));
+  DISPATCH_FT(x, "layer_norm_ct", {
+    hipLaunchKernelGGL(layer_norm_ct_kernel<scalar_t>, grid, dim3(threads), 0,
+                       cur_stream(), (const scalar_t*)x.data_ptr(),
+                       gamma_f.data_ptr<float>(), beta_f.data_ptr<float>(),
+                       (scalar_t*)out.data_ptr(), (int)C, T, (float)eps,
+                       B * T);
+  });
+  return out;
+}
+
+torch::Tensor fused_gate(torch::Tensor x, c10::optional<torch::Tensor> g,
+                         long n_channels) {
+  TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
+  const long B = x.size(0), C2 = x.size(1), T = x.size(2);
+  TORCH_CHECK(C2 == 2 * n_channels, "fused_gate: channel mismatch");
+  auto out = torch::empty({B, n_channels, T}, x.options());
+  const long n = B * n_channels * T;
+  const int threads = 256;
+  const long blocks = (n + threads - 1) / threads;
+  DISPATCH_FT(x, "fused_gate", {
+    if (g.has_value()) {
+      TORCH_CHECK(g->sizes() == x.sizes());
+      hipLaunchKernelGGL((fused_gate_kernel<scalar_t, true>), dim3(blocks),
+                         dim3(threads), 0, cur_stream(),
+                         (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)g->data_ptr(),
+                         (scalar_t*)out.data_ptr(), n_channels * T,
+                         2 * n_channels * T, n);
+    } else {
+      hipLaunchKernelGGL((fused_gate_kernel<scalar_t, false>), dim3(blocks),
+                         dim3(threads), 0, cur_stream(),
+                         (const scalar_t*)x.data_ptr(), (const scalar_t*)nullptr,
+                         (scalar_t*)out.data_ptr(), n_channels * T,
+                         2 * n_channels * T, n);
+    }
+  });
+  return out;
+}
+
+torch::Tensor prior_sample(torch::Tensor m, torch::Tensor logs,
+                           torch::Tensor mask, torch::Tensor noise,
+                           double noise_scale) {
+  TORCH_CHECK(m.is_cuda() && m.is_contiguous() && m.dim() == 3);
+  const long B = m.size(0), C = m.size(1), T = m.size(2);
+  auto out = torch::empty_like(m);
+  const long n = B * C * T;
+  const int threads = 256;
+  DISPATCH_FT(m, "prior_sample", {
+    hipLaunchKernelGGL(prior_sample_kernel<scalar_t>,
+                       dim3((n + threads - 1) / threads), dim3(threads), 0,
+                       cur_stream(), (const scalar_t*)m.data_ptr(),
+                       (const scalar_t*)logs.data_ptr(),
+                       (const scalar_t*)mask.data_ptr(),
+                       (const scalar_t*)noise.data_ptr(),
+                       (scalar_t*)out.data_ptr(), C * T, T,
+                       (float)noise_scale, n);
+  });
+  return out;
+}
+
+torch::Tensor expand_states(torch::Tensor stats, torch::Tensor durs,
+                            long F_max) {
+  TORCH_CHECK(stats.dim() == 3 && stats.is_cuda() && stats.is_contiguous());
+  TORCH_CHECK(durs.scalar_type() == at::kInt && durs.is_contiguous());
+  const long B = stats.size(0), C = stats.size(1), T = stats.size(2);
+  auto out = torch::empty({B, C, F_max}, stats.options());
+  size_t lds = F_max * sizeof(int);
+  TORCH_CHECK(lds <= 160 * 1024, "expand_states: F too large for LDS");
+  DISPATCH_FT(stats, "expand_states", {
+    hipLaunchKernelGGL(expand_states_kernel<scalar_t>, dim3(B), dim3(256),
+                       lds, cur_stream(), (const scalar_t*)stats.data_ptr(),
+                       durs.data_ptr<int>(), (scalar_t*)out.data_ptr(),
+                       (int)C, (int)T, (int)F_max);
+  });
+  return out;
+}

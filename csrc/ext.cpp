@@ -1,0 +1,34 @@
+// Python bindings for the sonata_amd CDNA4 kernel library.
+#include <torch/extension.h>
+
+// elementwise.hip
+torch::Tensor layer_norm_ct(torch::Tensor x, torch::Tensor gamma,
+                            torch::Tensor beta, double eps);
+torch::Tensor fused_gate(torch::Tensor x, c10::optional<torch::Tensor> g,
+                         long n_channels);
+torch::Tensor prior_sample(torch::Tensor m, torch::Tensor logs,
+                           torch::Tensor mask, torch::Tensor noise,
+                           double noise_scale);
+torch::Tensor expand_states(torch::Tensor stats, torch::Tensor durs,
+                            long F_max);
+// conv1d.hip
+torch::Tensor conv1d_fused(torch::Tensor x, torch::Tensor w_perm,
+                           c10::optional<torch::Tensor> bias, long Cout,
+                           long k, long stride, long padding, long dilation,
+                           long groups, double pre_lrelu, long act_mode,
+                           double post_slope);
+torch::Tensor convtranspose1d_fused(torch::Tensor x, torch::Tensor w_perm,
+                                    c10::optional<torch::Tensor> bias,
+                                    long Cout, long k, long stride,
+                                    long padding, double pre_lrelu);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "sonata_amd hand-written CDNA4 (gfx950) kernels";
+  m.def("layer_norm_ct", &layer_norm_ct, "LayerNorm over channels of [B,C,T]");
+  m.def("fused_gate", &fused_gate, "WaveNet tanh*sigmoid gate");
+  m.def("prior_sample", &prior_sample, "z=(m+eps*exp(logs)*ns)*mask");
+  m.def("expand_states", &expand_states, "duration length-regulator gather");
+  m.def("conv1d_fused", &conv1d_fused, "MFMA conv1d with fused activations");
+  m.def("convtranspose1d_fused", &convtranspose1d_fused,
+        "MFMA transposed conv1d (phase-decomposed GEMMs)");
+}

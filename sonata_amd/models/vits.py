@@ -21,6 +21,7 @@ import torch.nn.functional as F
 from torch import nn
 
 from ..ops import (
+    conv_mod,
     expand_states,
     fused_gate,
     layer_norm_ct,
@@ -109,9 +110,9 @@ class RelativeAttention(nn.Module):
 
     def forward(self, x: torch.Tensor, attn_mask: torch.Tensor) -> torch.Tensor:
         B, C, T = x.shape
-        q = self.conv_q(x).view(B, self.n_heads, self.head_dim, T).transpose(2, 3)
-        k = self.conv_k(x).view(B, self.n_heads, self.head_dim, T).transpose(2, 3)
-        v = self.conv_v(x).view(B, self.n_heads, self.head_dim, T).transpose(2, 3)
+        q = conv_mod(self.conv_q, x).view(B, self.n_heads, self.head_dim, T).transpose(2, 3)
+        k = conv_mod(self.conv_k, x).view(B, self.n_heads, self.head_dim, T).transpose(2, 3)
+        v = conv_mod(self.conv_v, x).view(B, self.n_heads, self.head_dim, T).transpose(2, 3)
         scale = self.head_dim ** -0.5
         scores = torch.matmul(q * scale, k.transpose(-2, -1))  # [B,h,T,T]
         rel_k = self._rel_embeddings(self.emb_rel_k, T)  # [1, 2T-1, d]
@@ -124,7 +125,7 @@ class RelativeAttention(nn.Module):
         rel_v = self._rel_embeddings(self.emb_rel_v, T)
         out = out + torch.matmul(rel_w, rel_v.unsqueeze(0))
         out = out.transpose(2, 3).contiguous().view(B, C, T)
-        return self.conv_o(out)
+        return conv_mod(self.conv_o, out)
 
 
 class FFN(nn.Module):
@@ -138,9 +139,9 @@ class FFN(nn.Module):
                                padding=kernel_size // 2)
 
     def forward(self, x: torch.Tensor, x_mask: torch.Tensor) -> torch.Tensor:
-        x = self.conv1(x * x_mask)
+        x = conv_mod(self.conv1, x * x_mask)
         x = torch.relu(x)
-        x = self.conv2(x * x_mask)
+        x = conv_mod(self.conv2, x * x_mask)
         return x * x_mask
 
 
@@ -177,7 +178,7 @@ class TextEncoder(nn.Module):
             x = n1(x + y)
             y = ffn(x, x_mask)
             x = n2(x + y)
-        stats = self.proj(x) * x_mask
+        stats = conv_mod(self.proj, x) * x_mask
         m, logs = stats.chunk(2, dim=1)
         return x, m, logs, x_mask
 
@@ -211,17 +212,17 @@ class WN(nn.Module):
                 g: Optional[torch.Tensor] = None) -> torch.Tensor:
         output = torch.zeros_like(x)
         if g is not None and self.cond_layer is not None:
-            g_all = self.cond_layer(g)
+            g_all = conv_mod(self.cond_layer, g)
         else:
             g_all = None
         for i in range(self.n_layers):
-            x_in = self.in_layers[i](x)
+            x_in = conv_mod(self.in_layers[i], x)
             g_l = (
                 g_all[:, i * 2 * self.hidden : (i + 1) * 2 * self.hidden]
                 if g_all is not None else None
             )
             acts = fused_gate(x_in, g_l, self.hidden)
-            res_skip = self.res_skip_layers[i](acts)
+            res_skip = conv_mod(self.res_skip_layers[i], acts)
             if i < self.n_layers - 1:
                 x = (x + res_skip[:, : self.hidden]) * x_mask
                 output = output + res_skip[:, self.hidden :]
@@ -247,9 +248,9 @@ class ResidualCouplingLayer(nn.Module):
                 g: Optional[torch.Tensor] = None,
                 reverse: bool = False) -> torch.Tensor:
         x0, x1 = x[:, : self.half], x[:, self.half :]
-        h = self.pre(x0) * x_mask
+        h = conv_mod(self.pre, x0) * x_mask
         h = self.enc(h, x_mask, g=g)
-        m = self.post(h) * x_mask
+        m = conv_mod(self.post, h) * x_mask
         if not reverse:
             x1 = (m + x1) * x_mask
         else:
@@ -419,10 +420,10 @@ class DDSConv(nn.Module):
             x = x + g
         for sep, one, n1, n2 in zip(self.convs_sep, self.convs_1x1,
                                     self.norms_1, self.norms_2):
-            y = sep(x * x_mask)
+            y = conv_mod(sep, x * x_mask)
             y = n1(y)
             y = F.gelu(y)
-            y = one(y)
+            y = conv_mod(one, y)
             y = n2(y)
             y = F.gelu(y)
             x = x + y
@@ -460,9 +461,9 @@ class ConvFlow(nn.Module):
 
     def forward(self, x, x_mask, g=None, reverse=False):
         x0, x1 = x[:, : self.half], x[:, self.half :]
-        h = self.pre(x0)
+        h = conv_mod(self.pre, x0)
         h = self.convs(h, x_mask, g=g)
-        h = self.proj(h) * x_mask
+        h = conv_mod(self.proj, h) * x_mask
         B, _, T = x0.shape
         h = h.reshape(B, self.half, 3 * self.num_bins - 1, T).permute(0, 1, 3, 2)
         scale = math.sqrt(self.filter_channels)
@@ -510,11 +511,11 @@ class StochasticDurationPredictor(nn.Module):
         """Reverse pass: sample log-durations. x: [B, H, T] text states.
         `noise` is [B, 2, T] standard-normal (masked per utterance so that
         batch composition cannot change an utterance's durations)."""
-        x = self.pre(x.detach())
+        x = conv_mod(self.pre, x.detach())
         if g is not None and self.cond is not None:
-            x = x + self.cond(g.detach())
+            x = x + conv_mod(self.cond, g.detach())
         x = self.convs(x, x_mask)
-        x = self.proj(x) * x_mask
+        x = conv_mod(self.proj, x) * x_mask
         flows = list(reversed(self.flows))
         flows = flows[:-2] + [flows[-1]]  # drop the final (unused) Flip pair
         if noise is None:
@@ -591,9 +592,9 @@ class Generator(nn.Module):
 
     def forward(self, x: torch.Tensor,
                 g: Optional[torch.Tensor] = None) -> torch.Tensor:
-        x = self.conv_pre(x)
+        x = conv_mod(self.conv_pre, x)
         if g is not None and self.cond is not None:
-            x = x + self.cond(g)
+            x = x + conv_mod(self.cond, g)
         for i, up in enumerate(self.ups):
             x = leaky_convtranspose1d(
                 x, up.weight, up.bias, stride=up.stride[0],

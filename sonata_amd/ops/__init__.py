@@ -62,4 +62,5 @@ from .functional import (  # noqa: F401,E402
     expand_states,
     leaky_conv1d,
     leaky_convtranspose1d,
+    conv_mod,
 )

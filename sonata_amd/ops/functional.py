@@ -110,6 +110,50 @@ def expand_states(
 # --------------------------------------------------------------------------- #
 # Conv1d (+ fused LeakyReLU on input or output) — the HiFi-GAN hot op
 # --------------------------------------------------------------------------- #
+def _round_up(v: int, m: int) -> int:
+    return (v + m - 1) // m * m
+
+
+def _conv_weight_mfma(weight: torch.Tensor) -> torch.Tensor:
+    """Cache the MFMA layout of a conv weight: [Cout,Cin,k] ->
+    [k, CoutP, CinP] bf16, Cout padded to the tile height, Cin to 32.
+    Weights are static at inference; the permuted copy lives on the
+    parameter object."""
+    cached = getattr(weight, "_sonata_perm", None)
+    if cached is not None:
+        return cached
+    Cout, Cin, k = weight.shape
+    bm = 64 if Cout >= 64 else 32
+    CoutP, CinP = _round_up(Cout, bm), _round_up(Cin, 32)
+    perm = torch.zeros((k, CoutP, CinP), dtype=torch.bfloat16,
+                       device=weight.device)
+    perm[:, :Cout, :Cin] = weight.detach().permute(2, 0, 1).to(torch.bfloat16)
+    perm = perm.contiguous()
+    weight._sonata_perm = perm
+    return perm
+
+
+def _convt_weight_mfma(weight: torch.Tensor, stride: int) -> torch.Tensor:
+    """ConvTranspose1d weight [Cin,Cout,k] -> phase layout
+    [s, kr_max, CoutP, CinP] bf16 where phase r tap m holds W[:, :, r+s*m]."""
+    cached = getattr(weight, "_sonata_perm_t", None)
+    if cached is not None:
+        return cached
+    Cin, Cout, k = weight.shape
+    kr_max = (k + stride - 1) // stride
+    bm = 64 if Cout >= 64 else 32
+    CoutP, CinP = _round_up(Cout, bm), _round_up(Cin, 32)
+    perm = torch.zeros((stride, kr_max, CoutP, CinP), dtype=torch.bfloat16,
+                       device=weight.device)
+    w = weight.detach().to(torch.bfloat16)
+    for r in range(stride):
+        for m in range((k - r + stride - 1) // stride):
+            perm[r, m, :Cout, :Cin] = w[:, :, r + stride * m].t()
+    perm = perm.contiguous()
+    weight._sonata_perm_t = perm
+    return perm
+
+
 def leaky_conv1d(
     x: torch.Tensor,
     weight: torch.Tensor,
@@ -117,6 +161,7 @@ def leaky_conv1d(
     stride: int = 1,
     padding: int = 0,
     dilation: int = 1,
+    groups: int = 1,
     pre_lrelu: float = 0.0,
     post_lrelu: float = 0.0,
 ) -> torch.Tensor:
@@ -126,15 +171,22 @@ def leaky_conv1d(
     (HiFi-GAN MRF pattern: y = conv(lrelu(x)))."""
     if use_hip(x):
         ext = hip_ext(required=True)
+        Cout, _, k = weight.shape
+        mfma = (
+            x.dtype == torch.bfloat16 and groups == 1 and stride == 1
+        )
+        w = _conv_weight_mfma(weight) if mfma else weight.contiguous()
         return ext.conv1d_fused(
-            x.contiguous(), weight.contiguous(),
+            x.contiguous(), w,
             bias.contiguous() if bias is not None else None,
-            stride, padding, dilation, float(pre_lrelu), float(post_lrelu),
+            Cout, k, stride, padding, dilation, groups,
+            pre_lrelu if pre_lrelu > 0.0 else -1.0,
+            1 if post_lrelu > 0.0 else 0, float(post_lrelu),
         )
     if pre_lrelu > 0.0:
         x = F.leaky_relu(x, pre_lrelu)
     y = F.conv1d(x, weight, bias, stride=stride, padding=padding,
-                 dilation=dilation)
+                 dilation=dilation, groups=groups)
     if post_lrelu > 0.0:
         y = F.leaky_relu(y, post_lrelu)
     return y
@@ -152,11 +204,29 @@ def leaky_convtranspose1d(
     upsampling stage (y = convT(lrelu(x)))."""
     if use_hip(x):
         ext = hip_ext(required=True)
+        Cin, Cout, k = weight.shape
+        mfma = x.dtype == torch.bfloat16
+        w = _convt_weight_mfma(weight, stride) if mfma else weight.contiguous()
         return ext.convtranspose1d_fused(
-            x.contiguous(), weight.contiguous(),
+            x.contiguous(), w,
             bias.contiguous() if bias is not None else None,
-            stride, padding, float(pre_lrelu),
+            Cout, k, stride, padding,
+            pre_lrelu if pre_lrelu > 0.0 else -1.0,
         )
     if pre_lrelu > 0.0:
         x = F.leaky_relu(x, pre_lrelu)
     return F.conv_transpose1d(x, weight, bias, stride=stride, padding=padding)
+
+
+def conv_mod(
+    mod: torch.nn.Conv1d,
+    x: torch.Tensor,
+    pre_lrelu: float = 0.0,
+    post_lrelu: float = 0.0,
+) -> torch.Tensor:
+    """Run an nn.Conv1d module through the dispatched conv op."""
+    return leaky_conv1d(
+        x, mod.weight, mod.bias, stride=mod.stride[0], padding=mod.padding[0],
+        dilation=mod.dilation[0], groups=mod.groups,
+        pre_lrelu=pre_lrelu, post_lrelu=post_lrelu,
+    )
