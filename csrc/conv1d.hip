@@ -38,11 +38,12 @@
 #define ACT_TANH 2
 
 template <int BM, int BN, int WGM, int WGN>
-__global__ __launch_bounds__(256) void conv1d_mfma_kernel(
+__global__ __launch_bounds__(512) void conv1d_mfma_kernel(
     const bf16* __restrict__ x,     // [B][Cin][Tin]
     const bf16* __restrict__ w,     // [ntaps][CoutP][CinP] pre-permuted
     const float* __restrict__ bias, // [Cout] or null
     bf16* __restrict__ out,         // [B][Cout][Tout]
+    const bf16* __restrict__ resid, // optional residual, same shape as out
     int Cin, int CinP, int Cout, int CoutP, long Tin, long Tout,
     int ntaps, int tap_in_stride,  // input offset step per tap (dil or -1)
     int in_off0,                   // input offset of n=0, tap=0
@@ -50,8 +51,7 @@ __global__ __launch_bounds__(256) void conv1d_mfma_kernel(
     int out_stride, int out_off,   // out t = n*out_stride + out_off
     float pre_slope,               // <0: no pre-act; else LeakyReLU slope
     int act_mode, float post_slope) {
-  // grid: (ceil(Nvirt/BN), ceil(Cout/BM) [, *s phases folded by caller],
-  //        B)
+  // grid: (ceil(Nvirt/BN), ceil(Cout/BM), B); 512 threads = 8 waves.
   const int n_tile = blockIdx.x;
   const int m_tile = blockIdx.y;
   const int b = blockIdx.z;
@@ -67,8 +67,8 @@ __global__ __launch_bounds__(256) void conv1d_mfma_kernel(
   const int wid = tid >> 6;
   const int wr = wid / WGN;  // wave row (M)
   const int wc = wid % WGN;  // wave col (N)
-  constexpr int WM = BM / WGM;  // per-wave M (32)
-  constexpr int WN = BN / WGN;  // per-wave N (32)
+  constexpr int WM = BM / WGM;  // per-wave M
+  constexpr int WN = BN / WGN;  // per-wave N
   constexpr int MT = WM / 16;   // m fragments per wave
   constexpr int NT = WN / 16;   // n fragments per wave
 
@@ -79,15 +79,17 @@ __global__ __launch_bounds__(256) void conv1d_mfma_kernel(
     for (int j = 0; j < NT; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const bf16* xb = x + (long)b * Cin * Tin;
+  const int kl = lane >> 4;  // k-lane group (0..3)
+  const int il = lane & 15;  // row/col within fragment
 
   for (int tap = 0; tap < ntaps; ++tap) {
     const long w0 = n0 + in_off0 + (long)tap * tap_in_stride;
     for (int c0 = 0; c0 < CinP; c0 += BK) {
       // ---- stage W tile: Ws[m][kk] = w[tap][m0+m][c0+kk] -------------
-      // 256 threads fill BM*BK bf16; 8 contiguous per thread.
       {
         const long wbase = ((long)tap * CoutP + m0) * CinP + c0;
-        for (int e = tid * 8; e < BM * BK; e += 256 * 8) {
+#pragma unroll
+        for (int e = tid * 8; e < BM * BK; e += 512 * 8) {
           int m = e / BK, kk = e % BK;
           *(ulonglong2*)&Ws[m][kk] =
               *(const ulonglong2*)&w[wbase + (long)m * CinP + kk];
@@ -95,7 +97,8 @@ __global__ __launch_bounds__(256) void conv1d_mfma_kernel(
       }
       // ---- stage X tile: Xs[r][c] = pre(x[c0+r][w0+c]) ---------------
       {
-        for (int e = tid * 8; e < BK * BN; e += 256 * 8) {
+#pragma unroll
+        for (int e = tid * 8; e < BK * BN; e += 512 * 8) {
           int r = e / BN, c = e % BN;
           int ci = c0 + r;
           bf16 vals[8];
@@ -124,32 +127,32 @@ __global__ __launch_bounds__(256) void conv1d_mfma_kernel(
       }
       __syncthreads();
 
-      // ---- MFMA ------------------------------------------------------
-      const int kl = lane >> 4;      // k-lane group (0..3)
-      const int il = lane & 15;      // row/col within fragment
+      // ---- MFMA: load B fragments once, reuse across all M tiles -----
+      bf16x8 b_frag[NT];
+#pragma unroll
+      for (int nj = 0; nj < NT; ++nj) {
+        const int ncol = wc * WN + nj * 16 + il;
+#pragma unroll
+        for (int q = 0; q < 8; ++q)
+          b_frag[nj][q] = *(__bf16*)&Xs[kl * 8 + q][ncol];
+      }
 #pragma unroll
       for (int mi = 0; mi < MT; ++mi) {
         bf16x8 a_frag =
             *(const bf16x8*)&Ws[wr * WM + mi * 16 + il][kl * 8];
 #pragma unroll
         for (int nj = 0; nj < NT; ++nj) {
-          bf16x8 b_frag;
-          const int ncol = wc * WN + nj * 16 + il;
-#pragma unroll
-          for (int q = 0; q < 8; ++q)
-            b_frag[q] = *(__bf16*)&Xs[kl * 8 + q][ncol];
           acc[mi][nj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag, b_frag, acc[mi][nj], 0, 0, 0);
+              a_frag, b_frag[nj], acc[mi][nj], 0, 0, 0);
         }
       }
       __syncthreads();
     }
   }
 
-  // ---- epilogue: bias + activation + store --------------------------
+  // ---- epilogue: bias + activation (+ residual) + store --------------
   bf16* ob = out + (long)b * Cout * Tout;
-  const int kl = lane >> 4;
-  const int il = lane & 15;
+  const bf16* rb = resid ? resid + (long)b * Cout * Tout : nullptr;
 #pragma unroll
   for (int mi = 0; mi < MT; ++mi) {
     const int row = m0 + wr * WM + mi * 16 + kl * 4;
@@ -167,6 +170,7 @@ __global__ __launch_bounds__(256) void conv1d_mfma_kernel(
         if (bias) v += bias[co];
         if (act_mode == ACT_LRELU) v = lrelu_(v, post_slope);
         else if (act_mode == ACT_TANH) v = tanhf(v);
+        if (rb) v += bf2f(rb[(long)co * Tout + t]);
         ob[(long)co * Tout + t] = f2bf(v);
       }
     }
@@ -253,23 +257,24 @@ static inline long conv_out_len(long Tin, int k, int stride, int pad,
 
 // Launch helper: pick tile config by Cout and launch the MFMA kernel.
 static void launch_conv_mfma(const bf16* x, const bf16* w, const float* bias,
-                             bf16* out, int B, int Cin, int CinP, int Cout,
-                             int CoutP, long Tin, long Tout, int ntaps,
-                             int tap_in_stride, int in_off0, long Nvirt,
-                             int out_stride, int out_off, float pre_slope,
-                             int act_mode, float post_slope,
-                             hipStream_t stream) {
+                             bf16* out, const bf16* resid, int B, int Cin,
+                             int CinP, int Cout, int CoutP, long Tin,
+                             long Tout, int ntaps, int tap_in_stride,
+                             int in_off0, long Nvirt, int out_stride,
+                             int out_off, float pre_slope, int act_mode,
+                             float post_slope, hipStream_t stream) {
 #define LAUNCH(BM, BN, WGM, WGN)                                            \
   do {                                                                      \
     dim3 grid(ceil_div(Nvirt, BN), ceil_div(Cout, BM), B);                  \
     hipLaunchKernelGGL((conv1d_mfma_kernel<BM, BN, WGM, WGN>), grid,        \
-                       dim3(256), 0, stream, x, w, bias, out, Cin, CinP,    \
-                       Cout, CoutP, Tin, Tout, ntaps, tap_in_stride,        \
+                       dim3(512), 0, stream, x, w, bias, out, resid, Cin,   \
+                       CinP, Cout, CoutP, Tin, Tout, ntaps, tap_in_stride,  \
                        in_off0, Nvirt, out_stride, out_off, pre_slope,      \
                        act_mode, post_slope);                               \
   } while (0)
-  if (Cout >= 64) LAUNCH(64, 64, 2, 2);
-  else LAUNCH(32, 128, 1, 4);
+  if (Cout >= 128) LAUNCH(128, 128, 4, 2);
+  else if (Cout >= 64) LAUNCH(64, 128, 2, 4);
+  else LAUNCH(32, 256, 1, 8);
 #undef LAUNCH
 }
 
@@ -277,7 +282,8 @@ torch::Tensor conv1d_fused(torch::Tensor x, torch::Tensor w_perm,
                            c10::optional<torch::Tensor> bias, long Cout,
                            long k, long stride, long padding, long dilation,
                            long groups, double pre_lrelu, long act_mode,
-                           double post_slope) {
+                           double post_slope,
+                           c10::optional<torch::Tensor> residual) {
   // w_perm: [k][CoutP][CinP] bf16 (pre-permuted+padded) for the MFMA path
   //         or the raw [Cout][Cin/g][k] tensor for the fallback path.
   TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
@@ -294,12 +300,17 @@ torch::Tensor conv1d_fused(torch::Tensor x, torch::Tensor w_perm,
   const bool mfma_ok = x.scalar_type() == at::kBFloat16 && groups == 1 &&
                        stride == 1 && w_perm.dim() == 3 &&
                        w_perm.size(0) == k;
+  const bf16* res_p = nullptr;
+  if (residual.has_value()) {
+    TORCH_CHECK(residual->sizes() == out.sizes() && residual->is_contiguous());
+    res_p = (const bf16*)residual->data_ptr();
+  }
   if (mfma_ok) {
     const int CoutP = w_perm.size(1), CinP = w_perm.size(2);
     launch_conv_mfma((const bf16*)x.data_ptr(),
                      (const bf16*)w_perm.data_ptr(), bias_p,
-                     (bf16*)out.data_ptr(), B, Cin, CinP, Cout, CoutP, Tin,
-                     Tout, k, dilation, -(int)padding, Tout, 1, 0,
+                     (bf16*)out.data_ptr(), res_p, B, Cin, CinP, Cout, CoutP,
+                     Tin, Tout, k, dilation, -(int)padding, Tout, 1, 0,
                      (float)pre_lrelu, (int)act_mode, (float)post_slope,
                      cur_stream2());
   } else {
@@ -315,6 +326,7 @@ torch::Tensor conv1d_fused(torch::Tensor x, torch::Tensor w_perm,
                          stride, padding, dilation, groups, (float)pre_lrelu,
                          (int)act_mode, (float)post_slope, n);
     });
+    if (res_p) out.add_(*residual);
   }
   return out;
 }
@@ -355,8 +367,8 @@ torch::Tensor convtranspose1d_fused(torch::Tensor x, torch::Tensor w_perm,
           (const bf16*)w_perm.data_ptr() + (long)r * kr_max * CoutP * CinP;
       // tap m reads x[v - m] -> in_off0 = v_lo, tap stride -1
       launch_conv_mfma((const bf16*)x.data_ptr(), wr, bias_p,
-                       (bf16*)out.data_ptr(), B, Cin, CinP, Cout, CoutP,
-                       Tin, Tout, ntaps, -1, (int)v_lo, Nvirt, stride,
+                       (bf16*)out.data_ptr(), nullptr, B, Cin, CinP, Cout,
+                       CoutP, Tin, Tout, ntaps, -1, (int)v_lo, Nvirt, stride,
                        (int)(v_lo * stride + r - padding), (float)pre_lrelu,
                        ACT_NONE, 0.f, cur_stream2());
     }

@@ -15,8 +15,9 @@
 // Thread t-major: lane i handles time position t0+i so every c-iteration
 // reads a contiguous [blockDim.x] segment (fully coalesced along T).
 // --------------------------------------------------------------------------
-template <typename T>
+template <typename T, bool HAS_RES>
 __global__ void layer_norm_ct_kernel(const T* __restrict__ x,
+                                     const T* __restrict__ res,
                                      const float* __restrict__ gamma,
                                      const float* __restrict__ beta,
                                      T* __restrict__ out, int C, long T_len,
@@ -25,20 +26,25 @@ __global__ void layer_norm_ct_kernel(const T* __restrict__ x,
   if (bt >= n_bt) return;
   const long b = bt / T_len;
   const long t = bt % T_len;
-  const T* xp = x + (b * C) * T_len + t;
-  float mean = 0.f;
-  for (int c = 0; c < C; ++c) mean += ld_f(xp + (long)c * T_len);
-  mean /= C;
-  float var = 0.f;
+  const long base = (b * C) * T_len + t;
+  const T* xp = x + base;
+  const T* rp = HAS_RES ? res + base : nullptr;
+  // single pass: sum + sum of squares (C is small, f32 accumulate)
+  float s = 0.f, ss = 0.f;
   for (int c = 0; c < C; ++c) {
-    float d = ld_f(xp + (long)c * T_len) - mean;
-    var += d * d;
+    float v = ld_f(xp + (long)c * T_len);
+    if (HAS_RES) v += ld_f(rp + (long)c * T_len);
+    s += v;
+    ss += v * v;
   }
-  float rstd = rsqrtf(var / C + eps);
-  T* op = out + (b * C) * T_len + t;
+  const float mean = s / C;
+  const float var = ss / C - mean * mean;
+  const float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+  T* op = out + base;
   for (int c = 0; c < C; ++c) {
-    float v = (ld_f(xp + (long)c * T_len) - mean) * rstd;
-    st_f(op + (long)c * T_len, v * gamma[c] + beta[c]);
+    float v = ld_f(xp + (long)c * T_len);
+    if (HAS_RES) v += ld_f(rp + (long)c * T_len);
+    st_f(op + (long)c * T_len, (v - mean) * rstd * gamma[c] + beta[c]);
   }
 }
 
@@ -142,21 +148,37 @@ static inline hipStream_t cur_stream() {
     }                                                                      \
   } while (0)
 
-torch::Tensor layer_norm_ct(torch::Tensor x, torch::Tensor gamma,
-                            torch::Tensor beta, double eps) {
+torch::Tensor layer_norm_ct(torch::Tensor x, c10::optional<torch::Tensor> res,
+                            torch::Tensor gamma, torch::Tensor beta,
+                            double eps) {
   TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
   const long B = x.size(0), C = x.size(1), T = x.size(2);
   auto out = torch::empty_like(x);
   auto gamma_f = gamma.to(at::kFloat).contiguous();
   auto beta_f = beta.to(at::kFloat).contiguous();
-  const int threads = 256;
+  // small blocks: B*T/threads workgroups must still fill 256 CUs for the
+  // short sequences of this graph
+  const int threads = 64;
   dim3 grid(ceil_div(B * T, threads));
   DISPATCH_FT(x, "layer_norm_ct", {
-    hipLaunchKernelGGL(layer_norm_ct_kernel<scalar_t>, grid, dim3(threads), 0,
-                       cur_stream(), (const scalar_t*)x.data_ptr(),
-                       gamma_f.data_ptr<float>(), beta_f.data_ptr<float>(),
-                       (scalar_t*)out.data_ptr(), (int)C, T, (float)eps,
-                       B * T);
+    if (res.has_value()) {
+      TORCH_CHECK(res->sizes() == x.sizes() && res->is_contiguous());
+      hipLaunchKernelGGL((layer_norm_ct_kernel<scalar_t, true>), grid,
+                         dim3(threads), 0, cur_stream(),
+                         (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)res->data_ptr(),
+                         gamma_f.data_ptr<float>(), beta_f.data_ptr<float>(),
+                         (scalar_t*)out.data_ptr(), (int)C, T, (float)eps,
+                         B * T);
+    } else {
+      hipLaunchKernelGGL((layer_norm_ct_kernel<scalar_t, false>), grid,
+                         dim3(threads), 0, cur_stream(),
+                         (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)nullptr,
+                         gamma_f.data_ptr<float>(), beta_f.data_ptr<float>(),
+                         (scalar_t*)out.data_ptr(), (int)C, T, (float)eps,
+                         B * T);
+    }
   });
   return out;
 }

@@ -2,8 +2,9 @@
 #include <torch/extension.h>
 
 // elementwise.hip
-torch::Tensor layer_norm_ct(torch::Tensor x, torch::Tensor gamma,
-                            torch::Tensor beta, double eps);
+torch::Tensor layer_norm_ct(torch::Tensor x, c10::optional<torch::Tensor> res,
+                            torch::Tensor gamma, torch::Tensor beta,
+                            double eps);
 torch::Tensor fused_gate(torch::Tensor x, c10::optional<torch::Tensor> g,
                          long n_channels);
 torch::Tensor prior_sample(torch::Tensor m, torch::Tensor logs,
@@ -16,7 +17,8 @@ torch::Tensor conv1d_fused(torch::Tensor x, torch::Tensor w_perm,
                            c10::optional<torch::Tensor> bias, long Cout,
                            long k, long stride, long padding, long dilation,
                            long groups, double pre_lrelu, long act_mode,
-                           double post_slope);
+                           double post_slope,
+                           c10::optional<torch::Tensor> residual);
 torch::Tensor convtranspose1d_fused(torch::Tensor x, torch::Tensor w_perm,
                                     c10::optional<torch::Tensor> bias,
                                     long Cout, long k, long stride,

@@ -54,8 +54,10 @@ class LayerNormCT(nn.Module):
         self.gamma = nn.Parameter(torch.ones(channels))
         self.beta = nn.Parameter(torch.zeros(channels))
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return layer_norm_ct(x, self.gamma, self.beta, self.eps)
+    def forward(self, x: torch.Tensor,
+                residual: "Optional[torch.Tensor]" = None) -> torch.Tensor:
+        return layer_norm_ct(x, self.gamma, self.beta, self.eps,
+                             residual=residual)
 
 
 # --------------------------------------------------------------------------- #
@@ -175,9 +177,9 @@ class TextEncoder(nn.Module):
         for attn, n1, ffn, n2 in zip(self.attn_layers, self.norm1,
                                      self.ffn_layers, self.norm2):
             y = attn(x * x_mask, attn_mask.unsqueeze(1))
-            x = n1(x + y)
+            x = n1(x, residual=y)   # fused residual-add + LayerNorm
             y = ffn(x, x_mask)
-            x = n2(x + y)
+            x = n2(x, residual=y)
         stats = conv_mod(self.proj, x) * x_mask
         m, logs = stats.chunk(2, dim=1)
         return x, m, logs, x_mask
@@ -555,11 +557,11 @@ class ResBlock1(nn.Module):
                 padding=c1.padding[0], dilation=c1.dilation[0],
                 pre_lrelu=LRELU_SLOPE,
             )
-            xt = leaky_conv1d(
+            # second conv fuses the residual add into its epilogue
+            x = leaky_conv1d(
                 xt, c2.weight, c2.bias, padding=c2.padding[0],
-                pre_lrelu=LRELU_SLOPE,
+                pre_lrelu=LRELU_SLOPE, residual=x,
             )
-            x = x + xt
         return x
 
 

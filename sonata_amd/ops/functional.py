@@ -26,11 +26,23 @@ from . import hip_ext, use_hip
 # LayerNorm over the channel dim of [B, C, T]
 # --------------------------------------------------------------------------- #
 def layer_norm_ct(
-    x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor, eps: float = 1e-5
+    x: torch.Tensor,
+    gamma: torch.Tensor,
+    beta: torch.Tensor,
+    eps: float = 1e-5,
+    residual: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
+    """LayerNorm across channels of [B,C,T]; optional fused residual:
+    normalizes (x + residual)."""
     if use_hip(x):
         ext = hip_ext(required=True)
-        return ext.layer_norm_ct(x.contiguous(), gamma, beta, eps)
+        return ext.layer_norm_ct(
+            x.contiguous(),
+            residual.contiguous() if residual is not None else None,
+            gamma, beta, eps,
+        )
+    if residual is not None:
+        x = x + residual
     # torch reference: normalize across C for each (b, t)
     mean = x.mean(dim=1, keepdim=True)
     var = x.var(dim=1, unbiased=False, keepdim=True)
@@ -123,7 +135,7 @@ def _conv_weight_mfma(weight: torch.Tensor) -> torch.Tensor:
     if cached is not None:
         return cached
     Cout, Cin, k = weight.shape
-    bm = 64 if Cout >= 64 else 32
+    bm = 128 if Cout >= 128 else (64 if Cout >= 64 else 32)
     CoutP, CinP = _round_up(Cout, bm), _round_up(Cin, 32)
     perm = torch.zeros((k, CoutP, CinP), dtype=torch.bfloat16,
                        device=weight.device)
@@ -141,7 +153,7 @@ def _convt_weight_mfma(weight: torch.Tensor, stride: int) -> torch.Tensor:
         return cached
     Cin, Cout, k = weight.shape
     kr_max = (k + stride - 1) // stride
-    bm = 64 if Cout >= 64 else 32
+    bm = 128 if Cout >= 128 else (64 if Cout >= 64 else 32)
     CoutP, CinP = _round_up(Cout, bm), _round_up(Cin, 32)
     perm = torch.zeros((stride, kr_max, CoutP, CinP), dtype=torch.bfloat16,
                        device=weight.device)
@@ -164,11 +176,13 @@ def leaky_conv1d(
     groups: int = 1,
     pre_lrelu: float = 0.0,
     post_lrelu: float = 0.0,
+    residual: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """Conv1d with optional fused LeakyReLU applied to the input
-    (pre_lrelu>0) and/or the output (post_lrelu>0).  Fusing the activation
-    into the producing GEMM keeps the activation tensor out of HBM
-    (HiFi-GAN MRF pattern: y = conv(lrelu(x)))."""
+    (pre_lrelu>0) and/or the output (post_lrelu>0), plus an optional fused
+    residual add (out = act(conv(x)) + residual).  Fusing keeps activation
+    and residual tensors out of HBM (HiFi-GAN MRF pattern:
+    x = x + conv(lrelu(conv(lrelu(x)))))."""
     if use_hip(x):
         ext = hip_ext(required=True)
         Cout, _, k = weight.shape
@@ -182,6 +196,7 @@ def leaky_conv1d(
             Cout, k, stride, padding, dilation, groups,
             pre_lrelu if pre_lrelu > 0.0 else -1.0,
             1 if post_lrelu > 0.0 else 0, float(post_lrelu),
+            residual.contiguous() if residual is not None else None,
         )
     if pre_lrelu > 0.0:
         x = F.leaky_relu(x, pre_lrelu)
@@ -189,6 +204,8 @@ def leaky_conv1d(
                  dilation=dilation, groups=groups)
     if post_lrelu > 0.0:
         y = F.leaky_relu(y, post_lrelu)
+    if residual is not None:
+        y = y + residual
     return y
 
 
