@@ -37,7 +37,10 @@
 #define ACT_LRELU 1
 #define ACT_TANH 2
 
-template <int BM, int BN, int WGM, int WGN>
+// Max halo: (ntaps-1)*|tap_stride| — VITS worst case is k=11, dil=5 -> 50.
+#define HALO_MAX 64
+
+template <int BM, int BN, int WGM, int WGN, int TC>
 __global__ __launch_bounds__(512) void conv1d_mfma_kernel(
     const bf16* __restrict__ x,     // [B][Cin][Tin]
     const bf16* __restrict__ w,     // [ntaps][CoutP][CinP] pre-permuted
@@ -59,8 +62,10 @@ __global__ __launch_bounds__(512) void conv1d_mfma_kernel(
   const long n0 = (long)n_tile * BN;
   const int m0 = m_tile * BM;
 
-  __shared__ bf16 Ws[BM][BK + 8];
-  __shared__ bf16 Xs[BK][BN + 8];
+  // X window staged ONCE per K-slice: all taps read from it at their own
+  // column offset (k-fold less LDS staging + barriers than per-tap tiles).
+  __shared__ bf16 Ws[TC][BM][BK + 8];
+  __shared__ bf16 Xs[BK][BN + HALO_MAX + 8];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -71,6 +76,7 @@ __global__ __launch_bounds__(512) void conv1d_mfma_kernel(
   constexpr int WN = BN / WGN;  // per-wave N
   constexpr int MT = WM / 16;   // m fragments per wave
   constexpr int NT = WN / 16;   // n fragments per wave
+  constexpr int XW = BN + HALO_MAX;  // staged X width
 
   f32x4 acc[MT][NT];
 #pragma unroll
@@ -82,68 +88,77 @@ __global__ __launch_bounds__(512) void conv1d_mfma_kernel(
   const int kl = lane >> 4;  // k-lane group (0..3)
   const int il = lane & 15;  // row/col within fragment
 
-  for (int tap = 0; tap < ntaps; ++tap) {
-    const long w0 = n0 + in_off0 + (long)tap * tap_in_stride;
-    for (int c0 = 0; c0 < CinP; c0 += BK) {
-      // ---- stage W tile: Ws[m][kk] = w[tap][m0+m][c0+kk] -------------
-      {
-        const long wbase = ((long)tap * CoutP + m0) * CinP + c0;
+  // window start: smallest input index any tap needs for col 0
+  const int min_tap_off =
+      tap_in_stride < 0 ? (ntaps - 1) * tap_in_stride : 0;
+  const long w0 = n0 + in_off0 + min_tap_off;
+
+  for (int c0 = 0; c0 < CinP; c0 += BK) {
+    // ---- stage X window: Xs[r][c] = pre(x[c0+r][w0+c]), c < XW -------
+    {
+#pragma unroll 2
+      for (int e = tid * 8; e < BK * XW; e += 512 * 8) {
+        int r = e / XW, c = e % XW;
+        int ci = c0 + r;
+        bf16 vals[8];
+        long p = w0 + c;
+        if (ci < Cin && p >= 0 && p + 7 < Tin) {
+          *(ulonglong2*)vals = *(const ulonglong2*)&xb[(long)ci * Tin + p];
+          if (pre_slope >= 0.f) {
+#pragma unroll
+            for (int q = 0; q < 8; ++q)
+              vals[q] = f2bf(lrelu_(bf2f(vals[q]), pre_slope));
+          }
+        } else {
+#pragma unroll
+          for (int q = 0; q < 8; ++q) {
+            long pq = p + q;
+            float v = (ci < Cin && pq >= 0 && pq < Tin)
+                          ? bf2f(xb[(long)ci * Tin + pq])
+                          : 0.f;
+            if (pre_slope >= 0.f) v = lrelu_(v, pre_slope);
+            vals[q] = f2bf(v);
+          }
+        }
+        // c is a multiple of 8 and rows are 16B-aligned -> one b128 write
+        *(ulonglong2*)&Xs[r][c] = *(ulonglong2*)vals;
+      }
+    }
+
+    for (int tap0 = 0; tap0 < ntaps; tap0 += TC) {
+      const int ntc = min(TC, ntaps - tap0);
+      // ---- stage W chunk: Ws[tc][m][kk] = w[tap0+tc][m0+m][c0+kk] ----
+      for (int tc = 0; tc < ntc; ++tc) {
+        const long wbase = ((long)(tap0 + tc) * CoutP + m0) * CinP + c0;
 #pragma unroll
         for (int e = tid * 8; e < BM * BK; e += 512 * 8) {
           int m = e / BK, kk = e % BK;
-          *(ulonglong2*)&Ws[m][kk] =
+          *(ulonglong2*)&Ws[tc][m][kk] =
               *(const ulonglong2*)&w[wbase + (long)m * CinP + kk];
-        }
-      }
-      // ---- stage X tile: Xs[r][c] = pre(x[c0+r][w0+c]) ---------------
-      {
-#pragma unroll
-        for (int e = tid * 8; e < BK * BN; e += 512 * 8) {
-          int r = e / BN, c = e % BN;
-          int ci = c0 + r;
-          bf16 vals[8];
-          long p = w0 + c;
-          if (ci < Cin && p >= 0 && p + 7 < Tin) {
-            *(ulonglong2*)vals = *(const ulonglong2*)&xb[(long)ci * Tin + p];
-            if (pre_slope >= 0.f) {
-#pragma unroll
-              for (int q = 0; q < 8; ++q)
-                vals[q] = f2bf(lrelu_(bf2f(vals[q]), pre_slope));
-            }
-          } else {
-#pragma unroll
-            for (int q = 0; q < 8; ++q) {
-              long pq = p + q;
-              float v = (ci < Cin && pq >= 0 && pq < Tin)
-                            ? bf2f(xb[(long)ci * Tin + pq])
-                            : 0.f;
-              if (pre_slope >= 0.f) v = lrelu_(v, pre_slope);
-              vals[q] = f2bf(v);
-            }
-          }
-          // c is a multiple of 8 and rows are 16B-aligned -> one b128 write
-          *(ulonglong2*)&Xs[r][c] = *(ulonglong2*)vals;
         }
       }
       __syncthreads();
 
-      // ---- MFMA: load B fragments once, reuse across all M tiles -----
-      bf16x8 b_frag[NT];
-#pragma unroll
-      for (int nj = 0; nj < NT; ++nj) {
-        const int ncol = wc * WN + nj * 16 + il;
-#pragma unroll
-        for (int q = 0; q < 8; ++q)
-          b_frag[nj][q] = *(__bf16*)&Xs[kl * 8 + q][ncol];
-      }
-#pragma unroll
-      for (int mi = 0; mi < MT; ++mi) {
-        bf16x8 a_frag =
-            *(const bf16x8*)&Ws[wr * WM + mi * 16 + il][kl * 8];
+      for (int tc = 0; tc < ntc; ++tc) {
+        // column offset of this tap inside the staged window
+        const int toff = (tap0 + tc) * tap_in_stride - min_tap_off;
+        bf16x8 b_frag[NT];
 #pragma unroll
         for (int nj = 0; nj < NT; ++nj) {
-          acc[mi][nj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag, b_frag[nj], acc[mi][nj], 0, 0, 0);
+          const int ncol = wc * WN + nj * 16 + il + toff;
+#pragma unroll
+          for (int q = 0; q < 8; ++q)
+            b_frag[nj][q] = *(__bf16*)&Xs[kl * 8 + q][ncol];
+        }
+#pragma unroll
+        for (int mi = 0; mi < MT; ++mi) {
+          bf16x8 a_frag =
+              *(const bf16x8*)&Ws[tc][wr * WM + mi * 16 + il][kl * 8];
+#pragma unroll
+          for (int nj = 0; nj < NT; ++nj) {
+            acc[mi][nj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag, b_frag[nj], acc[mi][nj], 0, 0, 0);
+          }
         }
       }
       __syncthreads();
@@ -263,18 +278,18 @@ static void launch_conv_mfma(const bf16* x, const bf16* w, const float* bias,
                              int in_off0, long Nvirt, int out_stride,
                              int out_off, float pre_slope, int act_mode,
                              float post_slope, hipStream_t stream) {
-#define LAUNCH(BM, BN, WGM, WGN)                                            \
+#define LAUNCH(BM, BN, WGM, WGN, TC)                                        \
   do {                                                                      \
     dim3 grid(ceil_div(Nvirt, BN), ceil_div(Cout, BM), B);                  \
-    hipLaunchKernelGGL((conv1d_mfma_kernel<BM, BN, WGM, WGN>), grid,        \
+    hipLaunchKernelGGL((conv1d_mfma_kernel<BM, BN, WGM, WGN, TC>), grid,    \
                        dim3(512), 0, stream, x, w, bias, out, resid, Cin,   \
                        CinP, Cout, CoutP, Tin, Tout, ntaps, tap_in_stride,  \
                        in_off0, Nvirt, out_stride, out_off, pre_slope,      \
                        act_mode, post_slope);                               \
   } while (0)
-  if (Cout >= 128) LAUNCH(128, 128, 4, 2);
-  else if (Cout >= 64) LAUNCH(64, 128, 2, 4);
-  else LAUNCH(32, 256, 1, 8);
+  if (Cout >= 128) LAUNCH(128, 128, 4, 2, 2);
+  else if (Cout >= 64) LAUNCH(64, 128, 2, 4, 4);
+  else LAUNCH(32, 256, 1, 8, 8);
 #undef LAUNCH
 }
 
@@ -299,7 +314,8 @@ torch::Tensor conv1d_fused(torch::Tensor x, torch::Tensor w_perm,
   }
   const bool mfma_ok = x.scalar_type() == at::kBFloat16 && groups == 1 &&
                        stride == 1 && w_perm.dim() == 3 &&
-                       w_perm.size(0) == k;
+                       w_perm.size(0) == k &&
+                       (k - 1) * dilation <= 64 /* HALO_MAX */;
   const bf16* res_p = nullptr;
   if (residual.has_value()) {
     TORCH_CHECK(residual->sizes() == out.sizes() && residual->is_contiguous());

@@ -43,12 +43,20 @@ def test_layer_norm_ct(dev, dtype):
     x = torch.randn(3, 192, 517, device=dev, dtype=dtype)
     g = torch.randn(192, device=dev)
     b = torch.randn(192, device=dev)
-    got = ext.layer_norm_ct(x, g, b, 1e-5)
+    got = ext.layer_norm_ct(x, None, g, b, 1e-5)
     xf = x.float()
     mean = xf.mean(1, keepdim=True)
     var = xf.var(1, unbiased=False, keepdim=True)
     ref = (xf - mean) * torch.rsqrt(var + 1e-5) * g.view(1, -1, 1) + b.view(1, -1, 1)
     assert _rel_err(got, ref) < (0.02 if dtype == torch.bfloat16 else 1e-4)
+    # fused residual variant
+    r = torch.randn_like(x)
+    got2 = ext.layer_norm_ct(x, r, g, b, 1e-5)
+    xr = (x.float() + r.float())
+    mean2 = xr.mean(1, keepdim=True)
+    var2 = xr.var(1, unbiased=False, keepdim=True)
+    ref2 = (xr - mean2) * torch.rsqrt(var2 + 1e-5) * g.view(1, -1, 1) + b.view(1, -1, 1)
+    assert _rel_err(got2, ref2) < (0.03 if dtype == torch.bfloat16 else 1e-4)
 
 
 @pytest.mark.parametrize("has_g", [False, True])
@@ -164,6 +172,20 @@ def test_conv1d_odd_sizes(dev):
 
 def test_conv1d_post_act(dev):
     _conv_case(dev, 1, 128, 128, 400, 3, post=0.1)
+
+
+def test_conv1d_fused_residual(dev):
+    from sonata_amd.ops.functional import leaky_conv1d
+
+    torch.manual_seed(77)
+    x = (torch.randn(2, 128, 300) / 4).to(torch.bfloat16)
+    w = (torch.randn(128, 128, 3) / 20).to(torch.bfloat16)
+    res = (torch.randn(2, 128, 300) / 4).to(torch.bfloat16)
+    got = leaky_conv1d(x.to(dev), w.to(dev), None, padding=1,
+                       pre_lrelu=0.1, residual=res.to(dev))
+    xf = torch.nn.functional.leaky_relu(x.float(), 0.1)
+    ref = torch.nn.functional.conv1d(xf, w.float(), None, padding=1) + res.float()
+    assert _rel_err(got, ref) < 0.02
 
 
 def test_conv1d_depthwise(dev):
