@@ -19,7 +19,15 @@ __global__ void probe(float* out, int pattern) {
     case 1: off = (lane & 15) * 4; break;            // lane row of 4
     case 2: off = (lane >> 4) * 64; break;           // group base
     case 3: off = (lane & 15) * 4 + (lane >> 4) * 64; break;
-    default: off = lane * 4; break;                  // fully linear
+    case 4: off = lane * 4; break;                   // fully linear
+    case 5: off = 1 + (lane & 15) * 4; break;        // UNALIGNED (2B) test
+    case 6: off = 2 + (lane & 15) * 4; break;        // 4B-aligned test
+    default: {                                       // intended conv use:
+      // B[k][n] tile rows pitch 40; lane m: row m/4 (in 16-group), col 4*(m%4)
+      int il = lane & 15, kl = lane >> 4;
+      off = (kl * 8 + il / 4) * 40 + 4 * (il % 4);
+      break;
+    }
   }
   auto p = (__LDS_ADDR bf16x4*)((__LDS_ADDR __bf16*)lds + off);
   bf16x4 v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
@@ -30,7 +38,7 @@ int main() {
   float* d;
   hipMalloc(&d, 64 * 4 * sizeof(float));
   float h[256];
-  for (int pat = 0; pat <= 4; ++pat) {
+  for (int pat = 0; pat <= 7; ++pat) {
     hipLaunchKernelGGL(probe, dim3(1), dim3(64), 0, 0, d, pat);
     hipMemcpy(h, d, sizeof(h), hipMemcpyDeviceToHost);
     printf("pattern %d:\n", pat);

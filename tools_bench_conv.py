@@ -1,0 +1,55 @@
+"""Microbench: time leaky_conv1d / leaky_convtranspose1d on the VITS hot
+shapes; prints TFLOP/s and GB/s per shape."""
+import sys, time, torch
+sys.path.insert(0, '.')
+from sonata_amd.ops.functional import leaky_conv1d, leaky_convtranspose1d
+
+B = 64
+F = 256
+shapes = [
+    # (name, Cin, Cout, T, k, dil, pre)
+    ("res@256 k3d1", 256, 256, 8*F, 3, 1, 0.1),
+    ("res@256 k11d5", 256, 256, 8*F, 11, 5, 0.1),
+    ("res@128 k3d1", 128, 128, 64*F, 3, 1, 0.1),
+    ("res@64 k3d1", 64, 64, 128*F, 3, 1, 0.1),
+    ("res@32 k3d1", 32, 32, 256*F, 3, 1, 0.1),
+    ("conv_pre k7", 192, 512, F, 7, 1, 0.0),
+    ("wn k5 192->384", 192, 384, F, 5, 1, 0.0),
+    ("proj1x1 192", 192, 192, F, 1, 1, 0.0),
+]
+dev = "cuda:0"
+for name, Cin, Cout, T, k, dil, pre in shapes:
+    x = (torch.randn(B, Cin, T)/4).to(torch.bfloat16).to(dev)
+    w = (torch.randn(Cout, Cin, k)/(Cin*k)**0.5).to(torch.bfloat16).to(dev)
+    bias = (torch.randn(Cout)/10).to(dev)
+    pad = (k-1)*dil//2
+    for _ in range(3):
+        y = leaky_conv1d(x, w, bias, padding=pad, dilation=dil, pre_lrelu=pre)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter(); N = 10
+    for _ in range(N):
+        y = leaky_conv1d(x, w, bias, padding=pad, dilation=dil, pre_lrelu=pre)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter()-t0)/N
+    fl = 2*B*Cin*Cout*k*T
+    gb = 2*B*T*(Cin+Cout)  # bf16 in+out
+    print(f"{name:18s} T={T:6d} {dt*1e3:8.3f} ms  {fl/dt/1e12:7.1f} TF  {gb/dt/1e9:7.0f} GB/s")
+
+# convT shapes
+for name, Cin, Cout, k, s, T in [("up0 512->256", 512, 256, 16, 8, F),
+                                 ("up1 256->128", 256, 128, 16, 8, 8*F),
+                                 ("up3 64->32", 64, 32, 4, 2, 128*F)]:
+    x = (torch.randn(B, Cin, T)/4).to(torch.bfloat16).to(dev)
+    w = (torch.randn(Cin, Cout, k)/(Cin*k)**0.5).to(torch.bfloat16).to(dev)
+    bias = (torch.randn(Cout)/10).to(dev)
+    for _ in range(3):
+        y = leaky_convtranspose1d(x, w, bias, s, (k-s)//2, pre_lrelu=0.1)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter(); N = 10
+    for _ in range(N):
+        y = leaky_convtranspose1d(x, w, bias, s, (k-s)//2, pre_lrelu=0.1)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter()-t0)/N
+    fl = 2*B*Cin*Cout*(k//s)*T*s
+    gb = 2*B*(T*Cin + T*s*Cout)
+    print(f"{name:18s} T={T:6d} {dt*1e3:8.3f} ms  {fl/dt/1e12:7.1f} TF  {gb/dt/1e9:7.0f} GB/s")
