@@ -53,11 +53,31 @@ __global__ __launch_bounds__(512) void conv1d_mfma_kernel(
     long Nvirt,                    // GEMM N size (Tout or V)
     int out_stride, int out_off,   // out t = n*out_stride + out_off
     float pre_slope,               // <0: no pre-act; else LeakyReLU slope
-    int act_mode, float post_slope) {
-  // grid: (ceil(Nvirt/BN), ceil(Cout/BM), B); 512 threads = 8 waves.
+    int act_mode, float post_slope,
+    // transposed-conv mode: all phases in ONE launch (convt_s > 0).
+    // grid.y = s * mtiles; per-block phase r is derived below and ntaps /
+    // offsets / weight base recomputed from (convt_k, convt_s, convt_pad).
+    int convt_s, int convt_k, int convt_pad, int kr_max) {
+  // grid: (ceil(Nvirt/BN), ceil(Cout/BM) [* s], B); 512 threads = 8 waves.
   const int n_tile = blockIdx.x;
-  const int m_tile = blockIdx.y;
+  int m_tile = blockIdx.y;
   const int b = blockIdx.z;
+
+  if (convt_s > 0) {
+    const int mtiles = gridDim.y / convt_s;
+    const int r = m_tile / mtiles;
+    m_tile = m_tile % mtiles;
+    ntaps = (convt_k - r + convt_s - 1) / convt_s;
+    int v_lo = (convt_pad - r + convt_s - 1) / convt_s;
+    if (v_lo < 0) v_lo = 0;
+    const long v_hi = (Tout - 1 + convt_pad - r) / convt_s;  // inclusive
+    Nvirt = v_hi - v_lo + 1;
+    if (Nvirt <= 0) return;
+    in_off0 = v_lo;
+    out_off = v_lo * convt_s + r - convt_pad;
+    out_stride = convt_s;
+    w += (long)r * kr_max * CoutP * CinP;
+  }
 
   const long n0 = (long)n_tile * BN;
   const int m0 = m_tile * BM;
@@ -93,9 +113,36 @@ __global__ __launch_bounds__(512) void conv1d_mfma_kernel(
       tap_in_stride < 0 ? (ntaps - 1) * tap_in_stride : 0;
   const long w0 = n0 + in_off0 + min_tap_off;
 
+  // thread -> (row, col-chunk) map for X staging, computed once: thread i
+  // handles chunks e = i*8 + s*4096 -> row/col via constexpr div (no
+  // per-iteration guards on the interior fast path).
+  const bool x_interior = (w0 >= 0) && (w0 + XW <= Tin);
+
   for (int c0 = 0; c0 < CinP; c0 += BK) {
     // ---- stage X window: Xs[r][c] = pre(x[c0+r][w0+c]), c < XW -------
-    {
+    const bool rows_ok = (c0 + BK) <= Cin;
+    if (x_interior && rows_ok) {
+      if (pre_slope < 0.f) {
+#pragma unroll
+        for (int e = tid * 8; e < BK * XW; e += 512 * 8) {
+          const int r = e / XW, c = e % XW;
+          *(ulonglong2*)&Xs[r][c] =
+              *(const ulonglong2*)&xb[(long)(c0 + r) * Tin + w0 + c];
+        }
+      } else {
+#pragma unroll
+        for (int e = tid * 8; e < BK * XW; e += 512 * 8) {
+          const int r = e / XW, c = e % XW;
+          bf16 vals[8];
+          *(ulonglong2*)vals =
+              *(const ulonglong2*)&xb[(long)(c0 + r) * Tin + w0 + c];
+#pragma unroll
+          for (int q = 0; q < 8; ++q)
+            vals[q] = f2bf(lrelu_(bf2f(vals[q]), pre_slope));
+          *(ulonglong2*)&Xs[r][c] = *(ulonglong2*)vals;
+        }
+      }
+    } else {
 #pragma unroll 2
       for (int e = tid * 8; e < BK * XW; e += 512 * 8) {
         int r = e / XW, c = e % XW;
@@ -277,15 +324,19 @@ static void launch_conv_mfma(const bf16* x, const bf16* w, const float* bias,
                              long Tout, int ntaps, int tap_in_stride,
                              int in_off0, long Nvirt, int out_stride,
                              int out_off, float pre_slope, int act_mode,
-                             float post_slope, hipStream_t stream) {
+                             float post_slope, hipStream_t stream,
+                             int convt_s = 0, int convt_k = 0,
+                             int convt_pad = 0, int kr_max = 0) {
+  const int phases = convt_s > 0 ? convt_s : 1;
 #define LAUNCH(BM, BN, WGM, WGN, TC)                                        \
   do {                                                                      \
-    dim3 grid(ceil_div(Nvirt, BN), ceil_div(Cout, BM), B);                  \
+    dim3 grid(ceil_div(Nvirt, BN), ceil_div(Cout, BM) * phases, B);         \
     hipLaunchKernelGGL((conv1d_mfma_kernel<BM, BN, WGM, WGN, TC>), grid,    \
                        dim3(512), 0, stream, x, w, bias, out, resid, Cin,   \
                        CinP, Cout, CoutP, Tin, Tout, ntaps, tap_in_stride,  \
                        in_off0, Nvirt, out_stride, out_off, pre_slope,      \
-                       act_mode, post_slope);                               \
+                       act_mode, post_slope, convt_s, convt_k, convt_pad,   \
+                       kr_max);                                             \
   } while (0)
   if (Cout >= 128) LAUNCH(128, 128, 4, 2, 2);
   else if (Cout >= 64) LAUNCH(64, 128, 2, 4, 4);
@@ -309,7 +360,9 @@ torch::Tensor conv1d_fused(torch::Tensor x, torch::Tensor w_perm,
   torch::Tensor bias_f;
   const float* bias_p = nullptr;
   if (bias.has_value()) {
-    bias_f = bias->to(at::kFloat).contiguous();
+    bias_f = bias->scalar_type() == at::kFloat
+                 ? *bias
+                 : bias->to(at::kFloat).contiguous();
     bias_p = bias_f.data_ptr<float>();
   }
   const bool mfma_ok = x.scalar_type() == at::kBFloat16 && groups == 1 &&
@@ -360,7 +413,9 @@ torch::Tensor convtranspose1d_fused(torch::Tensor x, torch::Tensor w_perm,
   torch::Tensor bias_f;
   const float* bias_p = nullptr;
   if (bias.has_value()) {
-    bias_f = bias->to(at::kFloat).contiguous();
+    bias_f = bias->scalar_type() == at::kFloat
+                 ? *bias
+                 : bias->to(at::kFloat).contiguous();
     bias_p = bias_f.data_ptr<float>();
   }
   const bool mfma_ok =
@@ -368,26 +423,14 @@ torch::Tensor convtranspose1d_fused(torch::Tensor x, torch::Tensor w_perm,
   if (mfma_ok) {
     const int kr_max = w_perm.size(1);
     const int CoutP = w_perm.size(2), CinP = w_perm.size(3);
-    // one GEMM per phase r: out[s*v + r - pad] over v in [0, Vmax)
-    for (int r = 0; r < stride; ++r) {
-      const int ntaps = (k - r + stride - 1) / stride;
-      // v range for valid t: 0 <= s*v + r - pad < Tout
-      // v_lo = ceil((pad - r)/s); fold into out_off by shifting v start.
-      long v_lo = (padding - r + stride - 1) / stride;
-      if (v_lo < 0) v_lo = 0;
-      long v_hi = (Tout - 1 + padding - r) / stride;  // inclusive
-      if (v_hi >= Tin + kr_max) v_hi = Tin + kr_max;
-      long Nvirt = v_hi - v_lo + 1;
-      if (Nvirt <= 0) continue;
-      const bf16* wr =
-          (const bf16*)w_perm.data_ptr() + (long)r * kr_max * CoutP * CinP;
-      // tap m reads x[v - m] -> in_off0 = v_lo, tap stride -1
-      launch_conv_mfma((const bf16*)x.data_ptr(), wr, bias_p,
-                       (bf16*)out.data_ptr(), nullptr, B, Cin, CinP, Cout,
-                       CoutP, Tin, Tout, ntaps, -1, (int)v_lo, Nvirt, stride,
-                       (int)(v_lo * stride + r - padding), (float)pre_lrelu,
-                       ACT_NONE, 0.f, cur_stream2());
-    }
+    // ALL phase GEMMs in one launch (phase folded into grid.y)
+    const long Nvirt_max = (Tout - 1 + padding) / stride + 1;
+    launch_conv_mfma((const bf16*)x.data_ptr(),
+                     (const bf16*)w_perm.data_ptr(), bias_p,
+                     (bf16*)out.data_ptr(), nullptr, B, Cin, CinP, Cout,
+                     CoutP, Tin, Tout, kr_max, -1, 0, Nvirt_max, stride, 0,
+                     (float)pre_lrelu, ACT_NONE, 0.f, cur_stream2(),
+                     (int)stride, (int)k, (int)padding, kr_max);
   } else {
     const long n = B * Cout * Tout;
     const int threads = 256;

@@ -166,6 +166,16 @@ def _convt_weight_mfma(weight: torch.Tensor, stride: int) -> torch.Tensor:
     return perm
 
 
+def _bias_f32(bias: Optional[torch.Tensor]) -> Optional[torch.Tensor]:
+    if bias is None:
+        return None
+    cached = getattr(bias, "_sonata_f32", None)
+    if cached is None:
+        cached = bias.detach().float().contiguous()
+        bias._sonata_f32 = cached
+    return cached
+
+
 def leaky_conv1d(
     x: torch.Tensor,
     weight: torch.Tensor,
@@ -191,8 +201,7 @@ def leaky_conv1d(
         )
         w = _conv_weight_mfma(weight) if mfma else weight.contiguous()
         return ext.conv1d_fused(
-            x.contiguous(), w,
-            bias.contiguous() if bias is not None else None,
+            x.contiguous(), w, _bias_f32(bias),
             Cout, k, stride, padding, dilation, groups,
             pre_lrelu if pre_lrelu > 0.0 else -1.0,
             1 if post_lrelu > 0.0 else 0, float(post_lrelu),
@@ -225,8 +234,7 @@ def leaky_convtranspose1d(
         mfma = x.dtype == torch.bfloat16
         w = _convt_weight_mfma(weight, stride) if mfma else weight.contiguous()
         return ext.convtranspose1d_fused(
-            x.contiguous(), w,
-            bias.contiguous() if bias is not None else None,
+            x.contiguous(), w, _bias_f32(bias),
             Cout, k, stride, padding,
             pre_lrelu if pre_lrelu > 0.0 else -1.0,
         )
