@@ -240,6 +240,170 @@ __global__ __launch_bounds__(512) void conv1d_mfma_kernel(
 }
 
 // --------------------------------------------------------------------------
+// ConvTranspose1d, phase-merged: each block computes ALL s phases of a
+// BM x BN(v) tile, so each lane's stores are CONTIGUOUS s-runs
+// (t = v*s + r).  The strided per-phase epilogue of the generic kernel
+// collapses write efficiency s-fold; this kernel fixes it.
+// Weights: [s][KR][CoutP][CinP] flattened to taps tau = r*KR + m;
+// tap tau reads x[v - m] and accumulates into acc[r].
+// --------------------------------------------------------------------------
+template <int BM, int WGM, int WGN, int S, int KR>
+__global__ __launch_bounds__(512) void convt1d_merged_kernel(
+    const bf16* __restrict__ x,   // [B][Cin][Tin]
+    const bf16* __restrict__ w,   // [S*KR][CoutP][CinP]
+    const float* __restrict__ bias,
+    bf16* __restrict__ out,       // [B][Cout][Tout]
+    int Cin, int CinP, int Cout, int CoutP, long Tin, long Tout,
+    int pad, long Vn,             // v in [0, Vn)
+    float pre_slope) {
+  constexpr int BN = 32;            // v columns per block
+  constexpr int TAUC = 4;           // tap chunk (Ws LDS budget)
+  constexpr int XW = BN + KR + 6;   // window + halo (align pad)
+  const long v0 = (long)blockIdx.x * BN;
+  const int m0 = blockIdx.y * BM;
+  const int b = blockIdx.z;
+
+  __shared__ bf16 Ws[TAUC][BM][BK + 8];
+  __shared__ bf16 Xs[BK][XW + 8];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid / WGN;
+  const int wc = wid % WGN;
+  constexpr int WM = BM / WGM;
+  constexpr int WN = BN / WGN;   // 16 when WGN=2
+  constexpr int MT = WM / 16;
+  static_assert(WN == 16, "convT merged kernel assumes NT=1");
+  const int kl = lane >> 4;
+  const int il = lane & 15;
+
+  f32x4 acc[S][MT];
+#pragma unroll
+  for (int r = 0; r < S; ++r)
+#pragma unroll
+    for (int i = 0; i < MT; ++i) acc[r][i] = {0.f, 0.f, 0.f, 0.f};
+
+  const bf16* xb = x + (long)b * Cin * Tin;
+  const long w0 = v0 - (KR - 1);  // tap m reads x[v - m]
+  const bool x_interior = (w0 >= 0) && (w0 + XW <= Tin);
+
+  for (int c0 = 0; c0 < CinP; c0 += BK) {
+    const bool rows_ok = (c0 + BK) <= Cin;
+    // ---- stage X window (BK x XW), one b128 chunk per thread ---------
+    {
+      // BK*XW elements; XW+8 row pitch keeps 16B alignment (XW mult of 8
+      // not guaranteed -> use element loop with 8-chunks over rows)
+      for (int e = tid * 8; e < BK * 48; e += 512 * 8) {
+        int r = e / 48, c = e % 48;  // 48 >= XW rounded to 8
+        if (c >= XW) continue;
+        int ci = c0 + r;
+        bf16 vals[8];
+        long p = w0 + c;
+        if (rows_ok && x_interior && ci < Cin && p + 7 < Tin && p >= 0) {
+          *(ulonglong2*)vals = *(const ulonglong2*)&xb[(long)ci * Tin + p];
+          if (pre_slope >= 0.f) {
+#pragma unroll
+            for (int q = 0; q < 8; ++q)
+              vals[q] = f2bf(lrelu_(bf2f(vals[q]), pre_slope));
+          }
+        } else {
+#pragma unroll
+          for (int q = 0; q < 8; ++q) {
+            long pq = p + q;
+            float v = (ci < Cin && pq >= 0 && pq < Tin)
+                          ? bf2f(xb[(long)ci * Tin + pq]) : 0.f;
+            if (pre_slope >= 0.f) v = lrelu_(v, pre_slope);
+            vals[q] = f2bf(v);
+          }
+        }
+        *(ulonglong2*)&Xs[r][c] = *(ulonglong2*)vals;
+      }
+    }
+
+    constexpr int NTAU = S * KR;
+    static_assert(NTAU % TAUC == 0, "tap chunking must divide evenly");
+    static_assert(XW + 8 == 48, "staging loop assumes 48-elem row pitch");
+#pragma unroll
+    for (int cc = 0; cc < NTAU / TAUC; ++cc) {
+#pragma unroll
+      for (int tc = 0; tc < TAUC; ++tc) {
+        const long wbase = ((long)(cc * TAUC + tc) * CoutP + m0) * CinP + c0;
+#pragma unroll
+        for (int e = tid * 8; e < BM * BK; e += 512 * 8) {
+          int m = e / BK, kk = e % BK;
+          *(ulonglong2*)&Ws[tc][m][kk] =
+              *(const ulonglong2*)&w[wbase + (long)m * CinP + kk];
+        }
+      }
+      __syncthreads();
+
+#pragma unroll
+      for (int tc = 0; tc < TAUC; ++tc) {
+        constexpr int dummy = 0;
+        const int tau = cc * TAUC + tc;  // compile-time (both loops unroll)
+        const int r = tau / KR;          // phase -> static acc index
+        const int m = tau % KR;          // tap
+        const int toff = (KR - 1) - m;
+        (void)dummy;
+        bf16x8 b_frag;
+        const int ncol = wc * WN + il + toff;
+#pragma unroll
+        for (int q = 0; q < 8; ++q)
+          b_frag[q] = *(__bf16*)&Xs[kl * 8 + q][ncol];
+#pragma unroll
+        for (int mi = 0; mi < MT; ++mi) {
+          bf16x8 a_frag =
+              *(const bf16x8*)&Ws[tc][wr * WM + mi * 16 + il][kl * 8];
+          acc[r][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag, b_frag, acc[r][mi], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // ---- epilogue: per (row, v) store the S-run contiguously -----------
+  bf16* ob = out + (long)b * Cout * Tout;
+  const long v = v0 + wc * WN + il;
+  if (v < Vn) {
+#pragma unroll
+    for (int mi = 0; mi < MT; ++mi) {
+      const int row = m0 + wr * WM + mi * 16 + kl * 4;
+#pragma unroll
+      for (int rg = 0; rg < 4; ++rg) {
+        const int co = row + rg;
+        if (co >= Cout) continue;
+        const float bv = bias ? bias[co] : 0.f;
+        bf16 run[S];
+#pragma unroll
+        for (int r = 0; r < S; ++r) run[r] = f2bf(acc[r][mi][rg] + bv);
+        const long t0 = v * S - pad;
+        if (t0 >= 0 && t0 + S <= Tout) {
+          // t0 is even -> byte address 8B-aligned (pad is even for k=2s
+          // upsamplers); use two b64 stores for S=8, scalars otherwise
+          if constexpr (S == 8) {
+            *(unsigned long long*)&ob[(long)co * Tout + t0] =
+                ((unsigned long long*)run)[0];
+            *(unsigned long long*)&ob[(long)co * Tout + t0 + 4] =
+                ((unsigned long long*)run)[1];
+          } else {
+#pragma unroll
+            for (int r = 0; r < S; ++r) ob[(long)co * Tout + t0 + r] = run[r];
+          }
+        } else {
+#pragma unroll
+          for (int r = 0; r < S; ++r) {
+            const long t = t0 + r;
+            if (t >= 0 && t < Tout) ob[(long)co * Tout + t] = run[r];
+          }
+        }
+      }
+    }
+  }
+}
+
+// --------------------------------------------------------------------------
 // Naive direct conv (grouped / f32 / stride>1 fallback + oracle).
 // One thread per (b, co, t_out).
 // --------------------------------------------------------------------------
@@ -423,14 +587,35 @@ torch::Tensor convtranspose1d_fused(torch::Tensor x, torch::Tensor w_perm,
   if (mfma_ok) {
     const int kr_max = w_perm.size(1);
     const int CoutP = w_perm.size(2), CinP = w_perm.size(3);
-    // ALL phase GEMMs in one launch (phase folded into grid.y)
-    const long Nvirt_max = (Tout - 1 + padding) / stride + 1;
-    launch_conv_mfma((const bf16*)x.data_ptr(),
-                     (const bf16*)w_perm.data_ptr(), bias_p,
-                     (bf16*)out.data_ptr(), nullptr, B, Cin, CinP, Cout,
-                     CoutP, Tin, Tout, kr_max, -1, 0, Nvirt_max, stride, 0,
-                     (float)pre_lrelu, ACT_NONE, 0.f, cur_stream2(),
-                     (int)stride, (int)k, (int)padding, kr_max);
+    const long Vn = (Tout - 1 + padding) / stride + 1;
+    const float pre = (float)pre_lrelu;
+    hipStream_t st = cur_stream2();
+    const bf16* xp = (const bf16*)x.data_ptr();
+    const bf16* wp = (const bf16*)w_perm.data_ptr();
+    bf16* op = (bf16*)out.data_ptr();
+    bool merged = (k == 2 * stride) && kr_max == 2 &&
+                  (stride == 8 || stride == 2);
+#define LAUNCH_M(BM, WGM, WGN, S)                                           \
+  hipLaunchKernelGGL((convt1d_merged_kernel<BM, WGM, WGN, S, 2>),           \
+                     dim3(ceil_div(Vn, 32), ceil_div(Cout, BM), B),         \
+                     dim3(512), 0, st, xp, wp, bias_p, op, Cin, CinP,       \
+                     Cout, CoutP, Tin, Tout, (int)padding, Vn, pre)
+    if (merged && stride == 8) {
+      if (Cout >= 128) LAUNCH_M(128, 4, 2, 8);
+      else if (Cout >= 64) LAUNCH_M(64, 4, 2, 8);
+      else LAUNCH_M(32, 4, 2, 8);
+    } else if (merged && stride == 2) {
+      if (Cout >= 128) LAUNCH_M(128, 4, 2, 2);
+      else if (Cout >= 64) LAUNCH_M(64, 4, 2, 2);
+      else LAUNCH_M(32, 4, 2, 2);
+    } else {
+      // generic: all phase GEMMs in one launch (phase folded into grid.y)
+      launch_conv_mfma(xp, wp, bias_p, op, nullptr, B, Cin, CinP, Cout,
+                       CoutP, Tin, Tout, kr_max, -1, 0, Vn, stride, 0,
+                       pre, ACT_NONE, 0.f, st,
+                       (int)stride, (int)k, (int)padding, kr_max);
+    }
+#undef LAUNCH_M
   } else {
     const long n = B * Cout * Tout;
     const int threads = 256;
