@@ -593,8 +593,10 @@ torch::Tensor convtranspose1d_fused(torch::Tensor x, torch::Tensor w_perm,
     const bf16* xp = (const bf16*)x.data_ptr();
     const bf16* wp = (const bf16*)w_perm.data_ptr();
     bf16* op = (bf16*)out.data_ptr();
+    // merged kernel needs WM = BM/WGM >= 16 (MT >= 1); BM=32 with the
+    // fixed 4x2 wave grid gives MT=0 -> Cout<64 uses the generic path.
     bool merged = (k == 2 * stride) && kr_max == 2 &&
-                  (stride == 8 || stride == 2);
+                  (stride == 8 || stride == 2) && Cout >= 64;
 #define LAUNCH_M(BM, WGM, WGN, S)                                           \
   hipLaunchKernelGGL((convt1d_merged_kernel<BM, WGM, WGN, S, 2>),           \
                      dim3(ceil_div(Vn, 32), ceil_div(Cout, BM), B),         \
@@ -603,11 +605,11 @@ torch::Tensor convtranspose1d_fused(torch::Tensor x, torch::Tensor w_perm,
     if (merged && stride == 8) {
       if (Cout >= 128) LAUNCH_M(128, 4, 2, 8);
       else if (Cout >= 64) LAUNCH_M(64, 4, 2, 8);
-      else LAUNCH_M(32, 4, 2, 8);
+      else LAUNCH_M(64, 4, 2, 8); /* unreachable: Cout>=64 guard */
     } else if (merged && stride == 2) {
       if (Cout >= 128) LAUNCH_M(128, 4, 2, 2);
       else if (Cout >= 64) LAUNCH_M(64, 4, 2, 2);
-      else LAUNCH_M(32, 4, 2, 2);
+      else LAUNCH_M(64, 4, 2, 2); /* unreachable: Cout>=64 guard */
     } else {
       // generic: all phase GEMMs in one launch (phase folded into grid.y)
       launch_conv_mfma(xp, wp, bias_p, op, nullptr, B, Cin, CinP, Cout,
