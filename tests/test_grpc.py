@@ -1,0 +1,87 @@
+"""gRPC frontend tests (CPU, loopback): version, voice load (idempotent),
+voice info, synthesis options round-trip, streamed synthesis — mirroring
+the reference server behavior (crates/frontends/grpc/src/main.rs)."""
+
+import grpc
+import pytest
+
+from sonata_amd.frontends.grpc import create_server
+from sonata_amd.frontends.grpc.client import SonataGrpcClient
+from sonata_amd.frontends.grpc.proto import MESSAGES
+from sonata_amd.models import create_random_voice
+
+
+@pytest.fixture(scope="module")
+def voice_pack(tmp_path_factory):
+    d = tmp_path_factory.mktemp("grpc_voice")
+    return create_random_voice(str(d), "grpc_voice", quality="x_low",
+                               num_speakers=2)
+
+
+@pytest.fixture(scope="module")
+def running(voice_pack):
+    server, port, service = create_server(port=0, device="cpu")
+    server.start()
+    client = SonataGrpcClient(f"127.0.0.1:{port}")
+    yield client, voice_pack
+    client.close()
+    server.stop(grace=None)
+
+
+def test_version(running):
+    client, _ = running
+    v = client.GetSonataVersion(MESSAGES["Empty"]())
+    assert v.version
+
+
+def test_load_voice_idempotent(running):
+    client, pack = running
+    a = client.LoadVoice(MESSAGES["VoicePath"](config_path=pack))
+    b = client.LoadVoice(MESSAGES["VoicePath"](config_path=pack))
+    assert a.voice_id == b.voice_id
+    assert a.audio.sample_rate == 16000  # x_low preset
+    assert a.supports_streaming_output
+    assert dict(a.speakers) == {0: "spk0", 1: "spk1"}
+    info = client.GetVoiceInfo(
+        MESSAGES["VoiceIdentifier"](voice_id=a.voice_id))
+    assert info.voice_id == a.voice_id
+
+
+def test_unknown_voice_not_found(running):
+    client, _ = running
+    with pytest.raises(grpc.RpcError) as e:
+        client.GetVoiceInfo(MESSAGES["VoiceIdentifier"](voice_id="nope"))
+    assert e.value.code() == grpc.StatusCode.NOT_FOUND
+
+
+def test_synthesis_options_roundtrip(running):
+    client, pack = running
+    vid = client.LoadVoice(MESSAGES["VoicePath"](config_path=pack)).voice_id
+    opts = MESSAGES["SynthesisOptions"](speaker="spk1", length_scale=1.3)
+    got = client.SetSynthesisOptions(MESSAGES["VoiceSynthesisOptions"](
+        voice_id=vid, synthesis_options=opts))
+    assert got.speaker == "spk1"
+    assert abs(got.length_scale - 1.3) < 1e-6
+    back = client.GetSynthesisOptions(
+        MESSAGES["VoiceIdentifier"](voice_id=vid))
+    assert back.speaker == "spk1"
+
+
+def test_synthesize_utterance_stream(running):
+    client, pack = running
+    vid = client.LoadVoice(MESSAGES["VoicePath"](config_path=pack)).voice_id
+    results = list(client.SynthesizeUtterance(MESSAGES["Utterance"](
+        voice_id=vid, text="wˈʌn. tˈuː.")))
+    assert len(results) == 2  # one per sentence
+    for r in results:
+        assert len(r.wav_samples) > 500
+        assert r.rtf > 0
+
+
+def test_synthesize_realtime_stream(running):
+    client, pack = running
+    vid = client.LoadVoice(MESSAGES["VoicePath"](config_path=pack)).voice_id
+    chunks = list(client.SynthesizeUtteranceRealtime(MESSAGES["Utterance"](
+        voice_id=vid, text="hˈɛloʊ ðˈɛr ˈɛvɹiwˌʌn.")))
+    assert len(chunks) >= 1
+    assert all(len(c.wav_samples) > 0 for c in chunks)
