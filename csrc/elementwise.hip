@@ -249,3 +249,33 @@ torch::Tensor expand_states(torch::Tensor stats, torch::Tensor durs,
   });
   return out;
 }
+
+// --------------------------------------------------------------------------
+// mask_tail_: in-place zero of x[b, :, lens[b]:] for padded batches.
+// Makes ragged-batch decoding bit-equal to single-utterance decoding when
+// applied after each conv stage (padding region never feeds valid taps).
+// One block per (b, c) row; threads sweep the tail only.
+// --------------------------------------------------------------------------
+template <typename T>
+__global__ void mask_tail_kernel(T* __restrict__ x,
+                                 const int* __restrict__ lens, int C,
+                                 long T_len) {
+  const int b = blockIdx.y;
+  const int c = blockIdx.x;
+  const long lo = lens[b];
+  for (long t = lo + threadIdx.x; t < T_len; t += blockDim.x)
+    x[((long)b * C + c) * T_len + t] = T(0);
+}
+
+torch::Tensor mask_tail_(torch::Tensor x, torch::Tensor lens) {
+  TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(lens.scalar_type() == at::kInt && lens.is_cuda() &&
+              lens.is_contiguous() && lens.size(0) == x.size(0));
+  const long B = x.size(0), C = x.size(1), T_len = x.size(2);
+  DISPATCH_FT(x, "mask_tail_", {
+    hipLaunchKernelGGL(mask_tail_kernel<scalar_t>, dim3(C, B), dim3(256), 0,
+                       cur_stream(), (scalar_t*)x.data_ptr(),
+                       lens.data_ptr<int>(), (int)C, T_len);
+  });
+  return x;
+}

@@ -12,6 +12,7 @@ torch::Tensor prior_sample(torch::Tensor m, torch::Tensor logs,
                            double noise_scale);
 torch::Tensor expand_states(torch::Tensor stats, torch::Tensor durs,
                             long F_max);
+torch::Tensor mask_tail_(torch::Tensor x, torch::Tensor lens);
 // conv1d.hip
 torch::Tensor conv1d_fused(torch::Tensor x, torch::Tensor w_perm,
                            c10::optional<torch::Tensor> bias, long Cout,
@@ -30,6 +31,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_gate", &fused_gate, "WaveNet tanh*sigmoid gate");
   m.def("prior_sample", &prior_sample, "z=(m+eps*exp(logs)*ns)*mask");
   m.def("expand_states", &expand_states, "duration length-regulator gather");
+  m.def("mask_tail_", &mask_tail_, "in-place zero of x[b,:,lens[b]:]");
   m.def("conv1d_fused", &conv1d_fused, "MFMA conv1d with fused activations");
   m.def("convtranspose1d_fused", &convtranspose1d_fused,
         "MFMA transposed conv1d (phase-decomposed GEMMs)");

@@ -27,6 +27,7 @@ from ..ops import (
     layer_norm_ct,
     leaky_conv1d,
     leaky_convtranspose1d,
+    mask_tail_,
     prior_sample,
 )
 from .config import VitsArchitecture
@@ -550,18 +551,21 @@ class ResBlock1(nn.Module):
                           padding=(kernel_size - 1) // 2)
             )
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor,
+                lengths: Optional[torch.Tensor] = None) -> torch.Tensor:
         for c1, c2 in zip(self.convs1, self.convs2):
             xt = leaky_conv1d(
                 x, c1.weight, c1.bias,
                 padding=c1.padding[0], dilation=c1.dilation[0],
                 pre_lrelu=LRELU_SLOPE,
             )
+            mask_tail_(xt, lengths)
             # second conv fuses the residual add into its epilogue
             x = leaky_conv1d(
                 xt, c2.weight, c2.bias, padding=c2.padding[0],
                 pre_lrelu=LRELU_SLOPE, residual=x,
             )
+            mask_tail_(x, lengths)
         return x
 
 
@@ -593,18 +597,30 @@ class Generator(nn.Module):
             self.cond = None
 
     def forward(self, x: torch.Tensor,
-                g: Optional[torch.Tensor] = None) -> torch.Tensor:
+                g: Optional[torch.Tensor] = None,
+                lengths: Optional[torch.Tensor] = None) -> torch.Tensor:
+        # `lengths`: valid frame count per batch row.  Masking the padded
+        # tail after every stage makes ragged-batch output bit-equal to
+        # single-utterance output (the tail would otherwise leak into the
+        # last receptive-field window of valid audio).  All-equal lengths
+        # short-circuit inside mask_tail_.
+        if lengths is not None and x.shape[0] == 1:
+            lengths = None
         x = conv_mod(self.conv_pre, x)
         if g is not None and self.cond is not None:
             x = x + conv_mod(self.cond, g)
+        mask_tail_(x, lengths)
         for i, up in enumerate(self.ups):
             x = leaky_convtranspose1d(
                 x, up.weight, up.bias, stride=up.stride[0],
                 padding=up.padding[0], pre_lrelu=LRELU_SLOPE,
             )
+            if lengths is not None:
+                lengths = lengths * up.stride[0]
+                mask_tail_(x, lengths)
             xs = None
             for j in range(self.num_kernels):
-                out = self.resblocks[i * self.num_kernels + j](x)
+                out = self.resblocks[i * self.num_kernels + j](x, lengths)
                 xs = out if xs is None else xs + out
             x = xs / self.num_kernels
         x = leaky_conv1d(x, self.conv_post.weight, None, padding=3,
@@ -708,9 +724,12 @@ class VitsModel(nn.Module):
         return z, y_mask, g
 
     def decode(self, z: torch.Tensor, y_mask: torch.Tensor,
-               g: Optional[torch.Tensor] = None) -> torch.Tensor:
-        """latent frames -> waveform [B, 1, F*hop]."""
-        return self.dec(z * y_mask, g=g)
+               g: Optional[torch.Tensor] = None,
+               lengths: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """latent frames -> waveform [B, 1, F*hop].  `lengths` (valid
+        frames per row) makes ragged-batch decode match single-utterance
+        decode exactly; omit for uniform batches / streaming chunks."""
+        return self.dec(z * y_mask, g=g, lengths=lengths)
 
     @torch.no_grad()
     def infer(self, ids, lengths, sid=None, noise_scale=0.667,
@@ -718,6 +737,6 @@ class VitsModel(nn.Module):
         z, y_mask, g = self.infer_encoder(
             ids, lengths, sid, noise_scale, length_scale, noise_w, generators
         )
-        audio = self.decode(z, y_mask, g)
         y_lengths = y_mask.squeeze(1).sum(-1).long()
+        audio = self.decode(z, y_mask, g, lengths=y_lengths)
         return audio, y_lengths * self.arch.hop_length

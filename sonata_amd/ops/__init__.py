@@ -63,4 +63,5 @@ from .functional import (  # noqa: F401,E402
     leaky_conv1d,
     leaky_convtranspose1d,
     conv_mod,
+    mask_tail_,
 )

@@ -255,3 +255,25 @@ def conv_mod(
         dilation=mod.dilation[0], groups=mod.groups,
         pre_lrelu=pre_lrelu, post_lrelu=post_lrelu,
     )
+
+
+# --------------------------------------------------------------------------- #
+# ragged-batch tail masking
+# --------------------------------------------------------------------------- #
+def mask_tail_(x: torch.Tensor, lengths: Optional[torch.Tensor]) -> torch.Tensor:
+    """In-place zero of x[b, :, lengths[b]:].  Applied between decoder
+    stages so padded-batch synthesis is numerically identical to
+    single-utterance synthesis (padding never feeds valid conv taps).
+    No-op when lengths is None or nothing is padded."""
+    if lengths is None:
+        return x
+    T = x.shape[-1]
+    if bool((lengths >= T).all()):
+        return x
+    if use_hip(x):
+        ext = hip_ext(required=True)
+        return ext.mask_tail_(x, lengths.to(device=x.device,
+                                            dtype=torch.int32).contiguous())
+    idx = torch.arange(T, device=x.device)
+    pad = idx.unsqueeze(0) >= lengths.to(x.device).unsqueeze(1)  # [B, T]
+    return x.masked_fill_(pad.unsqueeze(1), 0)
