@@ -1,0 +1,159 @@
+"""C ABI frontend tests: drive libsonata_amd.so through ctypes exactly as
+a C caller would (callback protocol, config round-trip, speak-to-file) —
+mirroring the reference C API semantics (crates/frontends/capi)."""
+
+import ctypes as C
+import os
+
+import pytest
+
+LIB = os.path.join(os.path.dirname(__file__), "..", "sonata_amd",
+                   "frontends", "libsonata_amd.so")
+
+pytestmark = pytest.mark.skipif(
+    not os.path.exists(LIB), reason="libsonata_amd.so not built")
+
+
+class ExternError(C.Structure):
+    _fields_ = [("code", C.c_int32), ("message", C.c_char_p)]
+
+
+class SynthesisEvent(C.Structure):
+    _fields_ = [("event_type", C.c_int32),
+                ("error_ptr", C.POINTER(ExternError)),
+                ("len", C.c_int64),
+                ("data", C.POINTER(C.c_uint8))]
+
+
+class AudioInfo(C.Structure):
+    _fields_ = [("sample_rate", C.c_uint32), ("num_channels", C.c_uint32),
+                ("sample_width", C.c_uint32)]
+
+
+class PiperSynthConfig(C.Structure):
+    _fields_ = [("speaker", C.c_uint32), ("length_scale", C.c_float),
+                ("noise_scale", C.c_float), ("noise_w", C.c_float)]
+
+
+CALLBACK = C.CFUNCTYPE(C.c_uint8, SynthesisEvent)
+
+
+class SynthesisParams(C.Structure):
+    _fields_ = [("mode", C.c_int32), ("rate", C.c_uint8),
+                ("volume", C.c_uint8), ("pitch", C.c_uint8),
+                ("appended_silence_ms", C.c_uint32),
+                ("callback", CALLBACK), ("nonblocking", C.c_uint8)]
+
+
+@pytest.fixture(scope="module")
+def lib():
+    lib = C.CDLL(LIB)
+    lib.libsonataLoadVoiceFromConfigPath.restype = C.c_void_p
+    lib.libsonataLoadVoiceFromConfigPath.argtypes = [
+        C.c_char_p, C.POINTER(ExternError)]
+    lib.libsonataUnloadSonataVoice.argtypes = [C.c_void_p]
+    lib.libsonataGetAudioInfo.argtypes = [
+        C.c_void_p, C.POINTER(AudioInfo), C.POINTER(ExternError)]
+    lib.libsonataGetPiperDefaultSynthConfig.restype = \
+        C.POINTER(PiperSynthConfig)
+    lib.libsonataGetPiperDefaultSynthConfig.argtypes = [
+        C.c_void_p, C.POINTER(ExternError)]
+    lib.libsonataSetPiperSynthConfig.argtypes = [
+        C.c_void_p, PiperSynthConfig, C.POINTER(ExternError)]
+    lib.libsonataSpeak.argtypes = [
+        C.c_void_p, C.c_char_p, SynthesisParams, C.POINTER(ExternError)]
+    lib.libsonataSpeakToFile.restype = C.c_uint8
+    lib.libsonataSpeakToFile.argtypes = [
+        C.c_void_p, C.c_char_p, SynthesisParams, C.c_char_p,
+        C.POINTER(ExternError)]
+    lib.libsonataFreePiperSynthConfig.argtypes = [
+        C.POINTER(PiperSynthConfig)]
+    return lib
+
+
+@pytest.fixture(scope="module")
+def voice(lib, tmp_path_factory):
+    from sonata_amd.models import create_random_voice
+
+    d = tmp_path_factory.mktemp("capi_voice")
+    pack = create_random_voice(str(d), "capi_voice", quality="x_low")
+    err = ExternError()
+    os.environ["SONATA_DEVICE"] = "cpu"
+    h = lib.libsonataLoadVoiceFromConfigPath(pack.encode(), C.byref(err))
+    assert err.code == 0, err.message
+    assert h
+    yield h
+    lib.libsonataUnloadSonataVoice(h)
+
+
+def test_load_bad_path(lib):
+    err = ExternError()
+    h = lib.libsonataLoadVoiceFromConfigPath(b"/nope/voice.json",
+                                             C.byref(err))
+    assert not h
+    assert err.code == 17  # FAILED_TO_LOAD_RESOURCE
+
+
+def test_audio_info(lib, voice):
+    err, info = ExternError(), AudioInfo()
+    lib.libsonataGetAudioInfo(voice, C.byref(info), C.byref(err))
+    assert err.code == 0
+    assert info.sample_rate == 16000
+    assert info.num_channels == 1 and info.sample_width == 2
+
+
+def test_synth_config_roundtrip(lib, voice):
+    err = ExternError()
+    cfg_p = lib.libsonataGetPiperDefaultSynthConfig(voice, C.byref(err))
+    assert err.code == 0
+    cfg = cfg_p.contents
+    cfg.length_scale = 1.25
+    lib.libsonataSetPiperSynthConfig(voice, cfg, C.byref(err))
+    assert err.code == 0
+    back = lib.libsonataGetPiperDefaultSynthConfig(voice, C.byref(err))
+    assert abs(back.contents.length_scale - 1.25) < 1e-6
+    lib.libsonataFreePiperSynthConfig(cfg_p)
+    lib.libsonataFreePiperSynthConfig(back)
+
+
+def test_speak_callback_protocol(lib, voice):
+    events = []
+
+    @CALLBACK
+    def cb(ev):
+        if ev.event_type == 0:  # SPEECH
+            events.append(bytes(C.cast(
+                ev.data, C.POINTER(C.c_uint8 * ev.len)).contents))
+        else:
+            events.append(ev.event_type)
+        return 0
+
+    err = ExternError()
+    params = SynthesisParams(mode=1, rate=0, volume=0, pitch=0,
+                             appended_silence_ms=0, callback=cb,
+                             nonblocking=0)
+    lib.libsonataSpeak(voice, "hˈɛloʊ wˈɜːld.".encode(), params,
+                       C.byref(err))
+    assert err.code == 0
+    assert events[-1] == 1  # FINISHED
+    chunks = [e for e in events if isinstance(e, bytes)]
+    assert len(chunks) >= 1 and sum(len(c) for c in chunks) > 500
+
+
+def test_speak_invalid_mode(lib, voice):
+    err = ExternError()
+    params = SynthesisParams(mode=9, callback=CALLBACK(lambda ev: 0),
+                             nonblocking=0)
+    lib.libsonataSpeak(voice, b"x.", params, C.byref(err))
+    assert err.code == 16  # INVALID_SYNTHESIS_MODE
+
+
+def test_speak_to_file(lib, voice, tmp_path):
+    err = ExternError()
+    out = str(tmp_path / "c.wav").encode()
+    params = SynthesisParams(mode=1, callback=CALLBACK(lambda ev: 0),
+                             nonblocking=0)
+    ok = lib.libsonataSpeakToFile(voice, "tˈɛst sˈɛntəns.".encode(),
+                                  params, out, C.byref(err))
+    assert ok == 1 and err.code == 0
+    assert open(out, "rb").read(4) == b"RIFF"
