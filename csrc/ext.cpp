@@ -24,6 +24,19 @@ torch::Tensor convtranspose1d_fused(torch::Tensor x, torch::Tensor w_perm,
                                     c10::optional<torch::Tensor> bias,
                                     long Cout, long k, long stride,
                                     long padding, double pre_lrelu);
+// conv1d_cl.hip (channel-last)
+torch::Tensor conv1d_cl_fused(torch::Tensor x, torch::Tensor w_perm,
+                              c10::optional<torch::Tensor> bias, long Cout,
+                              long k, long padding, long dilation,
+                              double pre_lrelu, long act_mode,
+                              double post_slope,
+                              c10::optional<torch::Tensor> residual,
+                              c10::optional<torch::Tensor> out_lens);
+torch::Tensor convtranspose1d_cl_fused(torch::Tensor x, torch::Tensor w_perm,
+                                       c10::optional<torch::Tensor> bias,
+                                       long Cout, long k, long stride,
+                                       long padding, double pre_lrelu,
+                                       c10::optional<torch::Tensor> out_lens);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "sonata_amd hand-written CDNA4 (gfx950) kernels";
@@ -35,4 +48,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv1d_fused", &conv1d_fused, "MFMA conv1d with fused activations");
   m.def("convtranspose1d_fused", &convtranspose1d_fused,
         "MFMA transposed conv1d (phase-decomposed GEMMs)");
+  m.def("conv1d_cl_fused", &conv1d_cl_fused,
+        "channel-last MFMA conv1d, fused act/residual/mask");
+  m.def("convtranspose1d_cl_fused", &convtranspose1d_cl_fused,
+        "channel-last MFMA transposed conv1d, phase-merged");
 }

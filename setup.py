@@ -21,6 +21,7 @@ ext = CUDAExtension(
         "csrc/ext.cpp",
         "csrc/elementwise.hip",
         "csrc/conv1d.hip",
+        "csrc/conv1d_cl.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -26,7 +26,9 @@ from ..ops import (
     fused_gate,
     layer_norm_ct,
     leaky_conv1d,
+    leaky_conv1d_cl,
     leaky_convtranspose1d,
+    leaky_convtranspose1d_cl,
     mask_tail_,
     prior_sample,
 )
@@ -568,6 +570,22 @@ class ResBlock1(nn.Module):
             mask_tail_(x, lengths)
         return x
 
+    def forward_cl(self, x: torch.Tensor,
+                   lengths: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Channel-last ([B,T,C]) path: fused-mask MFMA kernels
+        (csrc/conv1d_cl.hip); numerically equivalent to forward()."""
+        for c1, c2 in zip(self.convs1, self.convs2):
+            xt = leaky_conv1d_cl(
+                x, c1.weight, c1.bias,
+                padding=c1.padding[0], dilation=c1.dilation[0],
+                pre_lrelu=LRELU_SLOPE, out_lens=lengths,
+            )
+            x = leaky_conv1d_cl(
+                xt, c2.weight, c2.bias, padding=c2.padding[0],
+                pre_lrelu=LRELU_SLOPE, residual=x, out_lens=lengths,
+            )
+        return x
+
 
 class Generator(nn.Module):
     def __init__(self, arch: VitsArchitecture, gin_channels: int = 0):
@@ -606,6 +624,10 @@ class Generator(nn.Module):
         # short-circuit inside mask_tail_.
         if lengths is not None and x.shape[0] == 1:
             lengths = None
+        from ..ops import use_hip
+
+        if use_hip(x):
+            return self._forward_cl(x, g, lengths)
         x = conv_mod(self.conv_pre, x)
         if g is not None and self.cond is not None:
             x = x + conv_mod(self.cond, g)
@@ -626,6 +648,41 @@ class Generator(nn.Module):
         x = leaky_conv1d(x, self.conv_post.weight, None, padding=3,
                          pre_lrelu=LRELU_SLOPE)
         return torch.tanh(x)
+
+    def _forward_cl(self, x: torch.Tensor,
+                    g: Optional[torch.Tensor],
+                    lengths: Optional[torch.Tensor]) -> torch.Tensor:
+        """MI355X serving path: activations kept channel-last [B,T,C]
+        end-to-end so every conv is a k-contiguous MFMA GEMM; ragged-batch
+        masking fused into each conv epilogue."""
+        x = x.transpose(1, 2).contiguous()  # [B, F, C]
+        x = leaky_conv1d_cl(x, self.conv_pre.weight, self.conv_pre.bias,
+                            padding=3, out_lens=lengths)
+        if g is not None and self.cond is not None:
+            x = x + conv_mod(self.cond, g).transpose(1, 2)
+            if lengths is not None:  # re-mask: cond bias hit padded rows
+                idx = torch.arange(x.shape[1], device=x.device)
+                x = x.masked_fill(
+                    (idx.unsqueeze(0) >= lengths.unsqueeze(1)).unsqueeze(-1),
+                    0)
+        for i, up in enumerate(self.ups):
+            if lengths is not None:
+                lengths = lengths * up.stride[0]
+            x = leaky_convtranspose1d_cl(
+                x, up.weight, up.bias, stride=up.stride[0],
+                padding=up.padding[0], pre_lrelu=LRELU_SLOPE,
+                out_lens=lengths,
+            )
+            xs = None
+            for j in range(self.num_kernels):
+                out = self.resblocks[i * self.num_kernels + j].forward_cl(
+                    x, lengths)
+                xs = out if xs is None else xs + out
+            x = xs / self.num_kernels
+        x = leaky_conv1d_cl(x, self.conv_post.weight, None, padding=3,
+                            pre_lrelu=LRELU_SLOPE, post_tanh=True,
+                            out_lens=lengths)
+        return x.transpose(1, 2)  # [B, 1, T]
 
 
 # --------------------------------------------------------------------------- #

@@ -64,4 +64,6 @@ from .functional import (  # noqa: F401,E402
     leaky_convtranspose1d,
     conv_mod,
     mask_tail_,
+    leaky_conv1d_cl,
+    leaky_convtranspose1d_cl,
 )

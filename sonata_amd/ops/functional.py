@@ -277,3 +277,93 @@ def mask_tail_(x: torch.Tensor, lengths: Optional[torch.Tensor]) -> torch.Tensor
     idx = torch.arange(T, device=x.device)
     pad = idx.unsqueeze(0) >= lengths.to(x.device).unsqueeze(1)  # [B, T]
     return x.masked_fill_(pad.unsqueeze(1), 0)
+
+
+# --------------------------------------------------------------------------- #
+# channel-last conv ops (HiFi-GAN decode path)
+# --------------------------------------------------------------------------- #
+def _lens_i32(lengths: Optional[torch.Tensor], device) -> Optional[torch.Tensor]:
+    if lengths is None:
+        return None
+    return lengths.to(device=device, dtype=torch.int32).contiguous()
+
+
+def leaky_conv1d_cl(
+    x: torch.Tensor,  # [B, T, C] channel-last
+    weight: torch.Tensor,  # nn.Conv1d weight [Cout, Cin, k]
+    bias: Optional[torch.Tensor],
+    padding: int = 0,
+    dilation: int = 1,
+    pre_lrelu: float = 0.0,
+    post_lrelu: float = 0.0,
+    post_tanh: bool = False,
+    residual: Optional[torch.Tensor] = None,
+    out_lens: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """Channel-last stride-1 Conv1d with fused input/output LeakyReLU (or
+    tanh), residual add and ragged-batch row masking.  The MI355X-native
+    layout: Cin contiguous makes both MFMA operands k-contiguous
+    (csrc/conv1d_cl.hip)."""
+    if use_hip(x):
+        ext = hip_ext(required=True)
+        Cout, _, k = weight.shape
+        act = 2 if post_tanh else (1 if post_lrelu > 0.0 else 0)
+        return ext.conv1d_cl_fused(
+            x.contiguous(), _conv_weight_mfma(weight), _bias_f32(bias),
+            Cout, k, padding, dilation,
+            pre_lrelu if pre_lrelu > 0.0 else -1.0,
+            act, float(post_lrelu),
+            residual.contiguous() if residual is not None else None,
+            _lens_i32(out_lens, x.device),
+        )
+    # torch oracle: transpose to channel-first
+    xc = x.transpose(1, 2)
+    if pre_lrelu > 0.0:
+        xc = F.leaky_relu(xc, pre_lrelu)
+    y = F.conv1d(xc, weight, bias, padding=padding, dilation=dilation)
+    if post_tanh:
+        y = torch.tanh(y)
+    elif post_lrelu > 0.0:
+        y = F.leaky_relu(y, post_lrelu)
+    y = y.transpose(1, 2)
+    if residual is not None:
+        y = y + residual
+    if out_lens is not None:
+        idx = torch.arange(y.shape[1], device=y.device)
+        y = y.masked_fill(
+            (idx.unsqueeze(0) >= out_lens.to(y.device).unsqueeze(1))
+            .unsqueeze(-1), 0)
+    return y
+
+
+def leaky_convtranspose1d_cl(
+    x: torch.Tensor,  # [B, T, C]
+    weight: torch.Tensor,  # nn.ConvTranspose1d weight [Cin, Cout, k]
+    bias: Optional[torch.Tensor],
+    stride: int,
+    padding: int,
+    pre_lrelu: float = 0.0,
+    out_lens: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """Channel-last ConvTranspose1d (k = 2*stride HiFi-GAN upsampler),
+    phase-merged MFMA kernel with fused input LeakyReLU + row masking."""
+    if use_hip(x):
+        ext = hip_ext(required=True)
+        Cin, Cout, k = weight.shape
+        return ext.convtranspose1d_cl_fused(
+            x.contiguous(), _convt_weight_mfma(weight, stride),
+            _bias_f32(bias), Cout, k, stride, padding,
+            pre_lrelu if pre_lrelu > 0.0 else -1.0,
+            _lens_i32(out_lens, x.device),
+        )
+    xc = x.transpose(1, 2)
+    if pre_lrelu > 0.0:
+        xc = F.leaky_relu(xc, pre_lrelu)
+    y = F.conv_transpose1d(xc, weight, bias, stride=stride, padding=padding)
+    y = y.transpose(1, 2)
+    if out_lens is not None:
+        idx = torch.arange(y.shape[1], device=y.device)
+        y = y.masked_fill(
+            (idx.unsqueeze(0) >= out_lens.to(y.device).unsqueeze(1))
+            .unsqueeze(-1), 0)
+    return y

@@ -297,3 +297,93 @@ def test_ragged_batch_matches_single(dev):
         assert len(ref) == len(got)
         # bf16 kernels; identical launch shapes modulo batch -> tight tol
         assert float(abs(ref - got).max()) < 2e-2
+
+
+@pytest.mark.parametrize("Cin,Cout,k,dil", [
+    (128, 128, 3, 1), (128, 128, 11, 5), (256, 256, 7, 3),
+    (64, 64, 3, 1), (32, 32, 11, 5), (192, 512, 7, 1), (32, 1, 7, 1),
+])
+def test_conv1d_cl(dev, Cin, Cout, k, dil):
+    from sonata_amd.ops.functional import leaky_conv1d_cl
+
+    torch.manual_seed(Cin + k + dil)
+    B, T = 3, 211
+    pad = (k - 1) * dil // 2
+    x = (torch.randn(B, T, Cin) / 4).to(torch.bfloat16)
+    w = (torch.randn(Cout, Cin, k) / (Cin * k) ** 0.5).to(torch.bfloat16)
+    bias = torch.randn(Cout) / 10
+    res = (torch.randn(B, T, Cout) / 4).to(torch.bfloat16)
+    lens = torch.tensor([T, 150, 37])
+    got = leaky_conv1d_cl(x.to(dev), w.to(dev), bias.to(dev), padding=pad,
+                          dilation=dil, pre_lrelu=0.1, post_lrelu=0.2,
+                          residual=res.to(dev), out_lens=lens.to(dev))
+    ref = leaky_conv1d_cl(x.float(), w.float(), bias, padding=pad,
+                          dilation=dil, pre_lrelu=0.1, post_lrelu=0.2,
+                          residual=res.float(), out_lens=lens)
+    assert got.shape == ref.shape
+    err = _rel_err(got, ref)
+    assert err < 0.02, f"conv_cl parity {err}"
+    # masked rows are exactly zero
+    assert got[1, 150:].abs().max().item() == 0
+    assert got[2, 37:].abs().max().item() == 0
+
+
+def test_conv1d_cl_tanh(dev):
+    from sonata_amd.ops.functional import leaky_conv1d_cl
+
+    torch.manual_seed(5)
+    x = (torch.randn(2, 77, 32) / 4).to(torch.bfloat16)
+    w = (torch.randn(1, 32, 7) / 15).to(torch.bfloat16)
+    got = leaky_conv1d_cl(x.to(dev), w.to(dev), None, padding=3,
+                          pre_lrelu=0.1, post_tanh=True)
+    ref = leaky_conv1d_cl(x.float(), w.float(), None, padding=3,
+                          pre_lrelu=0.1, post_tanh=True)
+    assert _rel_err(got, ref) < 0.02
+
+
+@pytest.mark.parametrize("Cin,Cout,k,s", [
+    (512, 256, 16, 8), (256, 128, 16, 8), (128, 64, 4, 2), (64, 32, 4, 2),
+    (96, 48, 16, 8),
+])
+def test_convtranspose1d_cl(dev, Cin, Cout, k, s):
+    from sonata_amd.ops.functional import leaky_convtranspose1d_cl
+
+    torch.manual_seed(Cin + k)
+    pad = (k - s) // 2
+    B, T = 2, 97
+    x = (torch.randn(B, T, Cin) / 4).to(torch.bfloat16)
+    w = (torch.randn(Cin, Cout, k) / (Cin * k) ** 0.5).to(torch.bfloat16)
+    bias = torch.randn(Cout) / 10
+    lens = torch.tensor([T * s, 40 * s])
+    got = leaky_convtranspose1d_cl(x.to(dev), w.to(dev), bias.to(dev), s,
+                                   pad, pre_lrelu=0.1, out_lens=lens.to(dev))
+    ref = leaky_convtranspose1d_cl(x.float(), w.float(), bias, s, pad,
+                                   pre_lrelu=0.1, out_lens=lens)
+    assert got.shape == ref.shape
+    err = _rel_err(got, ref)
+    assert err < 0.02, f"convT_cl parity {err}"
+    assert got[1, 40 * s:].abs().max().item() == 0
+
+
+def test_generator_cl_matches_oracle(dev):
+    """Whole HiFi-GAN generator: channel-last HIP path vs fp32 CPU oracle
+    with ragged lengths."""
+    from sonata_amd.models.config import QUALITY_PRESETS, VitsArchitecture
+    from sonata_amd.models.vits import Generator
+
+    torch.manual_seed(11)
+    arch = VitsArchitecture(**QUALITY_PRESETS["x_low"]["arch"])
+    gen = Generator(arch).eval()
+    B, F = 3, 61
+    z = (torch.randn(B, arch.inter_channels, F) / 2)
+    lens = torch.tensor([F, 40, 23])
+    with torch.no_grad():
+        ref = gen(z.float(), lengths=lens)
+        got = gen.to(dev, torch.bfloat16)(z.to(dev, torch.bfloat16),
+                                          lengths=lens.to(dev))
+    err = _rel_err(got, ref)
+    assert err < 0.05, f"generator cl parity {err}"
+    hop = 1
+    for r in arch.upsample_rates:
+        hop *= r
+    assert got[1, :, 40 * hop:].abs().max().item() == 0

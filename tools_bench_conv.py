@@ -53,3 +53,60 @@ for name, Cin, Cout, k, s, T in [("up0 512->256", 512, 256, 16, 8, F),
     fl = 2*B*Cin*Cout*(k//s)*T*s
     gb = 2*B*(T*Cin + T*s*Cout)
     print(f"{name:18s} T={T:6d} {dt*1e3:8.3f} ms  {fl/dt/1e12:7.1f} TF  {gb/dt/1e9:7.0f} GB/s")
+
+# ---- channel-last kernels + T-sweep diagnostics --------------------------
+from sonata_amd.ops.functional import leaky_conv1d_cl, leaky_convtranspose1d_cl
+
+print("--- channel-last ---")
+for name, Cin, Cout, T, k, dil, pre in shapes:
+    x = (torch.randn(B, T, Cin)/4).to(torch.bfloat16).to(dev)
+    w = (torch.randn(Cout, Cin, k)/(Cin*k)**0.5).to(torch.bfloat16).to(dev)
+    bias = (torch.randn(Cout)/10).to(dev)
+    pad = (k-1)*dil//2
+    for _ in range(3):
+        y = leaky_conv1d_cl(x, w, bias, padding=pad, dilation=dil, pre_lrelu=pre)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter(); N = 10
+    for _ in range(N):
+        y = leaky_conv1d_cl(x, w, bias, padding=pad, dilation=dil, pre_lrelu=pre)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter()-t0)/N
+    fl = 2*B*Cin*Cout*k*T
+    gb = 2*B*T*(Cin+Cout)
+    print(f"cl {name:15s} T={T:6d} {dt*1e3:8.3f} ms  {fl/dt/1e12:7.1f} TF  {gb/dt/1e9:7.0f} GB/s")
+
+for name, Cin, Cout, k, s, T in [("up0 512->256", 512, 256, 16, 8, F),
+                                 ("up1 256->128", 256, 128, 16, 8, 8*F),
+                                 ("up2 128->64", 128, 64, 4, 2, 64*F),
+                                 ("up3 64->32", 64, 32, 4, 2, 128*F)]:
+    x = (torch.randn(B, T, Cin)/4).to(torch.bfloat16).to(dev)
+    w = (torch.randn(Cin, Cout, k)/(Cin*k)**0.5).to(torch.bfloat16).to(dev)
+    bias = (torch.randn(Cout)/10).to(dev)
+    for _ in range(3):
+        y = leaky_convtranspose1d_cl(x, w, bias, s, (k-s)//2, pre_lrelu=0.1)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter(); N = 10
+    for _ in range(N):
+        y = leaky_convtranspose1d_cl(x, w, bias, s, (k-s)//2, pre_lrelu=0.1)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter()-t0)/N
+    fl = 2*B*Cin*Cout*(k//s)*T*s
+    gb = 2*B*(T*Cin + T*s*Cout)
+    print(f"cl {name:15s} T={T:6d} {dt*1e3:8.3f} ms  {fl/dt/1e12:7.1f} TF  {gb/dt/1e9:7.0f} GB/s")
+
+# channel-FIRST convT T-sweep: localize the up1 anomaly (34 TF at T=2048)
+print("--- convT chfirst T-sweep (Cin=256 Cout=128 k16 s8) ---")
+for T in [256, 512, 1024, 2048]:
+    x = (torch.randn(B, 256, T)/4).to(torch.bfloat16).to(dev)
+    w = (torch.randn(256, 128, 16)/(256*16)**0.5).to(torch.bfloat16).to(dev)
+    bias = (torch.randn(128)/10).to(dev)
+    for _ in range(3):
+        y = leaky_convtranspose1d(x, w, bias, 8, 4, pre_lrelu=0.1)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter(); N = 10
+    for _ in range(N):
+        y = leaky_convtranspose1d(x, w, bias, 8, 4, pre_lrelu=0.1)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter()-t0)/N
+    fl = 2*B*256*128*2*T*8
+    print(f"cf up1 T={T:6d} {dt*1e3:8.3f} ms  {fl/dt/1e12:7.1f} TF")
