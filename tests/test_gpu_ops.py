@@ -381,9 +381,13 @@ def test_generator_cl_matches_oracle(dev):
         ref = gen(z.float(), lengths=lens)
         got = gen.to(dev, torch.bfloat16)(z.to(dev, torch.bfloat16),
                                           lengths=lens.to(dev))
-    err = _rel_err(got, ref)
-    assert err < 0.05, f"generator cl parity {err}"
     hop = 1
     for r in arch.upsample_rates:
         hop *= r
+    # compare the VALID region per batch row: the channel-last path zeroes
+    # the padded tail (fused mask) while the channel-first oracle leaves
+    # conv boundary bleed there — only [0, len*hop) is meaningful audio.
+    for b, ln in enumerate(lens.tolist()):
+        err = _rel_err(got[b, :, : ln * hop], ref[b, :, : ln * hop])
+        assert err < 0.05, f"generator cl parity row {b}: {err}"
     assert got[1, :, 40 * hop:].abs().max().item() == 0
