@@ -403,8 +403,10 @@ torch::Tensor conv1d_cl_fused(torch::Tensor x, torch::Tensor w_perm,
                      (int)Cout, CoutP, Tin, Tout, (int)k, (int)dilation,    \
                      (int)padding, (float)pre_lrelu, (int)act_mode,         \
                      (float)post_slope)
-  if (Cout >= 128) LAUNCH_CL(128, 128, 4, 2, 4);
-  else if (Cout >= 64) LAUNCH_CL(128, 64, 4, 2, 4);
+  // BM=256 t-tiles: skinny-K GEMMs (K = Cin*k <= 1408) are barrier-
+  // amortization-bound, so double the MFMA work per barrier window.
+  if (Cout >= 128) LAUNCH_CL(256, 128, 4, 2, 2);
+  else if (Cout >= 64) LAUNCH_CL(256, 64, 4, 2, 4);
   else LAUNCH_CL(256, 32, 4, 2, 4);
 #undef LAUNCH_CL
   return out;

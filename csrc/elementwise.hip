@@ -48,6 +48,62 @@ __global__ void layer_norm_ct_kernel(const T* __restrict__ x,
   }
 }
 
+// Lane-split variant for small B*T (encoder: ~8k tokens underfills the
+// 256-CU chip with one thread per token).  SPLIT lanes cooperate on one
+// token: lane = part*(64/SPLIT) + token, so each 64-lane wave holds
+// 64/SPLIT tokens; stats reduced with SPLIT-1 shfl_xor rounds.  Each
+// part owns a contiguous channel chunk (coalesced along T within a
+// part-group).
+template <typename T, bool HAS_RES, int SPLIT>
+__global__ void layer_norm_ct_split_kernel(const T* __restrict__ x,
+                                           const T* __restrict__ res,
+                                           const float* __restrict__ gamma,
+                                           const float* __restrict__ beta,
+                                           T* __restrict__ out, int C,
+                                           long T_len, float eps,
+                                           long n_bt) {
+  constexpr int TOK = 64 / SPLIT;  // tokens per wave
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int part = lane / TOK;
+  const int tok = lane % TOK;
+  const long bt = ((long)blockIdx.x * (blockDim.x >> 6) + wid) * TOK + tok;
+  const bool live = bt < n_bt;
+  const long b = live ? bt / T_len : 0;
+  const long t = live ? bt % T_len : 0;
+  const long base = (b * C) * T_len + t;
+  const T* xp = x + base;
+  const T* rp = HAS_RES ? res + base : nullptr;
+  const int cq = (C + SPLIT - 1) / SPLIT;
+  const int c_lo = part * cq;
+  const int c_hi = min(c_lo + cq, C);
+  float s = 0.f, ss = 0.f;
+  if (live) {
+    for (int c = c_lo; c < c_hi; ++c) {
+      float v = ld_f(xp + (long)c * T_len);
+      if (HAS_RES) v += ld_f(rp + (long)c * T_len);
+      s += v;
+      ss += v * v;
+    }
+  }
+  // combine the SPLIT part-lanes of each token (they sit TOK apart)
+#pragma unroll
+  for (int d = TOK; d < 64; d <<= 1) {
+    s += __shfl_xor(s, d, 64);
+    ss += __shfl_xor(ss, d, 64);
+  }
+  if (!live) return;
+  const float mean = s / C;
+  const float var = ss / C - mean * mean;
+  const float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+  T* op = out + base;
+  for (int c = c_lo; c < c_hi; ++c) {
+    float v = ld_f(xp + (long)c * T_len);
+    if (HAS_RES) v += ld_f(rp + (long)c * T_len);
+    st_f(op + (long)c * T_len, (v - mean) * rstd * gamma[c] + beta[c]);
+  }
+}
+
 // --------------------------------------------------------------------------
 // fused_gate: x [B, 2C, T] (+ optional g) -> tanh·sigmoid gate [B, C, T]
 // --------------------------------------------------------------------------
@@ -156,30 +212,61 @@ torch::Tensor layer_norm_ct(torch::Tensor x, c10::optional<torch::Tensor> res,
   auto out = torch::empty_like(x);
   auto gamma_f = gamma.to(at::kFloat).contiguous();
   auto beta_f = beta.to(at::kFloat).contiguous();
-  // small blocks: B*T/threads workgroups must still fill 256 CUs for the
-  // short sequences of this graph
-  const int threads = 64;
-  dim3 grid(ceil_div(B * T, threads));
-  DISPATCH_FT(x, "layer_norm_ct", {
-    if (res.has_value()) {
-      TORCH_CHECK(res->sizes() == x.sizes() && res->is_contiguous());
-      hipLaunchKernelGGL((layer_norm_ct_kernel<scalar_t, true>), grid,
-                         dim3(threads), 0, cur_stream(),
-                         (const scalar_t*)x.data_ptr(),
-                         (const scalar_t*)res->data_ptr(),
-                         gamma_f.data_ptr<float>(), beta_f.data_ptr<float>(),
-                         (scalar_t*)out.data_ptr(), (int)C, T, (float)eps,
-                         B * T);
-    } else {
-      hipLaunchKernelGGL((layer_norm_ct_kernel<scalar_t, false>), grid,
-                         dim3(threads), 0, cur_stream(),
-                         (const scalar_t*)x.data_ptr(),
-                         (const scalar_t*)nullptr,
-                         gamma_f.data_ptr<float>(), beta_f.data_ptr<float>(),
-                         (scalar_t*)out.data_ptr(), (int)C, T, (float)eps,
-                         B * T);
-    }
-  });
+  // Fill the chip: one thread per token underfills at encoder sizes
+  // (B*T ~ 8k vs 256 CUs), so split each token's channel reduction over
+  // SPLIT lanes when the token count is small.
+  const long n_bt = B * T;
+  const int threads = 256;
+#define LN_LAUNCH(KER, GRID)                                                \
+  DISPATCH_FT(x, "layer_norm_ct", {                                        \
+    if (res.has_value()) {                                                 \
+      TORCH_CHECK(res->sizes() == x.sizes() && res->is_contiguous());      \
+      hipLaunchKernelGGL((KER<scalar_t, true>), GRID, dim3(threads), 0,    \
+                         cur_stream(), (const scalar_t*)x.data_ptr(),      \
+                         (const scalar_t*)res->data_ptr(),                 \
+                         gamma_f.data_ptr<float>(),                        \
+                         beta_f.data_ptr<float>(),                         \
+                         (scalar_t*)out.data_ptr(), (int)C, T, (float)eps, \
+                         n_bt);                                            \
+    } else {                                                               \
+      hipLaunchKernelGGL((KER<scalar_t, false>), GRID, dim3(threads), 0,   \
+                         cur_stream(), (const scalar_t*)x.data_ptr(),      \
+                         (const scalar_t*)nullptr,                         \
+                         gamma_f.data_ptr<float>(),                        \
+                         beta_f.data_ptr<float>(),                         \
+                         (scalar_t*)out.data_ptr(), (int)C, T, (float)eps, \
+                         n_bt);                                            \
+    }                                                                      \
+  })
+  if (n_bt <= 16384) {
+    dim3 grid(ceil_div(n_bt, (long)threads / 8));
+    DISPATCH_FT(x, "layer_norm_ct", {
+      if (res.has_value()) {
+        TORCH_CHECK(res->sizes() == x.sizes() && res->is_contiguous());
+        hipLaunchKernelGGL((layer_norm_ct_split_kernel<scalar_t, true, 8>),
+                           grid, dim3(threads), 0, cur_stream(),
+                           (const scalar_t*)x.data_ptr(),
+                           (const scalar_t*)res->data_ptr(),
+                           gamma_f.data_ptr<float>(),
+                           beta_f.data_ptr<float>(),
+                           (scalar_t*)out.data_ptr(), (int)C, T, (float)eps,
+                           n_bt);
+      } else {
+        hipLaunchKernelGGL((layer_norm_ct_split_kernel<scalar_t, false, 8>),
+                           grid, dim3(threads), 0, cur_stream(),
+                           (const scalar_t*)x.data_ptr(),
+                           (const scalar_t*)nullptr,
+                           gamma_f.data_ptr<float>(),
+                           beta_f.data_ptr<float>(),
+                           (scalar_t*)out.data_ptr(), (int)C, T, (float)eps,
+                           n_bt);
+      }
+    });
+  } else {
+    dim3 grid(ceil_div(n_bt, (long)threads));
+    LN_LAUNCH(layer_norm_ct_kernel, grid);
+  }
+#undef LN_LAUNCH
   return out;
 }
 

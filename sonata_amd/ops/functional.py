@@ -39,7 +39,7 @@ def layer_norm_ct(
         return ext.layer_norm_ct(
             x.contiguous(),
             residual.contiguous() if residual is not None else None,
-            gamma, beta, eps,
+            _bias_f32(gamma), _bias_f32(beta), eps,
         )
     if residual is not None:
         x = x + residual
