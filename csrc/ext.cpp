@@ -1,5 +1,8 @@
-// Python bindings for the sonata_amd CDNA4 kernel library.
+// Python bindings for the sonata_amd CDNA4 kernel library + the C++
+// VitsEngine runtime.
 #include <torch/extension.h>
+
+#include "engine/vits_engine.h"
 
 // elementwise.hip
 torch::Tensor layer_norm_ct(torch::Tensor x, c10::optional<torch::Tensor> res,
@@ -52,4 +55,38 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "channel-last MFMA conv1d, fused act/residual/mask");
   m.def("convtranspose1d_cl_fused", &convtranspose1d_cl_fused,
         "channel-last MFMA transposed conv1d, phase-merged");
+
+  // C++ inference runtime (csrc/engine): the ort-replacement executor.
+  py::class_<sonata::VitsEngine>(m, "VitsEngine")
+      .def(py::init([](const std::string& path, const std::string& device,
+                       const std::string& dtype) {
+             torch::Device dev(device);
+             torch::Dtype dt =
+                 dtype == "bf16" ? torch::kBFloat16 : torch::kFloat32;
+             return new sonata::VitsEngine(path, dev, dt);
+           }),
+           py::arg("config_path"), py::arg("device") = "cpu",
+           py::arg("dtype") = "f32")
+      .def("infer",
+           [](sonata::VitsEngine& e, torch::Tensor ids,
+              torch::Tensor lengths, c10::optional<torch::Tensor> sid,
+              double ns, double ls, double nw,
+              std::vector<int64_t> seeds) {
+             auto r = e.infer(ids, lengths, sid, ns, ls, nw, seeds);
+             return py::make_tuple(r.first, r.second);
+           },
+           py::arg("ids"), py::arg("lengths"),
+           py::arg("sid") = py::none(), py::arg("noise_scale") = 0.667,
+           py::arg("length_scale") = 1.0, py::arg("noise_w") = 0.8,
+           py::arg("seeds") = std::vector<int64_t>{})
+      .def("phonemes_to_ids", &sonata::VitsEngine::phonemes_to_ids)
+      .def_property_readonly(
+          "sample_rate",
+          [](sonata::VitsEngine& e) { return e.config().sample_rate; })
+      .def_property_readonly(
+          "num_speakers",
+          [](sonata::VitsEngine& e) { return e.config().num_speakers; })
+      .def_property_readonly("hop", [](sonata::VitsEngine& e) {
+        return e.config().hop();
+      });
 }
