@@ -1,0 +1,64 @@
+"""Multi-voice serving (baseline config #5): ar (tashkeel diacritizer) +
+de voices co-resident in one gRPC server, synthesized interleaved —
+mirrors the reference's voice registry semantics (grpc main.rs:76-123)."""
+
+import pytest
+
+from sonata_amd.frontends.grpc import create_server
+from sonata_amd.frontends.grpc.client import SonataGrpcClient
+from sonata_amd.frontends.grpc.proto import MESSAGES
+from sonata_amd.models import create_random_voice
+
+
+@pytest.fixture(scope="module")
+def voices(tmp_path_factory):
+    d = tmp_path_factory.mktemp("mv")
+    ar = create_random_voice(str(d), "ar_JO_test", quality="x_low",
+                             language="ar")
+    de = create_random_voice(str(d), "de_DE_test", quality="x_low",
+                             language="de")
+    return ar, de
+
+
+def test_two_voices_coresident(voices):
+    ar_pack, de_pack = voices
+    server, port, service = create_server(port=0, device="cpu")
+    server.start()
+    try:
+        client = SonataGrpcClient(f"127.0.0.1:{port}")
+        ar_id = client.LoadVoice(
+            MESSAGES["VoicePath"](config_path=ar_pack)).voice_id
+        de_id = client.LoadVoice(
+            MESSAGES["VoicePath"](config_path=de_pack)).voice_id
+        assert ar_id != de_id
+        ar_info = client.GetVoiceInfo(
+            MESSAGES["VoiceIdentifier"](voice_id=ar_id))
+        de_info = client.GetVoiceInfo(
+            MESSAGES["VoiceIdentifier"](voice_id=de_id))
+        assert ar_info.language == "ar"
+        assert de_info.language == "de"
+
+        # interleaved synthesis on both voices; Arabic text goes through
+        # the tashkeel diacritizer inside phonemize (voice.py:89-93)
+        ar_results = list(client.SynthesizeUtterance(MESSAGES["Utterance"](
+            voice_id=ar_id, text="مرحبا بالعالم.")))
+        de_results = list(client.SynthesizeUtterance(MESSAGES["Utterance"](
+            voice_id=de_id, text="Hallo Welt. Wie geht es dir?")))
+        assert len(ar_results) >= 1
+        assert len(de_results) == 2
+        assert all(len(r.wav_samples) > 200 for r in ar_results + de_results)
+        client.close()
+    finally:
+        server.stop(grace=None)
+
+
+def test_tashkeel_applied_for_arabic(voices):
+    """Arabic voices instantiate the diacritizer (reference
+    piper/src/lib.rs:63-77,321-333)."""
+    from sonata_amd.models.voice import load_voice
+
+    ar_pack, _ = voices
+    v = load_voice(ar_pack, device="cpu")
+    assert v._tashkeel is not None
+    sents = v.phonemize_text("كتب الولد")
+    assert len(sents) >= 1 and len(sents[0]) > 0

@@ -110,3 +110,21 @@ for T in [256, 512, 1024, 2048]:
     dt = (time.perf_counter()-t0)/N
     fl = 2*B*256*128*2*T*8
     print(f"cf up1 T={T:6d} {dt*1e3:8.3f} ms  {fl/dt/1e12:7.1f} TF")
+
+# up1 flakiness probe: same shape, fresh tensors each trial
+print("--- cl up1 flakiness probe (5 fresh-tensor trials) ---")
+for trial in range(5):
+    x = (torch.randn(B, 2048, 256)/4).to(torch.bfloat16).to(dev)
+    w = (torch.randn(256, 128, 16)/(256*16)**0.5).to(torch.bfloat16).to(dev)
+    bias = (torch.randn(128)/10).to(dev)
+    for _ in range(3):
+        y = leaky_convtranspose1d_cl(x, w, bias, 8, 4, pre_lrelu=0.1)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter(); N = 10
+    for _ in range(N):
+        y = leaky_convtranspose1d_cl(x, w, bias, 8, 4, pre_lrelu=0.1)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter()-t0)/N
+    fl = 2*B*256*128*2*2048*8
+    print(f"trial {trial}: {dt*1e3:8.3f} ms  {fl/dt/1e12:7.1f} TF  x_ptr={x.data_ptr()%(1<<21)} w_ptr={w.data_ptr()%(1<<21)}")
+    del x, w, bias, y
