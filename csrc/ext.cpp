@@ -40,6 +40,11 @@ torch::Tensor convtranspose1d_cl_fused(torch::Tensor x, torch::Tensor w_perm,
                                        long Cout, long k, long stride,
                                        long padding, double pre_lrelu,
                                        c10::optional<torch::Tensor> out_lens);
+// resblock_cl.hip
+torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
+                                     torch::Tensor b1, torch::Tensor w2_perm,
+                                     torch::Tensor b2, long k, long dil,
+                                     c10::optional<torch::Tensor> out_lens);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "sonata_amd hand-written CDNA4 (gfx950) kernels";
@@ -55,6 +60,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "channel-last MFMA conv1d, fused act/residual/mask");
   m.def("convtranspose1d_cl_fused", &convtranspose1d_cl_fused,
         "channel-last MFMA transposed conv1d, phase-merged");
+  m.def("resblock_pair_cl_fused", &resblock_pair_cl_fused,
+        "fused HiFi-GAN resblock conv pair (xt stays in LDS)");
 
   // C++ inference runtime (csrc/engine): the ort-replacement executor.
   py::class_<sonata::VitsEngine>(m, "VitsEngine")

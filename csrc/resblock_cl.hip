@@ -1,0 +1,264 @@
+// Fused HiFi-GAN ResBlock conv pair, channel-last (gfx950).
+//
+//   out = conv2_{k,d=1}( lrelu( conv1_{k,d}( lrelu(x) ) ) ) + x
+//
+// One kernel per pair instead of two: the intermediate tensor xt never
+// touches HBM — conv1's output tile is written to LDS (pre-activated)
+// and conv2 consumes it in place.  The small-channel resblock stages
+// (C = 32..128 at T up to 256*F) are HBM-bandwidth-bound (measured
+// 2.5-2.9 TB/s, profiles/r01_conv_pmc_v2.txt), so removing xt's
+// write+read halves their traffic.
+//
+// Geometry: xt tile is a fixed 128 rows; the block's OUTPUT tile is
+// BM = 128-(k-1) rows (conv2 consumes (k-1)/2 halo rows per side of xt,
+// recomputed per block).  GEMM orientation identical to conv1d_cl.hip:
+// A = time rows (k-contiguous channel-last), B = [co][ci] weight taps,
+// K = Cin in 32-deep LDS slices.
+#include "common.h"
+
+#define BK 32
+#define BKP 40
+#define XTROWS 128
+
+template <int BN, int WGN, int TC>
+__global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
+    const bf16* __restrict__ x,    // [B][T][C]
+    const bf16* __restrict__ w1,   // [k][CP][CP] (dilated conv)
+    const float* __restrict__ b1,  // [C]
+    const bf16* __restrict__ w2,   // [k][CP][CP] (d=1 conv)
+    const float* __restrict__ b2,  // [C]
+    bf16* __restrict__ out,        // [B][T][C]
+    const int* __restrict__ out_lens,
+    int C, int CP, long T, int k, int dil) {
+  constexpr int WGM = 4;
+  constexpr int WM = XTROWS / WGM;  // 32
+  constexpr int MT = WM / 16;       // 2
+  constexpr int WN = BN / WGN;
+  constexpr int NT = WN / 16;
+  constexpr int XROWS_MAX = XTROWS + 64;  // + (k-1)*dil <= 50
+  constexpr int XTP = BN + 8;             // xt row pitch
+
+  const int h2 = (k - 1) / 2;        // conv2 halo per side
+  const int BM = XTROWS - (k - 1);   // output rows per block
+  const long t0 = (long)blockIdx.x * BM;
+  const int b = blockIdx.z;
+
+  __shared__ bf16 Xs[XROWS_MAX][BKP];
+  __shared__ bf16 Xt[XTROWS][XTP];
+  __shared__ bf16 Ws[TC][BN][BKP];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid / WGN;
+  const int wc = wid % WGN;
+  const int kl = lane >> 4;
+  const int il = lane & 15;
+
+  const bf16* xb = x + (long)b * T * C;
+  const int pad1 = (k - 1) * dil / 2;
+  // Xs[0] holds x row (t0 - h2 - pad1); GEMM1 xt row m taps rows m+j*dil
+  const long row0 = t0 - h2 - pad1;
+  const int xrows = XTROWS + (k - 1) * dil;
+  const bool t_interior = (row0 >= 0) && (row0 + xrows <= T);
+
+  // ================= GEMM1: xt = lrelu(conv1(lrelu(x))) ================
+  f32x4 acc[MT][NT];
+#pragma unroll
+  for (int i = 0; i < MT; ++i)
+#pragma unroll
+    for (int j = 0; j < NT; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int c0 = 0; c0 < CP; c0 += BK) {
+    const bool c_interior = (c0 + BK) <= C;
+    if (t_interior && c_interior) {
+      for (int u = tid; u < xrows * 4; u += 512) {
+        const int r = u >> 2, ch = (u & 3) * 8;
+        bf16 v8[8];
+        *(ulonglong2*)v8 =
+            *(const ulonglong2*)&xb[(row0 + r) * C + c0 + ch];
+#pragma unroll
+        for (int q = 0; q < 8; ++q)
+          v8[q] = f2bf(lrelu_(bf2f(v8[q]), 0.1f));
+        *(ulonglong2*)&Xs[r][ch] = *(ulonglong2*)v8;
+      }
+    } else {
+      for (int u = tid; u < xrows * 4; u += 512) {
+        const int r = u >> 2, ch = (u & 3) * 8;
+        const long t = row0 + r;
+        bf16 v8[8];
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+          const int c = c0 + ch + q;
+          float v = (t >= 0 && t < T && c < C) ? bf2f(xb[t * C + c]) : 0.f;
+          v8[q] = f2bf(lrelu_(v, 0.1f));
+        }
+        *(ulonglong2*)&Xs[r][ch] = *(ulonglong2*)v8;
+      }
+    }
+    for (int tap0 = 0; tap0 < k; tap0 += TC) {
+      const int ntc = min(TC, k - tap0);
+      for (int tc = 0; tc < ntc; ++tc) {
+        const long wbase = ((long)(tap0 + tc) * CP) * CP + c0;
+        for (int u = tid; u < BN * 4; u += 512) {
+          const int n = u >> 2, ch = (u & 3) * 8;
+          *(ulonglong2*)&Ws[tc][n][ch] =
+              *(const ulonglong2*)&w1[wbase + (long)n * CP + ch];
+        }
+      }
+      __syncthreads();
+      for (int tc = 0; tc < ntc; ++tc) {
+        const int toff = (tap0 + tc) * dil;
+        bf16x8 b_frag[NT];
+#pragma unroll
+        for (int nj = 0; nj < NT; ++nj)
+          b_frag[nj] =
+              *(const bf16x8*)&Ws[tc][wc * WN + nj * 16 + il][kl * 8];
+#pragma unroll
+        for (int mi = 0; mi < MT; ++mi) {
+          const bf16x8 a_frag =
+              *(const bf16x8*)&Xs[wr * WM + mi * 16 + il + toff][kl * 8];
+#pragma unroll
+          for (int nj = 0; nj < NT; ++nj)
+            acc[mi][nj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag, b_frag[nj], acc[mi][nj], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // epilogue1 -> LDS xt (bias1 + lrelu; zero outside [0,T) and >= lim)
+  const long lim = out_lens ? min((long)out_lens[b], T) : T;
+#pragma unroll
+  for (int mi = 0; mi < MT; ++mi) {
+#pragma unroll
+    for (int rg = 0; rg < 4; ++rg) {
+      const int m = wr * WM + mi * 16 + kl * 4 + rg;
+      const long t = t0 - h2 + m;
+      const bool live = (t >= 0) && (t < lim);
+#pragma unroll
+      for (int nj = 0; nj < NT; ++nj) {
+        const int co = wc * WN + nj * 16 + il;
+        float v = 0.f;
+        if (live && co < C)
+          v = lrelu_(acc[mi][nj][rg] + b1[co], 0.1f);
+        Xt[m][co] = f2bf(v);
+      }
+    }
+  }
+  __syncthreads();
+
+  // ================= GEMM2: out = conv2(xt) + x =======================
+#pragma unroll
+  for (int i = 0; i < MT; ++i)
+#pragma unroll
+    for (int j = 0; j < NT; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int c0 = 0; c0 < CP; c0 += BK) {
+    for (int tap0 = 0; tap0 < k; tap0 += TC) {
+      const int ntc = min(TC, k - tap0);
+      for (int tc = 0; tc < ntc; ++tc) {
+        const long wbase = ((long)(tap0 + tc) * CP) * CP + c0;
+        for (int u = tid; u < BN * 4; u += 512) {
+          const int n = u >> 2, ch = (u & 3) * 8;
+          *(ulonglong2*)&Ws[tc][n][ch] =
+              *(const ulonglong2*)&w2[wbase + (long)n * CP + ch];
+        }
+      }
+      __syncthreads();
+      for (int tc = 0; tc < ntc; ++tc) {
+        const int toff = tap0 + tc;  // d=1
+        bf16x8 b_frag[NT];
+#pragma unroll
+        for (int nj = 0; nj < NT; ++nj)
+          b_frag[nj] =
+              *(const bf16x8*)&Ws[tc][wc * WN + nj * 16 + il][kl * 8];
+#pragma unroll
+        for (int mi = 0; mi < MT; ++mi) {
+          // conv2 out row m2 taps xt rows m2 + toff (xt row 0 = t0-h2)
+          const bf16x8 a_frag =
+              *(const bf16x8*)&Xt[wr * WM + mi * 16 + il + toff][c0 + kl * 8];
+#pragma unroll
+          for (int nj = 0; nj < NT; ++nj)
+            acc[mi][nj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag, b_frag[nj], acc[mi][nj], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // epilogue2: bias2 + residual x + mask + store
+  bf16* ob = out + (long)b * T * C;
+#pragma unroll
+  for (int mi = 0; mi < MT; ++mi) {
+#pragma unroll
+    for (int rg = 0; rg < 4; ++rg) {
+      const int m2 = wr * WM + mi * 16 + kl * 4 + rg;
+      const long t = t0 + m2;
+      if (m2 >= BM || t >= T) continue;
+      const bool live = t < lim;
+#pragma unroll
+      for (int nj = 0; nj < NT; ++nj) {
+        const int co = wc * WN + nj * 16 + il;
+        if (co >= C) continue;
+        float v = 0.f;
+        if (live)
+          v = acc[mi][nj][rg] + b2[co] + bf2f(xb[t * C + co]);
+        ob[t * C + co] = f2bf(v);
+      }
+    }
+  }
+}
+
+// ========================================================================
+// host wrapper
+// ========================================================================
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+static inline hipStream_t cur_stream4() {
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
+                                     torch::Tensor b1, torch::Tensor w2_perm,
+                                     torch::Tensor b2, long k, long dil,
+                                     c10::optional<torch::Tensor> out_lens) {
+  TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "resblock_cl: bf16 only");
+  const long B = x.size(0), T = x.size(1), C = x.size(2);
+  TORCH_CHECK(w1_perm.size(0) == k && w2_perm.size(0) == k);
+  const int CP = w1_perm.size(2);
+  TORCH_CHECK(w1_perm.size(1) == CP && w2_perm.size(1) == CP &&
+              w2_perm.size(2) == CP, "resblock_cl: square channel conv");
+  TORCH_CHECK((k - 1) * dil <= 64, "resblock_cl: halo too large");
+  auto out = torch::empty_like(x);
+  if (out.numel() == 0) return out;
+  auto b1f = b1.scalar_type() == at::kFloat ? b1 : b1.to(at::kFloat).contiguous();
+  auto b2f = b2.scalar_type() == at::kFloat ? b2 : b2.to(at::kFloat).contiguous();
+  const int* lens_p = nullptr;
+  if (out_lens.has_value()) {
+    TORCH_CHECK(out_lens->scalar_type() == at::kInt && out_lens->is_cuda());
+    lens_p = out_lens->data_ptr<int>();
+  }
+  const long BM = XTROWS - (k - 1);
+  hipStream_t st = cur_stream4();
+#define LAUNCH_RB(BN, WGN, TC)                                              \
+  hipLaunchKernelGGL((resblock_pair_cl_kernel<BN, WGN, TC>),                \
+                     dim3(ceil_div(T, BM), 1, B), dim3(512), 0, st,         \
+                     (const bf16*)x.data_ptr(),                             \
+                     (const bf16*)w1_perm.data_ptr(),                       \
+                     b1f.data_ptr<float>(),                                 \
+                     (const bf16*)w2_perm.data_ptr(),                       \
+                     b2f.data_ptr<float>(), (bf16*)out.data_ptr(), lens_p,  \
+                     (int)C, CP, T, (int)k, (int)dil)
+  if (CP == 256) LAUNCH_RB(256, 2, 2);
+  else if (CP == 128) LAUNCH_RB(128, 2, 2);
+  else if (CP == 64) LAUNCH_RB(64, 2, 2);
+  else if (CP == 32) LAUNCH_RB(32, 2, 2);
+  else TORCH_CHECK(false, "resblock_cl: unsupported CP ", CP);
+#undef LAUNCH_RB
+  return out;
+}

@@ -22,6 +22,7 @@ ext = CUDAExtension(
         "csrc/elementwise.hip",
         "csrc/conv1d.hip",
         "csrc/conv1d_cl.hip",
+        "csrc/resblock_cl.hip",
         "csrc/engine/vits_engine.cpp",
     ],
     extra_compile_args={

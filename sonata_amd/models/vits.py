@@ -572,18 +572,14 @@ class ResBlock1(nn.Module):
 
     def forward_cl(self, x: torch.Tensor,
                    lengths: Optional[torch.Tensor] = None) -> torch.Tensor:
-        """Channel-last ([B,T,C]) path: fused-mask MFMA kernels
-        (csrc/conv1d_cl.hip); numerically equivalent to forward()."""
+        """Channel-last ([B,T,C]) path: each conv pair runs as ONE fused
+        kernel with the intermediate tensor in LDS
+        (csrc/resblock_cl.hip); numerically equivalent to forward()."""
+        from ..ops import resblock_pair_cl
+
         for c1, c2 in zip(self.convs1, self.convs2):
-            xt = leaky_conv1d_cl(
-                x, c1.weight, c1.bias,
-                padding=c1.padding[0], dilation=c1.dilation[0],
-                pre_lrelu=LRELU_SLOPE, out_lens=lengths,
-            )
-            x = leaky_conv1d_cl(
-                xt, c2.weight, c2.bias, padding=c2.padding[0],
-                pre_lrelu=LRELU_SLOPE, residual=x, out_lens=lengths,
-            )
+            x = resblock_pair_cl(x, c1.weight, c1.bias, c2.weight, c2.bias,
+                                 dilation=c1.dilation[0], out_lens=lengths)
         return x
 
 

@@ -66,4 +66,5 @@ from .functional import (  # noqa: F401,E402
     mask_tail_,
     leaky_conv1d_cl,
     leaky_convtranspose1d_cl,
+    resblock_pair_cl,
 )

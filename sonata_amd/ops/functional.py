@@ -367,3 +367,28 @@ def leaky_convtranspose1d_cl(
             (idx.unsqueeze(0) >= out_lens.to(y.device).unsqueeze(1))
             .unsqueeze(-1), 0)
     return y
+
+
+def resblock_pair_cl(
+    x: torch.Tensor,  # [B, T, C]
+    w1: torch.Tensor, b1: Optional[torch.Tensor],
+    w2: torch.Tensor, b2: Optional[torch.Tensor],
+    dilation: int,
+    out_lens: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """Fused resblock conv pair: conv2_{k,1}(lrelu(conv1_{k,d}(lrelu(x))))
+    + x, channel-last, intermediate tensor kept in LDS
+    (csrc/resblock_cl.hip).  Falls back to two conv calls elsewhere."""
+    Cout, Cin, k = w1.shape
+    if (use_hip(x) and Cin == Cout and b1 is not None and b2 is not None
+            and (k - 1) * dilation <= 64):
+        ext = hip_ext(required=True)
+        return ext.resblock_pair_cl_fused(
+            x.contiguous(), _conv_weight_mfma(w1), _bias_f32(b1),
+            _conv_weight_mfma(w2), _bias_f32(b2), k, dilation,
+            _lens_i32(out_lens, x.device),
+        )
+    xt = leaky_conv1d_cl(x, w1, b1, padding=(k - 1) * dilation // 2,
+                         dilation=dilation, pre_lrelu=0.1, out_lens=out_lens)
+    return leaky_conv1d_cl(xt, w2, b2, padding=(k - 1) // 2, pre_lrelu=0.1,
+                           residual=x, out_lens=out_lens)

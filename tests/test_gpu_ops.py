@@ -391,3 +391,28 @@ def test_generator_cl_matches_oracle(dev):
         err = _rel_err(got[b, :, : ln * hop], ref[b, :, : ln * hop])
         assert err < 0.05, f"generator cl parity row {b}: {err}"
     assert got[1, :, 40 * hop:].abs().max().item() == 0
+
+
+@pytest.mark.parametrize("C,k,dil", [
+    (128, 3, 1), (128, 7, 3), (128, 11, 5), (64, 3, 1), (32, 11, 5),
+    (256, 3, 1), (16, 7, 3),
+])
+def test_resblock_pair_cl(dev, C, k, dil):
+    from sonata_amd.ops.functional import resblock_pair_cl
+
+    torch.manual_seed(C + k)
+    B, T = 3, 300
+    x = (torch.randn(B, T, C) / 4).to(torch.bfloat16)
+    w1 = (torch.randn(C, C, k) / (C * k) ** 0.5).to(torch.bfloat16)
+    w2 = (torch.randn(C, C, k) / (C * k) ** 0.5).to(torch.bfloat16)
+    b1 = torch.randn(C) / 10
+    b2 = torch.randn(C) / 10
+    lens = torch.tensor([T, 200, 45])
+    got = resblock_pair_cl(x.to(dev), w1.to(dev), b1.to(dev), w2.to(dev),
+                           b2.to(dev), dilation=dil, out_lens=lens.to(dev))
+    ref = resblock_pair_cl(x.float(), w1.float(), b1, w2.float(), b2,
+                           dilation=dil, out_lens=lens)
+    err = _rel_err(got, ref)
+    assert err < 0.03, f"resblock pair parity {err}"
+    assert got[1, 200:].abs().max().item() == 0
+    assert got[2, 45:].abs().max().item() == 0

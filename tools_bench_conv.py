@@ -128,3 +128,24 @@ for trial in range(5):
     fl = 2*B*256*128*2*2048*8
     print(f"trial {trial}: {dt*1e3:8.3f} ms  {fl/dt/1e12:7.1f} TF  x_ptr={x.data_ptr()%(1<<21)} w_ptr={w.data_ptr()%(1<<21)}")
     del x, w, bias, y
+
+# fused resblock pair vs two convs
+from sonata_amd.ops.functional import resblock_pair_cl
+print("--- fused resblock pair (k3d1 + k3d1 equivalents) ---")
+for C, T, k, d in [(256, 8*F, 3, 1), (128, 64*F, 3, 1), (64, 128*F, 3, 1),
+                   (32, 256*F, 3, 1), (128, 64*F, 11, 5), (32, 256*F, 11, 5)]:
+    x = (torch.randn(B, T, C)/4).to(torch.bfloat16).to(dev)
+    w1 = (torch.randn(C, C, k)/(C*k)**0.5).to(torch.bfloat16).to(dev)
+    w2 = (torch.randn(C, C, k)/(C*k)**0.5).to(torch.bfloat16).to(dev)
+    b1 = (torch.randn(C)/10).to(dev); b2 = (torch.randn(C)/10).to(dev)
+    for _ in range(3):
+        y = resblock_pair_cl(x, w1, b1, w2, b2, dilation=d)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter(); N = 10
+    for _ in range(N):
+        y = resblock_pair_cl(x, w1, b1, w2, b2, dilation=d)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter()-t0)/N
+    fl = 2*2*B*C*C*k*T
+    gb = 2*B*T*(2*C + C)  # x read(+resid reread), out write
+    print(f"rbpair C={C:4d} k{k}d{d} T={T:6d} {dt*1e3:8.3f} ms  {fl/dt/1e12:7.1f} TF  {gb/dt/1e9:7.0f} GB/s")
