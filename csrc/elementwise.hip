@@ -111,6 +111,7 @@ template <typename T, bool HAS_G>
 __global__ void fused_gate_kernel(const T* __restrict__ x,
                                   const T* __restrict__ g,
                                   T* __restrict__ out, long C_T, long CT2,
+                                  long T_len, long g_T,  // g time size: T or 1
                                   long n) {
   // n = B*C*T ; C_T = C*T (half-channel offset); CT2 = 2*C*T (batch stride)
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -121,8 +122,13 @@ __global__ void fused_gate_kernel(const T* __restrict__ x,
   long ib = ia + C_T;
   float va = ld_f(x + ia), vb = ld_f(x + ib);
   if (HAS_G) {
-    va += ld_f(g + ia);
-    vb += ld_f(g + ib);
+    // g may be time-broadcast [B, 2C, 1] (speaker conditioning)
+    long c = r / T_len;
+    long t = g_T == 1 ? 0 : r % T_len;
+    long C = C_T / T_len;
+    long ga = (b * 2 * C + c) * g_T + t;
+    va += ld_f(g + ga);
+    vb += ld_f(g + (ga + C * g_T));
   }
   st_f(out + i, tanhf(va) * sigmoidf_(vb));
 }
@@ -281,19 +287,21 @@ torch::Tensor fused_gate(torch::Tensor x, c10::optional<torch::Tensor> g,
   const long blocks = (n + threads - 1) / threads;
   DISPATCH_FT(x, "fused_gate", {
     if (g.has_value()) {
-      TORCH_CHECK(g->sizes() == x.sizes());
+      TORCH_CHECK(g->dim() == 3 && g->size(0) == B && g->size(1) == C2 &&
+                      (g->size(2) == T || g->size(2) == 1),
+                  "fused_gate: g must be [B,2C,T] or [B,2C,1]");
       hipLaunchKernelGGL((fused_gate_kernel<scalar_t, true>), dim3(blocks),
                          dim3(threads), 0, cur_stream(),
                          (const scalar_t*)x.data_ptr(),
                          (const scalar_t*)g->data_ptr(),
                          (scalar_t*)out.data_ptr(), n_channels * T,
-                         2 * n_channels * T, n);
+                         2 * n_channels * T, T, g->size(2), n);
     } else {
       hipLaunchKernelGGL((fused_gate_kernel<scalar_t, false>), dim3(blocks),
                          dim3(threads), 0, cur_stream(),
                          (const scalar_t*)x.data_ptr(), (const scalar_t*)nullptr,
                          (scalar_t*)out.data_ptr(), n_channels * T,
-                         2 * n_channels * T, n);
+                         2 * n_channels * T, T, 1, n);
     }
   });
   return out;
