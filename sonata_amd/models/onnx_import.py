@@ -1,0 +1,204 @@
+"""Piper `.onnx` voice importer: weights-only compatibility path.
+
+The reference loads `voice.onnx` + `voice.onnx.json` and runs the graph
+through ONNX Runtime (piper/src/lib.rs:79-86).  This framework replaces
+the executor wholesale, so the compatibility path is a
+WEIGHT importer: parse the ONNX protobuf's initializers (no onnx package
+in this environment — a minimal wire-format reader below), map upstream
+VITS parameter names to sonata_amd module names, and write the
+`<stem>.safetensors` voice pack next to the config.
+
+Upstream names follow the canonical VITS `SynthesizerTrn` module tree
+(piper's training repo keeps them in the export):
+    enc_p.encoder.attn_layers.N.*   -> enc_p.attn_layers.N.*
+    enc_p.encoder.norm_layers_1.N.* -> enc_p.norm1.N.*
+    enc_p.encoder.ffn_layers.N.conv_1.* -> enc_p.ffn_layers.N.conv1.*
+    dp./flow./dec./emb_g.*          -> same names (minor renames below)
+
+Training-only tensors (posterior encoder `enc_q`, SDP `post_*` flows,
+discriminators) are skipped.
+"""
+
+from __future__ import annotations
+
+import re
+import struct
+import sys
+from typing import Dict, List, Tuple
+
+import numpy as np
+
+from ..core import ModelError
+
+# ONNX TensorProto data types we support
+_DTYPES = {1: np.float32, 10: np.float16, 16: None, 7: np.int64, 6: np.int32}
+
+
+def _read_varint(buf: bytes, pos: int) -> Tuple[int, int]:
+    result = 0
+    shift = 0
+    while True:
+        b = buf[pos]
+        pos += 1
+        result |= (b & 0x7F) << shift
+        if not (b & 0x80):
+            return result, pos
+        shift += 7
+
+
+def _walk_fields(buf: bytes):
+    """Yield (field_number, wire_type, value, raw) over a protobuf
+    message body."""
+    pos = 0
+    n = len(buf)
+    while pos < n:
+        tag, pos = _read_varint(buf, pos)
+        field, wt = tag >> 3, tag & 7
+        if wt == 0:  # varint
+            v, pos = _read_varint(buf, pos)
+            yield field, wt, v
+        elif wt == 1:  # 64-bit
+            v = struct.unpack_from("<q", buf, pos)[0]
+            pos += 8
+            yield field, wt, v
+        elif wt == 2:  # length-delimited
+            ln, pos = _read_varint(buf, pos)
+            yield field, wt, buf[pos : pos + ln]
+            pos += ln
+        elif wt == 5:  # 32-bit
+            v = struct.unpack_from("<i", buf, pos)[0]
+            pos += 4
+            yield field, wt, v
+        else:
+            raise ModelError(f"onnx: unsupported wire type {wt}")
+
+
+def _parse_tensor(buf: bytes) -> Tuple[str, np.ndarray]:
+    """TensorProto: dims=1, data_type=2, float_data=4, int64_data=7,
+    name=8, raw_data=9."""
+    dims: List[int] = []
+    dtype = 1
+    name = ""
+    raw = b""
+    float_data: List[float] = []
+    int64_data: List[int] = []
+    for field, wt, v in _walk_fields(buf):
+        if field == 1:
+            if wt == 0:
+                dims.append(v)
+            else:  # packed
+                p = 0
+                while p < len(v):
+                    d, p = _read_varint(v, p)
+                    dims.append(d)
+        elif field == 2:
+            dtype = v
+        elif field == 4:
+            if wt == 5:
+                float_data.append(struct.unpack("<f", struct.pack("<i", v))[0])
+            else:  # packed floats
+                float_data.extend(np.frombuffer(v, dtype=np.float32).tolist())
+        elif field == 7:
+            if wt == 0:
+                int64_data.append(v)
+            else:
+                p = 0
+                while p < len(v):
+                    d, p = _read_varint(v, p)
+                    int64_data.append(d)
+        elif field == 8:
+            name = v.decode("utf-8")
+        elif field == 9:
+            raw = v
+    np_dt = _DTYPES.get(dtype)
+    if np_dt is None:
+        raise ModelError(f"onnx: unsupported tensor dtype {dtype} ({name})")
+    if raw:
+        arr = np.frombuffer(raw, dtype=np_dt).reshape(dims or [-1]).copy()
+    elif float_data:
+        arr = np.asarray(float_data, dtype=np.float32).reshape(dims or [-1])
+    elif int64_data:
+        arr = np.asarray(int64_data, dtype=np.int64).reshape(dims or [-1])
+    else:
+        arr = np.zeros(dims or [0], dtype=np_dt)
+    return name, arr
+
+
+def parse_onnx_initializers(path: str) -> Dict[str, np.ndarray]:
+    """Extract all graph initializers from an ONNX file."""
+    with open(path, "rb") as f:
+        model = f.read()
+    out: Dict[str, np.ndarray] = {}
+    for field, wt, v in _walk_fields(model):
+        if field == 7 and wt == 2:  # ModelProto.graph
+            for gfield, gwt, gv in _walk_fields(v):
+                if gfield == 5 and gwt == 2:  # GraphProto.initializer
+                    name, arr = _parse_tensor(gv)
+                    if name:
+                        out[name] = arr
+    return out
+
+
+# ----- upstream VITS name -> sonata_amd name rules ----------------------- #
+_RULES = [
+    (r"^enc_p\.encoder\.attn_layers\.", "enc_p.attn_layers."),
+    (r"^enc_p\.encoder\.norm_layers_1\.", "enc_p.norm1."),
+    (r"^enc_p\.encoder\.norm_layers_2\.", "enc_p.norm2."),
+    (r"^enc_p\.encoder\.ffn_layers\.(\d+)\.conv_1\.", r"enc_p.ffn_layers.\1.conv1."),
+    (r"^enc_p\.encoder\.ffn_layers\.(\d+)\.conv_2\.", r"enc_p.ffn_layers.\1.conv2."),
+    (r"^dec\.cond\.", "dec.cond."),
+]
+_SKIP = re.compile(r"^(enc_q\.|dp\.post_|.*\.weight_[gv]$)")
+
+
+def map_vits_name(name: str) -> str:
+    for pat, repl in _RULES:
+        name = re.sub(pat, repl, name)
+    return name
+
+
+def import_onnx_voice(onnx_path: str, out_path: str = None,
+                      strict: bool = False) -> str:
+    """Convert a Piper `voice.onnx` into `<stem>.safetensors`.
+
+    Returns the written path.  With strict=True, unmapped initializers
+    raise instead of being reported and skipped."""
+    import torch
+    from safetensors.torch import save_file
+
+    inits = parse_onnx_initializers(onnx_path)
+    if not inits:
+        raise ModelError(f"no initializers found in {onnx_path}")
+    state: Dict[str, "torch.Tensor"] = {}
+    skipped: List[str] = []
+    for name, arr in inits.items():
+        if _SKIP.match(name):
+            continue
+        mapped = map_vits_name(name)
+        if not re.match(r"^(enc_p|dp|flow|dec|emb_g)\.|^emb_g$", mapped):
+            skipped.append(name)
+            continue
+        state[mapped] = torch.from_numpy(
+            np.ascontiguousarray(arr.astype(np.float32)))
+    if skipped:
+        msg = f"unmapped initializers ({len(skipped)}): {skipped[:8]}"
+        if strict:
+            raise ModelError(msg)
+        print(f"onnx_import warning: {msg}", file=sys.stderr)
+    if out_path is None:
+        stem = onnx_path
+        if stem.endswith(".onnx"):
+            stem = stem[: -len(".onnx")]
+        out_path = stem + ".safetensors"
+    save_file(state, out_path)
+    return out_path
+
+
+if __name__ == "__main__":
+    if len(sys.argv) < 2:
+        print("usage: python -m sonata_amd.models.onnx_import voice.onnx "
+              "[out.safetensors]")
+        sys.exit(2)
+    out = import_onnx_voice(sys.argv[1],
+                            sys.argv[2] if len(sys.argv) > 2 else None)
+    print(f"wrote {out}")
