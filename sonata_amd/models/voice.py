@@ -22,6 +22,7 @@ import numpy as np
 import torch
 
 from ..audio.samples import crossfade
+from ..utils.trace import stage_timer
 from ..core import Audio, AudioInfo, ModelError, Phonemes, SonataModel
 from ..text.ids import phonemes_to_ids
 from ..text.phonemizer import text_to_phonemes
@@ -87,10 +88,11 @@ class VitsVoice(SonataModel):
             self._synth_config = config.copy()
 
     def phonemize_text(self, text: str) -> Phonemes:
-        if self._tashkeel is not None:
-            text = self._tashkeel.diacritize(text)
-        sentences = text_to_phonemes(text, voice=self.config.espeak_voice)
-        return Phonemes(sentences)
+        with stage_timer("phonemize"):
+            if self._tashkeel is not None:
+                text = self._tashkeel.diacritize(text)
+            sentences = text_to_phonemes(text, voice=self.config.espeak_voice)
+            return Phonemes(sentences)
 
     # ------------------------------------------------------------------ #
     # inference
@@ -138,7 +140,8 @@ class VitsVoice(SonataModel):
         gens = self._generators(phonemes_batch, cfg.speaker_id)
 
         t0 = time.perf_counter()
-        audio, audio_lengths = self.net.infer(
+        with stage_timer("infer", self.device):
+            audio, audio_lengths = self.net.infer(
             ids,
             lengths,
             sid=self._sid_tensor(B, cfg.speaker_id),
@@ -146,7 +149,7 @@ class VitsVoice(SonataModel):
             length_scale=cfg.length_scale,
             noise_w=cfg.noise_w,
             generators=gens,
-        )
+            )
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
         infer_ms = (time.perf_counter() - t0) * 1000.0
@@ -179,11 +182,12 @@ class VitsVoice(SonataModel):
         ids = torch.tensor([ids_l], dtype=torch.long, device=self.device)
         lengths = torch.tensor([len(ids_l)], dtype=torch.long, device=self.device)
         gens = self._generators([phonemes], cfg.speaker_id)
-        z, y_mask, g = self.net.infer_encoder(
-            ids, lengths, sid=self._sid_tensor(1, cfg.speaker_id),
-            noise_scale=cfg.noise_scale, length_scale=cfg.length_scale,
-            noise_w=cfg.noise_w, generators=gens,
-        )
+        with stage_timer("encode", self.device):
+            z, y_mask, g = self.net.infer_encoder(
+                ids, lengths, sid=self._sid_tensor(1, cfg.speaker_id),
+                noise_scale=cfg.noise_scale, length_scale=cfg.length_scale,
+                noise_w=cfg.noise_w, generators=gens,
+            )
         hop = self.net.arch.hop_length
         num_frames = z.shape[-1]
         # Overlap-crossfade at seams without changing the timeline: each
@@ -196,7 +200,8 @@ class VitsVoice(SonataModel):
         for spec in chunk_plan(num_frames, chunk_size, chunk_padding):
             z_c = z[:, :, spec.mel_start : spec.mel_end]
             m_c = y_mask[:, :, spec.mel_start : spec.mel_end]
-            audio = self.net.decode(z_c, m_c, g)
+            with stage_timer("decode_chunk", self.device):
+                audio = self.net.decode(z_c, m_c, g)
             wav = audio[0, 0].float().cpu().numpy()
             lo = spec.trim_left_frames * hop
             hi = len(wav) - spec.trim_right_frames * hop
