@@ -72,16 +72,15 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
 #pragma unroll
     for (int j = 0; j < NT; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  // prologue: prefetch W1 chunk (0, 0) into registers.  Loads are
-  // UNCONDITIONAL with clamped addresses: a runtime branch around each
-  // load in an unrolled loop makes hipcc wait vmcnt(0) per element
-  // (guide trap: serialized L2 round trips).
+  // prologue: prefetch W1 chunk (0, 0) into registers
 #pragma unroll
   for (int r = 0; r < WUNITS; ++r) {
     const int u = tid + 512 * r;
-    const int tc = min(u / (BN * 4), (int)k - 1), rem = u % (BN * 4);
-    wpre[r] = *(const ulonglong2*)&w1[
-        ((long)tc * CP + (rem >> 2)) * CP + (rem & 3) * 8];
+    if (u < min(TC, k) * BN * 4) {
+      const int tc = u / (BN * 4), rem = u % (BN * 4);
+      wpre[r] = *(const ulonglong2*)&w1[
+          ((long)tc * CP + (rem >> 2)) * CP + (rem & 3) * 8];
+    }
   }
 
   for (int c0 = 0; c0 < CP; c0 += BK) {
@@ -123,19 +122,22 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
           *(ulonglong2*)&Ws[tc][rem >> 2][(rem & 3) * 8] = wpre[r];
         }
       }
-      // prefetch the NEXT chunk (possibly next K-slice, wrapping taps);
-      // unconditional clamped-address loads (see prologue note)
+      // prefetch the NEXT chunk (possibly next K-slice, wrapping taps)
       {
         int nc0 = c0, ntap0 = tap0 + TC;
         if (ntap0 >= k) { ntap0 = 0; nc0 = c0 + BK; }
-        nc0 = min(nc0, (int)CP - BK);
+        if (nc0 < CP) {
+          const int nntc = min(TC, k - ntap0);
 #pragma unroll
-        for (int r = 0; r < WUNITS; ++r) {
-          const int u = tid + 512 * r;
-          const int tc = min(ntap0 + u / (BN * 4), (int)k - 1);
-          const int rem = u % (BN * 4);
-          wpre[r] = *(const ulonglong2*)&w1[
-              ((long)tc * CP + (rem >> 2)) * CP + nc0 + (rem & 3) * 8];
+          for (int r = 0; r < WUNITS; ++r) {
+            const int u = tid + 512 * r;
+            if (u < nntc * BN * 4) {
+              const int tc = u / (BN * 4), rem = u % (BN * 4);
+              wpre[r] = *(const ulonglong2*)&w1[
+                  ((long)(ntap0 + tc) * CP + (rem >> 2)) * CP + nc0 +
+                  (rem & 3) * 8];
+            }
+          }
         }
       }
       __syncthreads();
@@ -187,13 +189,15 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
 #pragma unroll
     for (int j = 0; j < NT; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  // prologue for GEMM2\'s register ring (unconditional clamped loads)
+  // prologue for GEMM2\'s register ring
 #pragma unroll
   for (int r = 0; r < WUNITS; ++r) {
     const int u = tid + 512 * r;
-    const int tc = min(u / (BN * 4), (int)k - 1), rem = u % (BN * 4);
-    wpre[r] = *(const ulonglong2*)&w2[
-        ((long)tc * CP + (rem >> 2)) * CP + (rem & 3) * 8];
+    if (u < min(TC, k) * BN * 4) {
+      const int tc = u / (BN * 4), rem = u % (BN * 4);
+      wpre[r] = *(const ulonglong2*)&w2[
+          ((long)tc * CP + (rem >> 2)) * CP + (rem & 3) * 8];
+    }
   }
   for (int c0 = 0; c0 < CP; c0 += BK) {
     for (int tap0 = 0; tap0 < k; tap0 += TC) {
@@ -209,14 +213,18 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
       {
         int nc0 = c0, ntap0 = tap0 + TC;
         if (ntap0 >= k) { ntap0 = 0; nc0 = c0 + BK; }
-        nc0 = min(nc0, (int)CP - BK);
+        if (nc0 < CP) {
+          const int nntc = min(TC, k - ntap0);
 #pragma unroll
-        for (int r = 0; r < WUNITS; ++r) {
-          const int u = tid + 512 * r;
-          const int tc = min(ntap0 + u / (BN * 4), (int)k - 1);
-          const int rem = u % (BN * 4);
-          wpre[r] = *(const ulonglong2*)&w2[
-              ((long)tc * CP + (rem >> 2)) * CP + nc0 + (rem & 3) * 8];
+          for (int r = 0; r < WUNITS; ++r) {
+            const int u = tid + 512 * r;
+            if (u < nntc * BN * 4) {
+              const int tc = u / (BN * 4), rem = u % (BN * 4);
+              wpre[r] = *(const ulonglong2*)&w2[
+                  ((long)(ntap0 + tc) * CP + (rem >> 2)) * CP + nc0 +
+                  (rem & 3) * 8];
+            }
+          }
         }
       }
       __syncthreads();
