@@ -55,9 +55,6 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
   const int kl = lane >> 4;
   const int il = lane & 15;
 
-  constexpr int WUNITS = (TC * BN * 4 + 511) / 512;
-  ulonglong2 wpre[WUNITS];
-
   const bf16* xb = x + (long)b * T * C;
   const int pad1 = (k - 1) * dil / 2;
   // Xs[0] holds x row (t0 - h2 - pad1); GEMM1 xt row m taps rows m+j*dil
@@ -71,17 +68,6 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
   for (int i = 0; i < MT; ++i)
 #pragma unroll
     for (int j = 0; j < NT; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
-
-  // prologue: prefetch W1 chunk (0, 0) into registers
-#pragma unroll
-  for (int r = 0; r < WUNITS; ++r) {
-    const int u = tid + 512 * r;
-    if (u < min(TC, k) * BN * 4) {
-      const int tc = u / (BN * 4), rem = u % (BN * 4);
-      wpre[r] = *(const ulonglong2*)&w1[
-          ((long)tc * CP + (rem >> 2)) * CP + (rem & 3) * 8];
-    }
-  }
 
   for (int c0 = 0; c0 < CP; c0 += BK) {
     const bool c_interior = (c0 + BK) <= C;
@@ -112,32 +98,12 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
     }
     for (int tap0 = 0; tap0 < k; tap0 += TC) {
       const int ntc = min(TC, k - tap0);
-      // write the prefetched W chunk (register ring: loads were issued
-      // before the PREVIOUS chunk\'s MFMAs, so their latency is hidden)
-#pragma unroll
-      for (int r = 0; r < WUNITS; ++r) {
-        const int u = tid + 512 * r;
-        if (u < ntc * BN * 4) {
-          const int tc = u / (BN * 4), rem = u % (BN * 4);
-          *(ulonglong2*)&Ws[tc][rem >> 2][(rem & 3) * 8] = wpre[r];
-        }
-      }
-      // prefetch the NEXT chunk (possibly next K-slice, wrapping taps)
-      {
-        int nc0 = c0, ntap0 = tap0 + TC;
-        if (ntap0 >= k) { ntap0 = 0; nc0 = c0 + BK; }
-        if (nc0 < CP) {
-          const int nntc = min(TC, k - ntap0);
-#pragma unroll
-          for (int r = 0; r < WUNITS; ++r) {
-            const int u = tid + 512 * r;
-            if (u < nntc * BN * 4) {
-              const int tc = u / (BN * 4), rem = u % (BN * 4);
-              wpre[r] = *(const ulonglong2*)&w1[
-                  ((long)(ntap0 + tc) * CP + (rem >> 2)) * CP + nc0 +
-                  (rem & 3) * 8];
-            }
-          }
+      for (int tc = 0; tc < ntc; ++tc) {
+        const long wbase = ((long)(tap0 + tc) * CP) * CP + c0;
+        for (int u = tid; u < BN * 4; u += 512) {
+          const int n = u >> 2, ch = (u & 3) * 8;
+          *(ulonglong2*)&Ws[tc][n][ch] =
+              *(const ulonglong2*)&w1[wbase + (long)n * CP + ch];
         }
       }
       __syncthreads();
@@ -189,42 +155,15 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
 #pragma unroll
     for (int j = 0; j < NT; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  // prologue for GEMM2\'s register ring
-#pragma unroll
-  for (int r = 0; r < WUNITS; ++r) {
-    const int u = tid + 512 * r;
-    if (u < min(TC, k) * BN * 4) {
-      const int tc = u / (BN * 4), rem = u % (BN * 4);
-      wpre[r] = *(const ulonglong2*)&w2[
-          ((long)tc * CP + (rem >> 2)) * CP + (rem & 3) * 8];
-    }
-  }
   for (int c0 = 0; c0 < CP; c0 += BK) {
     for (int tap0 = 0; tap0 < k; tap0 += TC) {
       const int ntc = min(TC, k - tap0);
-#pragma unroll
-      for (int r = 0; r < WUNITS; ++r) {
-        const int u = tid + 512 * r;
-        if (u < ntc * BN * 4) {
-          const int tc = u / (BN * 4), rem = u % (BN * 4);
-          *(ulonglong2*)&Ws[tc][rem >> 2][(rem & 3) * 8] = wpre[r];
-        }
-      }
-      {
-        int nc0 = c0, ntap0 = tap0 + TC;
-        if (ntap0 >= k) { ntap0 = 0; nc0 = c0 + BK; }
-        if (nc0 < CP) {
-          const int nntc = min(TC, k - ntap0);
-#pragma unroll
-          for (int r = 0; r < WUNITS; ++r) {
-            const int u = tid + 512 * r;
-            if (u < nntc * BN * 4) {
-              const int tc = u / (BN * 4), rem = u % (BN * 4);
-              wpre[r] = *(const ulonglong2*)&w2[
-                  ((long)(ntap0 + tc) * CP + (rem >> 2)) * CP + nc0 +
-                  (rem & 3) * 8];
-            }
-          }
+      for (int tc = 0; tc < ntc; ++tc) {
+        const long wbase = ((long)(tap0 + tc) * CP) * CP + c0;
+        for (int u = tid; u < BN * 4; u += 512) {
+          const int n = u >> 2, ch = (u & 3) * 8;
+          *(ulonglong2*)&Ws[tc][n][ch] =
+              *(const ulonglong2*)&w2[wbase + (long)n * CP + ch];
         }
       }
       __syncthreads();
