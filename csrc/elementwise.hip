@@ -374,3 +374,67 @@ torch::Tensor mask_tail_(torch::Tensor x, torch::Tensor lens) {
   });
   return x;
 }
+
+// --------------------------------------------------------------------------
+// fused_gate_cl: channel-last WaveNet gate.
+// x [B, F, 2C] (+ optional g [B, 2C] speaker bias or [B, F, 2C]) ->
+// out [B, F, C] = tanh(xa + ga) * sigmoid(xb + gb).
+// Channel-last rows are contiguous: coalesced along C.
+// --------------------------------------------------------------------------
+template <typename T, int GMODE>  // 0: none, 1: [B,2C] broadcast, 2: full
+__global__ void fused_gate_cl_kernel(const T* __restrict__ x,
+                                     const T* __restrict__ g,
+                                     T* __restrict__ out, long C, long F_len,
+                                     long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;  // n = B*F*C
+  const long c = i % C;
+  const long bf = i / C;
+  const long b = bf / F_len;
+  const long ia = bf * 2 * C + c;
+  float va = ld_f(x + ia), vb = ld_f(x + ia + C);
+  if (GMODE == 1) {
+    va += ld_f(g + b * 2 * C + c);
+    vb += ld_f(g + b * 2 * C + C + c);
+  } else if (GMODE == 2) {
+    va += ld_f(g + ia);
+    vb += ld_f(g + ia + C);
+  }
+  st_f(out + i, tanhf(va) * sigmoidf_(vb));
+}
+
+torch::Tensor fused_gate_cl(torch::Tensor x, c10::optional<torch::Tensor> g,
+                            long n_channels) {
+  TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
+  const long B = x.size(0), F_len = x.size(1), C2 = x.size(2);
+  TORCH_CHECK(C2 == 2 * n_channels, "fused_gate_cl: channel mismatch");
+  auto out = torch::empty({B, F_len, n_channels}, x.options());
+  const long n = B * F_len * n_channels;
+  const int threads = 256;
+  const long blocks = (n + threads - 1) / threads;
+  DISPATCH_FT(x, "fused_gate_cl", {
+    if (!g.has_value()) {
+      hipLaunchKernelGGL((fused_gate_cl_kernel<scalar_t, 0>), dim3(blocks),
+                         dim3(threads), 0, cur_stream(),
+                         (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)nullptr,
+                         (scalar_t*)out.data_ptr(), n_channels, F_len, n);
+    } else if (g->dim() == 2 ||
+               (g->dim() == 3 && g->size(1) == 1)) {
+      TORCH_CHECK(g->numel() == B * C2, "fused_gate_cl: bad g");
+      hipLaunchKernelGGL((fused_gate_cl_kernel<scalar_t, 1>), dim3(blocks),
+                         dim3(threads), 0, cur_stream(),
+                         (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)g->contiguous().data_ptr(),
+                         (scalar_t*)out.data_ptr(), n_channels, F_len, n);
+    } else {
+      TORCH_CHECK(g->sizes() == x.sizes(), "fused_gate_cl: bad g");
+      hipLaunchKernelGGL((fused_gate_cl_kernel<scalar_t, 2>), dim3(blocks),
+                         dim3(threads), 0, cur_stream(),
+                         (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)g->contiguous().data_ptr(),
+                         (scalar_t*)out.data_ptr(), n_channels, F_len, n);
+    }
+  });
+  return out;
+}

@@ -10,6 +10,8 @@ torch::Tensor layer_norm_ct(torch::Tensor x, c10::optional<torch::Tensor> res,
                             double eps);
 torch::Tensor fused_gate(torch::Tensor x, c10::optional<torch::Tensor> g,
                          long n_channels);
+torch::Tensor fused_gate_cl(torch::Tensor x, c10::optional<torch::Tensor> g,
+                            long n_channels);
 torch::Tensor prior_sample(torch::Tensor m, torch::Tensor logs,
                            torch::Tensor mask, torch::Tensor noise,
                            double noise_scale);
@@ -50,6 +52,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "sonata_amd hand-written CDNA4 (gfx950) kernels";
   m.def("layer_norm_ct", &layer_norm_ct, "LayerNorm over channels of [B,C,T]");
   m.def("fused_gate", &fused_gate, "WaveNet tanh*sigmoid gate");
+  m.def("fused_gate_cl", &fused_gate_cl, "channel-last WaveNet gate");
   m.def("prior_sample", &prior_sample, "z=(m+eps*exp(logs)*ns)*mask");
   m.def("expand_states", &expand_states, "duration length-regulator gather");
   m.def("mask_tail_", &mask_tail_, "in-place zero of x[b,:,lens[b]:]");

@@ -24,6 +24,7 @@ from ..ops import (
     conv_mod,
     expand_states,
     fused_gate,
+    fused_gate_cl,
     layer_norm_ct,
     leaky_conv1d,
     leaky_conv1d_cl,
@@ -235,6 +236,35 @@ class WN(nn.Module):
                 output = output + res_skip
         return output * x_mask
 
+    def forward_cl(self, x: torch.Tensor, x_mask: torch.Tensor,
+                   g: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Channel-last WN: x [B,F,H], x_mask [B,F,1]; 1x1 convs as
+        hipBLASLt linears, k-tap convs as channel-last MFMA kernels,
+        gate fused (fused_gate_cl)."""
+        output = torch.zeros_like(x)
+        H = self.hidden
+        g_all = None
+        if g is not None and self.cond_layer is not None:
+            # g [B,gin,1] -> [B, 2H*n_layers] (per-utterance bias rows)
+            g_all = F.linear(g.squeeze(-1), self.cond_layer.weight.squeeze(-1),
+                             self.cond_layer.bias)
+        for i in range(self.n_layers):
+            conv = self.in_layers[i]
+            x_in = leaky_conv1d_cl(x, conv.weight, conv.bias,
+                                   padding=conv.padding[0],
+                                   dilation=conv.dilation[0])
+            g_l = (g_all[:, i * 2 * H:(i + 1) * 2 * H]
+                   if g_all is not None else None)
+            acts = fused_gate_cl(x_in, g_l, H)
+            rs = self.res_skip_layers[i]
+            res_skip = F.linear(acts, rs.weight.squeeze(-1), rs.bias)
+            if i < self.n_layers - 1:
+                x = (x + res_skip[..., :H]) * x_mask
+                output = output + res_skip[..., H:]
+            else:
+                output = output + res_skip
+        return output * x_mask
+
 
 class ResidualCouplingLayer(nn.Module):
     """Mean-only affine coupling with a WN conditioner."""
@@ -262,6 +292,17 @@ class ResidualCouplingLayer(nn.Module):
             x1 = (x1 - m) * x_mask
         return torch.cat([x0, x1], dim=1)
 
+    def reverse_cl(self, x: torch.Tensor, x_mask: torch.Tensor,
+                   g: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Channel-last inverse coupling: x [B,F,C], x_mask [B,F,1]."""
+        x0 = x[..., : self.half].contiguous()
+        x1 = x[..., self.half :]
+        h = F.linear(x0, self.pre.weight.squeeze(-1), self.pre.bias) * x_mask
+        h = self.enc.forward_cl(h, x_mask, g=g)
+        m = F.linear(h, self.post.weight.squeeze(-1), self.post.bias) * x_mask
+        x1 = (x1 - m) * x_mask
+        return torch.cat([x0, x1], dim=-1)
+
 
 class ResidualCouplingBlock(nn.Module):
     def __init__(self, channels: int, hidden: int, kernel_size: int,
@@ -285,6 +326,13 @@ class ResidualCouplingBlock(nn.Module):
             for flow in reversed(self.flows):
                 x = torch.flip(x, [1])
                 x = flow(x, x_mask, g=g, reverse=True)
+        return x
+
+    def reverse_cl(self, x: torch.Tensor, x_mask: torch.Tensor,
+                   g: Optional[torch.Tensor] = None) -> torch.Tensor:
+        for flow in reversed(self.flows):
+            x = torch.flip(x, [-1])
+            x = flow.reverse_cl(x, x_mask, g=g)
         return x
 
 
@@ -773,6 +821,16 @@ class VitsModel(nn.Module):
             x.device, x.dtype,
         )
         z_p = prior_sample(m_p_f, logs_p_f, y_mask, prior_noise, noise_scale)
+        from ..ops import use_hip
+
+        if use_hip(z_p):
+            # channel-last flow: one transpose in, one out (the decoder's
+            # cl path transposes again at entry; net cost ~zero, and all
+            # WN convs/gates/1x1s run on contiguous channel rows)
+            mask_cl = y_mask.transpose(1, 2)  # [B,F,1]
+            z_cl = self.flow.reverse_cl(
+                z_p.transpose(1, 2).contiguous(), mask_cl, g=g)
+            return z_cl.transpose(1, 2), y_mask, g
         z = self.flow(z_p, y_mask, g=g, reverse=True)
         return z, y_mask, g
 

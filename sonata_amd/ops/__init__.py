@@ -67,4 +67,5 @@ from .functional import (  # noqa: F401,E402
     leaky_conv1d_cl,
     leaky_convtranspose1d_cl,
     resblock_pair_cl,
+    fused_gate_cl,
 )

@@ -392,3 +392,19 @@ def resblock_pair_cl(
                          dilation=dilation, pre_lrelu=0.1, out_lens=out_lens)
     return leaky_conv1d_cl(xt, w2, b2, padding=(k - 1) // 2, pre_lrelu=0.1,
                            residual=x, out_lens=out_lens)
+
+
+def fused_gate_cl(
+    x: torch.Tensor, g: Optional[torch.Tensor], n_channels: int
+) -> torch.Tensor:
+    """Channel-last WaveNet gate: x [B,F,2C] (+ g [B,2C] speaker bias or
+    [B,F,2C]) -> tanh(xa+ga)*sigmoid(xb+gb) [B,F,C]."""
+    if use_hip(x):
+        ext = hip_ext(required=True)
+        return ext.fused_gate_cl(x.contiguous(), g, n_channels)
+    if g is not None:
+        if g.dim() == 2:
+            g = g.unsqueeze(1)
+        x = x + g
+    a, b = x[..., :n_channels], x[..., n_channels:]
+    return torch.tanh(a) * torch.sigmoid(b)

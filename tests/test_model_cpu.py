@@ -221,3 +221,26 @@ def test_vits_shapes_minimal():
     assert audio.shape[2] % 16 == 0  # hop = 4*4
     assert (alen <= audio.shape[2]).all()
     assert torch.isfinite(audio).all()
+
+
+def test_flow_reverse_cl_matches_channel_first():
+    """Channel-last flow inverse (GPU serving layout) == channel-first
+    oracle on CPU (torch fallback ops)."""
+    import torch
+
+    from sonata_amd.models.config import QUALITY_PRESETS, VitsArchitecture
+    from sonata_amd.models.vits import ResidualCouplingBlock, sequence_mask
+
+    torch.manual_seed(4)
+    arch = VitsArchitecture(**QUALITY_PRESETS["x_low"]["arch"])
+    flow = ResidualCouplingBlock(arch.inter_channels, arch.hidden_channels,
+                                 5, 1, 4).eval()
+    B, F = 2, 41
+    x = torch.randn(B, arch.inter_channels, F)
+    lens = torch.tensor([F, 30])
+    mask = sequence_mask(lens, F)
+    with torch.no_grad():
+        ref = flow(x * mask, mask, reverse=True)
+        got = flow.reverse_cl((x * mask).transpose(1, 2).contiguous(),
+                              mask.transpose(1, 2)).transpose(1, 2)
+    assert torch.allclose(got, ref, atol=1e-5)
