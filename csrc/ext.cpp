@@ -46,7 +46,9 @@ torch::Tensor convtranspose1d_cl_fused(torch::Tensor x, torch::Tensor w_perm,
 torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
                                      torch::Tensor b1, torch::Tensor w2_perm,
                                      torch::Tensor b2, long k, long dil,
-                                     c10::optional<torch::Tensor> out_lens);
+                                     c10::optional<torch::Tensor> out_lens,
+                                     c10::optional<torch::Tensor> accum,
+                                     double out_scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "sonata_amd hand-written CDNA4 (gfx950) kernels";

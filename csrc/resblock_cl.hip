@@ -28,8 +28,9 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
     const bf16* __restrict__ w2,   // [k][CP][CP] (d=1 conv)
     const float* __restrict__ b2,  // [C]
     bf16* __restrict__ out,        // [B][T][C]
+    const bf16* __restrict__ accum,  // optional: add (xs running sum)
     const int* __restrict__ out_lens,
-    int C, int CP, long T, int k, int dil) {
+    int C, int CP, long T, int k, int dil, float out_scale) {
   constexpr int WGM = 4;
   constexpr int WM = XTROWS / WGM;  // 32
   constexpr int MT = WM / 16;       // 2
@@ -189,8 +190,9 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
     }
   }
 
-  // epilogue2: bias2 + residual x + mask + store
+  // epilogue2: bias2 + residual x (+ xs accumulation) + scale + mask
   bf16* ob = out + (long)b * T * C;
+  const bf16* ab = accum ? accum + (long)b * T * C : nullptr;
 #pragma unroll
   for (int mi = 0; mi < MT; ++mi) {
 #pragma unroll
@@ -204,8 +206,11 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
         const int co = wc * WN + nj * 16 + il;
         if (co >= C) continue;
         float v = 0.f;
-        if (live)
+        if (live) {
           v = acc[mi][nj][rg] + b2[co] + bf2f(xb[t * C + co]);
+          if (ab) v += bf2f(ab[t * C + co]);
+          v *= out_scale;
+        }
         ob[t * C + co] = f2bf(v);
       }
     }
@@ -225,7 +230,9 @@ static inline hipStream_t cur_stream4() {
 torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
                                      torch::Tensor b1, torch::Tensor w2_perm,
                                      torch::Tensor b2, long k, long dil,
-                                     c10::optional<torch::Tensor> out_lens) {
+                                     c10::optional<torch::Tensor> out_lens,
+                                     c10::optional<torch::Tensor> accum,
+                                     double out_scale) {
   TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "resblock_cl: bf16 only");
   const long B = x.size(0), T = x.size(1), C = x.size(2);
@@ -243,6 +250,11 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
     TORCH_CHECK(out_lens->scalar_type() == at::kInt && out_lens->is_cuda());
     lens_p = out_lens->data_ptr<int>();
   }
+  const bf16* accum_p = nullptr;
+  if (accum.has_value()) {
+    TORCH_CHECK(accum->sizes() == x.sizes() && accum->is_contiguous());
+    accum_p = (const bf16*)accum->data_ptr();
+  }
   const long BM = XTROWS - (k - 1);
   hipStream_t st = cur_stream4();
 #define LAUNCH_RB(BN, WGN, TC)                                              \
@@ -252,8 +264,9 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
                      (const bf16*)w1_perm.data_ptr(),                       \
                      b1f.data_ptr<float>(),                                 \
                      (const bf16*)w2_perm.data_ptr(),                       \
-                     b2f.data_ptr<float>(), (bf16*)out.data_ptr(), lens_p,  \
-                     (int)C, CP, T, (int)k, (int)dil)
+                     b2f.data_ptr<float>(), (bf16*)out.data_ptr(), accum_p, \
+                     lens_p, (int)C, CP, T, (int)k, (int)dil,               \
+                     (float)out_scale)
   if (CP == 256) LAUNCH_RB(256, 2, 2);
   else if (CP == 128) LAUNCH_RB(128, 2, 2);
   else if (CP == 64) LAUNCH_RB(64, 2, 2);

@@ -619,15 +619,22 @@ class ResBlock1(nn.Module):
         return x
 
     def forward_cl(self, x: torch.Tensor,
-                   lengths: Optional[torch.Tensor] = None) -> torch.Tensor:
+                   lengths: Optional[torch.Tensor] = None,
+                   accum: Optional[torch.Tensor] = None,
+                   out_scale: float = 1.0) -> torch.Tensor:
         """Channel-last ([B,T,C]) path: each conv pair runs as ONE fused
-        kernel with the intermediate tensor in LDS
-        (csrc/resblock_cl.hip); numerically equivalent to forward()."""
+        kernel with the intermediate tensor in LDS (csrc/resblock_cl.hip).
+        `accum`/`out_scale` apply to the LAST pair (MRF sum fusion);
+        numerically equivalent to forward() then +accum, *out_scale."""
         from ..ops import resblock_pair_cl
 
-        for c1, c2 in zip(self.convs1, self.convs2):
+        n = len(self.convs1)
+        for i, (c1, c2) in enumerate(zip(self.convs1, self.convs2)):
+            last = i == n - 1
             x = resblock_pair_cl(x, c1.weight, c1.bias, c2.weight, c2.bias,
-                                 dilation=c1.dilation[0], out_lens=lengths)
+                                 dilation=c1.dilation[0], out_lens=lengths,
+                                 accum=accum if last else None,
+                                 out_scale=out_scale if last else 1.0)
         return x
 
 
@@ -719,10 +726,11 @@ class Generator(nn.Module):
             )
             xs = None
             for j in range(self.num_kernels):
-                out = self.resblocks[i * self.num_kernels + j].forward_cl(
-                    x, lengths)
-                xs = out if xs is None else xs + out
-            x = xs / self.num_kernels
+                last = j == self.num_kernels - 1
+                xs = self.resblocks[i * self.num_kernels + j].forward_cl(
+                    x, lengths, accum=xs,
+                    out_scale=1.0 / self.num_kernels if last else 1.0)
+            x = xs
         x = leaky_conv1d_cl(x, self.conv_post.weight, None, padding=3,
                             pre_lrelu=LRELU_SLOPE, post_tanh=True,
                             out_lens=lengths)

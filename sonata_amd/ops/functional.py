@@ -375,10 +375,13 @@ def resblock_pair_cl(
     w2: torch.Tensor, b2: Optional[torch.Tensor],
     dilation: int,
     out_lens: Optional[torch.Tensor] = None,
+    accum: Optional[torch.Tensor] = None,
+    out_scale: float = 1.0,
 ) -> torch.Tensor:
-    """Fused resblock conv pair: conv2_{k,1}(lrelu(conv1_{k,d}(lrelu(x))))
-    + x, channel-last, intermediate tensor kept in LDS
-    (csrc/resblock_cl.hip).  Falls back to two conv calls elsewhere."""
+    """Fused resblock conv pair: (conv2_{k,1}(lrelu(conv1_{k,d}(lrelu(x))))
+    + x [+ accum]) * out_scale, channel-last, intermediate tensor kept in
+    LDS (csrc/resblock_cl.hip).  `accum`/`out_scale` fold the MRF
+    cross-resblock sum and /num_kernels into the epilogue."""
     Cout, Cin, k = w1.shape
     if (use_hip(x) and Cin == Cout and b1 is not None and b2 is not None
             and (k - 1) * dilation <= 64):
@@ -387,11 +390,18 @@ def resblock_pair_cl(
             x.contiguous(), _conv_weight_mfma(w1), _bias_f32(b1),
             _conv_weight_mfma(w2), _bias_f32(b2), k, dilation,
             _lens_i32(out_lens, x.device),
+            accum.contiguous() if accum is not None else None,
+            float(out_scale),
         )
     xt = leaky_conv1d_cl(x, w1, b1, padding=(k - 1) * dilation // 2,
                          dilation=dilation, pre_lrelu=0.1, out_lens=out_lens)
-    return leaky_conv1d_cl(xt, w2, b2, padding=(k - 1) // 2, pre_lrelu=0.1,
-                           residual=x, out_lens=out_lens)
+    y = leaky_conv1d_cl(xt, w2, b2, padding=(k - 1) // 2, pre_lrelu=0.1,
+                        residual=x, out_lens=out_lens)
+    if accum is not None:
+        y = y + accum
+    if out_scale != 1.0:
+        y = y * out_scale
+    return y
 
 
 def fused_gate_cl(
