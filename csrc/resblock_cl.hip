@@ -20,7 +20,7 @@
 #define BKP 40
 #define XTROWS 128
 
-template <int BN, int WGN, int TC>
+template <int BN, int WGN, int TC, int XR>
 __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
     const bf16* __restrict__ x,    // [B][T][C]
     const bf16* __restrict__ w1,   // [k][CP][CP] (dilated conv)
@@ -36,8 +36,8 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
   constexpr int MT = WM / 16;       // 2
   constexpr int WN = BN / WGN;
   constexpr int NT = WN / 16;
-  constexpr int XROWS_MAX = XTROWS + 64;  // + (k-1)*dil <= 50
-  constexpr int XTP = BN + 8;             // xt row pitch
+  constexpr int XROWS_MAX = XR;     // 128 + (k-1)*dil, bucketed
+  constexpr int XTP = BN + 4;       // xt row pitch (16B-aligned, odd/2 dw)
 
   const int h2 = (k - 1) / 2;        // conv2 halo per side
   const int BM = XTROWS - (k - 1);   // output rows per block
@@ -257,8 +257,8 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
   }
   const long BM = XTROWS - (k - 1);
   hipStream_t st = cur_stream4();
-#define LAUNCH_RB(BN, WGN, TC)                                              \
-  hipLaunchKernelGGL((resblock_pair_cl_kernel<BN, WGN, TC>),                \
+#define LAUNCH_RB(BN, WGN, TC, XR)                                          \
+  hipLaunchKernelGGL((resblock_pair_cl_kernel<BN, WGN, TC, XR>),            \
                      dim3(ceil_div(T, BM), 1, B), dim3(512), 0, st,         \
                      (const bf16*)x.data_ptr(),                             \
                      (const bf16*)w1_perm.data_ptr(),                       \
@@ -267,11 +267,22 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
                      b2f.data_ptr<float>(), (bf16*)out.data_ptr(), accum_p, \
                      lens_p, (int)C, CP, T, (int)k, (int)dil,               \
                      (float)out_scale)
-  if (CP == 256) LAUNCH_RB(256, 2, 2);
-  else if (CP == 128) LAUNCH_RB(128, 2, 2);
-  else if (CP == 64) LAUNCH_RB(64, 2, 2);
-  else if (CP == 32) LAUNCH_RB(32, 2, 2);
+  // XR bucket = 128 + (k-1)*dil rounded up; TC=3 covers k=3 in one
+  // chunk and k=7/11 in 3/4 chunks (fewer barrier pairs); C=256 keeps
+  // TC=2 for LDS.
+  const int xrows = 128 + (int)((k - 1) * dil);
+#define RB_XR(BN, WGN, TC)                                                  \
+  do {                                                                      \
+    if (xrows <= 144) LAUNCH_RB(BN, WGN, TC, 144);                          \
+    else if (xrows <= 160) LAUNCH_RB(BN, WGN, TC, 160);                     \
+    else LAUNCH_RB(BN, WGN, TC, 192);                                       \
+  } while (0)
+  if (CP == 256) RB_XR(256, 2, 2);
+  else if (CP == 128) RB_XR(128, 2, 3);
+  else if (CP == 64) RB_XR(64, 2, 3);
+  else if (CP == 32) RB_XR(32, 2, 3);
   else TORCH_CHECK(false, "resblock_cl: unsupported CP ", CP);
+#undef RB_XR
 #undef LAUNCH_RB
   return out;
 }
