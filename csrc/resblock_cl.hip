@@ -279,8 +279,8 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
   } while (0)
   if (CP == 256) RB_XR(256, 2, 2);
   else if (CP == 128) RB_XR(128, 2, 3);
-  else if (CP == 64) RB_XR(64, 2, 3);
-  else if (CP == 32) RB_XR(32, 2, 3);
+  else if (CP == 64) RB_XR(64, 2, 2);
+  else if (CP == 32) RB_XR(32, 2, 2);
   else TORCH_CHECK(false, "resblock_cl: unsupported CP ", CP);
 #undef RB_XR
 #undef LAUNCH_RB
