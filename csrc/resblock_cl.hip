@@ -37,7 +37,8 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
   constexpr int WN = BN / WGN;
   constexpr int NT = WN / 16;
   constexpr int XROWS_MAX = XR;     // 128 + (k-1)*dil, bucketed
-  constexpr int XTP = BN + 4;       // xt row pitch (16B-aligned, odd/2 dw)
+  constexpr int XTP = BN + 8;       // xt row pitch: 16B-aligned rows,
+                                    // 16 distinct banks across il lanes
 
   const int h2 = (k - 1) / 2;        // conv2 halo per side
   const int BM = XTROWS - (k - 1);   // output rows per block
@@ -240,7 +241,7 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
   const int CP = w1_perm.size(2);
   TORCH_CHECK(w1_perm.size(1) == CP && w2_perm.size(1) == CP &&
               w2_perm.size(2) == CP, "resblock_cl: square channel conv");
-  TORCH_CHECK((k - 1) * dil <= 64, "resblock_cl: halo too large");
+  TORCH_CHECK((k - 1) * dil <= 52, "resblock_cl: halo too large");
   auto out = torch::empty_like(x);
   if (out.numel() == 0) return out;
   auto b1f = b1.scalar_type() == at::kFloat ? b1 : b1.to(at::kFloat).contiguous();
@@ -275,7 +276,7 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
   do {                                                                      \
     if (xrows <= 144) LAUNCH_RB(BN, WGN, TC, 144);                          \
     else if (xrows <= 160) LAUNCH_RB(BN, WGN, TC, 160);                     \
-    else LAUNCH_RB(BN, WGN, TC, 192);                                       \
+    else LAUNCH_RB(BN, WGN, TC, 180);                                       \
   } while (0)
   if (CP == 256) RB_XR(256, 2, 2);
   else if (CP == 128) RB_XR(128, 2, 3);
