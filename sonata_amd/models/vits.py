@@ -133,6 +133,35 @@ class RelativeAttention(nn.Module):
         out = out.transpose(2, 3).contiguous().view(B, C, T)
         return conv_mod(self.conv_o, out)
 
+    def forward_cl(self, x: torch.Tensor,
+                   attn_mask: torch.Tensor) -> torch.Tensor:
+        """Channel-last attention: x [B,T,C]; the [B,h,T,d] head view is a
+        free reshape of channel-last rows (no transpose of C against T),
+        and all projections are hipBLASLt linears."""
+        B, T, C = x.shape
+        h, d = self.n_heads, self.head_dim
+
+        def proj(conv, t):
+            return F.linear(t, conv.weight.squeeze(-1), conv.bias)
+
+        q = proj(self.conv_q, x).view(B, T, h, d).transpose(1, 2)
+        k = proj(self.conv_k, x).view(B, T, h, d).transpose(1, 2)
+        v = proj(self.conv_v, x).view(B, T, h, d).transpose(1, 2)
+        scale = d ** -0.5
+        scores = torch.matmul(q * scale, k.transpose(-2, -1))
+        rel_k = self._rel_embeddings(self.emb_rel_k, T)
+        rel_logits = torch.matmul(q * scale,
+                                  rel_k.unsqueeze(0).transpose(-2, -1))
+        scores = scores + self._rel_to_abs(rel_logits)
+        scores = scores.masked_fill(attn_mask == 0, -1e4)
+        p = torch.softmax(scores, dim=-1)
+        out = torch.matmul(p, v)
+        rel_w = self._abs_to_rel(p)
+        rel_v = self._rel_embeddings(self.emb_rel_v, T)
+        out = out + torch.matmul(rel_w, rel_v.unsqueeze(0))
+        out = out.transpose(1, 2).reshape(B, T, C)
+        return proj(self.conv_o, out)
+
 
 class FFN(nn.Module):
     """Conv1d(k) -> ReLU -> Conv1d(k), masked."""
@@ -173,6 +202,10 @@ class TextEncoder(nn.Module):
     def forward(
         self, ids: torch.Tensor, lengths: torch.Tensor
     ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
+        from ..ops import use_hip
+
+        if use_hip(ids):
+            return self._forward_cl(ids, lengths)
         x = self.emb(ids) * math.sqrt(self.hidden_channels)  # [B,T,H]
         x = x.transpose(1, 2)  # [B,H,T]
         x_mask = sequence_mask(lengths, ids.shape[1]).to(x.dtype)
@@ -187,6 +220,39 @@ class TextEncoder(nn.Module):
         stats = conv_mod(self.proj, x) * x_mask
         m, logs = stats.chunk(2, dim=1)
         return x, m, logs, x_mask
+
+    def _forward_cl(
+        self, ids: torch.Tensor, lengths: torch.Tensor
+    ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
+        """GPU path in the native channel-last layout: the embedding is
+        already [B,T,H]; attention/FFN/LN all run on contiguous rows and
+        only the (small) encoder outputs transpose to [B,C,T] for the
+        duration/prior stages.  Numerics match forward()."""
+        x = self.emb(ids) * math.sqrt(self.hidden_channels)  # [B,T,H]
+        x_mask_cl = sequence_mask(lengths, ids.shape[1]).to(
+            x.dtype).transpose(1, 2)  # [B,T,1]
+        attn_mask = (x_mask_cl * x_mask_cl.transpose(1, 2))  # [B,T,T]
+        x = x * x_mask_cl
+        for attn, n1, ffn, n2 in zip(self.attn_layers, self.norm1,
+                                     self.ffn_layers, self.norm2):
+            y = attn.forward_cl(x * x_mask_cl, attn_mask.unsqueeze(1))
+            x = F.layer_norm(x + y, (x.shape[-1],), n1.gamma, n1.beta,
+                             n1.eps)
+            f = leaky_conv1d_cl(x * x_mask_cl, ffn.conv1.weight,
+                                ffn.conv1.bias,
+                                padding=ffn.conv1.padding[0], post_relu=True,
+                                out_lens=lengths)
+            f = leaky_conv1d_cl(f, ffn.conv2.weight, ffn.conv2.bias,
+                                padding=ffn.conv2.padding[0],
+                                out_lens=lengths)
+            x = F.layer_norm(x + f, (x.shape[-1],), n2.gamma, n2.beta,
+                             n2.eps)
+        stats = F.linear(x, self.proj.weight.squeeze(-1), self.proj.bias)
+        stats = (stats * x_mask_cl).transpose(1, 2)  # [B,2C,T]
+        m, logs = stats.chunk(2, dim=1)
+        x_mask = x_mask_cl.transpose(1, 2)
+        return x.transpose(1, 2) * x_mask, m.contiguous(), \
+            logs.contiguous(), x_mask
 
 
 # --------------------------------------------------------------------------- #

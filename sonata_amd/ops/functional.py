@@ -297,17 +297,19 @@ def leaky_conv1d_cl(
     pre_lrelu: float = 0.0,
     post_lrelu: float = 0.0,
     post_tanh: bool = False,
+    post_relu: bool = False,
     residual: Optional[torch.Tensor] = None,
     out_lens: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """Channel-last stride-1 Conv1d with fused input/output LeakyReLU (or
-    tanh), residual add and ragged-batch row masking.  The MI355X-native
-    layout: Cin contiguous makes both MFMA operands k-contiguous
-    (csrc/conv1d_cl.hip)."""
+    tanh / relu), residual add and ragged-batch row masking.  The
+    MI355X-native layout: Cin contiguous makes both MFMA operands
+    k-contiguous (csrc/conv1d_cl.hip)."""
     if use_hip(x):
         ext = hip_ext(required=True)
         Cout, _, k = weight.shape
-        act = 2 if post_tanh else (1 if post_lrelu > 0.0 else 0)
+        act = 2 if post_tanh else (
+            1 if (post_lrelu > 0.0 or post_relu) else 0)
         return ext.conv1d_cl_fused(
             x.contiguous(), _conv_weight_mfma(weight), _bias_f32(bias),
             Cout, k, padding, dilation,
@@ -323,6 +325,8 @@ def leaky_conv1d_cl(
     y = F.conv1d(xc, weight, bias, padding=padding, dilation=dilation)
     if post_tanh:
         y = torch.tanh(y)
+    elif post_relu:
+        y = torch.relu(y)
     elif post_lrelu > 0.0:
         y = F.leaky_relu(y, post_lrelu)
     y = y.transpose(1, 2)

@@ -448,3 +448,25 @@ def test_multispeaker_gpu_synthesis(dev):
         # different speakers give different audio
         n = min(len(outs[0]), len(outs[1]))
         assert np.abs(outs[0][:n] - outs[1][:n]).max() > 1e-4
+
+
+def test_text_encoder_cl_matches_oracle(dev):
+    """Channel-last GPU text encoder vs fp32 CPU oracle."""
+    from sonata_amd.models.config import QUALITY_PRESETS, VitsArchitecture
+    from sonata_amd.models.vits import TextEncoder
+
+    torch.manual_seed(21)
+    arch = VitsArchitecture(**QUALITY_PRESETS["x_low"]["arch"])
+    enc = TextEncoder(130, arch.inter_channels, arch).eval()
+    B, T = 3, 57
+    ids = torch.randint(3, 120, (B, T))
+    ids[:, 0] = 1
+    lens = torch.tensor([T, 40, 22])
+    with torch.no_grad():
+        xr, mr, lr_, maskr = enc(ids, lens)
+        encg = enc.to(dev, torch.bfloat16)
+        xg, mg, lg, maskg = encg(ids.to(dev), lens.to(dev))
+    for b, ln in enumerate(lens.tolist()):
+        assert _rel_err(xg[b, :, :ln], xr[b, :, :ln]) < 0.05
+        assert _rel_err(mg[b, :, :ln], mr[b, :, :ln]) < 0.05
+        assert _rel_err(lg[b, :, :ln], lr_[b, :, :ln]) < 0.05
