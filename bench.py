@@ -52,8 +52,10 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--batch", type=int, default=32,
-                    help="utterances per rank per step")
+    ap.add_argument("--batch", type=int, default=64,
+                    help="utterances per rank per step (64/rank makes the "
+                         "8-GPU run exactly the 512-concurrent-utterance "
+                         "serving config of BASELINE.json)")
     ap.add_argument("--seq-len", type=int, default=256,
                     help="phoneme ids per utterance (PAD-interleaved)")
     ap.add_argument("--quality", default="medium")
