@@ -289,6 +289,54 @@ _AR_RULES = {
 }
 
 
+# --------------------------------------------------------------------------- #
+# French
+# --------------------------------------------------------------------------- #
+_FR_RULES = {
+    "eau": "o", "eaux": "o", "aux": "o", "eux": "ø", "oeu": "œ",
+    "ain": "ɛ̃", "ein": "ɛ̃", "aim": "ɛ̃", "oin": "wɛ̃",
+    "tion": "sjɔ̃", "ille": "ij", "gn": "ɲ", "ch": "ʃ", "ph": "f",
+    "qu": "k", "ou": "u", "oi": "wa", "au": "o", "ai": "ɛ", "ei": "ɛ",
+    "eu": "ø", "an": "ɑ̃", "am": "ɑ̃", "en": "ɑ̃", "em": "ɑ̃",
+    "on": "ɔ̃", "om": "ɔ̃", "in": "ɛ̃", "im": "ɛ̃", "un": "œ̃",
+    "er ": "e ", "ez ": "e ", "es ": " ", "é": "e", "è": "ɛ", "ê": "ɛ",
+    "à": "a", "â": "ɑ", "î": "i", "ô": "o", "û": "y", "ç": "s",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "ə", "f": "f",
+    "g": "ɡ", "h": "", "i": "i", "j": "ʒ", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "ɔ", "p": "p", "r": "ʁ", "s": "s",
+    "t": "t", "u": "y", "v": "v", "w": "w", "x": "ks", "y": "i",
+    "z": "z",
+}
+
+# --------------------------------------------------------------------------- #
+# Italian
+# --------------------------------------------------------------------------- #
+_IT_RULES = {
+    "gli": "ʎi", "gn": "ɲ", "sci": "ʃi", "sce": "ʃe", "chi": "ki",
+    "che": "ke", "ghi": "ɡi", "ghe": "ɡe", "ci": "tʃi", "ce": "tʃe",
+    "gi": "dʒi", "ge": "dʒe", "zz": "tts", "qu": "kw",
+    "à": "a", "è": "ɛ", "é": "e", "ì": "i", "ò": "ɔ", "ù": "u",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "", "i": "i", "l": "l", "m": "m", "n": "n",
+    "o": "o", "p": "p", "r": "r", "s": "s", "t": "t", "u": "u",
+    "v": "v", "z": "dz",
+}
+
+# --------------------------------------------------------------------------- #
+# Portuguese (pt-BR leaning)
+# --------------------------------------------------------------------------- #
+_PT_RULES = {
+    "lh": "ʎ", "nh": "ɲ", "ch": "ʃ", "qu": "k", "gu": "ɡ", "rr": "ʁ",
+    "ão": "ɐ̃w̃", "õe": "õj̃", "ã": "ɐ̃", "õ": "õ",
+    "á": "a", "â": "ɐ", "é": "ɛ", "ê": "e", "í": "i", "ó": "ɔ",
+    "ô": "o", "ú": "u", "ç": "s",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "", "i": "i", "j": "ʒ", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "o", "p": "p", "r": "ɾ", "s": "s",
+    "t": "t", "u": "u", "v": "v", "x": "ʃ", "z": "z",
+}
+
+
 _G2P_REGISTRY: Dict[str, RuleG2P] = {}
 
 
@@ -303,6 +351,12 @@ def _get_g2p(voice: str) -> RuleG2P:
         g = RuleG2P(_DE_RULES, letters="a-zA-Zäöüß")
     elif base == "es":
         g = RuleG2P(_ES_RULES, letters="a-zA-Zñáéíóúü")
+    elif base == "fr":
+        g = RuleG2P(_FR_RULES, letters="a-zA-Zàâçéèêëîïôûùüœ")
+    elif base == "it":
+        g = RuleG2P(_IT_RULES, letters="a-zA-Zàèéìòù")
+    elif base == "pt":
+        g = RuleG2P(_PT_RULES, letters="a-zA-Zàáâãçéêíóôõú")
     elif base == "ar":
         g = RuleG2P(
             _AR_RULES,
@@ -316,7 +370,7 @@ def _get_g2p(voice: str) -> RuleG2P:
 
 
 def available_languages() -> List[str]:
-    return ["en-us", "en", "de", "es", "ar"]
+    return ["en-us", "en", "de", "es", "fr", "it", "pt", "ar"]
 
 
 _LANG_SWITCH_RE = re.compile(r"\([a-z-]{2,10}\)")

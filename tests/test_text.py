@@ -126,3 +126,15 @@ def test_tashkeel_save_load(tmp_path):
     m.save(p)
     m2 = TashkeelModel.load(p)
     assert m.diacritize("سلام") == m2.diacritize("سلام")
+
+
+def test_additional_languages():
+    from sonata_amd.text.phonemizer import available_languages, text_to_phonemes
+
+    assert set(["fr", "it", "pt"]) <= set(available_languages())
+    for lang, text in [("fr", "Bonjour le monde. Comment allez-vous?"),
+                       ("it", "Ciao mondo. Come stai?"),
+                       ("pt", "Olá mundo. Tudo bem?")]:
+        sents = text_to_phonemes(text, voice=lang)
+        assert len(sents) == 2, (lang, sents)
+        assert all(len(x) > 2 for x in sents)
