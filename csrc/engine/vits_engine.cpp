@@ -44,6 +44,14 @@ torch::Tensor convtranspose1d_cl_fused(torch::Tensor x, torch::Tensor w_perm,
                                        long Cout, long k, long stride,
                                        long padding, double pre_lrelu,
                                        c10::optional<torch::Tensor> out_lens);
+torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
+                                     torch::Tensor b1, torch::Tensor w2_perm,
+                                     torch::Tensor b2, long k, long dil,
+                                     c10::optional<torch::Tensor> out_lens,
+                                     c10::optional<torch::Tensor> accum,
+                                     double out_scale);
+torch::Tensor fused_gate_cl(torch::Tensor x, c10::optional<torch::Tensor> g,
+                            long n_channels);
 
 namespace sonata {
 
@@ -459,6 +467,60 @@ torch::Tensor VitsEngine::wn(torch::Tensor x, torch::Tensor mask,
 torch::Tensor VitsEngine::flow_reverse(torch::Tensor x, torch::Tensor mask,
                                        c10::optional<torch::Tensor> g) const {
   long half = cfg_.inter / 2;
+  if (gpu() && x.scalar_type() == torch::kBFloat16) {
+    // channel-last flow (mirrors vits.py reverse_cl): WN k-tap convs as
+    // cl MFMA kernels, 1x1s as linears, gate via fused_gate_cl
+    auto lin = [&](torch::Tensor t, const std::string& mod) {
+      auto w = p(mod + ".weight").squeeze(-1);
+      return torch::linear(t, w, p(mod + ".bias"));
+    };
+    auto xc = x.transpose(1, 2).contiguous();       // [B,F,C]
+    auto mc = mask.transpose(1, 2).contiguous();    // [B,F,1]
+    const long H = cfg_.hidden;
+    for (long f = 3; f >= 0; --f) {
+      xc = torch::flip(xc, {-1});
+      std::string mod = "flow.flows." + std::to_string(f);
+      auto x0 = xc.narrow(-1, 0, half).contiguous();
+      auto x1 = xc.narrow(-1, half, half);
+      auto h = lin(x0, mod + ".pre") * mc;
+      // WN channel-last
+      {
+        auto output = torch::zeros_like(h);
+        c10::optional<torch::Tensor> g_all;
+        if (g.has_value() && has(mod + ".enc.cond_layer.weight"))
+          g_all = torch::linear(
+              g->squeeze(-1),
+              p(mod + ".enc.cond_layer.weight").squeeze(-1),
+              p(mod + ".enc.cond_layer.bias"));
+        for (long i = 0; i < 4; ++i) {
+          std::string li = std::to_string(i);
+          auto cw = p(mod + ".enc.in_layers." + li + ".weight");
+          auto x_in = conv1d_cl_fused(
+              h.contiguous(), perm_conv(mod + ".enc.in_layers." + li +
+                                        ".weight"),
+              bias_f32(mod + ".enc.in_layers." + li + ".bias"),
+              cw.size(0), cw.size(2), (cw.size(2) - 1) / 2, 1, -1.0, 0, 0.0,
+              c10::nullopt, c10::nullopt);
+          c10::optional<torch::Tensor> g_l;
+          if (g_all.has_value())
+            g_l = g_all->narrow(-1, i * 2 * H, 2 * H);
+          auto acts = fused_gate_cl(x_in, g_l, H);
+          auto res_skip = lin(acts, mod + ".enc.res_skip_layers." + li);
+          if (i < 3) {
+            h = (h + res_skip.narrow(-1, 0, H)) * mc;
+            output = output + res_skip.narrow(-1, H, H);
+          } else {
+            output = output + res_skip;
+          }
+        }
+        h = output * mc;
+      }
+      auto m = lin(h, mod + ".post") * mc;
+      x1 = (x1 - m) * mc;
+      xc = torch::cat({x0, x1}, -1);
+    }
+    return xc.transpose(1, 2).contiguous();
+  }
   for (long f = 3; f >= 0; --f) {
     x = torch::flip(x, {1});
     std::string mod = "flow.flows." + std::to_string(f);
@@ -700,28 +762,32 @@ torch::Tensor VitsEngine::generator(torch::Tensor x,
       xc = convtranspose1d_cl_fused(
           xc, perm_convt(ui + ".weight", s), bias_f32(ui + ".bias"),
           p(ui + ".weight").size(1), k, s, (k - s) / 2, kLRelu, to32(lens));
+      // fused resblock pairs; the MRF sum and /num_kernels fold into
+      // the last pair\'s epilogue (same as the Python path)
       torch::Tensor xs;
       for (long j = 0; j < n_kernels; ++j) {
         std::string rb =
             "dec.resblocks." + std::to_string(i * n_kernels + j);
         auto out = xc;
         long kk = cfg_.resblock_ks[j];
-        for (size_t di = 0; di < cfg_.resblock_dil[j].size(); ++di) {
+        const size_t npair = cfg_.resblock_dil[j].size();
+        for (size_t di = 0; di < npair; ++di) {
           long d = cfg_.resblock_dil[j][di];
           std::string c1 = rb + ".convs1." + std::to_string(di);
           std::string c2 = rb + ".convs2." + std::to_string(di);
-          auto xt = conv1d_cl_fused(
+          const bool last = (di == npair - 1);
+          const bool last_rb = (j == n_kernels - 1);
+          c10::optional<torch::Tensor> accum;
+          if (last && xs.defined()) accum = xs;
+          out = resblock_pair_cl_fused(
               out, perm_conv(c1 + ".weight"), bias_f32(c1 + ".bias"),
-              p(c1 + ".weight").size(0), kk, (kk - 1) * d / 2, d, kLRelu, 0,
-              0.0, c10::nullopt, to32(lens));
-          out = conv1d_cl_fused(
-              xt, perm_conv(c2 + ".weight"), bias_f32(c2 + ".bias"),
-              p(c2 + ".weight").size(0), kk, (kk - 1) / 2, 1, kLRelu, 0, 0.0,
-              out.contiguous(), to32(lens));
+              perm_conv(c2 + ".weight"), bias_f32(c2 + ".bias"), kk, d,
+              to32(lens), accum,
+              (last && last_rb) ? 1.0 / (double)n_kernels : 1.0);
         }
-        xs = xs.defined() ? xs + out : out;
+        xs = out;
       }
-      xc = xs / (double)n_kernels;
+      xc = xs;
     }
     xc = conv1d_cl_fused(xc, perm_conv("dec.conv_post.weight"), c10::nullopt,
                          1, 7, 3, 1, kLRelu, 2 /*tanh*/, 0.0, c10::nullopt,
