@@ -6,6 +6,7 @@
 #include "vits_engine.h"
 
 #include <ATen/ATen.h>
+#include <ATen/detail/CUDAHooksInterface.h>
 #include <torch/types.h>
 
 #include <cmath>
@@ -846,17 +847,18 @@ torch::Tensor VitsEngine::masked_noise(long B, long C, long T_max,
                                        torch::Tensor lengths,
                                        std::vector<torch::Generator>& gens)
     const {
-  // CPU fp32 generation (deterministic across devices), then cast.
-  auto out = torch::zeros({B, C, T_max}, torch::kFloat32);
+  // Per-utterance generators, on-device when serving a GPU (matches the
+  // Python path: voice.py _generators uses torch.Generator(device)); the
+  // CPU path generates fp32 host-side (the numerics oracle).
+  auto opts = torch::TensorOptions().device(device_).dtype(dtype_);
+  auto out = torch::zeros({B, C, T_max}, opts);
   auto lens_cpu = lengths.to(torch::kCPU);
   for (long b = 0; b < B; ++b) {
     long lb = lens_cpu[b].item<long>();
     if (lb <= 0) continue;
-    auto n = torch::randn({C, lb}, gens[b],
-                          torch::TensorOptions().dtype(torch::kFloat32));
-    out[b].narrow(1, 0, lb).copy_(n);
+    out[b].narrow(1, 0, lb) = torch::randn({C, lb}, gens[b], opts);
   }
-  return out.to(device_, dtype_);
+  return out;
 }
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>
@@ -870,7 +872,9 @@ VitsEngine::infer_encoder(torch::Tensor ids, torch::Tensor lengths,
   long B = ids.size(0);
   std::vector<torch::Generator> gens;
   for (long b = 0; b < B; ++b) {
-    auto g = at::detail::createCPUGenerator();
+    auto g = device_.is_cuda()
+                 ? at::detail::getCUDAHooks().getNewGenerator(device_.index())
+                 : at::detail::createCPUGenerator();
     g.set_current_seed(b < (long)seeds.size() ? seeds[b] : 1234 + b);
     gens.push_back(g);
   }
