@@ -60,6 +60,9 @@ def main():
                     help="phoneme ids per utterance (PAD-interleaved)")
     ap.add_argument("--quality", default="medium")
     ap.add_argument("--device", default=None)
+    ap.add_argument("--engine", choices=["cpp", "python"], default="cpp",
+                    help="serving runtime: the C++ VitsEngine (default) or "
+                         "the Python model path (same kernels either way)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -94,20 +97,46 @@ def main():
         architecture=arch,
     )
     torch.manual_seed(0)  # same weights on every rank
-    net = VitsModel(config.num_symbols, arch, n_speakers=1).eval()
-    dtype = torch.bfloat16 if use_gpu else torch.float32
-    net = net.to(device=device, dtype=dtype)
-
     id_map = default_phoneme_id_map()
     nv = num_symbols(id_map)
     ids, lengths = make_batch(args.batch, args.seq_len, nv, device,
                               seed=1234 + rank)
-
     sample_rate = config.sample_rate
+    dtype = torch.bfloat16 if use_gpu else torch.float32
+
+    engine = None
+    if args.engine == "cpp":
+        try:
+            import tempfile
+
+            from sonata_amd.models import create_random_voice
+            from sonata_amd.ops import hip_ext
+
+            ext = hip_ext(required=False)
+            if ext is not None and hasattr(ext, "VitsEngine"):
+                tmp = tempfile.mkdtemp(prefix=f"bench_voice_r{rank}_")
+                pack = create_random_voice(
+                    tmp, "bench", quality=args.quality, seed=0)
+                engine = ext.VitsEngine(
+                    pack, str(device), "bf16" if use_gpu else "f32")
+        except Exception as e:  # noqa: BLE001 - fall back to python path
+            print(f"# engine=cpp unavailable ({e}); using python path",
+                  flush=True)
+            engine = None
+
+    if engine is None:
+        net = VitsModel(config.num_symbols, arch, n_speakers=1).eval()
+        net = net.to(device=device, dtype=dtype)
+
+    seeds = [1234 * (rank + 1) + i for i in range(args.batch)]
 
     def one_step():
-        with torch.no_grad():
-            audio, audio_lengths = net.infer(ids, lengths)
+        if engine is not None:
+            audio, audio_lengths = engine.infer(
+                ids, lengths, None, 0.667, 1.0, 0.8, seeds)
+        else:
+            with torch.no_grad():
+                audio, audio_lengths = net.infer(ids, lengths)
         return float(audio_lengths.sum().item()) / sample_rate
 
     # ---- warmup --------------------------------------------------------- #
@@ -163,6 +192,7 @@ def main():
                 "global_batch": args.batch * world,
                 "seq_len": args.seq_len,
                 "parallelism": f"dp{world}",
+                "engine": "cpp" if engine is not None else "python",
                 "sample_rate": sample_rate,
                 "per_rank_rtf": round(rtf, 5),
             },
