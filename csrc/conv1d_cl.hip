@@ -464,3 +464,160 @@ torch::Tensor convtranspose1d_cl_fused(torch::Tensor x, torch::Tensor w_perm,
 #undef LAUNCH_TCL
   return out;
 }
+
+// --------------------------------------------------------------------------
+// Direct (zero-LDS, zero-barrier) channel-last conv: both MFMA operands
+// load straight from global memory as 16-B per-lane reads (channel-last
+// makes the K dim contiguous), served by L1/L2 — x tile rows (~8 KB per
+// K-slice) L1-fit and weight chunks are L2-hot.  No staging, no
+// __syncthreads: the WAIT_ANY barrier cost of the staged kernel
+// (56-73% of wave cycles, profiles/r01_conv_pmc_v2.txt) vanishes.
+// Requires Cin % 32 == 0 (full K-slices).
+// --------------------------------------------------------------------------
+__device__ __forceinline__ bf16x8 lrelu8_(bf16x8 v, float slope) {
+  bf16x8 r;
+#pragma unroll
+  for (int q = 0; q < 8; ++q) {
+    float f = __bfloat162float(((__hip_bfloat16*)&v)[q]);
+    ((__hip_bfloat16*)&r)[q] = __float2bfloat16(f > 0.f ? f : f * slope);
+  }
+  return r;
+}
+
+template <int BM, int BN, int WGM, int WGN>
+__global__ __launch_bounds__(512) void conv1d_direct_cl_kernel(
+    const bf16* __restrict__ x,     // [B][Tin][Cin]
+    const bf16* __restrict__ w,     // [ntaps][CoutP][CinP]
+    const float* __restrict__ bias,
+    bf16* __restrict__ out,         // [B][Tout][Cout]
+    const bf16* __restrict__ resid,
+    const int* __restrict__ out_lens,
+    int Cin, int CinP, int Cout, int CoutP, long Tin, long Tout,
+    int ntaps, int dil, int pad, float pre_slope, int act_mode,
+    float post_slope) {
+  const long t0 = (long)blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const int b = blockIdx.z;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid / WGN;
+  const int wc = wid % WGN;
+  constexpr int WM = BM / WGM;
+  constexpr int WN = BN / WGN;
+  constexpr int MT = WM / 16;
+  constexpr int NT = WN / 16;
+  const int kl = lane >> 4;
+  const int il = lane & 15;
+
+  f32x4 acc[MT][NT];
+#pragma unroll
+  for (int i = 0; i < MT; ++i)
+#pragma unroll
+    for (int j = 0; j < NT; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const bf16* xb = x + (long)b * Tin * Cin;
+  // per-lane row bases (constant across the K loop)
+  long trow[MT];
+  bool tok_base[MT];
+#pragma unroll
+  for (int mi = 0; mi < MT; ++mi)
+    trow[mi] = t0 + wr * WM + mi * 16 + il - pad;
+
+  const int wcol = n0 + wc * WN + il;  // this lane's co base (per nj +16)
+
+  for (int c0 = 0; c0 < Cin; c0 += 32) {
+    for (int tap = 0; tap < ntaps; ++tap) {
+      bf16x8 b_frag[NT];
+#pragma unroll
+      for (int nj = 0; nj < NT; ++nj)
+        b_frag[nj] = *(const bf16x8*)&w[
+            ((long)tap * CoutP + wcol + nj * 16) * CinP + c0 + kl * 8];
+      const int toff = tap * dil;
+#pragma unroll
+      for (int mi = 0; mi < MT; ++mi) {
+        const long t = trow[mi] + toff;
+        // clamped load + post-select (no branch around the load)
+        const long tc = t < 0 ? 0 : (t >= Tin ? Tin - 1 : t);
+        bf16x8 a = *(const bf16x8*)&xb[tc * Cin + c0 + kl * 8];
+        if (pre_slope >= 0.f) a = lrelu8_(a, pre_slope);
+        if (t < 0 || t >= Tin) a = bf16x8{};
+#pragma unroll
+        for (int nj = 0; nj < NT; ++nj)
+          acc[mi][nj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, b_frag[nj], acc[mi][nj], 0, 0, 0);
+      }
+    }
+  }
+
+  bf16* ob = out + (long)b * Tout * Cout;
+  const bf16* rb = resid ? resid + (long)b * Tout * Cout : nullptr;
+  const long lim = out_lens ? min((long)out_lens[b], Tout) : Tout;
+#pragma unroll
+  for (int mi = 0; mi < MT; ++mi) {
+#pragma unroll
+    for (int rg = 0; rg < 4; ++rg) {
+      const long t = t0 + wr * WM + mi * 16 + kl * 4 + rg;
+      if (t >= Tout) continue;
+      const bool live = t < lim;
+#pragma unroll
+      for (int nj = 0; nj < NT; ++nj) {
+        const int co = n0 + wc * WN + nj * 16 + il;
+        if (co >= Cout) continue;
+        float v = 0.f;
+        if (live) {
+          v = acc[mi][nj][rg];
+          if (bias) v += bias[co];
+          if (act_mode == ACT_LRELU) v = lrelu_(v, post_slope);
+          else if (act_mode == ACT_TANH) v = tanhf(v);
+          if (rb) v += bf2f(rb[t * Cout + co]);
+        }
+        ob[t * Cout + co] = f2bf(v);
+      }
+    }
+  }
+}
+
+torch::Tensor conv1d_direct_cl(torch::Tensor x, torch::Tensor w_perm,
+                               c10::optional<torch::Tensor> bias, long Cout,
+                               long k, long padding, long dilation,
+                               double pre_lrelu, long act_mode,
+                               double post_slope,
+                               c10::optional<torch::Tensor> residual,
+                               c10::optional<torch::Tensor> out_lens) {
+  TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16);
+  const long B = x.size(0), Tin = x.size(1), Cin = x.size(2);
+  TORCH_CHECK(Cin % 32 == 0, "conv_direct: Cin must be a multiple of 32");
+  const long Tout = Tin + 2 * padding - dilation * (k - 1);
+  const int CoutP = w_perm.size(1), CinP = w_perm.size(2);
+  auto out = torch::empty({B, Tout, Cout}, x.options());
+  if (out.numel() == 0) return out;
+  torch::Tensor bias_f;
+  const float* bias_p = nullptr;
+  if (bias.has_value()) {
+    bias_f = bias->scalar_type() == at::kFloat ? *bias
+                                               : bias->to(at::kFloat).contiguous();
+    bias_p = bias_f.data_ptr<float>();
+  }
+  const bf16* res_p = nullptr;
+  if (residual.has_value()) res_p = (const bf16*)residual->data_ptr();
+  const int* lens_p = nullptr;
+  if (out_lens.has_value()) lens_p = out_lens->data_ptr<int>();
+  hipStream_t st = cur_stream3();
+#define LAUNCH_DIR(BM, BN, WGM, WGN)                                        \
+  hipLaunchKernelGGL((conv1d_direct_cl_kernel<BM, BN, WGM, WGN>),           \
+                     dim3(ceil_div(Tout, BM), ceil_div(Cout, BN), B),       \
+                     dim3(512), 0, st, (const bf16*)x.data_ptr(),           \
+                     (const bf16*)w_perm.data_ptr(), bias_p,                \
+                     (bf16*)out.data_ptr(), res_p, lens_p, (int)Cin, CinP,  \
+                     (int)Cout, CoutP, Tin, Tout, (int)k, (int)dilation,    \
+                     (int)padding, (float)pre_lrelu, (int)act_mode,         \
+                     (float)post_slope)
+  if (Cout >= 128) LAUNCH_DIR(256, 128, 4, 2);
+  else if (Cout >= 64) LAUNCH_DIR(256, 64, 4, 2);
+  else LAUNCH_DIR(256, 32, 4, 2);
+#undef LAUNCH_DIR
+  return out;
+}

@@ -149,3 +149,33 @@ for C, T, k, d in [(256, 8*F, 3, 1), (128, 64*F, 3, 1), (64, 128*F, 3, 1),
     fl = 2*2*B*C*C*k*T
     gb = 2*B*T*(2*C + C)  # x read(+resid reread), out write
     print(f"rbpair C={C:4d} k{k}d{d} T={T:6d} {dt*1e3:8.3f} ms  {fl/dt/1e12:7.1f} TF  {gb/dt/1e9:7.0f} GB/s")
+
+# direct (zero-LDS) conv vs staged cl conv
+from sonata_amd.ops import hip_ext
+from sonata_amd.ops.functional import _conv_weight_mfma, _bias_f32
+ext = hip_ext(required=True)
+print("--- direct vs staged cl conv ---")
+for name, Cin, Cout, T, k, dil in [("res128 k3", 128, 128, 16384, 3, 1),
+                                   ("res128 k11d5", 128, 128, 16384, 11, 5),
+                                   ("res64 k3", 64, 64, 32768, 3, 1),
+                                   ("res32 k3", 32, 32, 65536, 3, 1),
+                                   ("res32 k11d5", 32, 32, 65536, 11, 5),
+                                   ("res256 k3", 256, 256, 2048, 3, 1)]:
+    x = (torch.randn(B, T, Cin)/4).to(torch.bfloat16).to(dev)
+    w = (torch.randn(Cout, Cin, k)/(Cin*k)**0.5).to(torch.bfloat16).to(dev)
+    bias = (torch.randn(Cout)/10).to(dev)
+    pad = (k-1)*dil//2
+    wp = _conv_weight_mfma(w); bf = _bias_f32(bias)
+    ref = leaky_conv1d_cl(x, w, bias, padding=pad, dilation=dil, pre_lrelu=0.1)
+    got = ext.conv1d_direct_cl(x, wp, bf, Cout, k, pad, dil, 0.1, 0, 0.0, None, None)
+    err = (got.float()-ref.float()).abs().max().item()/max(ref.float().abs().max().item(),1e-6)
+    for fn, tag in [(lambda: leaky_conv1d_cl(x, w, bias, padding=pad, dilation=dil, pre_lrelu=0.1), "staged"),
+                    (lambda: ext.conv1d_direct_cl(x, wp, bf, Cout, k, pad, dil, 0.1, 0, 0.0, None, None), "direct")]:
+        for _ in range(3): y = fn()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter(); N = 10
+        for _ in range(N): y = fn()
+        torch.cuda.synchronize()
+        dt = (time.perf_counter()-t0)/N
+        fl = 2*B*Cin*Cout*k*T
+        print(f"{tag} {name:14s} {dt*1e3:8.3f} ms  {fl/dt/1e12:7.1f} TF" + (f"  err={err:.4f}" if tag=="direct" else ""))
