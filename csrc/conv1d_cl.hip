@@ -466,13 +466,14 @@ torch::Tensor convtranspose1d_cl_fused(torch::Tensor x, torch::Tensor w_perm,
 }
 
 // --------------------------------------------------------------------------
-// Direct (zero-LDS, zero-barrier) channel-last conv: both MFMA operands
-// load straight from global memory as 16-B per-lane reads (channel-last
-// makes the K dim contiguous), served by L1/L2 — x tile rows (~8 KB per
-// K-slice) L1-fit and weight chunks are L2-hot.  No staging, no
-// __syncthreads: the WAIT_ANY barrier cost of the staged kernel
-// (56-73% of wave cycles, profiles/r01_conv_pmc_v2.txt) vanishes.
-// Requires Cin % 32 == 0 (full K-slices).
+// Direct (zero-LDS, zero-barrier) channel-last conv — MEASURED NEGATIVE
+// RESULT, kept as documentation: both MFMA operands load straight from
+// global (16-B per-lane reads) so the staged kernel\'s barrier cost
+// vanishes, but per-fragment VMEM issue + in-register LeakyReLU lose
+// 1.6-2.6x vs the staged kernel on every resblock shape
+// (gpurun A/B 2026-09-13: staged 323/689/470 TF vs direct 197/269/225 TF
+// on C=128 k3 / C=128 k11d5 / C=256 k3).  The LDS-staged structure wins
+// on this hardware; not dispatched anywhere.
 // --------------------------------------------------------------------------
 __device__ __forceinline__ bf16x8 lrelu8_(bf16x8 v, float slope) {
   bf16x8 r;

@@ -281,6 +281,17 @@ torch::Tensor VitsEngine::conv(torch::Tensor x, const std::string& mod,
                         pre_lrelu > 0 ? pre_lrelu : -1.0,
                         post_lrelu > 0 ? 1 : 0, post_lrelu, c10::nullopt);
   }
+  if (gpu() && groups > 1) {
+    // grouped/depthwise (DDSConv): the fused host wrapper\'s naive path
+    // beats MIOpen\'s im2col+GEMM at these tiny shapes
+    c10::optional<torch::Tensor> bias32;
+    if (b.has_value()) bias32 = bias_f32(mod + ".bias");
+    auto y = conv1d_fused(x.contiguous(), w.contiguous(), bias32, Cout, k,
+                          stride, pad, dil, groups,
+                          pre_lrelu > 0 ? pre_lrelu : -1.0,
+                          post_lrelu > 0 ? 1 : 0, post_lrelu, c10::nullopt);
+    return y;
+  }
   if (pre_lrelu > 0) x = torch::leaky_relu(x, pre_lrelu);
   auto y = torch::conv1d(x, w, b.has_value() ? *b : torch::Tensor(), stride,
                          pad, dil, groups);
@@ -406,6 +417,66 @@ torch::Tensor VitsEngine::attention(torch::Tensor x, torch::Tensor attn_mask,
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
 VitsEngine::text_encoder(torch::Tensor ids, torch::Tensor lengths) const {
+  if (gpu() && dtype_ == torch::kBFloat16) {
+    // channel-last encoder (mirrors vits.py TextEncoder._forward_cl):
+    // the embedding is already [B,T,H]; attention heads are a free view;
+    // FFN runs on cl MFMA convs with fused ReLU + ragged-row masking.
+    auto lin = [&](torch::Tensor t, const std::string& mod) {
+      return torch::linear(t, p(mod + ".weight").squeeze(-1),
+                           p(mod + ".bias"));
+    };
+    auto x = torch::embedding(p("enc_p.emb.weight"), ids) *
+             std::sqrt((double)cfg_.hidden);  // [B,T,H]
+    auto mask_cl = sequence_mask(lengths, ids.size(1))
+                       .to(x.dtype()).transpose(1, 2);  // [B,T,1]
+    auto attn_mask = mask_cl * mask_cl.transpose(1, 2);  // [B,T,T]
+    auto amask4 = attn_mask.unsqueeze(1);
+    x = x * mask_cl;
+    const long H = cfg_.n_heads, D = cfg_.hidden / cfg_.n_heads;
+    auto lens32 = lengths.to(torch::kInt32).contiguous();
+    for (long i = 0; i < cfg_.n_layers; ++i) {
+      std::string li = std::to_string(i);
+      std::string am = "enc_p.attn_layers." + li;
+      auto xm = x * mask_cl;
+      long B = x.size(0), T = x.size(1), C = x.size(2);
+      auto q = lin(xm, am + ".conv_q").view({B, T, H, D}).transpose(1, 2);
+      auto k = lin(xm, am + ".conv_k").view({B, T, H, D}).transpose(1, 2);
+      auto v = lin(xm, am + ".conv_v").view({B, T, H, D}).transpose(1, 2);
+      double scale = 1.0 / std::sqrt((double)D);
+      auto scores = torch::matmul(q * scale, k.transpose(-2, -1));
+      auto rel_k = rel_embeddings(p(am + ".emb_rel_k"), T, cfg_.window_size);
+      scores = scores + rel_to_abs(torch::matmul(
+                            q * scale, rel_k.unsqueeze(0).transpose(-2, -1)));
+      scores = scores.masked_fill(amask4 == 0, -1e4);
+      auto pr = torch::softmax(scores, -1);
+      auto outt = torch::matmul(pr, v);
+      auto rel_v = rel_embeddings(p(am + ".emb_rel_v"), T, cfg_.window_size);
+      outt = outt + torch::matmul(abs_to_rel(pr), rel_v.unsqueeze(0));
+      auto y = lin(outt.transpose(1, 2).reshape({B, T, C}), am + ".conv_o");
+      x = torch::layer_norm(x + y, {C}, p("enc_p.norm1." + li + ".gamma"),
+                            p("enc_p.norm1." + li + ".beta"), 1e-5);
+      std::string f1 = "enc_p.ffn_layers." + li + ".conv1";
+      std::string f2 = "enc_p.ffn_layers." + li + ".conv2";
+      long pad = cfg_.kernel_size / 2;
+      auto wf1 = p(f1 + ".weight");
+      auto f = conv1d_cl_fused((x * mask_cl).contiguous(), perm_conv(
+                                   f1 + ".weight"), bias_f32(f1 + ".bias"),
+                               wf1.size(0), wf1.size(2), pad, 1, -1.0,
+                               1 /*relu: lrelu slope 0*/, 0.0, c10::nullopt,
+                               lens32);
+      auto wf2 = p(f2 + ".weight");
+      f = conv1d_cl_fused(f, perm_conv(f2 + ".weight"),
+                          bias_f32(f2 + ".bias"), wf2.size(0), wf2.size(2),
+                          pad, 1, -1.0, 0, 0.0, c10::nullopt, lens32);
+      x = torch::layer_norm(x + f, {C}, p("enc_p.norm2." + li + ".gamma"),
+                            p("enc_p.norm2." + li + ".beta"), 1e-5);
+    }
+    auto stats = (lin(x, "enc_p.proj") * mask_cl).transpose(1, 2);
+    auto chunks = stats.chunk(2, 1);
+    auto x_mask = mask_cl.transpose(1, 2);
+    return {(x.transpose(1, 2) * x_mask).contiguous(),
+            chunks[0].contiguous(), chunks[1].contiguous(), x_mask};
+  }
   auto x = torch::embedding(p("enc_p.emb.weight"), ids) *
            std::sqrt((double)cfg_.hidden);
   x = x.transpose(1, 2).contiguous();  // [B, H, T]
