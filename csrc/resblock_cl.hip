@@ -20,7 +20,7 @@
 #define BKP 40
 #define XTROWS 128
 
-template <int BN, int WGN, int TC, int XR>
+template <int BN, int WGN, int TC, int XR, int XTR>
 __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
     const bf16* __restrict__ x,    // [B][T][C]
     const bf16* __restrict__ w1,   // [k][CP][CP] (dilated conv)
@@ -32,8 +32,8 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
     const int* __restrict__ out_lens,
     int C, int CP, long T, int k, int dil, float out_scale) {
   constexpr int WGM = 4;
-  constexpr int WM = XTROWS / WGM;  // 32
-  constexpr int MT = WM / 16;       // 2
+  constexpr int WM = XTR / WGM;
+  constexpr int MT = WM / 16;
   constexpr int WN = BN / WGN;
   constexpr int NT = WN / 16;
   constexpr int XROWS_MAX = XR;     // 128 + (k-1)*dil, bucketed
@@ -41,12 +41,12 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
                                     // 16 distinct banks across il lanes
 
   const int h2 = (k - 1) / 2;        // conv2 halo per side
-  const int BM = XTROWS - (k - 1);   // output rows per block
+  const int BM = XTR - (k - 1);      // output rows per block
   const long t0 = (long)blockIdx.x * BM;
   const int b = blockIdx.z;
 
   __shared__ bf16 Xs[XROWS_MAX][BKP];
-  __shared__ bf16 Xt[XTROWS][XTP];
+  __shared__ bf16 Xt[XTR][XTP];
   __shared__ bf16 Ws[TC][BN][BKP];
 
   const int tid = threadIdx.x;
@@ -61,7 +61,7 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
   const int pad1 = (k - 1) * dil / 2;
   // Xs[0] holds x row (t0 - h2 - pad1); GEMM1 xt row m taps rows m+j*dil
   const long row0 = t0 - h2 - pad1;
-  const int xrows = XTROWS + (k - 1) * dil;
+  const int xrows = XTR + (k - 1) * dil;
   const bool t_interior = (row0 >= 0) && (row0 + xrows <= T);
 
   // ================= GEMM1: xt = lrelu(conv1(lrelu(x))) ================
@@ -256,10 +256,13 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
     TORCH_CHECK(accum->sizes() == x.sizes() && accum->is_contiguous());
     accum_p = (const bf16*)accum->data_ptr();
   }
-  const long BM = XTROWS - (k - 1);
+  // small-C stages run at huge T with tiny per-block work: use taller
+  // 256-row xt tiles there (2x MFMA per block, occupancy still 2-3).
+  const long XTRh = (CP <= 64) ? 256 : 128;
+  const long BM = XTRh - (k - 1);
   hipStream_t st = cur_stream4();
-#define LAUNCH_RB(BN, WGN, TC, XR)                                          \
-  hipLaunchKernelGGL((resblock_pair_cl_kernel<BN, WGN, TC, XR>),            \
+#define LAUNCH_RB(BN, WGN, TC, XR, XTR)                                     \
+  hipLaunchKernelGGL((resblock_pair_cl_kernel<BN, WGN, TC, XR, XTR>),       \
                      dim3(ceil_div(T, BM), 1, B), dim3(512), 0, st,         \
                      (const bf16*)x.data_ptr(),                             \
                      (const bf16*)w1_perm.data_ptr(),                       \
@@ -271,19 +274,26 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
   // XR bucket = 128 + (k-1)*dil rounded up; TC=3 covers k=3 in one
   // chunk and k=7/11 in 3/4 chunks (fewer barrier pairs); C=256 keeps
   // TC=2 for LDS.
-  const int xrows = 128 + (int)((k - 1) * dil);
-#define RB_XR(BN, WGN, TC)                                                  \
+  const int xrows = (int)XTRh + (int)((k - 1) * dil);
+#define RB_XR128(BN, WGN, TC)                                               \
   do {                                                                      \
-    if (xrows <= 144) LAUNCH_RB(BN, WGN, TC, 144);                          \
-    else if (xrows <= 160) LAUNCH_RB(BN, WGN, TC, 160);                     \
-    else LAUNCH_RB(BN, WGN, TC, 180);                                       \
+    if (xrows <= 144) LAUNCH_RB(BN, WGN, TC, 144, 128);                     \
+    else if (xrows <= 160) LAUNCH_RB(BN, WGN, TC, 160, 128);                \
+    else LAUNCH_RB(BN, WGN, TC, 180, 128);                                  \
   } while (0)
-  if (CP == 256) RB_XR(256, 2, 2);
-  else if (CP == 128) RB_XR(128, 2, 3);
-  else if (CP == 64) RB_XR(64, 2, 2);
-  else if (CP == 32) RB_XR(32, 2, 2);
+#define RB_XR256(BN, WGN, TC)                                               \
+  do {                                                                      \
+    if (xrows <= 272) LAUNCH_RB(BN, WGN, TC, 272, 256);                     \
+    else if (xrows <= 288) LAUNCH_RB(BN, WGN, TC, 288, 256);                \
+    else LAUNCH_RB(BN, WGN, TC, 308, 256);                                  \
+  } while (0)
+  if (CP == 256) RB_XR128(256, 2, 2);
+  else if (CP == 128) RB_XR128(128, 2, 3);
+  else if (CP == 64) RB_XR256(64, 2, 2);
+  else if (CP == 32) RB_XR256(32, 2, 2);
   else TORCH_CHECK(false, "resblock_cl: unsupported CP ", CP);
-#undef RB_XR
+#undef RB_XR128
+#undef RB_XR256
 #undef LAUNCH_RB
   return out;
 }
