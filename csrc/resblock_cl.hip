@@ -258,7 +258,7 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
   }
   // small-C stages run at huge T with tiny per-block work: use taller
   // 256-row xt tiles there (2x MFMA per block, occupancy still 2-3).
-  const long XTRh = (CP <= 64) ? 256 : 128;
+  const long XTRh = (CP <= 32) ? 256 : 128;
   const long BM = XTRh - (k - 1);
   hipStream_t st = cur_stream4();
 #define LAUNCH_RB(BN, WGN, TC, XR, XTR)                                     \
@@ -289,7 +289,7 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
   } while (0)
   if (CP == 256) RB_XR128(256, 2, 2);
   else if (CP == 128) RB_XR128(128, 2, 3);
-  else if (CP == 64) RB_XR256(64, 2, 2);
+  else if (CP == 64) RB_XR128(64, 2, 2);
   else if (CP == 32) RB_XR256(32, 2, 2);
   else TORCH_CHECK(false, "resblock_cl: unsupported CP ", CP);
 #undef RB_XR128
