@@ -100,6 +100,27 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::arg("sid") = py::none(), py::arg("noise_scale") = 0.667,
            py::arg("length_scale") = 1.0, py::arg("noise_w") = 0.8,
            py::arg("seeds") = std::vector<int64_t>{})
+      .def("infer_encoder",
+           [](sonata::VitsEngine& e, torch::Tensor ids,
+              torch::Tensor lengths, c10::optional<torch::Tensor> sid,
+              double ns, double ls, double nw,
+              std::vector<int64_t> seeds) {
+             auto r = e.infer_encoder(ids, lengths, sid, ns, ls, nw, seeds);
+             return py::make_tuple(std::get<0>(r), std::get<1>(r),
+                                   std::get<2>(r));
+           },
+           py::arg("ids"), py::arg("lengths"),
+           py::arg("sid") = py::none(), py::arg("noise_scale") = 0.667,
+           py::arg("length_scale") = 1.0, py::arg("noise_w") = 0.8,
+           py::arg("seeds") = std::vector<int64_t>{})
+      .def("decode",
+           [](sonata::VitsEngine& e, torch::Tensor z, torch::Tensor y_mask,
+              c10::optional<torch::Tensor> g,
+              c10::optional<torch::Tensor> lengths) {
+             return e.decode(z, y_mask, g, lengths);
+           },
+           py::arg("z"), py::arg("y_mask"), py::arg("g") = py::none(),
+           py::arg("lengths") = py::none())
       .def("phonemes_to_ids", &sonata::VitsEngine::phonemes_to_ids)
       .def_property_readonly(
           "sample_rate",

@@ -51,11 +51,19 @@ class VitsVoice(SonataModel):
         net: VitsModel,
         device: str = "cpu",
         dtype: torch.dtype = torch.float32,
+        engine=None,
     ):
         self.config = config
         self.net = net.eval().to(device=device, dtype=dtype)
         self.device = torch.device(device)
         self.dtype = dtype
+        # optional C++ VitsEngine runtime (csrc/engine): when attached,
+        # speak_batch/stream_synthesis execute through it (same kernels,
+        # same per-utterance seeds -> identical audio; no Python in the
+        # graph loop).  NOTE: the engine holds its own weight copy — it
+        # attaches at load time, after which parallel.broadcast_module
+        # only affects self.net (ranks load identical packs anyway).
+        self._engine = engine
         self._synth_config = config.default_synthesis_config()
         self._cfg_lock = threading.Lock()
         self._tashkeel = None
@@ -140,6 +148,26 @@ class VitsVoice(SonataModel):
         gens = self._generators(phonemes_batch, cfg.speaker_id)
 
         t0 = time.perf_counter()
+        if self._engine is not None:
+            seeds = [_utterance_seed(p, cfg.speaker_id)
+                     for p in phonemes_batch]
+            sid_t = self._sid_tensor(B, cfg.speaker_id)
+            with stage_timer("infer", self.device):
+                audio, audio_lengths = self._engine.infer(
+                    ids, lengths, sid_t, cfg.noise_scale, cfg.length_scale,
+                    cfg.noise_w, seeds)
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize(self.device)
+            infer_ms = (time.perf_counter() - t0) * 1000.0
+            info = self.audio_output_info()
+            out: List[Audio] = []
+            audio = audio.float().cpu().numpy()
+            audio_lengths = audio_lengths.cpu().numpy()
+            for b in range(B):
+                n = int(audio_lengths[b])
+                out.append(Audio(audio[b, 0, :n], info,
+                                 inference_ms=infer_ms / B))
+            return out
         with stage_timer("infer", self.device):
             audio, audio_lengths = self.net.infer(
             ids,
@@ -188,6 +216,11 @@ class VitsVoice(SonataModel):
                 noise_scale=cfg.noise_scale, length_scale=cfg.length_scale,
                 noise_w=cfg.noise_w, generators=gens,
             )
+        yield from self._stream_decode(z, y_mask, g, chunk_size,
+                                       chunk_padding)
+
+    def _stream_decode(self, z, y_mask, g, chunk_size: int,
+                       chunk_padding: int) -> Iterator[np.ndarray]:
         hop = self.net.arch.hop_length
         num_frames = z.shape[-1]
         # Overlap-crossfade at seams without changing the timeline: each
@@ -198,10 +231,13 @@ class VitsVoice(SonataModel):
         tail: Optional[np.ndarray] = None
         prev_ext = 0
         for spec in chunk_plan(num_frames, chunk_size, chunk_padding):
-            z_c = z[:, :, spec.mel_start : spec.mel_end]
-            m_c = y_mask[:, :, spec.mel_start : spec.mel_end]
+            z_c = z[:, :, spec.mel_start : spec.mel_end].contiguous()
+            m_c = y_mask[:, :, spec.mel_start : spec.mel_end].contiguous()
             with stage_timer("decode_chunk", self.device):
-                audio = self.net.decode(z_c, m_c, g)
+                if self._engine is not None:
+                    audio = self._engine.decode(z_c, m_c, g, None)
+                else:
+                    audio = self.net.decode(z_c, m_c, g)
             wav = audio[0, 0].float().cpu().numpy()
             lo = spec.trim_left_frames * hop
             hi = len(wav) - spec.trim_right_frames * hop
@@ -232,7 +268,8 @@ def _weights_path_for(config_path: str) -> str:
 
 
 def load_voice(
-    config_path: str, device: str = "cpu", dtype: Optional[torch.dtype] = None
+    config_path: str, device: str = "cpu",
+    dtype: Optional[torch.dtype] = None, engine: str = "auto"
 ) -> VitsVoice:
     """Load a voice pack: `<stem>.json` + `<stem>.safetensors`.
 
@@ -258,7 +295,27 @@ def load_voice(
             )
     else:
         raise ModelError(f"voice weights not found: {wpath}")
-    return VitsVoice(config, net, device=device, dtype=dtype)
+    # serving runtime: the C++ VitsEngine by default (same kernels, same
+    # per-utterance seeds -> identical audio); SONATA_ENGINE=python or
+    # engine="python" keeps the torch-module path.
+    eng = None
+    choice = os.environ.get("SONATA_ENGINE", engine)
+    if choice == "auto" and not device.startswith("cuda"):
+        choice = "python"  # CPU stays on the torch oracle path
+    if choice in ("auto", "cpp"):
+        try:
+            from ..ops import hip_ext
+
+            ext = hip_ext(required=False)
+            if ext is not None and hasattr(ext, "VitsEngine"):
+                eng = ext.VitsEngine(
+                    config_path, device,
+                    "bf16" if (dtype == torch.bfloat16) else "f32")
+        except Exception:
+            if choice == "cpp":
+                raise
+            eng = None
+    return VitsVoice(config, net, device=device, dtype=dtype, engine=eng)
 
 
 def create_random_voice(
