@@ -438,3 +438,68 @@ torch::Tensor fused_gate_cl(torch::Tensor x, c10::optional<torch::Tensor> g,
   });
   return out;
 }
+
+// --------------------------------------------------------------------------
+// depthwise_cl: channel-last depthwise Conv1d (DDSConv separable stage).
+// out[b,t,c] = bias[c] + sum_j x[b, t + j*dil - pad, c] * w[c, j]
+// Channel-last rows are contiguous: thread handles 8 channels of one row
+// (b128 loads/stores); taps walk rows.  Memory-bound by design.
+// --------------------------------------------------------------------------
+template <typename T>
+__global__ void depthwise_cl_kernel(const T* __restrict__ x,
+                                    const T* __restrict__ w,  // [C][k]
+                                    const float* __restrict__ bias,
+                                    T* __restrict__ out, long Tlen, int C,
+                                    int k, int dil, int pad, long n8) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n8) return;  // n8 = B*T*C/8
+  const long c8 = (i % (C / 8)) * 8;
+  const long bt = i / (C / 8);
+  const long b = bt / Tlen;
+  const long t = bt % Tlen;
+  const T* xb = x + (b * Tlen) * C;
+  float acc[8];
+#pragma unroll
+  for (int q = 0; q < 8; ++q) acc[q] = bias ? bias[c8 + q] : 0.f;
+  for (int j = 0; j < k; ++j) {
+    const long tt = t + (long)j * dil - pad;
+    if (tt < 0 || tt >= Tlen) continue;
+    const T* xr = xb + tt * C + c8;
+#pragma unroll
+    for (int q = 0; q < 8; ++q)
+      acc[q] += ld_f(xr + q) * ld_f(w + (c8 + q) * k + j);
+  }
+  T* or_ = out + (b * Tlen + t) * C + c8;
+#pragma unroll
+  for (int q = 0; q < 8; ++q) st_f(or_ + q, acc[q]);
+}
+
+torch::Tensor depthwise_cl(torch::Tensor x, torch::Tensor w,
+                           c10::optional<torch::Tensor> bias, long dil,
+                           long pad) {
+  // x: [B, T, C] channel-last; w: [C, 1, k] (nn.Conv1d groups=C weight)
+  TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
+  const long B = x.size(0), Tlen = x.size(1), C = x.size(2);
+  TORCH_CHECK(C % 8 == 0, "depthwise_cl: C % 8 != 0");
+  auto w2 = w.reshape({C, w.size(-1)}).contiguous();
+  const long k = w2.size(1);
+  auto out = torch::empty_like(x);
+  torch::Tensor bias_f;
+  const float* bias_p = nullptr;
+  if (bias.has_value()) {
+    bias_f = bias->scalar_type() == at::kFloat
+                 ? *bias : bias->to(at::kFloat).contiguous();
+    bias_p = bias_f.data_ptr<float>();
+  }
+  const long n8 = B * Tlen * (C / 8);
+  const int threads = 256;
+  DISPATCH_FT(x, "depthwise_cl", {
+    hipLaunchKernelGGL(depthwise_cl_kernel<scalar_t>,
+                       dim3((n8 + threads - 1) / threads), dim3(threads), 0,
+                       cur_stream(), (const scalar_t*)x.data_ptr(),
+                       (const scalar_t*)w2.data_ptr(), bias_p,
+                       (scalar_t*)out.data_ptr(), Tlen, (int)C, (int)k,
+                       (int)dil, (int)pad, n8);
+  });
+  return out;
+}

@@ -53,6 +53,9 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
                                      double out_scale);
 torch::Tensor fused_gate_cl(torch::Tensor x, c10::optional<torch::Tensor> g,
                             long n_channels);
+torch::Tensor depthwise_cl(torch::Tensor x, torch::Tensor w,
+                           c10::optional<torch::Tensor> bias, long dil,
+                           long pad);
 
 namespace sonata {
 
@@ -736,6 +739,83 @@ torch::Tensor VitsEngine::sdp_infer(torch::Tensor x, torch::Tensor mask,
                                     c10::optional<torch::Tensor> g,
                                     double noise_w,
                                     torch::Tensor noise) const {
+  if (gpu() && x.scalar_type() == torch::kBFloat16) {
+    // channel-last SDP (mirrors vits.py _infer_cl): DDS/1x1 stages on
+    // [B,T,C] rows, 2-channel flow state + spline stay channel-first
+    auto lin = [&](torch::Tensor t, const std::string& mod) {
+      return torch::linear(t, p(mod + ".weight").squeeze(-1),
+                           p(mod + ".bias"));
+    };
+    auto dds_cl = [&](torch::Tensor t, torch::Tensor mc,
+                      c10::optional<torch::Tensor> gc,
+                      const std::string& mod) {
+      if (gc.has_value()) t = t + *gc;
+      const long C = t.size(-1);
+      for (long i = 0; i < 3; ++i) {
+        std::string li = std::to_string(i);
+        long dilation = 1;
+        for (long j = 0; j < i; ++j) dilation *= 3;
+        long pad = (3 - 1) * dilation / 2;
+        auto y = depthwise_cl((t * mc).contiguous(),
+                              p(mod + ".convs_sep." + li + ".weight"),
+                              bias_f32(mod + ".convs_sep." + li + ".bias"),
+                              dilation, pad);
+        y = torch::layer_norm(y, {C}, p(mod + ".norms_1." + li + ".gamma"),
+                              p(mod + ".norms_1." + li + ".beta"), 1e-5);
+        y = torch::gelu(y);
+        y = lin(y, mod + ".convs_1x1." + li);
+        y = torch::layer_norm(y, {C}, p(mod + ".norms_2." + li + ".gamma"),
+                              p(mod + ".norms_2." + li + ".beta"), 1e-5);
+        y = torch::gelu(y);
+        t = t + y;
+      }
+      return t * mc;
+    };
+    auto mc = mask.transpose(1, 2).contiguous();  // [B,T,1]
+    auto h = lin(x.detach().transpose(1, 2).contiguous(), "dp.pre");
+    if (g.has_value() && has("dp.cond.weight"))
+      h = h + lin(g->detach().squeeze(-1), "dp.cond").unsqueeze(1);
+    h = dds_cl(h, mc, c10::nullopt, "dp.convs");
+    h = lin(h, "dp.proj") * mc;
+
+    auto z = noise * noise_w * mask;
+    const long half = 1;
+    auto conv_flow_cl = [&](torch::Tensor z, long idx) {
+      std::string mod = "dp.flows." + std::to_string(idx);
+      long B = z.size(0), T = z.size(2);
+      auto z0 = z.narrow(1, 0, half);
+      auto z1 = z.narrow(1, half, half);
+      auto hz = lin(z0.transpose(1, 2).contiguous(), mod + ".pre");
+      hz = dds_cl(hz, mc, h, mod + ".convs");
+      hz = lin(hz, mod + ".proj") * mc;  // [B,T,half*(3b-1)]
+      long num_bins = 10;
+      auto h4 = hz.view({B, T, half, 3 * num_bins - 1})
+                    .permute({0, 2, 1, 3});
+      double scale = std::sqrt((double)192);
+      auto uw = h4.index({torch::indexing::Ellipsis,
+                          torch::indexing::Slice(0, num_bins)}) / scale;
+      auto uh = h4.index({torch::indexing::Ellipsis,
+                          torch::indexing::Slice(num_bins, 2 * num_bins)}) /
+                scale;
+      auto ud = h4.index({torch::indexing::Ellipsis,
+                          torch::indexing::Slice(2 * num_bins,
+                                                 torch::indexing::None)});
+      auto res = rq_spline(z1, uw, uh, ud, true, 5.0);
+      return torch::cat({z0, res.first}, 1) * mask;
+    };
+    auto flip = [&](torch::Tensor t) { return torch::flip(t, {1}); };
+    z = flip(z);
+    z = conv_flow_cl(z, 7);
+    z = flip(z);
+    z = conv_flow_cl(z, 5);
+    z = flip(z);
+    z = conv_flow_cl(z, 3);
+    z = flip(z);
+    auto m = p("dp.flows.0.m");
+    auto logs = p("dp.flows.0.logs");
+    z = (z - m) * torch::exp(-logs) * mask;
+    return z.narrow(1, 0, 1);
+  }
   x = conv(x.detach(), "dp.pre");
   if (g.has_value() && has("dp.cond.weight"))
     x = x + conv(g->detach(), "dp.cond");

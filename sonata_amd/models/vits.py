@@ -22,6 +22,7 @@ from torch import nn
 
 from ..ops import (
     conv_mod,
+    depthwise_conv1d_cl,
     expand_states,
     fused_gate,
     fused_gate_cl,
@@ -548,6 +549,25 @@ class DDSConv(nn.Module):
             x = x + y
         return x * x_mask
 
+    def forward_cl(self, x: torch.Tensor, x_mask: torch.Tensor,
+                   g: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Channel-last DDS stack: x [B,T,C], x_mask [B,T,1]; depthwise
+        via depthwise_cl kernel, 1x1s as linears, LN over rows."""
+        if g is not None:
+            x = x + g
+        C = x.shape[-1]
+        for sep, one, n1, n2 in zip(self.convs_sep, self.convs_1x1,
+                                    self.norms_1, self.norms_2):
+            y = depthwise_conv1d_cl(x * x_mask, sep.weight, sep.bias,
+                                    sep.dilation[0], sep.padding[0])
+            y = F.layer_norm(y, (C,), n1.gamma, n1.beta, n1.eps)
+            y = F.gelu(y)
+            y = F.linear(y, one.weight.squeeze(-1), one.bias)
+            y = F.layer_norm(y, (C,), n2.gamma, n2.beta, n2.eps)
+            y = F.gelu(y)
+            x = x + y
+        return x * x_mask
+
 
 class ElementwiseAffine(nn.Module):
     def __init__(self, channels: int):
@@ -598,8 +618,30 @@ class ConvFlow(nn.Module):
             return x, logdet
         return x
 
+    def reverse_cl(self, x: torch.Tensor, x_mask_cf: torch.Tensor,
+                   x_mask_cl: torch.Tensor,
+                   g_cl: torch.Tensor) -> torch.Tensor:
+        """Inverse spline coupling with the DDS conditioner running
+        channel-last; z itself stays [B,2,T] (tiny)."""
+        B, _, T = x.shape
+        x0, x1 = x[:, : self.half], x[:, self.half :]
+        h = F.linear(x0.transpose(1, 2), self.pre.weight.squeeze(-1),
+                     self.pre.bias)  # [B,T,F]
+        h = self.convs.forward_cl(h, x_mask_cl, g=g_cl)
+        h = F.linear(h, self.proj.weight.squeeze(-1),
+                     self.proj.bias) * x_mask_cl  # [B,T,half*(3b-1)]
+        h = h.view(B, T, self.half, 3 * self.num_bins - 1).permute(0, 2, 1, 3)
+        scale = math.sqrt(self.filter_channels)
+        uw = h[..., : self.num_bins] / scale
+        uh = h[..., self.num_bins : 2 * self.num_bins] / scale
+        ud = h[..., 2 * self.num_bins :]
+        x1, _ = rational_quadratic_spline(
+            x1, uw, uh, ud, inverse=True, tail_bound=self.tail_bound
+        )
+        return torch.cat([x0, x1], dim=1) * x_mask_cf
 
-class Flip(nn.Module):
+
+class Flip(nn.Module):  # noqa: E302
     def forward(self, x, *args, reverse=False, **kwargs):
         x = torch.flip(x, [1])
         if not reverse:
@@ -630,6 +672,10 @@ class StochasticDurationPredictor(nn.Module):
         """Reverse pass: sample log-durations. x: [B, H, T] text states.
         `noise` is [B, 2, T] standard-normal (masked per utterance so that
         batch composition cannot change an utterance's durations)."""
+        from ..ops import use_hip
+
+        if use_hip(x):
+            return self._infer_cl(x, x_mask, g, noise_scale, noise)
         x = conv_mod(self.pre, x.detach())
         if g is not None and self.cond is not None:
             x = x + conv_mod(self.cond, g.detach())
@@ -645,6 +691,36 @@ class StochasticDurationPredictor(nn.Module):
             z = flow(z, x_mask, g=x, reverse=True)
         z0, _ = z.chunk(2, dim=1)
         return z0  # logw [B, 1, T]
+
+    def _infer_cl(self, x: torch.Tensor, x_mask: torch.Tensor,
+                  g: Optional[torch.Tensor], noise_scale: float,
+                  noise: Optional[torch.Tensor]) -> torch.Tensor:
+        """GPU path: DDS/1x1 stages channel-last; the 2-channel flow state
+        and the spline stay channel-first (tiny tensors)."""
+        mc = x_mask.transpose(1, 2)  # [B,T,1]
+        h = F.linear(x.detach().transpose(1, 2),
+                     self.pre.weight.squeeze(-1), self.pre.bias)
+        if g is not None and self.cond is not None:
+            h = h + F.linear(g.detach().squeeze(-1),
+                             self.cond.weight.squeeze(-1),
+                             self.cond.bias).unsqueeze(1)
+        h = self.convs.forward_cl(h, mc)
+        h = F.linear(h, self.proj.weight.squeeze(-1), self.proj.bias) * mc
+        if noise is None:
+            noise = torch.randn((x.shape[0], 2, x.shape[2]),
+                                device=x.device, dtype=x.dtype)
+        z = noise * noise_scale * x_mask
+        flows = list(reversed(self.flows))
+        flows = flows[:-2] + [flows[-1]]
+        for flow in flows:
+            if isinstance(flow, ConvFlow):
+                z = flow.reverse_cl(z, x_mask, mc, h)
+            elif isinstance(flow, Flip):
+                z = torch.flip(z, [1])
+            else:  # ElementwiseAffine
+                z = flow(z, x_mask, reverse=True)
+        z0, _ = z.chunk(2, dim=1)
+        return z0
 
 
 # --------------------------------------------------------------------------- #

@@ -68,4 +68,5 @@ from .functional import (  # noqa: F401,E402
     leaky_convtranspose1d_cl,
     resblock_pair_cl,
     fused_gate_cl,
+    depthwise_conv1d_cl,
 )

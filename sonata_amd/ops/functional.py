@@ -422,3 +422,20 @@ def fused_gate_cl(
         x = x + g
     a, b = x[..., :n_channels], x[..., n_channels:]
     return torch.tanh(a) * torch.sigmoid(b)
+
+
+def depthwise_conv1d_cl(
+    x: torch.Tensor,  # [B, T, C]
+    weight: torch.Tensor,  # [C, 1, k]
+    bias: Optional[torch.Tensor],
+    dilation: int,
+    padding: int,
+) -> torch.Tensor:
+    """Channel-last depthwise Conv1d (DDSConv separable stage)."""
+    if use_hip(x):
+        ext = hip_ext(required=True)
+        return ext.depthwise_cl(x.contiguous(), weight, bias, dilation,
+                                padding)
+    y = F.conv1d(x.transpose(1, 2), weight, bias, padding=padding,
+                 dilation=dilation, groups=x.shape[-1])
+    return y.transpose(1, 2)
