@@ -42,7 +42,18 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
 
   const int h2 = (k - 1) / 2;        // conv2 halo per side
   const int BM = XTR - (k - 1);      // output rows per block
-  const long t0 = (long)blockIdx.x * BM;
+  // XCD-aware swizzle: consecutive T-tiles land on the SAME XCD so halo
+  // rows and the weight tensor stay hot in that XCD\'s L2 (the hardware
+  // round-robins blockIdx across the 8 XCDs).  Bijective remap per the
+  // CDNA4 guide (q/r split handles nwg % 8 != 0).
+  const int nwg = gridDim.x;
+  int tile = blockIdx.x;
+  if (nwg > 8) {
+    const int xcd = tile % 8, orig8 = tile / 8;
+    const int q = nwg / 8, r = nwg % 8;
+    tile = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig8;
+  }
+  const long t0 = (long)tile * BM;
   const int b = blockIdx.z;
 
   __shared__ bf16 Xs[XROWS_MAX][BKP];
