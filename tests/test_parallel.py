@@ -91,3 +91,13 @@ def test_dp_world2(tmp_path):
     assert not errs, errs
     # broadcast made weights identical
     assert abs(results[0][2] - results[1][2]) < 1e-6
+
+
+def test_shard_round_robin_uneven():
+    from sonata_amd.parallel import shard_round_robin
+
+    # partition property for any world size / count
+    for n, world in [(5, 3), (1, 8), (0, 2), (17, 4)]:
+        all_idx = sorted(sum((shard_round_robin(n, r, world)
+                              for r in range(world)), []))
+        assert all_idx == list(range(n))
