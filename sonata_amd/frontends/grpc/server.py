@@ -28,6 +28,7 @@ from typing import Dict, Optional
 import grpc
 
 from ...core import SonataError
+from ...synth.batcher import DynamicBatcher
 from ...synth.synthesizer import AudioOutputConfig, SonataSpeechSynthesizer
 from .proto import (MESSAGES, MODE_LAZY, QUALITY_VALUES, RPCS, SERVICE_NAME)
 
@@ -52,6 +53,9 @@ class _Voice:
     def __init__(self, voice_id: str, synth: SonataSpeechSynthesizer):
         self.voice_id = voice_id
         self.synth = synth
+        # dynamic batcher: concurrent RPCs coalesce into padded GPU
+        # batches (per-utterance seeding makes batching invisible)
+        self.batcher = DynamicBatcher(synth.model)
 
 
 class SonataGrpcService:
@@ -184,10 +188,22 @@ class SonataGrpcService:
         out_cfg = self._speech_args_to_config(
             request.speech_args if request.HasField("speech_args") else None)
         mode = request.synthesis_mode
-        it = (v.synth.synthesize_lazy(request.text, out_cfg)
-              if mode == MODE_LAZY
-              else v.synth.synthesize_parallel(request.text, out_cfg))
-        for audio in it:
+        if mode == MODE_LAZY:
+            it = v.synth.synthesize_lazy(request.text, out_cfg)
+            for audio in it:
+                yield MESSAGES["SynthesisResult"](
+                    wav_samples=audio.as_wave_bytes(),
+                    rtf=float(audio.real_time_factor),
+                )
+            return
+        # default (parallel/batched): sentences go through the dynamic
+        # batcher so CONCURRENT RPCs share padded GPU batches
+        sentences = list(v.synth.model.phonemize_text(request.text))
+        futures = [v.batcher.submit(sent) for sent in sentences]
+        for f in futures:
+            audio = f.result()
+            if out_cfg is not None:
+                audio = v.synth._post(audio, out_cfg)
             yield MESSAGES["SynthesisResult"](
                 wav_samples=audio.as_wave_bytes(),
                 rtf=float(audio.real_time_factor),

@@ -1,3 +1,4 @@
+from .batcher import DynamicBatcher  # noqa: F401
 from .synthesizer import (  # noqa: F401
     AudioOutputConfig,
     SonataSpeechSynthesizer,

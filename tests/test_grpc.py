@@ -85,3 +85,22 @@ def test_synthesize_realtime_stream(running):
         voice_id=vid, text="hˈɛloʊ ðˈɛr ˈɛvɹiwˌʌn.")))
     assert len(chunks) >= 1
     assert all(len(c.wav_samples) > 0 for c in chunks)
+
+
+def test_concurrent_rpcs_batch_together(running):
+    """Concurrent SynthesizeUtterance RPCs coalesce in the dynamic
+    batcher and all complete correctly."""
+    from concurrent.futures import ThreadPoolExecutor
+
+    client, pack = running
+    vid = client.LoadVoice(MESSAGES["VoicePath"](config_path=pack)).voice_id
+
+    def one(i):
+        res = list(client.SynthesizeUtterance(MESSAGES["Utterance"](
+            voice_id=vid, text=f"nˈʌmbɚ {'wˈʌn tˈuː '*(1+i%3)}.")))
+        assert len(res) == 1 and len(res[0].wav_samples) > 400
+        return len(res[0].wav_samples)
+
+    with ThreadPoolExecutor(max_workers=16) as ex:
+        sizes = list(ex.map(one, range(32)))
+    assert len(sizes) == 32
