@@ -9,7 +9,7 @@
 // Usage:
 //   sonata_infer <voice.json> [-o out.wav] [-d cuda:0|cpu] [-f phonemes.txt]
 //                [--length-scale F] [--noise-scale F] [--noise-w F]
-//                [--speaker N] [--bench N]
+//                [--speaker N] [--bench N] [--stream [chunk] [pad]]
 #include <chrono>
 #include <cstring>
 #include <fstream>
@@ -71,7 +71,7 @@ int main(int argc, char** argv) {
   }
   std::string config = argv[1], out = "out.wav", device_s, input_file;
   double ls = 0, ns = 0, nw = 0;
-  long speaker = -1, bench = 0;
+  long speaker = -1, bench = 0, stream_chunk = 0, stream_pad = 3;
   for (int i = 2; i < argc; ++i) {
     std::string a = argv[i];
     auto next = [&]() { return std::string(argv[++i]); };
@@ -83,6 +83,9 @@ int main(int argc, char** argv) {
     else if (a == "--noise-w") nw = std::stod(next());
     else if (a == "--speaker") speaker = std::stol(next());
     else if (a == "--bench") bench = std::stol(next());
+    else if (a == "--stream") stream_chunk = 45;
+    else if (a == "--stream-chunk") stream_chunk = std::stol(next());
+    else if (a == "--stream-pad") stream_pad = std::stol(next());
   }
   if (device_s.empty())
     device_s = torch::cuda::is_available() ? "cuda:0" : "cpu";
@@ -138,6 +141,65 @@ int main(int argc, char** argv) {
   c10::optional<torch::Tensor> sid;
   if (speaker >= 0)
     sid = torch::full({B}, speaker, torch::kLong);
+
+  if (stream_chunk > 0) {
+    // Native streaming: encoder once, HiFi-GAN decoded in adaptive
+    // chunks (ports models/chunker.py: growth x step, MIN 44 / MAX 1024
+    // frames, overlap-discard +-pad frames; reference AdaptiveMelChunker
+    // semantics, piper/src/lib.rs:860-913).  One utterance at a time.
+    std::vector<float> samples;
+    double first_ms = -1;
+    auto s0 = std::chrono::steady_clock::now();
+    for (long b = 0; b < B; ++b) {
+      auto idsb = ids.narrow(0, b, 1);
+      auto lensb = lengths.narrow(0, b, 1);
+      std::vector<int64_t> sb{seeds[(size_t)b]};
+      auto [z, y_mask, gv] = engine.infer_encoder(idsb, lensb, sid, ns, ls,
+                                                  nw, sb);
+      c10::optional<torch::Tensor> gopt;
+      if (gv.defined() && gv.numel()) gopt = gv;
+      const long F = z.size(2), hop = engine.config().hop();
+      const long MINC = 44, MAXC = 1024;
+      long start = 0, step = 1;
+      bool oneshot = F <= stream_chunk * 2 + stream_pad * 2;
+      while (start < F) {
+        long lo = 0, hi = F, pl = 0, pr = 0;
+        bool last = true;
+        if (!oneshot) {
+          long size = std::min(stream_chunk * step, MAXC);
+          ++step;
+          long end = std::min(start + size, F);
+          if (F - end < MINC) end = F;
+          pl = std::min(stream_pad, start);
+          pr = std::min(stream_pad, F - end);
+          last = end >= F;
+          lo = start - pl;
+          hi = end + pr;
+          start = end;
+        } else {
+          start = F;
+        }
+        auto zc = z.narrow(2, lo, hi - lo).contiguous();
+        auto mc = y_mask.narrow(2, lo, hi - lo).contiguous();
+        auto a = engine.decode(zc, mc, gopt, c10::nullopt)
+                     .to(torch::kFloat32).to(torch::kCPU).contiguous();
+        const float* ptr = a[0][0].data_ptr<float>();
+        long n = a.size(2);
+        samples.insert(samples.end(), ptr + pl * hop,
+                       ptr + (n - pr * hop));
+        if (first_ms < 0)
+          first_ms = std::chrono::duration<double, std::milli>(
+                         std::chrono::steady_clock::now() - s0).count();
+        if (last) break;
+      }
+    }
+    write_wav(out, samples.data(), samples.size(), cfg.sample_rate);
+    double dur_ms = 1000.0 * samples.size() / cfg.sample_rate;
+    std::cerr << "streamed " << out << ": " << samples.size()
+              << " samples (" << dur_ms << " ms audio), first chunk in "
+              << first_ms << " ms\n";
+    return 0;
+  }
 
   auto t0 = std::chrono::steady_clock::now();
   auto [audio, audio_lengths] = engine.infer(ids, lengths, sid, ns, ls, nw,

@@ -125,3 +125,17 @@ def test_engine_backed_voice_matches_python_voice(pack):
     # streaming path through the engine decode
     chunks = list(v_cpp.stream_synthesis(phon, 20, 2))
     assert sum(len(c) for c in chunks) == len(b)
+
+
+@pytest.mark.skipif(not os.path.exists(BIN), reason="sonata_infer not built")
+def test_cli_binary_streaming(pack, tmp_path):
+    inp = tmp_path / "phon.txt"
+    inp.write_text("hˈɛloʊ wˈɜːld tˈɛst sˈɛntəns lˈɔŋɡɚ ˈɛvɹiwˌʌn tʊdˈeɪ.\n")
+    out = tmp_path / "s.wav"
+    r = subprocess.run(
+        [BIN, pack, "-d", "cpu", "-f", str(inp), "-o", str(out),
+         "--stream", "--stream-chunk", "20"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr
+    assert "first chunk in" in r.stderr
+    assert out.read_bytes()[:4] == b"RIFF"
