@@ -246,11 +246,18 @@ def create_server(
     port: Optional[int] = None,
     device: Optional[str] = None,
     max_workers: int = 16,
+    reuse_port: bool = False,
 ):
     """Build a grpc.Server bound to 127.0.0.1 (reference binds loopback,
-    main.rs:437-445).  Returns (server, bound_port, service)."""
+    main.rs:437-445).  Returns (server, bound_port, service).
+
+    `reuse_port=True` sets SO_REUSEPORT so several server PROCESSES can
+    share one port (the kernel load-balances connections): the escape
+    from the single-process GIL ceiling on concurrent serving."""
     service = SonataGrpcService(device=device)
-    server = grpc.server(futures.ThreadPoolExecutor(max_workers=max_workers))
+    opts = [("grpc.so_reuseport", 1 if reuse_port else 0)]
+    server = grpc.server(futures.ThreadPoolExecutor(max_workers=max_workers),
+                         options=opts)
     server.add_generic_rpc_handlers((_generic_handler(service),))
     if port is None:
         port = int(os.environ.get("SONATA_GRPC_SERVER_PORT", DEFAULT_PORT))
@@ -258,14 +265,51 @@ def create_server(
     return server, bound, service
 
 
-def serve(port: Optional[int] = None, device: Optional[str] = None) -> None:
+def serve(port: Optional[int] = None, device: Optional[str] = None,
+          processes: int = 1) -> None:
     logging.basicConfig(
         level=os.environ.get("SONATA_GRPC", "INFO").upper())
+    if processes > 1:
+        # N identical server processes share the port via SO_REUSEPORT;
+        # each holds its own voice copies (cheap next to 288 GB HBM) and
+        # its own GIL, so handler throughput scales ~linearly.
+        import multiprocessing as mp
+
+        if port is None:
+            port = int(os.environ.get("SONATA_GRPC_SERVER_PORT",
+                                      DEFAULT_PORT))
+        ctx = mp.get_context("spawn")
+        procs = [ctx.Process(target=_serve_one, args=(port, device))
+                 for _ in range(processes - 1)]
+        for p in procs:
+            p.start()
+        try:
+            _serve_one(port, device)
+        finally:
+            for p in procs:
+                p.terminate()
+        return
     server, bound, _ = create_server(port=port, device=device)
     server.start()
     log.info("sonata_grpc serving on 127.0.0.1:%d", bound)
     server.wait_for_termination()
 
 
+def _serve_one(port: int, device: Optional[str]) -> None:
+    server, bound, _ = create_server(port=port, device=device,
+                                     reuse_port=True)
+    server.start()
+    log.info("sonata_grpc worker serving on 127.0.0.1:%d (pid shared port)",
+             bound)
+    server.wait_for_termination()
+
+
 if __name__ == "__main__":
-    serve()
+    import argparse
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--port", type=int, default=None)
+    ap.add_argument("--device", default=None)
+    ap.add_argument("--processes", type=int, default=1)
+    a = ap.parse_args()
+    serve(port=a.port, device=a.device, processes=a.processes)
