@@ -11,6 +11,7 @@
 //                [--length-scale F] [--noise-scale F] [--noise-w F]
 //                [--speaker N] [--bench N] [--stream [chunk] [pad]]
 #include <chrono>
+#include <cstdlib>
 #include <cstring>
 #include <fstream>
 #include <iostream>
@@ -59,6 +60,16 @@ uint64_t fnv1a(const std::string& s) {
   uint64_t h = 1469598103934665603ull;
   for (unsigned char c : s) { h ^= c; h *= 1099511628211ull; }
   return h & 0x7fffffffffffffffull;
+}
+
+// The HIP runtime\'s static destructors segfault at process exit when
+// torch_hip is loaded (observed on ROCm 7.2: __hip_module_dtor inside
+// __cxa_finalize).  All useful work is done by then; exit without
+// running static dtors.
+[[noreturn]] void clean_exit(int code) {
+  std::cout.flush();
+  std::cerr.flush();
+  std::_Exit(code);
 }
 
 }  // namespace
@@ -198,7 +209,7 @@ int main(int argc, char** argv) {
     std::cerr << "streamed " << out << ": " << samples.size()
               << " samples (" << dur_ms << " ms audio), first chunk in "
               << first_ms << " ms\n";
-    return 0;
+    clean_exit(0);
   }
 
   auto t0 = std::chrono::steady_clock::now();
@@ -226,7 +237,7 @@ int main(int argc, char** argv) {
     std::cout << "{\"bench_steps\": " << bench << ", \"audio_sec_per_s\": "
               << audio_sec / el << ", \"ms_per_step\": "
               << el * 1000.0 / bench << "}\n";
-    return 0;
+    clean_exit(0);
   }
 
   auto af = audio.to(torch::kFloat32).to(torch::kCPU).contiguous();
@@ -243,5 +254,5 @@ int main(int argc, char** argv) {
   std::cerr << "wrote " << out << ": " << samples.size() << " samples ("
             << dur_ms << " ms audio), infer " << infer_ms << " ms, rtf "
             << infer_ms / dur_ms << "\n";
-  return 0;
+  clean_exit(0);
 }
