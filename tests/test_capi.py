@@ -157,3 +157,32 @@ def test_speak_to_file(lib, voice, tmp_path):
                                   params, out, C.byref(err))
     assert ok == 1 and err.code == 0
     assert open(out, "rb").read(4) == b"RIFF"
+
+
+def test_speak_nonblocking(lib, voice):
+    """Nonblocking mode: libsonataSpeak returns immediately; events arrive
+    on a worker thread (reference capi lib.rs:366-386)."""
+    import threading
+    import time
+
+    done = threading.Event()
+    chunks = []
+
+    @CALLBACK
+    def cb(ev):
+        if ev.event_type == 0:
+            chunks.append(int(ev.len))
+        elif ev.event_type == 1:
+            done.set()
+        return 0
+
+    err = ExternError()
+    params = SynthesisParams(mode=1, callback=cb, nonblocking=1)
+    t0 = time.perf_counter()
+    lib.libsonataSpeak(voice, "hˈɛloʊ wˈɜːld ˈɛvɹiwˌʌn.".encode(), params,
+                       C.byref(err))
+    returned_in = time.perf_counter() - t0
+    assert err.code == 0
+    assert done.wait(timeout=120), "FINISHED event never arrived"
+    assert sum(chunks) > 500
+    assert returned_in < 5.0  # returned before synthesis completed
