@@ -104,3 +104,55 @@ def test_concurrent_rpcs_batch_together(running):
     with ThreadPoolExecutor(max_workers=16) as ex:
         sizes = list(ex.map(one, range(32)))
     assert len(sizes) == 32
+
+
+def _mp_worker(port, device):
+    from sonata_amd.frontends.grpc.server import _serve_one
+
+    _serve_one(port, device)
+
+
+def test_multiprocess_reuseport_serving(voice_pack):
+    """SO_REUSEPORT: two server processes share one port; requests land on
+    both and all succeed."""
+    import multiprocessing as mp
+    import time
+
+    from concurrent.futures import ThreadPoolExecutor
+
+    port = 49917
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_mp_worker, args=(port, "cpu"), daemon=True)
+             for _ in range(2)]
+    for p in procs:
+        p.start()
+    try:
+        # wait for bind (children import torch: allow a while)
+        deadline = time.time() + 120
+        client = None
+        while time.time() < deadline:
+            try:
+                c = SonataGrpcClient(f"127.0.0.1:{port}")
+                c.LoadVoice(MESSAGES["VoicePath"](config_path=voice_pack))
+                client = c
+                break
+            except grpc.RpcError:
+                time.sleep(2)
+        assert client is not None, "no worker bound the port"
+        # many fresh connections spread across workers; load + synthesize
+        def one(i):
+            c = SonataGrpcClient(f"127.0.0.1:{port}")
+            vid = c.LoadVoice(
+                MESSAGES["VoicePath"](config_path=voice_pack)).voice_id
+            res = list(c.SynthesizeUtterance(MESSAGES["Utterance"](
+                voice_id=vid, text="wˈʌn tˈuː.")))
+            c.close()
+            return len(res[0].wav_samples)
+
+        with ThreadPoolExecutor(max_workers=8) as ex:
+            sizes = list(ex.map(one, range(16)))
+        assert all(s > 300 for s in sizes)
+        client.close()
+    finally:
+        for p in procs:
+            p.terminate()
