@@ -337,6 +337,46 @@ _PT_RULES = {
 }
 
 
+# --------------------------------------------------------------------------- #
+# Russian (Cyrillic)
+# --------------------------------------------------------------------------- #
+_RU_RULES = {
+    "щ": "ɕː", "ч": "tɕ", "ш": "ʂ", "ж": "ʐ", "ц": "ts",
+    "а": "a", "б": "b", "в": "v", "г": "ɡ", "д": "d", "е": "je",
+    "ё": "jo", "з": "z", "и": "i", "й": "j", "к": "k", "л": "l",
+    "м": "m", "н": "n", "о": "o", "п": "p", "р": "r", "с": "s",
+    "т": "t", "у": "u", "ф": "f", "х": "x", "ъ": "", "ы": "ɨ",
+    "ь": "ʲ", "э": "ɛ", "ю": "ju", "я": "ja",
+}
+
+# --------------------------------------------------------------------------- #
+# Dutch
+# --------------------------------------------------------------------------- #
+_NL_RULES = {
+    "sch": "sx", "ij": "ɛi", "ei": "ɛi", "ui": "œy", "ou": "ʌu",
+    "au": "ʌu", "oe": "u", "eu": "ø", "ie": "i", "aa": "aː",
+    "ee": "eː", "oo": "oː", "uu": "y", "ch": "x", "ng": "ŋ",
+    "a": "ɑ", "b": "b", "c": "k", "d": "d", "e": "ɛ", "f": "f",
+    "g": "ɣ", "h": "ɦ", "i": "ɪ", "j": "j", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "ɔ", "p": "p", "r": "r", "s": "s",
+    "t": "t", "u": "ʏ", "v": "v", "w": "ʋ", "z": "z",
+}
+
+# --------------------------------------------------------------------------- #
+# Polish
+# --------------------------------------------------------------------------- #
+_PL_RULES = {
+    "szcz": "ʂtʂ", "sz": "ʂ", "cz": "tʂ", "rz": "ʐ", "dz": "dz",
+    "dź": "dʑ", "dż": "dʐ", "ch": "x", "ć": "tɕ", "ś": "ɕ",
+    "ź": "ʑ", "ż": "ʐ", "ł": "w", "ń": "ɲ", "ą": "ɔ̃", "ę": "ɛ̃",
+    "ó": "u", "w": "v", "y": "ɨ",
+    "a": "a", "b": "b", "c": "ts", "d": "d", "e": "ɛ", "f": "f",
+    "g": "ɡ", "h": "x", "i": "i", "j": "j", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "ɔ", "p": "p", "r": "r", "s": "s",
+    "t": "t", "u": "u", "z": "z",
+}
+
+
 _G2P_REGISTRY: Dict[str, RuleG2P] = {}
 
 
@@ -357,6 +397,12 @@ def _get_g2p(voice: str) -> RuleG2P:
         g = RuleG2P(_IT_RULES, letters="a-zA-Zàèéìòù")
     elif base == "pt":
         g = RuleG2P(_PT_RULES, letters="a-zA-Zàáâãçéêíóôõú")
+    elif base == "ru":
+        g = RuleG2P(_RU_RULES, letters="а-яА-ЯёЁ")
+    elif base == "nl":
+        g = RuleG2P(_NL_RULES, letters="a-zA-Z")
+    elif base == "pl":
+        g = RuleG2P(_PL_RULES, letters="a-zA-Ząćęłńóśźż")
     elif base == "ar":
         g = RuleG2P(
             _AR_RULES,
@@ -370,7 +416,7 @@ def _get_g2p(voice: str) -> RuleG2P:
 
 
 def available_languages() -> List[str]:
-    return ["en-us", "en", "de", "es", "fr", "it", "pt", "ar"]
+    return ["en-us", "en", "de", "es", "fr", "it", "pt", "nl", "pl", "ru", "ar"]
 
 
 _LANG_SWITCH_RE = re.compile(r"\([a-z-]{2,10}\)")

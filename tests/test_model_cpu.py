@@ -244,3 +244,25 @@ def test_flow_reverse_cl_matches_channel_first():
         got = flow.reverse_cl((x * mask).transpose(1, 2).contiguous(),
                               mask.transpose(1, 2)).transpose(1, 2)
     assert torch.allclose(got, ref, atol=1e-5)
+
+
+def test_synthesis_is_deterministic():
+    """Same text through the same voice pack gives bitwise-identical
+    audio, at any batch size (per-utterance seeding)."""
+    import tempfile
+
+    import numpy as np
+
+    from sonata_amd.models import create_random_voice
+    from sonata_amd.models.voice import load_voice
+
+    with tempfile.TemporaryDirectory() as d:
+        pack = create_random_voice(d, "det", quality="x_low")
+        v1 = load_voice(pack, device="cpu")
+        v2 = load_voice(pack, device="cpu")
+        phon = "hˈɛloʊ wˈɜːld."
+        a = v1.speak_one_sentence(phon).samples
+        b = v2.speak_one_sentence(phon).samples
+        np.testing.assert_array_equal(a, b)
+        c = v1.speak_batch([phon, "ˈʌðɚ sˈɛntəns lˈɔŋɡɚ hˈɪɹ."])[0].samples
+        np.testing.assert_allclose(a, c, atol=1e-5)

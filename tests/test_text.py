@@ -134,7 +134,10 @@ def test_additional_languages():
     assert set(["fr", "it", "pt"]) <= set(available_languages())
     for lang, text in [("fr", "Bonjour le monde. Comment allez-vous?"),
                        ("it", "Ciao mondo. Come stai?"),
-                       ("pt", "Olá mundo. Tudo bem?")]:
+                       ("pt", "Olá mundo. Tudo bem?"),
+                       ("ru", "Привет мир. Как дела?"),
+                       ("nl", "Hallo wereld. Hoe gaat het?"),
+                       ("pl", "Witaj świecie. Jak się masz?")]:
         sents = text_to_phonemes(text, voice=lang)
         assert len(sents) == 2, (lang, sents)
         assert all(len(x) > 2 for x in sents)
