@@ -62,3 +62,26 @@ def test_tashkeel_applied_for_arabic(voices):
     assert v._tashkeel is not None
     sents = v.phonemize_text("كتب الولد")
     assert len(sents) >= 1 and len(sents[0]) > 0
+
+
+def test_all_languages_end_to_end(tmp_path):
+    """Every supported G2P language synthesizes audio through its own
+    voice pack (text -> phonemes -> ids -> VITS -> waveform)."""
+    from sonata_amd.models import create_random_voice
+    from sonata_amd.models.voice import load_voice
+    from sonata_amd.text.phonemizer import available_languages
+
+    texts = {
+        "en-us": "Hello world.", "en": "Hello world.",
+        "de": "Hallo Welt.", "es": "Hola mundo.",
+        "fr": "Bonjour le monde.", "it": "Ciao mondo.",
+        "pt": "Olá mundo.", "nl": "Hallo wereld.",
+        "pl": "Witaj świecie.", "ru": "Привет мир.",
+        "ar": "مرحبا بالعالم.",
+    }
+    for lang in available_languages():
+        pack = create_random_voice(str(tmp_path), f"lang_{lang}",
+                                   quality="x_low", language=lang)
+        v = load_voice(pack, device="cpu")
+        audios = v.speak_batch(list(v.phonemize_text(texts[lang])))
+        assert audios and all(len(a.samples) > 200 for a in audios), lang
