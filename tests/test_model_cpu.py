@@ -266,3 +266,28 @@ def test_synthesis_is_deterministic():
         np.testing.assert_array_equal(a, b)
         c = v1.speak_batch([phon, "ˈʌðɚ sˈɛntəns lˈɔŋɡɚ hˈɪɹ."])[0].samples
         np.testing.assert_allclose(a, c, atol=1e-5)
+
+
+def test_streamed_equals_oneshot_length_and_tail():
+    """The streamed decode (adaptive chunks + overlap-discard + 42-sample
+    crossfade) preserves the EXACT one-shot timeline: same total length,
+    and samples away from seams match closely (reference SpeechStreamer
+    semantics, piper/src/lib.rs:765-858)."""
+    import tempfile
+
+    import numpy as np
+
+    from sonata_amd.models import create_random_voice
+    from sonata_amd.models.voice import load_voice
+
+    with tempfile.TemporaryDirectory() as d:
+        pack = create_random_voice(d, "stream_eq", quality="x_low")
+        v = load_voice(pack, device="cpu")
+        phon = "hˈɛloʊ wˈɜːld ˈɛvɹiwˌʌn tʊdˈeɪ wˈʌn tˈuː θɹˈiː fˈoːɹ."
+        oneshot = v.speak_one_sentence(phon).samples
+        streamed = np.concatenate(list(v.stream_synthesis(phon, 20, 2)))
+        assert len(streamed) == len(oneshot)
+        # identical away from crossfaded seams (interior samples equal)
+        diff = np.abs(streamed - oneshot)
+        assert float(np.median(diff)) < 1e-6
+        assert (diff < 1e-5).mean() > 0.95
