@@ -13,6 +13,20 @@ import numpy as np
 from .samples import to_i16
 
 
+def wav_header(data_bytes: int, sample_rate: int, num_channels: int = 1,
+               sample_width: int = 2) -> bytes:
+    """44-byte RIFF/WAVE header for a PCM payload of `data_bytes`."""
+    import struct
+
+    byte_rate = sample_rate * num_channels * sample_width
+    block_align = num_channels * sample_width
+    return (b"RIFF" + struct.pack("<I", 36 + data_bytes) + b"WAVE"
+            + b"fmt " + struct.pack("<IHHIIHH", 16, 1, num_channels,
+                                    sample_rate, byte_rate, block_align,
+                                    sample_width * 8)
+            + b"data" + struct.pack("<I", data_bytes))
+
+
 def wav_bytes(samples, sample_rate: int, num_channels: int = 1,
               peak_normalize: bool = True) -> bytes:
     pcm = to_i16(samples, peak_normalize=peak_normalize).astype("<i2").tobytes()

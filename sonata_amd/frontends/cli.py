@@ -144,10 +144,26 @@ def process_request(synth, args, req: SynthesisRequest, n: int,
     else:
         stdout = stdout if stdout is not None else sys.stdout.buffer
         if args.mode == "realtime":
-            pieces = list(synth.synthesize_streamed(
-                req.text, out_cfg, args.chunk_size, args.chunk_padding))
-            samples = (np.concatenate(pieces) if pieces
-                       else np.zeros(0, dtype=np.float32))
+            # stream chunks to stdout AS THEY ARRIVE (reference cli
+            # main.rs:160-176): WAV header first (size patched only when
+            # stdout is seekable), then i16 PCM per chunk
+            from ..audio.samples import to_i16_bytes
+            from ..audio.wav import wav_header
+
+            header_pos = stdout.tell() if stdout.seekable() else None
+            stdout.write(wav_header(0, info.sample_rate))
+            for chunk in synth.synthesize_streamed(
+                    req.text, out_cfg, args.chunk_size, args.chunk_padding):
+                stdout.write(to_i16_bytes(chunk, peak_normalize=False))
+                stdout.flush()
+                total += len(chunk)
+            if header_pos is not None:
+                end = stdout.tell()
+                stdout.seek(header_pos)
+                stdout.write(wav_header(total * 2, info.sample_rate))
+                stdout.seek(end)
+            stdout.flush()
+            return total
         else:
             it = (synth.synthesize_lazy(req.text, out_cfg)
                   if args.mode == "lazy"
