@@ -93,6 +93,58 @@ def test_dp_world2(tmp_path):
     assert abs(results[0][2] - results[1][2]) < 1e-6
 
 
+def test_dp_world4_gather(tmp_path):
+    """World-size-4 gather correctness (uneven shards, bigger world)."""
+    import torch.multiprocessing as mp2
+
+    from sonata_amd.models import create_random_voice
+
+    pack = create_random_voice(str(tmp_path), "dp4", quality="x_low")
+    ctx = mp2.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_run_dp4, args=(r, 4, 29621, pack, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    results = [q.get() for _ in range(4)]
+    for p in procs:
+        p.join(timeout=120)
+    errs = [r for r in results if r[0] != "ok"]
+    assert not errs, errs
+
+
+def _run_dp4(rank, world, port, pack, q):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+        "LOCAL_RANK": str(rank),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    try:
+        import torch.distributed as dist
+
+        from sonata_amd.models.voice import load_voice
+        from sonata_amd.parallel import (DistributedSynthesizer,
+                                         init_distributed)
+
+        init_distributed(backend="gloo")
+        voice = load_voice(pack, device="cpu")
+        ds = DistributedSynthesizer(voice, batch_size=3)
+        phon = [f"wˈʌn {'tˈuː ' * (1 + i % 3)}." for i in range(7)]
+        out = ds.synthesize_corpus(phon)
+        if rank == 0:
+            assert out is not None and len(out) == 7
+            assert all(len(o) > 100 for o in out)
+        else:
+            assert out is None
+        dist.barrier()
+        dist.destroy_process_group()
+        q.put(("ok", rank))
+    except Exception:
+        import traceback
+
+        q.put(("err", rank, traceback.format_exc()))
+
+
 def test_shard_round_robin_uneven():
     from sonata_amd.parallel import shard_round_robin
 
