@@ -138,6 +138,62 @@ class RuleG2P:
 # English (en-US)
 # --------------------------------------------------------------------------- #
 _EN_LEXICON = {
+    # irregular / loan words the rule table mispronounces
+    "machine": "məʃin", "machines": "məʃinz", "technology": "tɛknɑlədʒi",
+    "technique": "tɛknik", "unique": "junik", "antique": "æntik",
+    "genre": "ʒɑnɹə", "garage": "ɡəɹɑʒ", "massage": "məsɑʒ",
+    "measure": "mɛʒɚ", "pleasure": "plɛʒɚ", "treasure": "tɹɛʒɚ",
+    "usual": "juʒuəl", "usually": "juʒuəli", "vision": "vɪʒən",
+    "decision": "dəsɪʒən", "television": "tɛləvɪʒən",
+    "occasion": "əkeɪʒən", "version": "vɝʒən", "asia": "eɪʒə",
+    "europe": "jʊɹəp", "european": "jʊɹəpiən", "america": "əmɛɹɪkə",
+    "american": "əmɛɹɪkən", "africa": "æfɹɪkə", "australia": "ɔstɹeɪljə",
+    "russia": "ɹʌʃə", "russian": "ɹʌʃən",
+    "science": "saɪəns", "scientist": "saɪəntɪst", "scene": "sin",
+    "muscle": "mʌsəl", "island": "aɪlənd", "aisle": "aɪl",
+    "answer": "ænsɚ", "listen": "lɪsən", "often": "ɔfən",
+    "castle": "kæsəl", "whistle": "wɪsəl", "wednesday": "wɛnzdeɪ",
+    "february": "fɛbjuɛɹi", "colonel": "kɝnəl", "receipt": "ɹəsit",
+    "debt": "dɛt", "doubt": "daʊt", "subtle": "sʌtəl",
+    "tomb": "tum", "womb": "wum", "comb": "koʊm", "climb": "klaɪm",
+    "thumb": "θʌm", "lamb": "læm", "plumber": "plʌmɚ",
+    "business": "bɪznəs", "busy": "bɪzi", "beautiful": "bjutɪfəl",
+    "beauty": "bjuti", "language": "læŋɡwɪdʒ", "languages": "læŋɡwɪdʒəz",
+    "once": "wʌns", "only": "oʊnli", "own": "oʊn", "move": "muv",
+    "movie": "muvi", "prove": "pɹuv", "lose": "luz", "whose": "huz",
+    "shoe": "ʃu", "shoes": "ʃuz", "canoe": "kənu",
+    "sugar": "ʃʊɡɚ", "sure": "ʃʊɹ", "surely": "ʃʊɹli",
+    "ocean": "oʊʃən", "special": "spɛʃəl", "especially": "əspɛʃəli",
+    "social": "soʊʃəl", "official": "əfɪʃəl", "ancient": "eɪnʃənt",
+    "patient": "peɪʃənt", "efficient": "əfɪʃənt",
+    "question": "kwɛstʃən", "questions": "kwɛstʃənz",
+    "suggestion": "səɡdʒɛstʃən", "digestion": "daɪdʒɛstʃən",
+    "nature": "neɪtʃɚ", "natural": "nætʃɚəl", "picture": "pɪktʃɚ",
+    "future": "fjutʃɚ", "culture": "kʌltʃɚ", "capture": "kæptʃɚ",
+    "century": "sɛntʃɚi", "actual": "æktʃuəl", "actually": "æktʃuəli",
+    "iron": "aɪɚn", "choir": "kwaɪɚ", "heart": "hɑɹt",
+    "heard": "hɝd", "earth": "ɝθ", "early": "ɝli", "learn": "lɝn",
+    "search": "sɝtʃ", "research": "ɹisɝtʃ",
+    "eye": "aɪ", "eyes": "aɪz", "bury": "bɛɹi", "buried": "bɛɹid",
+    "blood": "blʌd", "flood": "flʌd", "door": "dɔɹ", "floor": "flɔɹ",
+    "poor": "pʊɹ", "tour": "tʊɹ", "your": "jʊɹ", "hour": "aʊɚ",
+    "hours": "aʊɚz", "honest": "ɑnəst", "honor": "ɑnɚ", "ghost": "ɡoʊst",
+    "friend": "fɹɛnd", "friends": "fɹɛndz", "again": "əɡɛn",
+    "against": "əɡɛnst", "says": "sɛz", "does": "dʌz", "done": "dʌn",
+    "gone": "ɡɔn", "none": "nʌn", "come": "kʌm", "becomes": "bəkʌmz",
+    "become": "bəkʌm", "above": "əbʌv", "love": "lʌv", "give": "ɡɪv",
+    "live": "lɪv", "lives": "lɪvz", "liver": "lɪvɚ",
+    "any": "ɛni", "anything": "ɛniθɪŋ", "every": "ɛvɹi",
+    "everything": "ɛvɹiθɪŋ", "everyone": "ɛvɹiwʌn",
+    "something": "sʌmθɪŋ", "someone": "sʌmwʌn", "nothing": "nʌθɪŋ",
+    "idea": "aɪdiə", "ideas": "aɪdiəz", "area": "ɛɹiə",
+    "real": "ɹiəl", "really": "ɹɪli", "create": "kɹieɪt",
+    "created": "kɹieɪtəd", "theater": "θiətɚ", "theatre": "θiətɚ",
+    "quiet": "kwaɪət", "quite": "kwaɪt", "guide": "ɡaɪd",
+    "guitar": "ɡɪtɑɹ", "building": "bɪldɪŋ", "build": "bɪld",
+    "built": "bɪlt", "juice": "dʒus", "fruit": "fɹut", "suit": "sut",
+    "engine": "ɛndʒən", "engineer": "ɛndʒənɪɹ", "examine": "ɪɡzæmən",
+    "medicine": "mɛdəsən", "determine": "dətɝmən", "imagine": "ɪmædʒən",
     "a": "ə", "an": "ən", "the": "ðə", "of": "əv", "to": "tu", "and": "ænd",
     "in": "ɪn", "is": "ɪz", "it": "ɪt", "you": "ju", "that": "ðæt",
     "he": "hi", "she": "ʃi", "was": "wəz", "for": "fɔɹ", "on": "ɑn",
