@@ -189,6 +189,150 @@ __global__ __launch_bounds__(512) void conv1d_cl_kernel(
 }
 
 // --------------------------------------------------------------------------
+// Hybrid conv: A (X window) staged in LDS, B (weight taps) loaded per
+// wave STRAIGHT from global memory.  Weights are tiny and shared by
+// thousands of blocks (L2-hot); skipping the Ws LDS stage removes the
+// per-tap-chunk barrier pair entirely (one barrier pair per K-slice)
+// and frees ~20 KB LDS -> more blocks/CU.  A-side stays staged: the
+// fully-direct variant measured 1.6-2.6x slower (see note above).
+// --------------------------------------------------------------------------
+template <int BM, int BN, int WGM, int WGN>
+__global__ __launch_bounds__(512) void conv1d_cl_wdirect_kernel(
+    const bf16* __restrict__ x,     // [B][Tin][Cin]
+    const bf16* __restrict__ w,     // [ntaps][CoutP][CinP]
+    const float* __restrict__ bias,
+    bf16* __restrict__ out,         // [B][Tout][Cout]
+    const bf16* __restrict__ resid,
+    const int* __restrict__ out_lens,
+    int Cin, int CinP, int Cout, int CoutP, long Tin, long Tout,
+    int ntaps, int dil, int pad, float pre_slope, int act_mode,
+    float post_slope) {
+  constexpr int ROWS = BM + HALO_MAX;
+  const long t0 = (long)blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const int b = blockIdx.z;
+
+  __shared__ bf16 Xs[ROWS][BKP];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid / WGN;
+  const int wc = wid % WGN;
+  constexpr int WM = BM / WGM;
+  constexpr int WN = BN / WGN;
+  constexpr int MT = WM / 16;
+  constexpr int NT = WN / 16;
+  const int kl = lane >> 4;
+  const int il = lane & 15;
+
+  f32x4 acc[MT][NT];
+#pragma unroll
+  for (int i = 0; i < MT; ++i)
+#pragma unroll
+    for (int j = 0; j < NT; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const bf16* xb = x + (long)b * Tin * Cin;
+  const long row0 = t0 - pad;
+  const int halo = (ntaps - 1) * dil;
+  const int rows_used = BM + halo;
+  const bool t_interior = (row0 >= 0) && (row0 + rows_used <= Tin);
+  // this lane's weight row base for each nj fragment
+  const int wcol = n0 + wc * WN + il;
+
+  for (int c0 = 0; c0 < CinP; c0 += BK) {
+    const bool c_interior = (c0 + BK) <= Cin;
+    __syncthreads();  // protect Xs against the previous slice's readers
+    if (t_interior && c_interior) {
+      for (int u = tid; u < rows_used * 4; u += 512) {
+        const int r = u >> 2, ch = (u & 3) * 8;
+        bf16 v8[8];
+        *(ulonglong2*)v8 =
+            *(const ulonglong2*)&xb[(row0 + r) * Cin + c0 + ch];
+        if (pre_slope >= 0.f) {
+#pragma unroll
+          for (int q = 0; q < 8; ++q)
+            v8[q] = f2bf(lrelu_(bf2f(v8[q]), pre_slope));
+        }
+        *(ulonglong2*)&Xs[r][ch] = *(ulonglong2*)v8;
+      }
+    } else {
+      for (int u = tid; u < rows_used * 4; u += 512) {
+        const int r = u >> 2, ch = (u & 3) * 8;
+        const long t = row0 + r;
+        bf16 v8[8];
+        if (t >= 0 && t < Tin && c_interior) {
+          *(ulonglong2*)v8 = *(const ulonglong2*)&xb[t * Cin + c0 + ch];
+          if (pre_slope >= 0.f) {
+#pragma unroll
+            for (int q = 0; q < 8; ++q)
+              v8[q] = f2bf(lrelu_(bf2f(v8[q]), pre_slope));
+          }
+        } else if (t >= 0 && t < Tin) {
+#pragma unroll
+          for (int q = 0; q < 8; ++q) {
+            const int c = c0 + ch + q;
+            float v = c < Cin ? bf2f(xb[t * Cin + c]) : 0.f;
+            if (pre_slope >= 0.f) v = lrelu_(v, pre_slope);
+            v8[q] = f2bf(v);
+          }
+        } else {
+#pragma unroll
+          for (int q = 0; q < 8; ++q) v8[q] = f2bf(0.f);
+        }
+        *(ulonglong2*)&Xs[r][ch] = *(ulonglong2*)v8;
+      }
+    }
+    __syncthreads();
+
+    for (int tap = 0; tap < ntaps; ++tap) {
+      bf16x8 b_frag[NT];
+#pragma unroll
+      for (int nj = 0; nj < NT; ++nj)
+        b_frag[nj] = *(const bf16x8*)&w[
+            ((long)tap * CoutP + wcol + nj * 16) * CinP + c0 + kl * 8];
+      const int toff = tap * dil;
+#pragma unroll
+      for (int mi = 0; mi < MT; ++mi) {
+        const bf16x8 a_frag =
+            *(const bf16x8*)&Xs[wr * WM + mi * 16 + il + toff][kl * 8];
+#pragma unroll
+        for (int nj = 0; nj < NT; ++nj)
+          acc[mi][nj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag, b_frag[nj], acc[mi][nj], 0, 0, 0);
+      }
+    }
+  }
+
+  bf16* ob = out + (long)b * Tout * Cout;
+  const bf16* rb = resid ? resid + (long)b * Tout * Cout : nullptr;
+  const long lim = out_lens ? min((long)out_lens[b], Tout) : Tout;
+#pragma unroll
+  for (int mi = 0; mi < MT; ++mi) {
+#pragma unroll
+    for (int rg = 0; rg < 4; ++rg) {
+      const long t = t0 + wr * WM + mi * 16 + kl * 4 + rg;
+      if (t >= Tout) continue;
+      const bool live = t < lim;
+#pragma unroll
+      for (int nj = 0; nj < NT; ++nj) {
+        const int co = n0 + wc * WN + nj * 16 + il;
+        if (co >= Cout) continue;
+        float v = 0.f;
+        if (live) {
+          v = acc[mi][nj][rg];
+          if (bias) v += bias[co];
+          if (act_mode == ACT_LRELU) v = lrelu_(v, post_slope);
+          else if (act_mode == ACT_TANH) v = tanhf(v);
+          if (rb) v += bf2f(rb[t * Cout + co]);
+        }
+        ob[t * Cout + co] = f2bf(v);
+      }
+    }
+  }
+}
+
+// --------------------------------------------------------------------------
 // ConvTranspose1d, channel-last, phase-merged (k = 2*stride, KR = 2):
 // each block computes ALL s phases of a BMV(v) x BN(co) tile.
 // D[r][v][co] = sum_{m,ci} X[v-m][ci] * W[r*KR+m][co][ci];
@@ -409,6 +553,49 @@ torch::Tensor conv1d_cl_fused(torch::Tensor x, torch::Tensor w_perm,
   else if (Cout >= 64) LAUNCH_CL(256, 64, 4, 2, 4);
   else LAUNCH_CL(256, 32, 4, 2, 4);
 #undef LAUNCH_CL
+  return out;
+}
+
+torch::Tensor conv1d_cl_wdirect(torch::Tensor x, torch::Tensor w_perm,
+                                c10::optional<torch::Tensor> bias, long Cout,
+                                long k, long padding, long dilation,
+                                double pre_lrelu, long act_mode,
+                                double post_slope,
+                                c10::optional<torch::Tensor> residual,
+                                c10::optional<torch::Tensor> out_lens) {
+  TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16);
+  const long B = x.size(0), Tin = x.size(1), Cin = x.size(2);
+  TORCH_CHECK((k - 1) * dilation <= HALO_MAX);
+  const long Tout = Tin + 2 * padding - dilation * (k - 1);
+  const int CoutP = w_perm.size(1), CinP = w_perm.size(2);
+  auto out = torch::empty({B, Tout, Cout}, x.options());
+  if (out.numel() == 0) return out;
+  torch::Tensor bias_f;
+  const float* bias_p = nullptr;
+  if (bias.has_value()) {
+    bias_f = bias->scalar_type() == at::kFloat ? *bias
+                                               : bias->to(at::kFloat).contiguous();
+    bias_p = bias_f.data_ptr<float>();
+  }
+  const bf16* res_p = nullptr;
+  if (residual.has_value()) res_p = (const bf16*)residual->data_ptr();
+  const int* lens_p = nullptr;
+  if (out_lens.has_value()) lens_p = out_lens->data_ptr<int>();
+  hipStream_t st = cur_stream3();
+#define LAUNCH_WD(BM, BN, WGM, WGN)                                         \
+  hipLaunchKernelGGL((conv1d_cl_wdirect_kernel<BM, BN, WGM, WGN>),          \
+                     dim3(ceil_div(Tout, BM), ceil_div(Cout, BN), B),       \
+                     dim3(512), 0, st, (const bf16*)x.data_ptr(),           \
+                     (const bf16*)w_perm.data_ptr(), bias_p,                \
+                     (bf16*)out.data_ptr(), res_p, lens_p, (int)Cin, CinP,  \
+                     (int)Cout, CoutP, Tin, Tout, (int)k, (int)dilation,    \
+                     (int)padding, (float)pre_lrelu, (int)act_mode,         \
+                     (float)post_slope)
+  if (Cout >= 128) LAUNCH_WD(256, 128, 4, 2);
+  else if (Cout >= 64) LAUNCH_WD(256, 64, 4, 2);
+  else LAUNCH_WD(256, 32, 4, 2);
+#undef LAUNCH_WD
   return out;
 }
 

@@ -45,6 +45,13 @@ torch::Tensor convtranspose1d_cl_fused(torch::Tensor x, torch::Tensor w_perm,
                                        long Cout, long k, long stride,
                                        long padding, double pre_lrelu,
                                        c10::optional<torch::Tensor> out_lens);
+torch::Tensor conv1d_cl_wdirect(torch::Tensor x, torch::Tensor w_perm,
+                                c10::optional<torch::Tensor> bias, long Cout,
+                                long k, long padding, long dilation,
+                                double pre_lrelu, long act_mode,
+                                double post_slope,
+                                c10::optional<torch::Tensor> residual,
+                                c10::optional<torch::Tensor> out_lens);
 torch::Tensor conv1d_direct_cl(torch::Tensor x, torch::Tensor w_perm,
                                c10::optional<torch::Tensor> bias, long Cout,
                                long k, long padding, long dilation,
@@ -76,6 +83,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "channel-last MFMA conv1d, fused act/residual/mask");
   m.def("convtranspose1d_cl_fused", &convtranspose1d_cl_fused,
         "channel-last MFMA transposed conv1d, phase-merged");
+  m.def("conv1d_cl_wdirect", &conv1d_cl_wdirect,
+        "hybrid: A staged in LDS, weights direct from L2 (no W barriers)");
   m.def("conv1d_direct_cl", &conv1d_direct_cl,
         "zero-LDS direct channel-last conv (L1/L2-fed MFMA)");
   m.def("resblock_pair_cl_fused", &resblock_pair_cl_fused,

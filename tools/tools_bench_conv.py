@@ -179,3 +179,30 @@ for name, Cin, Cout, T, k, dil in [("res128 k3", 128, 128, 16384, 3, 1),
         dt = (time.perf_counter()-t0)/N
         fl = 2*B*Cin*Cout*k*T
         print(f"{tag} {name:14s} {dt*1e3:8.3f} ms  {fl/dt/1e12:7.1f} TF" + (f"  err={err:.4f}" if tag=="direct" else ""))
+
+# hybrid: A staged, W direct-from-L2
+print("--- hybrid W-direct vs staged cl conv ---")
+for name, Cin, Cout, T, k, dil in [("res128 k3", 128, 128, 16384, 3, 1),
+                                   ("res128 k11d5", 128, 128, 16384, 11, 5),
+                                   ("res64 k3", 64, 64, 32768, 3, 1),
+                                   ("res32 k3", 32, 32, 65536, 3, 1),
+                                   ("res32 k11d5", 32, 32, 65536, 11, 5),
+                                   ("res256 k3", 256, 256, 2048, 3, 1)]:
+    x = (torch.randn(B, T, Cin)/4).to(torch.bfloat16).to(dev)
+    w = (torch.randn(Cout, Cin, k)/(Cin*k)**0.5).to(torch.bfloat16).to(dev)
+    bias = (torch.randn(Cout)/10).to(dev)
+    pad = (k-1)*dil//2
+    wp = _conv_weight_mfma(w); bf = _bias_f32(bias)
+    ref = leaky_conv1d_cl(x, w, bias, padding=pad, dilation=dil, pre_lrelu=0.1)
+    got = ext.conv1d_cl_wdirect(x, wp, bf, Cout, k, pad, dil, 0.1, 0, 0.0, None, None)
+    err = (got.float()-ref.float()).abs().max().item()/max(ref.float().abs().max().item(),1e-6)
+    for fn, tag in [(lambda: leaky_conv1d_cl(x, w, bias, padding=pad, dilation=dil, pre_lrelu=0.1), "staged"),
+                    (lambda: ext.conv1d_cl_wdirect(x, wp, bf, Cout, k, pad, dil, 0.1, 0, 0.0, None, None), "wdirect")]:
+        for _ in range(3): y = fn()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter(); N = 10
+        for _ in range(N): y = fn()
+        torch.cuda.synchronize()
+        dt = (time.perf_counter()-t0)/N
+        fl = 2*B*Cin*Cout*k*T
+        print(f"{tag:8s} {name:14s} {dt*1e3:8.3f} ms  {fl/dt/1e12:7.1f} TF" + (f"  err={err:.4f}" if tag=="wdirect" else ""))
