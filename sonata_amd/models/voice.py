@@ -221,6 +221,21 @@ class VitsVoice(SonataModel):
 
     def _stream_decode(self, z, y_mask, g, chunk_size: int,
                        chunk_padding: int) -> Iterator[np.ndarray]:
+        from ..utils.graphs import DecodeGraphCache, enabled as graphs_on
+
+        graph_cache = None
+        if (graphs_on() and self.device.type == "cuda" and g is None):
+            # hipGraph replay per chunk shape (launch-bound at small B);
+            # decode is capture-safe (no host syncs inside)
+            if not hasattr(self, "_decode_graphs"):
+                if self._engine is not None:
+                    self._decode_graphs = DecodeGraphCache(
+                        lambda zc, mc: self._engine.decode(zc, mc, None,
+                                                           None))
+                else:
+                    self._decode_graphs = DecodeGraphCache(
+                        lambda zc, mc: self.net.decode(zc, mc, None))
+            graph_cache = self._decode_graphs
         hop = self.net.arch.hop_length
         num_frames = z.shape[-1]
         # Overlap-crossfade at seams without changing the timeline: each
@@ -234,7 +249,9 @@ class VitsVoice(SonataModel):
             z_c = z[:, :, spec.mel_start : spec.mel_end].contiguous()
             m_c = y_mask[:, :, spec.mel_start : spec.mel_end].contiguous()
             with stage_timer("decode_chunk", self.device):
-                if self._engine is not None:
+                if graph_cache is not None:
+                    audio = graph_cache(z_c, m_c)
+                elif self._engine is not None:
                     audio = self._engine.decode(z_c, m_c, g, None)
                 else:
                     audio = self.net.decode(z_c, m_c, g)
