@@ -948,27 +948,35 @@ class VitsModel(nn.Module):
         `generators`: one torch.Generator per utterance (deterministic,
         batch-composition-independent sampling)."""
         B = ids.shape[0]
-        x, m_p, logs_p, x_mask = self.enc_p(ids, lengths)
         g = None
         if self.emb_g is not None:
             if sid is None:
                 sid = torch.zeros(ids.shape[0], dtype=torch.long,
                                   device=ids.device)
             g = self.emb_g(sid).unsqueeze(-1)  # [B, gin, 1]
+        dtype = self.enc_p.emb.weight.dtype
         sdp_noise = masked_noise_rows(B, 2, ids.shape[1], lengths, generators,
-                                      x.device, x.dtype)
-        logw = self.dp.infer(x, x_mask, g=g, noise_scale=noise_w,
-                             noise=sdp_noise)
+                                      ids.device, dtype)
+        x, m_p, logs_p, x_mask, logw = self.encode_phase1(
+            ids, lengths, g, noise_w, sdp_noise)
+        return self.encode_phase2(m_p, logs_p, x_mask, logw, g,
+                                  noise_scale, length_scale, generators)
+
+    def encode_phase2(self, m_p, logs_p, x_mask, logw, g,
+                      noise_scale, length_scale, generators):
+        """Duration -> frames -> prior -> flow: the shape-DYNAMIC suffix
+        of the encoder (frame count F is data-dependent; stays eager)."""
+        B = m_p.shape[0]
         w = torch.exp(logw) * x_mask * length_scale
         w_ceil = torch.ceil(w)
         y_lengths = torch.clamp_min(torch.sum(w_ceil, [1, 2]), 1).long()
-        y_mask = sequence_mask(y_lengths).to(x.dtype)
+        y_mask = sequence_mask(y_lengths).to(m_p.dtype)
         durations = w_ceil.squeeze(1).long()
         m_p_f = expand_states(m_p, durations, y_lengths)
         logs_p_f = expand_states(logs_p, durations, y_lengths)
         prior_noise = masked_noise_rows(
             B, m_p_f.shape[1], m_p_f.shape[2], y_lengths, generators,
-            x.device, x.dtype,
+            m_p.device, m_p.dtype,
         )
         z_p = prior_sample(m_p_f, logs_p_f, y_mask, prior_noise, noise_scale)
         from ..ops import use_hip
@@ -983,6 +991,16 @@ class VitsModel(nn.Module):
             return z_cl.transpose(1, 2), y_mask, g
         z = self.flow(z_p, y_mask, g=g, reverse=True)
         return z, y_mask, g
+
+    def encode_phase1(self, ids, lengths, g, noise_w, sdp_noise):
+        """Text encoder + duration predictor: the shape-static prefix of
+        the encoder (hipGraph-capturable per padded-T bucket — everything
+        after depends on the predicted frame count).  Padding invariance
+        holds exactly (masked everywhere), so callers may pad T freely."""
+        x, m_p, logs_p, x_mask = self.enc_p(ids, lengths)
+        logw = self.dp.infer(x, x_mask, g=g, noise_scale=noise_w,
+                             noise=sdp_noise)
+        return x, m_p, logs_p, x_mask, logw
 
     def decode(self, z: torch.Tensor, y_mask: torch.Tensor,
                g: Optional[torch.Tensor] = None,

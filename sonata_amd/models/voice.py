@@ -219,6 +219,37 @@ class VitsVoice(SonataModel):
         yield from self._stream_decode(z, y_mask, g, chunk_size,
                                        chunk_padding)
 
+    @torch.no_grad()
+    def _stream_graphed(self, phonemes, cfg, ids, lengths, chunk_size,
+                        chunk_padding):
+        """Realtime path with hipGraph-captured encoder phase 1 (text
+        encoder + duration predictor) per padded-T bucket; phase 2 (frame
+        expansion, flow) stays eager (F is data-dependent); chunk decode
+        replays per-shape graphs.  Padding invariance is exact (masked
+        everywhere), so bucketing ids costs nothing numerically."""
+        from ..utils.graphs import TupleGraphCache
+        from .vits import masked_noise_rows
+
+        T = ids.shape[1]
+        Tpad = (T + 31) // 32 * 32
+        if Tpad != T:
+            ids = torch.nn.functional.pad(ids, (0, Tpad - T))
+        gens = self._generators([phonemes], cfg.speaker_id)
+        noise = masked_noise_rows(1, 2, Tpad, lengths, gens,
+                                  self.device, self.dtype)
+        noise = noise * cfg.noise_w  # pre-scale: graph runs noise_w=1
+        if not hasattr(self, "_phase1_graphs"):
+            self._phase1_graphs = TupleGraphCache(
+                lambda i, l, n: self.net.encode_phase1(i, l, None, 1.0, n))
+        with stage_timer("encode_graph", self.device):
+            x, m_p, logs_p, x_mask, logw = self._phase1_graphs(
+                ids, lengths, noise)
+            z, y_mask, g = self.net.encode_phase2(
+                m_p, logs_p, x_mask, logw, None, cfg.noise_scale,
+                cfg.length_scale, gens)
+        yield from self._stream_decode(z, y_mask, g, chunk_size,
+                                       chunk_padding)
+
     def _stream_decode(self, z, y_mask, g, chunk_size: int,
                        chunk_padding: int) -> Iterator[np.ndarray]:
         from ..utils.graphs import DecodeGraphCache, enabled as graphs_on

@@ -60,3 +60,41 @@ class DecodeGraphCache:
         entry = (g, z_buf, m_buf, out_buf)
         self._graphs[key] = entry
         return entry
+
+
+class TupleGraphCache:
+    """Graph cache for fn(*tensors) -> tuple(tensors), keyed by input
+    shapes/dtypes."""
+
+    def __init__(self, fn: Callable, max_shapes: int = 16):
+        self.fn = fn
+        self.max_shapes = max_shapes
+        self._graphs: Dict[Tuple, tuple] = {}
+
+    def __call__(self, *inputs: torch.Tensor):
+        key = tuple((tuple(t.shape), t.dtype) for t in inputs)
+        entry = self._graphs.get(key)
+        if entry is None:
+            if len(self._graphs) >= self.max_shapes:
+                return self.fn(*inputs)
+            entry = self._capture(inputs, key)
+        g, in_bufs, out_bufs = entry
+        for buf, t in zip(in_bufs, inputs):
+            buf.copy_(t)
+        g.replay()
+        return out_bufs
+
+    def _capture(self, inputs, key):
+        in_bufs = tuple(t.clone() for t in inputs)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self.fn(*in_bufs)
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            out_bufs = self.fn(*in_bufs)
+        entry = (g, in_bufs, tuple(out_bufs))
+        self._graphs[key] = entry
+        return entry
