@@ -886,7 +886,7 @@ def masked_noise_rows(
     batch: int,
     channels: int,
     max_len: int,
-    lengths: torch.Tensor,
+    lengths,
     generators: Optional[List[torch.Generator]],
     device,
     dtype,
@@ -894,10 +894,15 @@ def masked_noise_rows(
     """Standard-normal noise [B, C, T] where row b is drawn from
     generators[b] on its own [C, len_b] grid and zero beyond — so an
     utterance's noise stream is independent of batch padding/composition
-    (SURVEY.md §7 hard part 7: seed per utterance, not per rank)."""
+    (SURVEY.md §7 hard part 7: seed per utterance, not per rank).
+
+    `lengths` may be a tensor or a plain int list (pre-fetched to avoid
+    per-row device syncs on the latency path)."""
+    if torch.is_tensor(lengths):
+        lengths = lengths.tolist()
     out = torch.zeros((batch, channels, max_len), device=device, dtype=dtype)
     for b in range(batch):
-        lb = int(lengths[b].item())
+        lb = int(lengths[b])
         gen = generators[b] if generators is not None else None
         out[b, :, :lb] = torch.randn((channels, lb), device=device,
                                      dtype=dtype, generator=gen)
@@ -970,12 +975,17 @@ class VitsModel(nn.Module):
         w = torch.exp(logw) * x_mask * length_scale
         w_ceil = torch.ceil(w)
         y_lengths = torch.clamp_min(torch.sum(w_ceil, [1, 2]), 1).long()
-        y_mask = sequence_mask(y_lengths).to(m_p.dtype)
+        # ONE host sync fetches all frame counts (the latency path was
+        # paying several pipeline drains: sequence_mask .max().item(),
+        # expand_states F_max, per-row noise .item()s)
+        lens_list = y_lengths.tolist()
+        F_max = max(lens_list)
+        y_mask = sequence_mask(y_lengths, F_max).to(m_p.dtype)
         durations = w_ceil.squeeze(1).long()
-        m_p_f = expand_states(m_p, durations, y_lengths)
-        logs_p_f = expand_states(logs_p, durations, y_lengths)
+        m_p_f = expand_states(m_p, durations, y_lengths, F_max)
+        logs_p_f = expand_states(logs_p, durations, y_lengths, F_max)
         prior_noise = masked_noise_rows(
-            B, m_p_f.shape[1], m_p_f.shape[2], y_lengths, generators,
+            B, m_p_f.shape[1], m_p_f.shape[2], lens_list, generators,
             m_p.device, m_p.dtype,
         )
         z_p = prior_sample(m_p_f, logs_p_f, y_mask, prior_noise, noise_scale)

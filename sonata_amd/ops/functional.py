@@ -92,22 +92,24 @@ def prior_sample(
 # Length regulator: expand phoneme states to frame states by durations
 # --------------------------------------------------------------------------- #
 def expand_states(
-    stats: torch.Tensor, durations: torch.Tensor, y_lengths: torch.Tensor
+    stats: torch.Tensor, durations: torch.Tensor, y_lengths: torch.Tensor,
+    F_max: Optional[int] = None,
 ) -> torch.Tensor:
     """stats: [B, C, T_ph]; durations: [B, T_ph] int frame counts;
     returns [B, C, F_max] where F_max = y_lengths.max().
 
     The attention path matrix of VITS inference: frame f copies phoneme p
     where cum_dur[p-1] <= f < cum_dur[p]."""
+    if F_max is None:
+        F_max = int(y_lengths.max().item())
     if use_hip(stats):
         ext = hip_ext(required=True)
         return ext.expand_states(
             stats.contiguous(),
             durations.to(torch.int32).contiguous(),
-            int(y_lengths.max().item()),
+            F_max,
         )
     B, C, T = stats.shape
-    F_max = int(y_lengths.max().item())
     out = stats.new_zeros((B, C, F_max))
     for b in range(B):
         cum = torch.cumsum(durations[b], dim=0)
