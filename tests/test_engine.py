@@ -139,3 +139,56 @@ def test_cli_binary_streaming(pack, tmp_path):
     assert r.returncode == 0, r.stderr
     assert "first chunk in" in r.stderr
     assert out.read_bytes()[:4] == b"RIFF"
+
+
+def test_engine_rejects_corrupt_inputs(ext, tmp_path):
+    """The native loaders (minijson + safetensors reader) fail with clean
+    errors, not crashes."""
+    bad_json = tmp_path / "bad.json"
+    bad_json.write_text("{not json")
+    with pytest.raises(RuntimeError):
+        ext.VitsEngine(str(bad_json), "cpu", "f32")
+
+    # valid config, truncated safetensors
+    import json as _json
+
+    cfg = tmp_path / "v.json"
+    cfg.write_text(_json.dumps({"audio": {"quality": "x_low"}}))
+    st = tmp_path / "v.safetensors"
+    st.write_bytes(b"\x00" * 4)  # shorter than the 8-byte header length
+    with pytest.raises(RuntimeError, match="safetensors"):
+        ext.VitsEngine(str(cfg), "cpu", "f32")
+
+    # header length pointing past EOF
+    import struct
+
+    st.write_bytes(struct.pack("<Q", 1 << 30) + b"{}")
+    with pytest.raises(RuntimeError, match="header"):
+        ext.VitsEngine(str(cfg), "cpu", "f32")
+
+    # well-formed header but bad offsets
+    hdr = _json.dumps({"w": {"dtype": "F32", "shape": [4],
+                             "data_offsets": [0, 999]}}).encode()
+    st.write_bytes(struct.pack("<Q", len(hdr)) + hdr + b"\x00" * 8)
+    with pytest.raises(RuntimeError):
+        ext.VitsEngine(str(cfg), "cpu", "f32")
+
+
+def test_engine_missing_weight_clean_error(ext, tmp_path):
+    """A pack with weights missing for the declared arch raises a clean
+    'missing weight' error on first inference."""
+    import json as _json
+    import struct
+
+    cfg = tmp_path / "m.json"
+    cfg.write_text(_json.dumps({"audio": {"quality": "x_low"}}))
+    hdr = _json.dumps({"enc_p.emb.weight": {
+        "dtype": "F32", "shape": [4, 4], "data_offsets": [0, 64]}}).encode()
+    (tmp_path / "m.safetensors").write_bytes(
+        struct.pack("<Q", len(hdr)) + hdr + b"\x00" * 64)
+    eng = ext.VitsEngine(str(cfg), "cpu", "f32")
+    import torch as _t
+
+    with pytest.raises(RuntimeError, match="missing weight"):
+        eng.infer(_t.tensor([[1, 5, 2]]), _t.tensor([3]), None,
+                  0.667, 1.0, 0.8, [1])
