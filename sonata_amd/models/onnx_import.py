@@ -148,7 +148,7 @@ _RULES = [
     (r"^enc_p\.encoder\.ffn_layers\.(\d+)\.conv_2\.", r"enc_p.ffn_layers.\1.conv2."),
     (r"^dec\.cond\.", "dec.cond."),
 ]
-_SKIP = re.compile(r"^(enc_q\.|dp\.post_|.*\.weight_[gv]$)")
+_SKIP = re.compile(r"^(enc_q\.|dp\.post_)")
 
 
 def map_vits_name(name: str) -> str:
@@ -169,10 +169,26 @@ def import_onnx_voice(onnx_path: str, out_path: str = None,
     inits = parse_onnx_initializers(onnx_path)
     if not inits:
         raise ModelError(f"no initializers found in {onnx_path}")
+    # fuse any weight-norm pairs the exporter left in:
+    # weight = g * v / ||v||  (norm over all dims but 0)
+    fused: Dict[str, np.ndarray] = {}
+    for name in list(inits):
+        if name.endswith(".weight_v"):
+            base = name[: -len(".weight_v")]
+            gname = base + ".weight_g"
+            if gname in inits:
+                v = inits[name].astype(np.float64)
+                gw = inits[gname].astype(np.float64)
+                norm = np.sqrt((v ** 2).sum(
+                    axis=tuple(range(1, v.ndim)), keepdims=True))
+                fused[base + ".weight"] = (gw * v / np.maximum(norm, 1e-12)
+                                           ).astype(np.float32)
+    for name, arr in fused.items():
+        inits.setdefault(name, arr)
     state: Dict[str, "torch.Tensor"] = {}
     skipped: List[str] = []
     for name, arr in inits.items():
-        if _SKIP.match(name):
+        if _SKIP.match(name) or name.endswith((".weight_g", ".weight_v")):
             continue
         mapped = map_vits_name(name)
         if not re.match(r"^(enc_p|dp|flow|dec|emb_g)\.|^emb_g$", mapped):

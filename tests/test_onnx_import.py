@@ -113,3 +113,30 @@ def test_import_synthetic_onnx(tmp_path):
     voice = load_voice(pack, device="cpu")
     audio = voice.speak_one_sentence("tˈɛst.")
     assert len(audio.samples) > 500
+
+
+def test_weight_norm_pairs_are_fused(tmp_path):
+    """Exports that kept weight_norm parametrization (weight_g/weight_v)
+    get fused into plain weights: w = g * v/||v||."""
+    rng = np.random.default_rng(0)
+    v = rng.standard_normal((6, 4, 3)).astype(np.float32)
+    g = rng.standard_normal((6, 1, 1)).astype(np.float32)
+    norm = np.sqrt((v.astype(np.float64) ** 2).sum(axis=(1, 2),
+                                                   keepdims=True))
+    expected = (g * v / norm).astype(np.float32)
+    tensors = [("flow.flows.0.enc.in_layers.0.weight_v", v),
+               ("flow.flows.0.enc.in_layers.0.weight_g", g),
+               ("flow.flows.0.enc.in_layers.0.bias",
+                np.zeros(6, np.float32))]
+    onnx_path = str(tmp_path / "wn.onnx")
+    with open(onnx_path, "wb") as f:
+        f.write(_onnx_bytes(tensors))
+    from safetensors.torch import load_file
+
+    out = import_onnx_voice(onnx_path, str(tmp_path / "wn.safetensors"))
+    got = load_file(out)
+    assert "flow.flows.0.enc.in_layers.0.weight" in got
+    np.testing.assert_allclose(
+        got["flow.flows.0.enc.in_layers.0.weight"].numpy(), expected,
+        rtol=1e-5)
+    assert "flow.flows.0.enc.in_layers.0.weight_v" not in got
