@@ -148,10 +148,16 @@ class SonataSpeechSynthesizer:
 
         def producer():
             try:
-                for sent in phonemes:
+                for n_done, sent in enumerate(phonemes):
+                    # chunk size grows with each processed sentence
+                    # (reference RealtimeSpeechStream, synth/src/
+                    # lib.rs:350-358): the first sentence streams with
+                    # low latency, later ones with bigger (faster)
+                    # chunks; the in-sentence chunker grows further.
+                    cs = min(chunk_size * (n_done + 1), 1024)
                     if self.model.supports_streaming_output:
                         it = self.model.stream_synthesis(
-                            sent, chunk_size, chunk_padding
+                            sent, cs, chunk_padding
                         )
                     else:
                         it = iter([self.model.speak_one_sentence(sent).samples])
