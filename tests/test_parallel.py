@@ -153,3 +153,28 @@ def test_shard_round_robin_uneven():
         all_idx = sorted(sum((shard_round_robin(n, r, world)
                               for r in range(world)), []))
         assert all_idx == list(range(n))
+
+
+@pytest.mark.gpu
+def test_rccl_world1_roundtrip():
+    """RCCL (nccl backend) initializes and runs the bench's collectives
+    (barrier + all_gather) at world_size 1 on a real GPU — the minimal
+    RCCL exercise possible on a single-GPU box; world>1 runs via the
+    driver's scaling bench."""
+    import os
+
+    import torch
+    import torch.distributed as dist
+
+    os.environ.update({"RANK": "0", "WORLD_SIZE": "1", "LOCAL_RANK": "0",
+                       "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29719"})
+    assert torch.cuda.is_available()
+    dist.init_process_group(backend="nccl")
+    try:
+        t = torch.tensor([3.5, 7.0], dtype=torch.float64, device="cuda:0")
+        out = [torch.zeros_like(t)]
+        dist.all_gather(out, t)
+        assert torch.equal(out[0], t)
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
