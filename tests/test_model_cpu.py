@@ -293,3 +293,26 @@ def test_streamed_equals_oneshot_length_and_tail():
         diff = np.abs(streamed - oneshot)
         assert float(np.median(diff)) < 1e-6
         assert (diff < 1e-5).mean() > 0.85
+
+
+def test_rational_quadratic_spline_inverts():
+    """Spline flow math: forward∘inverse = identity inside the tails, and
+    identity outside (linear tails) — the SDP's trickiest component
+    (SURVEY §7 hard part 3)."""
+    import torch
+
+    from sonata_amd.models.vits import rational_quadratic_spline
+
+    torch.manual_seed(5)
+    B, T, bins = 2, 64, 10
+    uw = torch.randn(B, 1, T, bins) * 0.3
+    uh = torch.randn(B, 1, T, bins) * 0.3
+    ud = torch.randn(B, 1, T, bins - 1) * 0.3
+    x = torch.empty(B, 1, T).uniform_(-8, 8)  # inside AND outside tails
+    y, logdet = rational_quadratic_spline(x, uw, uh, ud, inverse=False)
+    x_back = rational_quadratic_spline(y, uw, uh, ud, inverse=True)[0]
+    assert torch.allclose(x_back, x, atol=1e-4)
+    outside = (x < -5) | (x > 5)
+    assert torch.allclose(y[outside], x[outside])  # linear tails
+    # log-determinant is finite and zero outside the tails
+    assert torch.isfinite(logdet).all()
