@@ -49,3 +49,27 @@ def test_batcher_propagates_errors(voice):
     with pytest.raises(RuntimeError, match="boom"):
         b.synthesize("x")
     b.close()
+
+
+@pytest.mark.gpu
+def test_batcher_gpu_concurrent():
+    """Dynamic batcher on the GPU serving path (engine-backed voice):
+    concurrent results equal serial ones."""
+    import tempfile
+
+    import torch
+
+    assert torch.cuda.is_available()
+    with tempfile.TemporaryDirectory() as d:
+        v = load_voice(create_random_voice(d, "bg", quality="x_low"),
+                       device="cuda:0")
+        batcher = DynamicBatcher(v, max_batch=16, max_wait_ms=10)
+        phons = [f"wˈʌn {'tˈuː ' * (1 + i % 5)}." for i in range(24)]
+        serial = {p: v.speak_one_sentence(p).samples for p in set(phons)}
+        futures = [batcher.submit(p) for p in phons]
+        for p, f in zip(phons, futures):
+            got = f.result(timeout=120).samples
+            ref = serial[p]
+            assert len(got) == len(ref)
+            assert float(np.abs(got - ref).max()) < 1e-3
+        batcher.close()
