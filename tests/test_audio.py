@@ -143,3 +143,52 @@ def test_apply_prosody_volume():
     x = np.ones(100, dtype=np.float32) * 0.5
     y = apply_prosody(x, 22050, volume=0.5)
     assert np.allclose(y, 0.25)
+
+
+def test_crossfade_reference_semantics():
+    """Quarter-sine crossfade (reference samples.rs:144-157): output
+    length = len(a) + len(b) - n; constant-ish power through the seam."""
+    import numpy as np
+
+    from sonata_amd.audio.samples import crossfade
+
+    a = np.ones(100, np.float32)
+    b = np.ones(80, np.float32) * -1.0
+    out = crossfade(a, b, 20)
+    assert len(out) == 160
+    # ends untouched
+    np.testing.assert_array_equal(out[:80], a[:80])
+    np.testing.assert_array_equal(out[-60:], b[-60:])
+    # seam is a monotonic blend from +1 toward -1
+    seam = out[80:100]
+    assert seam[0] > 0.8 and seam[-1] < -0.8
+    assert (np.diff(seam) <= 1e-6).all()
+
+
+def test_to_i16_peak_normalization_reference():
+    """Peak normalization only kicks in ABOVE full scale (reference
+    samples.rs:51-75: scale = 32767/absmax when absmax > 1)."""
+    import numpy as np
+
+    from sonata_amd.audio.samples import to_i16
+
+    quiet = np.array([0.5, -0.25], np.float32)
+    out = to_i16(quiet)
+    assert out[0] == int(0.5 * 32767) and out[1] == int(-0.25 * 32767)
+    loud = np.array([2.0, -1.0], np.float32)
+    out = to_i16(loud)
+    assert out[0] == 32767 and out[1] == -16383  # scaled by 32767/2
+
+
+def test_overlap_with_sine_ramp():
+    """overlap_with joins with sine ramps (reference samples.rs:102-118):
+    length = len(a) + len(b) - n and the overlap mixes both."""
+    import numpy as np
+
+    from sonata_amd.audio.samples import overlap_with
+
+    a = np.full(50, 0.8, np.float32)
+    b = np.full(50, 0.2, np.float32)
+    out = overlap_with(a, b, 10)
+    assert len(out) == 90
+    assert np.all(out[40:50] <= 0.8 + 1e-6) and np.all(out[40:50] >= 0.2 - 1e-6)
