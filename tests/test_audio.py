@@ -166,18 +166,22 @@ def test_crossfade_reference_semantics():
 
 
 def test_to_i16_peak_normalization_reference():
-    """Peak normalization only kicks in ABOVE full scale (reference
-    samples.rs:51-75: scale = 32767/absmax when absmax > 1)."""
+    """Reference semantics (samples.rs:51-75): scale = 32767/absmax
+    ALWAYS — quiet audio is amplified to full scale, loud audio is
+    brought back into range."""
     import numpy as np
 
     from sonata_amd.audio.samples import to_i16
 
     quiet = np.array([0.5, -0.25], np.float32)
     out = to_i16(quiet)
-    assert out[0] == int(0.5 * 32767) and out[1] == int(-0.25 * 32767)
+    assert out[0] == 32767 and out[1] == -16383  # scaled by 32767/0.5
     loud = np.array([2.0, -1.0], np.float32)
     out = to_i16(loud)
     assert out[0] == 32767 and out[1] == -16383  # scaled by 32767/2
+    # normalization can be disabled (raw stream chunks)
+    raw = to_i16(quiet, peak_normalize=False)
+    assert raw[0] == int(0.5 * 32767)
 
 
 def test_overlap_with_sine_ramp():
