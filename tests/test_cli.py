@@ -62,3 +62,17 @@ def test_scales_flags_apply(voice_pack, tmp_path):
                    "--device", "cpu"])
     assert rc == 0
     assert out.exists()
+
+
+def test_cli_speaker_flag(tmp_path):
+    from sonata_amd.frontends import cli
+    from sonata_amd.models import create_random_voice
+
+    pack = create_random_voice(str(tmp_path), "spkcli", quality="x_low",
+                               num_speakers=3)
+    inp = tmp_path / "in.txt"
+    inp.write_text("hˈɛloʊ.")
+    out = tmp_path / "s.wav"
+    rc = cli.main([pack, "-f", str(inp), "-o", str(out), "-s", "2",
+                   "--device", "cpu"])
+    assert rc == 0 and out.exists()

@@ -156,3 +156,22 @@ def test_multiprocess_reuseport_serving(voice_pack):
     finally:
         for p in procs:
             p.terminate()
+
+
+def test_options_partial_update(running):
+    """SetSynthesisOptions only touches the fields present (proto3
+    explicit presence via optional)."""
+    client, pack = running
+    vid = client.LoadVoice(MESSAGES["VoicePath"](config_path=pack)).voice_id
+    base = client.SetSynthesisOptions(MESSAGES["VoiceSynthesisOptions"](
+        voice_id=vid,
+        synthesis_options=MESSAGES["SynthesisOptions"](
+            length_scale=1.5, noise_scale=0.4, noise_w=0.6)))
+    assert abs(base.noise_scale - 0.4) < 1e-6
+    # update ONLY length_scale; others must persist
+    got = client.SetSynthesisOptions(MESSAGES["VoiceSynthesisOptions"](
+        voice_id=vid,
+        synthesis_options=MESSAGES["SynthesisOptions"](length_scale=0.9)))
+    assert abs(got.length_scale - 0.9) < 1e-6
+    assert abs(got.noise_scale - 0.4) < 1e-6
+    assert abs(got.noise_w - 0.6) < 1e-6

@@ -316,3 +316,22 @@ def test_rational_quadratic_spline_inverts():
     assert torch.allclose(y[outside], x[outside])  # linear tails
     # log-determinant is finite and zero outside the tails
     assert torch.isfinite(logdet).all()
+
+
+def test_chunk_plan_covers_exactly():
+    """Chunk plans tile [0, F) exactly: trimmed cores are contiguous,
+    non-overlapping, and every frame is produced once."""
+    from sonata_amd.models.chunker import chunk_plan
+
+    for F in [1, 10, 44, 45, 89, 90, 91, 200, 1025, 5000]:
+        for cs, pad in [(45, 3), (20, 2), (100, 0), (1, 5)]:
+            covered = 0
+            for spec in chunk_plan(F, cs, pad):
+                core_lo = spec.mel_start + spec.trim_left_frames
+                core_hi = spec.mel_end - spec.trim_right_frames
+                assert core_lo == covered, (F, cs, pad)
+                assert spec.mel_start >= 0 and spec.mel_end <= F
+                covered = core_hi
+                if spec.is_last:
+                    break
+            assert covered == F, (F, cs, pad)
