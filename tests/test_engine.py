@@ -139,6 +139,13 @@ def test_cli_binary_streaming(pack, tmp_path):
     assert r.returncode == 0, r.stderr
     assert "first chunk in" in r.stderr
     assert out.read_bytes()[:4] == b"RIFF"
+    # the C++ adaptive chunker preserves the one-shot timeline exactly
+    one = tmp_path / "o.wav"
+    r2 = subprocess.run([BIN, pack, "-d", "cpu", "-f", str(inp),
+                         "-o", str(one)], capture_output=True, text=True,
+                        timeout=300)
+    assert r2.returncode == 0, r2.stderr
+    assert len(out.read_bytes()) == len(one.read_bytes())
 
 
 def test_engine_rejects_corrupt_inputs(ext, tmp_path):
