@@ -199,3 +199,21 @@ def test_overlap_with_sine_ramp():
     # bounded overshoot, ends exact
     assert np.all(out[40:50] <= 0.9) and np.all(out[40:50] >= 0.15)
     assert abs(out[40] - 0.8) < 0.05 and abs(out[49] - 0.2) < 0.05
+
+
+def test_wav_roundtrip_header():
+    """Our WAV writer produces a parseable PCM16 file (wave stdlib)."""
+    import io
+    import wave
+
+    import numpy as np
+
+    from sonata_amd.audio.wav import wav_bytes
+
+    samples = np.sin(np.linspace(0, 20, 4410)).astype(np.float32) * 0.7
+    data = wav_bytes(samples, 22050)
+    with wave.open(io.BytesIO(data)) as w:
+        assert w.getframerate() == 22050
+        assert w.getnchannels() == 1
+        assert w.getsampwidth() == 2
+        assert w.getnframes() == 4410

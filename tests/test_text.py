@@ -141,3 +141,24 @@ def test_additional_languages():
         sents = text_to_phonemes(text, voice=lang)
         assert len(sents) == 2, (lang, sents)
         assert all(len(x) > 2 for x in sents)
+
+
+def test_separator_option():
+    """Separator inserted between phonemes (reference
+    espeak-phonemizer/src/lib.rs:102-106)."""
+    from sonata_amd.text.phonemizer import text_to_phonemes
+
+    plain = text_to_phonemes("hi", voice="en")[0]
+    sep = text_to_phonemes("hi", voice="en", separator="|")[0]
+    assert "|" in sep
+    assert sep.replace("|", "") == plain
+
+
+def test_tashkeel_idempotent():
+    """Diacritizing already-diacritized text is stable."""
+    from sonata_amd.text.tashkeel import TashkeelModel
+
+    m = TashkeelModel.default()
+    once = m.diacritize("كتب الولد")
+    twice = m.diacritize(once)
+    assert twice == once
