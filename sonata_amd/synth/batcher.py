@@ -72,12 +72,25 @@ class DynamicBatcher:
             self._flush(batch)
 
     def _flush(self, batch: List[Tuple[str, Future]]) -> None:
-        phonemes = [p for p, _ in batch]
+        # Length-bucketed sub-batches: padded batching costs compute
+        # proportional to the LONGEST utterance, so a 3-word request
+        # coalesced with a 40-word one would pay 10x.  Sort by length
+        # and cut where the next item is >2x the bucket's minimum.
+        batch = sorted(batch, key=lambda it: len(it[0]))
+        start = 0
+        for i in range(1, len(batch) + 1):
+            if i == len(batch) or (
+                    len(batch[i][0]) > 2 * max(len(batch[start][0]), 8)):
+                self._run_bucket(batch[start:i])
+                start = i
+
+    def _run_bucket(self, bucket: List[Tuple[str, Future]]) -> None:
+        phonemes = [p for p, _ in bucket]
         try:
             audios = self.model.speak_batch(phonemes)
-            for (_, f), audio in zip(batch, audios):
+            for (_, f), audio in zip(bucket, audios):
                 f.set_result(audio)
         except BaseException as e:  # noqa: BLE001 - propagate to callers
-            for _, f in batch:
+            for _, f in bucket:
                 if not f.done():
                     f.set_exception(e)

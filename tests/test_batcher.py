@@ -73,3 +73,19 @@ def test_batcher_gpu_concurrent():
             assert len(got) == len(ref)
             assert float(np.abs(got - ref).max()) < 1e-3
         batcher.close()
+
+
+def test_length_bucketing_preserves_results(voice):
+    """Mixed short/long submissions split into length buckets; every
+    result still equals serial synthesis (per-utterance seeding)."""
+    batcher = DynamicBatcher(voice, max_batch=32, max_wait_ms=30)
+    phons = (["wˈʌn."] * 6
+             + ["wˈʌn tˈuː θɹˈiː fˈoːɹ fˈaɪv sˈɪks sˈɛvən ˈeɪt "
+                "nˈaɪn tˈɛn ˈeɪt nˈaɪn tˈɛn." ] * 6)
+    futures = [batcher.submit(p) for p in phons]
+    for p, f in zip(phons, futures):
+        got = f.result(timeout=120).samples
+        ref = voice.speak_one_sentence(p).samples
+        assert len(got) == len(ref)
+        np.testing.assert_allclose(got, ref, atol=1e-5)
+    batcher.close()
