@@ -433,6 +433,33 @@ _PL_RULES = {
 }
 
 
+# --------------------------------------------------------------------------- #
+# Turkish (highly regular orthography)
+# --------------------------------------------------------------------------- #
+_TR_RULES = {
+    "ç": "tʃ", "ş": "ʃ", "ğ": "ː", "ı": "ɯ", "ö": "ø", "ü": "y",
+    "c": "dʒ", "j": "ʒ", "y": "j", "v": "v",
+    "a": "a", "b": "b", "d": "d", "e": "e", "f": "f", "g": "ɡ",
+    "h": "h", "i": "i", "k": "k", "l": "l", "m": "m", "n": "n",
+    "o": "o", "p": "p", "r": "ɾ", "s": "s", "t": "t", "u": "u",
+    "z": "z",
+}
+
+# --------------------------------------------------------------------------- #
+# Czech
+# --------------------------------------------------------------------------- #
+_CS_RULES = {
+    "ch": "x", "č": "tʃ", "š": "ʃ", "ž": "ʒ", "ř": "r̝", "ď": "ɟ",
+    "ť": "c", "ň": "ɲ", "á": "aː", "é": "ɛː", "í": "iː", "ó": "oː",
+    "ú": "uː", "ů": "uː", "ý": "iː", "ě": "jɛ", "c": "ts", "j": "j",
+    "y": "i", "w": "v", "h": "ɦ",
+    "a": "a", "b": "b", "d": "d", "e": "ɛ", "f": "f", "g": "ɡ",
+    "i": "i", "k": "k", "l": "l", "m": "m", "n": "n", "o": "o",
+    "p": "p", "r": "r", "s": "s", "t": "t", "u": "u", "v": "v",
+    "z": "z",
+}
+
+
 _G2P_REGISTRY: Dict[str, RuleG2P] = {}
 
 
@@ -459,6 +486,10 @@ def _get_g2p(voice: str) -> RuleG2P:
         g = RuleG2P(_NL_RULES, letters="a-zA-Z")
     elif base == "pl":
         g = RuleG2P(_PL_RULES, letters="a-zA-Ząćęłńóśźż")
+    elif base == "tr":
+        g = RuleG2P(_TR_RULES, letters="a-zA-Zçğıöşü")
+    elif base == "cs":
+        g = RuleG2P(_CS_RULES, letters="a-zA-Začďéěíňóřšťúůýž")
     elif base == "ar":
         g = RuleG2P(
             _AR_RULES,
@@ -472,7 +503,8 @@ def _get_g2p(voice: str) -> RuleG2P:
 
 
 def available_languages() -> List[str]:
-    return ["en-us", "en", "de", "es", "fr", "it", "pt", "nl", "pl", "ru", "ar"]
+    return ["en-us", "en", "de", "es", "fr", "it", "pt", "nl", "pl",
+            "ru", "tr", "cs", "ar"]
 
 
 _LANG_SWITCH_RE = re.compile(r"\([a-z-]{2,10}\)")

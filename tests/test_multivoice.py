@@ -77,6 +77,7 @@ def test_all_languages_end_to_end(tmp_path):
         "fr": "Bonjour le monde.", "it": "Ciao mondo.",
         "pt": "Olá mundo.", "nl": "Hallo wereld.",
         "pl": "Witaj świecie.", "ru": "Привет мир.",
+        "tr": "Merhaba dünya.", "cs": "Ahoj světe.",
         "ar": "مرحبا بالعالم.",
     }
     for lang in available_languages():
