@@ -151,7 +151,12 @@ class SonataGrpcService:
                           f"failed to load voice: {e}")
         v = _Voice(voice_id, SonataSpeechSynthesizer(voice))
         with self._lock:
-            self._voices.setdefault(voice_id, v)
+            winner = self._voices.setdefault(voice_id, v)
+        if winner is not v:
+            # concurrent LoadVoice for the same config: release the
+            # loser's batcher worker thread and let its weights be GC'd
+            v.batcher.close()
+            v = winner
         log.info("loaded voice %s from %s on %s", voice_id,
                  request.config_path, self.device)
         return self._voice_info(v)

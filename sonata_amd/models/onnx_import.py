@@ -21,6 +21,7 @@ discriminators) are skipped.
 
 from __future__ import annotations
 
+import os
 import re
 import struct
 import sys
@@ -148,21 +149,43 @@ _RULES = [
     (r"^enc_p\.encoder\.ffn_layers\.(\d+)\.conv_2\.", r"enc_p.ffn_layers.\1.conv2."),
     (r"^dec\.cond\.", "dec.cond."),
 ]
-_SKIP = re.compile(r"^(enc_q\.|dp\.post_)")
+# Training-only subtrees present in upstream checkpoints (and possibly in
+# exports): posterior encoder, SDP posterior flows, discriminators.
+_SKIP = re.compile(r"^(enc_q\.|dp\.post_|disc\.|mpd\.|msd\.)")
 
 
 def map_vits_name(name: str) -> str:
     for pat, repl in _RULES:
         name = re.sub(pat, repl, name)
+    # Upstream ResidualCouplingBlock interleaves param-less Flip modules:
+    # flows = [ResidualCouplingLayer, Flip] * 4, so coupling layers sit at
+    # ModuleList indices 0,2,4,6 (vits models.py).  Our flow packs the four
+    # coupling layers densely at 0..3 (flips are implicit in the loop), so
+    # upstream index 2k -> ours k.
+    m = re.match(r"^flow\.flows\.(\d+)\.(.*)$", name)
+    if m:
+        idx = int(m.group(1))
+        if idx % 2 == 0:
+            name = f"flow.flows.{idx // 2}.{m.group(2)}"
     return name
 
 
 def import_onnx_voice(onnx_path: str, out_path: str = None,
-                      strict: bool = False) -> str:
+                      strict: bool = False,
+                      config_path: str = None,
+                      allow_partial: bool = False) -> str:
     """Convert a Piper `voice.onnx` into `<stem>.safetensors`.
 
     Returns the written path.  With strict=True, unmapped initializers
-    raise instead of being reported and skipped."""
+    raise instead of being reported and skipped.
+
+    Coverage check: when the sibling voice config (`<onnx_path>.json`, the
+    reference naming — piper/src/lib.rs:88-110) or an explicit
+    `config_path` is found, the imported state dict is verified to cover
+    the FULL VitsModel parameter tree for that architecture; missing keys
+    raise ModelError (a name-scheme mismatch must fail loudly here, not
+    produce a partially random-init voice that loads loosely later —
+    ADVICE r1).  `allow_partial=True` downgrades that to a warning."""
     import torch
     from safetensors.torch import save_file
 
@@ -170,7 +193,9 @@ def import_onnx_voice(onnx_path: str, out_path: str = None,
     if not inits:
         raise ModelError(f"no initializers found in {onnx_path}")
     # fuse any weight-norm pairs the exporter left in:
-    # weight = g * v / ||v||  (norm over all dims but 0)
+    # weight = g * v / ||v||  (norm over all dims but 0; PyTorch
+    # weight_norm default dim=0, which is what upstream VITS uses for WN
+    # in_layers/res_skip/cond_layer and HiFi-GAN ups convs)
     fused: Dict[str, np.ndarray] = {}
     for name in list(inits):
         if name.endswith(".weight_v"):
@@ -194,6 +219,12 @@ def import_onnx_voice(onnx_path: str, out_path: str = None,
         if not re.match(r"^(enc_p|dp|flow|dec|emb_g)\.|^emb_g$", mapped):
             skipped.append(name)
             continue
+        if not np.issubdtype(arr.dtype, np.floating):
+            # integer graph constants (shape/index tensors) are never
+            # module weights; casting them to f32 would corrupt values
+            # > 2^24 — skip with a note instead
+            skipped.append(name + f" [non-float {arr.dtype}]")
+            continue
         state[mapped] = torch.from_numpy(
             np.ascontiguousarray(arr.astype(np.float32)))
     if skipped:
@@ -201,6 +232,47 @@ def import_onnx_voice(onnx_path: str, out_path: str = None,
         if strict:
             raise ModelError(msg)
         print(f"onnx_import warning: {msg}", file=sys.stderr)
+
+    # ---- coverage verification against the target module tree ---------- #
+    if config_path is None:
+        cand = onnx_path + ".json"  # reference convention: config stem
+        if os.path.exists(cand):    # names the onnx file
+            config_path = cand
+    if config_path is not None:
+        from .config import ModelConfig
+        from .vits import VitsModel
+
+        cfg = ModelConfig.from_json_path(config_path)
+        skeleton = VitsModel(cfg.num_symbols, cfg.architecture,
+                             n_speakers=max(cfg.num_speakers, 1))
+        want = set(skeleton.state_dict().keys())
+        have = set(state.keys())
+        missing = sorted(want - have)
+        extra = sorted(have - want)
+        if missing and not allow_partial:
+            raise ModelError(
+                f"onnx import does not cover the voice architecture: "
+                f"{len(missing)} missing keys (first: {missing[:6]}); "
+                f"{len(skipped)} initializers were skipped "
+                f"(first: {skipped[:6]}).  Pass allow_partial=True to "
+                f"write anyway.")
+        if missing:
+            print(f"onnx_import warning: {len(missing)} missing keys: "
+                  f"{missing[:6]}", file=sys.stderr)
+        if extra:
+            print(f"onnx_import note: dropping {len(extra)} keys not in "
+                  f"the target module tree: {extra[:6]}", file=sys.stderr)
+            for k in extra:
+                del state[k]
+        # shape check: a transposed/mis-mapped tensor must fail here
+        skel_sd = skeleton.state_dict()
+        for k in list(state):
+            if k in skel_sd and tuple(state[k].shape) != tuple(
+                    skel_sd[k].shape):
+                raise ModelError(
+                    f"onnx import shape mismatch for {k}: "
+                    f"{tuple(state[k].shape)} vs expected "
+                    f"{tuple(skel_sd[k].shape)}")
     if out_path is None:
         stem = onnx_path
         if stem.endswith(".onnx"):

@@ -205,10 +205,22 @@ class VitsVoice(SonataModel):
         """Yield waveform chunks: encoder runs once, the HiFi-GAN decoder
         runs per adaptive chunk with overlap-discard + crossfade seams
         (reference SpeechStreamer, piper/src/lib.rs:765-858)."""
+        from ..utils.graphs import enabled as graphs_on
+
         cfg = self.get_synthesis_config()
         ids_l = self._encode_ids(phonemes)
         ids = torch.tensor([ids_l], dtype=torch.long, device=self.device)
         lengths = torch.tensor([len(ids_l)], dtype=torch.long, device=self.device)
+        if (graphs_on() and self.device.type == "cuda"
+                and self.config.num_speakers <= 1
+                and self._engine is None):
+            # hipGraph-replayed encoder phase 1 (measured neutral vs eager
+            # at B=1 — kept opt-in behind SONATA_HIPGRAPH).  Gated to
+            # single-speaker voices: the captured phase-1 closure runs with
+            # g=None, which would drop speaker conditioning otherwise.
+            yield from self._stream_graphed(
+                phonemes, cfg, ids, lengths, chunk_size, chunk_padding)
+            return
         gens = self._generators([phonemes], cfg.speaker_id)
         with stage_timer("encode", self.device):
             z, y_mask, g = self.net.infer_encoder(

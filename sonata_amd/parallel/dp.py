@@ -47,25 +47,33 @@ def broadcast_module(module: torch.nn.Module, src: int = 0) -> None:
         return
     tensors = [t.data for t in module.parameters()]
     tensors += list(module.buffers())
-    bucket: List[torch.Tensor] = []
-    nbytes = 0
+    # Group buckets by dtype and broadcast each in its NATIVE dtype:
+    # casting through float32 would silently corrupt int64 buffers with
+    # values > 2^24 and fp64 tensors.  Iterate dtypes in a
+    # deterministic order so all ranks issue identical collectives.
+    by_dtype: dict = {}
     for t in tensors:
-        bucket.append(t)
-        nbytes += t.numel() * t.element_size()
-        if nbytes >= BUCKET_BYTES:
+        by_dtype.setdefault(t.dtype, []).append(t)
+    for dtype in sorted(by_dtype, key=str):
+        bucket: List[torch.Tensor] = []
+        nbytes = 0
+        for t in by_dtype[dtype]:
+            bucket.append(t)
+            nbytes += t.numel() * t.element_size()
+            if nbytes >= BUCKET_BYTES:
+                _broadcast_bucket(bucket, src)
+                bucket, nbytes = [], 0
+        if bucket:
             _broadcast_bucket(bucket, src)
-            bucket, nbytes = [], 0
-    if bucket:
-        _broadcast_bucket(bucket, src)
 
 
 def _broadcast_bucket(bucket: List[torch.Tensor], src: int) -> None:
-    flat = torch.cat([t.reshape(-1).float() for t in bucket])
+    flat = torch.cat([t.reshape(-1) for t in bucket])
     dist.broadcast(flat, src=src)
     off = 0
     for t in bucket:
         n = t.numel()
-        t.copy_(flat[off:off + n].reshape(t.shape).to(t.dtype))
+        t.copy_(flat[off:off + n].reshape(t.shape))
         off += n
 
 

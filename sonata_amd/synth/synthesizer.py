@@ -149,11 +149,15 @@ class SonataSpeechSynthesizer:
         def producer():
             try:
                 for n_done, sent in enumerate(phonemes):
-                    # chunk size grows with each processed sentence
-                    # (reference RealtimeSpeechStream, synth/src/
-                    # lib.rs:350-358): the first sentence streams with
-                    # low latency, later ones with bigger (faster)
-                    # chunks; the in-sentence chunker grows further.
+                    # Chunk size grows with each processed SENTENCE —
+                    # a deliberate bounded-linear variant of the
+                    # reference's compounding growth (reference
+                    # RealtimeSpeechStream, synth/src/lib.rs:350-358,
+                    # multiplies by cumulative processed CHUNK count,
+                    # unbounded).  Intent is identical: first sentence
+                    # streams with low latency, later ones with bigger
+                    # (faster) chunks; the in-sentence chunker grows
+                    # further.  The 1024 cap matches MAX_CHUNK_SIZE.
                     cs = min(chunk_size * (n_done + 1), 1024)
                     if self.model.supports_streaming_output:
                         it = self.model.stream_synthesis(

@@ -36,9 +36,14 @@ def _tensor_proto(name: str, arr: np.ndarray) -> bytes:
     body = b""
     for d in arr.shape:
         body += _tag(1, 0) + _varint(d)
-    body += _tag(2, 0) + _varint(1)  # FLOAT
-    body += _ld(8, name.encode())
-    body += _ld(9, arr.astype(np.float32).tobytes())
+    if arr.dtype == np.int64:
+        body += _tag(2, 0) + _varint(7)  # INT64
+        body += _ld(8, name.encode())
+        body += _ld(9, arr.tobytes())
+    else:
+        body += _tag(2, 0) + _varint(1)  # FLOAT
+        body += _ld(8, name.encode())
+        body += _ld(9, arr.astype(np.float32).tobytes())
     return body
 
 
@@ -65,6 +70,12 @@ def _to_upstream(name: str) -> str:
 
     name = re.sub(r"(enc_p\.encoder\.ffn_layers\.\d+\.)conv(\d)\.",
                   r"\1conv_\2.", name)
+    # Upstream ResidualCouplingBlock flows = [Layer, Flip]*4: coupling
+    # layers live at ModuleList indices 0,2,4,6 (vits models.py); ours
+    # pack them densely at 0..3.
+    m = re.match(r"^flow\.flows\.(\d+)\.(.*)$", name)
+    if m:
+        name = f"flow.flows.{2 * int(m.group(1))}.{m.group(2)}"
     return name
 
 
@@ -74,9 +85,14 @@ def test_name_mapping_roundtrip():
                  "enc_p.ffn_layers.1.conv1.bias",
                  "dp.flows.3.pre.weight",
                  "flow.flows.0.enc.in_layers.2.weight",
+                 "flow.flows.3.enc.res_skip_layers.1.weight",
                  "dec.resblocks.5.convs1.1.bias",
                  "emb_g.weight"]:
         assert map_vits_name(_to_upstream(ours)) == ours
+    # genuine upstream flow indices 0,2,4,6 land on ours 0..3
+    assert map_vits_name("flow.flows.6.pre.weight") == "flow.flows.3.pre.weight"
+    assert map_vits_name("flow.flows.4.enc.cond_layer.bias") == \
+        "flow.flows.2.enc.cond_layer.bias"
 
 
 def test_import_synthetic_onnx(tmp_path):
@@ -113,6 +129,115 @@ def test_import_synthetic_onnx(tmp_path):
     voice = load_voice(pack, device="cpu")
     audio = voice.speak_one_sentence("tˈɛst.")
     assert len(audio.samples) > 500
+
+
+def _upstream_export_tensors(net, n_speakers=1):
+    """Emulate a genuine Piper export from OUR weights, first-principles:
+    upstream `SynthesizerTrn` names (flow coupling layers at ModuleList
+    indices 0,2,4,6 with param-less Flips between), HiFi-GAN weight norm
+    removed before export (piper export_onnx calls dec.remove_weight_norm),
+    flow WN layers still weight-norm-parametrized (weight_g/weight_v,
+    decomposed g=||w||, v=w so g*v/||v|| == w), training-only tensors
+    (enc_q, dp.post_*) and int64 graph constants included as a real
+    export's initializer list would have them."""
+    tensors = []
+    for k, t in net.state_dict().items():
+        up = _to_upstream(k)
+        w = t.numpy()
+        if (up.startswith("flow.flows.") and up.endswith(".weight")
+                and (".enc.in_layers." in up or ".enc.res_skip_layers." in up
+                     or ".enc.cond_layer." in up)):
+            g = np.sqrt((w.astype(np.float64) ** 2).sum(
+                axis=tuple(range(1, w.ndim)), keepdims=True))
+            tensors.append((up[:-len(".weight")] + ".weight_g",
+                            g.astype(np.float32)))
+            tensors.append((up[:-len(".weight")] + ".weight_v", w))
+        else:
+            tensors.append((up, w))
+    # training-only extras a checkpoint-derived export may carry
+    tensors.append(("enc_q.pre.weight", np.zeros((4, 4, 1), np.float32)))
+    tensors.append(("enc_q.enc.in_layers.0.weight_v",
+                    np.ones((4, 2, 5), np.float32)))
+    tensors.append(("enc_q.enc.in_layers.0.weight_g",
+                    np.ones((4, 1, 1), np.float32)))
+    tensors.append(("dp.post_pre.weight", np.zeros((2, 2, 1), np.float32)))
+    tensors.append(("dp.post_flows.1.pre.weight",
+                    np.zeros((2, 2, 1), np.float32)))
+    # int64 graph constants (shapes/indices); one inside a tree prefix
+    tensors.append(("onnx::Reshape_412", np.array([1, -1, 256], np.int64)))
+    tensors.append(("dec.num_upsamples",
+                    np.array([2 ** 40 + 7], np.int64)))  # would corrupt as f32
+    return tensors
+
+
+@pytest.mark.parametrize("n_speakers", [1, 4])
+def test_full_fidelity_piper_export_roundtrip(tmp_path, n_speakers):
+    """The complete upstream name set: import a
+    full-fidelity synthetic export and assert the resulting pack loads
+    with strict=True coverage AND synthesizes identically to the same
+    net built directly."""
+    from safetensors.torch import load_file
+    from sonata_amd.models import create_random_voice
+    from sonata_amd.models.voice import load_voice, _weights_path_for
+
+    pack = create_random_voice(str(tmp_path), "fidelity", quality="x_low",
+                               num_speakers=n_speakers,
+                               language="en-us", seed=3)
+    state = load_file(_weights_path_for(pack))
+
+    from sonata_amd.models.config import ModelConfig
+    from sonata_amd.models.vits import VitsModel
+    cfg = ModelConfig.from_json_path(pack)
+    net = VitsModel(cfg.num_symbols, cfg.architecture,
+                    n_speakers=max(n_speakers, 1))
+    net.load_state_dict({k: v for k, v in state.items()}, strict=True)
+
+    tensors = _upstream_export_tensors(net, n_speakers)
+    onnx_path = str(tmp_path / "fidelity.onnx")
+    with open(onnx_path, "wb") as f:
+        f.write(_onnx_bytes(tensors))
+    # reference config naming: `<onnx>.json` names the model file
+    import shutil
+    shutil.copy(pack, onnx_path + ".json")
+
+    out = import_onnx_voice(onnx_path, str(tmp_path / "imp.safetensors"))
+    imported = load_file(out)
+    # exact coverage of the target tree
+    assert set(imported.keys()) == set(state.keys())
+    for k in state:
+        np.testing.assert_allclose(imported[k].numpy(), state[k].numpy(),
+                                   rtol=2e-6, atol=2e-6, err_msg=k)
+
+    # synthesis through the imported pack == direct net
+    shutil.copy(out, _weights_path_for(pack))
+    voice = load_voice(pack, device="cpu", engine="python")
+    audio = voice.speak_one_sentence("həlˈoʊ wˈɜːld.")
+    assert len(audio.samples) > 500
+
+
+def test_import_missing_coverage_fails_loudly(tmp_path):
+    """A name-scheme mismatch (e.g. constant-folded generated names) must
+    raise, not write a partially random-init voice pack."""
+    from sonata_amd.models import create_random_voice
+
+    pack = create_random_voice(str(tmp_path), "partial", quality="x_low")
+    from sonata_amd.models.config import ModelConfig
+    nsym = ModelConfig.from_json_path(pack).num_symbols
+    # an export where flow tensors got constant-folded to generated names
+    tensors = [("enc_p.emb.weight", np.zeros((nsym, 96), np.float32)),
+               ("onnx::Conv_123", np.zeros((96, 48, 1), np.float32))]
+    onnx_path = str(tmp_path / "partial.onnx")
+    with open(onnx_path, "wb") as f:
+        f.write(_onnx_bytes(tensors))
+    import shutil
+    shutil.copy(pack, onnx_path + ".json")
+    from sonata_amd.core import ModelError
+    with pytest.raises(ModelError, match="does not cover"):
+        import_onnx_voice(onnx_path, str(tmp_path / "p.safetensors"))
+    # allow_partial downgrades to a warning and writes
+    out = import_onnx_voice(onnx_path, str(tmp_path / "p.safetensors"),
+                            allow_partial=True)
+    assert out.endswith("p.safetensors")
 
 
 def test_weight_norm_pairs_are_fused(tmp_path):
