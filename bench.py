@@ -79,11 +79,15 @@ def main():
         torch.cuda.set_device(device)
 
     dist = None
+    backend = None
     if world > 1:
         import torch.distributed as dist_mod
 
         dist = dist_mod
-        backend = "nccl" if use_gpu else "gloo"
+        # SONATA_BENCH_BACKEND=gloo lets world>1 rehearse on a 1-GPU box
+        # (both ranks compute on cuda:0, collectives go over gloo/host)
+        backend = os.environ.get(
+            "SONATA_BENCH_BACKEND", "nccl" if use_gpu else "gloo")
         dist.init_process_group(backend=backend)
 
     # ---- build the flagship voice (random init, medium preset) ---------- #
@@ -161,8 +165,9 @@ def main():
 
     # max elapsed over ranks; sum audio over ranks
     if dist:
+        comm_dev = device if (use_gpu and backend == "nccl") else "cpu"
         t = torch.tensor([elapsed, total_audio_sec], dtype=torch.float64,
-                         device=device if use_gpu else "cpu")
+                         device=comm_dev)
         gathered = [torch.zeros_like(t) for _ in range(world)]
         dist.all_gather(gathered, t)
         elapsed = max(float(g[0].item()) for g in gathered)

@@ -270,6 +270,17 @@ def create_server(
     return server, bound, service
 
 
+def worker_device(i: int, device: Optional[str], n_gpus: int
+                  ) -> Optional[str]:
+    """GPU pin for multi-process worker i: round-robin across visible
+    GPUs unless the caller pinned an explicit device."""
+    if device is not None and device not in ("cuda", "auto"):
+        return device  # explicit pin wins (e.g. "cuda:3", "cpu")
+    if n_gpus > 0:
+        return f"cuda:{i % n_gpus}"
+    return device if device != "auto" else None
+
+
 def serve(port: Optional[int] = None, device: Optional[str] = None,
           processes: int = 1) -> None:
     logging.basicConfig(
@@ -277,19 +288,27 @@ def serve(port: Optional[int] = None, device: Optional[str] = None,
     if processes > 1:
         # N identical server processes share the port via SO_REUSEPORT;
         # each holds its own voice copies (cheap next to 288 GB HBM) and
-        # its own GIL, so handler throughput scales ~linearly.
+        # its own GIL, so handler throughput scales ~linearly.  On a
+        # multi-GPU node the workers are pinned round-robin across the
+        # visible GPUs (worker i -> cuda:{i % n_gpus}), turning the front
+        # into the DP serving tier of baseline config #4: the kernel
+        # load-balances connections, each GPU serves its own engines.
         import multiprocessing as mp
+
+        import torch
 
         if port is None:
             port = int(os.environ.get("SONATA_GRPC_SERVER_PORT",
                                       DEFAULT_PORT))
+        n_gpus = torch.cuda.device_count()
         ctx = mp.get_context("spawn")
-        procs = [ctx.Process(target=_serve_one, args=(port, device))
-                 for _ in range(processes - 1)]
+        procs = [ctx.Process(target=_serve_one,
+                             args=(port, worker_device(i, device, n_gpus)))
+                 for i in range(1, processes)]
         for p in procs:
             p.start()
         try:
-            _serve_one(port, device)
+            _serve_one(port, worker_device(0, device, n_gpus))
         finally:
             for p in procs:
                 p.terminate()

@@ -82,6 +82,18 @@ def shard_round_robin(n_items: int, rank: int, world: int) -> List[int]:
     return list(range(rank, n_items, world))
 
 
+def comm_device(compute_device: torch.device) -> torch.device:
+    """Device collective tensors must live on for the active backend:
+    nccl/RCCL wants them on the GPU (xGMI path), gloo wants host memory
+    even when the model computes on cuda (GPU-box rehearsal of world>1
+    with gloo collectives)."""
+    if not (dist.is_available() and dist.is_initialized()):
+        return compute_device
+    if dist.get_backend() == "nccl" and compute_device.type == "cuda":
+        return compute_device
+    return torch.device("cpu")
+
+
 def gather_audio_to_rank0(
     pieces: Sequence[np.ndarray],
     indices: Sequence[int],
@@ -99,6 +111,7 @@ def gather_audio_to_rank0(
             out[i] = p
         return [p for p in out if p is not None]
 
+    device = comm_device(device)
     rank = dist.get_rank()
     world = dist.get_world_size()
     # header: [count, (index, length) * count]
