@@ -74,7 +74,10 @@ def main():
         device = torch.device(args.device)
         use_gpu = device.type == "cuda"
     else:
-        device = torch.device(f"cuda:{local_rank}" if use_gpu else "cpu")
+        # modulo device count so world>n_gpus rehearsal runs share GPUs
+        # (no-op on a real N-GPU node where local_rank < count)
+        dev_i = local_rank % max(torch.cuda.device_count(), 1)
+        device = torch.device(f"cuda:{dev_i}" if use_gpu else "cpu")
     if use_gpu:
         torch.cuda.set_device(device)
 
