@@ -53,6 +53,10 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
                                      double out_scale);
 torch::Tensor fused_gate_cl(torch::Tensor x, c10::optional<torch::Tensor> g,
                             long n_channels);
+torch::Tensor attn_relpos_cl(torch::Tensor qkv, torch::Tensor rel_k,
+                             torch::Tensor rel_v,
+                             c10::optional<torch::Tensor> lens, long H,
+                             long window, double scale);
 torch::Tensor depthwise_cl(torch::Tensor x, torch::Tensor w,
                            c10::optional<torch::Tensor> bias, long dil,
                            long pad);
@@ -432,8 +436,6 @@ VitsEngine::text_encoder(torch::Tensor ids, torch::Tensor lengths) const {
              std::sqrt((double)cfg_.hidden);  // [B,T,H]
     auto mask_cl = sequence_mask(lengths, ids.size(1))
                        .to(x.dtype()).transpose(1, 2);  // [B,T,1]
-    auto attn_mask = mask_cl * mask_cl.transpose(1, 2);  // [B,T,T]
-    auto amask4 = attn_mask.unsqueeze(1);
     x = x * mask_cl;
     const long H = cfg_.n_heads, D = cfg_.hidden / cfg_.n_heads;
     auto lens32 = lengths.to(torch::kInt32).contiguous();
@@ -442,20 +444,28 @@ VitsEngine::text_encoder(torch::Tensor ids, torch::Tensor lengths) const {
       std::string am = "enc_p.attn_layers." + li;
       auto xm = x * mask_cl;
       long B = x.size(0), T = x.size(1), C = x.size(2);
-      auto q = lin(xm, am + ".conv_q").view({B, T, H, D}).transpose(1, 2);
-      auto k = lin(xm, am + ".conv_k").view({B, T, H, D}).transpose(1, 2);
-      auto v = lin(xm, am + ".conv_v").view({B, T, H, D}).transpose(1, 2);
       double scale = 1.0 / std::sqrt((double)D);
-      auto scores = torch::matmul(q * scale, k.transpose(-2, -1));
-      auto rel_k = rel_embeddings(p(am + ".emb_rel_k"), T, cfg_.window_size);
-      scores = scores + rel_to_abs(torch::matmul(
-                            q * scale, rel_k.unsqueeze(0).transpose(-2, -1)));
-      scores = scores.masked_fill(amask4 == 0, -1e4);
-      auto pr = torch::softmax(scores, -1);
-      auto outt = torch::matmul(pr, v);
-      auto rel_v = rel_embeddings(p(am + ".emb_rel_v"), T, cfg_.window_size);
-      outt = outt + torch::matmul(abs_to_rel(pr), rel_v.unsqueeze(0));
-      auto y = lin(outt.transpose(1, 2).reshape({B, T, C}), am + ".conv_o");
+      // fused attention kernel (csrc/attention_cl.hip): one QKV GEMM +
+      // one kernel replace the QK^T/softmax/PV matmul chain + the
+      // [B,h,T,2T-1] rel pad/reshape traffic
+      auto wit = cache_.find("qkvw:" + am);
+      if (wit == cache_.end()) {
+        auto wq = p(am + ".conv_q.weight").squeeze(-1);
+        auto wk = p(am + ".conv_k.weight").squeeze(-1);
+        auto wv = p(am + ".conv_v.weight").squeeze(-1);
+        cache_["qkvw:" + am] = torch::cat({wq, wk, wv}).contiguous();
+        cache_["qkvb:" + am] =
+            torch::cat({p(am + ".conv_q.bias"), p(am + ".conv_k.bias"),
+                        p(am + ".conv_v.bias")}).contiguous();
+        cache_["relk:" + am] = p(am + ".emb_rel_k")[0].contiguous();
+        cache_["relv:" + am] = p(am + ".emb_rel_v")[0].contiguous();
+      }
+      auto qkv = torch::linear(xm, cache_["qkvw:" + am],
+                               cache_["qkvb:" + am]);
+      auto outt = attn_relpos_cl(qkv, cache_["relk:" + am],
+                                 cache_["relv:" + am], lens32, H,
+                                 cfg_.window_size, scale);
+      auto y = lin(outt, am + ".conv_o");
       x = torch::layer_norm(x + y, {C}, p("enc_p.norm1." + li + ".gamma"),
                             p("enc_p.norm1." + li + ".beta"), 1e-5);
       std::string f1 = "enc_p.ffn_layers." + li + ".conv1";

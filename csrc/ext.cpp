@@ -59,6 +59,11 @@ torch::Tensor conv1d_direct_cl(torch::Tensor x, torch::Tensor w_perm,
                                double post_slope,
                                c10::optional<torch::Tensor> residual,
                                c10::optional<torch::Tensor> out_lens);
+// attention_cl.hip
+torch::Tensor attn_relpos_cl(torch::Tensor qkv, torch::Tensor rel_k,
+                             torch::Tensor rel_v,
+                             c10::optional<torch::Tensor> lens, long H,
+                             long window, double scale);
 // resblock_cl.hip
 torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
                                      torch::Tensor b1, torch::Tensor w2_perm,
@@ -89,6 +94,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "zero-LDS direct channel-last conv (L1/L2-fed MFMA)");
   m.def("resblock_pair_cl_fused", &resblock_pair_cl_fused,
         "fused HiFi-GAN resblock conv pair (xt stays in LDS)");
+  m.def("attn_relpos_cl", &attn_relpos_cl,
+        "fused relative-position attention (QKT+band+softmax+PV+rel_v)",
+        py::arg("qkv"), py::arg("rel_k"), py::arg("rel_v"), py::arg("lens"),
+        py::arg("heads"), py::arg("window"), py::arg("scale"));
 
   // C++ inference runtime (csrc/engine): the ort-replacement executor.
   py::class_<sonata::VitsEngine>(m, "VitsEngine")

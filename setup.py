@@ -23,6 +23,7 @@ ext = CUDAExtension(
         "csrc/conv1d.hip",
         "csrc/conv1d_cl.hip",
         "csrc/resblock_cl.hip",
+        "csrc/attention_cl.hip",
         "csrc/engine/vits_engine.cpp",
     ],
     extra_compile_args={

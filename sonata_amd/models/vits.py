@@ -229,14 +229,24 @@ class TextEncoder(nn.Module):
         already [B,T,H]; attention/FFN/LN all run on contiguous rows and
         only the (small) encoder outputs transpose to [B,C,T] for the
         duration/prior stages.  Numerics match forward()."""
+        import os
+
+        from ..ops import attn_relpos_cl
+
+        fused_attn = os.environ.get("SONATA_FUSED_ATTN", "1") == "1"
         x = self.emb(ids) * math.sqrt(self.hidden_channels)  # [B,T,H]
         x_mask_cl = sequence_mask(lengths, ids.shape[1]).to(
             x.dtype).transpose(1, 2)  # [B,T,1]
-        attn_mask = (x_mask_cl * x_mask_cl.transpose(1, 2))  # [B,T,T]
+        attn_mask = None
+        if not fused_attn:
+            attn_mask = (x_mask_cl * x_mask_cl.transpose(1, 2))  # [B,T,T]
         x = x * x_mask_cl
         for attn, n1, ffn, n2 in zip(self.attn_layers, self.norm1,
                                      self.ffn_layers, self.norm2):
-            y = attn.forward_cl(x * x_mask_cl, attn_mask.unsqueeze(1))
+            if fused_attn:
+                y = attn_relpos_cl(x * x_mask_cl, attn, lengths)
+            else:
+                y = attn.forward_cl(x * x_mask_cl, attn_mask.unsqueeze(1))
             x = F.layer_norm(x + y, (x.shape[-1],), n1.gamma, n1.beta,
                              n1.eps)
             f = leaky_conv1d_cl(x * x_mask_cl, ffn.conv1.weight,

@@ -69,4 +69,5 @@ from .functional import (  # noqa: F401,E402
     resblock_pair_cl,
     fused_gate_cl,
     depthwise_conv1d_cl,
+    attn_relpos_cl,
 )

@@ -441,3 +441,37 @@ def depthwise_conv1d_cl(
     y = F.conv1d(x.transpose(1, 2), weight, bias, padding=padding,
                  dilation=dilation, groups=x.shape[-1])
     return y.transpose(1, 2)
+
+
+def attn_relpos_cl(
+    x: torch.Tensor,          # [B, T, C] channel-last (masked rows)
+    attn_module,              # RelativeAttention (weights + rel tables)
+    lengths: Optional[torch.Tensor],
+) -> torch.Tensor:
+    """Fused relative-position attention (csrc/attention_cl.hip): ONE
+    QKV projection GEMM + ONE kernel (QK^T + banded rel-k logits +
+    masked online softmax + PV + banded rel-v) + output projection.
+
+    Replaces the 10+ launch matmul/pad/reshape chain of
+    RelativeAttention.forward_cl; GPU-only (callers keep the torch
+    oracle for CPU)."""
+    ext = hip_ext(required=True)
+    m = attn_module
+    B, T, C = x.shape
+    qkv_w = getattr(m, "_qkv_w", None)
+    if qkv_w is None or qkv_w.dtype != x.dtype:
+        # cache the fused [3C, C] projection (inference-only weights)
+        m._qkv_w = torch.cat([
+            m.conv_q.weight.squeeze(-1), m.conv_k.weight.squeeze(-1),
+            m.conv_v.weight.squeeze(-1)]).to(x.dtype).contiguous()
+        m._qkv_b = torch.cat([
+            m.conv_q.bias, m.conv_k.bias, m.conv_v.bias]).to(x.dtype)
+        m._rel_k = m.emb_rel_k[0].to(x.dtype).contiguous()
+        m._rel_v = m.emb_rel_v[0].to(x.dtype).contiguous()
+        qkv_w = m._qkv_w
+    qkv = F.linear(x, qkv_w, m._qkv_b)  # [B, T, 3C]
+    out = ext.attn_relpos_cl(
+        qkv, m._rel_k, m._rel_v, _lens_i32(lengths, x.device),
+        heads=m.n_heads, window=m.window_size,
+        scale=m.head_dim ** -0.5)
+    return F.linear(out, m.conv_o.weight.squeeze(-1), m.conv_o.bias)
