@@ -230,6 +230,196 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
 }
 
 // ========================================================================
+// Persistent small-C pair kernel: BOTH conv weights LDS-resident.
+//
+// The C=32/64 resblock stages run at huge T (F*128 / F*256 samples) with
+// tiny per-window MFMA work (~8-16 MFMA between barriers in the generic
+// kernel): they are BARRIER/overhead-bound, not compute-bound.  Here one
+// block per CU stays resident, stages w1+w2 into LDS ONCE (fits for
+// CP=32 any k, CP=64 k=3: 2*k*CP*(CP+8)*2B <= 56 KB), and loops over
+// (b, t-tile)s with the whole K dim resident per row: each GEMM then
+// runs ALL taps x K-slices with ZERO interior barriers - 3 barriers per
+// tile total (vs ~24 in the generic kernel at k=11).
+// ========================================================================
+#define PERSIST_XTR 256
+#define PERSIST_XTROWS (PERSIST_XTR + 10)  // + (k-1) for k<=11
+
+template <int CP_T, int KMAX, int XR>
+__global__ __launch_bounds__(512) void resblock_pair_persist_kernel(
+    const bf16* __restrict__ x,    // [B][T][C]
+    const bf16* __restrict__ w1,   // [k][CP][CP]
+    const float* __restrict__ b1,
+    const bf16* __restrict__ w2,   // [k][CP][CP]
+    const float* __restrict__ b2,
+    bf16* __restrict__ out,
+    const bf16* __restrict__ accum,
+    const int* __restrict__ out_lens,
+    int C, long T, int k, int dil, float out_scale, int tiles_per_b,
+    int total_tiles) {
+  constexpr int CPP = CP_T + 8;        // row pitch (40 / 72: conflict-free)
+  constexpr int NT = CP_T / 16;
+  constexpr int KS = CP_T / 32;        // K slices per row
+  constexpr int MTMAX = 3;             // ceil(266/16/8) wave row-tiles
+
+  __shared__ bf16 Ws1[KMAX][CP_T][CPP];
+  __shared__ bf16 Ws2[KMAX][CP_T][CPP];
+  __shared__ bf16 Xs[XR][CPP];
+  __shared__ bf16 Xt[PERSIST_XTROWS][CPP];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int kl = lane >> 4;
+  const int il = lane & 15;
+
+  // ---- stage both weight tensors once ------------------------------- //
+  const int wchunks = k * CP_T * (CP_T / 8);
+  for (int u = tid; u < wchunks; u += 512) {
+    const int tap = u / (CP_T * CP_T / 8);
+    const int rem = u % (CP_T * CP_T / 8);
+    const int n = rem / (CP_T / 8), ch = (rem % (CP_T / 8)) * 8;
+    const long src = ((long)tap * CP_T + n) * CP_T + ch;
+    *(ulonglong2*)&Ws1[tap][n][ch] = *(const ulonglong2*)&w1[src];
+    *(ulonglong2*)&Ws2[tap][n][ch] = *(const ulonglong2*)&w2[src];
+  }
+
+  const int h2 = (k - 1) / 2;
+  const int BM = PERSIST_XTR - (k - 1);
+  const int xtrows = PERSIST_XTR + (k - 1);
+  const int pad1 = (k - 1) * dil / 2;
+  const int xrows = xtrows + (k - 1) * dil;
+  const int mtiles1 = (xtrows + 15) / 16;   // GEMM1 row tiles (<= 17)
+  const int mtiles2 = (BM + 15) / 16;       // GEMM2 row tiles (<= 16)
+
+  for (int idx = blockIdx.x; idx < total_tiles; idx += gridDim.x) {
+    const int b = idx / tiles_per_b;
+    const long t0 = (long)(idx % tiles_per_b) * BM;
+    const bf16* xb = x + (long)b * T * C;
+    const long row0 = t0 - h2 - pad1;
+    const long lim = out_lens ? min((long)out_lens[b], T) : T;
+
+    __syncthreads();  // prior tile's GEMM2/Xt reads complete
+    // ---- stage x rows (full K per row, pre-lrelu) ------------------- //
+    const bool interior = (row0 >= 0) && (row0 + xrows <= T);
+    if (interior) {
+      for (int u = tid; u < xrows * (CP_T / 8); u += 512) {
+        const int r = u / (CP_T / 8), ch = (u % (CP_T / 8)) * 8;
+        bf16 v8[8];
+        *(ulonglong2*)v8 = *(const ulonglong2*)&xb[(row0 + r) * C + ch];
+#pragma unroll
+        for (int q = 0; q < 8; ++q) v8[q] = f2bf(lrelu_(bf2f(v8[q]), 0.1f));
+        *(ulonglong2*)&Xs[r][ch] = *(ulonglong2*)v8;
+      }
+    } else {
+      for (int u = tid; u < xrows * (CP_T / 8); u += 512) {
+        const int r = u / (CP_T / 8), ch = (u % (CP_T / 8)) * 8;
+        const long t = row0 + r;
+        bf16 v8[8];
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+          float v = (t >= 0 && t < T) ? bf2f(xb[t * C + ch + q]) : 0.f;
+          v8[q] = f2bf(lrelu_(v, 0.1f));
+        }
+        *(ulonglong2*)&Xs[r][ch] = *(ulonglong2*)v8;
+      }
+    }
+    __syncthreads();
+
+    // ---- GEMM1: xt = lrelu(b1 + conv1(Xs)) — zero interior barriers - //
+    f32x4 acc[MTMAX][NT];
+    for (int mt = wid, mi = 0; mt < mtiles1; mt += 8, ++mi) {
+#pragma unroll
+      for (int nj = 0; nj < NT; ++nj) acc[mi][nj] = {0.f, 0.f, 0.f, 0.f};
+    }
+    for (int tap = 0; tap < k; ++tap) {
+      const int toff = tap * dil;
+#pragma unroll
+      for (int ks = 0; ks < KS; ++ks) {
+        bf16x8 b_frag[NT];
+#pragma unroll
+        for (int nj = 0; nj < NT; ++nj)
+          b_frag[nj] = *(const bf16x8*)&Ws1[tap][nj * 16 + il][ks * 32 + kl * 8];
+        for (int mt = wid, mi = 0; mt < mtiles1; mt += 8, ++mi) {
+          const int m = mt * 16 + il;
+          const bf16x8 a_frag =
+              *(const bf16x8*)&Xs[min(m + toff, XR - 1)][ks * 32 + kl * 8];
+#pragma unroll
+          for (int nj = 0; nj < NT; ++nj)
+            acc[mi][nj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag, b_frag[nj], acc[mi][nj], 0, 0, 0);
+        }
+      }
+    }
+    // epilogue1 -> Xt
+    for (int mt = wid, mi = 0; mt < mtiles1; mt += 8, ++mi) {
+#pragma unroll
+      for (int rg = 0; rg < 4; ++rg) {
+        const int m = mt * 16 + kl * 4 + rg;
+        if (m >= xtrows) continue;
+        const long t = t0 - h2 + m;
+        const bool live = (t >= 0) && (t < lim);
+#pragma unroll
+        for (int nj = 0; nj < NT; ++nj) {
+          const int co = nj * 16 + il;
+          float v = 0.f;
+          if (live && co < C) v = lrelu_(acc[mi][nj][rg] + b1[co], 0.1f);
+          Xt[m][co] = f2bf(v);
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- GEMM2: out = b2 + conv2(Xt) + x (+accum) ------------------- //
+    for (int mt = wid, mi = 0; mt < mtiles2; mt += 8, ++mi) {
+#pragma unroll
+      for (int nj = 0; nj < NT; ++nj) acc[mi][nj] = {0.f, 0.f, 0.f, 0.f};
+    }
+    for (int tap = 0; tap < k; ++tap) {
+#pragma unroll
+      for (int ks = 0; ks < KS; ++ks) {
+        bf16x8 b_frag[NT];
+#pragma unroll
+        for (int nj = 0; nj < NT; ++nj)
+          b_frag[nj] = *(const bf16x8*)&Ws2[tap][nj * 16 + il][ks * 32 + kl * 8];
+        for (int mt = wid, mi = 0; mt < mtiles2; mt += 8, ++mi) {
+          const int m = mt * 16 + il;
+          const bf16x8 a_frag =
+              *(const bf16x8*)&Xt[min(m + tap, PERSIST_XTROWS - 1)]
+                                 [ks * 32 + kl * 8];
+#pragma unroll
+          for (int nj = 0; nj < NT; ++nj)
+            acc[mi][nj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag, b_frag[nj], acc[mi][nj], 0, 0, 0);
+        }
+      }
+    }
+    bf16* ob = out + (long)b * T * C;
+    const bf16* ab = accum ? accum + (long)b * T * C : nullptr;
+    for (int mt = wid, mi = 0; mt < mtiles2; mt += 8, ++mi) {
+#pragma unroll
+      for (int rg = 0; rg < 4; ++rg) {
+        const int m2 = mt * 16 + kl * 4 + rg;
+        const long t = t0 + m2;
+        if (m2 >= BM || t >= T) continue;
+        const bool live = t < lim;
+#pragma unroll
+        for (int nj = 0; nj < NT; ++nj) {
+          const int co = nj * 16 + il;
+          if (co >= C) continue;
+          float v = 0.f;
+          if (live) {
+            v = acc[mi][nj][rg] + b2[co] + bf2f(xb[t * C + co]);
+            if (ab) v += bf2f(ab[t * C + co]);
+            v *= out_scale;
+          }
+          ob[t * C + co] = f2bf(v);
+        }
+      }
+    }
+  }
+}
+
+// ========================================================================
 // host wrapper
 // ========================================================================
 #include <torch/extension.h>
@@ -267,11 +457,53 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
     TORCH_CHECK(accum->sizes() == x.sizes() && accum->is_contiguous());
     accum_p = (const bf16*)accum->data_ptr();
   }
+  hipStream_t st = cur_stream4();
+  // Persistent W-resident variant for the barrier-bound small-C shapes
+  // (C=32 any k, C=64 k=3): both weight tensors live in LDS for the
+  // kernel's whole life, each GEMM runs tap x K with zero interior
+  // barriers.  Only pays off when there are enough tiles to keep all
+  // CUs busy (huge-T decode stages; streaming chunks fall through).
+  static int n_cu = 0;
+  if (n_cu == 0) {
+    hipDeviceProp_t prop;
+    HIP_CHECK(hipGetDeviceProperties(&prop, 0));
+    n_cu = prop.multiProcessorCount;
+  }
+  static const bool persist_on = [] {
+    const char* e = getenv("SONATA_PERSIST_RB");
+    return !(e && e[0] == '0');
+  }();
+  const bool persist_ok =
+      persist_on && ((CP == 32) || (CP == 64 && k == 3));
+  if (persist_ok) {
+    const int BMp = 256 - (int)(k - 1);
+    const int tiles_per_b = (int)((T + BMp - 1) / BMp);
+    const long total = (long)B * tiles_per_b;
+    if (total >= n_cu) {
+      const int grid = (int)std::min<long>(total, n_cu);
+      const int xrows = 256 + (int)(k - 1) * (1 + (int)dil);
+#define LAUNCH_PERSIST(CPT, KM, XR)                                         \
+  hipLaunchKernelGGL((resblock_pair_persist_kernel<CPT, KM, XR>),           \
+                     dim3(grid), dim3(512), 0, st,                          \
+                     (const bf16*)x.data_ptr(),                             \
+                     (const bf16*)w1_perm.data_ptr(),                       \
+                     b1f.data_ptr<float>(),                                 \
+                     (const bf16*)w2_perm.data_ptr(),                       \
+                     b2f.data_ptr<float>(), (bf16*)out.data_ptr(), accum_p, \
+                     lens_p, (int)C, T, (int)k, (int)dil,                   \
+                     (float)out_scale, tiles_per_b, (int)total)
+      if (CP == 64) LAUNCH_PERSIST(64, 3, 272);
+      else if (xrows <= 268) LAUNCH_PERSIST(32, 11, 268);
+      else if (xrows <= 292) LAUNCH_PERSIST(32, 11, 292);
+      else LAUNCH_PERSIST(32, 11, 316);
+#undef LAUNCH_PERSIST
+      return out;
+    }
+  }
   // small-C stages run at huge T with tiny per-block work: use taller
   // 256-row xt tiles there (2x MFMA per block, occupancy still 2-3).
   const long XTRh = (CP <= 32) ? 256 : 128;
   const long BM = XTRh - (k - 1);
-  hipStream_t st = cur_stream4();
 #define LAUNCH_RB(BN, WGN, TC, XR, XTR)                                     \
   hipLaunchKernelGGL((resblock_pair_cl_kernel<BN, WGN, TC, XR, XTR>),       \
                      dim3(ceil_div(T, BM), 1, B), dim3(512), 0, st,         \
