@@ -458,21 +458,25 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
     accum_p = (const bf16*)accum->data_ptr();
   }
   hipStream_t st = cur_stream4();
-  // Persistent W-resident variant for the barrier-bound small-C shapes
-  // (C=32 any k, C=64 k=3): both weight tensors live in LDS for the
-  // kernel's whole life, each GEMM runs tap x K with zero interior
-  // barriers.  Only pays off when there are enough tiles to keep all
-  // CUs busy (huge-T decode stages; streaming chunks fall through).
+  // Persistent W-resident variant (opt-in, SONATA_PERSIST_RB=1):
+  // DOCUMENTED NEGATIVE RESULT.  Both weight tensors LDS-resident, each
+  // GEMM runs tap x K with zero interior barriers (3 barriers/tile vs
+  // ~24) - but measured 2.4-16x SLOWER than the generic kernel
+  // (profiles/r02_rbpair_persist_ab.log): one serial block per CU
+  // (LDS-locked at 99-133 KB) exposes the x staging latency and every
+  // barrier drain that the generic kernel's 2-block-per-CU overlap
+  // hides.  Matches round-1's pipelining findings: cross-block overlap
+  // beats intra-block overhead elimination at these window sizes.
+  // Kept for documentation + as the substrate for a future prefetched
+  // variant; parity-tested (ragged/accum/scale) under the env flag.
+  const char* persist_env = getenv("SONATA_PERSIST_RB");
+  const bool persist_on = persist_env && persist_env[0] == '1';
   static int n_cu = 0;
-  if (n_cu == 0) {
+  if (persist_on && n_cu == 0) {
     hipDeviceProp_t prop;
     HIP_CHECK(hipGetDeviceProperties(&prop, 0));
     n_cu = prop.multiProcessorCount;
   }
-  static const bool persist_on = [] {
-    const char* e = getenv("SONATA_PERSIST_RB");
-    return !(e && e[0] == '0');
-  }();
   const bool persist_ok =
       persist_on && ((CP == 32) || (CP == 64 && k == 3));
   if (persist_ok) {

@@ -407,6 +407,13 @@ def resblock_pair_cl(
         y = y + accum
     if out_scale != 1.0:
         y = y * out_scale
+    if out_lens is not None and accum is not None:
+        # the kernels zero masked rows AFTER the accum add; re-mask so
+        # the oracle matches (accum rows past out_lens must not leak)
+        idx = torch.arange(y.shape[1], device=y.device)
+        y = y.masked_fill(
+            (idx.unsqueeze(0) >= out_lens.to(y.device).unsqueeze(1))
+            .unsqueeze(-1), 0)
     return y
 
 

@@ -565,8 +565,8 @@ def test_text_encoder_fused_attn_matches_eager_cl(dev):
 # persistent W-resident pair kernel (huge-T small-C shapes)
 # --------------------------------------------------------------------------- #
 @pytest.mark.parametrize("C,k,dil,B,T", [
-    (32, 3, 1, 2, 40000), (32, 7, 3, 2, 40000), (32, 11, 5, 2, 40000),
-    (64, 3, 5, 2, 40000), (32, 11, 1, 1, 70001), (64, 3, 1, 4, 20000),
+    (32, 3, 1, 2, 40000), (32, 11, 5, 2, 40000),
+    (64, 3, 5, 2, 40000), (32, 7, 3, 1, 70001),
 ])
 def test_resblock_pair_persistent_parity(dev, C, k, dil, B, T):
     """The persistent kernel engages when total tiles >= n_CU; compare
@@ -585,18 +585,17 @@ def test_resblock_pair_persistent_parity(dev, C, k, dil, B, T):
     lens[-1] = T - 3000
     accum = (torch.randn(B, T, C) / 8).to(torch.bfloat16)
 
-    got = resblock_pair_cl(x.to(dev), w1.to(dev), b1.to(dev), w2.to(dev),
-                           b2.to(dev), dilation=dil, out_lens=lens.to(dev),
-                           accum=accum.to(dev), out_scale=1.0 / 3)
-    os.environ["SONATA_PERSIST_RB"] = "0"
-    try:
-        # NOTE: the env gate is read once per process (static); this
-        # fallback comparison uses the fp32 oracle instead.
-        ref = resblock_pair_cl(x.float(), w1.float(), b1, w2.float(), b2,
-                               dilation=dil, out_lens=lens, accum=accum.float(),
+    os.environ["SONATA_PERSIST_RB"] = "1"  # opt-in (measured slower;
+    try:                                    # kept as documented variant)
+        got = resblock_pair_cl(x.to(dev), w1.to(dev), b1.to(dev),
+                               w2.to(dev), b2.to(dev), dilation=dil,
+                               out_lens=lens.to(dev), accum=accum.to(dev),
                                out_scale=1.0 / 3)
     finally:
         os.environ.pop("SONATA_PERSIST_RB", None)
+    ref = resblock_pair_cl(x.float(), w1.float(), b1, w2.float(), b2,
+                           dilation=dil, out_lens=lens, accum=accum.float(),
+                           out_scale=1.0 / 3)
     # sample-compare (full [B,40000,C] f32 compare is slow): borders +
     # random interior windows + the ragged tail
     idx = [0, 1, 250, 251, 252, 253, 254, 255, 256, 257, 5000, 19999,
