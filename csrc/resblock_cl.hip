@@ -424,6 +424,7 @@ __global__ __launch_bounds__(512) void resblock_pair_persist_kernel(
 // ========================================================================
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <cstring>
 
 static inline hipStream_t cur_stream4() {
   return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
@@ -506,7 +507,12 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
   }
   // small-C stages run at huge T with tiny per-block work: use taller
   // 256-row xt tiles there (2x MFMA per block, occupancy still 2-3).
-  const long XTRh = (CP <= 32) ? 256 : 128;
+  // SONATA_RB_GEOM=xtr64 (experiment): 64-row xt tiles for C=128/256 —
+  // Xt shrinks 2x (C=256's Xt[128][264]=66 KB locks it to ONE block/CU
+  // today), trading W-restage traffic (L2-hot) for occupancy.
+  const char* geom_env = getenv("SONATA_RB_GEOM");
+  const bool geom64 = geom_env && strcmp(geom_env, "xtr64") == 0;
+  const long XTRh = (CP <= 32) ? 256 : ((CP >= 128 && geom64) ? 64 : 128);
   const long BM = XTRh - (k - 1);
 #define LAUNCH_RB(BN, WGN, TC, XR, XTR)                                     \
   hipLaunchKernelGGL((resblock_pair_cl_kernel<BN, WGN, TC, XR, XTR>),       \
@@ -534,13 +540,24 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
     else if (xrows <= 288) LAUNCH_RB(BN, WGN, TC, 288, 256);                \
     else LAUNCH_RB(BN, WGN, TC, 308, 256);                                  \
   } while (0)
-  if (CP == 256) RB_XR128(256, 2, 2);
-  else if (CP == 128) RB_XR128(128, 2, 3);
-  else if (CP == 64) RB_XR128(64, 2, 2);
+#define RB_XR64(BN, WGN, TC)                                                \
+  do {                                                                      \
+    if (xrows <= 80) LAUNCH_RB(BN, WGN, TC, 80, 64);                        \
+    else if (xrows <= 96) LAUNCH_RB(BN, WGN, TC, 96, 64);                   \
+    else LAUNCH_RB(BN, WGN, TC, 124, 64);                                   \
+  } while (0)
+  if (CP == 256) {
+    if (geom64) RB_XR64(256, 2, 1);   // 63 KB -> 2 blocks/CU
+    else RB_XR128(256, 2, 2);         // 120 KB -> 1 block/CU
+  } else if (CP == 128) {
+    if (geom64) RB_XR64(128, 2, 2);   // 47 KB -> 3 blocks/CU
+    else RB_XR128(128, 2, 3);
+  } else if (CP == 64) RB_XR128(64, 2, 2);
   else if (CP == 32) RB_XR256(32, 2, 2);
   else TORCH_CHECK(false, "resblock_cl: unsupported CP ", CP);
 #undef RB_XR128
 #undef RB_XR256
+#undef RB_XR64
 #undef LAUNCH_RB
   return out;
 }

@@ -16,9 +16,10 @@ dev = "cuda:0"
 # (C, k, dil, T)
 shapes = [(32, 3, 1, 256 * F), (32, 3, 3, 256 * F), (32, 7, 3, 256 * F),
           (32, 11, 5, 256 * F), (64, 3, 1, 128 * F), (64, 3, 5, 128 * F),
-          (64, 7, 3, 128 * F), (128, 3, 1, 64 * F), (128, 11, 5, 64 * F)]
-mode = os.environ.get("SONATA_PERSIST_RB", "1")
-print(f"# persistent={mode}")
+          (64, 7, 3, 128 * F), (128, 3, 1, 64 * F), (128, 7, 3, 64 * F),
+          (128, 11, 5, 64 * F), (256, 3, 1, 8 * F), (256, 7, 3, 8 * F),
+          (256, 11, 5, 8 * F)]
+print(f"# persist={os.environ.get('SONATA_PERSIST_RB')} geom={os.environ.get('SONATA_RB_GEOM')}")
 for C, k, dil, T in shapes:
     x = (torch.randn(B // 8, T, C) / 4).to(torch.bfloat16).to(dev)
     # B//8 keeps memory sane; per-shape time scales linearly in B
