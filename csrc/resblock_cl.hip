@@ -510,9 +510,17 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
   // SONATA_RB_GEOM=xtr64 (experiment): 64-row xt tiles for C=128/256 —
   // Xt shrinks 2x (C=256's Xt[128][264]=66 KB locks it to ONE block/CU
   // today), trading W-restage traffic (L2-hot) for occupancy.
+  // Geometry per channel class (A/B in profiles/r02_rb256_ab.log):
+  //   C=256: 64-row xt tiles BY DEFAULT — Xt[128][264]=66 KB locked the
+  //     128-row variant to ONE block/CU (no cross-block overlap);
+  //     xt64 runs 2 blocks/CU: +24% k3, +6% k7, +1% k11 (medians).
+  //   C=128: 128-row stays — xt64 measured 10-38% SLOWER there.
+  // SONATA_RB_GEOM={xtr64,xtr128} overrides for experiments.
   const char* geom_env = getenv("SONATA_RB_GEOM");
-  const bool geom64 = geom_env && strcmp(geom_env, "xtr64") == 0;
-  const long XTRh = (CP <= 32) ? 256 : ((CP >= 128 && geom64) ? 64 : 128);
+  bool geom64 = CP >= 256;
+  if (geom_env && strcmp(geom_env, "xtr64") == 0) geom64 = CP >= 128;
+  if (geom_env && strcmp(geom_env, "xtr128") == 0) geom64 = false;
+  const long XTRh = (CP <= 32) ? 256 : (geom64 ? 64 : 128);
   const long BM = XTRh - (k - 1);
 #define LAUNCH_RB(BN, WGN, TC, XR, XTR)                                     \
   hipLaunchKernelGGL((resblock_pair_cl_kernel<BN, WGN, TC, XR, XTR>),       \
@@ -547,11 +555,11 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
     else LAUNCH_RB(BN, WGN, TC, 124, 64);                                   \
   } while (0)
   if (CP == 256) {
-    if (geom64) RB_XR64(256, 2, 1);   // 63 KB -> 2 blocks/CU
+    if (geom64) RB_XR64(256, 2, 1);   // 63 KB -> 2 blocks/CU (default)
     else RB_XR128(256, 2, 2);         // 120 KB -> 1 block/CU
   } else if (CP == 128) {
-    if (geom64) RB_XR64(128, 2, 2);   // 47 KB -> 3 blocks/CU
-    else RB_XR128(128, 2, 3);
+    if (geom64) RB_XR64(128, 2, 2);   // 47 KB -> 3 blocks (measured slower)
+    else RB_XR128(128, 2, 3);         // default
   } else if (CP == 64) RB_XR128(64, 2, 2);
   else if (CP == 32) RB_XR256(32, 2, 2);
   else TORCH_CHECK(false, "resblock_cl: unsupported CP ", CP);
