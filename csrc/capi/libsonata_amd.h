@@ -96,6 +96,11 @@ void libsonataSetPiperSynthConfig(struct SonataVoice *voice_ptr,
 void libsonataSpeak(struct SonataVoice *voice_ptr, FfiStr text_ptr,
                     struct SynthesisParams params,
                     struct ExternError *out_error);
+/* 1 when the voice runs on the native C++ engine (GIL-free synthesis),
+ * 0 when it fell back to the Python bridge.  sonata_amd extension (not
+ * in the reference API). */
+uint8_t libsonataIsNativeEngine(struct SonataVoice *voice_ptr);
+
 uint8_t libsonataSpeakToFile(struct SonataVoice *voice_ptr, FfiStr text_ptr,
                              struct SynthesisParams params,
                              FfiStr out_filename_ptr,

@@ -87,3 +87,62 @@ def speak_to_file(synth, text: str, path: str, rate: int, volume: int,
                   pitch: int, silence_ms: int) -> None:
     synth.synthesize_to_file(path, text,
                              _output_config(rate, volume, pitch, silence_ms))
+
+
+# --------------------------------------------------------------------- #
+# Native-engine support (csrc/capi/capi.cpp hot path): the C library
+# drives the C++ VitsEngine directly for synthesis; Python is entered
+# only for the TEXT front-end (phonemize + tashkeel + per-utterance
+# seeds) and optional prosody DSP.  A C caller never holds the GIL
+# while the neural graph runs.
+# --------------------------------------------------------------------- #
+class CFrontend:
+    """Lightweight text front-end: voice config + phonemizer (+ tashkeel
+    for Arabic voices).  Loads NO model weights."""
+
+    def __init__(self, config_path: str):
+        from ..models.config import ModelConfig
+
+        self.config = ModelConfig.from_json_path(config_path)
+        self._tashkeel = None
+        if self.config.espeak_voice.startswith("ar"):
+            from ..text.tashkeel import TashkeelModel
+
+            self._tashkeel = TashkeelModel.default(device="cpu")
+
+    def phonemize(self, text: str):
+        from ..text.phonemizer import text_to_phonemes
+
+        if self._tashkeel is not None:
+            text = self._tashkeel.diacritize(text)
+        return text_to_phonemes(text, voice=self.config.espeak_voice)
+
+
+def load_frontend(config_path: str) -> CFrontend:
+    return CFrontend(config_path)
+
+
+def phonemize_with_seeds(frontend: CFrontend, text: str, speaker_id: int):
+    """[(phonemes, seed), ...] — seeds use the SAME derivation as the
+    Python engine path (models/voice.py _utterance_seed) so native and
+    Python synthesis of the same text produce identical audio."""
+    from ..models.voice import _utterance_seed
+
+    sid = speaker_id if frontend.config.num_speakers > 1 else None
+    return [(p, _utterance_seed(p, sid)) for p in frontend.phonemize(text)]
+
+
+def apply_prosody(f32_bytes: bytes, sample_rate: int, rate: int,
+                  volume: int, pitch: int) -> bytes:
+    """rate/volume/pitch percents (0=unset) applied to f32 samples."""
+    import numpy as np
+
+    cfg = _output_config(rate, volume, pitch, 0)
+    s = np.frombuffer(f32_bytes, dtype=np.float32)
+    if cfg is None or cfg.is_noop:
+        return f32_bytes
+    return cfg.apply(s, sample_rate).astype(np.float32).tobytes()
+
+
+def silence_samples(ms: int, sample_rate: int) -> int:
+    return int(sample_rate * ms / 1000.0)
