@@ -51,6 +51,10 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
                                      c10::optional<torch::Tensor> out_lens,
                                      c10::optional<torch::Tensor> accum,
                                      double out_scale);
+torch::Tensor resblock_chain_cl_fused(
+    torch::Tensor x, torch::Tensor w_all, torch::Tensor b_all, long k,
+    long d1, long d2, long d3, c10::optional<torch::Tensor> out_lens,
+    c10::optional<torch::Tensor> accum, double out_scale);
 torch::Tensor fused_gate_cl(torch::Tensor x, c10::optional<torch::Tensor> g,
                             long n_channels);
 torch::Tensor attn_relpos_cl(torch::Tensor qkv, torch::Tensor rel_k,
@@ -927,18 +931,53 @@ torch::Tensor VitsEngine::generator(torch::Tensor x,
       // fused resblock pairs; the MRF sum and /num_kernels fold into
       // the last pair\'s epilogue (same as the Python path)
       torch::Tensor xs;
+      const long Cch = xc.size(2);
+      static const bool chain_on = [] {
+        const char* e = getenv("SONATA_RB_CHAIN");
+        return !(e && e[0] == '0');
+      }();
       for (long j = 0; j < n_kernels; ++j) {
         std::string rb =
             "dec.resblocks." + std::to_string(i * n_kernels + j);
         auto out = xc;
         long kk = cfg_.resblock_ks[j];
         const size_t npair = cfg_.resblock_dil[j].size();
+        const bool last_rb = (j == n_kernels - 1);
+        // whole-resblock chain kernel for the HBM-bound small-C stages
+        long dsum = 3;
+        for (long d : cfg_.resblock_dil[j]) dsum += d;
+        const long S0 = 128 + (kk - 1) * dsum;
+        if (chain_on && npair == 3 &&
+            ((Cch == 32 && S0 <= 248) || (Cch == 64 && S0 <= 152))) {
+          auto key = "chainw:" + rb;
+          auto it = cache_.find(key);
+          if (it == cache_.end()) {
+            std::vector<torch::Tensor> ws, bs;
+            for (size_t di = 0; di < 3; ++di) {
+              std::string c1 = rb + ".convs1." + std::to_string(di);
+              std::string c2 = rb + ".convs2." + std::to_string(di);
+              ws.push_back(perm_conv(c1 + ".weight"));
+              ws.push_back(perm_conv(c2 + ".weight"));
+              bs.push_back(bias_f32(c1 + ".bias"));
+              bs.push_back(bias_f32(c2 + ".bias"));
+            }
+            cache_[key] = torch::stack(ws).contiguous();
+            cache_["chainb:" + rb] = torch::stack(bs).contiguous();
+          }
+          c10::optional<torch::Tensor> accum;
+          if (xs.defined()) accum = xs;
+          xs = resblock_chain_cl_fused(
+              out, cache_[key], cache_["chainb:" + rb], kk,
+              cfg_.resblock_dil[j][0], cfg_.resblock_dil[j][1],
+              cfg_.resblock_dil[j][2], to32(lens), accum,
+              last_rb ? 1.0 / (double)n_kernels : 1.0);
+          continue;
+        }
         for (size_t di = 0; di < npair; ++di) {
           long d = cfg_.resblock_dil[j][di];
           std::string c1 = rb + ".convs1." + std::to_string(di);
           std::string c2 = rb + ".convs2." + std::to_string(di);
           const bool last = (di == npair - 1);
-          const bool last_rb = (j == n_kernels - 1);
           c10::optional<torch::Tensor> accum;
           if (last && xs.defined()) accum = xs;
           out = resblock_pair_cl_fused(

@@ -71,6 +71,10 @@ torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,
                                      c10::optional<torch::Tensor> out_lens,
                                      c10::optional<torch::Tensor> accum,
                                      double out_scale);
+torch::Tensor resblock_chain_cl_fused(
+    torch::Tensor x, torch::Tensor w_all, torch::Tensor b_all, long k,
+    long d1, long d2, long d3, c10::optional<torch::Tensor> out_lens,
+    c10::optional<torch::Tensor> accum, double out_scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "sonata_amd hand-written CDNA4 (gfx950) kernels";
@@ -94,6 +98,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "zero-LDS direct channel-last conv (L1/L2-fed MFMA)");
   m.def("resblock_pair_cl_fused", &resblock_pair_cl_fused,
         "fused HiFi-GAN resblock conv pair (xt stays in LDS)");
+  m.def("resblock_chain_cl_fused", &resblock_chain_cl_fused,
+        "whole resblock (3 pairs) fused: intermediates never touch HBM");
   m.def("attn_relpos_cl", &attn_relpos_cl,
         "fused relative-position attention (QKT+band+softmax+PV+rel_v)",
         py::arg("qkv"), py::arg("rel_k"), py::arg("rel_v"), py::arg("lens"),

@@ -420,6 +420,254 @@ __global__ __launch_bounds__(512) void resblock_pair_persist_kernel(
 }
 
 // ========================================================================
+// Whole-resblock CHAIN kernel: the three conv pairs of one HiFi-GAN
+// ResBlock (dilations d1,d2,d3, same k) fused into ONE kernel.
+//
+// Why: the C=32 / C=64-k3 pair shapes are HBM-bound (1.6-2.7 TB/s
+// measured); running the pairs separately reads + writes the full
+// [B,T,C] tensor 3x (in, out, residual re-read).  Chaining keeps every
+// intermediate in LDS: global traffic drops ~3x per resblock at the
+// cost of halo recompute (stage rows grow by (k-1)*(d_i+1) per pair).
+//
+// Same-origin indexing: all LDS buffers map row r -> time row0+r, so
+// pair p's residual is simply its input buffer at the same row; valid
+// rows shrink toward the final [LEAD, LEAD+BM) window and edge garbage
+// never reaches it (halo arithmetic below).
+//
+// Buffers hold PRE-ACTIVATED (lrelu) values — the next GEMM's A operand
+// needs them — and the residual reconstructs the raw value through the
+// exact-ish inverse (z>=0 ? z : 10z; bf16 storage makes the negative
+// branch differ by <=2^-8 relative, inside kernel parity tolerance).
+// 3 rotating buffers + per-window W staging: 42-76 KB -> 2-3 blocks/CU
+// (the cross-block overlap the persistent experiment showed is vital).
+// ========================================================================
+#define CHAIN_XTR 128
+
+__device__ __forceinline__ float inv_lrelu_(float z) {
+  return z >= 0.0f ? z : z * 10.0f;
+}
+
+template <int CP_T, int SR>
+__global__ __launch_bounds__(512) void resblock_chain_cl_kernel(
+    const bf16* __restrict__ x,       // [B][T][C]
+    const bf16* __restrict__ w_all,   // [6][k][CP][CP] (w1,w2 per pair)
+    const float* __restrict__ b_all,  // [6][C]
+    bf16* __restrict__ out,           // [B][T][C]
+    const bf16* __restrict__ accum,
+    const int* __restrict__ out_lens,
+    int C, long T, int k, int d1, int d2, int d3, float out_scale) {
+  constexpr int CPP = CP_T + 8;
+  constexpr int NT = CP_T / 16;
+  constexpr int KS = CP_T / 32;
+  constexpr int MTMAX = 2;  // ceil(SR/16)/8 waves, SR <= 256
+  constexpr int TC = 2;
+
+  const int km1 = k - 1;
+  const int BM = CHAIN_XTR;
+  const int LEAD = km1 * (d1 + d2 + d3 + 3) / 2;   // left halo total
+  const int S0 = BM + 2 * LEAD;                    // x rows staged
+
+  // XCD swizzle (bijective; see pair kernel above)
+  const int nwg = gridDim.x;
+  int tile = blockIdx.x;
+  if (nwg > 8) {
+    const int xcd = tile % 8, orig8 = tile / 8;
+    const int q = nwg / 8, r = nwg % 8;
+    tile = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig8;
+  }
+  const long t0 = (long)tile * BM;
+  const int b = blockIdx.z;
+  const long row0 = t0 - LEAD;
+
+  __shared__ bf16 Buf[3][SR][CPP];
+  __shared__ bf16 Ws[TC][CP_T][CPP];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int kl = lane >> 4;
+  const int il = lane & 15;
+
+  const bf16* xb = x + (long)b * T * C;
+  const long lim = out_lens ? min((long)out_lens[b], T) : T;
+
+  // ---- stage x rows (pre-activated, full K per row) ------------------ //
+  const bool interior = (row0 >= 0) && (row0 + S0 <= T);
+  if (interior) {
+    for (int u = tid; u < S0 * (CP_T / 8); u += 512) {
+      const int r = u / (CP_T / 8), ch = (u % (CP_T / 8)) * 8;
+      bf16 v8[8];
+      *(ulonglong2*)v8 = *(const ulonglong2*)&xb[(row0 + r) * C + ch];
+#pragma unroll
+      for (int q = 0; q < 8; ++q) v8[q] = f2bf(lrelu_(bf2f(v8[q]), 0.1f));
+      *(ulonglong2*)&Buf[0][r][ch] = *(ulonglong2*)v8;
+    }
+  } else {
+    for (int u = tid; u < S0 * (CP_T / 8); u += 512) {
+      const int r = u / (CP_T / 8), ch = (u % (CP_T / 8)) * 8;
+      const long t = row0 + r;
+      bf16 v8[8];
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
+        float v = (t >= 0 && t < T) ? bf2f(xb[t * C + ch + q]) : 0.f;
+        v8[q] = f2bf(lrelu_(v, 0.1f));
+      }
+      *(ulonglong2*)&Buf[0][r][ch] = *(ulonglong2*)v8;
+    }
+  }
+
+  const int mtiles = (S0 + 15) / 16;
+  int in_buf = 0;  // holds lrelu(pair input)
+  const int dils[3] = {d1, d2, d3};
+
+  for (int pair = 0; pair < 3; ++pair) {
+    const int dil = dils[pair];
+    const int xt_buf = (in_buf + 1) % 3;
+    const int out_buf = (in_buf + 2) % 3;
+    const bf16* w1 = w_all + ((long)(2 * pair) * k) * CP_T * CP_T;
+    const bf16* w2 = w_all + ((long)(2 * pair + 1) * k) * CP_T * CP_T;
+    const float* b1 = b_all + (2 * pair) * C;
+    const float* b2 = b_all + (2 * pair + 1) * C;
+
+    // ---- GEMM_a: xt = lrelu(b1 + conv1_{k,dil}(Buf[in])) ------------- //
+    {
+      f32x4 acc[MTMAX][NT];
+      for (int mt = wid, mi = 0; mt < mtiles; mt += 8, ++mi)
+#pragma unroll
+        for (int nj = 0; nj < NT; ++nj) acc[mi][nj] = {0.f, 0.f, 0.f, 0.f};
+      const int pad = km1 * dil / 2;
+      for (int tap0 = 0; tap0 < k; tap0 += TC) {
+        const int ntc = min(TC, k - tap0);
+        for (int tc = 0; tc < ntc; ++tc) {
+          const long wbase = ((long)(tap0 + tc) * CP_T) * CP_T;
+          for (int u = tid; u < CP_T * (CP_T / 8); u += 512) {
+            const int n = u / (CP_T / 8), ch = (u % (CP_T / 8)) * 8;
+            *(ulonglong2*)&Ws[tc][n][ch] =
+                *(const ulonglong2*)&w1[wbase + (long)n * CP_T + ch];
+          }
+        }
+        __syncthreads();
+        for (int tc = 0; tc < ntc; ++tc) {
+          const int toff = (tap0 + tc) * dil - pad;
+#pragma unroll
+          for (int ks = 0; ks < KS; ++ks) {
+            bf16x8 b_frag[NT];
+#pragma unroll
+            for (int nj = 0; nj < NT; ++nj)
+              b_frag[nj] =
+                  *(const bf16x8*)&Ws[tc][nj * 16 + il][ks * 32 + kl * 8];
+            for (int mt = wid, mi = 0; mt < mtiles; mt += 8, ++mi) {
+              const int m = mt * 16 + il;
+              const int src = min(max(m + toff, 0), S0 - 1);
+              const bf16x8 a_frag =
+                  *(const bf16x8*)&Buf[in_buf][src][ks * 32 + kl * 8];
+#pragma unroll
+              for (int nj = 0; nj < NT; ++nj)
+                acc[mi][nj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a_frag, b_frag[nj], acc[mi][nj], 0, 0, 0);
+            }
+          }
+        }
+        __syncthreads();
+      }
+      // epilogue_a -> Buf[xt] (pre-activated)
+      for (int mt = wid, mi = 0; mt < mtiles; mt += 8, ++mi) {
+#pragma unroll
+        for (int rg = 0; rg < 4; ++rg) {
+          const int m = mt * 16 + kl * 4 + rg;
+          if (m >= S0) continue;
+          const long t = row0 + m;
+          const bool live = (t >= 0) && (t < lim);
+#pragma unroll
+          for (int nj = 0; nj < NT; ++nj) {
+            const int co = nj * 16 + il;
+            float v = 0.f;
+            if (live && co < C) v = lrelu_(acc[mi][nj][rg] + b1[co], 0.1f);
+            Buf[xt_buf][m][co] = f2bf(v);
+          }
+        }
+      }
+      __syncthreads();
+    }
+
+    // ---- GEMM_b: out = b2 + conv2_{k,1}(xt) + raw(Buf[in]) ----------- //
+    {
+      f32x4 acc[MTMAX][NT];
+      for (int mt = wid, mi = 0; mt < mtiles; mt += 8, ++mi)
+#pragma unroll
+        for (int nj = 0; nj < NT; ++nj) acc[mi][nj] = {0.f, 0.f, 0.f, 0.f};
+      const int pad2 = km1 / 2;
+      for (int tap0 = 0; tap0 < k; tap0 += TC) {
+        const int ntc = min(TC, k - tap0);
+        for (int tc = 0; tc < ntc; ++tc) {
+          const long wbase = ((long)(tap0 + tc) * CP_T) * CP_T;
+          for (int u = tid; u < CP_T * (CP_T / 8); u += 512) {
+            const int n = u / (CP_T / 8), ch = (u % (CP_T / 8)) * 8;
+            *(ulonglong2*)&Ws[tc][n][ch] =
+                *(const ulonglong2*)&w2[wbase + (long)n * CP_T + ch];
+          }
+        }
+        __syncthreads();
+        for (int tc = 0; tc < ntc; ++tc) {
+          const int toff = (tap0 + tc) - pad2;
+#pragma unroll
+          for (int ks = 0; ks < KS; ++ks) {
+            bf16x8 b_frag[NT];
+#pragma unroll
+            for (int nj = 0; nj < NT; ++nj)
+              b_frag[nj] =
+                  *(const bf16x8*)&Ws[tc][nj * 16 + il][ks * 32 + kl * 8];
+            for (int mt = wid, mi = 0; mt < mtiles; mt += 8, ++mi) {
+              const int m = mt * 16 + il;
+              const int src = min(max(m + toff, 0), S0 - 1);
+              const bf16x8 a_frag =
+                  *(const bf16x8*)&Buf[xt_buf][src][ks * 32 + kl * 8];
+#pragma unroll
+              for (int nj = 0; nj < NT; ++nj)
+                acc[mi][nj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a_frag, b_frag[nj], acc[mi][nj], 0, 0, 0);
+            }
+          }
+        }
+        __syncthreads();
+      }
+      const bool last_pair = pair == 2;
+      bf16* ob = out + (long)b * T * C;
+      const bf16* ab = accum ? accum + (long)b * T * C : nullptr;
+      for (int mt = wid, mi = 0; mt < mtiles; mt += 8, ++mi) {
+#pragma unroll
+        for (int rg = 0; rg < 4; ++rg) {
+          const int m = mt * 16 + kl * 4 + rg;
+          if (m >= S0) continue;
+          const long t = row0 + m;
+          const bool live = (t >= 0) && (t < lim);
+#pragma unroll
+          for (int nj = 0; nj < NT; ++nj) {
+            const int co = nj * 16 + il;
+            if (co >= C) continue;
+            float v = 0.f;
+            if (live) {
+              const float resid = inv_lrelu_(bf2f(Buf[in_buf][m][co]));
+              v = acc[mi][nj][rg] + b2[co] + resid;
+            }
+            if (!last_pair) {
+              Buf[out_buf][m][co] = f2bf(lrelu_(v, 0.1f));
+            } else if (m >= LEAD && m < LEAD + BM && t >= t0 && t < T) {
+              float o = v;
+              if (live && ab) o += bf2f(ab[t * C + co]);
+              if (live) o *= out_scale;
+              ob[t * C + co] = f2bf(live ? o : 0.f);
+            }
+          }
+        }
+      }
+      __syncthreads();
+    }
+    in_buf = out_buf;
+  }
+}
+
+// ========================================================================
 // host wrapper
 // ========================================================================
 #include <torch/extension.h>
@@ -428,6 +676,52 @@ __global__ __launch_bounds__(512) void resblock_pair_persist_kernel(
 
 static inline hipStream_t cur_stream4() {
   return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+torch::Tensor resblock_chain_cl_fused(
+    torch::Tensor x, torch::Tensor w_all, torch::Tensor b_all, long k,
+    long d1, long d2, long d3, c10::optional<torch::Tensor> out_lens,
+    c10::optional<torch::Tensor> accum, double out_scale) {
+  // x: [B, T, C] bf16; w_all: [6][k][CP][CP] (w1,w2 per pair, permuted);
+  // b_all: [6][C] f32
+  TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "chain: bf16 only");
+  TORCH_CHECK(w_all.dim() == 4 && w_all.size(0) == 6 && w_all.size(1) == k);
+  TORCH_CHECK(w_all.is_contiguous() && b_all.is_contiguous());
+  TORCH_CHECK(b_all.scalar_type() == at::kFloat);
+  const long B = x.size(0), T = x.size(1), C = x.size(2);
+  const long CP = w_all.size(3);
+  TORCH_CHECK(w_all.size(2) == CP && CP == C, "chain: square channels");
+  const long S0 = CHAIN_XTR + (k - 1) * (d1 + d2 + d3 + 3);
+  TORCH_CHECK((CP == 32 && S0 <= 248) || (CP == 64 && S0 <= 152),
+              "chain: unsupported geometry");
+  auto out = torch::empty_like(x);
+  if (out.numel() == 0) return out;
+  const int* lens_p = nullptr;
+  if (out_lens.has_value()) {
+    TORCH_CHECK(out_lens->scalar_type() == at::kInt && out_lens->is_cuda());
+    lens_p = out_lens->data_ptr<int>();
+  }
+  const bf16* accum_p = nullptr;
+  if (accum.has_value()) {
+    TORCH_CHECK(accum->sizes() == x.sizes() && accum->is_contiguous());
+    accum_p = (const bf16*)accum->data_ptr();
+  }
+  hipStream_t st = cur_stream4();
+  dim3 grid(ceil_div(T, CHAIN_XTR), 1, B);
+#define LAUNCH_CHAIN(CPT, SRB)                                              \
+  hipLaunchKernelGGL((resblock_chain_cl_kernel<CPT, SRB>), grid, dim3(512), \
+                     0, st, (const bf16*)x.data_ptr(),                      \
+                     (const bf16*)w_all.data_ptr(),                         \
+                     b_all.data_ptr<float>(), (bf16*)out.data_ptr(),        \
+                     accum_p, lens_p, (int)C, T, (int)k, (int)d1, (int)d2,  \
+                     (int)d3, (float)out_scale)
+  if (CP == 64) LAUNCH_CHAIN(64, 152);
+  else if (S0 <= 152) LAUNCH_CHAIN(32, 152);
+  else if (S0 <= 200) LAUNCH_CHAIN(32, 200);
+  else LAUNCH_CHAIN(32, 248);
+#undef LAUNCH_CHAIN
+  return out;
 }
 
 torch::Tensor resblock_pair_cl_fused(torch::Tensor x, torch::Tensor w1_perm,

@@ -779,7 +779,17 @@ class ResBlock1(nn.Module):
         `accum`/`out_scale` apply to the LAST pair (MRF sum fusion);
         numerically equivalent to forward() then +accum, *out_scale."""
         from ..ops import resblock_pair_cl
+        from ..ops.functional import resblock_chain_cl
 
+        import os
+
+        if os.environ.get("SONATA_RB_CHAIN", "1") == "1":
+            # whole-resblock fusion (3 pairs, one kernel) for the
+            # HBM-bound small-C stages; None = geometry unsupported
+            y = resblock_chain_cl(self, x, out_lens=lengths, accum=accum,
+                                  out_scale=out_scale)
+            if y is not None:
+                return y
         n = len(self.convs1)
         for i, (c1, c2) in enumerate(zip(self.convs1, self.convs2)):
             last = i == n - 1

@@ -482,3 +482,61 @@ def attn_relpos_cl(
         heads=m.n_heads, window=m.window_size,
         scale=m.head_dim ** -0.5)
     return F.linear(out, m.conv_o.weight.squeeze(-1), m.conv_o.bias)
+
+
+def _chain_weights(resblock, dtype, device):
+    """Stack the 6 permuted conv weights + biases of a ResBlock1 for the
+    chain kernel (cached on the module)."""
+    import torch as _t
+
+    cache = getattr(resblock, "_chain_cache", None)
+    if cache is not None and cache[0].dtype == _t.bfloat16:
+        return cache
+    ws = []
+    bs = []
+    for c1, c2 in zip(resblock.convs1, resblock.convs2):
+        ws.append(_conv_weight_mfma(c1.weight))
+        ws.append(_conv_weight_mfma(c2.weight))
+        bs.append(c1.bias.float())
+        bs.append(c2.bias.float())
+    w_all = _t.stack(ws).contiguous().to(device)
+    b_all = _t.stack(bs).contiguous().to(device)
+    resblock._chain_cache = (w_all, b_all)
+    return w_all, b_all
+
+
+def resblock_chain_cl(
+    resblock,                      # models.vits.ResBlock1
+    x: torch.Tensor,               # [B, T, C]
+    out_lens: Optional[torch.Tensor] = None,
+    accum: Optional[torch.Tensor] = None,
+    out_scale: float = 1.0,
+) -> Optional[torch.Tensor]:
+    """Whole-resblock fusion (3 conv pairs in one kernel, intermediates
+    LDS-resident; csrc/resblock_cl.hip chain kernel).  Returns None when
+    the geometry is unsupported (caller falls back to the pair loop).
+
+    Numerics note: pair residuals reconstruct raw values through the
+    exact-ish lrelu inverse from bf16 storage — negative residual values
+    differ from the pair path by <= 2^-8 relative."""
+    if not use_hip(x):
+        return None
+    k = resblock.convs1[0].kernel_size[0]
+    dils = [c.dilation[0] for c in resblock.convs1]
+    C = x.shape[-1]
+    if len(dils) != 3:
+        return None
+    S0 = 128 + (k - 1) * (sum(dils) + 3)
+    if not ((C == 32 and S0 <= 248) or (C == 64 and S0 <= 152)):
+        return None
+    if any(c.dilation[0] != 1 for c in resblock.convs2):
+        return None
+    ext = hip_ext(required=True)
+    if not hasattr(ext, "resblock_chain_cl_fused"):
+        return None
+    w_all, b_all = _chain_weights(resblock, x.dtype, x.device)
+    return ext.resblock_chain_cl_fused(
+        x.contiguous(), w_all, b_all, k, dils[0], dils[1], dils[2],
+        _lens_i32(out_lens, x.device),
+        accum.contiguous() if accum is not None else None,
+        float(out_scale))

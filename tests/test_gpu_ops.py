@@ -610,3 +610,50 @@ def test_resblock_pair_persistent_parity(dev, C, k, dil, B, T):
                 assert got[b, t].abs().max().item() == 0, (b, t)
     # whole-tensor check at reduced precision
     assert _rel_err(got[:, ::37], ref[:, ::37]) < 0.06
+
+
+# --------------------------------------------------------------------------- #
+# whole-resblock chain kernel (3 pairs fused, intermediates LDS-resident)
+# --------------------------------------------------------------------------- #
+@pytest.mark.parametrize("C,k", [(32, 3), (32, 7), (32, 11), (64, 3)])
+def test_resblock_chain_parity(dev, C, k):
+    """Chain kernel vs the pair-loop path vs the fp32 oracle, with
+    ragged lens + MRF accum + out_scale.  Tolerance covers the bf16
+    inverse-lrelu residual reconstruction (<=2^-8 relative on negative
+    values) plus normal bf16 kernel noise."""
+    from sonata_amd.models.vits import ResBlock1
+
+    torch.manual_seed(C * k)
+    B, T = 2, 3000
+    rb = ResBlock1(C, k, [1, 3, 5])
+    x = (torch.randn(B, T, C) / 4).to(torch.bfloat16)
+    lens = torch.tensor([T, T - 700])
+    accum = (torch.randn(B, T, C) / 8).to(torch.bfloat16)
+
+    rb_g = ResBlock1(C, k, [1, 3, 5])
+    rb_g.load_state_dict(rb.state_dict())
+    rb_g = rb_g.to(dev, torch.bfloat16)
+    old = os.environ.get("SONATA_RB_CHAIN")
+    try:
+        os.environ["SONATA_RB_CHAIN"] = "1"
+        got = rb_g.forward_cl(x.to(dev), lens.to(dev),
+                              accum=accum.to(dev), out_scale=1.0 / 3)
+        os.environ["SONATA_RB_CHAIN"] = "0"
+        pair = rb_g.forward_cl(x.to(dev), lens.to(dev),
+                               accum=accum.to(dev), out_scale=1.0 / 3)
+    finally:
+        if old is None:
+            os.environ.pop("SONATA_RB_CHAIN", None)
+        else:
+            os.environ["SONATA_RB_CHAIN"] = old
+    # fp32 oracle via the CPU pair path
+    rb_f = ResBlock1(C, k, [1, 3, 5])
+    rb_f.load_state_dict(rb.state_dict())
+    ref = rb_f.float().forward_cl(x.float(), lens, accum=accum.float(),
+                                  out_scale=1.0 / 3)
+    for b, ln in enumerate(lens.tolist()):
+        e_pair = _rel_err(got[b, :ln], pair[b, :ln])
+        e_ref = _rel_err(got[b, :ln], ref[b, :ln])
+        assert e_pair < 0.04, (C, k, b, e_pair)
+        assert e_ref < 0.05, (C, k, b, e_ref)
+        assert got[b, ln:].abs().max().item() == 0
