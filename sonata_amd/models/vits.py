@@ -783,9 +783,13 @@ class ResBlock1(nn.Module):
 
         import os
 
-        if os.environ.get("SONATA_RB_CHAIN", "1") == "1":
-            # whole-resblock fusion (3 pairs, one kernel) for the
-            # HBM-bound small-C stages; None = geometry unsupported
+        if os.environ.get("SONATA_RB_CHAIN", "0") == "1":
+            # whole-resblock fusion (opt-in): DOCUMENTED NEGATIVE RESULT
+            # (profiles/r02_rbchain_ab.log, 2-3.4x slower) - the 6 GEMM
+            # stages serialize on each block's critical path (~32 W-load/
+            # barrier windows) and 2-3 blocks/CU of overlap cannot hide
+            # it across ~10k blocks; the 3x HBM traffic cut never pays.
+            # Kept for the kernel-structure record + parity tests.
             y = resblock_chain_cl(self, x, out_lens=lengths, accum=accum,
                                   out_scale=out_scale)
             if y is not None:
