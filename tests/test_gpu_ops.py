@@ -656,4 +656,5 @@ def test_resblock_chain_parity(dev, C, k):
         e_ref = _rel_err(got[b, :ln], ref[b, :ln])
         assert e_pair < 0.04, (C, k, b, e_pair)
         assert e_ref < 0.05, (C, k, b, e_ref)
-        assert got[b, ln:].abs().max().item() == 0
+        if ln < T:
+            assert got[b, ln:].abs().max().item() == 0
