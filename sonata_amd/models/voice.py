@@ -221,13 +221,24 @@ class VitsVoice(SonataModel):
             yield from self._stream_graphed(
                 phonemes, cfg, ids, lengths, chunk_size, chunk_padding)
             return
-        gens = self._generators([phonemes], cfg.speaker_id)
         with stage_timer("encode", self.device):
-            z, y_mask, g = self.net.infer_encoder(
-                ids, lengths, sid=self._sid_tensor(1, cfg.speaker_id),
-                noise_scale=cfg.noise_scale, length_scale=cfg.length_scale,
-                noise_w=cfg.noise_w, generators=gens,
-            )
+            if self._engine is not None:
+                # C++ engine encoder: same kernels/seeds, no per-launch
+                # Python overhead on the first-chunk latency path
+                z, y_mask, g = self._engine.infer_encoder(
+                    ids, lengths, self._sid_tensor(1, cfg.speaker_id),
+                    cfg.noise_scale, cfg.length_scale, cfg.noise_w,
+                    [_utterance_seed(phonemes, cfg.speaker_id)])
+                if g is not None and (not g.numel()):
+                    g = None
+            else:
+                gens = self._generators([phonemes], cfg.speaker_id)
+                z, y_mask, g = self.net.infer_encoder(
+                    ids, lengths, sid=self._sid_tensor(1, cfg.speaker_id),
+                    noise_scale=cfg.noise_scale,
+                    length_scale=cfg.length_scale,
+                    noise_w=cfg.noise_w, generators=gens,
+                )
         yield from self._stream_decode(z, y_mask, g, chunk_size,
                                        chunk_padding)
 
