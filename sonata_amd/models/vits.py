@@ -432,26 +432,26 @@ def rational_quadratic_spline(
     min_derivative: float = 1e-3,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Monotonic piecewise rational-quadratic spline with linear tails
-    (Durkan et al., Neural Spline Flows).  Returns (outputs, logabsdet)."""
+    (Durkan et al., Neural Spline Flows).  Returns (outputs, logabsdet).
+
+    Sync-free formulation: instead of boolean-mask compaction (whose
+    `inside.any()` host-syncs on every flow call — 4 per utterance on
+    the latency path — and whose index_put breaks hipGraph capture),
+    the spline is evaluated on ALL elements with tail-clamped inputs
+    and blended with the identity tails via torch.where.  Per-element
+    math is identical to the compacted version."""
     inside = (inputs >= -tail_bound) & (inputs <= tail_bound)
-    outputs = torch.zeros_like(inputs)
-    logabsdet = torch.zeros_like(inputs)
-    # linear tails: identity outside
-    outputs[~inside] = inputs[~inside]
 
     # pad derivatives so boundary derivative == 1
     constant = math.log(math.exp(1.0 - min_derivative) - 1.0)
     unnormalized_derivatives = F.pad(unnormalized_derivatives, (1, 1),
                                      value=constant)
 
-    if not bool(inside.any()):
-        return outputs, logabsdet
-
     num_bins = unnormalized_widths.shape[-1]
-    uw = unnormalized_widths[inside]
-    uh = unnormalized_heights[inside]
-    ud = unnormalized_derivatives[inside]
-    x = inputs[inside]
+    uw = unnormalized_widths
+    uh = unnormalized_heights
+    ud = unnormalized_derivatives
+    x = torch.clamp(inputs, -tail_bound, tail_bound)
 
     widths = torch.softmax(uw, dim=-1)
     widths = min_bin_width + (1 - min_bin_width * num_bins) * widths
@@ -477,6 +477,8 @@ def rational_quadratic_spline(
         bin_idx = _searchsorted(cumheights, x)[..., None]
     else:
         bin_idx = _searchsorted(cumwidths, x)[..., None]
+    # x clamped exactly onto +tail_bound lands one past the last bin
+    bin_idx = bin_idx.clamp(0, num_bins - 1)
 
     in_cumwidths = cumwidths.gather(-1, bin_idx)[..., 0]
     in_widths = widths.gather(-1, bin_idx)[..., 0]
@@ -504,8 +506,8 @@ def rational_quadratic_spline(
             + in_deriv * (1 - root).pow(2)
         )
         lad = torch.log(deriv_num) - 2 * torch.log(denom)
-        outputs[inside] = out
-        logabsdet[inside] = -lad
+        outputs = torch.where(inside, out, inputs)
+        logabsdet = torch.where(inside, -lad, torch.zeros_like(lad))
     else:
         theta = (x - in_cumwidths) / in_widths
         theta_one_minus_theta = theta * (1 - theta)
@@ -519,8 +521,8 @@ def rational_quadratic_spline(
             + in_deriv * (1 - theta).pow(2)
         )
         lad = torch.log(deriv_num) - 2 * torch.log(denom)
-        outputs[inside] = out
-        logabsdet[inside] = lad
+        outputs = torch.where(inside, out, inputs)
+        logabsdet = torch.where(inside, lad, torch.zeros_like(lad))
     return outputs, logabsdet
 
 
