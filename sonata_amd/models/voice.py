@@ -205,19 +205,18 @@ class VitsVoice(SonataModel):
         """Yield waveform chunks: encoder runs once, the HiFi-GAN decoder
         runs per adaptive chunk with overlap-discard + crossfade seams
         (reference SpeechStreamer, piper/src/lib.rs:765-858)."""
-        from ..utils.graphs import enabled as graphs_on
+        from ..utils.graphs import phase1_enabled
 
         cfg = self.get_synthesis_config()
         ids_l = self._encode_ids(phonemes)
         ids = torch.tensor([ids_l], dtype=torch.long, device=self.device)
         lengths = torch.tensor([len(ids_l)], dtype=torch.long, device=self.device)
-        if (graphs_on() and self.device.type == "cuda"
-                and self.config.num_speakers <= 1
-                and self._engine is None):
-            # hipGraph-replayed encoder phase 1 (measured neutral vs eager
-            # at B=1 — kept opt-in behind SONATA_HIPGRAPH).  Gated to
-            # single-speaker voices: the captured phase-1 closure runs with
-            # g=None, which would drop speaker conditioning otherwise.
+        if (phase1_enabled() and self.device.type == "cuda"
+                and self.config.num_speakers <= 1):
+            # hipGraph-replayed encoder phase 1 (default ON: 9.1 -> 6.8 ms
+            # first chunk at B=1; the sync-free spline made capture legal).
+            # Gated to single-speaker voices: the captured phase-1 closure
+            # runs with g=None, which would drop speaker conditioning.
             yield from self._stream_graphed(
                 phonemes, cfg, ids, lengths, chunk_size, chunk_padding)
             return

@@ -21,7 +21,21 @@ import torch
 
 
 def enabled() -> bool:
+    """Decode-chunk graph replay: measured NEUTRAL (chunk decode is
+    GPU-time-bound, not launch-bound) — stays opt-in."""
     return os.environ.get("SONATA_HIPGRAPH", "0") not in ("0", "", "false")
+
+
+def phase1_enabled() -> bool:
+    """Encoder phase-1 graph replay (text encoder + SDP): ON by default
+    since the sync-free spline made capture legal — replaying the ~200
+    small SDP launches as one submission cut B=1 first-chunk latency
+    9.1 -> 6.8 ms (profiles/r02_latency_ladder.json).  SONATA_HIPGRAPH=0
+    disables all graph use."""
+    v = os.environ.get("SONATA_HIPGRAPH", "phase1")
+    if v in ("0", "", "false"):
+        return False
+    return True
 
 
 class DecodeGraphCache:

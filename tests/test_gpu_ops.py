@@ -658,3 +658,30 @@ def test_resblock_chain_parity(dev, C, k):
         assert e_ref < 0.05, (C, k, b, e_ref)
         if ln < T:
             assert got[b, ln:].abs().max().item() == 0
+
+
+def test_stream_graphed_matches_eager(dev, tmp_path):
+    """Default streaming path (hipGraph-replayed encoder phase 1) must
+    produce the same audio as the fully-eager stream."""
+    import numpy as np
+
+    from sonata_amd.models import create_random_voice
+    from sonata_amd.models.voice import load_voice
+
+    pack = create_random_voice(str(tmp_path), "graphed", quality="medium")
+    v = load_voice(pack, device="cuda:0")
+    ph = "ðɪs ɪz ə tˈɛst ʌv ðə ɡɹˈæft pˈæθweɪ."
+    old = os.environ.get("SONATA_HIPGRAPH")
+    try:
+        os.environ["SONATA_HIPGRAPH"] = "phase1"  # default path
+        a = np.concatenate(list(v.stream_synthesis(ph, 45, 3)))
+        os.environ["SONATA_HIPGRAPH"] = "0"       # fully eager
+        b = np.concatenate(list(v.stream_synthesis(ph, 45, 3)))
+    finally:
+        if old is None:
+            os.environ.pop("SONATA_HIPGRAPH", None)
+        else:
+            os.environ["SONATA_HIPGRAPH"] = old
+    assert len(a) == len(b)
+    denom = max(np.abs(b).max(), 1e-6)
+    assert float(np.abs(a - b).max() / denom) < 0.05
