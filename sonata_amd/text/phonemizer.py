@@ -206,6 +206,8 @@ class RuleG2P:
         return starts
 
     def _stress_rules_output(self, word: str, ipa: str) -> str:
+        if "ˈ" in ipa:  # rules already placed stress (e.g. Greek accents)
+            return ipa
         starts = self._vowel_clusters(ipa)
         if not starts:
             return ipa
@@ -630,14 +632,25 @@ def _get_g2p(voice: str) -> RuleG2P:
             stress=False,
         )
     else:
-        raise PhonemizationError(f"unsupported phonemizer language: {voice!r}")
+        # round-2 expansion tables (g2p_tables.py): 13 more languages
+        from .g2p_tables import ALIASES, LETTERS, TABLES
+
+        key2 = ALIASES.get(base, base)
+        if key2 in TABLES:
+            g = RuleG2P(TABLES[key2], letters=LETTERS[key2])
+        else:
+            raise PhonemizationError(
+                f"unsupported phonemizer language: {voice!r}")
     _G2P_REGISTRY[key] = g
     return g
 
 
 def available_languages() -> List[str]:
     return ["en-us", "en", "de", "es", "fr", "it", "pt", "nl", "pl",
-            "ru", "tr", "cs", "ar"]
+            "ru", "tr", "cs", "ar",
+            # round-2 expansion (g2p_tables.py)
+            "sv", "no", "nb", "nn", "da", "fi", "hu", "ro", "el", "bg",
+            "uk", "hr", "sr", "sk", "id", "ms", "sw"]
 
 
 _LANG_SWITCH_RE = re.compile(r"\([a-z-]{2,10}\)")

@@ -216,3 +216,36 @@ def test_tr_accuracy():
     wrong = [(w, g.word_to_ipa(w), want) for w, want in TR_CASES
              if g.word_to_ipa(w) != want]
     assert len(wrong) <= 1, wrong
+
+
+# ---- round-2 language expansion (g2p_tables.py) ------------------------ #
+def test_expanded_languages_smoke():
+    """Every expansion language phonemizes common words into IPA with
+    exactly one primary stress per content word and its load-bearing
+    digraphs mapped."""
+    from sonata_amd.text.phonemizer import text_to_phonemes
+
+    checks = [
+        ("sv", "stjärna", "ɧ"), ("no", "skjorte", "ʃ"),
+        ("da", "søster", "ø"), ("fi", "kiitos", "iː"),
+        ("hu", "gyerek", "ɟ"), ("hu", "szép", "s"),
+        ("ro", "ceva", "tʃ"), ("el", "ευχαριστώ", "vx"),  # context-free ev (real: ef before voiceless)
+        ("bg", "благодаря", "ɡ"), ("uk", "дякую", "dj"),
+        ("hr", "džep", "dʒ"), ("sk", "ďakujem", "ɟ"),
+        ("id", "nyamuk", "ɲ"), ("sw", "ng'ombe", "ŋ"),
+        ("sr", "ljudi", "ʎ"), ("ms", "pagi", "ɡ"),
+    ]
+    for lang, word, frag in checks:
+        out = text_to_phonemes(word, voice=lang)[0]
+        assert frag in out.replace("ˈ", ""), (lang, word, out, frag)
+        assert out.count("ˈ") == 1, (lang, word, out)
+
+
+def test_greek_accent_is_stress():
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    g = _get_g2p("el")
+    assert g.word_to_ipa("καλημέρα") == "kalimˈera"
+    # unaccented word falls back to first-syllable stress
+    ipa = g.word_to_ipa("και")
+    assert ipa.count("ˈ") == 1
