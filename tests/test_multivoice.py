@@ -79,6 +79,14 @@ def test_all_languages_end_to_end(tmp_path):
         "pl": "Witaj świecie.", "ru": "Привет мир.",
         "tr": "Merhaba dünya.", "cs": "Ahoj světe.",
         "ar": "مرحبا بالعالم.",
+        # round-2 expansion languages (g2p_tables.py)
+        "sv": "Hej världen.", "no": "Hei verden.", "nb": "Hei verden.",
+        "nn": "Hei verda.", "da": "Hej verden.", "fi": "Hei maailma.",
+        "hu": "Helló világ.", "ro": "Salut lume.", "el": "Γεια σου κόσμε.",
+        "bg": "Здравей свят.", "uk": "Привіт світе.",
+        "hr": "Pozdrav svijete.", "sr": "Pozdrav svete.",
+        "sk": "Ahoj svet.", "id": "Halo dunia.", "ms": "Halo dunia.",
+        "sw": "Habari dunia.",
     }
     for lang in available_languages():
         pack = create_random_voice(str(tmp_path), f"lang_{lang}",
