@@ -298,17 +298,19 @@ class VitsVoice(SonataModel):
         # length exactly (streamed output == one-shot length).
         tail: Optional[np.ndarray] = None
         prev_ext = 0
-        for spec in chunk_plan(num_frames, chunk_size, chunk_padding):
+
+        def decode_one(spec):
             z_c = z[:, :, spec.mel_start : spec.mel_end].contiguous()
             m_c = y_mask[:, :, spec.mel_start : spec.mel_end].contiguous()
             with stage_timer("decode_chunk", self.device):
                 if graph_cache is not None:
-                    audio = graph_cache(z_c, m_c)
-                elif self._engine is not None:
-                    audio = self._engine.decode(z_c, m_c, g, None)
-                else:
-                    audio = self.net.decode(z_c, m_c, g)
-            wav = audio[0, 0].float().cpu().numpy()
+                    return graph_cache(z_c, m_c)
+                if self._engine is not None:
+                    return self._engine.decode(z_c, m_c, g, None)
+                return self.net.decode(z_c, m_c, g)
+
+        def trim_and_emit(spec, wav):
+            nonlocal tail, prev_ext
             lo = spec.trim_left_frames * hop
             hi = len(wav) - spec.trim_right_frames * hop
             ext = 0 if spec.is_last else min(
@@ -318,11 +320,57 @@ class VitsVoice(SonataModel):
             if tail is not None:
                 cur = crossfade(tail, cur, prev_ext)
             cut = len(cur) - ext
-            yield cur[:cut].astype(np.float32)
-            tail = cur[cut:] if ext else None
+            tail_next = cur[cut:] if ext else None
+            out = cur[:cut].astype(np.float32)
+            tail = tail_next
             prev_ext = ext
+            return out
+
+        if self.device.type != "cuda":
+            for spec in chunk_plan(num_frames, chunk_size, chunk_padding):
+                wav = decode_one(spec)[0, 0].float().cpu().numpy()
+                yield trim_and_emit(spec, wav)
+                if spec.is_last:
+                    return
+            return
+
+        # GPU: 1-deep pipeline — chunk i+1's decode is enqueued before
+        # chunk i's host copy is consumed, so GPU decode overlaps the
+        # D2H transfer + Python trim/crossfade/emit of the previous
+        # chunk.  Double-buffered pinned staging; stream order makes the
+        # graph-replay output safe (the copy is enqueued before the next
+        # replay can overwrite its capture buffer).
+        if not hasattr(self, "_pin_bufs"):
+            max_samples = (1024 + 2 * 16) * hop
+            self._pin_bufs = [
+                torch.empty(max_samples, dtype=torch.float32,
+                            pin_memory=True) for _ in range(2)]
+            self._pin_events = [torch.cuda.Event(), torch.cuda.Event()]
+        pending = None  # (spec, buf_idx, n_samples)
+        which = 0
+        for spec in chunk_plan(num_frames, chunk_size, chunk_padding):
+            audio = decode_one(spec)
+            n = audio.shape[-1]
+            buf = self._pin_bufs[which]
+            if n > buf.shape[0]:  # defensive: unexpected chunk size
+                buf = torch.empty(n, dtype=torch.float32, pin_memory=True)
+                self._pin_bufs[which] = buf
+            buf[:n].copy_(audio[0, 0].float(), non_blocking=True)
+            self._pin_events[which].record()
+            if pending is not None:
+                pspec, pwhich, pn = pending
+                self._pin_events[pwhich].synchronize()
+                wav = self._pin_bufs[pwhich][:pn].numpy().copy()
+                yield trim_and_emit(pspec, wav)
+            pending = (spec, which, n)
+            which ^= 1
             if spec.is_last:
-                return
+                break
+        if pending is not None:
+            pspec, pwhich, pn = pending
+            self._pin_events[pwhich].synchronize()
+            wav = self._pin_bufs[pwhich][:pn].numpy().copy()
+            yield trim_and_emit(pspec, wav)
 
 
 # --------------------------------------------------------------------------- #
