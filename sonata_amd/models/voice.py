@@ -348,6 +348,7 @@ class VitsVoice(SonataModel):
             self._pin_events = [torch.cuda.Event(), torch.cuda.Event()]
         pending = None  # (spec, buf_idx, n_samples)
         which = 0
+        first = True
         for spec in chunk_plan(num_frames, chunk_size, chunk_padding):
             audio = decode_one(spec)
             n = audio.shape[-1]
@@ -357,12 +358,20 @@ class VitsVoice(SonataModel):
                 self._pin_bufs[which] = buf
             buf[:n].copy_(audio[0, 0].float(), non_blocking=True)
             self._pin_events[which].record()
-            if pending is not None:
-                pspec, pwhich, pn = pending
-                self._pin_events[pwhich].synchronize()
-                wav = self._pin_bufs[pwhich][:pn].numpy().copy()
-                yield trim_and_emit(pspec, wav)
-            pending = (spec, which, n)
+            if first:
+                # latency priority: emit chunk 0 before enqueuing chunk
+                # 1's ~150 eager launches; pipeline from chunk 1 on
+                self._pin_events[which].synchronize()
+                wav = self._pin_bufs[which][:n].numpy().copy()
+                yield trim_and_emit(spec, wav)
+                first = False
+            else:
+                if pending is not None:
+                    pspec, pwhich, pn = pending
+                    self._pin_events[pwhich].synchronize()
+                    wav = self._pin_bufs[pwhich][:pn].numpy().copy()
+                    yield trim_and_emit(pspec, wav)
+                pending = (spec, which, n)
             which ^= 1
             if spec.is_last:
                 break
