@@ -503,3 +503,72 @@ torch::Tensor depthwise_cl(torch::Tensor x, torch::Tensor w,
   });
   return out;
 }
+
+// ------------------------------------------------------------------------- //
+// seeded_noise: per-utterance deterministic standard-normal noise in ONE
+// launch.  Replaces the per-row torch::randn loops (B x 2 generator
+// launches per batch + per-row .item() syncs) that showed up as 1.4k
+// distribution kernels per bench step.  Value at (b, c, t) depends ONLY
+// on (seeds[b], c, t): noise is independent of batch composition, rank
+// AND padding by construction (SURVEY.md section 7 hard part 7).
+// RNG: splitmix64 counter hash -> Box-Muller.
+// ------------------------------------------------------------------------- //
+__device__ __forceinline__ unsigned long long sm64_(unsigned long long z) {
+  z += 0x9E3779B97F4A7C15ULL;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
+  return z ^ (z >> 31);
+}
+
+template <typename T>
+__global__ void seeded_noise_kernel(T* __restrict__ out,
+                                    const long* __restrict__ seeds,
+                                    const int* __restrict__ lens,
+                                    long B, long C, long Tm) {
+  const long total = B * C * Tm;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const long b = i / (C * Tm);
+    const long rem = i - b * C * Tm;
+    const long c = rem / Tm, t = rem - (rem / Tm) * Tm;
+    float v = 0.f;
+    if (t < (long)lens[b]) {
+      const unsigned long long ctr =
+          ((unsigned long long)(c + 1) << 40) ^ (unsigned long long)t;
+      const unsigned long long r1 =
+          sm64_((unsigned long long)seeds[b] ^ sm64_(ctr));
+      const unsigned long long r2 = sm64_(r1);
+      const float u1 = (float)((r1 >> 11) + 1) * 1.1102230246251565e-16f;
+      const float u2 = (float)(r2 >> 11) * 1.1102230246251565e-16f;
+      v = sqrtf(-2.0f * __logf(u1)) * __cosf(6.28318530717958f * u2);
+    }
+    st_f(out + i, v);
+  }
+}
+
+torch::Tensor seeded_noise(long B, long C, long T_max, torch::Tensor lens,
+                           torch::Tensor seeds, torch::ScalarType dtype) {
+  TORCH_CHECK(lens.is_cuda() && lens.scalar_type() == at::kInt);
+  TORCH_CHECK(seeds.is_cuda() && seeds.scalar_type() == at::kLong);
+  TORCH_CHECK(lens.numel() == B && seeds.numel() == B);
+  auto out = torch::empty(
+      {B, C, T_max},
+      torch::TensorOptions().device(lens.device()).dtype(dtype));
+  if (out.numel() == 0) return out;
+  const long total = B * C * T_max;
+  const int blocks = (int)std::min<long>((total + 255) / 256, 4096);
+  if (dtype == at::kBFloat16) {
+    hipLaunchKernelGGL(seeded_noise_kernel<bf16>, dim3(blocks), dim3(256), 0,
+                       cur_stream(), (bf16*)out.data_ptr(),
+                       seeds.data_ptr<long>(), lens.data_ptr<int>(), B, C,
+                       T_max);
+  } else {
+    TORCH_CHECK(dtype == at::kFloat, "seeded_noise: f32/bf16 only");
+    hipLaunchKernelGGL(seeded_noise_kernel<float>, dim3(blocks), dim3(256),
+                       0, cur_stream(), (float*)out.data_ptr(),
+                       seeds.data_ptr<long>(), lens.data_ptr<int>(), B, C,
+                       T_max);
+  }
+  return out;
+}

@@ -27,6 +27,8 @@ torch::Tensor prior_sample(torch::Tensor m, torch::Tensor logs,
                            double noise_scale);
 torch::Tensor expand_states(torch::Tensor stats, torch::Tensor durs,
                             long F_max);
+torch::Tensor seeded_noise(long B, long C, long T_max, torch::Tensor lens,
+                           torch::Tensor seeds, torch::ScalarType dtype);
 torch::Tensor conv1d_fused(torch::Tensor x, torch::Tensor w_perm,
                            c10::optional<torch::Tensor> bias, long Cout,
                            long k, long stride, long padding, long dilation,
@@ -1047,9 +1049,22 @@ torch::Tensor VitsEngine::masked_noise(long B, long C, long T_max,
                                        torch::Tensor lengths,
                                        std::vector<torch::Generator>& gens)
     const {
-  // Per-utterance generators, on-device when serving a GPU (matches the
-  // Python path: voice.py _generators uses torch.Generator(device)); the
-  // CPU path generates fp32 host-side (the numerics oracle).
+  if (gpu()) {
+    // ONE launch: counter-based normal noise keyed by (seed, c, t) —
+    // replaces B per-row torch::randn launches + B .item() host syncs
+    // per noise tensor (csrc/elementwise.hip seeded_noise_kernel; the
+    // Python GPU path uses the same kernel, so engine == python holds)
+    std::vector<long> sv;
+    for (auto& g : gens) sv.push_back((long)g.current_seed());
+    auto seeds = torch::from_blob(sv.data(), {(long)sv.size()},
+                                  torch::kLong).clone().to(device_);
+    auto lens32 = lengths.to(device_).to(torch::kInt);
+    return seeded_noise(B, C, T_max, lens32.contiguous(),
+                        seeds.contiguous(),
+                        dtype_ == torch::kBFloat16 ? at::kBFloat16
+                                                   : at::kFloat);
+  }
+  // CPU: per-utterance torch generators (the fp32 numerics oracle)
   auto opts = torch::TensorOptions().device(device_).dtype(dtype_);
   auto out = torch::zeros({B, C, T_max}, opts);
   auto lens_cpu = lengths.to(torch::kCPU);

@@ -18,6 +18,8 @@ torch::Tensor prior_sample(torch::Tensor m, torch::Tensor logs,
 torch::Tensor expand_states(torch::Tensor stats, torch::Tensor durs,
                             long F_max);
 torch::Tensor mask_tail_(torch::Tensor x, torch::Tensor lens);
+torch::Tensor seeded_noise(long B, long C, long T_max, torch::Tensor lens,
+                           torch::Tensor seeds, torch::ScalarType dtype);
 torch::Tensor depthwise_cl(torch::Tensor x, torch::Tensor w,
                            c10::optional<torch::Tensor> bias, long dil,
                            long pad);
@@ -84,6 +86,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("prior_sample", &prior_sample, "z=(m+eps*exp(logs)*ns)*mask");
   m.def("expand_states", &expand_states, "duration length-regulator gather");
   m.def("mask_tail_", &mask_tail_, "in-place zero of x[b,:,lens[b]:]");
+  m.def("seeded_noise",
+        [](long B, long C, long T, torch::Tensor lens, torch::Tensor seeds,
+           const std::string& dt) {
+          return seeded_noise(B, C, T, lens, seeds,
+                              dt == "bf16" ? at::kBFloat16 : at::kFloat);
+        },
+        "per-utterance counter-based normal noise, one launch");
   m.def("depthwise_cl", &depthwise_cl, "channel-last depthwise conv1d");
   m.def("conv1d_fused", &conv1d_fused, "MFMA conv1d with fused activations");
   m.def("convtranspose1d_fused", &convtranspose1d_fused,

@@ -14,6 +14,7 @@ Inference-only: no dropout, no posterior encoder, no discriminators.
 from __future__ import annotations
 
 import math
+import os
 from typing import List, Optional, Tuple
 
 import torch
@@ -924,6 +925,25 @@ def masked_noise_rows(
 
     `lengths` may be a tensor or a plain int list (pre-fetched to avoid
     per-row device syncs on the latency path)."""
+    from ..ops import hip_ext, use_hip
+
+    if (generators is not None and str(device).startswith("cuda")
+            and os.environ.get("SONATA_FORCE_TORCH", "0") != "1"):
+        # ONE launch (csrc/elementwise.hip seeded_noise): counter-based
+        # normal noise keyed by (seed, c, t) — replaces B per-row randn
+        # launches; the C++ engine uses the same kernel, so engine ==
+        # python parity holds on GPU
+        ext = hip_ext(required=True)
+        seeds = torch.tensor([g.initial_seed() for g in generators],
+                             dtype=torch.long, device=device)
+        if torch.is_tensor(lengths):
+            lens32 = lengths.to(device=device, dtype=torch.int32)
+        else:
+            lens32 = torch.tensor(lengths, dtype=torch.int32, device=device)
+        return ext.seeded_noise(
+            batch, channels, max_len, lens32.contiguous(),
+            seeds.contiguous(),
+            "bf16" if dtype == torch.bfloat16 else "f32")
     if torch.is_tensor(lengths):
         lengths = lengths.tolist()
     out = torch.zeros((batch, channels, max_len), device=device, dtype=dtype)
