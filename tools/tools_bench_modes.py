@@ -25,8 +25,16 @@ TEXT = ("Hello there everyone. This is a longer paragraph of text. "
         "It contains several sentences of varying length. "
         "The streaming modes chunk it differently. Goodbye now.")
 
-# warmup
+# warmup: parallel AND one full realtime pass (graph captures are a
+# one-time first-request cost, reported separately as cold_first below)
 list(synth.synthesize_parallel(TEXT))
+t_cold = time.perf_counter()
+cold_first = None
+for _c in synth.synthesize_streamed(TEXT, chunk_size=45, chunk_padding=3):
+    if cold_first is None:
+        cold_first = (time.perf_counter() - t_cold) * 1000
+print(json.dumps({"mode": "realtime-cold-first-request",
+                  "first_chunk_ms": round(cold_first or 0, 2)}))
 
 N = 5
 for mode in ["lazy", "parallel", "realtime"]:
