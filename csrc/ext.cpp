@@ -130,7 +130,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
               torch::Tensor lengths, c10::optional<torch::Tensor> sid,
               double ns, double ls, double nw,
               std::vector<int64_t> seeds) {
-             auto r = e.infer(ids, lengths, sid, ns, ls, nw, seeds);
+             std::pair<torch::Tensor, torch::Tensor> r;
+             {
+               // release the GIL for the whole graph: concurrent server
+               // handler threads serialize WAVs while the GPU runs
+               py::gil_scoped_release rel;
+               r = e.infer(ids, lengths, sid, ns, ls, nw, seeds);
+             }
              return py::make_tuple(r.first, r.second);
            },
            py::arg("ids"), py::arg("lengths"),
@@ -142,7 +148,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
               torch::Tensor lengths, c10::optional<torch::Tensor> sid,
               double ns, double ls, double nw,
               std::vector<int64_t> seeds) {
-             auto r = e.infer_encoder(ids, lengths, sid, ns, ls, nw, seeds);
+             std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> r;
+             {
+               py::gil_scoped_release rel;
+               r = e.infer_encoder(ids, lengths, sid, ns, ls, nw, seeds);
+             }
              return py::make_tuple(std::get<0>(r), std::get<1>(r),
                                    std::get<2>(r));
            },
@@ -154,6 +164,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            [](sonata::VitsEngine& e, torch::Tensor z, torch::Tensor y_mask,
               c10::optional<torch::Tensor> g,
               c10::optional<torch::Tensor> lengths) {
+             py::gil_scoped_release rel;
              return e.decode(z, y_mask, g, lengths);
            },
            py::arg("z"), py::arg("y_mask"), py::arg("g") = py::none(),
