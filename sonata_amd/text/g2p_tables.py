@@ -209,3 +209,84 @@ TABLES = {
 }
 # sr (Serbian latin) shares the hr table; nb/nn map to no
 ALIASES = {"sr": "hr", "nb": "no", "nn": "no", "ms": "id"}
+
+
+# --------------------------------------------------------------------- #
+# French support layer (round 2): the rule table alone mispronounces the
+# silent final consonants that dominate real French text.  A word
+# PREPROCESS strips silent finals before the rules, and a lexicon pins
+# the top function words (whose vowels the rules cannot guess).
+# --------------------------------------------------------------------- #
+import re as _re
+
+_FR_KEEP_FINAL = {
+    "avec", "fils", "mars", "bus", "sud", "est", "ouest", "net", "sept",
+    "huit", "six", "dix", "lys", "sens", "chef", "bref", "neuf", "sac",
+    "lac", "parc", "truc", "chic", "bac",
+}
+
+
+def fr_preprocess(w: str) -> str:
+    """Strip silent final consonants / e-muet before rule application."""
+    if w in _FR_KEEP_FINAL or len(w) <= 2:
+        return w
+    w = _re.sub(r"(er|ez)$", "é", w)       # parler -> parlé (= /e/)
+    w = _re.sub(r"[stdxzp]$", "", w)       # petit, temps, grand, prix...
+    w = _re.sub(r"[stdxzp]$", "", w)       # temps: strip s then p
+    if len(w) > 2 and w.endswith("e") and w[-2] not in "aeiouéè":
+        w = w[:-1]                          # e-muet: chose -> chos
+    return w or w
+
+
+FR_LEXICON = {
+    # articles / pronouns / function words (rule-resistant vowels)
+    "le": "lə", "la": "la", "les": "le", "un": "œ̃", "une": "yn",
+    "des": "de", "du": "dy", "de": "də", "au": "o", "aux": "o",
+    "ce": "sə", "ces": "se", "cet": "sɛt", "cette": "sɛt",
+    "je": "ʒə", "tu": "ty", "il": "il", "elle": "ɛl", "on": "ɔ̃",
+    "nous": "nu", "vous": "vu", "ils": "il", "elles": "ɛl",
+    "mon": "mɔ̃", "ma": "ma", "mes": "me", "ton": "tɔ̃", "tes": "te",
+    "son": "sɔ̃", "sa": "sa", "ses": "se", "notre": "nɔtʁ",
+    "votre": "vɔtʁ", "leur": "lœʁ", "leurs": "lœʁ",
+    "et": "e", "ou": "u", "où": "u", "mais": "mɛ", "donc": "dɔ̃k",
+    "or": "ɔʁ", "ni": "ni", "car": "kaʁ", "si": "si", "que": "kə",
+    "qui": "ki", "quoi": "kwa", "dont": "dɔ̃", "quand": "kɑ̃",
+    "comme": "kɔm", "comment": "kɔmˈɑ̃", "pourquoi": "puʁkwˈa",
+    "est": "ɛ", "es": "ɛ", "sont": "sɔ̃", "suis": "sɥi", "êtes": "ɛt",
+    "sommes": "sɔm", "était": "etˈɛ", "être": "ɛtʁ", "été": "etˈe",
+    "a": "a", "as": "a", "ont": "ɔ̃", "avons": "avˈɔ̃", "avez": "avˈe",
+    "avoir": "avwˈaʁ", "avait": "avˈɛ", "eu": "y",
+    "fait": "fɛ", "faire": "fɛʁ", "fais": "fɛ", "font": "fɔ̃",
+    "va": "va", "vais": "vɛ", "vont": "vɔ̃", "aller": "alˈe",
+    "dit": "di", "dire": "diʁ", "peut": "pø", "peux": "pø",
+    "pouvoir": "puvwˈaʁ", "veut": "vø", "veux": "vø",
+    "vouloir": "vulwˈaʁ", "doit": "dwa", "devoir": "dəvwˈaʁ",
+    "sait": "sɛ", "savoir": "savwˈaʁ", "voit": "vwa", "voir": "vwaʁ",
+    "pas": "pa", "ne": "nə", "non": "nɔ̃", "oui": "wi", "plus": "ply",
+    "moins": "mwɛ̃", "très": "tʁɛ", "trop": "tʁo", "peu": "pø",
+    "beaucoup": "bokˈu", "bien": "bjɛ̃", "mal": "mal", "tout": "tu",
+    "tous": "tus", "toute": "tut", "toutes": "tut", "rien": "ʁjɛ̃",
+    "dans": "dɑ̃", "sur": "syʁ", "sous": "su", "avant": "avˈɑ̃",
+    "après": "apʁˈɛ", "pendant": "pɑ̃dˈɑ̃", "depuis": "dəpɥˈi",
+    "pour": "puʁ", "par": "paʁ", "sans": "sɑ̃", "chez": "ʃe",
+    "entre": "ɑ̃tʁ", "vers": "vɛʁ", "contre": "kɔ̃tʁ",
+    "ici": "isˈi", "là": "la", "aujourd'hui": "oʒuʁdɥˈi",
+    "hier": "jɛʁ", "demain": "dəmˈɛ̃", "maintenant": "mɛ̃tnˈɑ̃",
+    "toujours": "tuʒˈuʁ", "jamais": "ʒamˈɛ", "souvent": "suvˈɑ̃",
+    "déjà": "deʒˈa", "encore": "ɑ̃kˈɔʁ", "aussi": "osˈi",
+    "alors": "alˈɔʁ", "ainsi": "ɛ̃sˈi", "puis": "pɥi",
+    "monsieur": "məsjˈø", "madame": "madˈam", "merci": "mɛʁsˈi",
+    "bonjour": "bɔ̃ʒˈuʁ", "bonsoir": "bɔ̃swˈaʁ", "salut": "salˈy",
+    "temps": "tɑ̃", "fois": "fwa", "jour": "ʒuʁ", "nuit": "nɥi",
+    "an": "ɑ̃", "ans": "ɑ̃", "année": "anˈe", "monde": "mɔ̃d",
+    "gens": "ʒɑ̃", "femme": "fam", "homme": "ɔm", "enfant": "ɑ̃fˈɑ̃",
+    "eau": "o", "ville": "vil", "pays": "peˈi", "france": "fʁɑ̃s",
+    "français": "fʁɑ̃sˈɛ", "deux": "dø", "trois": "tʁwa",
+    "quatre": "katʁ", "cinq": "sɛ̃k", "huit": "ɥit", "vingt": "vɛ̃",
+    "cent": "sɑ̃", "mille": "mil", "premier": "pʁəmjˈe",
+    "grand": "ɡʁɑ̃", "grande": "ɡʁɑ̃d", "petit": "pətˈi",
+    "petite": "pətˈit", "bon": "bɔ̃", "bonne": "bɔn", "beau": "bo",
+    "belle": "bɛl", "nouveau": "nuvˈo", "nouvelle": "nuvˈɛl",
+    "vieux": "vjø", "jeune": "ʒœn", "autre": "otʁ", "même": "mɛm",
+    "seul": "sœl", "chose": "ʃoz", "choses": "ʃoz",
+}

@@ -75,10 +75,12 @@ class RuleG2P:
         letters: str = "a-z",
         stress: bool = True,
         unstressed: Optional[set] = None,
+        preprocess=None,
     ):
         self.lexicon = lexicon or {}
         self.stress = stress
         self.unstressed = unstressed or set()
+        self.preprocess = preprocess  # word -> word, before rules
         # sort patterns by length desc for longest match
         self._patterns = sorted(rules.items(), key=lambda kv: -len(kv[0]))
         self._rules = rules
@@ -107,7 +109,8 @@ class RuleG2P:
         if ipa is None:
             ipa = self._inflect(w)
         if ipa is None:
-            ipa = self._apply_rules(w)
+            ipa = self._apply_rules(
+                self.preprocess(w) if self.preprocess else w)
             if self.stress:
                 ipa = self._stress_rules_output(w, ipa)
         if (self.stress and ipa and w not in self.unstressed
@@ -610,7 +613,11 @@ def _get_g2p(voice: str) -> RuleG2P:
     elif base == "es":
         g = RuleG2P(_ES_RULES, letters="a-zA-Zñáéíóúü")
     elif base == "fr":
-        g = RuleG2P(_FR_RULES, letters="a-zA-Zàâçéèêëîïôûùüœ")
+        from .g2p_tables import FR_LEXICON, fr_preprocess
+
+        g = RuleG2P(_FR_RULES, FR_LEXICON,
+                    letters="a-zA-Zàâçéèêëîïôûùüœ'",
+                    preprocess=fr_preprocess)
     elif base == "it":
         g = RuleG2P(_IT_RULES, letters="a-zA-Zàèéìòù")
     elif base == "pt":

@@ -249,3 +249,33 @@ def test_greek_accent_is_stress():
     # unaccented word falls back to first-syllable stress
     ipa = g.word_to_ipa("και")
     assert ipa.count("ˈ") == 1
+
+
+# ---- French (lexicon + silent-final preprocess) ------------------------ #
+FR_CASES = [
+    ("est", "ɛ"), ("les", "le"), ("vous", "vu"), ("dans", "dɑ̃"),
+    ("pas", "pa"), ("tout", "tu"), ("beaucoup", "bokˈu"),
+    ("temps", "tɑ̃"), ("petit", "pətˈi"), ("grand", "ɡʁɑ̃"),
+    ("bonjour", "bɔ̃ʒˈuʁ"), ("merci", "mɛʁsˈi"), ("être", "ɛtʁ"),
+    ("faire", "fɛʁ"), ("france", "fʁɑ̃s"), ("monde", "mɔ̃d"),
+    ("toujours", "tuʒˈuʁ"), ("aujourd'hui", "oʒuʁdɥˈi"),
+]
+
+
+def test_fr_lexicon_accuracy():
+    g = _get_g2p("fr")
+    wrong = []
+    for w, want in FR_CASES:
+        got = g.word_to_ipa(w).replace("ˈ", "")
+        if got != want.replace("ˈ", ""):
+            wrong.append((w, got, want))
+    assert len(wrong) <= 1, wrong
+
+
+def test_fr_silent_finals_rule_path():
+    """Out-of-lexicon words still drop silent finals (preprocess)."""
+    g = _get_g2p("fr")
+    for w, absent in [("chats", "s"), ("parlait", "t"),
+                      ("normand", "d"), ("galop", "p")]:
+        ipa = g.word_to_ipa(w)
+        assert not ipa.rstrip(".").endswith(absent), (w, ipa)
