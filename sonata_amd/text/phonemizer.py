@@ -76,11 +76,15 @@ class RuleG2P:
         stress: bool = True,
         unstressed: Optional[set] = None,
         preprocess=None,
+        stress_default: str = "first",
     ):
         self.lexicon = lexicon or {}
         self.stress = stress
         self.unstressed = unstressed or set()
         self.preprocess = preprocess  # word -> word, before rules
+        # "first" | "es-penult" (Spanish: penult when the word ends in a
+        # vowel/n/s, final syllable otherwise; accents override upstream)
+        self.stress_default = stress_default
         # sort patterns by length desc for longest match
         self._patterns = sorted(rules.items(), key=lambda kv: -len(kv[0]))
         self._rules = rules
@@ -220,7 +224,12 @@ class RuleG2P:
                 idx = max(len(starts) - 1 - from_end, 0)
                 break
         else:
-            if len(starts) >= 4:
+            if self.stress_default == "es-penult":
+                if word[-1] in "aeiouns" and len(starts) >= 2:
+                    idx = len(starts) - 2
+                else:
+                    idx = len(starts) - 1
+            elif len(starts) >= 4:
                 idx = 1
         pos = starts[min(idx, len(starts) - 1)]
         return ipa[:pos] + "ˈ" + ipa[pos:]
@@ -447,6 +456,7 @@ _DE_RULES = {
 _ES_RULES = {
     "ch": "tʃ", "ll": "ʝ", "rr": "r", "qu": "k", "gu": "ɡ",
     "ñ": "ɲ", "j": "x", "v": "b", "z": "θ", "ce": "θe", "ci": "θi",
+    "á": "ˈa", "é": "ˈe", "í": "ˈi", "ó": "ˈo", "ú": "ˈu", "ü": "w",
     "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
     "g": "ɡ", "h": "", "i": "i", "k": "k", "l": "l", "m": "m",
     "n": "n", "o": "o", "p": "p", "r": "ɾ", "s": "s", "t": "t",
@@ -611,7 +621,8 @@ def _get_g2p(voice: str) -> RuleG2P:
     elif base == "de":
         g = RuleG2P(_DE_RULES, letters="a-zA-Zäöüß")
     elif base == "es":
-        g = RuleG2P(_ES_RULES, letters="a-zA-Zñáéíóúü")
+        g = RuleG2P(_ES_RULES, letters="a-zA-Zñáéíóúü",
+                    stress_default="es-penult")
     elif base == "fr":
         from .g2p_tables import FR_LEXICON, fr_preprocess
 
