@@ -150,6 +150,9 @@ class SonataGrpcService:
             context.abort(grpc.StatusCode.INVALID_ARGUMENT,
                           f"failed to load voice: {e}")
         v = _Voice(voice_id, SonataSpeechSynthesizer(voice))
+        # warm graph captures in the background: the first real request
+        # shouldn't pay the one-time capture cost (~36 ms measured)
+        threading.Thread(target=voice.warmup, daemon=True).start()
         with self._lock:
             winner = self._voices.setdefault(voice_id, v)
         if winner is not v:

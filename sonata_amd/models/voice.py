@@ -272,6 +272,21 @@ class VitsVoice(SonataModel):
         yield from self._stream_decode(z, y_mask, g, chunk_size,
                                        chunk_padding)
 
+    @torch.no_grad()
+    def warmup(self) -> None:
+        """Pay one-time costs (hipGraph captures for the default stream
+        path, kernel/module caches) at LOAD time so the first real
+        request doesn't (cold first request measured ~36 ms of capture;
+        profiles/r02_modes_final.json).  Safe no-op on CPU."""
+        if self.device.type != "cuda":
+            return
+        try:
+            for _ in self.stream_synthesis("wˈɔːm ˈʌp sˈɛntəns.", 45, 3):
+                pass
+            self.speak_one_sentence("wˈɔːm.")
+        except Exception:  # warmup must never break loading
+            pass
+
     def _stream_decode(self, z, y_mask, g, chunk_size: int,
                        chunk_padding: int) -> Iterator[np.ndarray]:
         from ..utils.graphs import DecodeGraphCache, enabled as graphs_on
@@ -425,6 +440,9 @@ def load_voice(
     # serving runtime: the C++ VitsEngine by default (same kernels, same
     # per-utterance seeds -> identical audio); SONATA_ENGINE=python or
     # engine="python" keeps the torch-module path.
+    #
+    # warmup(): see VitsVoice.warmup — capture/compile costs paid at
+    # load time instead of the first request.
     eng = None
     choice = os.environ.get("SONATA_ENGINE", engine)
     if choice == "auto" and not device.startswith("cuda"):
