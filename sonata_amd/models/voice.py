@@ -66,6 +66,15 @@ class VitsVoice(SonataModel):
         self._engine = engine
         self._synth_config = config.default_synthesis_config()
         self._cfg_lock = threading.Lock()
+        # Serializes GPU inference on THIS voice: the hipGraph caches
+        # (capture + replay via shared capture buffers), the pinned
+        # staging buffers of the stream pipeline, and the C++ engine's
+        # lazily-built weight cache are all per-voice mutable state.
+        # Concurrent callers on one voice otherwise race (a 4-minute
+        # mixed-load soak surfaced rare hipErrorInvalidConfiguration).
+        # Cross-voice concurrency is unaffected; the gRPC batcher
+        # already funnels per-voice work through one thread.
+        self._infer_lock = threading.Lock()
         self._tashkeel = None
         if config.espeak_voice.startswith("ar"):
             from ..text.tashkeel import TashkeelModel
@@ -152,7 +161,7 @@ class VitsVoice(SonataModel):
             seeds = [_utterance_seed(p, cfg.speaker_id)
                      for p in phonemes_batch]
             sid_t = self._sid_tensor(B, cfg.speaker_id)
-            with stage_timer("infer", self.device):
+            with self._infer_lock, stage_timer("infer", self.device):
                 audio, audio_lengths = self._engine.infer(
                     ids, lengths, sid_t, cfg.noise_scale, cfg.length_scale,
                     cfg.noise_w, seeds)
@@ -168,7 +177,7 @@ class VitsVoice(SonataModel):
                 out.append(Audio(audio[b, 0, :n], info,
                                  inference_ms=infer_ms / B))
             return out
-        with stage_timer("infer", self.device):
+        with self._infer_lock, stage_timer("infer", self.device):
             audio, audio_lengths = self.net.infer(
             ids,
             lengths,
@@ -211,15 +220,26 @@ class VitsVoice(SonataModel):
         ids_l = self._encode_ids(phonemes)
         ids = torch.tensor([ids_l], dtype=torch.long, device=self.device)
         lengths = torch.tensor([len(ids_l)], dtype=torch.long, device=self.device)
-        if (phase1_enabled() and self.device.type == "cuda"
-                and self.config.num_speakers <= 1):
-            # hipGraph-replayed encoder phase 1 (default ON: 9.1 -> 6.8 ms
-            # first chunk at B=1; the sync-free spline made capture legal).
-            # Gated to single-speaker voices: the captured phase-1 closure
-            # runs with g=None, which would drop speaker conditioning.
-            yield from self._stream_graphed(
-                phonemes, cfg, ids, lengths, chunk_size, chunk_padding)
-            return
+        # the lock spans the whole stream (graph caches + pinned staging
+        # are per-voice state); released when the generator finishes or
+        # is closed/GC'd
+        with self._infer_lock:
+            if (phase1_enabled() and self.device.type == "cuda"
+                    and self.config.num_speakers <= 1):
+                # hipGraph-replayed encoder phase 1 (default ON: 9.1 ->
+                # 6.8 ms first chunk at B=1; the sync-free spline made
+                # capture legal).  Gated to single-speaker voices: the
+                # captured phase-1 closure runs with g=None, which would
+                # drop speaker conditioning.
+                yield from self._stream_graphed(
+                    phonemes, cfg, ids, lengths, chunk_size, chunk_padding)
+                return
+            yield from self._stream_eager(phonemes, cfg, ids, lengths,
+                                          chunk_size, chunk_padding)
+
+    @torch.no_grad()
+    def _stream_eager(self, phonemes, cfg, ids, lengths, chunk_size,
+                      chunk_padding):
         with stage_timer("encode", self.device):
             if self._engine is not None:
                 # C++ engine encoder: same kernels/seeds, no per-launch
