@@ -117,3 +117,51 @@ def test_synth_config_delegation(synth):
     assert synth.get_synthesis_config().length_scale == 1.5
     cfg.length_scale = 1.0
     synth.set_synthesis_config(cfg)
+
+
+def test_concurrent_streams_one_voice(tmp_path):
+    """Regression for the soak-found race: concurrent streams + batch
+    calls on ONE voice must serialize cleanly through the per-voice
+    inference lock (incl. a consumer that abandons its generator)."""
+    import threading
+
+    from sonata_amd.models import create_random_voice
+    from sonata_amd.models.voice import load_voice
+
+    pack = create_random_voice(str(tmp_path), "cc", quality="x_low")
+    v = load_voice(pack, device="cpu")
+    errs = []
+
+    def stream_worker():
+        try:
+            for _ in range(3):
+                total = sum(len(c) for c in
+                            v.stream_synthesis("wˈʌn tˈuː θɹˈiː.", 45, 3))
+                assert total > 100
+        except Exception as e:  # noqa: BLE001
+            errs.append(e)
+
+    def abandon_worker():
+        try:
+            for _ in range(3):
+                it = v.stream_synthesis("fˈoːɹ fˈaɪv.", 45, 3)
+                next(it)
+                it.close()  # abandon mid-stream: lock must release
+        except Exception as e:  # noqa: BLE001
+            errs.append(e)
+
+    def batch_worker():
+        try:
+            for _ in range(3):
+                v.speak_batch(["sˈɪks sˈɛvən.", "ˈeɪt."])
+        except Exception as e:  # noqa: BLE001
+            errs.append(e)
+
+    threads = [threading.Thread(target=f) for f in
+               (stream_worker, stream_worker, abandon_worker, batch_worker)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=300)
+    assert not errs, errs
+    assert not any(t.is_alive() for t in threads), "deadlock"
