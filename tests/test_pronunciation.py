@@ -77,6 +77,25 @@ EN_LEXICON_CASES = [
     ("especially", "əspˈɛʃəli"), ("certainly", "sˈɝtənli"),
     ("immediately", "ɪmˈidiətli"), ("opportunity", "ɑpɚtˈunəti"),
     ("responsibility", "ɹɪspɑnsəbˈɪləti"), ("environment", "ɪnvˈaɪɹənmənt"),
+    # entries from the r2 expansion batches
+    ("achieve", "ətʃˈiv"), ("analysis", "ənˈæləsɪs"),
+    ("appropriate", "əpɹˈoʊpɹiət"), ("committee", "kəmˈɪti"),
+    ("communicate", "kəmjˈunəkeɪt"), ("conclusion", "kənklˈuʒən"),
+    ("democracy", "dɪmˈɑkɹəsi"), ("economic", "ɛkənˈɑmɪk"),
+    ("emergency", "ɪmˈɝdʒənsi"), ("equipment", "ɪkwˈɪpmənt"),
+    ("executive", "ɪɡzˈɛkjətɪv"), ("foundation", "faʊndˈeɪʃən"),
+    ("guarantee", "ɡɛɹəntˈi"), ("hypothesis", "haɪpˈɑθəsɪs"),
+    ("identity", "aɪdˈɛntəti"), ("individual", "ɪndəvˈɪdʒuəl"),
+    ("intelligence", "ɪntˈɛlədʒəns"), ("interview", "ˈɪntɚvju"),
+    ("laboratory", "lˈæbɹətɔɹi"), ("literature", "lˈɪtɚətʃɚ"),
+    ("majority", "mədʒˈɔɹəti"), ("mechanism", "mˈɛkənɪzəm"),
+    ("negotiate", "nəɡˈoʊʃieɪt"), ("phenomenon", "fənˈɑmənɑn"),
+    ("priority", "pɹaɪˈɔɹəti"), ("procedure", "pɹəsˈidʒɚ"),
+    ("recognize", "ɹˈɛkəɡnaɪz"), ("schedule", "skˈɛdʒul"),
+    ("significant", "sɪɡnˈɪfɪkənt"), ("statistics", "stətˈɪstɪks"),
+    ("strategy", "stɹˈætədʒi"), ("sufficient", "səfˈɪʃənt"),
+    ("temporary", "tˈɛmpɚɛɹi"), ("variety", "vɚɹˈaɪəti"),
+    ("vehicle", "vˈiəkəl"), ("volunteer", "vɑləntˈɪɹ"),
 ]
 
 # ---- English tier B: inflection layer --------------------------------- #
