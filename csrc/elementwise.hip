@@ -572,3 +572,86 @@ torch::Tensor seeded_noise(long B, long C, long T_max, torch::Tensor lens,
   }
   return out;
 }
+
+// ------------------------------------------------------------------------- //
+// row_ln_cl: LayerNorm over the last (channel) dim of channel-last rows
+// with FUSED residual add: y = LN(x + r) * gamma + beta.  The encoder
+// calls torch::layer_norm(x + y) ~400x per step (two launches + an
+// intermediate tensor each); this is one launch, one pass.
+// One wave per row (C <= 8*64 via per-lane accumulation + butterfly).
+// ------------------------------------------------------------------------- //
+template <typename T>
+__global__ void row_ln_cl_kernel(const T* __restrict__ x,
+                                 const T* __restrict__ r,
+                                 const float* __restrict__ gamma,
+                                 const float* __restrict__ beta,
+                                 T* __restrict__ out, long rows, int C,
+                                 float eps) {
+  const long row = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= rows) return;
+  const int lane = threadIdx.x & 63;
+  const T* xr = x + row * C;
+  const T* rr = r ? r + row * C : nullptr;
+  float v[8];
+  float sum = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    const int c = lane + i * 64;
+    float t = 0.f;
+    if (c < C) {
+      t = ld_f(xr + c);
+      if (rr) t += ld_f(rr + c);
+    }
+    v[i] = t;
+    sum += t;
+  }
+#pragma unroll
+  for (int sh = 32; sh > 0; sh >>= 1) sum += __shfl_xor(sum, sh, 64);
+  const float mean = sum / C;
+  float var = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    const int c = lane + i * 64;
+    if (c < C) {
+      v[i] -= mean;
+      var += v[i] * v[i];
+    }
+  }
+#pragma unroll
+  for (int sh = 32; sh > 0; sh >>= 1) var += __shfl_xor(var, sh, 64);
+  const float inv = rsqrtf(var / C + eps);
+  T* orow = out + row * C;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    const int c = lane + i * 64;
+    if (c < C) st_f(orow + c, v[i] * inv * gamma[c] + beta[c]);
+  }
+}
+
+torch::Tensor row_ln_cl(torch::Tensor x, c10::optional<torch::Tensor> resid,
+                        torch::Tensor gamma, torch::Tensor beta,
+                        double eps) {
+  TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous());
+  const long B = x.size(0), T = x.size(1), C = x.size(2);
+  TORCH_CHECK(C <= 512, "row_ln_cl: C too large");
+  auto g32 = gamma.scalar_type() == at::kFloat ? gamma.contiguous()
+                                               : gamma.to(at::kFloat).contiguous();
+  auto b32 = beta.scalar_type() == at::kFloat ? beta.contiguous()
+                                              : beta.to(at::kFloat).contiguous();
+  auto out = torch::empty_like(x);
+  const long rows = B * T;
+  if (!rows) return out;
+  const void* rp = nullptr;
+  if (resid.has_value()) {
+    TORCH_CHECK(resid->is_contiguous() && resid->sizes() == x.sizes());
+    rp = resid->data_ptr();
+  }
+  const int waves_per_block = 4;  // 256 threads
+  const int blocks = (int)((rows + waves_per_block - 1) / waves_per_block);
+  DISPATCH_FT_CONV(x, hipLaunchKernelGGL(
+      row_ln_cl_kernel<scalar_t>, dim3(blocks), dim3(64 * waves_per_block),
+      0, cur_stream(), (const scalar_t*)x.data_ptr(),
+      (const scalar_t*)rp, g32.data_ptr<float>(), b32.data_ptr<float>(),
+      (scalar_t*)out.data_ptr(), rows, (int)C, (float)eps));
+  return out;
+}

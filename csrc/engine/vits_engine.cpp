@@ -29,6 +29,8 @@ torch::Tensor expand_states(torch::Tensor stats, torch::Tensor durs,
                             long F_max);
 torch::Tensor seeded_noise(long B, long C, long T_max, torch::Tensor lens,
                            torch::Tensor seeds, torch::ScalarType dtype);
+torch::Tensor row_ln_cl(torch::Tensor x, c10::optional<torch::Tensor> resid,
+                        torch::Tensor gamma, torch::Tensor beta, double eps);
 torch::Tensor conv1d_fused(torch::Tensor x, torch::Tensor w_perm,
                            c10::optional<torch::Tensor> bias, long Cout,
                            long k, long stride, long padding, long dilation,
@@ -472,8 +474,9 @@ VitsEngine::text_encoder(torch::Tensor ids, torch::Tensor lengths) const {
                                  cache_["relv:" + am], lens32, H,
                                  cfg_.window_size, scale);
       auto y = lin(outt, am + ".conv_o");
-      x = torch::layer_norm(x + y, {C}, p("enc_p.norm1." + li + ".gamma"),
-                            p("enc_p.norm1." + li + ".beta"), 1e-5);
+      x = row_ln_cl(x.contiguous(), y.contiguous(),
+                    p("enc_p.norm1." + li + ".gamma"),
+                    p("enc_p.norm1." + li + ".beta"), 1e-5);
       std::string f1 = "enc_p.ffn_layers." + li + ".conv1";
       std::string f2 = "enc_p.ffn_layers." + li + ".conv2";
       long pad = cfg_.kernel_size / 2;
@@ -487,8 +490,9 @@ VitsEngine::text_encoder(torch::Tensor ids, torch::Tensor lengths) const {
       f = conv1d_cl_fused(f, perm_conv(f2 + ".weight"),
                           bias_f32(f2 + ".bias"), wf2.size(0), wf2.size(2),
                           pad, 1, -1.0, 0, 0.0, c10::nullopt, lens32);
-      x = torch::layer_norm(x + f, {C}, p("enc_p.norm2." + li + ".gamma"),
-                            p("enc_p.norm2." + li + ".beta"), 1e-5);
+      x = row_ln_cl(x.contiguous(), f.contiguous(),
+                    p("enc_p.norm2." + li + ".gamma"),
+                    p("enc_p.norm2." + li + ".beta"), 1e-5);
     }
     auto stats = (lin(x, "enc_p.proj") * mask_cl).transpose(1, 2);
     auto chunks = stats.chunk(2, 1);
@@ -776,11 +780,13 @@ torch::Tensor VitsEngine::sdp_infer(torch::Tensor x, torch::Tensor mask,
                               p(mod + ".convs_sep." + li + ".weight"),
                               bias_f32(mod + ".convs_sep." + li + ".bias"),
                               dilation, pad);
-        y = torch::layer_norm(y, {C}, p(mod + ".norms_1." + li + ".gamma"),
+        y = row_ln_cl(y.contiguous(), c10::nullopt,
+                      p(mod + ".norms_1." + li + ".gamma"),
                               p(mod + ".norms_1." + li + ".beta"), 1e-5);
         y = torch::gelu(y);
         y = lin(y, mod + ".convs_1x1." + li);
-        y = torch::layer_norm(y, {C}, p(mod + ".norms_2." + li + ".gamma"),
+        y = row_ln_cl(y.contiguous(), c10::nullopt,
+                      p(mod + ".norms_2." + li + ".gamma"),
                               p(mod + ".norms_2." + li + ".beta"), 1e-5);
         y = torch::gelu(y);
         t = t + y;

@@ -20,6 +20,8 @@ torch::Tensor expand_states(torch::Tensor stats, torch::Tensor durs,
 torch::Tensor mask_tail_(torch::Tensor x, torch::Tensor lens);
 torch::Tensor seeded_noise(long B, long C, long T_max, torch::Tensor lens,
                            torch::Tensor seeds, torch::ScalarType dtype);
+torch::Tensor row_ln_cl(torch::Tensor x, c10::optional<torch::Tensor> resid,
+                        torch::Tensor gamma, torch::Tensor beta, double eps);
 torch::Tensor depthwise_cl(torch::Tensor x, torch::Tensor w,
                            c10::optional<torch::Tensor> bias, long dil,
                            long pad);
@@ -93,6 +95,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                               dt == "bf16" ? at::kBFloat16 : at::kFloat);
         },
         "per-utterance counter-based normal noise, one launch");
+  m.def("row_ln_cl", &row_ln_cl,
+        "LayerNorm over channel-last rows with fused residual add");
   m.def("depthwise_cl", &depthwise_cl, "channel-last depthwise conv1d");
   m.def("conv1d_fused", &conv1d_fused, "MFMA conv1d with fused activations");
   m.def("convtranspose1d_fused", &convtranspose1d_fused,

@@ -23,6 +23,7 @@ from torch import nn
 
 from ..ops import (
     conv_mod,
+    row_ln_cl,
     depthwise_conv1d_cl,
     expand_states,
     fused_gate,
@@ -248,8 +249,7 @@ class TextEncoder(nn.Module):
                 y = attn_relpos_cl(x * x_mask_cl, attn, lengths)
             else:
                 y = attn.forward_cl(x * x_mask_cl, attn_mask.unsqueeze(1))
-            x = F.layer_norm(x + y, (x.shape[-1],), n1.gamma, n1.beta,
-                             n1.eps)
+            x = row_ln_cl(x, n1.gamma, n1.beta, n1.eps, residual=y)
             f = leaky_conv1d_cl(x * x_mask_cl, ffn.conv1.weight,
                                 ffn.conv1.bias,
                                 padding=ffn.conv1.padding[0], post_relu=True,
@@ -257,8 +257,7 @@ class TextEncoder(nn.Module):
             f = leaky_conv1d_cl(f, ffn.conv2.weight, ffn.conv2.bias,
                                 padding=ffn.conv2.padding[0],
                                 out_lens=lengths)
-            x = F.layer_norm(x + f, (x.shape[-1],), n2.gamma, n2.beta,
-                             n2.eps)
+            x = row_ln_cl(x, n2.gamma, n2.beta, n2.eps, residual=f)
         stats = F.linear(x, self.proj.weight.squeeze(-1), self.proj.bias)
         stats = (stats * x_mask_cl).transpose(1, 2)  # [B,2C,T]
         m, logs = stats.chunk(2, dim=1)
@@ -573,10 +572,10 @@ class DDSConv(nn.Module):
                                     self.norms_1, self.norms_2):
             y = depthwise_conv1d_cl(x * x_mask, sep.weight, sep.bias,
                                     sep.dilation[0], sep.padding[0])
-            y = F.layer_norm(y, (C,), n1.gamma, n1.beta, n1.eps)
+            y = row_ln_cl(y, n1.gamma, n1.beta, n1.eps)
             y = F.gelu(y)
             y = F.linear(y, one.weight.squeeze(-1), one.bias)
-            y = F.layer_norm(y, (C,), n2.gamma, n2.beta, n2.eps)
+            y = row_ln_cl(y, n2.gamma, n2.beta, n2.eps)
             y = F.gelu(y)
             x = x + y
         return x * x_mask

@@ -70,4 +70,5 @@ from .functional import (  # noqa: F401,E402
     fused_gate_cl,
     depthwise_conv1d_cl,
     attn_relpos_cl,
+    row_ln_cl,
 )

@@ -540,3 +540,21 @@ def resblock_chain_cl(
         _lens_i32(out_lens, x.device),
         accum.contiguous() if accum is not None else None,
         float(out_scale))
+
+
+
+def row_ln_cl(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
+              eps: float = 1e-5,
+              residual: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """LayerNorm over the channel-last row dim with fused residual:
+    LN(x [+ residual]) * gamma + beta (csrc/elementwise.hip row_ln_cl).
+    One launch instead of add + layer_norm."""
+    if use_hip(x) and x.shape[-1] <= 512:
+        ext = hip_ext(required=True)
+        return ext.row_ln_cl(
+            x.contiguous(),
+            residual.contiguous() if residual is not None else None,
+            gamma, beta, eps)
+    if residual is not None:
+        x = x + residual
+    return F.layer_norm(x, (x.shape[-1],), gamma, beta, eps)

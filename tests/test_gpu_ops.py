@@ -731,3 +731,20 @@ def test_engine_python_noise_parity(dev, tmp_path):
     b = vp.speak_one_sentence("wˈʌn tˈuː θɹˈiː.").samples
     assert len(a) == len(b)
     assert float(np.abs(a - b).max()) < 0.05
+
+
+@pytest.mark.parametrize("C,resid", [(192, True), (192, False), (96, True),
+                                     (384, False)])
+def test_row_ln_cl(dev, C, resid):
+    ext = _ext()
+    torch.manual_seed(C)
+    B, T = 3, 127
+    x = torch.randn(B, T, C, device=dev, dtype=torch.bfloat16)
+    r = torch.randn_like(x) if resid else None
+    g = torch.randn(C, device=dev) * 0.5 + 1
+    b = torch.randn(C, device=dev) * 0.1
+    got = ext.row_ln_cl(x, r, g, b, 1e-5)
+    ref_in = (x + r) if resid else x
+    ref = torch.nn.functional.layer_norm(
+        ref_in.float(), (C,), g, b, 1e-5)
+    assert _rel_err(got, ref) < 0.02
