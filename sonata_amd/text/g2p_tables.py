@@ -290,3 +290,66 @@ FR_LEXICON = {
     "vieux": "vjø", "jeune": "ʒœn", "autre": "otʁ", "même": "mɛm",
     "seul": "sœl", "chose": "ʃoz", "choses": "ʃoz",
 }
+
+
+# --------------------------------------------------------------------- #
+# German support layer: final devoicing (Tag -> /k/), -er reduction,
+# double-consonant collapse, schwa endings + a small function-word
+# lexicon for the vowels rules cannot guess.
+# --------------------------------------------------------------------- #
+def de_postprocess(ipa: str) -> str:
+    """IPA-level fixes after the rule table."""
+    # final obstruent devoicing
+    if ipa.endswith("b"):
+        ipa = ipa[:-1] + "p"
+    elif ipa.endswith("d"):
+        ipa = ipa[:-1] + "t"
+    elif ipa.endswith("ɡ"):
+        ipa = ipa[:-1] + "k"
+    # -er coda -> vocalized ɐ
+    if ipa.endswith("ɛʁ"):
+        ipa = ipa[:-2] + "ɐ"
+    # unstressed -en / -e endings use schwa
+    if ipa.endswith("ɛn"):
+        ipa = ipa[:-2] + "ən"
+    return ipa
+
+
+def de_preprocess(w: str) -> str:
+    """Collapse double consonants (they mark the previous vowel short,
+    not a geminate): wasser -> waser."""
+    out = []
+    for ch in w:
+        if out and out[-1] == ch and ch not in "aeiouäöü":
+            continue
+        out.append(ch)
+    return "".join(out)
+
+
+DE_LEXICON = {
+    "der": "dɛʁ", "die": "diː", "das": "das", "ein": "ˈaɪn",
+    "eine": "ˈaɪnə", "und": "ʊnt", "ist": "ɪst", "sind": "zɪnt",
+    "war": "vaːʁ", "ich": "ɪç", "du": "duː", "er": "ɛʁ", "sie": "ziː",
+    "es": "ɛs", "wir": "viːʁ", "ihr": "iːʁ", "nicht": "nɪçt",
+    "mit": "mɪt", "auf": "ˈaʊf", "für": "fyːʁ", "von": "fɔn",
+    "zu": "tsuː", "im": "ɪm", "in": "ɪn", "an": "an", "am": "am",
+    "bei": "baɪ", "nach": "naːx", "über": "ˈyːbɐ", "unter": "ˈʊntɐ",
+    "aus": "ˈaʊs", "vor": "foːʁ", "durch": "dʊʁç", "gegen": "ɡˈeːɡən",
+    "ohne": "ˈoːnə", "um": "ʊm", "als": "als", "auch": "ˈaʊx",
+    "aber": "ˈaːbɐ", "oder": "ˈoːdɐ", "wenn": "vɛn", "dann": "dan",
+    "noch": "nɔx", "nur": "nuːʁ", "schon": "ʃoːn", "sehr": "zeːʁ",
+    "so": "zoː", "wie": "viː", "was": "vas", "wer": "veːʁ",
+    "wo": "voː", "ja": "jaː", "nein": "naɪn", "gut": "ɡuːt",
+    "haben": "hˈaːbən", "hat": "hat", "hatte": "hˈatə",
+    "werden": "vˈeːʁdən", "wird": "vɪʁt", "wurde": "vˈʊʁdə",
+    "kann": "kan", "können": "kˈœnən", "muss": "mʊs",
+    "machen": "mˈaxən", "sagen": "zˈaːɡən", "sagt": "zaːkt",
+    "gehen": "ɡˈeːən", "geht": "ɡeːt", "kommen": "kˈɔmən",
+    "kommt": "kɔmt", "sehen": "zˈeːən", "geben": "ɡˈeːbən",
+    "jahr": "jaːʁ", "jahre": "jˈaːʁə", "zeit": "tsaɪt", "tag": "taːk",
+    "mann": "man", "frau": "fʁaʊ", "kind": "kɪnt", "haus": "haʊs",
+    "stadt": "ʃtat", "land": "lant", "welt": "vɛlt", "leben": "lˈeːbən",
+    "wasser": "vˈasɐ", "heute": "hˈɔʏtə", "morgen": "mˈɔʁɡən",
+    "deutschland": "dˈɔʏtʃlant", "deutsch": "dɔʏtʃ", "hallo": "halˈoː",
+    "danke": "dˈaŋkə", "bitte": "bˈɪtə",
+}

@@ -117,6 +117,9 @@ class RuleG2P:
                 self.preprocess(w) if self.preprocess else w)
             if self.stress:
                 ipa = self._stress_rules_output(w, ipa)
+        post = getattr(self, "postprocess", None)
+        if post is not None and ipa:
+            ipa = post(ipa)
         if (self.stress and ipa and w not in self.unstressed
                 and "ˈ" not in ipa and "ˌ" not in ipa):
             # fallback: primary stress before the first vowel
@@ -619,7 +622,11 @@ def _get_g2p(voice: str) -> RuleG2P:
         g = RuleG2P(_EN_RULES, merged, letters="a-zA-Z",
                     unstressed=_EN_UNSTRESSED)
     elif base == "de":
-        g = RuleG2P(_DE_RULES, letters="a-zA-Zäöüß")
+        from .g2p_tables import DE_LEXICON, de_postprocess, de_preprocess
+
+        g = RuleG2P(_DE_RULES, DE_LEXICON, letters="a-zA-Zäöüß",
+                    preprocess=de_preprocess)
+        g.postprocess = de_postprocess
     elif base == "es":
         g = RuleG2P(_ES_RULES, letters="a-zA-Zñáéíóúü",
                     stress_default="es-penult")

@@ -298,3 +298,23 @@ def test_fr_silent_finals_rule_path():
                       ("normand", "d"), ("galop", "p")]:
         ipa = g.word_to_ipa(w)
         assert not ipa.rstrip(".").endswith(absent), (w, ipa)
+
+
+# ---- German (lexicon + devoicing/reduction postprocess) ---------------- #
+DE_CASES = [
+    ("tag", "taːk"), ("und", "ʊnt"), ("hund", "hʊnt"), ("weg", "vɛk"),
+    ("berg", "bɛʁk"), ("aber", "aːbɐ"), ("wasser", "vasɐ"),
+    ("schneller", "ʃnɛlɐ"), ("haben", "haːbən"), ("lieben", "liːbən"),
+    ("nicht", "nɪçt"), ("deutschland", "dɔʏtʃlant"), ("ich", "ɪç"),
+    ("heute", "hɔʏtə"), ("zeit", "tsaɪt"), ("stadt", "ʃtat"),
+]
+
+
+def test_de_accuracy():
+    g = _get_g2p("de")
+    wrong = []
+    for w, want in DE_CASES:
+        got = g.word_to_ipa(w).replace("ˈ", "")
+        if got != want:
+            wrong.append((w, got, want))
+    assert len(wrong) <= 1, wrong
