@@ -325,3 +325,59 @@ def test_native_concurrent_voices(lib, tmp_path_factory):
     for i, h in enumerate(handles):
         assert results[i] and results[i] > 1000, results
         lib.libsonataUnloadSonataVoice(h)
+
+
+@pytest.mark.gpu
+def test_capi_native_on_gpu(lib, tmp_path_factory):
+    """Native C-API synthesis on the GPU engine (bf16 kernels), driven
+    through ctypes exactly as a C caller would."""
+    import numpy as np
+
+    from sonata_amd.models import create_random_voice
+
+    d = tmp_path_factory.mktemp("capi_gpu")
+    pack = create_random_voice(str(d), "gpuv", quality="medium")
+    os.environ["SONATA_DEVICE"] = "cuda:0"
+    try:
+        err = ExternError()
+        h = lib.libsonataLoadVoiceFromConfigPath(pack.encode(),
+                                                 C.byref(err))
+        assert err.code == 0 and h
+        lib.libsonataIsNativeEngine.restype = C.c_uint8
+        lib.libsonataIsNativeEngine.argtypes = [C.c_void_p]
+        assert lib.libsonataIsNativeEngine(h) == 1
+
+        chunks = []
+
+        @CALLBACK
+        def cb(ev):
+            if ev.event_type == 0 and ev.len:
+                chunks.append(bytes(C.cast(
+                    ev.data, C.POINTER(C.c_uint8 * ev.len)).contents))
+            return 0
+
+        for mode in (0, 2):  # lazy one-shot + realtime streaming
+            chunks.clear()
+            params = SynthesisParams(mode=mode, rate=0, volume=0, pitch=0,
+                                     appended_silence_ms=0, callback=cb,
+                                     nonblocking=0)
+            err = ExternError()
+            lib.libsonataSpeak(
+                h, "One two three four five six seven.".encode(),
+                params, C.byref(err))
+            assert err.code == 0, err.message
+            pcm = np.frombuffer(b"".join(chunks), dtype=np.int16)
+            assert len(pcm) > 10000
+            assert np.abs(pcm).max() > 1000  # real audio, not zeros
+        # to-file path
+        out = str(d / "gpu.wav")
+        params = SynthesisParams(mode=0, rate=0, volume=0, pitch=0,
+                                 appended_silence_ms=0,
+                                 callback=C.cast(None, CALLBACK),
+                                 nonblocking=0)
+        ok = lib.libsonataSpeakToFile(h, b"Writing a file.", params,
+                                      out.encode(), C.byref(err))
+        assert ok == 1 and os.path.getsize(out) > 1000
+        lib.libsonataUnloadSonataVoice(h)
+    finally:
+        os.environ["SONATA_DEVICE"] = "cpu"
