@@ -512,10 +512,15 @@ _FR_RULES = {
 # Italian
 # --------------------------------------------------------------------------- #
 _IT_RULES = {
-    "gli": "ʎi", "gn": "ɲ", "sci": "ʃi", "sce": "ʃe", "chi": "ki",
-    "che": "ke", "ghi": "ɡi", "ghe": "ɡe", "ci": "tʃi", "ce": "tʃe",
-    "gi": "dʒi", "ge": "dʒe", "zz": "tts", "qu": "kw",
-    "à": "a", "è": "ɛ", "é": "e", "ì": "i", "ò": "ɔ", "ù": "u",
+    "glia": "ʎa", "glie": "ʎe", "glio": "ʎo", "gliu": "ʎu",
+    "gli": "ʎi", "gn": "ɲ",
+    "scia": "ʃa", "scio": "ʃo", "sciu": "ʃu", "sci": "ʃi", "sce": "ʃe",
+    "chi": "ki", "che": "ke", "ghi": "ɡi", "ghe": "ɡe",
+    "cia": "tʃa", "cio": "tʃo", "ciu": "tʃu", "ci": "tʃi", "ce": "tʃe",
+    "gia": "dʒa", "gio": "dʒo", "giu": "dʒu", "gi": "dʒi", "ge": "dʒe",
+    "zz": "tts", "qu": "kw",
+    # accented vowels mark stress (word-final in native orthography)
+    "à": "ˈa", "è": "ˈɛ", "é": "ˈe", "ì": "ˈi", "ò": "ˈɔ", "ù": "ˈu",
     "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
     "g": "ɡ", "h": "", "i": "i", "l": "l", "m": "m", "n": "n",
     "o": "o", "p": "p", "r": "r", "s": "s", "t": "t", "u": "u",
@@ -527,9 +532,13 @@ _IT_RULES = {
 # --------------------------------------------------------------------------- #
 _PT_RULES = {
     "lh": "ʎ", "nh": "ɲ", "ch": "ʃ", "qu": "k", "gu": "ɡ", "rr": "ʁ",
-    "ão": "ɐ̃w̃", "õe": "õj̃", "ã": "ɐ̃", "õ": "õ",
-    "á": "a", "â": "ɐ", "é": "ɛ", "ê": "e", "í": "i", "ó": "ɔ",
-    "ô": "o", "ú": "u", "ç": "s",
+    "ão": "ˈɐ̃w̃", "õe": "ˈõj̃", "ã": "ɐ̃", "õ": "õ",
+    "ce": "se", "ci": "si", "cé": "sˈɛ", "cê": "sˈe", "cí": "sˈi",
+    "ge": "ʒe", "gi": "ʒi", "gé": "ʒˈɛ", "gê": "ʒˈe", "gí": "ʒˈi",
+    "ss": "s",
+    # accented vowels mark stress
+    "á": "ˈa", "â": "ˈɐ", "é": "ˈɛ", "ê": "ˈe", "í": "ˈi", "ó": "ˈɔ",
+    "ô": "ˈo", "ú": "ˈu", "ç": "s",
     "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
     "g": "ɡ", "h": "", "i": "i", "j": "ʒ", "k": "k", "l": "l",
     "m": "m", "n": "n", "o": "o", "p": "p", "r": "ɾ", "s": "s",
@@ -637,9 +646,20 @@ def _get_g2p(voice: str) -> RuleG2P:
                     letters="a-zA-Zàâçéèêëîïôûùüœ'",
                     preprocess=fr_preprocess)
     elif base == "it":
-        g = RuleG2P(_IT_RULES, letters="a-zA-Zàèéìòù")
+        g = RuleG2P(_IT_RULES, letters="a-zA-Zàèéìòù",
+                    stress_default="es-penult")
     elif base == "pt":
-        g = RuleG2P(_PT_RULES, letters="a-zA-Zàáâãçéêíóôõú")
+        def _pt_post(ipa: str) -> str:
+            # BR final-vowel reduction: -o -> u, unstressed final -e -> i
+            if ipa.endswith("o") and "ˈo" != ipa[-2:]:
+                ipa = ipa[:-1] + "u"
+            elif ipa.endswith("e") and "ˈe" != ipa[-2:]:
+                ipa = ipa[:-1] + "i"
+            return ipa
+
+        g = RuleG2P(_PT_RULES, letters="a-zA-Zàáâãçéêíóôõú",
+                    stress_default="es-penult")
+        g.postprocess = _pt_post
     elif base == "ru":
         g = RuleG2P(_RU_RULES, letters="а-яА-ЯёЁ")
     elif base == "nl":

@@ -318,3 +318,25 @@ def test_de_accuracy():
         if got != want:
             wrong.append((w, got, want))
     assert len(wrong) <= 1, wrong
+
+
+# ---- Italian / Portuguese stress + digraphs ---------------------------- #
+def test_it_stress_and_digraphs():
+    g = _get_g2p("it")
+    cases = [("ciao", "tʃˈao"), ("parlare", "parlˈare"),
+             ("città", "tʃittˈa"), ("giorno", "dʒˈorno"),
+             ("molto", "mˈolto"), ("bambino", "bambˈino")]
+    wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
+             if g.word_to_ipa(w) != want]
+    assert len(wrong) <= 1, wrong
+
+
+def test_pt_stress_and_reduction():
+    g = _get_g2p("pt")
+    cases = [("obrigado", "obɾiɡˈadu"), ("você", "vosˈe"),
+             ("cidade", "sidˈadi"), ("coração", "koɾasˈɐ̃w̃"),
+             ("falar", "falˈaɾ"), ("bonito", "bonˈitu"),
+             ("gente", "ʒˈenti")]
+    wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
+             if g.word_to_ipa(w) != want]
+    assert len(wrong) <= 1, wrong
