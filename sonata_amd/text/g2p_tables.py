@@ -312,6 +312,8 @@ def de_postprocess(ipa: str) -> str:
     # unstressed -en / -e endings use schwa
     if ipa.endswith("ɛn"):
         ipa = ipa[:-2] + "ən"
+    elif ipa.endswith("ɛ"):
+        ipa = ipa[:-1] + "ə"
     return ipa
 
 

@@ -715,8 +715,11 @@ def text_to_phonemes(
     split per line, each line into sentences; each sentence's terminator
     (./,/?/!) is preserved at its clause position; optional separator is
     inserted between phonemes; stress marks optionally stripped."""
+    from .normalize import normalize as _normalize
+
     g2p = _get_g2p(voice)
-    out: List[str] = []
+    text = _normalize(text, voice)  # digits/abbrevs -> words (espeak
+    out: List[str] = []             # does this inside TranslateNumber)
     for line in text.splitlines() or [text]:
         if not line.strip():
             continue

@@ -162,3 +162,51 @@ def test_tashkeel_idempotent():
     once = m.diacritize("كتب الولد")
     twice = m.diacritize(once)
     assert twice == once
+
+
+# --------------------------------------------------------------------- #
+# text normalization (espeak expands numbers inside TranslateNumber;
+# the rule G2P previously dropped digit tokens)
+# --------------------------------------------------------------------- #
+def test_numbers_are_spoken_en():
+    from sonata_amd.text.phonemizer import text_to_phonemes
+
+    out = text_to_phonemes("I have 3 cats and 25 dogs.", voice="en-us")[0]
+    assert "θɹˈi" in out and "twˈɛnti fˈaɪv" in out
+
+
+def test_number_grammar_en():
+    from sonata_amd.text.normalize import (normalize_en, num_to_words_en,
+                                           ordinal_to_words_en,
+                                           year_to_words_en)
+
+    assert num_to_words_en(0) == "zero"
+    assert num_to_words_en(17) == "seventeen"
+    assert num_to_words_en(42) == "forty two"
+    assert num_to_words_en(105) == "one hundred five"
+    assert num_to_words_en(3211) == "three thousand two hundred eleven"
+    assert num_to_words_en(1000000) == "one million"
+    assert ordinal_to_words_en(1) == "first"
+    assert ordinal_to_words_en(3) == "third"
+    assert ordinal_to_words_en(22) == "twenty second"
+    assert ordinal_to_words_en(30) == "thirtieth"
+    assert year_to_words_en(1984) == "nineteen eighty four"
+    assert year_to_words_en(1900) == "nineteen hundred"
+    assert year_to_words_en(2007) == "two thousand seven"
+    assert year_to_words_en(2024) == "twenty twenty four"
+    assert normalize_en("$5") == "five dollars"
+    assert normalize_en("15%") == "fifteen percent"
+    assert normalize_en("Dr. Who met Mr. Jones.") == \
+        "doctor Who met mister Jones."
+    assert normalize_en("3.14 pies") == "three point one four pies"
+
+
+def test_numbers_other_languages_digitwise():
+    from sonata_amd.text.phonemizer import text_to_phonemes
+
+    de = text_to_phonemes("Ich habe 42 Katzen.", voice="de")[0]
+    assert "fˈiːʁ" in de and "tsvˈaɪ" in de  # vier zwei (digit-by-digit)
+    es = text_to_phonemes("Tengo 7 gatos.", voice="es")[0]
+    assert "sˈiete" in es
+    ru = text_to_phonemes("У меня 5 кошек.", voice="ru")[0]
+    assert ru  # no digit table for ru: number silently skipped, text ok
