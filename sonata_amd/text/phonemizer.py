@@ -85,6 +85,7 @@ class RuleG2P:
         # "first" | "es-penult" (Spanish: penult when the word ends in a
         # vowel/n/s, final syllable otherwise; accents override upstream)
         self.stress_default = stress_default
+        self.spell_acronyms = False  # letter names for all-caps tokens
         # sort patterns by length desc for longest match
         self._patterns = sorted(rules.items(), key=lambda kv: -len(kv[0]))
         self._rules = rules
@@ -263,7 +264,19 @@ class RuleG2P:
             kept = "".join(c if c in ",;:" else " " for c in between)
             if kept.strip(",;:") or kept:
                 parts.append(kept)
-            parts.append(self.word_to_ipa(m.group(0)))
+            tok = m.group(0)
+            if (self.spell_acronyms and tok.isupper()
+                    and 2 <= len(tok) <= 8
+                    and tok.lower() not in self.lexicon
+                    and (len(tok) <= 3
+                         or not any(c in "AEIOU" for c in tok))):
+                # all-caps token that doesn't look pronounceable ->
+                # letter names (espeak's acronym behavior)
+                parts.append(" ".join(_EN_LETTERS[c.lower()]
+                                      for c in tok
+                                      if c.lower() in _EN_LETTERS))
+            else:
+                parts.append(self.word_to_ipa(tok))
             pos = m.end()
         tail = text[pos:]
         parts.append("".join(c if c in ",;:" else " " for c in tail))
@@ -613,6 +626,18 @@ _CS_RULES = {
 }
 
 
+# English letter names for acronym spelling (espeak speaks all-caps
+# tokens letter-by-letter when they don't look pronounceable)
+_EN_LETTERS = {
+    "a": "ˈeɪ", "b": "bˈi", "c": "sˈi", "d": "dˈi", "e": "ˈi",
+    "f": "ˈɛf", "g": "dʒˈi", "h": "ˈeɪtʃ", "i": "ˈaɪ", "j": "dʒˈeɪ",
+    "k": "kˈeɪ", "l": "ˈɛl", "m": "ˈɛm", "n": "ˈɛn", "o": "ˈoʊ",
+    "p": "pˈi", "q": "kjˈu", "r": "ˈɑɹ", "s": "ˈɛs", "t": "tˈi",
+    "u": "jˈu", "v": "vˈi", "w": "dˈʌbəlju", "x": "ˈɛks", "y": "wˈaɪ",
+    "z": "zˈi",
+}
+
+
 _G2P_REGISTRY: Dict[str, RuleG2P] = {}
 
 
@@ -630,6 +655,7 @@ def _get_g2p(voice: str) -> RuleG2P:
         merged.update(_EN_STRESSED)  # stressed entries take precedence
         g = RuleG2P(_EN_RULES, merged, letters="a-zA-Z",
                     unstressed=_EN_UNSTRESSED)
+        g.spell_acronyms = True
     elif base == "de":
         from .g2p_tables import DE_LEXICON, de_postprocess, de_preprocess
 

@@ -210,3 +210,17 @@ def test_numbers_other_languages_digitwise():
     assert "sˈiete" in es
     ru = text_to_phonemes("У меня 5 кошек.", voice="ru")[0]
     assert ru  # no digit table for ru: number silently skipped, text ok
+
+
+def test_acronym_spelling_en():
+    from sonata_amd.text.phonemizer import text_to_phonemes
+
+    out = text_to_phonemes("The BBC uses HTML.", voice="en-us")[0]
+    assert "bˈi bˈi sˈi" in out
+    assert "ˈeɪtʃ tˈi ˈɛm ˈɛl" in out
+    # pronounceable all-caps (vowels, len>3) stays a word
+    out2 = text_to_phonemes("NASA launched.", voice="en-us")[0]
+    assert "ˈɛn ˈeɪ" not in out2
+    # mixed case / lowercase unaffected
+    out3 = text_to_phonemes("the cat", voice="en-us")[0]
+    assert out3.startswith("ðə")
