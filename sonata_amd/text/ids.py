@@ -25,6 +25,10 @@ _IPA_SYMBOLS = (
     "ˈˌːˑ",
     "ʲʷˤ",
     "ɪ̯ʰ",
+    # round-2 language expansion (sv/no/da/fi/hu/el/bg/uk/hr/sk + es/pt
+    # quality layers): appended so existing ids stay stable
+    "ɕɟɤɥɦɧɨɯʂʉʋʎʐʑ",
+    "̝̃",  # combining tilde (nasal) + raised diacritics
 )
 
 

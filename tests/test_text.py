@@ -224,3 +224,26 @@ def test_acronym_spelling_en():
     # mixed case / lowercase unaffected
     out3 = text_to_phonemes("the cat", voice="en-us")[0]
     assert out3.startswith("ðə")
+
+
+def test_id_map_covers_all_g2p_output():
+    """Every IPA char any bundled G2P table/lexicon can emit must be in
+    the default phoneme id map (unknown chars are silently dropped at
+    synthesis — r2 found 14 expansion symbols missing)."""
+    from sonata_amd.text import g2p_tables as G
+    from sonata_amd.text import phonemizer as P
+    from sonata_amd.text.en_lexicon import LEXICON as EN
+    from sonata_amd.text.ids import default_phoneme_id_map
+
+    chars = set()
+    tables = [P._EN_RULES, P._DE_RULES, P._ES_RULES, P._AR_RULES,
+              P._FR_RULES, P._IT_RULES, P._PT_RULES, P._RU_RULES,
+              P._NL_RULES, P._PL_RULES, P._TR_RULES, P._CS_RULES,
+              P._EN_LEXICON, EN, G.FR_LEXICON, G.DE_LEXICON,
+              P._EN_LETTERS] + list(G.TABLES.values())
+    for tbl in tables:
+        for v in tbl.values():
+            chars.update(v)
+    m = default_phoneme_id_map()
+    missing = sorted(c for c in chars if c not in m and c != " ")
+    assert not missing, missing
