@@ -117,7 +117,14 @@ def _expand_number_en(tok: str) -> str:
     return num_to_words_en(n)
 
 
+_DOTTED_ACRO_RE = re.compile(r"\b(?:[A-Z]\.){2,}")
+
+
 def normalize_en(text: str) -> str:
+    # dotted initialisms: U.S.A. -> USA (then spelled letter-by-letter
+    # by the phonemizer's acronym path)
+    text = _DOTTED_ACRO_RE.sub(
+        lambda m: m.group(0).replace(".", ""), text)
     text = _ABBR_RE.sub(
         lambda m: _EN_ABBREV[m.group(1).lower()], text)
     text = _CURRENCY_RE.sub(

@@ -247,3 +247,10 @@ def test_id_map_covers_all_g2p_output():
     m = default_phoneme_id_map()
     missing = sorted(c for c in chars if c not in m and c != " ")
     assert not missing, missing
+
+
+def test_dotted_initialisms():
+    from sonata_amd.text.phonemizer import text_to_phonemes
+
+    out = text_to_phonemes("The U.S.A. won.", voice="en-us")[0]
+    assert "jˈu ˈɛs ˈeɪ" in out
