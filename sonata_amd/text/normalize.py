@@ -120,11 +120,27 @@ def _expand_number_en(tok: str) -> str:
 _DOTTED_ACRO_RE = re.compile(r"\b(?:[A-Z]\.){2,}")
 
 
+_TIME_RE = re.compile(r"\b(\d{1,2}):(\d{2})\b")
+
+
+def _time_words(m: re.Match) -> str:
+    h, mm = int(m.group(1)), int(m.group(2))
+    if h > 23 or mm > 59:
+        return m.group(0)  # not a clock time; later passes handle digits
+    out = num_to_words_en(h if h else 12)
+    if mm == 0:
+        return out + " o'clock"
+    if mm < 10:
+        return out + " oh " + num_to_words_en(mm)
+    return out + " " + num_to_words_en(mm)
+
+
 def normalize_en(text: str) -> str:
     # dotted initialisms: U.S.A. -> USA (then spelled letter-by-letter
     # by the phonemizer's acronym path)
     text = _DOTTED_ACRO_RE.sub(
         lambda m: m.group(0).replace(".", ""), text)
+    text = _TIME_RE.sub(_time_words, text)
     text = _ABBR_RE.sub(
         lambda m: _EN_ABBREV[m.group(1).lower()], text)
     text = _CURRENCY_RE.sub(

@@ -254,3 +254,11 @@ def test_dotted_initialisms():
 
     out = text_to_phonemes("The U.S.A. won.", voice="en-us")[0]
     assert "jˈu ˈɛs ˈeɪ" in out
+
+
+def test_clock_times_en():
+    from sonata_amd.text.normalize import normalize_en
+
+    assert normalize_en("at 3:30") == "at three thirty"
+    assert normalize_en("12:00 sharp") == "twelve o'clock sharp"
+    assert normalize_en("9:05 train") == "nine oh five train"
