@@ -140,6 +140,9 @@ def normalize_en(text: str) -> str:
     # by the phonemizer's acronym path)
     text = _DOTTED_ACRO_RE.sub(
         lambda m: m.group(0).replace(".", ""), text)
+    # am/pm only in time context (the bare word "am" is a verb)
+    text = re.sub(r"(\d|o'clock)\s*[aA][mM]\b", r"\1 ay em", text)
+    text = re.sub(r"(\d|o'clock)\s*[pP][mM]\b", r"\1 pee em", text)
     text = _TIME_RE.sub(_time_words, text)
     text = _ABBR_RE.sub(
         lambda m: _EN_ABBREV[m.group(1).lower()], text)
