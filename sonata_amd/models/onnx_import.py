@@ -290,3 +290,91 @@ if __name__ == "__main__":
     out = import_onnx_voice(sys.argv[1],
                             sys.argv[2] if len(sys.argv) > 2 else None)
     print(f"wrote {out}")
+
+
+def import_streaming_pack(config_path: str, out_path: str = None,
+                          allow_partial: bool = False) -> str:
+    """Import a reference STREAMING voice pack: `config.json` with
+    sibling `encoder.onnx` + `decoder.onnx` (piper/src/lib.rs:90-96 —
+    selected by the config's `"streaming": true` key).  The two files'
+    initializers merge into one safetensors (this build serves one-shot
+    and streaming from a single net; the encoder/decoder split is a
+    method boundary, models/voice.py)."""
+    d = os.path.dirname(os.path.abspath(config_path))
+    enc = os.path.join(d, "encoder.onnx")
+    dec = os.path.join(d, "decoder.onnx")
+    for f in (enc, dec):
+        if not os.path.exists(f):
+            raise ModelError(f"streaming pack file missing: {f}")
+    import numpy as np
+    import torch
+    from safetensors.torch import save_file
+
+    inits: Dict[str, "np.ndarray"] = {}
+    inits.update(parse_onnx_initializers(enc))
+    dec_inits = parse_onnx_initializers(dec)
+    for name, arr in dec_inits.items():
+        # decoder tensors may be exported without the `dec.` prefix
+        if not name.startswith(("dec.", "onnx::")) and not _SKIP.match(name):
+            cand = "dec." + name
+            if name not in inits:
+                inits.setdefault(cand, arr)
+                continue
+        inits.setdefault(name, arr)
+    if not inits:
+        raise ModelError("no initializers found in streaming pack")
+    # reuse the single-file pipeline by writing through a temp merge
+    import tempfile
+
+    with tempfile.NamedTemporaryFile(suffix=".onnx", delete=False) as tf:
+        tmp = tf.name
+    try:
+        _write_min_onnx(tmp, inits)
+        if out_path is None:
+            stem = config_path
+            if stem.endswith(".json"):
+                stem = stem[: -len(".json")]
+            out_path = stem + ".safetensors"
+        return import_onnx_voice(tmp, out_path, config_path=config_path,
+                                 allow_partial=allow_partial)
+    finally:
+        os.unlink(tmp)
+
+
+def _write_min_onnx(path: str, inits) -> None:
+    """Serialize initializers back into a minimal ModelProto (merge
+    helper for import_streaming_pack)."""
+    import numpy as np
+
+    def varint(v: int) -> bytes:
+        out = b""
+        while True:
+            b7 = v & 0x7F
+            v >>= 7
+            if v:
+                out += bytes([b7 | 0x80])
+            else:
+                return out + bytes([b7])
+
+    def tag(field: int, wt: int) -> bytes:
+        return varint((field << 3) | wt)
+
+    def ld(field: int, payload: bytes) -> bytes:
+        return tag(field, 2) + varint(len(payload)) + payload
+
+    graph = b""
+    for name, arr in inits.items():
+        body = b""
+        for dim in arr.shape:
+            body += tag(1, 0) + varint(dim)
+        if np.issubdtype(arr.dtype, np.integer):
+            body += tag(2, 0) + varint(7)  # INT64
+            arr = arr.astype(np.int64)
+        else:
+            body += tag(2, 0) + varint(1)  # FLOAT
+            arr = arr.astype(np.float32)
+        body += ld(8, name.encode())
+        body += ld(9, arr.tobytes())
+        graph += ld(5, body)
+    with open(path, "wb") as f:
+        f.write(ld(7, graph))

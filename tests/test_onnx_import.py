@@ -265,3 +265,47 @@ def test_weight_norm_pairs_are_fused(tmp_path):
         got["flow.flows.0.enc.in_layers.0.weight"].numpy(), expected,
         rtol=1e-5)
     assert "flow.flows.0.enc.in_layers.0.weight_v" not in got
+
+
+def test_streaming_pack_import(tmp_path):
+    """Reference streaming packs = config.json + encoder.onnx +
+    decoder.onnx (piper lib.rs:90-96): both files' weights merge into
+    one pack that loads strict and synthesizes."""
+    import shutil
+
+    from sonata_amd.models import create_random_voice
+    from sonata_amd.models.config import ModelConfig
+    from sonata_amd.models.onnx_import import import_streaming_pack
+    from sonata_amd.models.vits import VitsModel
+    from sonata_amd.models.voice import load_voice, _weights_path_for
+
+    pack = create_random_voice(str(tmp_path), "rt", quality="x_low")
+    cfg = ModelConfig.from_json_path(pack)
+    from safetensors.torch import load_file
+
+    state = load_file(_weights_path_for(pack))
+    net = VitsModel(cfg.num_symbols, cfg.architecture, n_speakers=1)
+    net.load_state_dict(state, strict=True)
+    tensors = _upstream_export_tensors(net)
+    enc_t = [(n, a) for n, a in tensors if not n.startswith("dec.")]
+    # decoder exported WITHOUT the dec. prefix (upstream Generator tree)
+    dec_t = [(n[len("dec."):], a) for n, a in tensors
+             if n.startswith("dec.")]
+    # streaming layout: config.json + encoder.onnx + decoder.onnx
+    d = tmp_path / "rt_pack"
+    d.mkdir()
+    shutil.copy(pack, d / "config.json")
+    with open(d / "encoder.onnx", "wb") as f:
+        f.write(_onnx_bytes(enc_t))
+    with open(d / "decoder.onnx", "wb") as f:
+        f.write(_onnx_bytes(dec_t))
+    shutil.copy(_weights_path_for(pack).replace(".safetensors",
+                                                ".safetensors"),
+                d / "ignore.bin")  # decoy; importer must not need it
+
+    out = import_streaming_pack(str(d / "config.json"))
+    imported = load_file(out)
+    assert set(imported.keys()) == set(state.keys())
+    v = load_voice(str(d / "config.json"), device="cpu", engine="python")
+    chunks = list(v.stream_synthesis("tˈɛst strˈiːm.", 45, 3))
+    assert sum(len(c) for c in chunks) > 500
