@@ -51,8 +51,10 @@ def main():
     ctx = mp.get_context("spawn")
 
     from sonata_amd.frontends.grpc.server import _serve_one
+    import os
+    n_srv = int(os.environ.get("SRV_PROCS", "1"))
     servers = [ctx.Process(target=_serve_one, args=(port, device),
-                           daemon=True) for _ in range(2)]
+                           daemon=True) for _ in range(n_srv)]
     for p in servers:
         p.start()
     time.sleep(8)
@@ -93,7 +95,7 @@ def main():
         wall = time.perf_counter() - t0
         total_bytes = sum(r[0] for r in results)
         print(json.dumps({
-            "client_procs": n_cli, "server_procs": 2,
+            "client_procs": n_cli, "server_procs": n_srv,
             "utts": per * n_cli,
             "audio_sec_per_s": round(total_bytes / 2 / 22050 / wall, 1),
         }), flush=True)
