@@ -10,7 +10,7 @@ import time
 sys.path.insert(0, ".")
 
 
-def client_proc(port, vid, n_utts, conc, q):
+def client_proc(port, vid, n_utts, conc, q, bar):
     import grpc  # noqa: E402
 
     from sonata_amd.frontends.grpc.client import SonataGrpcClient
@@ -30,6 +30,7 @@ def client_proc(port, vid, n_utts, conc, q):
 
     with ThreadPoolExecutor(max_workers=conc) as ex:
         list(ex.map(one, range(conc * 2)))  # warm
+    bar.wait()  # all clients warm before anyone starts the timed run
     t0 = time.perf_counter()
     with ThreadPoolExecutor(max_workers=conc) as ex:
         sizes = list(ex.map(one, range(n_utts)))
@@ -82,13 +83,15 @@ def main():
     assert vid
     for n_cli in (1, 4):
         q = ctx.SimpleQueue()
+        bar = ctx.Barrier(n_cli + 1)
         per = 512 // n_cli
         procs = [ctx.Process(target=client_proc,
-                             args=(port, vid, per, 32, q))
+                             args=(port, vid, per, 32, q, bar))
                  for _ in range(n_cli)]
-        t0 = time.perf_counter()
         for p in procs:
             p.start()
+        bar.wait()  # released once every client finished its warmup
+        t0 = time.perf_counter()
         results = [q.get() for _ in range(n_cli)]
         for p in procs:
             p.join(timeout=300)
