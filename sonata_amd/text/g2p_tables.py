@@ -201,6 +201,362 @@ LETTERS = {
     "id": "a-zA-Z", "sw": "a-zA-Z'",
 }
 
+# ===================================================================== #
+# Second expansion batch: 23 more regular-orthography languages.
+# Stress behavior per language is in STRESS_DEFAULTS below (penult /
+# antepenult / final / first); approximations are documented inline and
+# in PARITY.md.  Accuracy samples: tests/test_pronunciation.py.
+# ===================================================================== #
+
+# Esperanto (eo): fully phonemic by design; penultimate stress
+EO_RULES = {
+    "ĉ": "tʃ", "ĝ": "dʒ", "ĥ": "x", "ĵ": "ʒ", "ŝ": "ʃ", "ŭ": "w",
+    "a": "a", "b": "b", "c": "ts", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "h", "i": "i", "j": "j", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "o", "p": "p", "r": "r", "s": "s",
+    "t": "t", "u": "u", "v": "v", "z": "z",
+}
+
+# Catalan (ca): central dialect leaning; accented vowels carry stress
+CA_RULES = {
+    "l·l": "l", "tx": "tʃ", "tj": "dʒ", "tg": "dʒ", "ll": "ʎ",
+    "ny": "ɲ", "ss": "s", "ix": "iʃ", "qu": "k", "gu": "ɡ",
+    "ce": "sɛ", "ci": "si", "ge": "ʒɛ", "gi": "ʒi", "ig": "itʃ",
+    "à": "ˈa", "è": "ˈɛ", "é": "ˈe", "í": "ˈi", "ò": "ˈɔ",
+    "ó": "ˈo", "ú": "ˈu", "ï": "i", "ü": "u", "ç": "s",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "ɛ", "f": "f",
+    "g": "ɡ", "h": "", "i": "i", "j": "ʒ", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "o", "p": "p", "r": "r", "s": "s",
+    "t": "t", "u": "u", "v": "b", "w": "w", "x": "ʃ", "y": "j",
+    "z": "z",
+}
+
+# Galician (gl): Spanish-adjacent; x = /ʃ/, c+e/i = /θ/
+GL_RULES = {
+    "ch": "tʃ", "ll": "ʎ", "nh": "ŋ", "qu": "k", "gu": "ɡ",
+    "ce": "θe", "ci": "θi", "rr": "r",
+    "á": "ˈa", "é": "ˈe", "í": "ˈi", "ó": "ˈo", "ú": "ˈu",
+    "ñ": "ɲ", "ü": "u",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "", "i": "i", "j": "x", "l": "l", "m": "m",
+    "n": "n", "o": "o", "p": "p", "r": "ɾ", "s": "s", "t": "t",
+    "u": "u", "v": "b", "x": "ʃ", "z": "θ",
+}
+
+# Basque (eu): Batua; j = /x/ (widespread), z/s sibilant merger approx
+EU_RULES = {
+    "tx": "tʃ", "ts": "ts", "tz": "ts", "ll": "ʎ", "rr": "r",
+    "ñ": "ɲ", "x": "ʃ",
+    "a": "a", "b": "b", "d": "d", "e": "e", "f": "f", "g": "ɡ",
+    "h": "", "i": "i", "j": "x", "k": "k", "l": "l", "m": "m",
+    "n": "n", "o": "o", "p": "p", "r": "ɾ", "s": "s", "t": "t",
+    "u": "u", "z": "s",
+}
+
+# Azerbaijani (az): Turkish-like Latin; final stress
+AZ_RULES = {
+    "ç": "tʃ", "ş": "ʃ", "ğ": "ɣ", "ə": "æ", "ı": "ɯ", "ö": "ø",
+    "ü": "y", "c": "dʒ", "j": "ʒ", "q": "ɡ", "x": "x",
+    "a": "ɑ", "b": "b", "d": "d", "e": "e", "f": "f", "g": "ɟ",
+    "h": "h", "i": "i", "k": "k", "l": "l", "m": "m", "n": "n",
+    "o": "o", "p": "p", "r": "r", "s": "s", "t": "t", "u": "u",
+    "v": "v", "y": "j", "z": "z",
+}
+
+# Kazakh (kk, Cyrillic): vowel harmony language, final stress
+KK_RULES = {
+    "ә": "æ", "ғ": "ʁ", "қ": "q", "ң": "ŋ", "ө": "ø", "ұ": "ʊ",
+    "ү": "y", "һ": "h", "і": "ɪ", "ы": "ə", "ё": "jo", "ю": "ju",
+    "я": "ja", "э": "e", "щ": "ʃ", "ъ": "", "ь": "",
+    "а": "ɑ", "б": "b", "в": "v", "г": "ɡ", "д": "d", "е": "e",
+    "ж": "ʒ", "з": "z", "и": "i", "й": "j", "к": "k", "л": "l",
+    "м": "m", "н": "n", "о": "o", "п": "p", "р": "r", "с": "s",
+    "т": "t", "у": "w", "ф": "f", "х": "x", "ц": "ts", "ч": "tʃ",
+    "ш": "ʃ",
+}
+
+# Kyrgyz (ky, Cyrillic)
+KY_RULES = {
+    "ң": "ŋ", "ө": "ø", "ү": "y", "ё": "jo", "ю": "ju", "я": "ja",
+    "э": "e", "щ": "ʃ", "ъ": "", "ь": "", "ы": "ɯ",
+    "а": "ɑ", "б": "b", "в": "v", "г": "ɡ", "д": "d", "е": "e",
+    "ж": "dʒ", "з": "z", "и": "i", "й": "j", "к": "k", "л": "l",
+    "м": "m", "н": "n", "о": "o", "п": "p", "р": "r", "с": "s",
+    "т": "t", "у": "u", "ф": "f", "х": "x", "ц": "ts", "ч": "tʃ",
+    "ш": "ʃ",
+}
+
+# Uzbek (uz, Latin)
+UZ_RULES = {
+    "oʻ": "o", "o'": "o", "gʻ": "ʁ", "g'": "ʁ", "sh": "ʃ",
+    "ch": "tʃ", "ng": "ŋ",
+    "a": "a", "b": "b", "d": "d", "e": "e", "f": "f", "g": "ɡ",
+    "h": "h", "i": "i", "j": "dʒ", "k": "k", "l": "l", "m": "m",
+    "n": "n", "o": "ɒ", "p": "p", "q": "q", "r": "r", "s": "s",
+    "t": "t", "u": "u", "v": "v", "x": "x", "y": "j", "z": "z",
+}
+
+# Macedonian (mk, Cyrillic): fixed antepenultimate stress
+MK_RULES = {
+    "џ": "dʒ", "ѕ": "dz", "љ": "ʎ", "њ": "ɲ", "ѓ": "ɟ", "ќ": "c",
+    "а": "a", "б": "b", "в": "v", "г": "ɡ", "д": "d", "е": "ɛ",
+    "ж": "ʒ", "з": "z", "и": "i", "ј": "j", "к": "k", "л": "l",
+    "м": "m", "н": "n", "о": "ɔ", "п": "p", "р": "r", "с": "s",
+    "т": "t", "у": "u", "ф": "f", "х": "x", "ц": "ts", "ч": "tʃ",
+    "ш": "ʃ",
+}
+
+# Belarusian (be, Cyrillic): akanne is written, so rules stay simple
+BE_RULES = {
+    "дж": "dʒ", "дз": "dz", "ьо": "ʲo",
+    "а": "a", "б": "b", "в": "v", "г": "ɦ", "д": "d", "е": "jɛ",
+    "ё": "jo", "ж": "ʒ", "з": "z", "і": "i", "й": "j", "к": "k",
+    "л": "l", "м": "m", "н": "n", "о": "o", "п": "p", "р": "r",
+    "с": "s", "т": "t", "у": "u", "ў": "w", "ф": "f", "х": "x",
+    "ц": "ts", "ч": "tʃ", "ш": "ʃ", "ы": "ɨ", "ь": "ʲ", "э": "ɛ",
+    "ю": "ju", "я": "ja", "'": "",
+}
+
+# Slovenian (sl)
+SL_RULES = {
+    "č": "tʃ", "š": "ʃ", "ž": "ʒ", "dž": "dʒ",
+    "a": "a", "b": "b", "c": "ts", "d": "d", "e": "ɛ", "f": "f",
+    "g": "ɡ", "h": "x", "i": "i", "j": "j", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "ɔ", "p": "p", "r": "r", "s": "s",
+    "t": "t", "u": "u", "v": "ʋ", "z": "z",
+}
+
+# Lithuanian (lt): mobile stress approximated word-initial
+LT_RULES = {
+    "ch": "x", "dž": "dʒ", "dz": "dz",
+    "ą": "aː", "ę": "ɛː", "ė": "eː", "į": "iː", "ų": "uː",
+    "ū": "uː", "č": "tʃ", "š": "ʃ", "ž": "ʒ",
+    "a": "a", "b": "b", "c": "ts", "d": "d", "e": "ɛ", "f": "f",
+    "g": "ɡ", "h": "ɣ", "i": "ɪ", "j": "j", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "oː", "p": "p", "r": "r", "s": "s",
+    "t": "t", "u": "ʊ", "v": "ʋ", "y": "iː", "z": "z",
+}
+
+# Latvian (lv): fixed initial stress
+LV_RULES = {
+    "dz": "dz", "dž": "dʒ",
+    "ā": "aː", "ē": "ɛː", "ī": "iː", "ū": "uː", "č": "tʃ",
+    "š": "ʃ", "ž": "ʒ", "ģ": "ɟ", "ķ": "c", "ļ": "ʎ", "ņ": "ɲ",
+    "a": "a", "b": "b", "c": "ts", "d": "d", "e": "ɛ", "f": "f",
+    "g": "ɡ", "h": "x", "i": "i", "j": "j", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "uɔ", "p": "p", "r": "r", "s": "s",
+    "t": "t", "u": "u", "v": "v", "z": "z",
+}
+
+# Estonian (et): fixed initial stress; õ = /ɤ/
+ET_RULES = {
+    "aa": "ɑː", "ee": "eː", "ii": "iː", "oo": "oː", "uu": "uː",
+    "õõ": "ɤː", "ää": "æː", "öö": "øː", "üü": "yː",
+    "õ": "ɤ", "ä": "æ", "ö": "ø", "ü": "y", "š": "ʃ", "ž": "ʒ",
+    "a": "ɑ", "b": "b", "d": "d", "e": "e", "f": "f", "g": "ɡ",
+    "h": "h", "i": "i", "j": "j", "k": "k", "l": "l", "m": "m",
+    "n": "n", "o": "o", "p": "p", "r": "r", "s": "s", "t": "t",
+    "u": "u", "v": "v",
+}
+
+# Icelandic (is): fixed initial stress; key digraph approximations
+IS_RULES = {
+    "hv": "kv", "ll": "tl", "ei": "ei", "ey": "ei", "au": "øy",
+    "pp": "ʰp", "tt": "ʰt", "kk": "ʰk",  # preaspirated geminates
+    "þ": "θ", "ð": "ð", "æ": "ai", "á": "au", "é": "jɛ",
+    "í": "i", "ó": "ou", "ú": "u", "ý": "i", "ö": "œ",
+    "a": "a", "b": "p", "d": "t", "e": "ɛ", "f": "f", "g": "k",
+    "h": "h", "i": "ɪ", "j": "j", "k": "kʰ", "l": "l", "m": "m",
+    "n": "n", "o": "ɔ", "p": "pʰ", "r": "r", "s": "s", "t": "tʰ",
+    "u": "ʏ", "v": "v", "x": "ks", "y": "ɪ", "z": "s",
+}
+
+# Albanian (sq): near-regular; penult stress
+SQ_RULES = {
+    "dh": "ð", "gj": "ɟ", "ll": "l", "nj": "ɲ", "rr": "r",
+    "sh": "ʃ", "th": "θ", "xh": "dʒ", "zh": "ʒ",
+    "ç": "tʃ", "ë": "ə",
+    "a": "a", "b": "b", "c": "ts", "d": "d", "e": "ɛ", "f": "f",
+    "g": "ɡ", "h": "h", "i": "i", "j": "j", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "ɔ", "p": "p", "q": "c", "r": "ɾ",
+    "s": "s", "t": "t", "u": "u", "v": "v", "x": "dz", "y": "y",
+    "z": "z",
+}
+
+# Armenian (hy, Eastern): phonemic script; final stress
+HY_RULES = {
+    "ու": "u", "և": "ɛv",
+    "ա": "ɑ", "բ": "b", "գ": "ɡ", "դ": "d", "ե": "ɛ", "զ": "z",
+    "է": "ɛ", "ը": "ə", "թ": "tʰ", "ժ": "ʒ", "ի": "i", "լ": "l",
+    "խ": "x", "ծ": "ts", "կ": "k", "հ": "h", "ձ": "dz", "ղ": "ʁ",
+    "ճ": "tʃ", "մ": "m", "յ": "j", "ն": "n", "շ": "ʃ", "ո": "o",
+    "չ": "tʃ", "պ": "p", "ջ": "dʒ", "ռ": "r", "ս": "s", "վ": "v",
+    "տ": "t", "ր": "ɾ", "ց": "ts", "ւ": "v", "փ": "pʰ", "ք": "kʰ",
+    "օ": "o", "ֆ": "f",
+}
+
+# Georgian (ka): phonemic script; ejectives approximated plain
+KA_RULES = {
+    "ა": "ɑ", "ბ": "b", "გ": "ɡ", "დ": "d", "ე": "ɛ", "ვ": "v",
+    "ზ": "z", "თ": "tʰ", "ი": "i", "კ": "k", "ლ": "l", "მ": "m",
+    "ნ": "n", "ო": "ɔ", "პ": "p", "ჟ": "ʒ", "რ": "r", "ს": "s",
+    "ტ": "t", "უ": "u", "ფ": "pʰ", "ქ": "kʰ", "ღ": "ʁ", "ყ": "q",
+    "შ": "ʃ", "ჩ": "tʃ", "ც": "ts", "ძ": "dz", "წ": "ts",
+    "ჭ": "tʃ", "ხ": "x", "ჯ": "dʒ", "ჰ": "h",
+}
+
+# Afrikaans (af): Dutch-derived, g = /x/, v = /f/, w = /v/
+AF_RULES = {
+    "oei": "ui", "eeu": "iu", "aai": "aːi",
+    "aa": "aː", "ee": "eː", "oo": "oː", "uu": "yː", "oe": "u",
+    "ie": "i", "eu": "øː", "ui": "œy", "ou": "əu", "ei": "ei",
+    "y": "ei", "ng": "ŋ", "sj": "ʃ", "tj": "tʃ", "dj": "dʒ",
+    "ê": "ɛː", "ô": "ɔː", "ë": "e",
+    "a": "a", "b": "b", "d": "d", "e": "ɛ", "f": "f", "g": "x",
+    "h": "ɦ", "i": "ə", "j": "j", "k": "k", "l": "l", "m": "m",
+    "n": "n", "o": "ɔ", "p": "p", "r": "r", "s": "s", "t": "t",
+    "u": "œ", "v": "f", "w": "v", "z": "z",
+}
+
+# Welsh (cy): ll = /ɬ/; penult stress
+CY_RULES = {
+    "ll": "ɬ", "dd": "ð", "ff": "f", "ph": "f", "th": "θ",
+    "ch": "x", "rh": "r", "si": "ʃ", "ngh": "ŋh", "ng": "ŋ",
+    "â": "aː", "ê": "eː", "î": "iː", "ô": "oː", "û": "iː",
+    "ŵ": "uː", "ŷ": "iː",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "ɛ", "f": "v",
+    "g": "ɡ", "h": "h", "i": "i", "j": "dʒ", "l": "l", "m": "m",
+    "n": "n", "o": "ɔ", "p": "p", "r": "r", "s": "s", "t": "t",
+    "u": "i", "w": "u", "y": "ə",
+}
+
+# Maltese (mt): Semitic with Latin script; għ silent, q = /ʔ/
+MT_RULES = {
+    "għ": "", "ie": "iː", "ċ": "tʃ", "ġ": "dʒ", "ħ": "ħ",
+    "ż": "z", "x": "ʃ", "q": "ʔ", "z": "ts",
+    "a": "a", "b": "b", "d": "d", "e": "ɛ", "f": "f", "g": "ɡ",
+    "h": "", "i": "i", "j": "j", "k": "k", "l": "l", "m": "m",
+    "n": "n", "o": "ɔ", "p": "p", "r": "r", "s": "s", "t": "t",
+    "u": "u", "v": "v", "w": "w",
+}
+
+# Haitian Creole (ht): regular French-derived orthography; final stress
+HT_RULES = {
+    "tch": "tʃ", "dj": "dʒ", "ou": "u", "an": "ã", "en": "ɛ̃",
+    "on": "ɔ̃", "ui": "wi", "ch": "ʃ", "ng": "ŋ",
+    "è": "ɛ", "ò": "ɔ",
+    "a": "a", "b": "b", "d": "d", "e": "e", "f": "f", "g": "ɡ",
+    "h": "h", "i": "i", "j": "ʒ", "k": "k", "l": "l", "m": "m",
+    "n": "n", "o": "o", "p": "p", "r": "ʁ", "s": "s", "t": "t",
+    "v": "v", "w": "w", "y": "j", "z": "z",
+}
+
+# Latin (la, classical-leaning): penult stress approximation
+LA_RULES = {
+    "ae": "ai", "oe": "oi", "au": "au", "qu": "kw", "gn": "ŋn",
+    "ph": "f", "th": "t", "ch": "k", "x": "ks",
+    "ā": "aː", "ē": "eː", "ī": "iː", "ō": "oː", "ū": "uː",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "h", "i": "i", "j": "j", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "o", "p": "p", "r": "r", "s": "s",
+    "t": "t", "u": "u", "v": "w", "y": "y", "z": "z",
+}
+
+# Hindi (hi, Devanagari): inherent schwa + final schwa deletion handled
+# in hi_preprocess-free form: the virama and matras drive vowels; the
+# inherent /ə/ is inserted after bare consonants by hi_expand below.
+_HI_CONS = {
+    "क": "k", "ख": "kʰ", "ग": "ɡ", "घ": "ɡʰ", "ङ": "ŋ",
+    "च": "tʃ", "छ": "tʃʰ", "ज": "dʒ", "झ": "dʒʰ", "ञ": "ɲ",
+    "ट": "ʈ", "ठ": "ʈʰ", "ड": "ɖ", "ढ": "ɖʰ", "ण": "ɳ",
+    "त": "t", "थ": "tʰ", "द": "d", "ध": "dʰ", "न": "n",
+    "प": "p", "फ": "pʰ", "ब": "b", "भ": "bʰ", "म": "m",
+    "य": "j", "र": "r", "ल": "l", "व": "ʋ", "श": "ʃ",
+    "ष": "ʂ", "स": "s", "ह": "h", "ळ": "l", "क़": "q",
+    "ख़": "x", "ग़": "ɣ", "ज़": "z", "ड़": "ɾ", "ढ़": "ɾʰ",
+    "फ़": "f", "य़": "j",
+}
+_HI_VOWELS = {
+    "अ": "ə", "आ": "ɑː", "इ": "ɪ", "ई": "iː", "उ": "ʊ",
+    "ऊ": "uː", "ऋ": "rɪ", "ए": "eː", "ऐ": "ɛː", "ओ": "oː",
+    "औ": "ɔː",
+}
+_HI_MATRAS = {
+    "ा": "ɑː", "ि": "ɪ", "ी": "iː", "ु": "ʊ", "ू": "uː",
+    "ृ": "rɪ", "े": "eː", "ै": "ɛː", "ो": "oː", "ौ": "ɔː",
+}
+
+
+def hi_word_to_ipa(w: str) -> str:
+    """Devanagari -> IPA with inherent-schwa insertion and word-final
+    schwa deletion (the standard Hindi rule)."""
+    out = []
+    chars = list(w)
+    i = 0
+    n = len(chars)
+    while i < n:
+        ch = chars[i]
+        if ch in _HI_CONS:
+            out.append(_HI_CONS[ch])
+            nxt = chars[i + 1] if i + 1 < n else None
+            if nxt in _HI_MATRAS:
+                out.append(_HI_MATRAS[nxt])
+                i += 2
+                continue
+            if nxt == "्":  # virama: no vowel
+                i += 2
+                continue
+            # inherent schwa, deleted word-finally
+            if i + 1 < n:
+                out.append("ə")
+            i += 1
+        elif ch in _HI_VOWELS:
+            out.append(_HI_VOWELS[ch])
+            i += 1
+        elif ch == "ं" or ch == "ँ":  # anusvara/chandrabindu: nasal
+            out.append("n")
+            i += 1
+        elif ch == "ः":
+            out.append("h")
+            i += 1
+        else:
+            i += 1
+    return "".join(out)
+
+
+TABLES2 = {
+    "eo": EO_RULES, "ca": CA_RULES, "gl": GL_RULES, "eu": EU_RULES,
+    "az": AZ_RULES, "kk": KK_RULES, "ky": KY_RULES, "uz": UZ_RULES,
+    "mk": MK_RULES, "be": BE_RULES, "sl": SL_RULES, "lt": LT_RULES,
+    "lv": LV_RULES, "et": ET_RULES, "is": IS_RULES, "sq": SQ_RULES,
+    "hy": HY_RULES, "ka": KA_RULES, "af": AF_RULES, "cy": CY_RULES,
+    "mt": MT_RULES, "ht": HT_RULES, "la": LA_RULES,
+}
+
+LETTERS2 = {
+    "eo": "a-zĉĝĥĵŝŭA-ZĈĜĤĴŜŬ", "ca": "a-zA-Zàèéíòóúïüç·",
+    "gl": "a-zA-Záéíóúñü", "eu": "a-zA-Zñ",
+    "az": "a-zA-Zçəğıöşüİ", "kk": "а-яА-ЯәғқңөұүһіӘҒҚҢӨҰҮҺІ",
+    "ky": "а-яА-ЯңөүҢӨҮ", "uz": "a-zA-Z'ʻ",
+    "mk": "а-шА-Шџѕљњѓќј", "be": "а-яА-Яёіўэюя'ЁІЎ",
+    "sl": "a-zA-Zčšž", "lt": "a-zA-Ząčęėįšųūž",
+    "lv": "a-zA-Zāčēģīķļņšūž", "et": "a-zA-Zõäöüšž",
+    "is": "a-zA-Zþðæáéíóúýö", "sq": "a-zA-Zçë",
+    "hy": "ա-ֆԱ-Ֆ", "ka": "ა-ჰ", "af": "a-zA-Zêôë",
+    "cy": "a-zA-Zâêîôûŵŷ", "mt": "a-zA-Zċġħż",
+    "ht": "a-zA-Zèò", "la": "a-zA-Zāēīōū",
+}
+
+# stress placement per second-batch language (RuleG2P stress_default)
+STRESS_DEFAULTS = {
+    "eo": "penult", "ca": "es-penult", "gl": "es-penult",
+    "eu": "penult", "az": "final", "kk": "final", "ky": "final",
+    "uz": "final", "mk": "antepenult", "be": "first", "sl": "penult",
+    "lt": "first", "lv": "first", "et": "first", "is": "first",
+    "sq": "penult", "hy": "final", "ka": "first", "af": "first",
+    "cy": "penult", "mt": "penult", "ht": "final", "la": "penult",
+}
+
 TABLES = {
     "sv": SV_RULES, "no": NO_RULES, "da": DA_RULES, "fi": FI_RULES,
     "hu": HU_RULES, "ro": RO_RULES, "el": EL_RULES, "bg": BG_RULES,
@@ -208,7 +564,7 @@ TABLES = {
     "sw": SW_RULES,
 }
 # sr (Serbian latin) shares the hr table; nb/nn map to no
-ALIASES = {"sr": "hr", "nb": "no", "nn": "no", "ms": "id"}
+ALIASES = {"sr": "hr", "bs": "hr", "nb": "no", "nn": "no", "ms": "id"}
 
 
 # --------------------------------------------------------------------- #

@@ -29,6 +29,8 @@ _IPA_SYMBOLS = (
     # quality layers): appended so existing ids stay stable
     "ɕɟɤɥɦɧɨɯʂʉʋʎʐʑ",
     "̝̃",  # combining tilde (nasal) + raised diacritics
+    # second expansion batch (cy/hi): lateral fricative + retroflexes
+    "ɬʈɖɳ",
 )
 
 

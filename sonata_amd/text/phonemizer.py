@@ -92,7 +92,7 @@ class RuleG2P:
         self._max_pat = max((len(p) for p in rules), default=1)
         self._word_re = re.compile(rf"[{letters}']+", re.IGNORECASE)
 
-    _VOWELS = "aeiouɑæʌɔəɛɪiʊuɜɚɝoʏøyɶɒãõɐ"
+    _VOWELS = "aeiouɑæʌɔəɛɪiʊuɜɚɝoʏøyɶɒãõɐɯɤɨʉœ"
     _SIBILANT_END = ("s", "z", "ʃ", "ʒ", "tʃ", "dʒ")
     _VOICELESS_END = ("p", "t", "k", "f", "θ")
 
@@ -233,6 +233,12 @@ class RuleG2P:
                     idx = len(starts) - 2
                 else:
                     idx = len(starts) - 1
+            elif self.stress_default == "penult":
+                idx = max(len(starts) - 2, 0)
+            elif self.stress_default == "antepenult":
+                idx = max(len(starts) - 3, 0)
+            elif self.stress_default == "final":
+                idx = len(starts) - 1
             elif len(starts) >= 4:
                 idx = 1
         pos = starts[min(idx, len(starts) - 1)]
@@ -702,13 +708,24 @@ def _get_g2p(voice: str) -> RuleG2P:
             letters="؀-ۿ",
             stress=False,
         )
+    elif base == "hi":
+        # Devanagari needs abugida handling (inherent schwa, matras,
+        # virama) that a flat rule table cannot express.
+        from .g2p_tables import hi_word_to_ipa
+
+        g = RuleG2P({}, letters="ऀ-ॿ", stress=False)
+        g._apply_rules = hi_word_to_ipa  # plain function, no self
     else:
-        # round-2 expansion tables (g2p_tables.py): 13 more languages
-        from .g2p_tables import ALIASES, LETTERS, TABLES
+        # expansion tables (g2p_tables.py): 13 + 23 more languages
+        from .g2p_tables import (ALIASES, LETTERS, LETTERS2,
+                                 STRESS_DEFAULTS, TABLES, TABLES2)
 
         key2 = ALIASES.get(base, base)
         if key2 in TABLES:
             g = RuleG2P(TABLES[key2], letters=LETTERS[key2])
+        elif key2 in TABLES2:
+            g = RuleG2P(TABLES2[key2], letters=LETTERS2[key2],
+                        stress_default=STRESS_DEFAULTS[key2])
         else:
             raise PhonemizationError(
                 f"unsupported phonemizer language: {voice!r}")
@@ -721,7 +738,11 @@ def available_languages() -> List[str]:
             "ru", "tr", "cs", "ar",
             # round-2 expansion (g2p_tables.py)
             "sv", "no", "nb", "nn", "da", "fi", "hu", "ro", "el", "bg",
-            "uk", "hr", "sr", "sk", "id", "ms", "sw"]
+            "uk", "hr", "sr", "bs", "sk", "id", "ms", "sw",
+            # second expansion batch (g2p_tables.py TABLES2 + hi)
+            "eo", "ca", "gl", "eu", "az", "kk", "ky", "uz", "mk", "be",
+            "sl", "lt", "lv", "et", "is", "sq", "hy", "ka", "af", "cy",
+            "mt", "ht", "la", "hi"]
 
 
 _LANG_SWITCH_RE = re.compile(r"\([a-z-]{2,10}\)")

@@ -86,7 +86,20 @@ def test_all_languages_end_to_end(tmp_path):
         "bg": "Здравей свят.", "uk": "Привіт світе.",
         "hr": "Pozdrav svijete.", "sr": "Pozdrav svete.",
         "sk": "Ahoj svet.", "id": "Halo dunia.", "ms": "Halo dunia.",
-        "sw": "Habari dunia.",
+        "sw": "Habari dunia.", "bs": "Pozdrav svijete.",
+        # second expansion batch
+        "eo": "Saluton mondo.", "ca": "Hola món.", "gl": "Ola mundo.",
+        "eu": "Kaixo mundua.", "az": "Salam dünya.",
+        "kk": "Сәлем әлем.", "ky": "Салам дүйнө.",
+        "uz": "Salom dunyo.", "mk": "Здраво свету.",
+        "be": "Прывітанне свет.", "sl": "Pozdravljen svet.",
+        "lt": "Labas pasauli.", "lv": "Sveika pasaule.",
+        "et": "Tere maailm.", "is": "Halló heimur.",
+        "sq": "Përshëndetje botë.", "hy": "Բարև աշխարհ.",
+        "ka": "გამარჯობა მსოფლიო.", "af": "Hallo wêreld.",
+        "cy": "Helo byd.", "mt": "Bongu dinja.",
+        "ht": "Bonjou monn.", "la": "Salve munde.",
+        "hi": "नमस्ते दुनिया.",
     }
     for lang in available_languages():
         pack = create_random_voice(str(tmp_path), f"lang_{lang}",

@@ -340,3 +340,81 @@ def test_pt_stress_and_reduction():
     wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
              if g.word_to_ipa(w) != want]
     assert len(wrong) <= 1, wrong
+
+
+# ---- second expansion batch (g2p_tables.py TABLES2 + hi) --------------- #
+def test_expanded_languages_batch2_smoke():
+    """23 more rule-table languages + Hindi: load-bearing digraphs map
+    and exactly one primary stress per content word."""
+    from sonata_amd.text.phonemizer import text_to_phonemes
+
+    checks = [
+        ("eo", "ĉambro", "tʃ"), ("eo", "ŝipo", "ʃ"),
+        ("ca", "cançó", "s"), ("ca", "llibre", "ʎ"),
+        ("gl", "xente", "ʃ"), ("gl", "cidade", "θi"),
+        ("eu", "etxea", "tʃ"), ("eu", "jan", "x"),
+        ("az", "çörək", "tʃ"), ("az", "yaxşı", "ʃ"),
+        ("kk", "қазақ", "q"), ("kk", "сәлем", "æ"),
+        ("ky", "кыргыз", "ɯ"), ("uz", "o'zbek", "o"),
+        ("uz", "yaxshi", "ʃ"), ("mk", "џеб", "dʒ"),
+        ("be", "мова", "v"), ("be", "ўсё", "w"),
+        ("sl", "človek", "tʃ"), ("lt", "ačiū", "tʃ"),
+        ("lv", "paldies", "d"), ("et", "tänan", "æ"),
+        ("is", "þakka", "θ"), ("is", "hvað", "kv"),
+        ("sq", "shqip", "ʃc"), ("hy", "շնորհակալ", "ʃ"),
+        ("ka", "მადლობა", "dl"), ("af", "goeie", "x"),
+        ("cy", "llyfr", "ɬ"), ("cy", "bedd", "ð"),
+        ("mt", "ħobż", "ħ"), ("ht", "bonjou", "ɔ̃"),
+        ("la", "quattuor", "kw"),
+    ]
+    for lang, word, frag in checks:
+        out = text_to_phonemes(word, voice=lang)[0]
+        assert frag in out.replace("ˈ", ""), (lang, word, out, frag)
+        assert out.count("ˈ") == 1, (lang, word, out)
+
+
+def test_batch2_stress_modes():
+    """penult / antepenult / final stress_default place primary stress
+    on the right vowel cluster."""
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    # Macedonian: fixed antepenultimate
+    assert _get_g2p("mk").word_to_ipa("планина") == "plˈanina"
+    # Armenian: final stress
+    ipa = _get_g2p("hy").word_to_ipa("հայերեն")
+    assert ipa.endswith("ɾˈɛn"), ipa
+    # Esperanto: penultimate
+    assert _get_g2p("eo").word_to_ipa("esperanto") == "esperˈanto"
+    # Azerbaijani: final
+    ipa = _get_g2p("az").word_to_ipa("kitablar")
+    assert ipa.endswith("lˈɑr"), ipa
+
+
+def test_hindi_schwa_rules():
+    """Devanagari: inherent schwa inserted medially, deleted finally;
+    matras and virama override it."""
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    g = _get_g2p("hi")
+    assert g.word_to_ipa("कमल") == "kəməl"      # ka-ma-l(a): final deleted
+    assert g.word_to_ipa("हिंदी") == "hɪndiː"   # matra + anusvara
+    assert g.word_to_ipa("नमस्ते") == "nəməsteː"  # virama joins s-t
+    assert g.word_to_ipa("बोलता") == "boːlətɑː"
+
+
+def test_batch2_symbol_coverage():
+    """Every phoneme emitted for batch-2 sample sentences is in the
+    default id map (no silently-dropped symbols)."""
+    from sonata_amd.text.ids import default_phoneme_id_map
+    from sonata_amd.text.phonemizer import text_to_phonemes
+
+    m = default_phoneme_id_map()
+    sents = {
+        "cy": "dw i'n siarad cymraeg llan", "hi": "मैं हिंदी बोलता हूँ",
+        "ht": "mwen pale kreyòl", "ka": "მე ვლაპარაკობ ქართულად",
+        "is": "ég tala íslensku", "mt": "jien nitkellem bil-malti",
+    }
+    for lang, txt in sents.items():
+        ph = text_to_phonemes(txt, voice=lang)[0]
+        missing = {c for c in ph if c not in m and c not in " ˈˌ"}
+        assert not missing, (lang, ph, missing)
