@@ -711,3 +711,128 @@ DE_LEXICON = {
     "deutschland": "dˈɔʏtʃlant", "deutsch": "dɔʏtʃ", "hallo": "halˈoː",
     "danke": "dˈaŋkə", "bitte": "bˈɪtə",
 }
+
+
+# --------------------------------------------------------------------- #
+# Russian quality layer: palatalization digraphs, a frequent-word
+# stressed lexicon (stress is lexical in Russian — the rule default
+# cannot guess it), akanye/ikanye vowel reduction and final devoicing.
+# Parity: the reference's espeak-ng ru_dict carries per-word stress;
+# this is the same idea at smaller scale.
+# --------------------------------------------------------------------- #
+
+def ru_palatal_rules(base_rules):
+    """Generate consonant+front-vowel digraphs: те -> tʲe etc.
+    (context-free single-letter rules cannot express palatalization)."""
+    out = dict(base_rules)
+    cons = {"б": "b", "в": "v", "г": "ɡ", "д": "d", "з": "z",
+            "к": "k", "л": "l", "м": "m", "н": "n", "п": "p",
+            "р": "r", "с": "s", "т": "t", "ф": "f", "х": "x"}
+    soft = {"е": "e", "ё": "o", "ю": "u", "я": "a", "и": "i"}
+    for c, ci in cons.items():
+        for v, vi in soft.items():
+            out[c + v] = ci + "ʲ" + vi
+        out[c + "ь"] = ci + "ʲ"
+    # hushers are inherently hard/soft: no ʲ, no spurious j from е/ё/ю/я
+    hush = {"ч": "tɕ", "щ": "ɕː", "ж": "ʐ", "ш": "ʂ", "ц": "ts"}
+    hard = {"ж", "ш", "ц"}  # жи/ши/ци -> ɨ
+    for c, ci in hush.items():
+        out[c + "е"] = ci + ("ɛ" if c in hard else "e")
+        out[c + "и"] = ci + ("ɨ" if c in hard else "i")
+        out[c + "ё"] = ci + "o"
+        out[c + "ю"] = ci + "u"
+        out[c + "я"] = ci + "a"
+        out[c + "ь"] = ci
+    return out
+
+
+# word -> 1-based stressed vowel-cluster index (frequent words whose
+# stress the first-syllable default gets wrong, plus common anchors)
+RU_STRESS = {
+    "она": 2, "оно": 2, "они": 2, "меня": 2, "тебя": 2, "себя": 2,
+    "её": 2, "тебе": 2, "себе": 2, "ему": 2, "мою": 2, "моя": 2,
+    "твоя": 2, "свою": 2, "когда": 2, "тогда": 2, "потом": 2,
+    "почему": 3, "потому": 3, "сейчас": 2, "теперь": 2, "ещё": 2,
+    "уже": 2, "всегда": 2, "никогда": 3, "иногда": 3, "хорошо": 3,
+    "спасибо": 2, "привет": 2, "пока": 2,
+    "человек": 3, "язык": 2, "семья": 2, "страна": 2, "вода": 2,
+    "земля": 2, "рука": 2, "нога": 2, "голова": 3, "глаза": 2,
+    "окно": 2, "вопрос": 2, "ответ": 2, "работа": 2, "москва": 2,
+    "россия": 2, "молоко": 3, "мужчина": 2, "ребёнок": 2,
+    "девушка": 1, "женщина": 1, "деньги": 1, "люди": 1,
+    "была": 2, "былo": 1, "иду": 2, "идёт": 2, "пошёл": 2,
+    "пришёл": 2, "говорить": 3, "говорю": 3, "говорит": 3,
+    "сказать": 2, "сказал": 2, "сказала": 2, "хотеть": 2, "хочу": 2,
+    "хочет": 1, "нельзя": 2, "видеть": 1, "вижу": 1, "смотреть": 2,
+    "смотрю": 2, "понимать": 3, "понимаю": 3, "любить": 2,
+    "люблю": 2, "любит": 1, "живу": 2, "живёт": 2, "работать": 2,
+    "работаю": 2, "стоять": 2, "сидеть": 2, "лежать": 2,
+    "прийти": 2, "уйти": 2, "найти": 2, "помочь": 2, "играть": 2,
+    "читать": 2, "писать": 2, "учить": 2, "купить": 2, "начать": 2,
+    "открыть": 2, "закрыть": 2, "использовать": 2,
+    "большой": 2, "хороший": 2, "плохой": 2, "красивый": 2,
+    "другой": 2, "второй": 2, "последний": 2, "который": 2,
+    "которая": 2, "чёрный": 1, "всё": 1,
+    "один": 2, "одна": 2, "четыре": 2, "тринадцать": 2,
+    "пятнадцать": 2, "двадцать": 1, "тридцать": 1, "пятьдесят": 3,
+    "опять": 2, "вместе": 1, "может": 1, "конечно": 2,
+    "любовь": 2, "отец": 2, "жена": 2, "сестра": 2, "число": 2,
+    "письмо": 2, "столы": 2, "цветы": 2, "часы": 2, "цена": 2,
+    "дела": 2, "дома": 1, "утром": 1,
+}
+
+# full-IPA overrides: orthography-irregular words (г -> в genitives,
+# что/конечно ч -> ш, silent clusters); stored in final reduced form
+RU_IPA = {
+    "что": "ʂtˈo", "чтобы": "ʂtˈobɨ", "конечно": "kɐnʲˈeʂnɐ",
+    "его": "jɪvˈo", "сегодня": "sʲɪvˈodnʲɐ", "ничего": "nʲitɕɪvˈo",
+    "чего": "tɕɪvˈo", "кого": "kɐvˈo", "того": "tɐvˈo",
+    "здравствуйте": "zdrˈastvujtʲɪ", "здравствуй": "zdrˈastvuj",
+    "пожалуйста": "pɐʐˈalustɐ", "солнце": "sˈontsɨ",
+    "счастье": "ɕːˈastʲe",
+}
+
+
+def ru_reduce(ipa: str) -> str:
+    """Akanye/ikanye + final obstruent devoicing on stressed IPA."""
+    out = []
+    for i, ch in enumerate(ipa):
+        stressed = i > 0 and ipa[i - 1] == "ˈ"
+        if ch == "o" and not stressed:
+            out.append("ɐ")
+        elif ch == "e" and not stressed:
+            out.append("ɪ")
+        else:
+            out.append(ch)
+    s = "".join(out)
+    # final devoicing (look through a trailing palatalization mark)
+    devoice = {"b": "p", "d": "t", "ɡ": "k", "v": "f", "z": "s",
+               "ʐ": "ʂ"}
+    j = len(s) - 1
+    while j >= 0 and s[j] == "ʲ":
+        j -= 1
+    if j >= 0 and s[j] in devoice:
+        s = s[:j] + devoice[s[j]] + s[j + 1:]
+    return s
+
+
+def ru_build_lexicon(apply_rules, vowels) -> dict:
+    """Materialize RU_STRESS into stressed-IPA lexicon entries using the
+    same rule table the OOV path uses (guarantees consistency)."""
+    lex = dict(RU_IPA)
+    for word, idx in RU_STRESS.items():
+        if word in lex:
+            continue
+        ipa = apply_rules(word)
+        starts = []
+        prev_v = False
+        for i, ch in enumerate(ipa):
+            v = ch in vowels
+            if v and not prev_v:
+                starts.append(i)
+            prev_v = v
+        if not starts:
+            continue
+        pos = starts[min(idx - 1, len(starts) - 1)]
+        lex[word] = ipa[:pos] + "ˈ" + ipa[pos:]
+    return lex

@@ -693,7 +693,12 @@ def _get_g2p(voice: str) -> RuleG2P:
                     stress_default="es-penult")
         g.postprocess = _pt_post
     elif base == "ru":
-        g = RuleG2P(_RU_RULES, letters="а-яА-ЯёЁ")
+        from .g2p_tables import (ru_build_lexicon, ru_palatal_rules,
+                                 ru_reduce)
+
+        g = RuleG2P(ru_palatal_rules(_RU_RULES), letters="а-яА-ЯёЁ")
+        g.lexicon = ru_build_lexicon(g._apply_rules, g._VOWELS)
+        g.postprocess = ru_reduce
     elif base == "nl":
         g = RuleG2P(_NL_RULES, letters="a-zA-Z")
     elif base == "pl":

@@ -418,3 +418,34 @@ def test_batch2_symbol_coverage():
         ph = text_to_phonemes(txt, voice=lang)[0]
         missing = {c for c in ph if c not in m and c not in " ˈˌ"}
         assert not missing, (lang, ph, missing)
+
+
+# ---- Russian quality layer (palatalization, stress lexicon, akanye) ---- #
+RU_CASES = [
+    ("привет", "prʲivʲˈet"), ("хорошо", "xɐrɐʂˈo"),
+    ("говорит", "ɡɐvɐrʲˈit"), ("молоко", "mɐlɐkˈo"),
+    ("что", "ʂtˈo"), ("конечно", "kɐnʲˈeʂnɐ"),
+    ("сегодня", "sʲɪvˈodnʲɐ"), ("человек", "tɕɪlɐvʲˈek"),
+    ("день", "dʲˈenʲ"), ("хлеб", "xlʲˈep"), ("друг", "drˈuk"),
+    ("москва", "mɐskvˈa"), ("она", "ɐnˈa"), ("вода", "vɐdˈa"),
+    ("жизнь", "ʐˈɨznʲ"), ("ещё", "jɪɕːˈo"), ("город", "ɡˈorɐt"),
+]
+
+
+def test_ru_quality_layer():
+    """Palatalized consonants before front vowels, lexical stress for
+    frequent words, akanye/ikanye reduction, final devoicing."""
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    g = _get_g2p("ru")
+    wrong = [(w, g.word_to_ipa(w), want) for w, want in RU_CASES
+             if g.word_to_ipa(w) != want]
+    assert not wrong, wrong
+
+
+def test_ru_oov_still_works():
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    g = _get_g2p("ru")
+    ipa = g.word_to_ipa("электрификация")
+    assert ipa and "ˈ" in ipa
