@@ -159,11 +159,252 @@ def normalize_en(text: str) -> str:
     return text
 
 
+
+
+# --------------------------------------------------------------------- #
+# Full cardinal grammars (parity: espeak TranslateNumber speaks real
+# number words in every language; digit-by-digit was the round-2
+# approximation for de/es/fr/it/pt — these replace it).
+# --------------------------------------------------------------------- #
+
+_DE_ONES = ["null", "eins", "zwei", "drei", "vier", "fünf", "sechs",
+            "sieben", "acht", "neun", "zehn", "elf", "zwölf",
+            "dreizehn", "vierzehn", "fünfzehn", "sechzehn",
+            "siebzehn", "achtzehn", "neunzehn"]
+_DE_TENS = ["", "", "zwanzig", "dreißig", "vierzig", "fünfzig",
+            "sechzig", "siebzig", "achtzig", "neunzig"]
+
+
+def num_to_words_de(n: int) -> str:
+    if n < 0:
+        return "minus " + num_to_words_de(-n)
+    if n < 20:
+        return _DE_ONES[n]
+    if n < 100:
+        t, r = divmod(n, 10)
+        if r == 0:
+            return _DE_TENS[t]
+        unit = "ein" if r == 1 else _DE_ONES[r]
+        return unit + "und" + _DE_TENS[t]
+    if n < 1000:
+        h, r = divmod(n, 100)
+        head = ("ein" if h == 1 else _DE_ONES[h]) + "hundert"
+        return head if r == 0 else head + num_to_words_de(r)
+    if n < 10 ** 6:
+        t, r = divmod(n, 1000)
+        head = ("ein" if t == 1 else num_to_words_de(t)) + "tausend"
+        return head if r == 0 else head + num_to_words_de(r)
+    if n < 10 ** 9:
+        m, r = divmod(n, 10 ** 6)
+        head = ("eine Million" if m == 1
+                else num_to_words_de(m) + " Millionen")
+        return head if r == 0 else head + " " + num_to_words_de(r)
+    m, r = divmod(n, 10 ** 9)
+    head = ("eine Milliarde" if m == 1
+            else num_to_words_de(m) + " Milliarden")
+    return head if r == 0 else head + " " + num_to_words_de(r)
+
+
+_ES_ONES = ["cero", "uno", "dos", "tres", "cuatro", "cinco", "seis",
+            "siete", "ocho", "nueve", "diez", "once", "doce", "trece",
+            "catorce", "quince", "dieciséis", "diecisiete",
+            "dieciocho", "diecinueve", "veinte", "veintiuno",
+            "veintidós", "veintitrés", "veinticuatro", "veinticinco",
+            "veintiséis", "veintisiete", "veintiocho", "veintinueve"]
+_ES_TENS = ["", "", "veinte", "treinta", "cuarenta", "cincuenta",
+            "sesenta", "setenta", "ochenta", "noventa"]
+_ES_HUNDREDS = ["", "ciento", "doscientos", "trescientos",
+                "cuatrocientos", "quinientos", "seiscientos",
+                "setecientos", "ochocientos", "novecientos"]
+
+
+def num_to_words_es(n: int) -> str:
+    if n < 0:
+        return "menos " + num_to_words_es(-n)
+    if n < 30:
+        return _ES_ONES[n]
+    if n < 100:
+        t, r = divmod(n, 10)
+        return _ES_TENS[t] + ("" if r == 0 else " y " + _ES_ONES[r])
+    if n == 100:
+        return "cien"
+    if n < 1000:
+        h, r = divmod(n, 100)
+        return _ES_HUNDREDS[h] + ("" if r == 0
+                                  else " " + num_to_words_es(r))
+    if n < 10 ** 6:
+        t, r = divmod(n, 1000)
+        head = "mil" if t == 1 else num_to_words_es(t) + " mil"
+        return head if r == 0 else head + " " + num_to_words_es(r)
+    m, r = divmod(n, 10 ** 6)
+    head = ("un millón" if m == 1
+            else num_to_words_es(m) + " millones")
+    return head if r == 0 else head + " " + num_to_words_es(r)
+
+
+_FR_ONES = ["zéro", "un", "deux", "trois", "quatre", "cinq", "six",
+            "sept", "huit", "neuf", "dix", "onze", "douze", "treize",
+            "quatorze", "quinze", "seize", "dix-sept", "dix-huit",
+            "dix-neuf"]
+_FR_TENS = ["", "", "vingt", "trente", "quarante", "cinquante",
+            "soixante"]
+
+
+def num_to_words_fr(n: int) -> str:
+    if n < 0:
+        return "moins " + num_to_words_fr(-n)
+    if n < 20:
+        return _FR_ONES[n]
+    if n < 70:
+        t, r = divmod(n, 10)
+        if r == 0:
+            return _FR_TENS[t]
+        if r == 1:
+            return _FR_TENS[t] + " et un"
+        return _FR_TENS[t] + "-" + _FR_ONES[r]
+    if n < 80:  # soixante-dix .. soixante-dix-neuf
+        r = n - 60
+        if r == 11:
+            return "soixante et onze"
+        return "soixante-" + _FR_ONES[r]
+    if n < 100:  # quatre-vingts
+        r = n - 80
+        if r == 0:
+            return "quatre-vingts"
+        return "quatre-vingt-" + _FR_ONES[r]
+    if n < 1000:
+        h, r = divmod(n, 100)
+        head = "cent" if h == 1 else _FR_ONES[h] + " cent"
+        if r == 0:
+            return head + ("s" if h > 1 else "")
+        return head + " " + num_to_words_fr(r)
+    if n < 10 ** 6:
+        t, r = divmod(n, 1000)
+        head = "mille" if t == 1 else num_to_words_fr(t) + " mille"
+        return head if r == 0 else head + " " + num_to_words_fr(r)
+    m, r = divmod(n, 10 ** 6)
+    head = ("un million" if m == 1
+            else num_to_words_fr(m) + " millions")
+    return head if r == 0 else head + " " + num_to_words_fr(r)
+
+
+_IT_ONES = ["zero", "uno", "due", "tre", "quattro", "cinque", "sei",
+            "sette", "otto", "nove", "dieci", "undici", "dodici",
+            "tredici", "quattordici", "quindici", "sedici",
+            "diciassette", "diciotto", "diciannove"]
+_IT_TENS = ["", "", "venti", "trenta", "quaranta", "cinquanta",
+            "sessanta", "settanta", "ottanta", "novanta"]
+
+
+def num_to_words_it(n: int) -> str:
+    if n < 0:
+        return "meno " + num_to_words_it(-n)
+    if n < 20:
+        return _IT_ONES[n]
+    if n < 100:
+        t, r = divmod(n, 10)
+        tens = _IT_TENS[t]
+        if r == 0:
+            return tens
+        if r in (1, 8):  # elision: ventuno, ventotto
+            tens = tens[:-1]
+        return tens + _IT_ONES[r]
+    if n < 1000:
+        h, r = divmod(n, 100)
+        head = "cento" if h == 1 else _IT_ONES[h] + "cento"
+        if r == 0:
+            return head
+        tail = num_to_words_it(r)
+        if tail.startswith("o"):  # elision: cento+ottanta = centottanta
+            head = head[:-1]
+        return head + tail
+    if n < 10 ** 6:
+        t, r = divmod(n, 1000)
+        head = "mille" if t == 1 else num_to_words_it(t) + "mila"
+        return head if r == 0 else head + num_to_words_it(r)
+    m, r = divmod(n, 10 ** 6)
+    head = ("un milione" if m == 1
+            else num_to_words_it(m) + " milioni")
+    return head if r == 0 else head + " " + num_to_words_it(r)
+
+
+_PT_ONES = ["zero", "um", "dois", "três", "quatro", "cinco", "seis",
+            "sete", "oito", "nove", "dez", "onze", "doze", "treze",
+            "catorze", "quinze", "dezesseis", "dezessete", "dezoito",
+            "dezenove"]
+_PT_TENS = ["", "", "vinte", "trinta", "quarenta", "cinquenta",
+            "sessenta", "setenta", "oitenta", "noventa"]
+_PT_HUNDREDS = ["", "cento", "duzentos", "trezentos", "quatrocentos",
+                "quinhentos", "seiscentos", "setecentos",
+                "oitocentos", "novecentos"]
+
+
+def num_to_words_pt(n: int) -> str:
+    if n < 0:
+        return "menos " + num_to_words_pt(-n)
+    if n < 20:
+        return _PT_ONES[n]
+    if n < 100:
+        t, r = divmod(n, 10)
+        return _PT_TENS[t] + ("" if r == 0 else " e " + _PT_ONES[r])
+    if n == 100:
+        return "cem"
+    if n < 1000:
+        h, r = divmod(n, 100)
+        return _PT_HUNDREDS[h] + ("" if r == 0
+                                  else " e " + num_to_words_pt(r))
+    if n < 10 ** 6:
+        t, r = divmod(n, 1000)
+        head = "mil" if t == 1 else num_to_words_pt(t) + " mil"
+        if r == 0:
+            return head
+        joiner = " e " if r < 100 or r % 100 == 0 else " "
+        return head + joiner + num_to_words_pt(r)
+    m, r = divmod(n, 10 ** 6)
+    head = ("um milhão" if m == 1
+            else num_to_words_pt(m) + " milhões")
+    return head if r == 0 else head + " e " + num_to_words_pt(r)
+
+
+_CARDINALS = {
+    "de": num_to_words_de, "es": num_to_words_es,
+    "fr": num_to_words_fr, "it": num_to_words_it,
+    "pt": num_to_words_pt,
+}
+# decimal separator word per language (12.5 -> "douze virgule cinq")
+_DECIMAL_WORD = {"de": "Komma", "es": "coma", "fr": "virgule",
+                 "it": "virgola", "pt": "vírgula"}
+
+
+_GROUPED_DOT_RE = re.compile(r"\b\d{1,3}(?:\.\d{3})+\b")
+_DEC_COMMA_RE = re.compile(r"\b(\d+),(\d+)\b")
+_INT_RE = re.compile(r"\d+")
+
+
 def normalize(text: str, language: str) -> str:
-    """Expand digits/abbreviations for `language` (base code)."""
+    """Expand digits/abbreviations for `language` (base code).
+
+    en: full grammar (cardinals/ordinals/years/currency/percent/times).
+    de/es/fr/it/pt: full cardinal grammar + decimal-comma reading.
+    other covered languages: digit-by-digit (documented approximation).
+    """
     base = language.lower().replace("_", "-").split("-")[0]
     if base == "en":
         return normalize_en(text)
+    card = _CARDINALS.get(base)
+    if card is not None:
+        digits = _DIGITS[base]
+        # 1.234.567 grouping dots -> plain integer
+        text = _GROUPED_DOT_RE.sub(
+            lambda m: m.group(0).replace(".", ""), text)
+        # decimal comma: 12,5 -> "douze virgule cinq"
+        dec = _DECIMAL_WORD[base]
+        text = _DEC_COMMA_RE.sub(
+            lambda m: card(int(m.group(1))) + " " + dec + " "
+            + " ".join(digits[int(d)] for d in m.group(2)), text)
+        return _INT_RE.sub(
+            lambda m: card(int(m.group(0))) if len(m.group(0)) <= 12
+            else " ".join(digits[int(d)] for d in m.group(0)), text)
     digits = _DIGITS.get(base)
     if digits is None:
         return text  # scripts where our tables have no digit names

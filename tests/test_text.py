@@ -205,7 +205,7 @@ def test_numbers_other_languages_digitwise():
     from sonata_amd.text.phonemizer import text_to_phonemes
 
     de = text_to_phonemes("Ich habe 42 Katzen.", voice="de")[0]
-    assert "fˈiːʁ" in de and "tsvˈaɪ" in de  # vier zwei (digit-by-digit)
+    assert "tsvˈaɪʊndfiːʁtsɪk" in de, de  # zweiundvierzig
     es = text_to_phonemes("Tengo 7 gatos.", voice="es")[0]
     assert "sˈiete" in es
     ru = text_to_phonemes("У меня 5 кошек.", voice="ru")[0]
@@ -262,3 +262,29 @@ def test_clock_times_en():
     assert normalize_en("at 3:30") == "at three thirty"
     assert normalize_en("12:00 sharp") == "twelve o'clock sharp"
     assert normalize_en("9:05 train") == "nine oh five train"
+
+
+def test_cardinal_grammars():
+    """Full number words for de/es/fr/it/pt (espeak TranslateNumber
+    parity; previously digit-by-digit)."""
+    from sonata_amd.text.normalize import (normalize, num_to_words_de,
+                                           num_to_words_es,
+                                           num_to_words_fr,
+                                           num_to_words_it,
+                                           num_to_words_pt)
+
+    assert num_to_words_de(21) == "einundzwanzig"
+    assert num_to_words_de(1984) == "eintausendneunhundertvierundachtzig"
+    assert num_to_words_es(42) == "cuarenta y dos"
+    assert num_to_words_es(500) == "quinientos"
+    assert num_to_words_es(100) == "cien"
+    assert num_to_words_fr(71) == "soixante et onze"
+    assert num_to_words_fr(80) == "quatre-vingts"
+    assert num_to_words_fr(99) == "quatre-vingt-dix-neuf"
+    assert num_to_words_it(28) == "ventotto"
+    assert num_to_words_it(1984) == "millenovecentottantaquattro"
+    assert num_to_words_pt(42) == "quarenta e dois"
+    assert num_to_words_pt(100) == "cem"
+    # decimal comma + grouping dots
+    assert normalize("12,5", "fr") == "douze virgule cinq"
+    assert normalize("1.000.000", "de") == "eine Million"
