@@ -95,6 +95,55 @@ _DIGITS = {
            "sete", "oito", "nove"],
 }
 
+
+# digit names for every other covered language (digit-by-digit reading;
+# written in each language's own script so its G2P letters match)
+_DIGITS.update({
+    "ru": "ноль один два три четыре пять шесть семь восемь девять".split(),
+    "uk": "нуль один два три чотири п'ять шість сім вісім дев'ять".split(),
+    "be": "нуль адзін два тры чатыры пяць шэсць сем восем дзевяць".split(),
+    "pl": "zero jeden dwa trzy cztery pięć sześć siedem osiem dziewięć".split(),
+    "cs": "nula jedna dva tři čtyři pět šest sedm osm devět".split(),
+    "sk": "nula jeden dva tri štyri päť šesť sedem osem deväť".split(),
+    "nl": "nul een twee drie vier vijf zes zeven acht negen".split(),
+    "sv": "noll ett två tre fyra fem sex sju åtta nio".split(),
+    "no": "null en to tre fire fem seks sju åtte ni".split(),
+    "da": "nul en to tre fire fem seks syv otte ni".split(),
+    "fi": "nolla yksi kaksi kolme neljä viisi kuusi seitsemän kahdeksan yhdeksän".split(),
+    "hu": "nulla egy kettő három négy öt hat hét nyolc kilenc".split(),
+    "ro": "zero unu doi trei patru cinci șase șapte opt nouă".split(),
+    "el": "μηδέν ένα δύο τρία τέσσερα πέντε έξι επτά οκτώ εννέα".split(),
+    "bg": "нула едно две три четири пет шест седем осем девет".split(),
+    "hr": "nula jedan dva tri četiri pet šest sedam osam devet".split(),
+    "sl": "nič ena dva tri štiri pet šest sedem osem devet".split(),
+    "lt": "nulis vienas du trys keturi penki šeši septyni aštuoni devyni".split(),
+    "lv": "nulle viens divi trīs četri pieci seši septiņi astoņi deviņi".split(),
+    "et": "null üks kaks kolm neli viis kuus seitse kaheksa üheksa".split(),
+    "is": "núll einn tveir þrír fjórir fimm sex sjö átta níu".split(),
+    "sq": "zero një dy tre katër pesë gjashtë shtatë tetë nëntë".split(),
+    "mk": "нула еден два три четири пет шест седум осум девет".split(),
+    "tr": "sıfır bir iki üç dört beş altı yedi sekiz dokuz".split(),
+    "az": "sıfır bir iki üç dörd beş altı yeddi səkkiz doqquz".split(),
+    "kk": "нөл бір екі үш төрт бес алты жеті сегіз тоғыз".split(),
+    "ky": "нөл бир эки үч төрт беш алты жети сегиз тогуз".split(),
+    "uz": "nol bir ikki uch to'rt besh olti yetti sakkiz to'qqiz".split(),
+    "id": "nol satu dua tiga empat lima enam tujuh delapan sembilan".split(),
+    "sw": "sifuri moja mbili tatu nne tano sita saba nane tisa".split(),
+    "eo": "nul unu du tri kvar kvin ses sep ok naŭ".split(),
+    "ca": "zero u dos tres quatre cinc sis set vuit nou".split(),
+    "gl": "cero un dous tres catro cinco seis sete oito nove".split(),
+    "eu": "zero bat bi hiru lau bost sei zazpi zortzi bederatzi".split(),
+    "af": "nul een twee drie vier vyf ses sewe agt nege".split(),
+    "cy": "dim un dau tri pedwar pump chwech saith wyth naw".split(),
+    "mt": "żero wieħed tnejn tlieta erbgħa ħamsa sitta sebgħa tmienja disgħa".split(),
+    "ht": "zewo en de twa kat senk sis sèt uit nèf".split(),
+    "la": "nihil unus duo tres quattuor quinque sex septem octo novem".split(),
+    "hy": "զրո մեկ երկու երեք չորս հինգ վեց յոթ ութ ինը".split(),
+    "ka": "ნული ერთი ორი სამი ოთხი ხუთი ექვსი შვიდი რვა ცხრა".split(),
+    "hi": "शून्य एक दो तीन चार पाँच छह सात आठ नौ".split(),
+    "ar": "صفر واحد اثنان ثلاثة أربعة خمسة ستة سبعة ثمانية تسعة".split(),
+})
+
 _NUM_RE = re.compile(r"\d[\d,]*(?:\.\d+)?")
 _ORD_RE = re.compile(r"\b(\d+)(st|nd|rd|th)\b", re.IGNORECASE)
 _ABBR_RE = re.compile(r"\b(Mr|Mrs|Ms|Dr|St|Jr|Sr|Prof|etc|vs)\.",
