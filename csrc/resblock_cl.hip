@@ -82,41 +82,55 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
 #pragma unroll
     for (int j = 0; j < NT; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+  // Staging lane map: each QUAD (4 lanes) keeps one row's 4 chunks
+  // contiguous (64B global coalescing), but the 4 quads of a 16-lane
+  // LDS phase group cover rows {0,4,8,12}+g instead of {0,1,2,3}: at
+  // BKP=40 (row stride 20 banks) the bank slot is (5r+c) mod 16, and
+  // rows 4 apart give slots 4q+c -- a clean permutation, so staging
+  // writes are conflict-free (the row=u>>2 map collided rows 0 and 3
+  // of each group; profiles/r02_pmc_rbpair.txt: LDS issue-stall 17%).
+  const int sr_l = 4 * ((lane >> 2) & 3) + (lane >> 4);  // row-in-group
+  const int sr_c = (lane & 3) * 8;     // staging chunk (bf16 offset)
+  const int sr_base0 = wid * 16;       // wave's first row; stride 128
+
   for (int c0 = 0; c0 < CP; c0 += BK) {
     const bool c_interior = (c0 + BK) <= C;
     if (t_interior && c_interior) {
-      for (int u = tid; u < xrows * 4; u += 512) {
-        const int r = u >> 2, ch = (u & 3) * 8;
-        bf16 v8[8];
-        *(ulonglong2*)v8 =
-            *(const ulonglong2*)&xb[(row0 + r) * C + c0 + ch];
+      for (int base = sr_base0; base < xrows; base += 128) {
+        const int r = base + sr_l;
+        if (r < xrows) {
+          bf16 v8[8];
+          *(ulonglong2*)v8 =
+              *(const ulonglong2*)&xb[(row0 + r) * C + c0 + sr_c];
 #pragma unroll
-        for (int q = 0; q < 8; ++q)
-          v8[q] = f2bf(lrelu_(bf2f(v8[q]), 0.1f));
-        *(ulonglong2*)&Xs[r][ch] = *(ulonglong2*)v8;
+          for (int q = 0; q < 8; ++q)
+            v8[q] = f2bf(lrelu_(bf2f(v8[q]), 0.1f));
+          *(ulonglong2*)&Xs[r][sr_c] = *(ulonglong2*)v8;
+        }
       }
     } else {
-      for (int u = tid; u < xrows * 4; u += 512) {
-        const int r = u >> 2, ch = (u & 3) * 8;
+      for (int base = sr_base0; base < xrows; base += 128) {
+        const int r = base + sr_l;
+        if (r >= xrows) continue;
         const long t = row0 + r;
         bf16 v8[8];
 #pragma unroll
         for (int q = 0; q < 8; ++q) {
-          const int c = c0 + ch + q;
+          const int c = c0 + sr_c + q;
           float v = (t >= 0 && t < T && c < C) ? bf2f(xb[t * C + c]) : 0.f;
           v8[q] = f2bf(lrelu_(v, 0.1f));
         }
-        *(ulonglong2*)&Xs[r][ch] = *(ulonglong2*)v8;
+        *(ulonglong2*)&Xs[r][sr_c] = *(ulonglong2*)v8;
       }
     }
     for (int tap0 = 0; tap0 < k; tap0 += TC) {
       const int ntc = min(TC, k - tap0);
       for (int tc = 0; tc < ntc; ++tc) {
         const long wbase = ((long)(tap0 + tc) * CP) * CP + c0;
-        for (int u = tid; u < BN * 4; u += 512) {
-          const int n = u >> 2, ch = (u & 3) * 8;
-          *(ulonglong2*)&Ws[tc][n][ch] =
-              *(const ulonglong2*)&w1[wbase + (long)n * CP + ch];
+        for (int base = sr_base0; base < BN; base += 128) {
+          const int n = base + sr_l;
+          *(ulonglong2*)&Ws[tc][n][sr_c] =
+              *(const ulonglong2*)&w1[wbase + (long)n * CP + sr_c];
         }
       }
       __syncthreads();
@@ -173,10 +187,10 @@ __global__ __launch_bounds__(512) void resblock_pair_cl_kernel(
       const int ntc = min(TC, k - tap0);
       for (int tc = 0; tc < ntc; ++tc) {
         const long wbase = ((long)(tap0 + tc) * CP) * CP + c0;
-        for (int u = tid; u < BN * 4; u += 512) {
-          const int n = u >> 2, ch = (u & 3) * 8;
-          *(ulonglong2*)&Ws[tc][n][ch] =
-              *(const ulonglong2*)&w2[wbase + (long)n * CP + ch];
+        for (int base = sr_base0; base < BN; base += 128) {
+          const int n = base + sr_l;
+          *(ulonglong2*)&Ws[tc][n][sr_c] =
+              *(const ulonglong2*)&w2[wbase + (long)n * CP + sr_c];
         }
       }
       __syncthreads();
