@@ -68,6 +68,12 @@ __global__ __launch_bounds__(512) void conv1d_cl_kernel(
   constexpr int NT = WN / 16;
   const int kl = lane >> 4;
   const int il = lane & 15;
+  // conflict-free staging lane map (see resblock_cl.hip: quads keep a
+  // row's 4 chunks contiguous for 64B coalescing; phase-group quads
+  // cover rows {0,4,8,12}+g so LDS bank slots (5r+c)%16 = 4q+c)
+  const int sr_l = 4 * ((lane >> 2) & 3) + (lane >> 4);
+  const int sr_c = (lane & 3) * 8;
+  const int sr_base0 = wid * 16;
 
   f32x4 acc[MT][NT];
 #pragma unroll
@@ -86,8 +92,10 @@ __global__ __launch_bounds__(512) void conv1d_cl_kernel(
     const bool c_interior = (c0 + BK) <= Cin;
     // ---- stage X rows: Xs[r][0..31] = pre(x[row0+r][c0..c0+31]) ------
     if (t_interior && c_interior) {
-      for (int u = tid; u < rows_used * 4; u += 512) {
-        const int r = u >> 2, ch = (u & 3) * 8;
+      for (int base = sr_base0; base < rows_used; base += 128) {
+        const int r = base + sr_l;
+        if (r >= rows_used) continue;
+        const int ch = sr_c;
         bf16 v8[8];
         *(ulonglong2*)v8 =
             *(const ulonglong2*)&xb[(row0 + r) * Cin + c0 + ch];
@@ -99,8 +107,10 @@ __global__ __launch_bounds__(512) void conv1d_cl_kernel(
         *(ulonglong2*)&Xs[r][ch] = *(ulonglong2*)v8;
       }
     } else {
-      for (int u = tid; u < rows_used * 4; u += 512) {
-        const int r = u >> 2, ch = (u & 3) * 8;
+      for (int base = sr_base0; base < rows_used; base += 128) {
+        const int r = base + sr_l;
+        if (r >= rows_used) continue;
+        const int ch = sr_c;
         const long t = row0 + r;
         bf16 v8[8];
         if (t >= 0 && t < Tin && c_interior) {
@@ -131,8 +141,9 @@ __global__ __launch_bounds__(512) void conv1d_cl_kernel(
       // ---- stage W taps: Ws[tc][n][kk] = w[tap][n0+n][c0+kk] ---------
       for (int tc = 0; tc < ntc; ++tc) {
         const long wbase = ((long)(tap0 + tc) * CoutP + n0) * CinP + c0;
-        for (int u = tid; u < BN * 4; u += 512) {
-          const int n = u >> 2, ch = (u & 3) * 8;
+        for (int base = sr_base0; base < BN; base += 128) {
+          const int n = base + sr_l;
+          const int ch = sr_c;
           *(ulonglong2*)&Ws[tc][n][ch] =
               *(const ulonglong2*)&w[wbase + (long)n * CinP + ch];
         }
@@ -225,6 +236,12 @@ __global__ __launch_bounds__(512) void conv1d_cl_wdirect_kernel(
   constexpr int NT = WN / 16;
   const int kl = lane >> 4;
   const int il = lane & 15;
+  // conflict-free staging lane map (see resblock_cl.hip: quads keep a
+  // row's 4 chunks contiguous for 64B coalescing; phase-group quads
+  // cover rows {0,4,8,12}+g so LDS bank slots (5r+c)%16 = 4q+c)
+  const int sr_l = 4 * ((lane >> 2) & 3) + (lane >> 4);
+  const int sr_c = (lane & 3) * 8;
+  const int sr_base0 = wid * 16;
 
   f32x4 acc[MT][NT];
 #pragma unroll
@@ -244,8 +261,10 @@ __global__ __launch_bounds__(512) void conv1d_cl_wdirect_kernel(
     const bool c_interior = (c0 + BK) <= Cin;
     __syncthreads();  // protect Xs against the previous slice's readers
     if (t_interior && c_interior) {
-      for (int u = tid; u < rows_used * 4; u += 512) {
-        const int r = u >> 2, ch = (u & 3) * 8;
+      for (int base = sr_base0; base < rows_used; base += 128) {
+        const int r = base + sr_l;
+        if (r >= rows_used) continue;
+        const int ch = sr_c;
         bf16 v8[8];
         *(ulonglong2*)v8 =
             *(const ulonglong2*)&xb[(row0 + r) * Cin + c0 + ch];
@@ -257,8 +276,10 @@ __global__ __launch_bounds__(512) void conv1d_cl_wdirect_kernel(
         *(ulonglong2*)&Xs[r][ch] = *(ulonglong2*)v8;
       }
     } else {
-      for (int u = tid; u < rows_used * 4; u += 512) {
-        const int r = u >> 2, ch = (u & 3) * 8;
+      for (int base = sr_base0; base < rows_used; base += 128) {
+        const int r = base + sr_l;
+        if (r >= rows_used) continue;
+        const int ch = sr_c;
         const long t = row0 + r;
         bf16 v8[8];
         if (t >= 0 && t < Tin && c_interior) {
@@ -367,6 +388,12 @@ __global__ __launch_bounds__(512) void convt1d_cl_kernel(
   constexpr int NT = WN / 16;
   const int kl = lane >> 4;
   const int il = lane & 15;
+  // conflict-free staging lane map (see resblock_cl.hip: quads keep a
+  // row's 4 chunks contiguous for 64B coalescing; phase-group quads
+  // cover rows {0,4,8,12}+g so LDS bank slots (5r+c)%16 = 4q+c)
+  const int sr_l = 4 * ((lane >> 2) & 3) + (lane >> 4);
+  const int sr_c = (lane & 3) * 8;
+  const int sr_base0 = wid * 16;
 
   f32x4 acc[S][MT][NT];
 #pragma unroll
@@ -383,8 +410,10 @@ __global__ __launch_bounds__(512) void convt1d_cl_kernel(
   for (int c0 = 0; c0 < CinP; c0 += BK) {
     const bool c_interior = (c0 + BK) <= Cin;
     if (t_interior && c_interior) {
-      for (int u = tid; u < ROWS * 4; u += 512) {
-        const int r = u >> 2, ch = (u & 3) * 8;
+      for (int base = sr_base0; base < ROWS; base += 128) {
+        const int r = base + sr_l;
+        if (r >= ROWS) continue;
+        const int ch = sr_c;
         bf16 v8[8];
         *(ulonglong2*)v8 =
             *(const ulonglong2*)&xb[(row0 + r) * Cin + c0 + ch];
@@ -396,8 +425,10 @@ __global__ __launch_bounds__(512) void convt1d_cl_kernel(
         *(ulonglong2*)&Xs[r][ch] = *(ulonglong2*)v8;
       }
     } else {
-      for (int u = tid; u < ROWS * 4; u += 512) {
-        const int r = u >> 2, ch = (u & 3) * 8;
+      for (int base = sr_base0; base < ROWS; base += 128) {
+        const int r = base + sr_l;
+        if (r >= ROWS) continue;
+        const int ch = sr_c;
         const long t = row0 + r;
         bf16 v8[8];
         if (t >= 0 && t < Tin && c_interior) {
@@ -430,8 +461,9 @@ __global__ __launch_bounds__(512) void convt1d_cl_kernel(
 #pragma unroll
       for (int tc = 0; tc < TAUC; ++tc) {
         const long wbase = ((long)(cc * TAUC + tc) * CoutP + n0) * CinP + c0;
-        for (int u = tid; u < BN * 4; u += 512) {
-          const int n = u >> 2, ch = (u & 3) * 8;
+        for (int base = sr_base0; base < BN; base += 128) {
+          const int n = base + sr_l;
+          const int ch = sr_c;
           *(ulonglong2*)&Ws[tc][n][ch] =
               *(const ulonglong2*)&w[wbase + (long)n * CinP + ch];
         }
@@ -698,6 +730,12 @@ __global__ __launch_bounds__(512) void conv1d_direct_cl_kernel(
   constexpr int NT = WN / 16;
   const int kl = lane >> 4;
   const int il = lane & 15;
+  // conflict-free staging lane map (see resblock_cl.hip: quads keep a
+  // row's 4 chunks contiguous for 64B coalescing; phase-group quads
+  // cover rows {0,4,8,12}+g so LDS bank slots (5r+c)%16 = 4q+c)
+  const int sr_l = 4 * ((lane >> 2) & 3) + (lane >> 4);
+  const int sr_c = (lane & 3) * 8;
+  const int sr_base0 = wid * 16;
 
   f32x4 acc[MT][NT];
 #pragma unroll
