@@ -836,3 +836,44 @@ def ru_build_lexicon(apply_rules, vowels) -> dict:
         pos = starts[min(idx - 1, len(starts) - 1)]
         lex[word] = ipa[:pos] + "ˈ" + ipa[pos:]
     return lex
+
+
+# --------------------------------------------------------------------- #
+# Ukrainian / Belarusian palatalization (same mechanism as Russian:
+# consonant + soft vowel digraphs; the bare single-letter rules keep
+# the word-initial/post-vocalic j-glide forms)
+# --------------------------------------------------------------------- #
+
+def _palatalize(rules, cons, soft):
+    out = dict(rules)
+    for c, ci in cons.items():
+        for v, vi in soft.items():
+            out[c + v] = ci + "ʲ" + vi
+        out[c + "ь"] = ci + "ʲ"
+    return out
+
+
+UK_RULES.update(_palatalize({}, {
+    "д": "d", "т": "t", "з": "z", "с": "s", "ц": "ts",
+    "л": "l", "н": "n", "р": "r",
+}, {"я": "ɑ", "ю": "u", "є": "ɛ", "і": "i"}))
+
+BE_RULES.update(_palatalize({}, {
+    "б": "b", "в": "v", "з": "z", "с": "s", "л": "l", "н": "n",
+    "м": "m", "п": "p", "ф": "f", "к": "k", "г": "ɦ", "х": "x",
+    "ц": "ts", "дз": "dz",
+}, {"я": "a", "ю": "u", "е": "ɛ", "ё": "o", "і": "i"}))
+
+
+# small per-language exception lexicons for table languages (stress is
+# lexical in uk; extend per language as corpora grow)
+LEXICONS = {
+    "uk": {
+        "привіт": "prɪʋʲˈit", "будь": "bˈudʲ", "ласка": "lˈɑskɑ",
+        "добре": "dˈɔbrɛ", "дякую": "dʲˈɑkuju", "вона": "ʋɔnˈɑ",
+        "воно": "ʋɔnˈɔ", "вони": "ʋɔnˈɪ", "мене": "mɛnˈɛ",
+        "тебе": "tɛbˈɛ", "себе": "sɛbˈɛ", "язик": "jɑzˈɪk",
+        "завжди": "zˈɑʋʐdɪ", "тепер": "tɛpˈɛr", "тому": "tɔmˈu",
+        "вода": "ʋɔdˈɑ", "земля": "zɛmlʲˈɑ", "зараз": "zˈɑrɑz",
+    },
+}

@@ -725,11 +725,15 @@ def _get_g2p(voice: str) -> RuleG2P:
         from .g2p_tables import (ALIASES, LETTERS, LETTERS2,
                                  STRESS_DEFAULTS, TABLES, TABLES2)
 
+        from .g2p_tables import LEXICONS
+
         key2 = ALIASES.get(base, base)
         if key2 in TABLES:
-            g = RuleG2P(TABLES[key2], letters=LETTERS[key2])
+            g = RuleG2P(TABLES[key2], LEXICONS.get(key2),
+                        letters=LETTERS[key2])
         elif key2 in TABLES2:
-            g = RuleG2P(TABLES2[key2], letters=LETTERS2[key2],
+            g = RuleG2P(TABLES2[key2], LEXICONS.get(key2),
+                        letters=LETTERS2[key2],
                         stress_default=STRESS_DEFAULTS[key2])
         else:
             raise PhonemizationError(

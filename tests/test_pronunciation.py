@@ -449,3 +449,17 @@ def test_ru_oov_still_works():
     g = _get_g2p("ru")
     ipa = g.word_to_ipa("электрификация")
     assert ipa and "ˈ" in ipa
+
+
+def test_uk_be_palatalization():
+    """Ukrainian/Belarusian consonant+soft-vowel digraphs palatalize
+    (no spurious j-glide after consonants)."""
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    uk = _get_g2p("uk")
+    assert uk.word_to_ipa("дякую") == "dʲˈɑkuju"
+    assert uk.word_to_ipa("привіт") == "prɪʋʲˈit"   # lexicon stress
+    assert uk.word_to_ipa("сьогодні") == "sʲˈɔɦɔdnʲi"
+    be = _get_g2p("be")
+    assert be.word_to_ipa("дзякуй") == "dzʲˈakuj"
+    assert be.word_to_ipa("дзень") == "dzʲˈɛnʲ"
