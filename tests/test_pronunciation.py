@@ -249,7 +249,7 @@ def test_expanded_languages_smoke():
         ("da", "søster", "ø"), ("fi", "kiitos", "iː"),
         ("hu", "gyerek", "ɟ"), ("hu", "szép", "s"),
         ("ro", "ceva", "tʃ"), ("el", "ευχαριστώ", "vx"),  # context-free ev (real: ef before voiceless)
-        ("bg", "благодаря", "ɡ"), ("uk", "дякую", "dj"),
+        ("bg", "благодаря", "ɡ"), ("uk", "дякую", "dʲ"),
         ("hr", "džep", "dʒ"), ("sk", "ďakujem", "ɟ"),
         ("id", "nyamuk", "ɲ"), ("sw", "ng'ombe", "ŋ"),
         ("sr", "ljudi", "ʎ"), ("ms", "pagi", "ɡ"),
