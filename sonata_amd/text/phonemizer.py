@@ -598,6 +598,33 @@ _PL_RULES = {
     "dź": "dʑ", "dż": "dʐ", "ch": "x", "ć": "tɕ", "ś": "ɕ",
     "ź": "ʑ", "ż": "ʐ", "ł": "w", "ń": "ɲ", "ą": "ɔ̃", "ę": "ɛ̃",
     "ó": "u", "w": "v", "y": "ɨ",
+    # i-softening: c/s/z/n/dz + i are alveolo-palatal; the i is silent
+    # before another vowel ("ciebie" tɕɛbjɛ) and voiced otherwise
+    "dzia": "dʑa", "dzią": "dʑɔ̃", "dzie": "dʑɛ", "dzię": "dʑɛ̃",
+    "dzio": "dʑɔ", "dziu": "dʑu", "dzi": "dʑi",
+    "cia": "tɕa", "cią": "tɕɔ̃", "cie": "tɕɛ", "cię": "tɕɛ̃",
+    "cio": "tɕɔ", "ciu": "tɕu", "ci": "tɕi",
+    "sia": "ɕa", "sią": "ɕɔ̃", "sie": "ɕɛ", "się": "ɕɛ̃",
+    "sio": "ɕɔ", "siu": "ɕu", "si": "ɕi",
+    "zia": "ʑa", "zią": "ʑɔ̃", "zie": "ʑɛ", "zię": "ʑɛ̃",
+    "zio": "ʑɔ", "ziu": "ʑu", "zi": "ʑi",
+    "nia": "ɲa", "nią": "ɲɔ̃", "nie": "ɲɛ", "nię": "ɲɛ̃",
+    "nio": "ɲɔ", "niu": "ɲu", "ni": "ɲi",
+    # labials/velars + i + vowel carry a j-glide (kobieta, wiem)
+    "bia": "bja", "bią": "bjɔ̃", "bie": "bjɛ",
+    "bię": "bjɛ̃", "bio": "bjɔ", "biu": "bju",
+    "pia": "pja", "pią": "pjɔ̃", "pie": "pjɛ",
+    "pię": "pjɛ̃", "pio": "pjɔ", "piu": "pju",
+    "mia": "mja", "mią": "mjɔ̃", "mie": "mjɛ",
+    "mię": "mjɛ̃", "mio": "mjɔ", "miu": "mju",
+    "fia": "fja", "fią": "fjɔ̃", "fie": "fjɛ",
+    "fię": "fjɛ̃", "fio": "fjɔ", "fiu": "fju",
+    "wia": "vja", "wią": "vjɔ̃", "wie": "vjɛ",
+    "wię": "vjɛ̃", "wio": "vjɔ", "wiu": "vju",
+    "kia": "kja", "kią": "kjɔ̃", "kie": "kjɛ",
+    "kię": "kjɛ̃", "kio": "kjɔ", "kiu": "kju",
+    "gia": "ɡja", "gią": "ɡjɔ̃", "gie": "ɡjɛ",
+    "gię": "ɡjɛ̃", "gio": "ɡjɔ", "giu": "ɡju",
     "a": "a", "b": "b", "c": "ts", "d": "d", "e": "ɛ", "f": "f",
     "g": "ɡ", "h": "x", "i": "i", "j": "j", "k": "k", "l": "l",
     "m": "m", "n": "n", "o": "ɔ", "p": "p", "r": "r", "s": "s",
@@ -702,7 +729,8 @@ def _get_g2p(voice: str) -> RuleG2P:
     elif base == "nl":
         g = RuleG2P(_NL_RULES, letters="a-zA-Z")
     elif base == "pl":
-        g = RuleG2P(_PL_RULES, letters="a-zA-Ząćęłńóśźż")
+        g = RuleG2P(_PL_RULES, letters="a-zA-Ząćęłńóśźż",
+                    stress_default="penult")
     elif base == "tr":
         g = RuleG2P(_TR_RULES, letters="a-zA-Zçğıöşü")
     elif base == "cs":

@@ -463,3 +463,20 @@ def test_uk_be_palatalization():
     be = _get_g2p("be")
     assert be.word_to_ipa("дзякуй") == "dzʲˈakuj"
     assert be.word_to_ipa("дзень") == "dzʲˈɛnʲ"
+
+
+def test_pl_softening_and_stress():
+    """Polish i-softening digraphs + uniform penultimate stress."""
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    g = _get_g2p("pl")
+    cases = [
+        ("dziękuję", "dʑɛ̃kˈujɛ̃"), ("ciebie", "tɕˈɛbjɛ"),
+        ("siedem", "ɕˈɛdɛm"), ("zielony", "ʑɛlˈɔnɨ"),
+        ("nie", "ɲˈɛ"), ("kobieta", "kɔbjˈɛta"),
+        ("wiem", "vjˈɛm"), ("warszawa", "varʂˈava"),
+        ("kiedy", "kjˈɛdɨ"), ("pies", "pjˈɛs"),
+    ]
+    wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
+             if g.word_to_ipa(w) != want]
+    assert not wrong, wrong
