@@ -652,6 +652,10 @@ _CS_RULES = {
     "ť": "c", "ň": "ɲ", "á": "aː", "é": "ɛː", "í": "iː", "ó": "oː",
     "ú": "uː", "ů": "uː", "ý": "iː", "ě": "jɛ", "c": "ts", "j": "j",
     "y": "i", "w": "v", "h": "ɦ",
+    # softening: d/t/n before ě/i/í are palatal; mě = mɲɛ
+    "dě": "ɟɛ", "tě": "cɛ", "ně": "ɲɛ", "mě": "mɲɛ",
+    "di": "ɟi", "ti": "ci", "ni": "ɲi",
+    "dí": "ɟiː", "tí": "ciː", "ní": "ɲiː",
     "a": "a", "b": "b", "d": "d", "e": "ɛ", "f": "f", "g": "ɡ",
     "i": "i", "k": "k", "l": "l", "m": "m", "n": "n", "o": "o",
     "p": "p", "r": "r", "s": "s", "t": "t", "u": "u", "v": "v",
@@ -732,7 +736,8 @@ def _get_g2p(voice: str) -> RuleG2P:
         g = RuleG2P(_PL_RULES, letters="a-zA-Ząćęłńóśźż",
                     stress_default="penult")
     elif base == "tr":
-        g = RuleG2P(_TR_RULES, letters="a-zA-Zçğıöşü")
+        g = RuleG2P(_TR_RULES, letters="a-zA-Zçğıöşü",
+                    stress_default="final")
     elif base == "cs":
         g = RuleG2P(_CS_RULES, letters="a-zA-Začďéěíňóřšťúůýž")
     elif base == "ar":

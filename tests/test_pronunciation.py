@@ -224,9 +224,10 @@ def test_it_basics():
 
 # ---- Turkish (one-to-one orthography) ---------------------------------- #
 TR_CASES = [
-    ("ev", "ˈev"), ("su", "sˈu"), ("kitap", "kˈitap"),
-    ("çocuk", "tʃˈodʒuk"), ("şehir", "ʃˈehiɾ"), ("güzel", "ɡˈyzel"),
-    ("ılık", "ˈɯlɯk"), ("yol", "jˈol"), ("cam", "dʒˈam"),
+    # Turkish stress is (regularly) word-final
+    ("ev", "ˈev"), ("su", "sˈu"), ("kitap", "kitˈap"),
+    ("çocuk", "tʃodʒˈuk"), ("şehir", "ʃehˈiɾ"), ("güzel", "ɡyzˈel"),
+    ("ılık", "ɯlˈɯk"), ("yol", "jˈol"), ("cam", "dʒˈam"),
 ]
 
 
