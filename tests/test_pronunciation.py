@@ -481,3 +481,25 @@ def test_pl_softening_and_stress():
     wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
              if g.word_to_ipa(w) != want]
     assert not wrong, wrong
+
+
+def test_en_irregular_probe_batches():
+    """Round-2 probe batches: silent-letter words, French/Greek loans,
+    -gue/-que endings resolve via lexicon, and inflections propagate."""
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    g = _get_g2p("en")
+    cases = [
+        ("queue", "kjˈu"), ("straight", "stɹˈeɪt"),
+        ("christmas", "kɹˈɪsməs"), ("yacht", "jˈɑt"),
+        ("colleague", "kˈɑliɡ"), ("suite", "swˈit"),
+        ("salmon", "sˈæmən"), ("psalm", "sˈɑm"), ("hymn", "hˈɪm"),
+        ("ballet", "bælˈeɪ"), ("pint", "pˈaɪnt"), ("heir", "ˈɛɹ"),
+        ("chef", "ʃˈɛf"), ("ache", "ˈeɪk"), ("steak", "stˈeɪk"),
+    ]
+    wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
+             if g.word_to_ipa(w) != want]
+    assert not wrong, wrong
+    # inflections derive from the new entries
+    assert g.word_to_ipa("aches") == "ˈeɪks"
+    assert g.word_to_ipa("gauges") == "ɡˈeɪdʒəz"
