@@ -675,7 +675,10 @@ def de_postprocess(ipa: str) -> str:
 
 def de_preprocess(w: str) -> str:
     """Collapse double consonants (they mark the previous vowel short,
-    not a geminate): wasser -> waser."""
+    not a geminate): wasser -> waser.  Word-final -ig is [ɪç]
+    (zwanzig), unlike genuine -ik loans (Musik)."""
+    if w.endswith("ig"):
+        w = w[:-2] + "ich"
     out = []
     for ch in w:
         if out and out[-1] == ch and ch not in "aeiouäöü":
@@ -685,6 +688,17 @@ def de_preprocess(w: str) -> str:
 
 
 DE_LEXICON = {
+    # loans (French/Greek/English) the native rules cannot derive
+    "computer": "kɔmpjˈuːtɐ", "restaurant": "ʁɛstoʁˈaŋ",
+    "garage": "ɡaʁˈaːʒə", "chance": "ʃˈaŋsə", "cousin": "kuzˈɛŋ",
+    "orange": "oʁˈaŋʒə", "genie": "ʒenˈiː", "etage": "etˈaːʒə",
+    "regie": "ʁeʒˈiː", "journalist": "ʒʊʁnalˈɪst",
+    "ingenieur": "ɪnʒenjˈøːɐ", "balkon": "balkˈoŋ",
+    "chemie": "çemˈiː", "china": "çˈiːna", "theater": "teˈaːtɐ",
+    "musik": "muzˈiːk",
+    "thema": "tˈeːma", "familie": "famˈiːliə", "linie": "lˈiːniə",
+    "italien": "itˈaːliən", "europa": "ɔʏʁˈoːpa",
+    "straße": "ʃtʁˈaːsə", "nation": "natsjˈoːn",
     "der": "dɛʁ", "die": "diː", "das": "das", "ein": "ˈaɪn",
     "eine": "ˈaɪnə", "und": "ʊnt", "ist": "ɪst", "sind": "zɪnt",
     "war": "vaːʁ", "ich": "ɪç", "du": "duː", "er": "ɛʁ", "sie": "ziː",

@@ -216,6 +216,16 @@ class RuleG2P:
             prev_v = v
         return starts
 
+    # orthographic prefixes that never carry stress (e.g. German
+    # ver-/be-: stress moves to the stem); applied only when the word
+    # has >= 3 vowel clusters to avoid short false positives
+    stress_skip_prefixes: tuple = ()
+
+    @property
+    def suffix_stress(self):
+        # per-language override (set as instance attr); English default
+        return getattr(self, "_suffix_stress", self._SUFFIX_STRESS)
+
     def _stress_rules_output(self, word: str, ipa: str) -> str:
         if "ˈ" in ipa:  # rules already placed stress (e.g. Greek accents)
             return ipa
@@ -223,7 +233,10 @@ class RuleG2P:
         if not starts:
             return ipa
         idx = 0  # default: first syllable
-        for suf, from_end in self._SUFFIX_STRESS:
+        if len(starts) >= 3 and word.startswith(self.stress_skip_prefixes):
+            pos = starts[1]
+            return ipa[:pos] + "ˈ" + ipa[pos:]
+        for suf, from_end in self.suffix_stress:
             if word.endswith(suf):
                 idx = max(len(starts) - 1 - from_end, 0)
                 break
@@ -461,7 +474,10 @@ _EN_RULES = {
 # --------------------------------------------------------------------------- #
 _DE_RULES = {
     "sch": "ʃ", "tsch": "tʃ", "chs": "ks", "ung": "ʊŋ",
+    "tion": "tsjoːn", "eh": "eː",
     "ei": "aɪ", "ai": "aɪ", "au": "aʊ", "eu": "ɔʏ", "äu": "ɔʏ",
+    # ch is [x] after back vowels (Bach-laut), [ç] elsewhere
+    "auch": "aʊx", "ach": "ax", "och": "ɔx", "uch": "uːx",
     "ie": "iː", "ch": "ç", "ck": "k", "sp": "ʃp", "st": "ʃt",
     "th": "t", "ph": "f", "qu": "kv", "ß": "s",
     "ä": "ɛ", "ö": "ø", "ü": "y",
@@ -699,6 +715,12 @@ def _get_g2p(voice: str) -> RuleG2P:
         g = RuleG2P(_DE_RULES, DE_LEXICON, letters="a-zA-Zäöüß",
                     preprocess=de_preprocess)
         g.postprocess = de_postprocess
+        g.stress_skip_prefixes = ("ver", "be", "er", "ent", "zer",
+                                  "emp", "miss")
+        # German stressed suffixes (loan morphology)
+        g._suffix_stress = [("tion", 0), ("sion", 0), ("tät", 0),
+                            ("enz", 0), ("anz", 0), ("ieren", 1),
+                            ("ieret", 1), ("ei", 0)]
     elif base == "es":
         g = RuleG2P(_ES_RULES, letters="a-zA-Zñáéíóúü",
                     stress_default="es-penult")

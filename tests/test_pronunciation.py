@@ -503,3 +503,23 @@ def test_en_irregular_probe_batches():
     # inflections derive from the new entries
     assert g.word_to_ipa("aches") == "ˈeɪks"
     assert g.word_to_ipa("gauges") == "ɡˈeɪdʒəz"
+
+
+def test_de_quality_layer_round2b():
+    """German: -ig finals, ch after back vowels, stressed loan
+    suffixes, unstressed verb prefixes, loan lexicon."""
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    g = _get_g2p("de")
+    cases = [
+        ("zwanzig", "tsvˈantsɪç"), ("richtig", "ʁˈɪçtɪç"),
+        ("buch", "bˈuːx"), ("sache", "zˈaxə"), ("bücher", "bˈyçɐ"),
+        ("musik", "muzˈiːk"), ("geben", "ɡˈeːbən"),
+        ("information", "ɪnfɔʁmatsjˈoːn"),
+        ("universität", "ʊnɪfɛʁzɪtˈɛt"), ("studieren", "ʃtʊdˈiːʁən"),
+        ("verstehen", "fɛʁʃtˈeːən"), ("bekommen", "bɛkˈɔmən"),
+        ("computer", "kɔmpjˈuːtɐ"), ("chemie", "çemˈiː"),
+    ]
+    wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
+             if g.word_to_ipa(w) != want]
+    assert not wrong, wrong
