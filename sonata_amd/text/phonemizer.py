@@ -492,7 +492,10 @@ _DE_RULES = {
 # Spanish
 # --------------------------------------------------------------------------- #
 _ES_RULES = {
-    "ch": "tʃ", "ll": "ʝ", "rr": "r", "qu": "k", "gu": "ɡ",
+    "ch": "tʃ", "ll": "ʝ", "rr": "r", "qu": "k",
+    # g: [x] before e/i, [ɡ] with silent u in gue/gui (ü keeps the w)
+    "güe": "ɡwe", "güi": "ɡwi", "gue": "ɡe", "gui": "ɡi",
+    "ge": "xe", "gi": "xi",
     "ñ": "ɲ", "j": "x", "v": "b", "z": "θ", "ce": "θe", "ci": "θi",
     "á": "ˈa", "é": "ˈe", "í": "ˈi", "ó": "ˈo", "ú": "ˈu", "ü": "w",
     "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
@@ -549,7 +552,8 @@ _FR_RULES = {
 _IT_RULES = {
     "glia": "ʎa", "glie": "ʎe", "glio": "ʎo", "gliu": "ʎu",
     "gli": "ʎi", "gn": "ɲ",
-    "scia": "ʃa", "scio": "ʃo", "sciu": "ʃu", "sci": "ʃi", "sce": "ʃe",
+    "scia": "ʃa", "scie": "ʃe", "scio": "ʃo", "sciu": "ʃu",
+    "sci": "ʃi", "sce": "ʃe",
     "chi": "ki", "che": "ke", "ghi": "ɡi", "ghe": "ɡe",
     "cia": "tʃa", "cio": "tʃo", "ciu": "tʃu", "ci": "tʃi", "ce": "tʃe",
     "gia": "dʒa", "gio": "dʒo", "giu": "dʒu", "gi": "dʒi", "ge": "dʒe",
@@ -724,6 +728,12 @@ def _get_g2p(voice: str) -> RuleG2P:
     elif base == "es":
         g = RuleG2P(_ES_RULES, letters="a-zA-Zñáéíóúü",
                     stress_default="es-penult")
+
+        def _es_post(ipa: str) -> str:
+            # word-final y is the vowel [i] (muy, hoy)
+            return ipa[:-1] + "i" if ipa.endswith("ʝ") else ipa
+
+        g.postprocess = _es_post
     elif base == "fr":
         from .g2p_tables import FR_LEXICON, fr_preprocess
 

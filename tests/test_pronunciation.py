@@ -523,3 +523,20 @@ def test_de_quality_layer_round2b():
     wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
              if g.word_to_ipa(w) != want]
     assert not wrong, wrong
+
+
+def test_es_g_contexts():
+    """Spanish g: [x] before e/i, silent u in gue/gui, ü = [w];
+    word-final y = [i]."""
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    g = _get_g2p("es")
+    cases = [
+        ("gente", "xˈente"), ("girasol", "xiɾasˈol"),
+        ("guerra", "ɡˈera"), ("agua", "ˈaɡua"),
+        ("vergüenza", "beɾɡwˈenθa"), ("muy", "mˈui"),
+        ("hoy", "ˈoi"), ("gato", "ɡˈato"),
+    ]
+    wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
+             if g.word_to_ipa(w) != want]
+    assert not wrong, wrong
