@@ -601,9 +601,11 @@ _RU_RULES = {
 # Dutch
 # --------------------------------------------------------------------------- #
 _NL_RULES = {
-    "sch": "sx", "ij": "ɛi", "ei": "ɛi", "ui": "œy", "ou": "ʌu",
-    "au": "ʌu", "oe": "u", "eu": "ø", "ie": "i", "aa": "aː",
-    "ee": "eː", "oo": "oː", "uu": "y", "ch": "x", "ng": "ŋ",
+    "sch": "sx", "ij": "ɛi", "ei": "ɛi", "ui": "œy",
+    "ouw": "ʌu", "auw": "ʌu", "ieuw": "iu", "eeuw": "eːu",
+    "ou": "ʌu", "au": "ʌu", "oe": "u", "eu": "ø", "ie": "i",
+    "aa": "aː", "ee": "eː", "oo": "oː", "uu": "y", "ch": "x",
+    "ng": "ŋ",
     "a": "ɑ", "b": "b", "c": "k", "d": "d", "e": "ɛ", "f": "f",
     "g": "ɣ", "h": "ɦ", "i": "ɪ", "j": "j", "k": "k", "l": "l",
     "m": "m", "n": "n", "o": "ɔ", "p": "p", "r": "r", "s": "s",
@@ -764,6 +766,27 @@ def _get_g2p(voice: str) -> RuleG2P:
         g.postprocess = ru_reduce
     elif base == "nl":
         g = RuleG2P(_NL_RULES, letters="a-zA-Z")
+        g.stress_skip_prefixes = ("ge", "be", "ver", "ont", "her")
+
+        _NL_DEVOICE = {"b": "p", "d": "t", "ɣ": "x", "v": "f",
+                       "z": "s"}
+
+        def _nl_post(ipa: str) -> str:
+            # -ig = [əx]; unstressed -en/-er/-e endings are schwa;
+            # Dutch final obstruents devoice
+            if ipa.endswith("ɪɣ"):
+                ipa = ipa[:-2] + "əx"
+            elif ipa.endswith("ɛn"):
+                ipa = ipa[:-2] + "ən"
+            elif ipa.endswith("ɛr"):
+                ipa = ipa[:-2] + "ər"
+            elif ipa.endswith("ɛ") and "ˈɛ" != ipa[-2:]:
+                ipa = ipa[:-1] + "ə"
+            if ipa and ipa[-1] in _NL_DEVOICE:
+                ipa = ipa[:-1] + _NL_DEVOICE[ipa[-1]]
+            return ipa
+
+        g.postprocess = _nl_post
     elif base == "pl":
         g = RuleG2P(_PL_RULES, letters="a-zA-Ząćęłńóśźż",
                     stress_default="penult")

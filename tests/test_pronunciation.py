@@ -540,3 +540,19 @@ def test_es_g_contexts():
     wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
              if g.word_to_ipa(w) != want]
     assert not wrong, wrong
+
+
+def test_nl_quality_layer():
+    """Dutch: final devoicing, schwa endings, -ig = əx, ouw/ieuw."""
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    g = _get_g2p("nl")
+    cases = [
+        ("goed", "ɣˈut"), ("vrouw", "vrˈʌu"), ("nieuw", "nˈiu"),
+        ("meisje", "mˈɛisjə"), ("spreken", "sprˈɛkən"),
+        ("gezellig", "ɣɛzˈɛlləx"), ("hond", "ɦˈɔnt"),
+        ("dag", "dˈɑx"),
+    ]
+    wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
+             if g.word_to_ipa(w) != want]
+    assert not wrong, wrong
