@@ -570,7 +570,10 @@ _IT_RULES = {
 # Portuguese (pt-BR leaning)
 # --------------------------------------------------------------------------- #
 _PT_RULES = {
-    "lh": "ʎ", "nh": "ɲ", "ch": "ʃ", "qu": "k", "gu": "ɡ", "rr": "ʁ",
+    "lh": "ʎ", "nh": "ɲ", "ch": "ʃ", "rr": "ʁ",
+    # qu/gu: silent u before e/i, [w] before a/o
+    "qua": "kwa", "quo": "kwo", "qu": "k",
+    "gua": "ɡwa", "guo": "ɡwo", "gu": "ɡ",
     "ão": "ˈɐ̃w̃", "õe": "ˈõj̃", "ã": "ɐ̃", "õ": "õ",
     "ce": "se", "ci": "si", "cé": "sˈɛ", "cê": "sˈe", "cí": "sˈi",
     "ge": "ʒe", "gi": "ʒi", "gé": "ʒˈɛ", "gê": "ʒˈe", "gí": "ʒˈi",
@@ -752,9 +755,20 @@ def _get_g2p(voice: str) -> RuleG2P:
                 ipa = ipa[:-1] + "u"
             elif ipa.endswith("e") and "ˈe" != ipa[-2:]:
                 ipa = ipa[:-1] + "i"
+            # BR palatalization of the reduced final syllable: ti/di
+            if ipa.endswith("ti"):
+                ipa = ipa[:-2] + "tʃi"
+            elif ipa.endswith("di"):
+                ipa = ipa[:-2] + "dʒi"
             return ipa
 
-        g = RuleG2P(_PT_RULES, letters="a-zA-Zàáâãçéêíóôõú",
+        _PT_LEX = {
+            "muito": "mˈũitu", "muita": "mˈũita", "bem": "bˈẽi",
+            "também": "tɐ̃bˈẽi", "quem": "kˈẽi", "sem": "sˈẽi",
+            "tem": "tˈẽi", "em": "ẽi", "um": "ˈũ", "uma": "ˈuma",
+            "com": "kˈõ", "bom": "bˈõ", "são": "sˈɐ̃w̃",
+        }
+        g = RuleG2P(_PT_RULES, _PT_LEX, letters="a-zA-Zàáâãçéêíóôõú",
                     stress_default="es-penult")
         g.postprocess = _pt_post
     elif base == "ru":

@@ -556,3 +556,21 @@ def test_nl_quality_layer():
     wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
              if g.word_to_ipa(w) != want]
     assert not wrong, wrong
+
+
+def test_pt_quality_layer():
+    """Brazilian Portuguese: qu/gu contexts, nasal function words,
+    final-syllable ti/di palatalization."""
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    g = _get_g2p("pt")
+    cases = [
+        ("água", "ˈaɡwa"), ("quando", "kwˈandu"),
+        ("quente", "kˈentʃi"), ("muito", "mˈũitu"),
+        ("bem", "bˈẽi"), ("cidade", "sidˈadʒi"),
+        ("noite", "nˈoitʃi"), ("gente", "ʒˈentʃi"),
+        ("obrigado", "obɾiɡˈadu"),
+    ]
+    wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
+             if g.word_to_ipa(w) != want]
+    assert not wrong, wrong
