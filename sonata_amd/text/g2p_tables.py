@@ -589,12 +589,30 @@ def fr_preprocess(w: str) -> str:
     w = _re.sub(r"(er|ez)$", "é", w)       # parler -> parlé (= /e/)
     w = _re.sub(r"[stdxzp]$", "", w)       # petit, temps, grand, prix...
     w = _re.sub(r"[stdxzp]$", "", w)       # temps: strip s then p
+    # context markers (uppercase = direct rules in _FR_RULES): protect
+    # intervocalic n/m from the nasal-vowel rules (ami, animal),
+    # intervocalic s = [z] (maison), soft c/g before front vowels,
+    # hard gu before front vowels.  BEFORE the e-muet strip so the
+    # final e still conditions them (image -> imaj).
+    V = "aeiouyâäéèêëîïôöûù"
+    w = _re.sub(r"nn", "N", w)
+    w = _re.sub(r"mm", "M", w)
+    w = _re.sub(rf"(?<=[{V}])n(?=[{V}h])", "N", w)
+    w = _re.sub(rf"(?<=[{V}])m(?=[{V}h])", "M", w)
+    w = _re.sub(rf"(?<=[{V}])s(?=[{V}])", "z", w)
+    w = _re.sub(r"gu(?=[eiyéèê])", "G", w)
+    w = _re.sub(r"g(?=[eiyéèê])", "j", w)
+    w = _re.sub(r"c(?=[eiyéèê])", "ç", w)
     if len(w) > 2 and w.endswith("e") and w[-2] not in "aeiouéè":
         w = w[:-1]                          # e-muet: chose -> chos
     return w or w
 
 
 FR_LEXICON = {
+    "ville": "vˈil", "mille": "mˈil", "tranquille": "tʁɑ̃kˈil",
+    "village": "vilˈaʒ", "million": "miljˈɔ̃", "villa": "vilˈa",
+    "guerre": "ɡˈɛʁ", "question": "kɛstjˈɔ̃", "monsieur": "məsjˈø",
+    "femme": "fˈam", "fils": "fˈis", "oeil": "ˈœj", "eau": "ˈo",
     # articles / pronouns / function words (rule-resistant vowels)
     "le": "lə", "la": "la", "les": "le", "un": "œ̃", "une": "yn",
     "des": "de", "du": "dy", "de": "də", "au": "o", "aux": "o",

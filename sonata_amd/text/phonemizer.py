@@ -534,6 +534,10 @@ _FR_RULES = {
     "eau": "o", "eaux": "o", "aux": "o", "eux": "ø", "oeu": "œ",
     "ain": "ɛ̃", "ein": "ɛ̃", "aim": "ɛ̃", "oin": "wɛ̃",
     "tion": "sjɔ̃", "ille": "ij", "gn": "ɲ", "ch": "ʃ", "ph": "f",
+    "eille": "ɛj", "aille": "aj", "ouille": "uj", "euille": "œj",
+    "eil": "ɛj", "ail": "aj", "euil": "œj", "ueil": "œj",
+    "ien": "jɛ̃", "ss": "s", "N": "n", "M": "m", "G": "ɡ",
+    "ill": "ij", "rr": "ʁ", "que": "k", "er": "ɛʁ",
     "qu": "k", "ou": "u", "oi": "wa", "au": "o", "ai": "ɛ", "ei": "ɛ",
     "eu": "ø", "an": "ɑ̃", "am": "ɑ̃", "en": "ɑ̃", "em": "ɑ̃",
     "on": "ɔ̃", "om": "ɔ̃", "in": "ɛ̃", "im": "ɛ̃", "un": "œ̃",
@@ -744,7 +748,9 @@ def _get_g2p(voice: str) -> RuleG2P:
 
         g = RuleG2P(_FR_RULES, FR_LEXICON,
                     letters="a-zA-Zàâçéèêëîïôûùüœ'",
-                    preprocess=fr_preprocess)
+                    preprocess=fr_preprocess,
+                    stress_default="final")
+        g._suffix_stress = []  # French stress is uniformly final
     elif base == "it":
         g = RuleG2P(_IT_RULES, letters="a-zA-Zàèéìòù",
                     stress_default="es-penult")

@@ -574,3 +574,24 @@ def test_pt_quality_layer():
     wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
              if g.word_to_ipa(w) != want]
     assert not wrong, wrong
+
+
+def test_fr_context_markers():
+    """French context handling: intervocalic n/m/s, soft c/g, -ille,
+    final stress."""
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    g = _get_g2p("fr")
+    cases = [
+        ("ami", "amˈi"), ("animal", "animˈal"), ("maison", "mɛzˈɔ̃"),
+        ("image", "imˈaʒ"), ("famille", "famˈij"), ("fille", "fˈij"),
+        ("ville", "vˈil"), ("village", "vilˈaʒ"),
+        ("travail", "tʁavˈaj"), ("soleil", "sɔlˈɛj"),
+        ("chien", "ʃjˈɛ̃"), ("cinéma", "sinemˈa"),
+        ("guerre", "ɡˈɛʁ"), ("langue", "lˈɑ̃ɡ"),
+        ("musique", "myzˈik"), ("manger", "mɑ̃ʒˈe"),
+        ("personne", "pɛʁsˈɔn"), ("université", "ynivɛʁsitˈe"),
+    ]
+    wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
+             if g.word_to_ipa(w) != want]
+    assert not wrong, wrong
