@@ -31,6 +31,9 @@ _IPA_SYMBOLS = (
     "̝̃",  # combining tilde (nasal) + raised diacritics
     # second expansion batch (cy/hi): lateral fricative + retroflexes
     "ɬʈɖɳ",
+    # third expansion batch (Indic/Hangul/Vietnamese): retroflex
+    # lateral + approximant, velar approximant (Korean ㅢ)
+    "ɭɻɰ",
 )
 
 

@@ -706,6 +706,9 @@ _EN_LETTERS = {
 
 _G2P_REGISTRY: Dict[str, RuleG2P] = {}
 
+_INDIC = ("mr", "ne", "bn", "as", "gu", "pa", "or", "ta", "te", "kn",
+          "ml")
+
 
 def _get_g2p(voice: str) -> RuleG2P:
     key = voice.lower().replace("_", "-")
@@ -821,13 +824,17 @@ def _get_g2p(voice: str) -> RuleG2P:
             letters="؀-ۿ",
             stress=False,
         )
-    elif base == "hi":
-        # Devanagari needs abugida handling (inherent schwa, matras,
-        # virama) that a flat rule table cannot express.
-        from .g2p_tables import hi_word_to_ipa
+    elif base == "hi" or base in _INDIC or base == "si":
+        # Brahmic abugidas share one engine (g2p_indic.py): tables are
+        # generated from the Devanagari base via the ISCII-parallel
+        # block offsets; Sinhala has hand tables.
+        from .g2p_indic import (INDIC_LETTERS, make_engine,
+                                make_si_engine)
 
-        g = RuleG2P({}, letters="ऀ-ॿ", stress=False)
-        g._apply_rules = hi_word_to_ipa  # plain function, no self
+        eng = make_si_engine() if base == "si" else make_engine(base)
+        letters = "ऀ-ॿ" if base == "hi" else INDIC_LETTERS[base]
+        g = RuleG2P({}, letters=letters, stress=False)
+        g._apply_rules = eng.word_to_ipa
     else:
         # expansion tables (g2p_tables.py): 13 + 23 more languages
         from .g2p_tables import (ALIASES, LETTERS, LETTERS2,
