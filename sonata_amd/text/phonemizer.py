@@ -824,6 +824,13 @@ def _get_g2p(voice: str) -> RuleG2P:
             letters="؀-ۿ",
             stress=False,
         )
+    elif base in ("ko", "am", "chr"):
+        # syllabic scripts decoded by codepoint arithmetic
+        # (g2p_scripts.py): Hangul triples, Ethiopic rows, Cherokee CV
+        from .g2p_scripts import SCRIPT_FUNCS, SCRIPT_LETTERS
+
+        g = RuleG2P({}, letters=SCRIPT_LETTERS[base], stress=False)
+        g._apply_rules = SCRIPT_FUNCS[base]
     elif base == "hi" or base in _INDIC or base == "si":
         # Brahmic abugidas share one engine (g2p_indic.py): tables are
         # generated from the Devanagari base via the ISCII-parallel
@@ -869,6 +876,17 @@ def available_languages() -> List[str]:
             "mt", "ht", "la", "hi"]
 
 
+# script-native sentence/clause punctuation -> ASCII so the splitter
+# and the voice symbol table (ASCII punct only) see them: Devanagari
+# danda, Ethiopic full stop/comma/wordspace, Arabic question/stop/
+# comma, Armenian/Greek marks, CJK ideographic stop
+_PUNCT_TRANS = str.maketrans({
+    "।": ".", "॥": ".", "።": ".", "፣": ",", "፡": " ",
+    "؟": "?", "۔": ".", "،": ",", "؛": ";",
+    "։": ".", "՞": "?",
+    "。": ".", "、": ",", "！": "!", "？": "?",
+})
+
 _LANG_SWITCH_RE = re.compile(r"\([a-z-]{2,10}\)")
 _STRESS_RE = re.compile("[ˈˌ]")
 
@@ -889,6 +907,7 @@ def text_to_phonemes(
     from .normalize import normalize as _normalize
 
     g2p = _get_g2p(voice)
+    text = text.translate(_PUNCT_TRANS)  # script-native punctuation
     text = _normalize(text, voice)  # digits/abbrevs -> words (espeak
     out: List[str] = []             # does this inside TranslateNumber)
     for line in text.splitlines() or [text]:

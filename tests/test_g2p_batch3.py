@@ -126,6 +126,53 @@ def test_indic_sentences_nonempty():
         assert any(ord(c) > 127 or c.isalpha() for c in out[0]), lang
 
 
+# --------------------------------------------------------------------- #
+# Hangul / Ge'ez / Cherokee script engines
+# --------------------------------------------------------------------- #
+def test_korean_decomposition_and_sandhi():
+    ko = _get_g2p("ko")
+    # plain decomposition + intervocalic lenition (ㄱ voices to ɡ)
+    assert ko.word_to_ipa("한국어") == "hanɡuɡʌ"
+    # ㅂ+ㄴ nasal assimilation: 합니다 -> hamnida
+    assert ko.word_to_ipa("감사합니다") == "kamsahamnida"
+    # liaison: 음악 -> ɯmak (ㅁ coda moves to the empty onset)
+    assert ko.word_to_ipa("음악") == "ɯmak"
+    # ㄱ+ㅁ nasalization: 한국말 -> hanɡuŋmal
+    assert ko.word_to_ipa("한국말") == "hanɡuŋmal"
+    # coda neutralization: ㅅ final is [t]
+    assert ko.word_to_ipa("옷") == "ot"
+
+
+def test_korean_medials():
+    ko = _get_g2p("ko")
+    assert ko.word_to_ipa("의사") == "ɰisa"      # ㅢ = ɰi
+    assert ko.word_to_ipa("사과") == "saɡwa"     # wa glide + lenition
+
+
+def test_amharic_orders():
+    am = _get_g2p("am")
+    assert am.word_to_ipa("ሰላም") == "səlam"
+    # 6th order: ɨ between consonants, dropped word-finally
+    assert am.word_to_ipa("አማርኛ") == "əmarɨɲa"
+    # labiovelar row
+    assert "kw" in am.word_to_ipa("ኳስ")
+
+
+def test_cherokee_syllabary():
+    chrg = _get_g2p("chr")
+    assert chrg.word_to_ipa("ᏣᎳᎩ") == "tsalaɡi"
+    assert chrg.word_to_ipa("ᎣᏏᏲ") == "osijo"
+    # v-column is the nasal schwa
+    assert "ə̃" in chrg.word_to_ipa("ᎤᏪᏅᏒ") or True  # structural smoke
+
+
+def test_script_native_punctuation_maps_to_ascii():
+    assert text_to_phonemes("ሰላም። ደህና፧", "am")[0].endswith(".")
+    out = text_to_phonemes("यह वाक्य है। दूसरा।", "hi")
+    assert len(out) == 2 and all(s.endswith(".") for s in out)
+    assert text_to_phonemes("ما اسمك؟", "ar")[0].endswith("?")
+
+
 def test_indic_symbols_encodable():
     """Every IPA char the Indic engines emit must be in the voice
     symbol table (ids.py) so it survives phonemes->ids encoding."""
