@@ -1,0 +1,599 @@
+"""Third-batch rule tables: Arabic-script, Hebrew, kana-Japanese and
+the remaining Latin/Cyrillic regular orthographies.
+
+Reference bar: the same language codes in espeak-ng's dictionary set
+(deps/dev/espeak-ng-data/{fa,ur,ug,he,ja,vi,ga,…}_dict, reached through
+crates/text/espeak-phonemizer/src/lib.rs:65-156).  Coverage tier per
+language is documented in PARITY.md:
+
+- fa/ur/he write no short vowels: a consonant map plus an epenthetic
+  vowel between cluster-internal consonants and a function-word lexicon
+  gives an intelligible approximation (espeak itself uses per-word
+  dictionaries here).
+- ug (Uyghur Arabic script) and the Latin/Cyrillic tables are fully
+  written orthographies — plain longest-match rules work.
+- ja covers kana exactly (a syllabary); kanji needs a dictionary and is
+  dropped, stated honestly in PARITY.md.
+"""
+
+from __future__ import annotations
+
+import re as _re
+import unicodedata
+from typing import Dict, Optional
+
+# --------------------------------------------------------------------- #
+# Epenthesis for consonant-skeleton scripts (fa/ur/he)
+# --------------------------------------------------------------------- #
+_MULTI_PHONES = ("tʃʰ", "dʒʰ", "tʃ", "dʒ", "ts", "dz", "kʰ", "ɡʰ",
+                 "pʰ", "bʰ", "tʰ", "dʰ", "ʈʰ", "ɖʰ", "sˤ", "dˤ", "tˤ",
+                 "ðˤ", "aː", "iː", "uː", "eː", "oː", "ɒː", "æː", "ɛː")
+_VOWEL_START = set("aeiouəɑɔæɛɪʊʌɒɯɨyøœɶʏɤ")
+
+
+def _tokenize_ipa(ipa: str):
+    toks = []
+    i, n = 0, len(ipa)
+    while i < n:
+        for m in _MULTI_PHONES:
+            if ipa.startswith(m, i):
+                toks.append(m)
+                i += len(m)
+                break
+        else:
+            toks.append(ipa[i])
+            i += 1
+    return toks
+
+
+def epenthesize(ipa: str, vowel: str) -> str:
+    """Insert an epenthetic vowel between consecutive consonants.
+
+    The word-final CC cluster is kept (Persian allows final clusters:
+    دوست duːst) unless the word is just two consonants (من mæn) or has
+    4+ consonants (keeping the cluster would leave CC after an inserted
+    vowel: هستم h-s-t-m -> hæsætæm, not *hæsætm)."""
+    toks = _tokenize_ipa(ipa)
+    out = []
+    n = len(toks)
+    n_cons = sum(1 for t in toks
+                 if t[0] not in _VOWEL_START and t != "ʔ")
+    for i, t in enumerate(toks):
+        out.append(t)
+        if i + 1 >= n:
+            break
+        cur_c = t[0] not in _VOWEL_START and t not in ("ʔ",)
+        nxt = toks[i + 1]
+        nxt_c = nxt[0] not in _VOWEL_START
+        if cur_c and nxt_c:
+            final_pair = (i + 2 == n)
+            if not final_pair or n <= 2 or n_cons >= 4:
+                out.append(vowel)
+    return "".join(out)
+
+
+# --------------------------------------------------------------------- #
+# Persian (fa): Arabic script + پ چ ژ گ; unwritten short vowels
+# --------------------------------------------------------------------- #
+FA_RULES = {
+    "خوا": "xɒː", "ای": "iː",  # silent-vav and word-initial i
+    "آ": "ɒː", "ا": "ɒː", "ب": "b", "پ": "p", "ت": "t", "ث": "s",
+    "ج": "dʒ", "چ": "tʃ", "ح": "h", "خ": "x", "د": "d", "ذ": "z",
+    "ر": "r", "ز": "z", "ژ": "ʒ", "س": "s", "ش": "ʃ", "ص": "s",
+    "ض": "z", "ط": "t", "ظ": "z", "ع": "ʔ", "غ": "ɢ", "ف": "f",
+    "ق": "ɢ", "ک": "k", "ك": "k", "گ": "ɡ", "ل": "l", "م": "m",
+    "ن": "n", "و": "uː", "ه": "h", "ی": "iː", "ي": "iː", "ئ": "ʔ",
+    "ء": "ʔ", "أ": "ʔ", "ؤ": "ʔ",
+    # harakat if present
+    "َ": "æ", "ُ": "o", "ِ": "e", "ّ": "", "ْ": "",
+}
+
+FA_LEXICON = {
+    "است": "æst", "این": "iːn", "آن": "ɒːn", "که": "ke", "به": "be",
+    "از": "æz", "در": "dær", "را": "rɒː", "و": "væ", "با": "bɒː",
+    "من": "mæn", "تو": "to", "او": "uː", "ما": "mɒː", "شما": "ʃomɒː",
+    "چه": "tʃe", "بود": "buːd", "شد": "ʃod", "می": "miː",
+    "یک": "jek", "دو": "do", "سه": "se", "نه": "næ", "بله": "bæle",
+    "سلام": "sælɒːm", "خوب": "xuːb", "بزرگ": "bozorɡ",
+    "ایران": "iːrɒːn", "فارسی": "fɒːrsiː", "زبان": "zæbɒːn",
+    "خانه": "xɒːne", "آب": "ɒːb", "روز": "ruːz", "شب": "ʃæb",
+    "سال": "sɒːl", "مرد": "mærd", "زن": "zæn", "بچه": "bætʃtʃe",
+    "کتاب": "ketɒːb", "دوست": "duːst", "کار": "kɒːr", "وقت": "væɢt",
+    "دست": "dæst", "سر": "sær", "دل": "del", "چشم": "tʃeʃm",
+    "خدا": "xodɒː", "مردم": "mærdom", "شهر": "ʃæhr", "راه": "rɒːh",
+}
+
+
+def fa_postprocess(ipa: str) -> str:
+    # word-final ه is the vowel /e/ (خانه), not [h]
+    if ipa.endswith("h") and len(ipa) > 2:
+        ipa = ipa[:-1] + "e"
+    return epenthesize(ipa, "æ")
+
+
+# --------------------------------------------------------------------- #
+# Urdu (ur): Persian set + retroflexes, aspiration with ھ, ے and ں
+# --------------------------------------------------------------------- #
+UR_RULES = {
+    "کھ": "kʰ", "گھ": "ɡʰ", "چھ": "tʃʰ", "جھ": "dʒʰ", "ٹھ": "ʈʰ",
+    "ڈھ": "ɖʰ", "تھ": "tʰ", "دھ": "dʰ", "پھ": "pʰ", "بھ": "bʰ",
+    "ڑھ": "ɾʰ",
+    "آ": "ɑː", "ا": "ɑː", "ب": "b", "پ": "p", "ت": "t", "ٹ": "ʈ",
+    "ث": "s", "ج": "dʒ", "چ": "tʃ", "ح": "h", "خ": "x", "د": "d",
+    "ڈ": "ɖ", "ذ": "z", "ر": "r", "ڑ": "ɾ", "ز": "z", "ژ": "ʒ",
+    "س": "s", "ش": "ʃ", "ص": "s", "ض": "z", "ط": "t", "ظ": "z",
+    "ع": "ʔ", "غ": "ɣ", "ف": "f", "ق": "q", "ک": "k", "گ": "ɡ",
+    "ل": "l", "م": "m", "ن": "n", "ں": "n", "و": "oː", "ہ": "h",
+    "ھ": "h", "ء": "ʔ", "ی": "iː", "ے": "eː", "ئ": "ʔ",
+    "َ": "ə", "ُ": "ʊ", "ِ": "ɪ", "ّ": "", "ْ": "",
+}
+
+UR_LEXICON = {
+    "ہے": "hɛː", "ہیں": "hɛ̃ː", "کا": "kɑː", "کی": "kiː",
+    "کے": "keː", "میں": "mẽː", "سے": "seː", "کو": "koː",
+    "پر": "pər", "اور": "ɔːr", "یہ": "jeh", "وہ": "voh",
+    "ایک": "eːk", "نے": "neː", "نہیں": "nəhĩː", "کیا": "kjɑː",
+    "اردو": "ʊrduː", "پاکستان": "pɑːkɪstɑːn", "سلام": "səlɑːm",
+    "شکریہ": "ʃʊkrijə", "پانی": "pɑːniː", "دن": "dɪn",
+    "رات": "rɑːt", "گھر": "ɡʰər", "لوگ": "loːɡ", "بات": "bɑːt",
+}
+
+
+def ur_postprocess(ipa: str) -> str:
+    return epenthesize(ipa, "ə")
+
+
+# --------------------------------------------------------------------- #
+# Uyghur (ug): fully vocalized Arabic script — plain rules suffice
+# --------------------------------------------------------------------- #
+UG_RULES = {
+    # hamza carrier + vowel (word-initial)
+    "ئا": "a", "ئە": "æ", "ئې": "e", "ئى": "i", "ئو": "o",
+    "ئۇ": "u", "ئۆ": "ø", "ئۈ": "y",
+    "ا": "a", "ە": "æ", "ې": "e", "ى": "i", "و": "o", "ۇ": "u",
+    "ۆ": "ø", "ۈ": "y",
+    "ب": "b", "پ": "p", "ت": "t", "ج": "dʒ", "چ": "tʃ", "خ": "x",
+    "د": "d", "ر": "r", "ز": "z", "ژ": "ʒ", "س": "s", "ش": "ʃ",
+    "غ": "ʁ", "ف": "f", "ق": "q", "ك": "k", "ک": "k", "گ": "ɡ",
+    "ڭ": "ŋ", "ل": "l", "م": "m", "ن": "n", "ھ": "h", "ۋ": "w",
+    "ي": "j", "ئ": "",
+}
+
+# --------------------------------------------------------------------- #
+# Hebrew (he): unvocalized; nikud handled when present
+# --------------------------------------------------------------------- #
+HE_RULES = {
+    "א": "ʔ", "ב": "v", "ג": "ɡ", "ד": "d", "ה": "h", "ו": "v",
+    "ז": "z", "ח": "x", "ט": "t", "י": "j", "כ": "x", "ך": "x",
+    "ל": "l", "מ": "m", "ם": "m", "נ": "n", "ן": "n", "ס": "s",
+    "ע": "ʔ", "פ": "f", "ף": "f", "צ": "ts", "ץ": "ts", "ק": "k",
+    "ר": "ʁ", "ש": "ʃ", "ת": "t",
+    # nikud (when present, it wins over epenthesis)
+    "ַ": "a", "ָ": "a", "ֵ": "e", "ֶ": "e", "ִ": "i", "ֹ": "o",
+    "ֻ": "u", "ְ": "", "ּ": "", "ׁ": "", "ׂ": "",
+    "וֹ": "o", "וּ": "u",
+}
+
+HE_LEXICON = {
+    "של": "ʃel", "את": "et", "לא": "lo", "זה": "ze", "אני": "ani",
+    "הוא": "hu", "היא": "hi", "מה": "ma", "כן": "ken", "על": "al",
+    "עם": "im", "אל": "el", "כל": "kol", "יש": "jeʃ", "אין": "ejn",
+    "גם": "ɡam", "רק": "ʁak", "אם": "im", "או": "o", "כי": "ki",
+    "שלום": "ʃalom", "תודה": "toda", "בוקר": "bokeʁ", "טוב": "tov",
+    "ערב": "eʁev", "לילה": "lajla", "יום": "jom", "שנה": "ʃana",
+    "בית": "bajit", "ילד": "jeled", "אישה": "iʃa", "איש": "iʃ",
+    "מים": "majim", "עברית": "ivʁit", "ישראל": "jisʁael",
+    "אדם": "adam", "עיר": "iʁ", "דרך": "deʁex", "עכשיו": "axʃav",
+}
+
+
+def he_postprocess(ipa: str) -> str:
+    # glottal carriers vanish except word-initially before a vowel
+    ipa = epenthesize(ipa, "a")
+    ipa = ipa.replace("ʔ", "")
+    # word-final ה is silent (written vowel marker)
+    if ipa.endswith("h") and len(ipa) > 2:
+        ipa = ipa[:-1]
+    return ipa
+
+
+def he_preprocess(w: str) -> str:
+    # medial י = /i/, medial ו = /o/ when flanked by consonants —
+    # approximate matres lectionis before the consonant rules run
+    if len(w) > 2:
+        core = w[1:-1]
+        core = core.replace("ו", "ֺ")  # private: map to vowel o
+        w = w[0] + core + w[-1]
+    return w.replace("ֺ", "ֹ")
+
+
+# --------------------------------------------------------------------- #
+# Japanese kana (ja): hiragana/katakana are exact syllabaries.
+# Kanji requires a reading dictionary — dropped (documented).
+# --------------------------------------------------------------------- #
+_KANA_BASE = {
+    "あ": "a", "い": "i", "う": "u", "え": "e", "お": "o",
+    "か": "ka", "き": "ki", "く": "ku", "け": "ke", "こ": "ko",
+    "が": "ɡa", "ぎ": "ɡi", "ぐ": "ɡu", "げ": "ɡe", "ご": "ɡo",
+    "さ": "sa", "し": "ʃi", "す": "su", "せ": "se", "そ": "so",
+    "ざ": "za", "じ": "dʒi", "ず": "zu", "ぜ": "ze", "ぞ": "zo",
+    "た": "ta", "ち": "tʃi", "つ": "tsu", "て": "te", "と": "to",
+    "だ": "da", "ぢ": "dʒi", "づ": "zu", "で": "de", "ど": "do",
+    "な": "na", "に": "ni", "ぬ": "nu", "ね": "ne", "の": "no",
+    "は": "ha", "ひ": "çi", "ふ": "ɸu", "へ": "he", "ほ": "ho",
+    "ば": "ba", "び": "bi", "ぶ": "bu", "べ": "be", "ぼ": "bo",
+    "ぱ": "pa", "ぴ": "pi", "ぷ": "pu", "ぺ": "pe", "ぽ": "po",
+    "ま": "ma", "み": "mi", "む": "mu", "め": "me", "も": "mo",
+    "や": "ja", "ゆ": "ju", "よ": "jo",
+    "ら": "ɾa", "り": "ɾi", "る": "ɾu", "れ": "ɾe", "ろ": "ɾo",
+    "わ": "wa", "を": "o", "ん": "n",
+    "ぁ": "a", "ぃ": "i", "ぅ": "u", "ぇ": "e", "ぉ": "o",
+}
+_KANA_SMALL_Y = {"ゃ": "ja", "ゅ": "ju", "ょ": "jo"}
+
+
+def _build_kana() -> Dict[str, str]:
+    t: Dict[str, str] = {}
+    for k, v in _KANA_BASE.items():
+        t[k] = v
+        kk = chr(ord(k) + 0x60)  # katakana is hiragana + 0x60
+        t[kk] = v
+    for sm, glide in _KANA_SMALL_Y.items():
+        for base in "きぎしじちにひびぴみり":
+            onset = _KANA_BASE[base][:-1]
+            if onset.endswith("ʃ") or onset.endswith("ʒ"):
+                t[base + sm] = onset + glide[1:]
+            else:
+                t[base + sm] = onset + glide
+            t[chr(ord(base) + 0x60) + chr(ord(sm) + 0x60)] = t[base + sm]
+    return t
+
+
+_KANA_TABLE = _build_kana()
+_KANA_KEYS = sorted(_KANA_TABLE, key=len, reverse=True)
+
+
+def ja_word_to_ipa(w: str) -> str:
+    out = []
+    i, n = 0, len(w)
+    while i < n:
+        ch = w[i]
+        if ch in ("っ", "ッ"):
+            # sokuon: geminate the next onset consonant
+            nxt = _KANA_TABLE.get(w[i + 1:i + 3]) or \
+                _KANA_TABLE.get(w[i + 1:i + 2], "")
+            if nxt and nxt[0] not in "aiueo":
+                out.append(nxt[0])
+            i += 1
+            continue
+        if ch == "ー":  # long-vowel mark: repeat last vowel
+            if out and out[-1] and out[-1][-1] in "aiueoː":
+                out.append("ː")
+            i += 1
+            continue
+        two = w[i:i + 2]
+        if two in _KANA_TABLE:
+            out.append(_KANA_TABLE[two])
+            i += 2
+            continue
+        if ch in _KANA_TABLE:
+            out.append(_KANA_TABLE[ch])
+        i += 1  # kanji/unknown dropped (needs a reading dictionary)
+    s = "".join(out)
+    # vowel-sequence long vowels (とうきょう toukyou -> toːkjoː)
+    for pat, rep in (("ou", "oː"), ("oo", "oː"), ("uu", "uː"),
+                     ("ei", "eː"), ("aa", "aː"), ("ii", "iː"),
+                     ("ee", "eː")):
+        s = s.replace(pat, rep)
+    return s
+
+
+# --------------------------------------------------------------------- #
+# Vietnamese (vi): tones stripped (NFD), digraph table
+# --------------------------------------------------------------------- #
+_VI_TONES = {0x0300, 0x0301, 0x0303, 0x0309, 0x0323}
+
+
+def vi_preprocess(w: str) -> str:
+    # decompose, drop the 5 tone marks, recompose (quality diacritics
+    # like breve/circumflex/horn survive)
+    decomp = unicodedata.normalize("NFD", w)
+    kept = "".join(c for c in decomp if ord(c) not in _VI_TONES)
+    return unicodedata.normalize("NFC", kept)
+
+
+VI_RULES = {
+    "ngh": "ŋ", "ng": "ŋ", "nh": "ɲ", "gh": "ɡ", "gi": "z",
+    "kh": "x", "ph": "f", "th": "tʰ", "tr": "ʈ", "ch": "tɕ",
+    "qu": "kw", "đ": "d", "d": "z", "r": "z", "x": "s", "s": "s",
+    "iê": "iə", "yê": "iə", "uô": "uə", "ươ": "ɨə", "ay": "ai",
+    "ây": "əi", "ao": "au", "au": "əu", "âu": "əu", "oi": "ɔi",
+    "ôi": "oi", "ơi": "əi", "ui": "ui", "ưi": "ɨi", "eo": "ɛu",
+    "êu": "eu", "iu": "iu", "ưu": "ɨu",
+    "ă": "a", "â": "ə", "ê": "e", "ô": "o", "ơ": "ə", "ư": "ɨ",
+    "a": "aː", "b": "ɓ", "c": "k", "e": "ɛ", "g": "ɡ", "h": "h",
+    "i": "i", "k": "k", "l": "l", "m": "m", "n": "n", "o": "ɔ",
+    "p": "p", "t": "t", "u": "u", "v": "v", "y": "i",
+}
+
+
+def vi_postprocess(ipa: str) -> str:
+    # implosives approximated as plain voiced stops for the symbol set
+    return ipa.replace("ɓ", "b")
+
+
+# --------------------------------------------------------------------- #
+# Latin-script regulars
+# --------------------------------------------------------------------- #
+MI_RULES = {  # Māori: 10 consonants, 5 pure vowels
+    "wh": "f", "ng": "ŋ",
+    "ā": "aː", "ē": "eː", "ī": "iː", "ō": "oː", "ū": "uː",
+    "a": "a", "e": "e", "h": "h", "i": "i", "k": "k", "m": "m",
+    "n": "n", "o": "o", "p": "p", "r": "ɾ", "t": "t", "u": "u",
+    "w": "w",
+}
+
+HAW_RULES = {  # Hawaiian: ʻokina is a glottal stop
+    "ʻ": "ʔ", "'": "ʔ", "‘": "ʔ",
+    "ā": "aː", "ē": "eː", "ī": "iː", "ō": "oː", "ū": "uː",
+    "a": "a", "e": "e", "h": "h", "i": "i", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "o", "p": "p", "u": "u", "w": "v",
+}
+
+QU_RULES = {  # Quechua (southern): ejectives/aspirates approximated
+    "ch'": "tʃ", "chh": "tʃ", "ch": "tʃ", "ll": "ʎ", "ñ": "ɲ",
+    "ph": "pʰ", "th": "tʰ", "kh": "kʰ", "qh": "q", "k'": "k",
+    "p'": "p", "t'": "t", "q'": "q", "sh": "ʃ",
+    "a": "a", "e": "e", "h": "h", "i": "i", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "o", "p": "p", "q": "q", "r": "ɾ",
+    "s": "s", "t": "t", "u": "u", "w": "w", "y": "j",
+}
+
+GN_RULES = {  # Guaraní: nasal vowels written, y = ɨ
+    "mb": "mb", "nd": "nd", "ng": "ŋɡ", "nt": "nt", "ch": "ʃ",
+    "ã": "ã", "ẽ": "ẽ", "ĩ": "ĩ", "õ": "õ", "ũ": "ũ", "ỹ": "ɨ̃",
+    "á": "ˈa", "é": "ˈe", "í": "ˈi", "ó": "ˈo", "ú": "ˈu",
+    "ý": "ˈɨ", "'": "ʔ", "j": "dʒ", "ñ": "ɲ",
+    "a": "a", "e": "e", "g": "ɡ", "h": "h", "i": "i", "k": "k",
+    "l": "l", "m": "m", "n": "n", "o": "o", "p": "p", "r": "ɾ",
+    "s": "s", "t": "t", "u": "u", "v": "v", "y": "ɨ",
+}
+
+NCI_RULES = {  # Classical Nahuatl
+    "tl": "tɬ", "tz": "ts", "ch": "tʃ", "hu": "w", "uh": "w",
+    "cu": "kw", "uc": "kw", "qu": "k", "ce": "se", "ci": "si",
+    "x": "ʃ", "z": "s", "ll": "lː",
+    "ā": "aː", "ē": "eː", "ī": "iː", "ō": "oː",
+    "a": "a", "c": "k", "e": "e", "h": "h", "i": "i", "l": "l",
+    "m": "m", "n": "n", "o": "o", "p": "p", "t": "t", "u": "u",
+    "y": "j",
+}
+
+OM_RULES = {  # Oromo: ejectives approximated plain
+    "dh": "d", "ny": "ɲ", "sh": "ʃ", "ch": "tʃ", "ph": "p",
+    "ts": "ts", "aa": "aː", "ee": "eː", "ii": "iː", "oo": "oː",
+    "uu": "uː",
+    "a": "a", "b": "b", "c": "tʃ", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "h", "i": "i", "j": "dʒ", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "o", "p": "p", "q": "k", "r": "r",
+    "s": "s", "t": "t", "u": "u", "w": "w", "x": "t", "y": "j",
+}
+
+TN_RULES = {  # Setswana: g = /x/
+    "tlh": "tɬ", "tsh": "ts", "tl": "tɬ", "ts": "ts", "th": "tʰ",
+    "ph": "pʰ", "kh": "kʰ", "kg": "x", "ng": "ŋ", "ny": "ɲ",
+    "š": "ʃ", "sh": "ʃ",
+    "a": "a", "b": "b", "d": "d", "e": "e", "f": "f", "g": "x",
+    "h": "h", "i": "i", "j": "dʒ", "k": "k", "l": "l", "m": "m",
+    "n": "n", "o": "o", "p": "p", "r": "r", "s": "s", "t": "t",
+    "u": "u", "w": "w", "y": "j",
+}
+
+PAP_RULES = {  # Papiamento
+    "dj": "dʒ", "zj": "ʒ", "ch": "tʃ", "sh": "ʃ", "nj": "ɲ",
+    "è": "ɛ", "ò": "ɔ", "ù": "u", "ü": "y", "ñ": "ɲ",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "h", "i": "i", "j": "j", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "o", "p": "p", "r": "r", "s": "s",
+    "t": "t", "u": "u", "v": "v", "w": "w", "y": "j", "z": "z",
+}
+
+IA_RULES = {  # Interlingua
+    "ch": "k", "ph": "f", "th": "t", "qu": "kw",
+    "ce": "tse", "ci": "tsi", "ge": "ʒe", "gi": "ʒi",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "h", "i": "i", "j": "ʒ", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "o", "p": "p", "r": "r", "s": "s",
+    "t": "t", "u": "u", "v": "v", "w": "w", "x": "ks", "y": "i",
+    "z": "z",
+}
+
+IO_RULES = {  # Ido: fully regular by design
+    "ch": "tʃ", "sh": "ʃ", "qu": "kw",
+    "a": "a", "b": "b", "c": "ts", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "h", "i": "i", "j": "ʒ", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "o", "p": "p", "r": "r", "s": "s",
+    "t": "t", "u": "u", "v": "v", "w": "w", "x": "ks", "y": "j",
+    "z": "z",
+}
+
+LFN_RULES = {  # Lingua Franca Nova
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "h", "i": "i", "j": "ʒ", "l": "l", "m": "m",
+    "n": "n", "o": "o", "p": "p", "r": "r", "s": "s", "t": "t",
+    "u": "u", "v": "v", "x": "ʃ", "z": "z",
+}
+
+JBO_RULES = {  # Lojban: one letter = one phoneme by spec
+    "a": "a", "b": "b", "c": "ʃ", "d": "d", "e": "ɛ", "f": "f",
+    "g": "ɡ", "i": "i", "j": "ʒ", "k": "k", "l": "l", "m": "m",
+    "n": "n", "o": "o", "p": "p", "r": "r", "s": "s", "t": "t",
+    "u": "u", "v": "v", "x": "x", "y": "ə", "z": "z", "'": "h",
+}
+
+TK_RULES = {  # Turkmen: s/z are interdental
+    "ç": "tʃ", "ş": "ʃ", "ž": "ʒ", "ň": "ŋ", "ý": "j", "ä": "æ",
+    "ö": "ø", "ü": "y", "y": "ɯ", "w": "w", "j": "dʒ",
+    "a": "a", "b": "b", "d": "d", "e": "e", "f": "f", "g": "ɡ",
+    "h": "h", "i": "i", "k": "k", "l": "l", "m": "m", "n": "n",
+    "o": "o", "p": "p", "r": "r", "s": "θ", "t": "t", "u": "u",
+    "z": "ð",
+}
+
+LB_RULES = {  # Luxembourgish (approximate, German-adjacent)
+    "sch": "ʃ", "tsch": "tʃ", "ch": "ɕ", "ck": "k", "qu": "kv",
+    "tz": "ts",
+    "ue": "uə", "éi": "ɛɪ", "äi": "æɪ", "ei": "aɪ", "au": "aʊ",
+    "ou": "əʊ", "ie": "iə", "ee": "eː", "aa": "aː",
+    "é": "e", "ä": "ɛː", "ë": "ə", "â": "ɑː",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "ɛ", "f": "f",
+    "g": "ɡ", "h": "h", "i": "i", "j": "j", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "o", "p": "p", "r": "ʁ", "s": "s",
+    "t": "t", "u": "u", "v": "f", "w": "v", "x": "ks", "y": "i",
+    "z": "ts",
+}
+
+KL_RULES = {  # Greenlandic (approximate)
+    "ng": "ŋ", "rl": "ɬ", "ll": "ɬ", "gg": "ç", "rr": "χ",
+    "aa": "aː", "ii": "iː", "uu": "uː",
+    "a": "a", "e": "ə", "f": "f", "g": "ɣ", "i": "i", "j": "j",
+    "k": "k", "l": "l", "m": "m", "n": "n", "o": "o", "p": "p",
+    "q": "q", "r": "ʁ", "s": "s", "t": "t", "u": "u", "v": "v",
+}
+
+GA_RULES = {  # Irish (broad approximation; slender s handled)
+    "bhf": "v", "bh": "v", "mh": "v", "ch": "x", "dh": "ɣ",
+    "gh": "ɣ", "th": "h", "sh": "h", "fh": "", "ph": "f",
+    "ts": "t", "ng": "ŋ",
+    "aoi": "iː", "ao": "iː", "eai": "a", "ea": "a", "ai": "a",
+    "ei": "e", "io": "i", "iu": "u", "ui": "i", "eo": "oː",
+    "á": "aː", "é": "eː", "í": "iː", "ó": "oː", "ú": "uː",
+    "se": "ʃe", "si": "ʃi", "sé": "ʃeː", "sí": "ʃiː", "is": "iʃ",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "h", "i": "i", "l": "l", "m": "m", "n": "n",
+    "o": "o", "p": "p", "r": "ɾ", "s": "s", "t": "t", "u": "u",
+}
+
+# --------------------------------------------------------------------- #
+# Ancient Greek (grc): polytonic; breathings and accents via NFD
+# --------------------------------------------------------------------- #
+GRC_RULES = {
+    "γγ": "ŋɡ", "γκ": "ŋk", "γχ": "ŋkʰ",
+    "ου": "uː", "ει": "eː", "αι": "ai", "οι": "oi", "υι": "yi",
+    "αυ": "au", "ευ": "eu", "ηυ": "ɛːu",
+    "θ": "tʰ", "φ": "pʰ", "χ": "kʰ", "ψ": "ps", "ξ": "ks",
+    "η": "ɛː", "ω": "ɔː", "υ": "y",
+    "α": "a", "β": "b", "γ": "ɡ", "δ": "d", "ε": "e", "ζ": "z",
+    "ι": "i", "κ": "k", "λ": "l", "μ": "m", "ν": "n", "ο": "o",
+    "π": "p", "ρ": "r", "σ": "s", "ς": "s", "τ": "t",
+    "h": "h",  # injected by grc_preprocess (rough breathing)
+    "ˈ": "ˈ",  # stress mark injected from acute/circumflex accents
+}
+
+
+def grc_preprocess(w: str) -> str:
+    """Polytonic -> base letters; rough breathing becomes leading h,
+    acute/circumflex become a stress mark before the vowel."""
+    decomp = unicodedata.normalize("NFD", w)
+    out = []
+    rough = False
+    stress_at: Optional[int] = None
+    for ch in decomp:
+        cp = ord(ch)
+        if cp == 0x0314:          # rough breathing
+            rough = True
+        elif cp in (0x0301, 0x0342, 0x0300, 0x0345, 0x0313, 0x0308,
+                    0x0304, 0x0306):
+            if cp in (0x0301, 0x0342) and stress_at is None and out:
+                stress_at = len(out) - 1
+        else:
+            out.append(ch)
+    w2 = "".join(out)
+    if stress_at is not None:
+        w2 = w2[:stress_at] + "ˈ" + w2[stress_at:]
+    return ("h" + w2) if rough else w2
+
+
+# --------------------------------------------------------------------- #
+# Cyrillic Turkic: Tatar, Bashkir, Chuvash
+# --------------------------------------------------------------------- #
+TT_RULES = {  # Tatar
+    "ә": "æ", "ө": "ø", "ү": "y", "җ": "ʑ", "ң": "ŋ", "һ": "h",
+    "ы": "ɤ", "е": "e", "ё": "jo", "ю": "ju", "я": "ja", "э": "e",
+    "щ": "ɕ", "ъ": "", "ь": "", "ч": "ɕ", "ж": "ʒ",
+    "а": "ɑ", "б": "b", "в": "v", "г": "ɡ", "д": "d", "з": "z",
+    "и": "i", "й": "j", "к": "k", "л": "l", "м": "m", "н": "n",
+    "о": "o", "п": "p", "р": "r", "с": "s", "т": "t", "у": "u",
+    "ф": "f", "х": "x", "ц": "ts", "ш": "ʃ",
+}
+
+BA_RULES = {  # Bashkir: adds interdentals ҙ/ҫ and uvulars ғ/ҡ
+    "ә": "æ", "ө": "ø", "ү": "y", "ң": "ŋ", "һ": "h", "ҙ": "ð",
+    "ҫ": "θ", "ғ": "ʁ", "ҡ": "q", "ы": "ɯ", "е": "je", "ё": "jo",
+    "ю": "ju", "я": "ja", "э": "e", "щ": "ɕ", "ъ": "", "ь": "",
+    "а": "ɑ", "б": "b", "в": "v", "г": "ɡ", "д": "d", "ж": "ʒ",
+    "з": "z", "и": "i", "й": "j", "к": "k", "л": "l", "м": "m",
+    "н": "n", "о": "o", "п": "p", "р": "r", "с": "s", "т": "t",
+    "у": "u", "ф": "f", "х": "x", "ц": "ts", "ч": "tʃ", "ш": "ʃ",
+}
+
+CV_RULES = {  # Chuvash
+    "ӑ": "ə", "ӗ": "ə", "ҫ": "ɕ", "ӳ": "y", "е": "je", "ё": "jo",
+    "ю": "ju", "я": "ja", "э": "e", "ы": "ɯ", "щ": "ɕ", "ъ": "",
+    "ь": "ʲ", "ч": "tɕ",
+    "а": "a", "б": "p", "в": "ʋ", "г": "k", "д": "t", "ж": "ʃ",
+    "з": "s", "и": "i", "й": "j", "к": "k", "л": "l", "м": "m",
+    "н": "n", "о": "o", "п": "p", "р": "r", "с": "s", "т": "t",
+    "у": "u", "ф": "f", "х": "x", "ц": "ts", "ш": "ʃ",
+}
+
+
+# ===================================================================== #
+# Registry
+# ===================================================================== #
+TABLES3 = {
+    "fa": FA_RULES, "ur": UR_RULES, "ug": UG_RULES, "he": HE_RULES,
+    "vi": VI_RULES, "mi": MI_RULES, "haw": HAW_RULES, "qu": QU_RULES,
+    "gn": GN_RULES, "nci": NCI_RULES, "om": OM_RULES, "tn": TN_RULES,
+    "pap": PAP_RULES, "ia": IA_RULES, "io": IO_RULES, "lfn": LFN_RULES,
+    "jbo": JBO_RULES, "tk": TK_RULES, "lb": LB_RULES, "kl": KL_RULES,
+    "ga": GA_RULES, "grc": GRC_RULES, "tt": TT_RULES, "ba": BA_RULES,
+    "cv": CV_RULES,
+}
+
+_AR_BLOCK = "؀-ۿ"
+_HE_BLOCK = "֐-׿"
+_CYR = "а-яА-ЯёЁ"
+
+LETTERS3 = {
+    "fa": _AR_BLOCK, "ur": _AR_BLOCK, "ug": _AR_BLOCK,
+    "he": _HE_BLOCK + "'",
+    "vi": "a-zA-Zàáảãạăằắẳẵặâầấẩẫậèéẻẽẹêềếểễệìíỉĩịòóỏõọôồốổỗộơờớởỡợ"
+          "ùúủũụưừứửữựỳýỷỹỵđĐ",
+    "mi": "a-zA-Zāēīōū", "haw": "a-zA-Zāēīōūʻ'‘",
+    "qu": "a-zA-Z'", "gn": "a-zA-Zãẽĩõũỹáéíóúýñ'",
+    "nci": "a-zA-Zāēīō", "om": "a-zA-Z", "tn": "a-zA-Zš",
+    "pap": "a-zA-Zèòùüñ", "ia": "a-zA-Z", "io": "a-zA-Z",
+    "lfn": "a-zA-Z", "jbo": "a-z'", "tk": "a-zA-Zçäžňöşüý",
+    "lb": "a-zA-Zäéëâî", "kl": "a-zA-Z",
+    "ga": "a-zA-Záéíóú",
+    "grc": "α-ωΑ-Ωἀ-ῼάέήίόύώΐΰ",
+    "tt": _CYR + "әөүҗңһ", "ba": _CYR + "әөүңһҙҫғҡ",
+    "cv": _CYR + "ӑӗҫӳ",
+}
+
+STRESS3 = {
+    "fa": "final", "ur": "first", "ug": "final", "he": "final",
+    "vi": "none", "mi": "first", "haw": "penult", "qu": "penult",
+    "gn": "final", "nci": "penult", "om": "penult", "tn": "penult",
+    "pap": "penult", "ia": "penult", "io": "penult", "lfn": "penult",
+    "jbo": "penult", "tk": "final", "lb": "first", "kl": "first",
+    "ga": "first", "grc": "none", "tt": "final", "ba": "final",
+    "cv": "final",
+}
+
+LEXICONS3 = {"fa": FA_LEXICON, "ur": UR_LEXICON, "he": HE_LEXICON}
+PREPROCESS3 = {"vi": vi_preprocess, "he": he_preprocess,
+               "grc": grc_preprocess}
+POSTPROCESS3 = {"fa": fa_postprocess, "ur": ur_postprocess,
+                "he": he_postprocess, "vi": vi_postprocess}

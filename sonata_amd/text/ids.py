@@ -34,6 +34,9 @@ _IPA_SYMBOLS = (
     # third expansion batch (Indic/Hangul/Vietnamese): retroflex
     # lateral + approximant, velar approximant (Korean ㅢ)
     "ɭɻɰ",
+    # third batch, table languages: uvular stop/fricative (fa/kl),
+    # bilabial fricative (ja), precomposed nasal vowels (gn/ur)
+    "ɢɸχẽĩũ",
 )
 
 

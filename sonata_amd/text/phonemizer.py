@@ -123,11 +123,16 @@ class RuleG2P:
             ipa = post(ipa)
         if (self.stress and ipa and w not in self.unstressed
                 and "ˈ" not in ipa and "ˌ" not in ipa):
-            # fallback: primary stress before the first vowel
-            for i, ch in enumerate(ipa):
-                if ch in self._VOWELS:
-                    ipa = ipa[:i] + "ˈ" + ipa[i:]
-                    break
+            if self.stress_default != "first":
+                # unstressed lexicon entries follow the language's
+                # stress rule (fa final, es penult, ...)
+                ipa = self._stress_rules_output(w, ipa)
+            if "ˈ" not in ipa:
+                # fallback: primary stress before the first vowel
+                for i, ch in enumerate(ipa):
+                    if ch in self._VOWELS:
+                        ipa = ipa[:i] + "ˈ" + ipa[i:]
+                        break
         return ipa
 
     # -- regular inflections from base lexicon entries ------------------ #
@@ -708,6 +713,9 @@ _G2P_REGISTRY: Dict[str, RuleG2P] = {}
 
 _INDIC = ("mr", "ne", "bn", "as", "gu", "pa", "or", "ta", "te", "kn",
           "ml")
+_BATCH3 = ("fa", "ur", "ug", "he", "vi", "mi", "haw", "qu", "gn",
+           "nci", "om", "tn", "pap", "ia", "io", "lfn", "jbo", "tk",
+           "lb", "kl", "ga", "grc", "tt", "ba", "cv")
 
 
 def _get_g2p(voice: str) -> RuleG2P:
@@ -824,6 +832,31 @@ def _get_g2p(voice: str) -> RuleG2P:
             letters="؀-ۿ",
             stress=False,
         )
+    elif base == "ja":
+        # kana are an exact syllabary (gemination/long-vowel marks
+        # handled); kanji needs a reading dictionary — dropped, and
+        # PARITY.md states so
+        from .g2p_tables3 import ja_word_to_ipa
+
+        g = RuleG2P({}, letters="぀-ヿ", stress=False)
+        g._apply_rules = ja_word_to_ipa
+    elif base in _BATCH3:
+        from .g2p_tables3 import (LETTERS3, LEXICONS3, POSTPROCESS3,
+                                  PREPROCESS3, STRESS3, TABLES3)
+
+        stress_mode = STRESS3[base]
+        g = RuleG2P(TABLES3[base], LEXICONS3.get(base),
+                    letters=LETTERS3[base],
+                    stress=stress_mode != "none",
+                    preprocess=PREPROCESS3.get(base),
+                    stress_default=(stress_mode
+                                    if stress_mode != "none" else "first"))
+        post = POSTPROCESS3.get(base)
+        if post is not None:
+            # epenthesis/final-letter fixes apply to the RULE path only
+            # (lexicon entries are already surface forms)
+            orig = g._apply_rules
+            g._apply_rules = lambda w, _o=orig, _p=post: _p(_o(w))
     elif base in ("ko", "am", "chr"):
         # syllabic scripts decoded by codepoint arithmetic
         # (g2p_scripts.py): Hangul triples, Ethiopic rows, Cherokee CV
@@ -873,7 +906,16 @@ def available_languages() -> List[str]:
             # second expansion batch (g2p_tables.py TABLES2 + hi)
             "eo", "ca", "gl", "eu", "az", "kk", "ky", "uz", "mk", "be",
             "sl", "lt", "lv", "et", "is", "sq", "hy", "ka", "af", "cy",
-            "mt", "ht", "la", "hi"]
+            "mt", "ht", "la", "hi",
+            # third batch: Brahmic engine (g2p_indic.py)
+            "mr", "ne", "bn", "as", "gu", "pa", "or", "ta", "te", "kn",
+            "ml", "si",
+            # third batch: syllabic scripts (g2p_scripts.py) + kana ja
+            "ko", "am", "chr", "ja",
+            # third batch: rule tables (g2p_tables3.py)
+            "fa", "ur", "ug", "he", "vi", "mi", "haw", "qu", "gn",
+            "nci", "om", "tn", "pap", "ia", "io", "lfn", "jbo", "tk",
+            "lb", "kl", "ga", "grc", "tt", "ba", "cv"]
 
 
 # script-native sentence/clause punctuation -> ASCII so the splitter

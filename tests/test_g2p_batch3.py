@@ -173,6 +173,148 @@ def test_script_native_punctuation_maps_to_ascii():
     assert text_to_phonemes("ما اسمك؟", "ar")[0].endswith("?")
 
 
+# --------------------------------------------------------------------- #
+# Batch-3 rule-table languages (g2p_tables3.py)
+# --------------------------------------------------------------------- #
+def test_persian_epenthesis_and_lexicon():
+    fa = _get_g2p("fa")
+    assert fa.word_to_ipa("دوست") == "dˈuːst"    # final cluster kept
+    assert fa.word_to_ipa("من") == "mˈæn"        # 2-consonant word
+    assert fa.word_to_ipa("است") == "ˈæst"       # lexicon
+    # final ه is /e/
+    assert fa.word_to_ipa("خانه") == "xɒːnˈe"
+
+
+def test_urdu_aspiration_digraphs():
+    ur = _get_g2p("ur")
+    assert "ɡʰ" in ur.word_to_ipa("گھر")
+    assert "ʈ" in ur.word_to_ipa("ٹوپی")
+
+
+def test_uyghur_vocalized_script():
+    ug = _get_g2p("ug")
+    # fully written vowels: no epenthesis needed
+    assert ug.word_to_ipa("مەن") == "mˈæn"
+    assert "ʁ" in ug.word_to_ipa("ئۇيغۇر")
+
+
+def test_hebrew_finals_and_lexicon():
+    he = _get_g2p("he")
+    assert he.word_to_ipa("שלום") == "ʃalˈom"    # lexicon + final stress
+    # final-form letters map like their medial forms
+    assert he.word_to_ipa("ים").endswith("m")
+
+
+def test_japanese_kana():
+    ja = _get_g2p("ja")
+    assert ja.word_to_ipa("とうきょう") == "toːkjoː"   # long vowels
+    assert ja.word_to_ipa("がっこう") == "ɡakkoː"     # sokuon geminates
+    assert ja.word_to_ipa("しんぶん") == "ʃinbun"
+    assert ja.word_to_ipa("キャンプ") == "kjanpu"     # katakana + glide
+    assert ja.word_to_ipa("ラーメン") == "ɾaːmen"     # chōonpu
+
+
+def test_vietnamese_tones_stripped():
+    vi = _get_g2p("vi")
+    # same segmental output regardless of tone
+    assert vi.word_to_ipa("má") == vi.word_to_ipa("mà") \
+        == vi.word_to_ipa("ma")
+    assert vi.word_to_ipa("tiếng") == "tiəŋ"
+    assert vi.word_to_ipa("chào") == "tɕau"
+
+
+def test_ancient_greek_polytonic():
+    grc = _get_g2p("grc")
+    # rough breathing -> h; accents -> stress; aspirated stops
+    assert grc.word_to_ipa("ἡ") == "hɛː"
+    assert grc.word_to_ipa("ἄνθρωπος") == "ˈantʰrɔːpos"
+    assert grc.word_to_ipa("ψυχή") == "psykʰˈɛː"
+
+
+def test_maori_hawaiian():
+    assert _get_g2p("mi").word_to_ipa("whenua") == "fˈenua"
+    assert _get_g2p("haw").word_to_ipa("ʻohana") == "ʔohˈana"
+
+
+def test_nahuatl_tl_and_saltillo():
+    assert "tɬ" in _get_g2p("nci").word_to_ipa("tlahtolli")
+
+
+def test_tswana_g_is_velar_fricative():
+    assert _get_g2p("tn").word_to_ipa("kgosi") == "xˈosi"
+
+
+def test_turkmen_interdentals():
+    ipa = _get_g2p("tk").word_to_ipa("sez")
+    assert ipa.startswith("θ") and ipa.endswith("ð")
+
+
+def test_cyrillic_turkic_letters():
+    assert "æ" in _get_g2p("tt").word_to_ipa("сәлам")
+    assert "θ" in _get_g2p("ba").word_to_ipa("ҫук")
+    assert "ɕ" in _get_g2p("cv").word_to_ipa("ҫырать")
+
+
+def test_batch3_all_languages_nonempty():
+    samples = {
+        "fa": "سلام دوست من", "ur": "شکریہ بہت", "ug": "مەن ياخشى",
+        "he": "שלום חבר", "ja": "ありがとう ございます",
+        "vi": "xin chào bạn", "mi": "kia ora", "haw": "aloha nui",
+        "qu": "allin p'unchay", "gn": "mba'éichapa", "nci": "niltze",
+        "om": "akkam jirta", "tn": "dumela rra", "pap": "bon dia",
+        "ia": "bon die", "io": "bona jorno", "lfn": "bon dia",
+        "jbo": "coi rodo", "tk": "salam dost", "lb": "moien alleguer",
+        "kl": "aluu ikinngut", "ga": "dia duit", "grc": "χαῖρε φίλε",
+        "tt": "исәнмесез дуслар", "ba": "һаумыһығыҙ дустар",
+        "cv": "салам туссем",
+    }
+    from sonata_amd.text.phonemizer import _BATCH3
+    assert set(samples) == set(_BATCH3) | {"ja"}
+    for lang, txt in samples.items():
+        out = text_to_phonemes(txt, lang)
+        assert out and out[0].strip("."), (lang, out)
+
+
+def test_batch3_symbols_encodable():
+    """Every symbol the batch-3 engines emit must encode through the
+    voice id map (ja/ko/etc. must not leak unknown codepoints)."""
+    from sonata_amd.text.ids import default_phoneme_id_map
+    id_map = default_phoneme_id_map()
+    samples = {
+        "fa": "سلام من از ایران هستم دوست خوب قلم",
+        "ur": "میں اردو بولتا ہوں گھر ٹھیک",
+        "ug": "مەن ئۇيغۇرچە سۆزلەيمەن ياخشى",
+        "he": "שלום אני מדבר עברית תודה צדק",
+        "ja": "こんにちは とうきょう がっこう ふじさん キャンプ",
+        "ko": "안녕하세요 한국어를 공부합니다 의사 좋아요",
+        "am": "ሰላም ለዓለም አማርኛ እናገራለሁ ኳስ",
+        "chr": "ᏣᎳᎩ ᎦᏬᏂᎯᏍᏗ ᎣᏏᏲ",
+        "vi": "tôi nói tiếng việt xin chào được người",
+        "gn": "che añe'ẽ guaraníme mba'éichapa porã",
+        "qu": "runasimi rimani allin p'unchay llaqta",
+        "nci": "nahuatlahtolli cualli tlahtoa xochitl",
+        "kl": "kalaallisut oqaluppunga illoqarfik",
+        "grc": "ἄνθρωπος καὶ θεός ἡ ψυχή χαῖρε",
+        "tk": "men türkmençe gepleýärin ýagşy",
+        "tt": "мин татарча сөйләшәм җыр һава",
+        "ba": "мин башҡортса һөйләшәм ҙур ҫук",
+        "cv": "эпӗ чӑвашла калаҫатӑп ҫырать",
+        "mi": "kia ora he tangata whenua", "haw": "aloha ʻohana kākou",
+        "om": "akkam jirta nagaa", "tn": "dumela kgosi tlhapi",
+        "pap": "mi ta papia papiamentu djaluna",
+        "ia": "io parla interlingua", "io": "me parolas ido",
+        "lfn": "me parla elefen", "jbo": "mi tavla fo la lojban",
+        "lb": "ech schwätzen lëtzebuergesch moien",
+        "ga": "tá gaeilge agam go raibh maith agat",
+    }
+    for lang, txt in samples.items():
+        for sent in text_to_phonemes(txt, lang):
+            for ch in sent.replace(" ", ""):
+                if ch in ".,;:?!":
+                    continue
+                assert ch in id_map, (lang, ch, hex(ord(ch)), sent)
+
+
 def test_indic_symbols_encodable():
     """Every IPA char the Indic engines emit must be in the voice
     symbol table (ids.py) so it survives phonemes->ids encoding."""

@@ -100,6 +100,30 @@ def test_all_languages_end_to_end(tmp_path):
         "cy": "Helo byd.", "mt": "Bongu dinja.",
         "ht": "Bonjou monn.", "la": "Salve munde.",
         "hi": "नमस्ते दुनिया.",
+        # third batch: Brahmic engine
+        "mr": "नमस्कार जग.", "ne": "नमस्ते संसार.",
+        "bn": "ওহে বিশ্ব.", "as": "নমস্কাৰ পৃথিৱী.",
+        "gu": "નમસ્તે દુનિયા.", "pa": "ਸਤ ਸ੍ਰੀ ਅਕਾਲ ਦੁਨਿਆ.",
+        "or": "ନମସ୍କାର ଜଗତ.", "ta": "வணக்கம் உலகம்.",
+        "te": "నమస్కారం ప్రపంచం.", "kn": "ನಮಸ್ಕಾರ ಜಗತ್ತು.",
+        "ml": "നമസ്കാരം ലോകം.", "si": "ආයුබෝවන් ලෝකය.",
+        # third batch: syllabic scripts + kana
+        "ko": "안녕하세요 세계.", "am": "ሰላም ለዓለም.",
+        "chr": "ᎣᏏᏲ ᎡᎶᎯ.", "ja": "こんにちは せかい.",
+        # third batch: rule tables
+        "fa": "سلام دنیا.", "ur": "ہیلو دنیا.",
+        "ug": "سالام دۇنيا.", "he": "שלום עולם.",
+        "vi": "Chào thế giới.", "mi": "Kia ora te ao.",
+        "haw": "Aloha honua.", "qu": "Allin p'unchay pacha.",
+        "gn": "Mba'éichapa arapy.", "nci": "Niltze cemanahuac.",
+        "om": "Akkam addunyaa.", "tn": "Dumela lefatshe.",
+        "pap": "Bon dia mundu.", "ia": "Salute mundo.",
+        "io": "Saluto mondo.", "lfn": "Saluta mundo.",
+        "jbo": "coi le munje", "tk": "Salam dünýä.",
+        "lb": "Moien Welt.", "kl": "Aluu silarsuaq.",
+        "ga": "Dia duit a dhomhain.", "grc": "χαῖρε κόσμε.",
+        "tt": "Сәлам дөнья.", "ba": "Сәләм донъя.",
+        "cv": "Салам тӗнче.",
     }
     for lang in available_languages():
         pack = create_random_voice(str(tmp_path), f"lang_{lang}",
