@@ -195,7 +195,7 @@ def _bn_engine(assamese: bool = False) -> BrahmicG2P:
 
 
 def make_engine(lang: str) -> BrahmicG2P:
-    if lang in ("hi", "mr", "ne"):
+    if lang in ("hi", "mr", "ne", "kok"):
         # Devanagari: Hindi vowel quality (ɪ/ʊ lax short vowels, ɑː)
         return BrahmicG2P(
             0x0, "ə", final_del=True,
@@ -234,14 +234,15 @@ def make_engine(lang: str) -> BrahmicG2P:
 
 # word-regex letter ranges per script block
 INDIC_LETTERS = {
-    "mr": "ऀ-ॿ", "ne": "ऀ-ॿ", "bn": "ঀ-৿", "as": "ঀ-৿",
+    "mr": "ऀ-ॿ", "ne": "ऀ-ॿ", "kok": "ऀ-ॿ",
+    "bn": "ঀ-৿", "as": "ঀ-৿",
     "gu": "઀-૿", "pa": "਀-੿", "or": "଀-୿",
     "ta": "஀-௿", "te": "ఀ-౿", "kn": "ಀ-೿",
     "ml": "ഀ-ൿ", "si": "඀-෿",
 }
 
-INDIC_LANGS = ("mr", "ne", "bn", "as", "gu", "pa", "or", "ta", "te",
-               "kn", "ml")
+INDIC_LANGS = ("mr", "ne", "kok", "bn", "as", "gu", "pa", "or", "ta",
+               "te", "kn", "ml")
 
 
 # --------------------------------------------------------------------- #

@@ -177,13 +177,151 @@ def chr_word_to_ipa(w: str) -> str:
     return "".join(out)
 
 
+# --------------------------------------------------------------------- #
+# Burmese (my): abugida with inherent /a/; Unicode stores logical order
+# (the e-vowel sign U+1031 follows its consonant in memory even though
+# it renders before it), so a left-to-right scan works.  Tones and the
+# stacked-consonant rhyme changes are approximated away.
+# --------------------------------------------------------------------- #
+_MY_CONS = {
+    "က": "k", "ခ": "kʰ", "ဂ": "ɡ", "ဃ": "ɡ", "င": "ŋ",
+    "စ": "s", "ဆ": "s", "ဇ": "z", "ဈ": "z", "ဉ": "ɲ", "ည": "ɲ",
+    "ဋ": "t", "ဌ": "tʰ", "ဍ": "d", "ဎ": "d", "ဏ": "n",
+    "တ": "t", "ထ": "tʰ", "ဒ": "d", "ဓ": "d", "န": "n",
+    "ပ": "p", "ဖ": "pʰ", "ဗ": "b", "ဘ": "b", "မ": "m",
+    "ယ": "j", "ရ": "j", "လ": "l", "ဝ": "w", "သ": "θ",
+    "ဟ": "h", "ဠ": "l", "အ": "",
+}
+_MY_VOWEL_SIGNS = {
+    "ါ": "aː", "ာ": "aː", "ိ": "i", "ီ": "iː", "ု": "u",
+    "ူ": "uː", "ေ": "eː", "ဲ": "ɛː",
+}
+_MY_MEDIALS = {"ျ": "j", "ြ": "j", "ွ": "w", "ှ": "h"}
+_MY_INDEP = {"ဣ": "i", "ဤ": "iː", "ဥ": "u", "ဦ": "uː", "ဧ": "eː",
+             "ဩ": "ɔː", "ဪ": "ɔː"}
+_MY_ASAT = "်"     # U+103A: kills the inherent vowel (syllable coda)
+_MY_ANUSVARA = "ံ"  # U+1036
+
+
+def my_word_to_ipa(w: str) -> str:
+    out = []
+    chars = list(w)
+    i, n = 0, len(chars)
+    while i < n:
+        ch = chars[i]
+        if ch in _MY_CONS:
+            seg = _MY_CONS[ch]
+            i += 1
+            while i < n and chars[i] in _MY_MEDIALS:
+                seg += _MY_MEDIALS[chars[i]]
+                i += 1
+            vowel = None
+            # ော (e + aa) = ɔː; otherwise single signs
+            if i + 1 < n and chars[i] == "ေ" and chars[i + 1] in "ာါ":
+                vowel = "ɔː"
+                i += 2
+            elif i < n and chars[i] in _MY_VOWEL_SIGNS:
+                vowel = _MY_VOWEL_SIGNS[chars[i]]
+                i += 1
+                if vowel == "i" and i < n and chars[i] == "ု":
+                    vowel = "o"  # ို
+                    i += 1
+            if i < n and chars[i] == _MY_ASAT:
+                i += 1
+                out.append(seg)  # coda consonant, no vowel
+                continue
+            out.append(seg + (vowel if vowel is not None else "a"))
+        elif ch in _MY_INDEP:
+            out.append(_MY_INDEP[ch])
+            i += 1
+        elif ch == _MY_ANUSVARA:
+            out.append("n")
+            i += 1
+        else:
+            i += 1  # tones (့ း), virama stacking, digits: dropped
+    return "".join(out)
+
+
+# --------------------------------------------------------------------- #
+# Thai (th): abugida, no spaces; prefix vowels (เแโใไ) are stored
+# BEFORE their consonant so they are stashed and emitted after it.
+# Unwritten inherent vowels are approximated by epenthesis.
+# --------------------------------------------------------------------- #
+_TH_CONS = {
+    "ก": "k", "ข": "kʰ", "ฃ": "kʰ", "ค": "kʰ", "ฅ": "kʰ", "ฆ": "kʰ",
+    "ง": "ŋ", "จ": "tɕ", "ฉ": "tɕʰ", "ช": "tɕʰ", "ซ": "s",
+    "ฌ": "tɕʰ", "ญ": "j", "ฎ": "d", "ฏ": "t", "ฐ": "tʰ", "ฑ": "tʰ",
+    "ฒ": "tʰ", "ณ": "n", "ด": "d", "ต": "t", "ถ": "tʰ", "ท": "tʰ",
+    "ธ": "tʰ", "น": "n", "บ": "b", "ป": "p", "ผ": "pʰ", "ฝ": "f",
+    "พ": "pʰ", "ฟ": "f", "ภ": "pʰ", "ม": "m", "ย": "j", "ร": "r",
+    "ล": "l", "ว": "w", "ศ": "s", "ษ": "s", "ส": "s", "ห": "h",
+    "ฬ": "l", "อ": "ʔ", "ฮ": "h",
+}
+_TH_AFTER = {
+    "ะ": "a", "ั": "a", "า": "aː", "ำ": "am", "ิ": "i", "ี": "iː",
+    "ึ": "ɯ", "ื": "ɯː", "ุ": "u", "ู": "uː",
+}
+_TH_PREFIX = {"เ": "eː", "แ": "ɛː", "โ": "oː", "ใ": "aj", "ไ": "aj"}
+# prefix เ + following sign combos
+_TH_E_COMBOS = {"า": "aw", "ิ": "ɤː", "ี": "ia", "ื": "ɯa", "อ": "ɤː"}
+_TH_TONES = {"่", "้", "๊", "๋", "็", "์", "ๆ", "ฯ"}
+
+
+def th_word_to_ipa(w: str) -> str:
+    toks = []
+    chars = [c for c in w if c not in _TH_TONES]
+    i, n = 0, len(chars)
+    pending = None  # stashed prefix vowel
+    while i < n:
+        ch = chars[i]
+        if ch in _TH_PREFIX:
+            pending = _TH_PREFIX[ch]
+            i += 1
+            continue
+        if ch in _TH_CONS:
+            onset = _TH_CONS[ch]
+            i += 1
+            # second consonant of a cluster (กร, กล, กว)
+            if (pending is not None and i < n
+                    and chars[i] in ("ร", "ล", "ว")
+                    and i + 1 < n and chars[i + 1] in _TH_AFTER):
+                onset += _TH_CONS[chars[i]]
+                i += 1
+            vowel = None
+            if pending == "eː" and i < n and chars[i] in _TH_E_COMBOS:
+                vowel = _TH_E_COMBOS[chars[i]]
+                i += 1
+                pending = None
+            elif i < n and chars[i] in _TH_AFTER:
+                vowel = _TH_AFTER[chars[i]]
+                i += 1
+            if pending is not None:
+                vowel = pending + (vowel or "")
+                pending = None
+            toks.append(onset if onset != "ʔ" else "")
+            if vowel:
+                toks.append(vowel)
+                # ไ-C-ย: the trailing ย is silent (ไทย = tʰaj)
+                if vowel == "aj" and i < n and chars[i] == "ย":
+                    i += 1
+        else:
+            i += 1
+    ipa = "".join(toks)
+    from .g2p_tables3 import epenthesize
+    return epenthesize(ipa, "a")
+
+
 SCRIPT_LETTERS = {
     "ko": "가-힣",
     "am": "ሀ-ፚ",
     "chr": "Ꭰ-Ᏼ",
+    "my": "က-ဿ",
+    "th": "ก-๛",
 }
 SCRIPT_FUNCS = {
     "ko": ko_word_to_ipa,
     "am": am_word_to_ipa,
     "chr": chr_word_to_ipa,
+    "my": my_word_to_ipa,
+    "th": th_word_to_ipa,
 }

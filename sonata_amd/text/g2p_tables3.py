@@ -27,8 +27,9 @@ from typing import Dict, Optional
 # --------------------------------------------------------------------- #
 _MULTI_PHONES = ("tʃʰ", "dʒʰ", "tʃ", "dʒ", "ts", "dz", "kʰ", "ɡʰ",
                  "pʰ", "bʰ", "tʰ", "dʰ", "ʈʰ", "ɖʰ", "sˤ", "dˤ", "tˤ",
-                 "ðˤ", "aː", "iː", "uː", "eː", "oː", "ɒː", "æː", "ɛː")
-_VOWEL_START = set("aeiouəɑɔæɛɪʊʌɒɯɨyøœɶʏɤ")
+                 "ðˤ", "aː", "iː", "uː", "eː", "oː", "ɒː", "æː", "ɛː",
+                 "ɑː", "ɔː", "ɯː", "øː", "yː", "ʊː", "ɪː")
+_VOWEL_START = set("aeiouəɑɔæɛɪʊʌɒɯɨyøœɶʏɤː")
 
 
 def _tokenize_ipa(ipa: str):
@@ -548,6 +549,88 @@ CV_RULES = {  # Chuvash
 }
 
 
+AN_RULES = {  # Aragonese (Spanish-adjacent; x = /ʃ/)
+    "ch": "tʃ", "ll": "ʎ", "ny": "ɲ", "rr": "r", "qu": "k",
+    "gue": "ɡe", "gui": "ɡi", "ce": "θe", "ci": "θi",
+    "á": "ˈa", "é": "ˈe", "í": "ˈi", "ó": "ˈo", "ú": "ˈu", "ñ": "ɲ",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "", "i": "i", "j": "x", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "o", "p": "p", "r": "ɾ", "s": "s",
+    "t": "t", "u": "u", "v": "b", "w": "w", "x": "ʃ", "y": "j",
+    "z": "θ",
+}
+
+KU_RULES = {  # Kurdish (Kurmanji, Hawar Latin alphabet — regular)
+    "ç": "tʃ", "ş": "ʃ", "ê": "eː", "î": "iː", "û": "uː",
+    "c": "dʒ", "j": "ʒ", "x": "x", "q": "q",
+    "a": "aː", "b": "b", "d": "d", "e": "ɛ", "f": "f", "g": "ɡ",
+    "h": "h", "i": "ɪ", "k": "k", "l": "l", "m": "m", "n": "n",
+    "o": "o", "p": "p", "r": "r", "s": "s", "t": "t", "u": "u",
+    "v": "v", "w": "w", "y": "j", "z": "z",
+}
+
+GD_RULES = {  # Scottish Gaelic (broad approximation, like ga)
+    "bh": "v", "mh": "v", "ch": "x", "dh": "ɣ", "gh": "ɣ",
+    "th": "h", "sh": "h", "fh": "", "ph": "f", "chd": "xk",
+    "ao": "ɯː", "eu": "ia", "ia": "iə", "ua": "uə",
+    "à": "aː", "è": "ɛː", "é": "eː", "ì": "iː", "ò": "ɔː",
+    "ó": "oː", "ù": "uː", "ai": "a", "ea": "ɛ", "ei": "e",
+    "io": "i", "ui": "u",
+    "se": "ʃe", "si": "ʃi", "sì": "ʃiː", "sè": "ʃɛː",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "h", "i": "i", "l": "l", "m": "m", "n": "n",
+    "o": "ɔ", "p": "p", "r": "ɾ", "s": "s", "t": "t", "u": "u",
+}
+
+QUC_RULES = {  # K'iche' (glottalized series approximated plain)
+    "tz'": "ts", "ch'": "tʃ", "tz": "ts", "ch": "tʃ", "b'": "b",
+    "k'": "k", "q'": "q", "t'": "t", "x": "ʃ", "j": "x", "'": "ʔ",
+    "ä": "ə",
+    "a": "a", "b": "b", "e": "e", "i": "i", "k": "k", "l": "l",
+    "m": "m", "n": "n", "o": "o", "p": "p", "q": "q", "r": "r",
+    "s": "s", "t": "t", "u": "u", "w": "w", "y": "j",
+}
+
+SD_RULES = {  # Sindhi (Arabic script; implosives approximated voiced)
+    "ڪھ": "kʰ", "گھ": "ɡʰ", "جھ": "dʒʰ", "ڙھ": "ɾʰ",
+    "آ": "ɑː", "ا": "ɑː", "ب": "b", "ٻ": "b", "ڀ": "bʰ", "ت": "t",
+    "ٿ": "tʰ", "ٽ": "ʈ", "ٺ": "ʈʰ", "ث": "s", "پ": "p", "ج": "dʒ",
+    "ڄ": "dʒ", "جهہ": "dʒʰ", "ڃ": "ɲ", "چ": "tʃ", "ڇ": "tʃʰ",
+    "ح": "h", "خ": "x", "د": "d", "ڌ": "dʰ", "ڏ": "ɖ", "ڊ": "ɖ",
+    "ڍ": "ɖʰ", "ذ": "z", "ر": "r", "ڙ": "ɾ", "ز": "z", "س": "s",
+    "ش": "ʃ", "ص": "s", "ض": "z", "ط": "t", "ظ": "z", "ع": "ʔ",
+    "غ": "ɣ", "ف": "f", "ڦ": "pʰ", "ق": "q", "ڪ": "k", "ک": "kʰ",
+    "گ": "ɡ", "ڳ": "ɡ", "ڱ": "ŋ", "ل": "l", "م": "m", "ن": "n",
+    "ڻ": "ɳ", "ڽ": "ɲ", "و": "uː", "ه": "h", "ھ": "h", "ء": "ʔ",
+    "ي": "iː", "ی": "iː", "ے": "eː",
+    "َ": "ə", "ُ": "ʊ", "ِ": "ɪ", "ّ": "", "ْ": "",
+}
+
+
+def sd_postprocess(ipa: str) -> str:
+    return epenthesize(ipa, "ə")
+
+
+NOG_RULES = {  # Nogai (Cyrillic; аь/оь/уь front-vowel digraphs)
+    "аь": "æ", "оь": "ø", "уь": "y", "нъ": "ŋ", "ё": "jo",
+    "ю": "ju", "я": "ja", "э": "e", "щ": "ɕ", "ъ": "", "ь": "",
+    "а": "ɑ", "б": "b", "в": "v", "г": "ɡ", "д": "d", "е": "e",
+    "ж": "ʒ", "з": "z", "и": "i", "й": "j", "к": "k", "л": "l",
+    "м": "m", "н": "n", "о": "o", "п": "p", "р": "r", "с": "s",
+    "т": "t", "у": "u", "ф": "f", "х": "x", "ц": "ts", "ч": "tʃ",
+    "ш": "ʃ", "ы": "ɯ",
+}
+
+SMJ_RULES = {  # Lule Sami (approximate)
+    "tj": "tʃ", "dj": "dʒ", "nj": "ɲ", "sj": "ʃ", "ts": "ts",
+    "á": "aː", "å": "oː", "ŋ": "ŋ", "æ": "æ", "ä": "æ",
+    "a": "a", "b": "b", "d": "d", "e": "e", "f": "f", "g": "ɡ",
+    "h": "h", "i": "i", "j": "j", "k": "k", "l": "l", "m": "m",
+    "n": "n", "o": "o", "p": "p", "r": "r", "s": "s", "t": "t",
+    "u": "u", "v": "v",
+}
+
+
 # ===================================================================== #
 # Registry
 # ===================================================================== #
@@ -559,6 +642,8 @@ TABLES3 = {
     "jbo": JBO_RULES, "tk": TK_RULES, "lb": LB_RULES, "kl": KL_RULES,
     "ga": GA_RULES, "grc": GRC_RULES, "tt": TT_RULES, "ba": BA_RULES,
     "cv": CV_RULES,
+    "an": AN_RULES, "ku": KU_RULES, "gd": GD_RULES, "quc": QUC_RULES,
+    "sd": SD_RULES, "nog": NOG_RULES, "smj": SMJ_RULES,
 }
 
 _AR_BLOCK = "؀-ۿ"
@@ -580,6 +665,9 @@ LETTERS3 = {
     "grc": "α-ωΑ-Ωἀ-ῼάέήίόύώΐΰ",
     "tt": _CYR + "әөүҗңһ", "ba": _CYR + "әөүңһҙҫғҡ",
     "cv": _CYR + "ӑӗҫӳ",
+    "an": "a-zA-Zñáéíóú", "ku": "a-zA-Zçşêîû",
+    "gd": "a-zA-Zàèéìòóù", "quc": "a-zA-Zä'",
+    "sd": _AR_BLOCK, "nog": _CYR, "smj": "a-zA-Záåŋæä",
 }
 
 STRESS3 = {
@@ -590,10 +678,13 @@ STRESS3 = {
     "jbo": "penult", "tk": "final", "lb": "first", "kl": "first",
     "ga": "first", "grc": "none", "tt": "final", "ba": "final",
     "cv": "final",
+    "an": "es-penult", "ku": "final", "gd": "first", "quc": "final",
+    "sd": "first", "nog": "final", "smj": "first",
 }
 
 LEXICONS3 = {"fa": FA_LEXICON, "ur": UR_LEXICON, "he": HE_LEXICON}
 PREPROCESS3 = {"vi": vi_preprocess, "he": he_preprocess,
                "grc": grc_preprocess}
 POSTPROCESS3 = {"fa": fa_postprocess, "ur": ur_postprocess,
-                "he": he_postprocess, "vi": vi_postprocess}
+                "he": he_postprocess, "vi": vi_postprocess,
+                "sd": sd_postprocess}

@@ -711,11 +711,12 @@ _EN_LETTERS = {
 
 _G2P_REGISTRY: Dict[str, RuleG2P] = {}
 
-_INDIC = ("mr", "ne", "bn", "as", "gu", "pa", "or", "ta", "te", "kn",
-          "ml")
+_INDIC = ("mr", "ne", "kok", "bn", "as", "gu", "pa", "or", "ta", "te",
+          "kn", "ml")
 _BATCH3 = ("fa", "ur", "ug", "he", "vi", "mi", "haw", "qu", "gn",
            "nci", "om", "tn", "pap", "ia", "io", "lfn", "jbo", "tk",
-           "lb", "kl", "ga", "grc", "tt", "ba", "cv")
+           "lb", "kl", "ga", "grc", "tt", "ba", "cv",
+           "an", "ku", "gd", "quc", "sd", "nog", "smj")
 
 
 def _get_g2p(voice: str) -> RuleG2P:
@@ -857,7 +858,7 @@ def _get_g2p(voice: str) -> RuleG2P:
             # (lexicon entries are already surface forms)
             orig = g._apply_rules
             g._apply_rules = lambda w, _o=orig, _p=post: _p(_o(w))
-    elif base in ("ko", "am", "chr"):
+    elif base in ("ko", "am", "chr", "my", "th"):
         # syllabic scripts decoded by codepoint arithmetic
         # (g2p_scripts.py): Hangul triples, Ethiopic rows, Cherokee CV
         from .g2p_scripts import SCRIPT_FUNCS, SCRIPT_LETTERS
@@ -908,14 +909,15 @@ def available_languages() -> List[str]:
             "sl", "lt", "lv", "et", "is", "sq", "hy", "ka", "af", "cy",
             "mt", "ht", "la", "hi",
             # third batch: Brahmic engine (g2p_indic.py)
-            "mr", "ne", "bn", "as", "gu", "pa", "or", "ta", "te", "kn",
-            "ml", "si",
+            "mr", "ne", "kok", "bn", "as", "gu", "pa", "or", "ta",
+            "te", "kn", "ml", "si",
             # third batch: syllabic scripts (g2p_scripts.py) + kana ja
-            "ko", "am", "chr", "ja",
+            "ko", "am", "chr", "my", "th", "ja",
             # third batch: rule tables (g2p_tables3.py)
             "fa", "ur", "ug", "he", "vi", "mi", "haw", "qu", "gn",
             "nci", "om", "tn", "pap", "ia", "io", "lfn", "jbo", "tk",
-            "lb", "kl", "ga", "grc", "tt", "ba", "cv"]
+            "lb", "kl", "ga", "grc", "tt", "ba", "cv",
+            "an", "ku", "gd", "quc", "sd", "nog", "smj"]
 
 
 # script-native sentence/clause punctuation -> ASCII so the splitter

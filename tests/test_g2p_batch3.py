@@ -255,6 +255,36 @@ def test_cyrillic_turkic_letters():
     assert "ɕ" in _get_g2p("cv").word_to_ipa("ҫырать")
 
 
+def test_burmese_engine():
+    my = _get_g2p("my")
+    # medial ြ = j, asat coda, inherent a
+    assert my.word_to_ipa("မြန်မာ") == "mjanmaː"
+    assert my.word_to_ipa("မင်္ဂလာပါ").startswith("m")
+
+
+def test_thai_engine():
+    th = _get_g2p("th")
+    # prefix vowel emitted after its consonant; silent ย after ไ
+    assert th.word_to_ipa("ไทย") == "tʰaj"
+    assert th.word_to_ipa("เขา") == "kʰaw"
+    out = th.word_to_ipa("สวัสดี")
+    assert out.startswith("s") and "diː" in out
+
+
+def test_konkani_rides_devanagari_engine():
+    assert _get_g2p("kok").word_to_ipa("कमल") == "kəməl"
+
+
+def test_batch3b_tables():
+    assert _get_g2p("an").word_to_ipa("muller") == "muʎˈeɾ"
+    assert _get_g2p("ku").word_to_ipa("cîhan") == "dʒiːhˈaːn"
+    assert "x" in _get_g2p("gd").word_to_ipa("loch")
+    assert "ʃ" in _get_g2p("quc").word_to_ipa("xela")
+    assert _get_g2p("sd").word_to_ipa("سنڌ").startswith("s")
+    assert "æ" in _get_g2p("nog").word_to_ipa("аьел")
+    assert "ʃ" in _get_g2p("smj").word_to_ipa("sjaddat")
+
+
 def test_batch3_all_languages_nonempty():
     samples = {
         "fa": "سلام دوست من", "ur": "شکریہ بہت", "ug": "مەن ياخشى",
@@ -267,6 +297,9 @@ def test_batch3_all_languages_nonempty():
         "kl": "aluu ikinngut", "ga": "dia duit", "grc": "χαῖρε φίλε",
         "tt": "исәнмесез дуслар", "ba": "һаумыһығыҙ дустар",
         "cv": "салам туссем",
+        "an": "ola mundo", "ku": "silav cîhan", "gd": "halò saoghail",
+        "quc": "saqarik uleew", "sd": "سلام دنيا",
+        "nog": "салам дуныя", "smj": "buoris",
     }
     from sonata_amd.text.phonemizer import _BATCH3
     assert set(samples) == set(_BATCH3) | {"ja"}
@@ -306,6 +339,12 @@ def test_batch3_symbols_encodable():
         "lfn": "me parla elefen", "jbo": "mi tavla fo la lojban",
         "lb": "ech schwätzen lëtzebuergesch moien",
         "ga": "tá gaeilge agam go raibh maith agat",
+        "my": "မြန်မာဘာသာ ပြောတယ် မင်္ဂလာပါ",
+        "th": "สวัสดีครับ ผมพูดภาษาไทย",
+        "kok": "नमस्कार संसार", "an": "ola mundo muller",
+        "ku": "silav cîhan ez kurdî", "gd": "halò a shaoghail loch",
+        "quc": "saqarik uleew utz awach", "sd": "سلام دنيا سنڌي",
+        "nog": "салам дуныя аьел", "smj": "buoris sjaddat",
     }
     for lang, txt in samples.items():
         for sent in text_to_phonemes(txt, lang):

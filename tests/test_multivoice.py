@@ -124,6 +124,11 @@ def test_all_languages_end_to_end(tmp_path):
         "ga": "Dia duit a dhomhain.", "grc": "χαῖρε κόσμε.",
         "tt": "Сәлам дөнья.", "ba": "Сәләм донъя.",
         "cv": "Салам тӗнче.",
+        "kok": "नमस्कार संसार.", "my": "မင်္ဂလာပါ ကမ္ဘာ.",
+        "th": "สวัสดีโลก.", "an": "Ola mundo.", "ku": "Silav cîhan.",
+        "gd": "Halò a shaoghail.", "quc": "Saqarik uleew.",
+        "sd": "سلام دنيا.", "nog": "Салам дуныя.",
+        "smj": "Buoris væráldda.",
     }
     for lang in available_languages():
         pack = create_random_voice(str(tmp_path), f"lang_{lang}",
