@@ -201,7 +201,8 @@ def make_engine(lang: str) -> BrahmicG2P:
             0x0, "ə", final_del=True,
             vowel_overrides={"आ": "ɑː", "इ": "ɪ", "उ": "ʊ", "ऋ": "rɪ"},
             matra_overrides={"ा": "ɑː", "ि": "ɪ", "ु": "ʊ", "ृ": "rɪ"})
-    if lang == "bn":
+    if lang in ("bn", "bpy"):
+        # Bishnupriya Manipuri is written in the Bengali script
         return _bn_engine()
     if lang == "as":
         return _bn_engine(assamese=True)
@@ -235,14 +236,14 @@ def make_engine(lang: str) -> BrahmicG2P:
 # word-regex letter ranges per script block
 INDIC_LETTERS = {
     "mr": "ऀ-ॿ", "ne": "ऀ-ॿ", "kok": "ऀ-ॿ",
-    "bn": "ঀ-৿", "as": "ঀ-৿",
+    "bn": "ঀ-৿", "as": "ঀ-৿", "bpy": "ঀ-৿",
     "gu": "઀-૿", "pa": "਀-੿", "or": "଀-୿",
     "ta": "஀-௿", "te": "ఀ-౿", "kn": "ಀ-೿",
     "ml": "ഀ-ൿ", "si": "඀-෿",
 }
 
-INDIC_LANGS = ("mr", "ne", "kok", "bn", "as", "gu", "pa", "or", "ta",
-               "te", "kn", "ml")
+INDIC_LANGS = ("mr", "ne", "kok", "bn", "as", "bpy", "gu", "pa", "or",
+               "ta", "te", "kn", "ml")
 
 
 # --------------------------------------------------------------------- #

@@ -191,10 +191,15 @@ _MY_CONS = {
     "ပ": "p", "ဖ": "pʰ", "ဗ": "b", "ဘ": "b", "မ": "m",
     "ယ": "j", "ရ": "j", "လ": "l", "ဝ": "w", "သ": "θ",
     "ဟ": "h", "ဠ": "l", "အ": "",
+    # Shan consonant extensions (shn shares this engine)
+    "ၵ": "k", "ၶ": "kʰ", "ၷ": "ɡ", "ၸ": "ts", "ၹ": "z",
+    "ၺ": "ɲ", "ၻ": "d", "ၼ": "n", "ၽ": "pʰ", "ၾ": "f",
+    "ႀ": "θ", "ႁ": "h",
 }
 _MY_VOWEL_SIGNS = {
     "ါ": "aː", "ာ": "aː", "ိ": "i", "ီ": "iː", "ု": "u",
     "ူ": "uː", "ေ": "eː", "ဲ": "ɛː",
+    "ႃ": "aː", "ႄ": "ɛː",  # Shan signs
 }
 _MY_MEDIALS = {"ျ": "j", "ြ": "j", "ွ": "w", "ှ": "h"}
 _MY_INDEP = {"ဣ": "i", "ဤ": "iː", "ဥ": "u", "ဦ": "uː", "ဧ": "eː",
@@ -315,8 +320,9 @@ SCRIPT_LETTERS = {
     "ko": "가-힣",
     "am": "ሀ-ፚ",
     "chr": "Ꭰ-Ᏼ",
-    "my": "က-ဿ",
+    "my": "က-ႏ",   # includes the Shan extensions (U+1075-1081)
     "th": "ก-๛",
+    "shn": "က-ႏ",
 }
 SCRIPT_FUNCS = {
     "ko": ko_word_to_ipa,
@@ -324,4 +330,5 @@ SCRIPT_FUNCS = {
     "chr": chr_word_to_ipa,
     "my": my_word_to_ipa,
     "th": th_word_to_ipa,
+    "shn": my_word_to_ipa,  # Shan shares the Myanmar script machinery
 }

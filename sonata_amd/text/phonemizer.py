@@ -711,8 +711,8 @@ _EN_LETTERS = {
 
 _G2P_REGISTRY: Dict[str, RuleG2P] = {}
 
-_INDIC = ("mr", "ne", "kok", "bn", "as", "gu", "pa", "or", "ta", "te",
-          "kn", "ml")
+_INDIC = ("mr", "ne", "kok", "bn", "as", "bpy", "gu", "pa", "or",
+          "ta", "te", "kn", "ml")
 _BATCH3 = ("fa", "ur", "ug", "he", "vi", "mi", "haw", "qu", "gn",
            "nci", "om", "tn", "pap", "ia", "io", "lfn", "jbo", "tk",
            "lb", "kl", "ga", "grc", "tt", "ba", "cv",
@@ -858,7 +858,7 @@ def _get_g2p(voice: str) -> RuleG2P:
             # (lexicon entries are already surface forms)
             orig = g._apply_rules
             g._apply_rules = lambda w, _o=orig, _p=post: _p(_o(w))
-    elif base in ("ko", "am", "chr", "my", "th"):
+    elif base in ("ko", "am", "chr", "my", "th", "shn"):
         # syllabic scripts decoded by codepoint arithmetic
         # (g2p_scripts.py): Hangul triples, Ethiopic rows, Cherokee CV
         from .g2p_scripts import SCRIPT_FUNCS, SCRIPT_LETTERS
@@ -917,7 +917,7 @@ def available_languages() -> List[str]:
             "fa", "ur", "ug", "he", "vi", "mi", "haw", "qu", "gn",
             "nci", "om", "tn", "pap", "ia", "io", "lfn", "jbo", "tk",
             "lb", "kl", "ga", "grc", "tt", "ba", "cv",
-            "an", "ku", "gd", "quc", "sd", "nog", "smj"]
+            "an", "ku", "gd", "quc", "sd", "nog", "smj", "bpy", "shn"]
 
 
 # script-native sentence/clause punctuation -> ASCII so the splitter
