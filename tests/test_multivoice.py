@@ -129,6 +129,7 @@ def test_all_languages_end_to_end(tmp_path):
         "gd": "Halò a shaoghail.", "quc": "Saqarik uleew.",
         "sd": "سلام دنيا.", "nog": "Салам дуныя.",
         "smj": "Buoris væráldda.",
+        "bpy": "আমার ঠার.", "shn": "မႂ်ႇသုင် ၵမ်ႇၽႃႇ.",
     }
     for lang in available_languages():
         pack = create_random_voice(str(tmp_path), f"lang_{lang}",
