@@ -25,7 +25,8 @@ from typing import Dict, Optional
 # --------------------------------------------------------------------- #
 # Epenthesis for consonant-skeleton scripts (fa/ur/he)
 # --------------------------------------------------------------------- #
-_MULTI_PHONES = ("tʃʰ", "dʒʰ", "tʃ", "dʒ", "ts", "dz", "kʰ", "ɡʰ",
+_MULTI_PHONES = ("tʃʰ", "dʒʰ", "tɕʰ", "tɕ", "dʑ", "tʃ", "dʒ", "ts",
+                 "dz", "kʰ", "ɡʰ",
                  "pʰ", "bʰ", "tʰ", "dʰ", "ʈʰ", "ɖʰ", "sˤ", "dˤ", "tˤ",
                  "ðˤ", "aː", "iː", "uː", "eː", "oː", "ɒː", "æː", "ɛː",
                  "ɑː", "ɔː", "ɯː", "øː", "yː", "ʊː", "ɪː")

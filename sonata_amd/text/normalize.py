@@ -144,6 +144,84 @@ _DIGITS.update({
     "ar": "صفر واحد اثنان ثلاثة أربعة خمسة ستة سبعة ثمانية تسعة".split(),
 })
 
+# third-batch languages (digit-by-digit reading, own script; note
+# Python's \d and int() accept native digits — ०३ / ٣ / ๕ — so these
+# fire for script-native numerals too)
+_DIGITS.update({
+    "mr": "शून्य एक दोन तीन चार पाच सहा सात आठ नऊ".split(),
+    "ne": "शून्य एक दुई तीन चार पाँच छ सात आठ नौ".split(),
+    "kok": "शून्य एक दोन तीन चार पांच सहा सात आठ नऊ".split(),
+    "bn": "শূন্য এক দুই তিন চার পাঁচ ছয় সাত আট নয়".split(),
+    "bpy": "শূন্য এক দুই তিন চার পাঁচ ছয় সাত আট নয়".split(),
+    "as": "শূন্য এক দুই তিনি চাৰি পাঁচ ছয় সাত আঠ ন".split(),
+    "gu": "શૂન્ય એક બે ત્રણ ચાર પાંચ છ સાત આઠ નવ".split(),
+    "pa": "ਸਿਫਰ ਇੱਕ ਦੋ ਤਿੰਨ ਚਾਰ ਪੰਜ ਛੇ ਸੱਤ ਅੱਠ ਨੌਂ".split(),
+    "or": "ଶୂନ ଏକ ଦୁଇ ତିନି ଚାରି ପାଞ୍ଚ ଛଅ ସାତ ଆଠ ନଅ".split(),
+    "ta": ("பூஜ்ஜியம் ஒன்று இரண்டு மூன்று நான்கு ஐந்து ஆறு ஏழு "
+           "எட்டு ஒன்பது").split(),
+    "te": ("సున్నా ఒకటి రెండు మూడు నాలుగు ఐదు ఆరు ఏడు ఎనిమిది "
+           "తొమ్మిది").split(),
+    "kn": ("ಸೊನ್ನೆ ಒಂದು ಎರಡು ಮೂರು ನಾಲ್ಕು ಐದು ಆರು ಏಳು ಎಂಟು "
+           "ಒಂಬತ್ತು").split(),
+    "ml": ("പൂജ്യം ഒന്ന് രണ്ട് മൂന്ന് നാല് അഞ്ച് ആറ് ഏഴ് എട്ട് "
+           "ഒമ്പത്").split(),
+    "si": "බින්දුව එක දෙක තුන හතර පහ හය හත අට නවය".split(),
+    "ko": "영 일 이 삼 사 오 육 칠 팔 구".split(),
+    "ja": "ゼロ いち に さん よん ご ろく なな はち きゅう".split(),
+    "am": "ዜሮ አንድ ሁለት ሶስት አራት አምስት ስድስት ሰባት ስምንት ዘጠኝ".split(),
+    "chr": "ᏏᎶ ᏌᏊ ᏔᎵ ᏦᎢ ᏅᎩ ᎯᏍᎩ ᏑᏓᎵ ᎦᎵᏉᎩ ᏧᏁᎳ ᏐᏁᎳ".split(),
+    "my": "သုည တစ် နှစ် သုံး လေး ငါး ခြောက် ခုနစ် ရှစ် ကိုး".split(),
+    "shn": "သုၼ် ၼိုင်ႈ သွင် သၢမ် သီႇ ႁႃႈ ႁူၵ်း ၸဵတ်း ပႅတ်ႈ ၵဝ်ႈ".split(),
+    "th": "ศูนย์ หนึ่ง สอง สาม สี่ ห้า หก เจ็ด แปด เก้า".split(),
+    "fa": "صفر یک دو سه چهار پنج شش هفت هشت نه".split(),
+    "ur": "صفر ایک دو تین چار پانچ چھ سات آٹھ نو".split(),
+    "sd": "ٻڙي هڪ ٻه ٽي چار پنج ڇهه ست اٺ نو".split(),
+    "ug": ("نۆل بىر ئىككى ئۈچ تۆت بەش ئالتە يەتتە سەككىز "
+           "توققۇز").split(),
+    "he": "אפס אחת שתיים שלוש ארבע חמש שש שבע שמונה תשע".split(),
+    "vi": "không một hai ba bốn năm sáu bảy tám chín".split(),
+    "mi": "kore tahi rua toru whā rima ono whitu waru iwa".split(),
+    "haw": ("ʻole ʻekahi ʻelua ʻekolu ʻehā ʻelima ʻeono ʻehiku "
+            "ʻewalu ʻeiwa").split(),
+    "qu": ("ch'usaq huk iskay kinsa tawa pichqa suqta qanchis "
+           "pusaq isqun").split(),
+    "gn": ("mba'eve peteĩ mokõi mbohapy irundy po poteĩ pokõi "
+           "poapy porundy").split(),
+    "nci": ("ahtle ce ome eyi nahui macuilli chicuace chicome "
+            "chicuei chicnahui").split(),
+    "om": "duwwaa tokko lama sadii afur shan jaha torba saddeet sagal".split(),
+    "tn": ("lefela nngwe pedi tharo nne tlhano thataro supa "
+           "robedi robongwe").split(),
+    "pap": "sero un dos tres kuater sinku seis shete ocho nuebe".split(),
+    "ia": "zero un duo tres quatro cinque sex septe octo novem".split(),
+    "io": "zero un du tri quar kin sis sep ok non".split(),
+    "lfn": "zero un du tre cuatro sinco ses sete oto nove".split(),
+    "jbo": "no pa re ci vo mu xa ze bi so".split(),
+    "tk": "nol bir iki üç dört bäş alty ýedi sekiz dokuz".split(),
+    "lb": "null eent zwee dräi véier fënnef sechs siwen aacht néng".split(),
+    "kl": ("nul ataaseq marluk pingasut sisamat tallimat arfinillit "
+           "arfineq-marluk arfineq-pingasut qulingiluat").split(),
+    "ga": "náid aon dó trí ceathair cúig sé seacht ocht naoi".split(),
+    "gd": "neoni aon dà trì ceithir còig sia seachd ochd naoi".split(),
+    "grc": "οὐδέν εἷς δύο τρεῖς τέσσαρες πέντε ἕξ ἑπτά ὀκτώ ἐννέα".split(),
+    "tt": "нуль бер ике өч дүрт биш алты җиде сигез тугыз".split(),
+    "ba": "нуль бер ике өс дүрт биш алты ете һигеҙ туғыҙ".split(),
+    "cv": ("нуль пӗрре иккӗ виҫҫӗ тӑваттӑ пиллӗк улттӑ ҫиччӗ "
+           "саккӑр тӑххӑр").split(),
+    "nog": "ноль бир эки уьш доьрт бес алты ети сегиз тогыз".split(),
+    "ku": "sifir yek du sê çar pênc şeş heft heşt neh".split(),
+    "an": "zero un dos tres quatre cinco seis siet ueito nueu".split(),
+    "quc": ("maj jun keb oxib kajib job waqib wuqub wajxaqib "
+            "belejeb").split(),
+    "smj": ("nolla akta guokta golmma nielja vihtta guhtta gietjav "
+            "gáktsa aktse").split(),
+})
+
+# orthography aliases share digit tables
+for _alias, _src in (("nb", "no"), ("nn", "no"), ("sr", "hr"),
+                     ("bs", "hr"), ("ms", "id")):
+    _DIGITS[_alias] = _DIGITS[_src]
+
 _NUM_RE = re.compile(r"\d[\d,]*(?:\.\d+)?")
 _ORD_RE = re.compile(r"\b(\d+)(st|nd|rd|th)\b", re.IGNORECASE)
 _ABBR_RE = re.compile(r"\b(Mr|Mrs|Ms|Dr|St|Jr|Sr|Prof|etc|vs)\.",
