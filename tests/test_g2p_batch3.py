@@ -354,6 +354,27 @@ def test_batch3_symbols_encodable():
                 assert ch in id_map, (lang, ch, hex(ord(ch)), sent)
 
 
+def test_digit_names_all_languages():
+    """Every covered language reads digits in its own words — ASCII
+    and script-native numerals (३ ٣ ๕ ৫) both expand."""
+    from sonata_amd.text.normalize import _DIGITS
+    from sonata_amd.text.phonemizer import available_languages
+    for lang in available_languages():
+        base = lang.split("-")[0]
+        assert base in _DIGITS, lang
+        assert len(_DIGITS[base]) == 10, lang
+    assert text_to_phonemes("मेरे पास ३ किताबें", "hi")[0].count("tiːn")
+    assert "haː" in text_to_phonemes("๕", "th")[0]
+    assert "sam" in text_to_phonemes("3", "ko")[0]
+    # digit words must encode through the id map
+    from sonata_amd.text.ids import default_phoneme_id_map
+    idm = default_phoneme_id_map()
+    for lang in available_languages():
+        for s in text_to_phonemes("3 7 9", lang):
+            for ch in s.replace(" ", ""):
+                assert ch in idm or ch in ".,;:?!", (lang, ch)
+
+
 def test_indic_symbols_encodable():
     """Every IPA char the Indic engines emit must be in the voice
     symbol table (ids.py) so it survives phonemes->ids encoding."""
