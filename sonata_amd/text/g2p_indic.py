@@ -194,13 +194,55 @@ def _bn_engine(assamese: bool = False) -> BrahmicG2P:
                       anusvara_ipa="ŋ")
 
 
+_HI_VOWEL_IPA = set("əɑaeiouɛɔɪʊ")
+_HI_MULTI = ("tʃʰ", "dʒʰ", "tʃ", "dʒ", "kʰ", "ɡʰ", "ʈʰ", "ɖʰ", "tʰ",
+             "dʰ", "pʰ", "bʰ", "ɾʰ", "ɑː", "aː", "iː", "uː", "eː",
+             "oː", "ɛː", "ɔː", "rɪ")
+
+
+def _hi_tokens(ipa: str):
+    toks = []
+    i, n = 0, len(ipa)
+    while i < n:
+        for m in _HI_MULTI:
+            if ipa.startswith(m, i):
+                toks.append(m)
+                i += len(m)
+                break
+        else:
+            toks.append(ipa[i])
+            i += 1
+    return toks
+
+
+def hi_schwa_deletion(ipa: str) -> str:
+    """Medial schwa deletion (Ohala's rule): delete ə in a V C ə C V
+    context, applying right-to-left (नमकीन nəməkiːn -> nəmkiːn, but
+    नमस्कार nəməskaːr keeps its schwa — s is followed by a consonant)."""
+
+    def is_v(t: str) -> bool:
+        return t[0] in _HI_VOWEL_IPA
+
+    toks = _hi_tokens(ipa)
+    i = len(toks) - 1
+    while i >= 2:
+        if (toks[i] == "ə"
+                and not is_v(toks[i - 1]) and is_v(toks[i - 2])
+                and i + 2 < len(toks)
+                and not is_v(toks[i + 1]) and is_v(toks[i + 2])):
+            del toks[i]
+        i -= 1
+    return "".join(toks)
+
+
 def make_engine(lang: str) -> BrahmicG2P:
     if lang in ("hi", "mr", "ne", "kok"):
         # Devanagari: Hindi vowel quality (ɪ/ʊ lax short vowels, ɑː)
         return BrahmicG2P(
             0x0, "ə", final_del=True,
             vowel_overrides={"आ": "ɑː", "इ": "ɪ", "उ": "ʊ", "ऋ": "rɪ"},
-            matra_overrides={"ा": "ɑː", "ि": "ɪ", "ु": "ʊ", "ृ": "rɪ"})
+            matra_overrides={"ा": "ɑː", "ि": "ɪ", "ु": "ʊ", "ृ": "rɪ"},
+            postprocess=hi_schwa_deletion if lang == "hi" else None)
     if lang in ("bn", "bpy"):
         # Bishnupriya Manipuri is written in the Bengali script
         return _bn_engine()

@@ -400,7 +400,11 @@ def test_hindi_schwa_rules():
     assert g.word_to_ipa("कमल") == "kəməl"      # ka-ma-l(a): final deleted
     assert g.word_to_ipa("हिंदी") == "hɪndiː"   # matra + anusvara
     assert g.word_to_ipa("नमस्ते") == "nəməsteː"  # virama joins s-t
-    assert g.word_to_ipa("बोलता") == "boːlətɑː"
+    # medial schwa deletion (Ohala's VC_CV rule, right-to-left)
+    assert g.word_to_ipa("बोलता") == "boːltɑː"
+    assert g.word_to_ipa("नमकीन") == "nəmkiːn"
+    assert g.word_to_ipa("नमस्कार") == "nəməskɑːr"  # s+k blocks it
+    assert g.word_to_ipa("जानवर") == "dʒɑːnʋər"
 
 
 def test_batch2_symbol_coverage():
