@@ -237,11 +237,17 @@ def hi_schwa_deletion(ipa: str) -> str:
 
 def make_engine(lang: str) -> BrahmicG2P:
     if lang in ("hi", "mr", "ne", "kok"):
-        # Devanagari: Hindi vowel quality (ɪ/ʊ lax short vowels, ɑː)
+        # Devanagari: Hindi vowel quality (ɪ/ʊ lax short vowels, ɑː).
+        # In Modern Standard Hindi ऐ/औ are monophthongs ɛː/ɔː; Marathi
+        # and Nepali keep the əi/əu-style diphthongs.
+        vow = {"आ": "ɑː", "इ": "ɪ", "उ": "ʊ", "ऋ": "rɪ"}
+        mat = {"ा": "ɑː", "ि": "ɪ", "ु": "ʊ", "ृ": "rɪ"}
+        if lang == "hi":
+            vow.update({"ऐ": "ɛː", "औ": "ɔː"})
+            mat.update({"ै": "ɛː", "ौ": "ɔː"})
         return BrahmicG2P(
             0x0, "ə", final_del=True,
-            vowel_overrides={"आ": "ɑː", "इ": "ɪ", "उ": "ʊ", "ऋ": "rɪ"},
-            matra_overrides={"ा": "ɑː", "ि": "ɪ", "ु": "ʊ", "ृ": "rɪ"},
+            vowel_overrides=vow, matra_overrides=mat,
             postprocess=hi_schwa_deletion if lang == "hi" else None)
     if lang in ("bn", "bpy"):
         # Bishnupriya Manipuri is written in the Bengali script
