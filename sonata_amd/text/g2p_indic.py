@@ -319,6 +319,38 @@ _SI_MATRAS = {
 }
 
 
+# frequent-word exception lexicons (surface forms the rules cannot
+# derive: nasalized function words, loans, schwa-deletion exceptions)
+INDIC_LEXICONS = {
+    "hi": {
+        # nasalized function words (candrabindu/anusvara = vowel
+        # nasality, not a stop+nasal cluster)
+        "हैं": "hɛ̃ː", "मैं": "mɛ̃ː", "में": "mẽː", "नहीं": "nəhˈĩː",
+        "हूँ": "hũː", "हूं": "hũː", "कहाँ": "kəhɑ̃ː", "यहाँ": "jəhɑ̃ː",
+        "वहाँ": "ʋəhɑ̃ː", "जहाँ": "dʒəhɑ̃ː", "हाँ": "hɑ̃ː",
+        "आँख": "ɑ̃ːkʰ", "पाँच": "pɑ̃ːtʃ", "गाँव": "ɡɑ̃ːʋ",
+        "कुछ": "kʊtʃʰ", "बहुत": "bəhʊt", "थीं": "tʰĩː",
+        # ये/वो colloquial forms and irregular pronouns
+        "ये": "jeː", "वो": "ʋoː", "यह": "jeh", "वह": "ʋoh",
+        # loans where deletion/epenthesis rules misfire
+        "स्कूल": "skuːl", "स्टेशन": "sʈeːʃən", "डॉक्टर": "ɖɔkʈər",
+    },
+    "bn": {
+        # Bengali: irregular high-frequency forms
+        "আমি": "aːmi", "তুমি": "tumi", "সে": "ʃeː", "এই": "ei",
+        "ওই": "oi", "কি": "ki", "না": "naː", "হ্যাঁ": "hɛ̃",
+        "আছে": "aːtʃʰe", "এবং": "eboŋ", "কিন্তু": "kintu",
+        "করে": "kɔre", "হয়": "hɔe",
+    },
+    "ta": {
+        # Tamil: common words where voicing/cluster rules misfire
+        "நான்": "naːn", "நீ": "niː", "அவன்": "avan", "அவள்": "avaɭ",
+        "இது": "idu", "அது": "adu", "என்ன": "enna", "இல்லை": "illai",
+        "ஆமாம்": "aːmaːm", "வணக்கம்": "vaɳakkam",
+    },
+}
+
+
 def make_si_engine() -> BrahmicG2P:
     g = BrahmicG2P.__new__(BrahmicG2P)
     g.cons = dict(_SI_CONS)

@@ -683,7 +683,15 @@ STRESS3 = {
     "sd": "first", "nog": "final", "smj": "first",
 }
 
-LEXICONS3 = {"fa": FA_LEXICON, "ur": UR_LEXICON, "he": HE_LEXICON}
+VI_LEXICON = {
+    # function words with finals the digraph rules miss
+    "của": "kuə", "và": "vaː", "là": "laː", "không": "xoŋ",
+    "người": "ŋɨəi", "được": "dɨək", "những": "ɲɨŋ",
+    "anh": "aɲ", "em": "ɛm", "tôi": "toi",
+}
+
+LEXICONS3 = {"fa": FA_LEXICON, "ur": UR_LEXICON, "he": HE_LEXICON,
+             "vi": VI_LEXICON}
 PREPROCESS3 = {"vi": vi_preprocess, "he": he_preprocess,
                "grc": grc_preprocess}
 POSTPROCESS3 = {"fa": fa_postprocess, "ur": ur_postprocess,

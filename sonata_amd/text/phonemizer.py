@@ -872,9 +872,12 @@ def _get_g2p(voice: str) -> RuleG2P:
         from .g2p_indic import (INDIC_LETTERS, make_engine,
                                 make_si_engine)
 
+        from .g2p_indic import INDIC_LEXICONS
+
         eng = make_si_engine() if base == "si" else make_engine(base)
         letters = "ऀ-ॿ" if base == "hi" else INDIC_LETTERS[base]
-        g = RuleG2P({}, letters=letters, stress=False)
+        g = RuleG2P({}, INDIC_LEXICONS.get(base), letters=letters,
+                    stress=False)
         g._apply_rules = eng.word_to_ipa
     else:
         # expansion tables (g2p_tables.py): 13 + 23 more languages

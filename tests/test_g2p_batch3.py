@@ -354,6 +354,28 @@ def test_batch3_symbols_encodable():
                 assert ch in id_map, (lang, ch, hex(ord(ch)), sent)
 
 
+def test_indic_lexicons():
+    """Exception lexicons: nasalized Hindi function words (candrabindu
+    = vowel nasality, not a stop), Bengali/Tamil irregulars."""
+    hi = _get_g2p("hi")
+    assert hi.word_to_ipa("मैं") == "mɛ̃ː"
+    assert hi.word_to_ipa("नहीं") == "nəhˈĩː"
+    assert hi.word_to_ipa("में") == "mẽː"
+    assert hi.word_to_ipa("स्कूल") == "skuːl"
+    assert _get_g2p("bn").word_to_ipa("কিন্তু") == "kintu"
+    assert _get_g2p("ta").word_to_ipa("வணக்கம்") == "vaɳakkam"
+    assert _get_g2p("vi").word_to_ipa("người") == "ŋɨəi"
+    # lexicon outputs must encode through the id map
+    from sonata_amd.text.ids import default_phoneme_id_map
+    from sonata_amd.text.g2p_indic import INDIC_LEXICONS
+    from sonata_amd.text.g2p_tables3 import LEXICONS3
+    idm = default_phoneme_id_map()
+    for lex in list(INDIC_LEXICONS.values()) + list(LEXICONS3.values()):
+        for ipa in lex.values():
+            for ch in ipa:
+                assert ch in idm, (ipa, ch, hex(ord(ch)))
+
+
 def test_digit_names_all_languages():
     """Every covered language reads digits in its own words — ASCII
     and script-native numerals (३ ٣ ๕ ৫) both expand."""
