@@ -734,6 +734,26 @@ def _get_g2p(voice: str) -> RuleG2P:
         g = RuleG2P(_EN_RULES, merged, letters="a-zA-Z",
                     unstressed=_EN_UNSTRESSED)
         g.spell_acronyms = True
+
+        def _en_fix(w: str, ipa: str) -> str:
+            # orthography-aware ending fixes the context-free rules
+            # cannot express (rule path only; lexicon is surface truth)
+            if w.endswith("ous") and ipa.endswith("aʊs"):
+                ipa = ipa[:-3] + "əs"        # famous, previous
+            if len(w) > 4 and w.endswith("ble") and ipa.endswith("blɛ"):
+                ipa = ipa[:-3] + "bəl"       # possible, comparable
+            elif (len(w) > 4 and w.endswith("le") and ipa.endswith("lɛ")
+                    and w[-3] not in "aeiou"):
+                ipa = ipa[:-2] + "əl"        # circle, obstacle
+            if len(w) > 4 and w.endswith("age") and ipa.endswith("ædʒ"):
+                ipa = ipa[:-3] + "ɪdʒ"       # manage, luggage
+            if (len(w) > 3 and w.endswith("e") and w[-2] not in "aeiouy"
+                    and ipa.endswith("ɛ")):
+                ipa = ipa[:-1]               # silent final e (blouse)
+            return ipa
+
+        _en_rules_orig = g._apply_rules
+        g._apply_rules = lambda w, _o=_en_rules_orig: _en_fix(w, _o(w))
     elif base == "de":
         from .g2p_tables import DE_LEXICON, de_postprocess, de_preprocess
 

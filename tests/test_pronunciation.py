@@ -119,6 +119,12 @@ EN_INFLECTION_CASES = [
 
 # ---- English tier C: rule path + suffix stress ------------------------ #
 EN_RULE_CASES = [
+    # ending-fix classes (-ous, -Cle, -age, silent e) through rules
+    ("famous", "fˈeɪməs"), ("nervous", "nˈɝvəs"),
+    ("various", "vˈɛɹiəs"), ("previous", "pɹˈiviəs"),
+    ("responsible", "ɹɛspˈɑnsɪbəl"), ("obstacle", "ˈɑbstækəl"),
+    ("miracle", "mˈɪɹəkəl"), ("manage", "mˈænɪdʒ"),
+    ("luggage", "lˈʌɡɡɪdʒ"), ("blouse", "blˈaʊs"),
     # regular words deliberately NOT in the lexicon
     ("blasting", "blˈæstɪŋ"), ("grandstand", "ɡɹˈændstænd"),
     ("fantastic", "fæntˈæstɪk"), ("septic", "sˈɛptɪk"),
