@@ -255,7 +255,126 @@ _KANA_TABLE = _build_kana()
 _KANA_KEYS = sorted(_KANA_TABLE, key=len, reverse=True)
 
 
+# Common-word kanji readings (word -> kana), longest-match.  This is a
+# deliberately small high-frequency dictionary — ~250 entries — not a
+# morphological analyzer: compounds not listed here are still dropped
+# (espeak-ng's ja_dict is the reference bar; full kanji coverage needs
+# a real reading dictionary, stated in PARITY.md).
+JA_KANJI = {
+    # numbers
+    "一": "いち", "二": "に", "三": "さん", "四": "よん", "五": "ご",
+    "六": "ろく", "七": "なな", "八": "はち", "九": "きゅう",
+    "十": "じゅう", "百": "ひゃく", "千": "せん", "万": "まん",
+    "円": "えん", "年": "ねん", "月": "つき", "日": "ひ",
+    # time
+    "今日": "きょう", "明日": "あした", "昨日": "きのう",
+    "今": "いま", "時間": "じかん", "時": "とき", "分": "ふん",
+    "今年": "ことし", "去年": "きょねん", "来年": "らいねん",
+    "毎日": "まいにち", "朝": "あさ", "昼": "ひる", "夜": "よる",
+    "午前": "ごぜん", "午後": "ごご", "週": "しゅう", "春": "はる",
+    "夏": "なつ", "秋": "あき", "冬": "ふゆ",
+    # people / pronouns
+    "私": "わたし", "僕": "ぼく", "人": "ひと", "人々": "ひとびと",
+    "友達": "ともだち", "先生": "せんせい", "学生": "がくせい",
+    "子供": "こども", "男": "おとこ", "女": "おんな",
+    "家族": "かぞく", "父": "ちち", "母": "はは", "名前": "なまえ",
+    "皆": "みんな", "彼": "かれ", "彼女": "かのじょ",
+    # places
+    "日本": "にほん", "日本語": "にほんご", "東京": "とうきょう",
+    "学校": "がっこう", "大学": "だいがく", "会社": "かいしゃ",
+    "家": "いえ", "国": "くに", "店": "みせ", "駅": "えき",
+    "道": "みち", "町": "まち", "市": "し", "世界": "せかい",
+    "部屋": "へや", "場所": "ばしょ", "外": "そと", "中": "なか",
+    "上": "うえ", "下": "した", "前": "まえ", "後": "あと",
+    # nature
+    "水": "みず", "火": "ひ", "山": "やま", "川": "かわ",
+    "海": "うみ", "空": "そら", "雨": "あめ", "雪": "ゆき",
+    "風": "かぜ", "花": "はな", "木": "き", "天気": "てんき",
+    "太陽": "たいよう", "光": "ひかり", "石": "いし",
+    # things
+    "本": "ほん", "車": "くるま", "電車": "でんしゃ",
+    "電話": "でんわ", "手紙": "てがみ", "写真": "しゃしん",
+    "音楽": "おんがく", "映画": "えいが", "料理": "りょうり",
+    "食べ物": "たべもの", "飲み物": "のみもの", "お金": "おかね",
+    "金": "かね", "仕事": "しごと", "言葉": "ことば",
+    "物": "もの", "事": "こと", "話": "はなし", "歌": "うた",
+    "声": "こえ", "音": "おと", "字": "じ", "絵": "え",
+    "机": "つくえ", "椅子": "いす", "窓": "まど", "戸": "と",
+    "犬": "いぬ", "猫": "ねこ", "鳥": "とり", "魚": "さかな",
+    "牛": "うし", "馬": "うま", "卵": "たまご", "肉": "にく",
+    "茶": "ちゃ", "米": "こめ", "酒": "さけ",
+    # body
+    "手": "て", "足": "あし", "目": "め", "耳": "みみ",
+    "口": "くち", "頭": "あたま", "顔": "かお", "心": "こころ",
+    "体": "からだ", "気": "き",
+    # verbs / stems (okurigana follows in kana)
+    "食べ": "たべ", "飲み": "のみ", "飲ん": "のん", "行き": "いき",
+    "行っ": "いっ", "行く": "いく", "来る": "くる", "来て": "きて",
+    "来ま": "きま", "見": "み", "見る": "みる", "聞き": "きき",
+    "聞く": "きく", "話し": "はなし", "話す": "はなす",
+    "読み": "よみ", "読む": "よむ", "書き": "かき", "書く": "かく",
+    "買い": "かい", "買う": "かう", "売り": "うり",
+    "立っ": "たっ", "座っ": "すわっ", "歩き": "あるき",
+    "走り": "はしり", "泳ぎ": "およぎ", "帰り": "かえり",
+    "帰る": "かえる", "出": "で", "入り": "はいり", "入っ": "はいっ",
+    "作り": "つくり", "作る": "つくる", "使い": "つかい",
+    "使う": "つかう", "思い": "おもい", "思う": "おもう",
+    "知り": "しり", "知っ": "しっ", "分かり": "わかり",
+    "分かる": "わかる", "言い": "いい", "言う": "いう",
+    "言っ": "いっ", "会い": "あい", "会う": "あう",
+    "待ち": "まち", "待つ": "まつ", "持ち": "もち", "持っ": "もっ",
+    "住ん": "すん", "死ん": "しん", "生き": "いき",
+    "働き": "はたらき", "休み": "やすみ", "遊び": "あそび",
+    "始め": "はじめ", "終わり": "おわり", "開け": "あけ",
+    "閉め": "しめ", "教え": "おしえ", "覚え": "おぼえ",
+    "忘れ": "わすれ", "寝": "ね", "起き": "おき",
+    # adjective stems
+    "大き": "おおき", "小さ": "ちいさ", "新し": "あたらし",
+    "古": "ふる", "高": "たか", "安": "やす", "長": "なが",
+    "短": "みじか", "早": "はや", "遅": "おそ", "多": "おお",
+    "少な": "すくな", "良": "よ", "悪": "わる", "白": "しろ",
+    "黒": "くろ", "赤": "あか", "青": "あお", "暑": "あつ",
+    "寒": "さむ", "強": "つよ", "弱": "よわ", "難し": "むずかし",
+    "易し": "やさし", "楽し": "たのし", "嬉し": "うれし",
+    "美し": "うつくし", "面白": "おもしろ", "元気": "げんき",
+    "大切": "たいせつ", "大丈夫": "だいじょうぶ",
+    "好き": "すき", "嫌い": "きらい", "静か": "しずか",
+    "有名": "ゆうめい", "便利": "べんり", "簡単": "かんたん",
+    # misc frequent
+    "何": "なに", "誰": "だれ", "一つ": "ひとつ", "二つ": "ふたつ",
+    "三つ": "みっつ", "一人": "ひとり", "二人": "ふたり",
+    "一番": "いちばん", "全部": "ぜんぶ", "少し": "すこし",
+    "本当": "ほんとう", "勉強": "べんきょう", "旅行": "りょこう",
+    "質問": "しつもん", "問題": "もんだい", "答え": "こたえ",
+    "意味": "いみ", "説明": "せつめい", "最初": "さいしょ",
+    "最後": "さいご", "次": "つぎ", "他": "ほか", "別": "べつ",
+}
+_JA_KANJI_MAX = max(len(k) for k in JA_KANJI)
+
+
+def _ja_kanji_to_kana(w: str) -> str:
+    out = []
+    i, n = 0, len(w)
+    while i < n:
+        for ln in range(min(_JA_KANJI_MAX, n - i), 0, -1):
+            seg = w[i:i + ln]
+            if seg in JA_KANJI:
+                out.append(JA_KANJI[seg])
+                i += ln
+                break
+        else:
+            ch = w[i]
+            if ch == "々" and out:
+                out.append(out[-1])  # iteration mark repeats reading
+            elif not (0x4E00 <= ord(ch) <= 0x9FFF):
+                out.append(ch)  # kana and marks pass through
+            # unknown kanji: dropped (needs a reading dictionary)
+            i += 1
+    return "".join(out)
+
+
 def ja_word_to_ipa(w: str) -> str:
+    w = _ja_kanji_to_kana(w)
     out = []
     i, n = 0, len(w)
     while i < n:

@@ -859,7 +859,9 @@ def _get_g2p(voice: str) -> RuleG2P:
         # PARITY.md states so
         from .g2p_tables3 import ja_word_to_ipa
 
-        g = RuleG2P({}, letters="぀-ヿ", stress=False)
+        # kana blocks + CJK unified (common-word readings via JA_KANJI;
+        # unlisted kanji drop) + 々 iteration mark
+        g = RuleG2P({}, letters="぀-ヿ一-鿿々", stress=False)
         g._apply_rules = ja_word_to_ipa
     elif base in _BATCH3:
         from .g2p_tables3 import (LETTERS3, LEXICONS3, POSTPROCESS3,

@@ -214,6 +214,17 @@ def test_japanese_kana():
     assert ja.word_to_ipa("ラーメン") == "ɾaːmen"     # chōonpu
 
 
+def test_japanese_kanji_readings():
+    ja = _get_g2p("ja")
+    # common-word dictionary, longest match (日本語 before 日本/日)
+    assert ja.word_to_ipa("日本語") == "nihonɡo"
+    assert ja.word_to_ipa("東京") == "toːkjoː"
+    assert ja.word_to_ipa("人々") == "çitobito"
+    assert ja.word_to_ipa("勉強します") == "benkjoːʃimasu"
+    # unknown kanji drop; kana around them survive
+    assert ja.word_to_ipa("鸞です") == "desu"
+
+
 def test_vietnamese_tones_stripped():
     vi = _get_g2p("vi")
     # same segmental output regardless of tone
