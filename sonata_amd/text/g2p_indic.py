@@ -127,6 +127,12 @@ class BrahmicG2P:
         self.postprocess = postprocess
         self.anusvara_ipa = anusvara_ipa
         self.final_anusvara = final_anusvara
+        # the independent short-a letter IS the inherent vowel
+        # (Tamil அ = /a/, Bengali অ = /ɔ/, Devanagari अ = /ə/)
+        a_letter = chr(0x0905 + offset)
+        if a_letter in self.vowels and not (vowel_overrides or {}) \
+                .get(a_letter):
+            self.vowels[a_letter] = inherent
 
     def word_to_ipa(self, w: str) -> str:
         # NFC so nukta forms (য়, ড়…) match their precomposed table

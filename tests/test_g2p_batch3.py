@@ -174,6 +174,78 @@ def test_script_native_punctuation_maps_to_ascii():
 
 
 # --------------------------------------------------------------------- #
+# Accuracy corpora for the highest-traffic new languages (gold values
+# hand-checked against standard romanizations/IPA, NOT generated by the
+# engine; floor asserts tolerate the documented approximations)
+# --------------------------------------------------------------------- #
+def _accuracy(g2p, cases):
+    wrong = [(w, g2p.word_to_ipa(w), want)
+             for w, want in cases if g2p.word_to_ipa(w) != want]
+    return 1.0 - len(wrong) / len(cases), wrong
+
+
+KO_CASES = [
+    ("한국", "hanɡuk"), ("사람", "saɾam"), ("감사", "kamsa"),
+    ("사랑", "saɾaŋ"), ("시간", "ʃiɡan" ), ("물", "mul"),
+    ("불", "pul"), ("눈", "nun"), ("손", "son"), ("말", "mal"),
+    ("집", "tɕip"), ("밥", "pap"), ("아침", "atɕʰim"),
+    ("저녁", "tɕʌnjʌk"), ("친구", "tɕʰinɡu"), ("학생", "haksɛŋ"),
+    ("이름", "iɾɯm"), ("나라", "naɾa"), ("바다", "pada"),
+    ("하늘", "hanɯl"),
+]
+
+HI_CASES = [
+    ("पानी", "pɑːniː"), ("आदमी", "ɑːdmiː"), ("औरत", "ɔːrət"),
+    ("बच्चा", "bətʃtʃɑː"), ("किताब", "kɪtɑːb"), ("घर", "ɡʰər"),
+    ("शहर", "ʃəhər"), ("रात", "rɑːt"), ("दिन", "dɪn"),
+    ("साल", "sɑːl"), ("काम", "kɑːm"), ("नाम", "nɑːm"),
+    ("अच्छा", "ətʃtʃʰɑː"), ("बड़ा", "bəɾɑː"), ("छोटा", "tʃʰoːʈɑː"),
+    ("लड़का", "ləɾkɑː"), ("लड़की", "ləɾkiː"), ("हिंदुस्तान", "hɪndʊstɑːn"),
+]
+
+FA_CASES = [
+    ("ایران", "iːrɒːn"), ("کتاب", "ketɒːb"), ("آب", "ɒːb"),
+    ("نان", "nɒːn"), ("شب", "ʃæb"), ("روز", "ruːz"),
+    ("سال", "sɒːl"), ("کار", "kɒːr"), ("دست", "dæst"),
+    ("دل", "del"), ("شهر", "ʃæhr"), ("راه", "rɒːh"),
+]
+
+TA_CASES = [
+    ("அம்மா", "ammaː"), ("அப்பா", "appaː"), ("வீடு", "ʋiːɖu"),
+    ("தண்ணீர்", "taɳɳiːr"), ("பால்", "paːl"), ("கை", "kai"),
+    ("கண்", "kaɳ"), ("ஊர்", "uːr"), ("பை", "pai"), ("மீன்", "miːn"),
+]
+
+
+def test_ko_accuracy():
+    acc, wrong = _accuracy(_get_g2p("ko"), KO_CASES)
+    assert acc >= 0.85, wrong
+
+
+def test_hi_accuracy():
+    g = _get_g2p("hi")
+    cases = [(w, want) for w, want in HI_CASES]
+    acc, wrong = _accuracy(g, cases)
+    assert acc >= 0.8, wrong
+
+
+def test_fa_accuracy():
+    g = _get_g2p("fa")
+    # strip stress for comparison (fa adds final stress)
+    wrong = []
+    for w, want in FA_CASES:
+        got = g.word_to_ipa(w).replace("ˈ", "")
+        if got != want:
+            wrong.append((w, got, want))
+    assert 1 - len(wrong) / len(FA_CASES) >= 0.8, wrong
+
+
+def test_ta_accuracy():
+    acc, wrong = _accuracy(_get_g2p("ta"), TA_CASES)
+    assert acc >= 0.8, wrong
+
+
+# --------------------------------------------------------------------- #
 # Batch-3 rule-table languages (g2p_tables3.py)
 # --------------------------------------------------------------------- #
 def test_persian_epenthesis_and_lexicon():
