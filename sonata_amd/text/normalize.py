@@ -507,6 +507,34 @@ _GROUPED_DOT_RE = re.compile(r"\b\d{1,3}(?:\.\d{3})+\b")
 _DEC_COMMA_RE = re.compile(r"\b(\d+),(\d+)\b")
 _INT_RE = re.compile(r"\d+")
 
+# the word for "%" per language (own script so the G2P letters match);
+# languages not listed keep the current behavior ("%" dropped)
+_PERCENT_WORDS = {
+    "de": "Prozent", "es": "por ciento", "fr": "pour cent",
+    "it": "per cento", "pt": "por cento", "nl": "procent",
+    "pl": "procent", "cs": "procent", "sk": "percent",
+    "sv": "procent", "no": "prosent", "da": "procent",
+    "fi": "prosenttia", "hu": "százalék", "ro": "la sută",
+    "el": "τοις εκατό", "bg": "процента", "ru": "процентов",
+    "uk": "відсотків", "be": "працэнтаў", "tr": "yüzde",
+    "az": "faiz", "kk": "пайыз", "ky": "пайыз", "uz": "foiz",
+    "id": "persen", "sw": "asilimia", "hr": "posto", "sl": "odstotkov",
+    "mk": "проценти", "sq": "për qind", "et": "protsenti",
+    "lv": "procenti", "lt": "procentų", "is": "prósent",
+    "af": "persent", "ca": "per cent", "gl": "por cento",
+    "eu": "ehuneko", "eo": "procento", "cy": "y cant",
+    "hi": "प्रतिशत", "mr": "टक्के", "ne": "प्रतिशत",
+    "bn": "শতাংশ", "gu": "ટકા", "pa": "ਫੀਸਦੀ", "ta": "சதவீதம்",
+    "te": "శాతం", "kn": "ಶೇಕಡಾ", "ml": "ശതമാനം",
+    "ko": "퍼센트", "ja": "パーセント", "vi": "phần trăm",
+    "th": "เปอร์เซ็นต์", "fa": "درصد", "ur": "فیصد", "ar": "بالمئة",
+    "he": "אחוז", "am": "ፐርሰንት", "ka": "პროცენტი",
+    "hy": "տոկոս", "tt": "процент", "ba": "процент",
+    "ku": "ji sedî", "tk": "göterim", "lb": "Prozent",
+    "ga": "faoin gcéad", "gd": "sa cheud", "mt": "fil-mija",
+    "ht": "pousan", "la": "centesimae",
+}
+
 
 def normalize(text: str, language: str) -> str:
     """Expand digits/abbreviations for `language` (base code).
@@ -518,6 +546,9 @@ def normalize(text: str, language: str) -> str:
     base = language.lower().replace("_", "-").split("-")[0]
     if base == "en":
         return normalize_en(text)
+    pw = _PERCENT_WORDS.get(base)
+    if pw is not None:
+        text = _PERCENT_RE.sub(lambda m: m.group(1) + " " + pw, text)
     card = _CARDINALS.get(base)
     if card is not None:
         digits = _DIGITS[base]
