@@ -502,6 +502,17 @@ _CARDINALS = {
 _DECIMAL_WORD = {"de": "Komma", "es": "coma", "fr": "virgule",
                  "it": "virgola", "pt": "vírgula"}
 
+# second grammar batch (numbers2.py): ru/pl/tr/id/nl/sv + dot-decimal
+# ko/ja
+from .numbers2 import (CARDINALS2, DECIMAL_WORDS2,  # noqa: E402
+                       DOT_DECIMAL2)
+
+_CARDINALS.update(CARDINALS2)
+_DECIMAL_WORD.update(DECIMAL_WORDS2)
+_DOT_DECIMAL = dict(DOT_DECIMAL2)
+_GROUP_COMMA_RE = re.compile(r"(?<=\d),(?=\d\d\d)")
+_DEC_DOT_RE = re.compile(r"(\d+)\.(\d+)")
+
 
 _GROUPED_DOT_RE = re.compile(r"\b\d{1,3}(?:\.\d{3})+\b")
 _DEC_COMMA_RE = re.compile(r"\b(\d+),(\d+)\b")
@@ -552,14 +563,21 @@ def normalize(text: str, language: str) -> str:
     card = _CARDINALS.get(base)
     if card is not None:
         digits = _DIGITS[base]
-        # 1.234.567 grouping dots -> plain integer
-        text = _GROUPED_DOT_RE.sub(
-            lambda m: m.group(0).replace(".", ""), text)
-        # decimal comma: 12,5 -> "douze virgule cinq"
-        dec = _DECIMAL_WORD[base]
-        text = _DEC_COMMA_RE.sub(
-            lambda m: card(int(m.group(1))) + " " + dec + " "
-            + " ".join(digits[int(d)] for d in m.group(2)), text)
+        if base in _DOT_DECIMAL:
+            # en-style locale (ko/ja): comma grouping, dot decimals
+            text = _GROUP_COMMA_RE.sub("", text)
+            dec = _DOT_DECIMAL[base]
+            text = _DEC_DOT_RE.sub(
+                lambda m: card(int(m.group(1))) + " " + dec + " "
+                + " ".join(digits[int(d)] for d in m.group(2)), text)
+        else:
+            # European locale: dot grouping, comma decimals
+            text = _GROUPED_DOT_RE.sub(
+                lambda m: m.group(0).replace(".", ""), text)
+            dec = _DECIMAL_WORD[base]
+            text = _DEC_COMMA_RE.sub(
+                lambda m: card(int(m.group(1))) + " " + dec + " "
+                + " ".join(digits[int(d)] for d in m.group(2)), text)
         return _INT_RE.sub(
             lambda m: card(int(m.group(0))) if len(m.group(0)) <= 12
             else " ".join(digits[int(d)] for d in m.group(0)), text)

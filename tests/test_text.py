@@ -288,3 +288,35 @@ def test_cardinal_grammars():
     # decimal comma + grouping dots
     assert normalize("12,5", "fr") == "douze virgule cinq"
     assert normalize("1.000.000", "de") == "eine Million"
+
+
+def test_cardinal_grammars_batch2():
+    """Second cardinal-grammar batch: ru/pl (case-suffix plurals),
+    tr/id/nl/sv (compounding), ko/ja (sino-xenic with rendaku)."""
+    from sonata_amd.text.normalize import normalize
+    from sonata_amd.text.numbers2 import (num_to_words_ja,
+                                          num_to_words_ko,
+                                          num_to_words_nl,
+                                          num_to_words_pl,
+                                          num_to_words_ru,
+                                          num_to_words_tr)
+
+    assert num_to_words_ru(2000) == "две тысячи"       # feminine two
+    assert num_to_words_ru(5000) == "пять тысяч"       # genitive plural
+    assert num_to_words_ru(21) == "двадцать один"
+    assert num_to_words_pl(2000) == "dwa tysiące"
+    assert num_to_words_pl(5000) == "pięć tysięcy"
+    assert num_to_words_tr(2500) == "iki bin beş yüz"  # no "bir" on bin
+    assert num_to_words_nl(22) == "tweeëntwintig"      # diaeresis join
+    assert num_to_words_ko(123456) == "십이만 삼천사백오십육"
+    assert num_to_words_ja(300) == "さんびゃく"          # rendaku
+    assert num_to_words_ja(10000) == "いちまん"
+    # locale decimal styles: European comma vs ko/ja dot
+    assert normalize("3,5", "ru") == "три запятая пять"
+    assert normalize("3.5", "ko") == "삼 점 오"
+    assert normalize("1,000", "ja") == "せん"
+    # end-to-end through the phonemizer
+    from sonata_amd.text.phonemizer import text_to_phonemes
+    assert "tɯl" not in text_to_phonemes("25", "ko")[0]
+    out = text_to_phonemes("Это 21 год.", "ru")[0]
+    assert "adʲˈin" in out or "ɐdʲˈin" in out
