@@ -509,6 +509,11 @@ from .numbers2 import (CARDINALS2, DECIMAL_WORDS2,  # noqa: E402
 
 _CARDINALS.update(CARDINALS2)
 _DECIMAL_WORD.update(DECIMAL_WORDS2)
+
+from .numbers3 import CARDINALS3, DECIMAL_WORDS3  # noqa: E402
+
+_CARDINALS.update(CARDINALS3)
+_DECIMAL_WORD.update(DECIMAL_WORDS3)
 _DOT_DECIMAL = dict(DOT_DECIMAL2)
 _GROUP_COMMA_RE = re.compile(r"(?<=\d),(?=\d\d\d)")
 _DEC_DOT_RE = re.compile(r"(\d+)\.(\d+)")

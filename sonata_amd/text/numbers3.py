@@ -1,0 +1,395 @@
+"""Cardinal number grammars, batch 3: uk (Slavic plurals), no/da
+(Scandinavian, incl. Danish vigesimal tens), fi/hu (agglutinative
+compounds), el (Greek hundreds), cs (Slavic), ro (și-composition), and
+a simplified-MSA ar (masculine nominative forms, gender agreement
+approximated — espeak's ar number reading is the parity target).
+"""
+
+from __future__ import annotations
+
+from .numbers2 import _ru_plural
+
+# --------------------------------------------------------------------- #
+# Ukrainian
+# --------------------------------------------------------------------- #
+_UK_ONES = ("нуль один два три чотири п'ять шість сім вісім дев'ять "
+            "десять одинадцять дванадцять тринадцять чотирнадцять "
+            "п'ятнадцять шістнадцять сімнадцять вісімнадцять "
+            "дев'ятнадцять").split()
+_UK_TENS = ["", "", "двадцять", "тридцять", "сорок", "п'ятдесят",
+            "шістдесят", "сімдесят", "вісімдесят", "дев'яносто"]
+_UK_HUNDREDS = ["", "сто", "двісті", "триста", "чотириста", "п'ятсот",
+                "шістсот", "сімсот", "вісімсот", "дев'ятсот"]
+
+
+def _uk_under_1000(n: int, feminine: bool = False) -> str:
+    parts = []
+    h, r = divmod(n, 100)
+    if h:
+        parts.append(_UK_HUNDREDS[h])
+    if r >= 20:
+        t, u = divmod(r, 10)
+        parts.append(_UK_TENS[t])
+        r = u
+    if r:
+        if feminine and r == 1:
+            parts.append("одна")
+        elif feminine and r == 2:
+            parts.append("дві")
+        else:
+            parts.append(_UK_ONES[r])
+    return " ".join(parts)
+
+
+def num_to_words_uk(n: int) -> str:
+    if n < 0:
+        return "мінус " + num_to_words_uk(-n)
+    if n == 0:
+        return "нуль"
+    parts = []
+    for div, one, few, many, fem in (
+            (10 ** 9, "мільярд", "мільярди", "мільярдів", False),
+            (10 ** 6, "мільйон", "мільйони", "мільйонів", False),
+            (1000, "тисяча", "тисячі", "тисяч", True)):
+        g, n = divmod(n, div)
+        if g:
+            parts.append(_uk_under_1000(g, fem))
+            parts.append(_ru_plural(g, one, few, many))
+    if n:
+        parts.append(_uk_under_1000(n))
+    return " ".join(p for p in parts if p)
+
+
+# --------------------------------------------------------------------- #
+# Norwegian (bokmål, modern compounds: tjueen)
+# --------------------------------------------------------------------- #
+_NO_ONES = ("null en to tre fire fem seks sju åtte ni ti elleve tolv "
+            "tretten fjorten femten seksten sytten atten nitten").split()
+_NO_TENS = ["", "", "tjue", "tretti", "førti", "femti", "seksti",
+            "sytti", "åtti", "nitti"]
+
+
+def num_to_words_no(n: int) -> str:
+    if n < 0:
+        return "minus " + num_to_words_no(-n)
+    if n < 20:
+        return _NO_ONES[n]
+    if n < 100:
+        t, r = divmod(n, 10)
+        return _NO_TENS[t] + ("" if r == 0 else _NO_ONES[r])
+    if n < 1000:
+        h, r = divmod(n, 100)
+        head = ("hundre" if h == 1 else _NO_ONES[h] + " hundre")
+        return head if r == 0 else head + " og " + num_to_words_no(r)
+    if n < 10 ** 6:
+        t, r = divmod(n, 1000)
+        head = ("tusen" if t == 1 else num_to_words_no(t) + " tusen")
+        return head if r == 0 else head + " " + num_to_words_no(r)
+    if n < 10 ** 9:
+        m, r = divmod(n, 10 ** 6)
+        head = ("en million" if m == 1
+                else num_to_words_no(m) + " millioner")
+        return head if r == 0 else head + " " + num_to_words_no(r)
+    m, r = divmod(n, 10 ** 9)
+    head = ("en milliard" if m == 1
+            else num_to_words_no(m) + " milliarder")
+    return head if r == 0 else head + " " + num_to_words_no(r)
+
+
+# --------------------------------------------------------------------- #
+# Danish (vigesimal tens, unit-og-tens inversion)
+# --------------------------------------------------------------------- #
+_DA_ONES = ("nul en to tre fire fem seks syv otte ni ti elleve tolv "
+            "tretten fjorten femten seksten sytten atten nitten").split()
+_DA_TENS = ["", "", "tyve", "tredive", "fyrre", "halvtreds", "tres",
+            "halvfjerds", "firs", "halvfems"]
+
+
+def num_to_words_da(n: int) -> str:
+    if n < 0:
+        return "minus " + num_to_words_da(-n)
+    if n < 20:
+        return _DA_ONES[n]
+    if n < 100:
+        t, r = divmod(n, 10)
+        if r == 0:
+            return _DA_TENS[t]
+        return _DA_ONES[r] + "og" + _DA_TENS[t]
+    if n < 1000:
+        h, r = divmod(n, 100)
+        head = ("hundrede" if h == 1 else _DA_ONES[h] + " hundrede")
+        return head if r == 0 else head + " og " + num_to_words_da(r)
+    if n < 10 ** 6:
+        t, r = divmod(n, 1000)
+        head = ("tusind" if t == 1 else num_to_words_da(t) + " tusind")
+        return head if r == 0 else head + " " + num_to_words_da(r)
+    m, r = divmod(n, 10 ** 6)
+    head = ("en million" if m == 1
+            else num_to_words_da(m) + " millioner")
+    return head if r == 0 else head + " " + num_to_words_da(r)
+
+
+# --------------------------------------------------------------------- #
+# Finnish (agglutinative: kaksikymmentäyksi)
+# --------------------------------------------------------------------- #
+_FI_ONES = ("nolla yksi kaksi kolme neljä viisi kuusi seitsemän "
+            "kahdeksan yhdeksän kymmenen").split()
+
+
+def num_to_words_fi(n: int) -> str:
+    if n < 0:
+        return "miinus " + num_to_words_fi(-n)
+    if n <= 10:
+        return _FI_ONES[n]
+    if n < 20:
+        return _FI_ONES[n - 10] + "toista"
+    if n < 100:
+        t, r = divmod(n, 10)
+        head = _FI_ONES[t] + "kymmentä"
+        return head if r == 0 else head + _FI_ONES[r]
+    if n < 1000:
+        h, r = divmod(n, 100)
+        head = "sata" if h == 1 else _FI_ONES[h] + "sataa"
+        return head if r == 0 else head + num_to_words_fi(r)
+    if n < 10 ** 6:
+        t, r = divmod(n, 1000)
+        head = ("tuhat" if t == 1
+                else num_to_words_fi(t) + "tuhatta")
+        return head if r == 0 else head + " " + num_to_words_fi(r)
+    m, r = divmod(n, 10 ** 6)
+    head = ("miljoona" if m == 1
+            else num_to_words_fi(m) + " miljoonaa")
+    return head if r == 0 else head + " " + num_to_words_fi(r)
+
+
+# --------------------------------------------------------------------- #
+# Hungarian (tizenX/huszonX, compounds; két- in multiples)
+# --------------------------------------------------------------------- #
+_HU_ONES = ("nulla egy kettő három négy öt hat hét nyolc kilenc "
+            "tíz").split()
+_HU_TENS = ["", "tíz", "húsz", "harminc", "negyven", "ötven",
+            "hatvan", "hetven", "nyolcvan", "kilencven"]
+_HU_TENS_C = ["", "tizen", "huszon", "harminc", "negyven", "ötven",
+              "hatvan", "hetven", "nyolcvan", "kilencven"]
+
+
+def _hu_mult(d: int) -> str:
+    return "két" if d == 2 else _HU_ONES[d]
+
+
+def num_to_words_hu(n: int) -> str:
+    if n < 0:
+        return "mínusz " + num_to_words_hu(-n)
+    if n <= 10:
+        return _HU_ONES[n]
+    if n < 100:
+        t, r = divmod(n, 10)
+        if r == 0:
+            return _HU_TENS[t]
+        return _HU_TENS_C[t] + _HU_ONES[r]
+    if n < 1000:
+        h, r = divmod(n, 100)
+        head = "száz" if h == 1 else _hu_mult(h) + "száz"
+        return head if r == 0 else head + num_to_words_hu(r)
+    if n < 10 ** 6:
+        t, r = divmod(n, 1000)
+        head = ("ezer" if t == 1 else num_to_words_hu(t) + "ezer")
+        return head if r == 0 else head + " " + num_to_words_hu(r)
+    m, r = divmod(n, 10 ** 6)
+    head = num_to_words_hu(m) + " millió"
+    return head if r == 0 else head + " " + num_to_words_hu(r)
+
+
+# --------------------------------------------------------------------- #
+# Greek (neuter counting forms)
+# --------------------------------------------------------------------- #
+_EL_ONES = ("μηδέν ένα δύο τρία τέσσερα πέντε έξι επτά οκτώ εννέα "
+            "δέκα έντεκα δώδεκα").split()
+_EL_TENS = ["", "δέκα", "είκοσι", "τριάντα", "σαράντα", "πενήντα",
+            "εξήντα", "εβδομήντα", "ογδόντα", "ενενήντα"]
+_EL_HUNDREDS = ["", "εκατόν", "διακόσια", "τριακόσια", "τετρακόσια",
+                "πεντακόσια", "εξακόσια", "επτακόσια", "οκτακόσια",
+                "εννιακόσια"]
+
+
+def num_to_words_el(n: int) -> str:
+    if n < 0:
+        return "μείον " + num_to_words_el(-n)
+    if n <= 12:
+        return _EL_ONES[n]
+    if n < 20:
+        return "δεκα" + _EL_ONES[n - 10]
+    if n < 100:
+        t, r = divmod(n, 10)
+        return _EL_TENS[t] + ("" if r == 0 else " " + _EL_ONES[r])
+    if n == 100:
+        return "εκατό"
+    if n < 1000:
+        h, r = divmod(n, 100)
+        head = _EL_HUNDREDS[h]
+        return head if r == 0 else head + " " + num_to_words_el(r)
+    if n < 2000:
+        r = n - 1000
+        return "χίλια" if r == 0 else "χίλια " + num_to_words_el(r)
+    if n < 10 ** 6:
+        t, r = divmod(n, 1000)
+        head = num_to_words_el(t) + " χιλιάδες"
+        return head if r == 0 else head + " " + num_to_words_el(r)
+    m, r = divmod(n, 10 ** 6)
+    head = ("ένα εκατομμύριο" if m == 1
+            else num_to_words_el(m) + " εκατομμύρια")
+    return head if r == 0 else head + " " + num_to_words_el(r)
+
+
+# --------------------------------------------------------------------- #
+# Czech (dvě stě / tři sta / pět set)
+# --------------------------------------------------------------------- #
+_CS_ONES = ("nula jedna dva tři čtyři pět šest sedm osm devět deset "
+            "jedenáct dvanáct třináct čtrnáct patnáct šestnáct "
+            "sedmnáct osmnáct devatenáct").split()
+_CS_TENS = ["", "", "dvacet", "třicet", "čtyřicet", "padesát",
+            "šedesát", "sedmdesát", "osmdesát", "devadesát"]
+
+
+def _cs_hundreds(h: int) -> str:
+    if h == 1:
+        return "sto"
+    if h == 2:
+        return "dvě stě"
+    if h <= 4:
+        return _CS_ONES[h] + " sta"
+    return _CS_ONES[h] + " set"
+
+
+def num_to_words_cs(n: int) -> str:
+    if n < 0:
+        return "mínus " + num_to_words_cs(-n)
+    if n < 20:
+        return _CS_ONES[n]
+    if n < 100:
+        t, r = divmod(n, 10)
+        head = _CS_TENS[t]
+        return head if r == 0 else head + " " + _CS_ONES[r]
+    if n < 1000:
+        h, r = divmod(n, 100)
+        head = _cs_hundreds(h)
+        return head if r == 0 else head + " " + num_to_words_cs(r)
+    if n < 10 ** 6:
+        t, r = divmod(n, 1000)
+        word = _ru_plural(t, "tisíc", "tisíce", "tisíc")
+        head = ("tisíc" if t == 1
+                else num_to_words_cs(t) + " " + word)
+        return head if r == 0 else head + " " + num_to_words_cs(r)
+    m, r = divmod(n, 10 ** 6)
+    word = _ru_plural(m, "milion", "miliony", "milionů")
+    head = ("milion" if m == 1 else num_to_words_cs(m) + " " + word)
+    return head if r == 0 else head + " " + num_to_words_cs(r)
+
+
+# --------------------------------------------------------------------- #
+# Romanian (și-composition, o sută / două sute)
+# --------------------------------------------------------------------- #
+_RO_ONES = ("zero unu doi trei patru cinci șase șapte opt nouă "
+            "zece").split()
+_RO_TEENS = ["", "unsprezece", "doisprezece", "treisprezece",
+             "paisprezece", "cincisprezece", "șaisprezece",
+             "șaptesprezece", "optsprezece", "nouăsprezece"]
+_RO_TENS = ["", "zece", "douăzeci", "treizeci", "patruzeci",
+            "cincizeci", "șaizeci", "șaptezeci", "optzeci", "nouăzeci"]
+
+
+def num_to_words_ro(n: int) -> str:
+    if n < 0:
+        return "minus " + num_to_words_ro(-n)
+    if n <= 10:
+        return _RO_ONES[n]
+    if n < 20:
+        return _RO_TEENS[n - 10]
+    if n < 100:
+        t, r = divmod(n, 10)
+        head = _RO_TENS[t]
+        return head if r == 0 else head + " și " + _RO_ONES[r]
+    if n < 1000:
+        h, r = divmod(n, 100)
+        head = ("o sută" if h == 1 else
+                ("două sute" if h == 2 else _RO_ONES[h] + " sute"))
+        return head if r == 0 else head + " " + num_to_words_ro(r)
+    if n < 10 ** 6:
+        t, r = divmod(n, 1000)
+        head = ("o mie" if t == 1 else
+                ("două mii" if t == 2
+                 else num_to_words_ro(t) + " mii"))
+        return head if r == 0 else head + " " + num_to_words_ro(r)
+    m, r = divmod(n, 10 ** 6)
+    head = ("un milion" if m == 1
+            else num_to_words_ro(m) + " milioane")
+    return head if r == 0 else head + " " + num_to_words_ro(r)
+
+
+# --------------------------------------------------------------------- #
+# Arabic (simplified MSA, masculine nominative; units precede tens
+# with و; gender/case agreement approximated — documented)
+# --------------------------------------------------------------------- #
+_AR_ONES = ("صفر واحد اثنان ثلاثة أربعة خمسة ستة سبعة ثمانية تسعة "
+            "عشرة").split()
+_AR_TEENS = ["", "أحد عشر", "اثنا عشر", "ثلاثة عشر", "أربعة عشر",
+             "خمسة عشر", "ستة عشر", "سبعة عشر", "ثمانية عشر",
+             "تسعة عشر"]
+_AR_TENS = ["", "عشرة", "عشرون", "ثلاثون", "أربعون", "خمسون",
+            "ستون", "سبعون", "ثمانون", "تسعون"]
+
+
+def _ar_under_100(n: int) -> str:
+    if n <= 10:
+        return _AR_ONES[n]
+    if n < 20:
+        return _AR_TEENS[n - 10]
+    t, r = divmod(n, 10)
+    if r == 0:
+        return _AR_TENS[t]
+    return _AR_ONES[r] + " و" + _AR_TENS[t]
+
+
+def _ar_hundreds(h: int) -> str:
+    if h == 1:
+        return "مئة"
+    if h == 2:
+        return "مئتان"
+    return _AR_ONES[h] + "مئة"
+
+
+def num_to_words_ar(n: int) -> str:
+    if n < 0:
+        return "ناقص " + num_to_words_ar(-n)
+    if n < 100:
+        return _ar_under_100(n)
+    if n < 1000:
+        h, r = divmod(n, 100)
+        head = _ar_hundreds(h)
+        return head if r == 0 else head + " و" + num_to_words_ar(r)
+    if n < 10 ** 6:
+        t, r = divmod(n, 1000)
+        if t == 1:
+            head = "ألف"
+        elif t == 2:
+            head = "ألفان"
+        elif t <= 10:
+            head = _AR_ONES[t] + " آلاف"
+        else:
+            head = num_to_words_ar(t) + " ألف"
+        return head if r == 0 else head + " و" + num_to_words_ar(r)
+    m, r = divmod(n, 10 ** 6)
+    head = ("مليون" if m == 1 else num_to_words_ar(m) + " مليون")
+    return head if r == 0 else head + " و" + num_to_words_ar(r)
+
+
+CARDINALS3 = {
+    "uk": num_to_words_uk, "no": num_to_words_no,
+    "da": num_to_words_da, "fi": num_to_words_fi,
+    "hu": num_to_words_hu, "el": num_to_words_el,
+    "cs": num_to_words_cs, "ro": num_to_words_ro,
+    "ar": num_to_words_ar,
+}
+DECIMAL_WORDS3 = {"uk": "кома", "no": "komma", "da": "komma",
+                  "fi": "pilkku", "hu": "egész", "el": "κόμμα",
+                  "cs": "celá", "ro": "virgulă", "ar": "فاصلة"}

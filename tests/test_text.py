@@ -320,3 +320,30 @@ def test_cardinal_grammars_batch2():
     assert "tɯl" not in text_to_phonemes("25", "ko")[0]
     out = text_to_phonemes("Это 21 год.", "ru")[0]
     assert "adʲˈin" in out or "ɐdʲˈin" in out
+
+
+def test_cardinal_grammars_batch3():
+    """Third batch: uk/no/da/fi/hu/el/cs/ro/ar full number reading."""
+    from sonata_amd.text.normalize import normalize
+    from sonata_amd.text.numbers3 import (num_to_words_ar,
+                                          num_to_words_cs,
+                                          num_to_words_da,
+                                          num_to_words_el,
+                                          num_to_words_fi,
+                                          num_to_words_hu,
+                                          num_to_words_no,
+                                          num_to_words_ro,
+                                          num_to_words_uk)
+
+    assert num_to_words_uk(2000) == "дві тисячі"
+    assert num_to_words_no(21) == "tjueen"
+    assert num_to_words_da(95) == "femoghalvfems"   # vigesimal tens
+    assert num_to_words_fi(21) == "kaksikymmentäyksi"
+    assert num_to_words_hu(200) == "kétszáz"        # két- multiple
+    assert num_to_words_el(2000) == "δύο χιλιάδες"
+    assert num_to_words_cs(300) == "tři sta"        # sta/set forms
+    assert num_to_words_cs(500) == "pět set"
+    assert num_to_words_ro(21) == "douăzeci și unu"
+    assert num_to_words_ar(23) == "ثلاثة وعشرون"    # unit و tens order
+    assert normalize("3,5", "uk") == "три кома п'ять"
+    assert normalize("42", "fi") == "neljäkymmentäkaksi"
