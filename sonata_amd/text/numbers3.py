@@ -393,3 +393,45 @@ CARDINALS3 = {
 DECIMAL_WORDS3 = {"uk": "кома", "no": "komma", "da": "komma",
                   "fi": "pilkku", "hu": "egész", "el": "κόμμα",
                   "cs": "celá", "ro": "virgulă", "ar": "فاصلة"}
+
+
+# --------------------------------------------------------------------- #
+# Hindi: 0-99 are lexical; Indian grouping सौ/हज़ार/लाख/करोड़
+# --------------------------------------------------------------------- #
+_HI_0_99 = (
+    "शून्य एक दो तीन चार पाँच छह सात आठ नौ दस "
+    "ग्यारह बारह तेरह चौदह पंद्रह सोलह सत्रह अठारह उन्नीस बीस "
+    "इक्कीस बाईस तेईस चौबीस पच्चीस छब्बीस सत्ताईस अट्ठाईस उनतीस तीस "
+    "इकतीस बत्तीस तैंतीस चौंतीस पैंतीस छत्तीस सैंतीस अड़तीस उनतालीस चालीस "
+    "इकतालीस बयालीस तैंतालीस चौवालीस पैंतालीस छियालीस सैंतालीस अड़तालीस "
+    "उनचास पचास "
+    "इक्यावन बावन तिरपन चौवन पचपन छप्पन सत्तावन अट्ठावन उनसठ साठ "
+    "इकसठ बासठ तिरसठ चौंसठ पैंसठ छियासठ सड़सठ अड़सठ उनहत्तर सत्तर "
+    "इकहत्तर बहत्तर तिहत्तर चौहत्तर पचहत्तर छिहत्तर सतहत्तर अठहत्तर "
+    "उनासी अस्सी "
+    "इक्यासी बयासी तिरासी चौरासी पचासी छियासी सत्तासी अट्ठासी नवासी नब्बे "
+    "इक्यानवे बानवे तिरानवे चौरानवे पचानवे छियानवे सत्तानवे अट्ठानवे "
+    "निन्यानवे"
+).split()
+
+
+def num_to_words_hi(n: int) -> str:
+    if n < 0:
+        return "माइनस " + num_to_words_hi(-n)
+    if n < 100:
+        return _HI_0_99[n]
+    parts = []
+    for div, name in ((10 ** 7, "करोड़"), (10 ** 5, "लाख"),
+                      (1000, "हज़ार"), (100, "सौ")):
+        g, n = divmod(n, div)
+        if g:
+            parts.append(_HI_0_99[g] if g < 100
+                         else num_to_words_hi(g))
+            parts.append(name)
+    if n:
+        parts.append(_HI_0_99[n])
+    return " ".join(parts)
+
+
+CARDINALS3["hi"] = num_to_words_hi
+DECIMAL_WORDS3["hi"] = "दशमलव"

@@ -347,3 +347,16 @@ def test_cardinal_grammars_batch3():
     assert num_to_words_ar(23) == "ثلاثة وعشرون"    # unit و tens order
     assert normalize("3,5", "uk") == "три кома п'ять"
     assert normalize("42", "fi") == "neljäkymmentäkaksi"
+
+
+def test_cardinal_grammar_hindi():
+    """Hindi: lexical 0-99 + Indian grouping (सौ/हज़ार/लाख/करोड़)."""
+    from sonata_amd.text.numbers3 import num_to_words_hi
+    from sonata_amd.text.phonemizer import text_to_phonemes
+
+    assert num_to_words_hi(21) == "इक्कीस"
+    assert num_to_words_hi(99) == "निन्यानवे"
+    assert num_to_words_hi(345) == "तीन सौ पैंतालीस"
+    assert num_to_words_hi(123456) == "एक लाख तेईस हज़ार चार सौ छप्पन"
+    out = text_to_phonemes("मेरे पास 25 किताबें हैं।", "hi")[0]
+    assert "pətʃtʃiːs" in out
