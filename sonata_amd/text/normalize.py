@@ -552,6 +552,44 @@ _PERCENT_WORDS = {
 }
 
 
+# currency symbol -> unit word, per language (plural/base counting
+# form; case agreement approximated for the Slavic languages)
+_CURRENCY_WORDS = {
+    "de": {"€": "Euro", "$": "Dollar", "£": "Pfund"},
+    "fr": {"€": "euros", "$": "dollars", "£": "livres"},
+    "es": {"€": "euros", "$": "dólares", "£": "libras"},
+    "it": {"€": "euro", "$": "dollari", "£": "sterline"},
+    "pt": {"€": "euros", "$": "dólares", "£": "libras"},
+    "nl": {"€": "euro", "$": "dollar", "£": "pond"},
+    "sv": {"€": "euro", "$": "dollar", "kr": "kronor"},
+    "no": {"€": "euro", "$": "dollar", "kr": "kroner"},
+    "da": {"€": "euro", "$": "dollar", "kr": "kroner"},
+    "fi": {"€": "euroa", "$": "dollaria"},
+    "pl": {"€": "euro", "$": "dolarów", "zł": "złotych"},
+    "cs": {"€": "eur", "$": "dolarů", "Kč": "korun"},
+    "ru": {"€": "евро", "$": "долларов", "₽": "рублей"},
+    "uk": {"€": "євро", "$": "доларів", "₴": "гривень"},
+    "tr": {"€": "avro", "$": "dolar", "₺": "lira"},
+    "el": {"€": "ευρώ", "$": "δολάρια"},
+    "ro": {"€": "euro", "$": "dolari"},
+    "hu": {"€": "euró", "$": "dollár", "Ft": "forint"},
+    "id": {"$": "dolar", "Rp": "rupiah"},
+    "hi": {"₹": "रुपये", "$": "डॉलर"},
+    "ja": {"¥": "えん", "円": "えん", "$": "ドル"},
+    "ko": {"₩": "원", "$": "달러"},
+    "ar": {"$": "دولار", "€": "يورو"},
+}
+# the word between hours and minutes in clock times (14:30); only
+# languages where "H <word> MM" is a natural reading — Turkish (saat
+# precedes), Polish (ordinal hours) and Korean (native-numeral hours)
+# are left to digit reading rather than said wrongly
+_TIME_WORDS = {
+    "de": "Uhr", "fr": "heures", "es": "horas", "it": "e",
+    "pt": "horas", "nl": "uur", "sv": "och", "ru": "часов",
+    "uk": "годин", "fi": "ja", "ja": "じ",
+}
+
+
 def normalize(text: str, language: str) -> str:
     """Expand digits/abbreviations for `language` (base code).
 
@@ -565,6 +603,25 @@ def normalize(text: str, language: str) -> str:
     pw = _PERCENT_WORDS.get(base)
     if pw is not None:
         text = _PERCENT_RE.sub(lambda m: m.group(1) + " " + pw, text)
+    cw = _CURRENCY_WORDS.get(base)
+    if cw is not None:
+        # €5 / 5€ / $5 / £5 -> "5 <unit>" before number expansion
+        num = r"(\d[\d.,]*\d|\d)"  # cannot end in punctuation
+        for sym, word in cw.items():
+            text = re.sub(
+                rf"{re.escape(sym)}\s?{num}", rf"\1 {word}", text)
+            text = re.sub(
+                rf"{num}\s?{re.escape(sym)}", rf"\1 {word}", text)
+    tw = _TIME_WORDS.get(base)
+    if tw is not None:
+        # 14:30 -> "14 <hour-word> 30" (espeak-style clock reading)
+        def _clock(m: re.Match) -> str:
+            h, mm = int(m.group(1)), int(m.group(2))
+            if h > 23 or mm > 59:
+                return m.group(0)
+            out = f"{h} {tw}"
+            return out if mm == 0 else f"{out} {mm}"
+        text = _TIME_RE.sub(_clock, text)
     card = _CARDINALS.get(base)
     if card is not None:
         digits = _DIGITS[base]

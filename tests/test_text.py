@@ -360,3 +360,17 @@ def test_cardinal_grammar_hindi():
     assert num_to_words_hi(123456) == "एक लाख तेईस हज़ार चार सौ छप्पन"
     out = text_to_phonemes("मेरे पास 25 किताबें हैं।", "hi")[0]
     assert "pətʃtʃiːs" in out
+
+
+def test_currency_and_time_other_languages():
+    from sonata_amd.text.normalize import normalize
+
+    assert normalize("Das kostet 25€.", "de") == \
+        "Das kostet fünfundzwanzig Euro."
+    assert normalize("Es ist 14:30.", "de") == \
+        "Es ist vierzehn Uhr dreißig."
+    assert normalize("Il est 14:30.", "fr") == \
+        "Il est quatorze heures trente."
+    assert normalize("Стоит 100₽.", "ru") == "Стоит сто рублей."
+    assert normalize("₹250", "hi") == "दो सौ पचास रुपये"
+    assert normalize("가격은 ₩5000.", "ko") == "가격은 오천 원."
