@@ -590,6 +590,21 @@ _TIME_WORDS = {
 }
 
 
+# Germanic year-style reading of bare 1100-1999 ("neunzehnhundert...",
+# also idiomatic for counts: "fünfzehnhundert Meter")
+_TEEN_HUNDRED_RE = re.compile(r"\b(1[1-9])(\d\d)\b")
+
+
+def _teen_hundreds(base: str, card, text: str) -> str:
+    word = {"de": "hundert", "nl": "honderd", "sv": "hundra"}[base]
+
+    def _sub(m: re.Match) -> str:
+        h, r = int(m.group(1)), int(m.group(2))
+        return card(h) + word + ("" if r == 0 else card(r))
+
+    return _TEEN_HUNDRED_RE.sub(_sub, text)
+
+
 def normalize(text: str, language: str) -> str:
     """Expand digits/abbreviations for `language` (base code).
 
@@ -640,6 +655,8 @@ def normalize(text: str, language: str) -> str:
             text = _DEC_COMMA_RE.sub(
                 lambda m: card(int(m.group(1))) + " " + dec + " "
                 + " ".join(digits[int(d)] for d in m.group(2)), text)
+        if base in ("de", "nl", "sv"):
+            text = _teen_hundreds(base, card, text)
         return _INT_RE.sub(
             lambda m: card(int(m.group(0))) if len(m.group(0)) <= 12
             else " ".join(digits[int(d)] for d in m.group(0)), text)

@@ -374,3 +374,15 @@ def test_currency_and_time_other_languages():
     assert normalize("Стоит 100₽.", "ru") == "Стоит сто рублей."
     assert normalize("₹250", "hi") == "दो सौ पचास रुपये"
     assert normalize("가격은 ₩5000.", "ko") == "가격은 오천 원."
+
+
+def test_teen_hundreds_germanic():
+    from sonata_amd.text.normalize import normalize
+
+    assert normalize("Im Jahr 1984.", "de") == \
+        "Im Jahr neunzehnhundertvierundachtzig."
+    assert normalize("In 1923.", "nl") == \
+        "In negentienhonderddrieëntwintig."
+    assert normalize("År 1950.", "sv") == "År nittonhundrafemtio."
+    # 2000s stay plain cardinals
+    assert normalize("2024", "de") == "zweitausendvierundzwanzig"
