@@ -137,3 +137,29 @@ def test_all_languages_end_to_end(tmp_path):
         v = load_voice(pack, device="cpu")
         audios = v.speak_batch(list(v.phonemize_text(texts[lang])))
         assert audios and all(len(a.samples) > 200 for a in audios), lang
+
+
+@pytest.mark.gpu
+def test_gpu_multilingual_synthesis(tmp_path):
+    """Third-batch languages run the full GPU path (text -> script
+    engine -> ids -> C++ engine -> waveform): one utterance each for a
+    Brahmic, Hangul, Arabic(+tashkeel), Tamil, kana and Persian voice."""
+    from sonata_amd.models.voice import load_voice
+
+    texts = {
+        "hi": "नमस्ते दुनिया, मैं हिंदी बोलता हूँ।",
+        "ko": "안녕하세요, 한국어를 말합니다.",
+        "ar": "مرحبا بالعالم.",
+        "ta": "வணக்கம் உலகம்.",
+        "ja": "こんにちは、日本語を話します。",
+        "fa": "سلام دنیا، فارسی حرف می‌زنم.",
+    }
+    for lang, text in texts.items():
+        pack = create_random_voice(str(tmp_path), f"g_{lang}",
+                                   quality="x_low", language=lang)
+        v = load_voice(pack, device="cuda")
+        sents = v.phonemize_text(text)
+        assert sents and sents[0], lang
+        audio = v.speak_one_sentence(sents[0])
+        assert len(audio.samples) > 1000, lang
+        assert float(abs(audio.samples).max()) > 0, lang
