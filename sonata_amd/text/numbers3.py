@@ -352,10 +352,14 @@ def _ar_under_100(n: int) -> str:
 
 def _ar_hundreds(h: int) -> str:
     if h == 1:
-        return "مئة"
+        return "مائة"
     if h == 2:
-        return "مئتان"
-    return _AR_ONES[h] + "مئة"
+        return "مائتان"
+    # the unit's ta-marbuta drops in the compound: ثلاثمائة
+    unit = _AR_ONES[h]
+    if unit.endswith("ة"):
+        unit = unit[:-1]
+    return unit + "مائة"
 
 
 def num_to_words_ar(n: int) -> str:

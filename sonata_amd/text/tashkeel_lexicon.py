@@ -99,6 +99,13 @@ _WORDS: List[str] = [
     "سِتَّة", "سَبْعَة", "ثَمَانِيَة", "تِسْعَة", "عَشَرَة",
     "عِشْرُونَ", "مِائَة", "أَلْف", "مِلْيُون", "أَوَّل", "آخِر",
     "نِصْف", "رُبْع",
+    # number-grammar forms (normalize emits these for digits)
+    "صِفْر", "عَشَرَ", "ثَلَاثُونَ", "أَرْبَعُونَ", "خَمْسُونَ",
+    "سِتُّونَ", "سَبْعُونَ", "ثَمَانُونَ", "تِسْعُونَ",
+    "مِائَتَانِ", "أَلْفَانِ", "آلَاف", "مِلْيَار", "نَاقِص",
+    "فَاصِلَة", "أَحَدَ", "اِثْنَا",
+    "ثَلَاثُمِائَة", "أَرْبَعُمِائَة", "خَمْسُمِائَة", "سِتُّمِائَة",
+    "سَبْعُمِائَة", "ثَمَانِيمِائَة", "تِسْعُمِائَة",
 ]
 
 # first occurrence wins: _WORDS is frequency-ordered, so for ambiguous

@@ -159,3 +159,17 @@ def test_tashkeel_onnx_unknown_layout_fails_loudly(tmp_path):
     _emit_onnx(tensors, onnx_path)
     with pytest.raises(ModelError, match="not recognized"):
         import_tashkeel_onnx(onnx_path, str(tmp_path / "x.safetensors"))
+
+
+def test_number_words_diacritized():
+    """normalize's Arabic number grammar output is covered by the
+    lexicon (incl. compound hundreds), so digits in ar text get real
+    vowels instead of net guesses."""
+    from sonata_amd.text.numbers3 import num_to_words_ar
+    from sonata_amd.text.tashkeel import TashkeelModel
+
+    t = TashkeelModel.default()
+    assert t.diacritize(num_to_words_ar(23)) == "ثَلَاثَة وَعِشْرُونَ"
+    assert t.diacritize(num_to_words_ar(345)) == \
+        "ثَلَاثُمِائَة وَخَمْسَة وَأَرْبَعُونَ"
+    assert t.diacritize(num_to_words_ar(1000)) == "أَلْف"
