@@ -751,6 +751,41 @@ SMJ_RULES = {  # Lule Sami (approximate)
 }
 
 
+QYA_RULES = {  # Quenya (Tolkien's published phonology; qu = kw)
+    "qu": "kw", "hw": "ʍ", "hy": "ç", "ty": "c", "ny": "ɲ",
+    "ly": "ʎ", "th": "θ", "ch": "x", "x": "ks",
+    "ai": "ai", "au": "au", "oi": "oi", "ui": "ui", "eu": "eu",
+    "iu": "ju",
+    "á": "aː", "é": "eː", "í": "iː", "ó": "oː", "ú": "uː",
+    "ë": "e", "ñ": "ŋ",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "h", "i": "i", "l": "l", "m": "m", "n": "n",
+    "o": "o", "p": "p", "r": "r", "s": "s", "t": "t", "u": "u",
+    "v": "v", "w": "w", "y": "j",
+}
+
+SJN_RULES = {  # Sindarin: c always /k/, ch = /x/, dh = /ð/
+    "ch": "x", "dh": "ð", "th": "θ", "lh": "ɬ", "rh": "r̥",
+    "ph": "f", "ng": "ŋ",
+    "ae": "ae", "ai": "ai", "au": "au", "ei": "ei", "ui": "ui",
+    "oe": "oe",
+    "á": "aː", "é": "eː", "í": "iː", "ó": "oː", "ú": "uː",
+    "û": "uː", "î": "iː", "ê": "eː", "ŷ": "yː", "ý": "yː",
+    "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
+    "g": "ɡ", "h": "h", "i": "i", "l": "l", "m": "m", "n": "n",
+    "o": "o", "p": "p", "r": "r", "s": "s", "t": "t", "u": "u",
+    "v": "v", "w": "w", "y": "y",
+}
+
+PIQD_RULES = {  # Klingon (after case-folding: D/H/I/S/Q lose case)
+    "tlh": "tɬ", "ch": "tʃ", "gh": "ɣ", "ng": "ŋ",
+    "a": "ɑ", "b": "b", "d": "ɖ", "e": "ɛ", "h": "x", "i": "ɪ",
+    "j": "dʒ", "l": "l", "m": "m", "n": "n", "o": "o", "p": "pʰ",
+    "q": "qʰ", "r": "r", "s": "ʂ", "t": "tʰ", "u": "u", "v": "v",
+    "w": "w", "y": "j", "'": "ʔ",
+}
+
+
 # ===================================================================== #
 # Registry
 # ===================================================================== #
@@ -764,6 +799,7 @@ TABLES3 = {
     "cv": CV_RULES,
     "an": AN_RULES, "ku": KU_RULES, "gd": GD_RULES, "quc": QUC_RULES,
     "sd": SD_RULES, "nog": NOG_RULES, "smj": SMJ_RULES,
+    "qya": QYA_RULES, "sjn": SJN_RULES, "piqd": PIQD_RULES,
 }
 
 _AR_BLOCK = "؀-ۿ"
@@ -785,6 +821,8 @@ LETTERS3 = {
     "grc": "α-ωΑ-Ωἀ-ῼάέήίόύώΐΰ",
     "tt": _CYR + "әөүҗңһ", "ba": _CYR + "әөүңһҙҫғҡ",
     "cv": _CYR + "ӑӗҫӳ",
+    "qya": "a-zA-Záéíóúëñ", "sjn": "a-zA-Záéíóúûîêŷý",
+    "piqd": "a-zA-Z'",
     "an": "a-zA-Zñáéíóú", "ku": "a-zA-Zçşêîû",
     "gd": "a-zA-Zàèéìòóù", "quc": "a-zA-Zä'",
     "sd": _AR_BLOCK, "nog": _CYR, "smj": "a-zA-Záåŋæä",
@@ -800,6 +838,7 @@ STRESS3 = {
     "cv": "final",
     "an": "es-penult", "ku": "final", "gd": "first", "quc": "final",
     "sd": "first", "nog": "final", "smj": "first",
+    "qya": "penult", "sjn": "first", "piqd": "final",
 }
 
 VI_LEXICON = {

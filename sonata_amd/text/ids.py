@@ -37,6 +37,8 @@ _IPA_SYMBOLS = (
     # third batch, table languages: uvular stop/fricative (fa/kl),
     # bilabial fricative (ja), precomposed nasal vowels (gn/ur)
     "ɢɸχẽĩũ",
+    # conlang batch (qya/sjn): voiceless w, combining ring (r̥)
+    "ʍ̥",
 )
 
 

@@ -217,6 +217,14 @@ _DIGITS.update({
             "gáktsa aktse").split(),
 })
 
+_DIGITS.update({
+    # conlang batch (attested forms; zero approximated where the
+    # corpus has none: Quenya munta "nothing", Sindarin ú negation)
+    "qya": "munta minë atta neldë canta lempë enquë otso tolto nertë".split(),
+    "sjn": "ú min tad neled canad leben eneg odog toloth neder".split(),
+    "piqd": "pagh wa' cha' wej loS vagh jav Soch chorgh Hut".split(),
+})
+
 # orthography aliases share digit tables
 for _alias, _src in (("nb", "no"), ("nn", "no"), ("sr", "hr"),
                      ("bs", "hr"), ("ms", "id")):

@@ -716,7 +716,8 @@ _INDIC = ("mr", "ne", "kok", "bn", "as", "bpy", "gu", "pa", "or",
 _BATCH3 = ("fa", "ur", "ug", "he", "vi", "mi", "haw", "qu", "gn",
            "nci", "om", "tn", "pap", "ia", "io", "lfn", "jbo", "tk",
            "lb", "kl", "ga", "grc", "tt", "ba", "cv",
-           "an", "ku", "gd", "quc", "sd", "nog", "smj")
+           "an", "ku", "gd", "quc", "sd", "nog", "smj",
+           "qya", "sjn", "piqd")
 
 
 def _get_g2p(voice: str) -> RuleG2P:
@@ -942,7 +943,8 @@ def available_languages() -> List[str]:
             "fa", "ur", "ug", "he", "vi", "mi", "haw", "qu", "gn",
             "nci", "om", "tn", "pap", "ia", "io", "lfn", "jbo", "tk",
             "lb", "kl", "ga", "grc", "tt", "ba", "cv",
-            "an", "ku", "gd", "quc", "sd", "nog", "smj", "bpy", "shn"]
+            "an", "ku", "gd", "quc", "sd", "nog", "smj", "bpy", "shn",
+            "qya", "sjn", "piqd"]
 
 
 # script-native sentence/clause punctuation -> ASCII so the splitter

@@ -383,6 +383,9 @@ def test_batch3_all_languages_nonempty():
         "an": "ola mundo", "ku": "silav cîhan", "gd": "halò saoghail",
         "quc": "saqarik uleew", "sd": "سلام دنيا",
         "nog": "салам дуныя", "smj": "buoris",
+        "qya": "elen síla lúmenn omentielvo",
+        "sjn": "mae govannen mellon nîn",
+        "piqd": "tlhIngan Hol vIjatlh Qapla'",
     }
     from sonata_amd.text.phonemizer import _BATCH3
     assert set(samples) == set(_BATCH3) | {"ja"}
@@ -428,6 +431,8 @@ def test_batch3_symbols_encodable():
         "ku": "silav cîhan ez kurdî", "gd": "halò a shaoghail loch",
         "quc": "saqarik uleew utz awach", "sd": "سلام دنيا سنڌي",
         "nog": "салам дуныя аьел", "smj": "buoris sjaddat",
+        "qya": "elen síla lúmenn hwesta", "sjn": "mae govannen lhaw",
+        "piqd": "tlhIngan Hol Qapla'",
     }
     for lang, txt in samples.items():
         for sent in text_to_phonemes(txt, lang):

@@ -130,6 +130,9 @@ def test_all_languages_end_to_end(tmp_path):
         "sd": "سلام دنيا.", "nog": "Салам дуныя.",
         "smj": "Buoris væráldda.",
         "bpy": "আমার ঠার.", "shn": "မႂ်ႇသုင် ၵမ်ႇၽႃႇ.",
+        "qya": "Elen síla lúmenn omentielvo.",
+        "sjn": "Mae govannen, mellon nîn.",
+        "piqd": "tlhIngan Hol vIjatlh.",
     }
     for lang in available_languages():
         pack = create_random_voice(str(tmp_path), f"lang_{lang}",
