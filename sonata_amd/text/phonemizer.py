@@ -751,6 +751,9 @@ def _get_g2p(voice: str) -> RuleG2P:
             if (len(w) > 3 and w.endswith("e") and w[-2] not in "aeiouy"
                     and ipa.endswith("ɛ")):
                 ipa = ipa[:-1]               # silent final e (blouse)
+            if len(w) > 6 and w.endswith(("ence", "ance")):
+                if ipa.endswith(("ɛns", "æns")):
+                    ipa = ipa[:-3] + "əns"   # difference, importance
             return ipa
 
         _en_rules_orig = g._apply_rules
