@@ -811,6 +811,18 @@ RU_STRESS = {
     "любовь": 2, "отец": 2, "жена": 2, "сестра": 2, "число": 2,
     "письмо": 2, "столы": 2, "цветы": 2, "часы": 2, "цена": 2,
     "дела": 2, "дома": 1, "утром": 1,
+    # second batch (r2 final sessions): frequent non-initial stress
+    "возможно": 2, "например": 3, "вопросы": 2, "проблема": 2,
+    "машина": 2, "дорога": 2, "минута": 2, "неделя": 2,
+    "погода": 2, "собака": 2, "газета": 2, "квартира": 2,
+    "картина": 2, "бумага": 2, "столица": 2, "граница": 2,
+    "больница": 2, "учитель": 2, "магазин": 3, "автобус": 2,
+    "природа": 2, "свобода": 2, "наука": 2, "культура": 2,
+    "история": 2, "программа": 2, "система": 2, "секунда": 2,
+    "победа": 2, "надежда": 2, "ошибка": 2, "улыбка": 2,
+    "вчера": 2, "весна": 2, "зима": 2, "гроза": 2,
+    "интересно": 3, "красиво": 2, "приятно": 2, "огромный": 2,
+    "дорогой": 3, "молодой": 3, "простой": 2, "сложный": 1,
 }
 
 # full-IPA overrides: orthography-irregular words (г -> в genitives,

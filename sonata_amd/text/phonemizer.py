@@ -820,6 +820,23 @@ def _get_g2p(voice: str) -> RuleG2P:
         g = RuleG2P(ru_palatal_rules(_RU_RULES), letters="а-яА-ЯёЁ")
         g.lexicon = ru_build_lexicon(g._apply_rules, g._VOWELS)
         g.postprocess = ru_reduce
+
+        # ё is always the stressed syllable in Russian — place stress
+        # there for out-of-lexicon words before any other heuristic
+        _cyr_v = "аеёиоуыэюя"
+        _orig_stress = g._stress_rules_output
+
+        def _ru_stress(word, ipa, _o=_orig_stress, _g=g):
+            if "ё" in word:
+                idx = sum(1 for c in word[:word.index("ё")]
+                          if c in _cyr_v)
+                starts = _g._vowel_clusters(ipa)
+                if idx < len(starts):
+                    pos = starts[idx]
+                    return ipa[:pos] + "ˈ" + ipa[pos:]
+            return _o(word, ipa)
+
+        g._stress_rules_output = _ru_stress
     elif base == "nl":
         g = RuleG2P(_NL_RULES, letters="a-zA-Z")
         g.stress_skip_prefixes = ("ge", "be", "ver", "ont", "her")

@@ -605,3 +605,15 @@ def test_fr_context_markers():
     wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
              if g.word_to_ipa(w) != want]
     assert not wrong, wrong
+
+
+def test_ru_yo_always_stressed():
+    """ё marks the stressed syllable in Russian — rule-path words with
+    ё get stress there, not on the first-syllable default."""
+    g = _get_g2p("ru")
+    assert g.word_to_ipa("самолёт") == "samɐlʲˈot"
+    assert g.word_to_ipa("тёплый") == "tʲˈoplɨj"
+    assert g.word_to_ipa("зелёный") == "zʲɪlʲˈonɨj"
+    # second stress-lexicon batch
+    assert g.word_to_ipa("машина") == "maʂˈɨna"
+    assert g.word_to_ipa("возможно") == "vɐzmˈoʐnɐ"
