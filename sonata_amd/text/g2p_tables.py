@@ -586,7 +586,10 @@ def fr_preprocess(w: str) -> str:
     """Strip silent final consonants / e-muet before rule application."""
     if w in _FR_KEEP_FINAL or len(w) <= 2:
         return w
+    w = _re.sub(r"ier$", "Jé", w)          # dernier -> dernJé (= /je/)
     w = _re.sub(r"(er|ez)$", "é", w)       # parler -> parlé (= /e/)
+    w = _re.sub(r"ie$", "i", w)            # vie -> vi (no schwa)
+    w = _re.sub(r"ée$", "é", w)            # musée -> musé
     w = _re.sub(r"[stdxzp]$", "", w)       # petit, temps, grand, prix...
     w = _re.sub(r"[stdxzp]$", "", w)       # temps: strip s then p
     # context markers (uppercase = direct rules in _FR_RULES): protect
@@ -663,6 +666,15 @@ FR_LEXICON = {
     "belle": "bɛl", "nouveau": "nuvˈo", "nouvelle": "nuvˈɛl",
     "vieux": "vjø", "jeune": "ʒœn", "autre": "otʁ", "même": "mɛm",
     "seul": "sœl", "chose": "ʃoz", "choses": "ʃoz",
+    # r2 final batch: rule-resistant everyday words
+    "avec": "avˈɛk", "hôtel": "otˈɛl", "exemple": "ɛɡzˈɑ̃pl",
+    "examen": "ɛɡzamˈɛ̃", "exact": "ɛɡzˈakt",
+    "juillet": "ʒɥijˈɛ", "août": "ˈut", "yeux": "jø",
+    "messieurs": "mesjˈø", "mesdames": "medˈam",
+    "second": "səɡˈɔ̃", "seconde": "səɡˈɔ̃d",
+    "photo": "fotˈo", "vélo": "velˈo", "numéro": "nymeʁˈo",
+    "enfants": "ɑ̃fˈɑ̃", "gens": "ʒɑ̃", "corps": "kˈɔʁ",
+    "temps": "tɑ̃", "printemps": "pʁɛ̃tˈɑ̃",
 }
 
 

@@ -108,10 +108,15 @@ class RuleG2P:
         ("ic", 1),
     ]
 
+    # English-specific derivational layer (-s/-ed/-ing voicing rules);
+    # MUST stay off for other languages (French plural -s is silent —
+    # "enfants" must not get an English /z/)
+    english_inflections = False
+
     def word_to_ipa(self, word: str) -> str:
         w = word.lower()
         ipa = self.lexicon.get(w)
-        if ipa is None:
+        if ipa is None and self.english_inflections:
             ipa = self._inflect(w)
         if ipa is None:
             ipa = self._apply_rules(
@@ -543,6 +548,10 @@ _FR_RULES = {
     "eil": "ɛj", "ail": "aj", "euil": "œj", "ueil": "œj",
     "ien": "jɛ̃", "ss": "s", "N": "n", "M": "m", "G": "ɡ",
     "ill": "ij", "rr": "ʁ", "que": "k", "er": "ɛʁ",
+    "ier": "je", "ion": "jɔ̃", "ia": "ja", "tioN": "sjɔn",
+    "J": "j",
+    "tt": "t", "pp": "p", "ff": "f", "bb": "b", "dd": "d",
+    "ll": "l",
     "qu": "k", "ou": "u", "oi": "wa", "au": "o", "ai": "ɛ", "ei": "ɛ",
     "eu": "ø", "an": "ɑ̃", "am": "ɑ̃", "en": "ɑ̃", "em": "ɑ̃",
     "on": "ɔ̃", "om": "ɔ̃", "in": "ɛ̃", "im": "ɛ̃", "un": "œ̃",
@@ -735,6 +744,7 @@ def _get_g2p(voice: str) -> RuleG2P:
         g = RuleG2P(_EN_RULES, merged, letters="a-zA-Z",
                     unstressed=_EN_UNSTRESSED)
         g.spell_acronyms = True
+        g.english_inflections = True
 
         def _en_fix(w: str, ipa: str) -> str:
             # orthography-aware ending fixes the context-free rules

@@ -617,3 +617,18 @@ def test_ru_yo_always_stressed():
     # second stress-lexicon batch
     assert g.word_to_ipa("машина") == "maʂˈɨna"
     assert g.word_to_ipa("возможно") == "vɐzmˈoʐnɐ"
+
+
+def test_fr_quality_round2_final():
+    """-ier = /je/, -tion- before vowels = /sjɔn/, double consonants
+    collapse, silent plural -s (the English inflection layer must NOT
+    fire for French)."""
+    g = _get_g2p("fr")
+    assert g.word_to_ipa("dernier") == "dɛʁnjˈe"
+    assert g.word_to_ipa("premier") == "pʁəmjˈe"
+    assert g.word_to_ipa("national") == "nasjɔnˈal"
+    assert g.word_to_ipa("attention") == "atɑ̃sjˈɔ̃"
+    assert g.word_to_ipa("enfants") == "ɑ̃fˈɑ̃"   # silent -s, no /z/
+    assert g.word_to_ipa("vie") == "vˈi"
+    assert g.word_to_ipa("musée") == "myzˈe"
+    assert g.word_to_ipa("avion") == "avjˈɔ̃"
