@@ -695,20 +695,31 @@ def de_postprocess(ipa: str) -> str:
     # -er coda -> vocalized ɐ
     if ipa.endswith("ɛʁ"):
         ipa = ipa[:-2] + "ɐ"
-    # unstressed -en / -e endings use schwa
+    # unstressed -en / -e / -es endings use schwa; final s devoices
     if ipa.endswith("ɛn"):
         ipa = ipa[:-2] + "ən"
+    elif ipa.endswith("ɛz"):
+        ipa = ipa[:-2] + "əs"
     elif ipa.endswith("ɛ"):
         ipa = ipa[:-1] + "ə"
+    if ipa.endswith("z"):
+        ipa = ipa[:-1] + "s"
     return ipa
 
 
 def de_preprocess(w: str) -> str:
     """Collapse double consonants (they mark the previous vowel short,
     not a geminate): wasser -> waser.  Word-final -ig is [ɪç]
-    (zwanzig), unlike genuine -ik loans (Musik)."""
+    (zwanzig), unlike genuine -ik loans (Musik).  Morpheme-initial
+    st/sp (word start or after an unstressed verb prefix) become the
+    St/Sp markers = [ʃt]/[ʃp]; elsewhere they stay plain (Dienstag,
+    lustig, August)."""
     if w.endswith("ig"):
         w = w[:-2] + "ich"
+    w = _re.sub(
+        r"^(ver|vor|be|er|ent|zer|ge|an|auf|aus|ab|über|unter|ein|"
+        r"mit|miss|emp|durch|um|nach|weg|zu)?s([tp])",
+        lambda m: (m.group(1) or "") + "S" + m.group(2), w)
     out = []
     for ch in w:
         if out and out[-1] == ch and ch not in "aeiouäöü":
@@ -754,6 +765,10 @@ DE_LEXICON = {
     "wasser": "vˈasɐ", "heute": "hˈɔʏtə", "morgen": "mˈɔʁɡən",
     "deutschland": "dˈɔʏtʃlant", "deutsch": "dɔʏtʃ", "hallo": "halˈoː",
     "danke": "dˈaŋkə", "bitte": "bˈɪtə",
+    # superlatives with plain [st] (the be- prefix rule would give ʃt)
+    "beste": "bˈɛstə", "besten": "bˈɛstən", "bester": "bˈɛstɐ",
+    "bestes": "bˈɛstəs", "am": "am", "erste": "ˈeːʁstə",
+    "ersten": "ˈeːʁstən", "erster": "ˈeːʁstɐ",
 }
 
 

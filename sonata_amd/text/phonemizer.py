@@ -488,7 +488,12 @@ _DE_RULES = {
     "ei": "aɪ", "ai": "aɪ", "au": "aʊ", "eu": "ɔʏ", "äu": "ɔʏ",
     # ch is [x] after back vowels (Bach-laut), [ç] elsewhere
     "auch": "aʊx", "ach": "ax", "och": "ɔx", "uch": "uːx",
-    "ie": "iː", "ch": "ç", "ck": "k", "sp": "ʃp", "st": "ʃt",
+    "ie": "iː", "ch": "ç", "ck": "k", "tz": "ts",
+    # st/sp are [ʃt]/[ʃp] only morpheme-initially; de_preprocess marks
+    # those positions with St/Sp — elsewhere they are plain [st]/[sp]
+    "St": "ʃt", "Sp": "ʃp", "sp": "sp", "st": "st",
+    "äh": "ɛː", "öh": "øː", "üh": "yː", "ah": "aː", "oh": "oː",
+    "ih": "iː", "uh": "uː",
     "th": "t", "ph": "f", "qu": "kv", "ß": "s",
     "ä": "ɛ", "ö": "ø", "ü": "y",
     "a": "a", "b": "b", "c": "k", "d": "d", "e": "ɛ", "f": "f",

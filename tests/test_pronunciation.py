@@ -632,3 +632,20 @@ def test_fr_quality_round2_final():
     assert g.word_to_ipa("vie") == "vˈi"
     assert g.word_to_ipa("musée") == "myzˈe"
     assert g.word_to_ipa("avion") == "avjˈɔ̃"
+
+
+def test_de_st_sp_context():
+    """st/sp are [ʃt]/[ʃp] only morpheme-initially (word start or
+    after an unstressed prefix); plain [st]/[sp] elsewhere."""
+    g = _get_g2p("de")
+    assert g.word_to_ipa("stehen") == "ʃtˈeːən"
+    assert g.word_to_ipa("verstehen") == "fɛʁʃtˈeːən"
+    assert g.word_to_ipa("lustig") == "lˈʊstɪç"
+    assert g.word_to_ipa("dienstag") == "dˈiːnstak"
+    assert g.word_to_ipa("fenster") == "fˈɛnstɐ"
+    assert g.word_to_ipa("besten") == "bˈɛstən"    # superlative, lexicon
+    assert g.word_to_ipa("bestehen") == "bɛʃtˈeːən"  # be+stehen
+    # vowel-length h, tz, final -es
+    assert g.word_to_ipa("ruhig") == "ʁˈuːɪç"
+    assert g.word_to_ipa("trotz") == "tʁˈɔts"
+    assert g.word_to_ipa("dieses") == "dˈiːzəs"
