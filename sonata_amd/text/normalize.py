@@ -598,6 +598,38 @@ _TIME_WORDS = {
 }
 
 
+# German ordinals for date-style "3. Mai" (digit + period directly
+# before a capitalized word — German nouns are capitalized, so this is
+# a reliable ordinal signal; sentence-final "3." stays a cardinal)
+_DE_ORD_STEMS = {1: "ers", 3: "drit", 7: "sieb", 8: "ach"}
+_DE_ORD_RE = re.compile(
+    r"\b(\d{1,2})\.(?=\s+(?:Januar|Februar|März|April|Mai|Juni|Juli|"
+    r"August|September|Oktober|November|Dezember|Montag|Dienstag|"
+    r"Mittwoch|Donnerstag|Freitag|Samstag|Sonntag)\b)")
+_DE_DATIVE_RE = re.compile(r"\b(am|vom|zum|dem|den)$", re.IGNORECASE)
+
+
+def _de_ordinal(n: int, dative: bool) -> str:
+    from .numbers2 import _ru_plural  # noqa: F401 (module load order)
+    if n in _DE_ORD_STEMS:
+        stem = _DE_ORD_STEMS[n] + "t"
+    elif n < 20:
+        stem = num_to_words_de(n) + "t"
+    else:
+        stem = num_to_words_de(n) + "st"
+    return stem + ("en" if dative else "e")
+
+
+def _de_expand_ordinals(text: str) -> str:
+    def sub(m: re.Match) -> str:
+        prefix = text[:m.start()].rstrip()
+        prev = prefix.split()[-1] if prefix.split() else ""
+        dative = bool(_DE_DATIVE_RE.search(prev))
+        return _de_ordinal(int(m.group(1)), dative)
+
+    return _DE_ORD_RE.sub(sub, text)
+
+
 # Germanic year-style reading of bare 1100-1999 ("neunzehnhundert...",
 # also idiomatic for counts: "fünfzehnhundert Meter")
 _TEEN_HUNDRED_RE = re.compile(r"\b(1[1-9])(\d\d)\b")
@@ -663,6 +695,8 @@ def normalize(text: str, language: str) -> str:
             text = _DEC_COMMA_RE.sub(
                 lambda m: card(int(m.group(1))) + " " + dec + " "
                 + " ".join(digits[int(d)] for d in m.group(2)), text)
+        if base == "de":
+            text = _de_expand_ordinals(text)
         if base in ("de", "nl", "sv"):
             text = _teen_hundreds(base, card, text)
         return _INT_RE.sub(

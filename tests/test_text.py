@@ -386,3 +386,15 @@ def test_teen_hundreds_germanic():
     assert normalize("År 1950.", "sv") == "År nittonhundrafemtio."
     # 2000s stay plain cardinals
     assert normalize("2024", "de") == "zweitausendvierundzwanzig"
+
+
+def test_de_ordinal_dates():
+    from sonata_amd.text.normalize import normalize
+
+    assert normalize("Am 3. Mai beginnt es.", "de") == \
+        "Am dritten Mai beginnt es."
+    assert normalize("Der 1. Januar.", "de") == "Der erste Januar."
+    assert normalize("Der 21. Dezember.", "de") == \
+        "Der einundzwanzigste Dezember."
+    # non-date "N." stays a cardinal
+    assert normalize("Kapitel 7. Ende.", "de") == "Kapitel sieben. Ende."
