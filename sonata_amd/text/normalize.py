@@ -697,6 +697,21 @@ def normalize(text: str, language: str) -> str:
                 + " ".join(digits[int(d)] for d in m.group(2)), text)
         if base == "de":
             text = _de_expand_ordinals(text)
+        elif base == "fr":
+            # the only written French ordinal digits: 1er/1re/1ère
+            text = re.sub(r"\b1er\b", "premier", text)
+            text = re.sub(r"\b1(?:ère|re)\b", "première", text)
+            def _fr_ord(m: re.Match) -> str:
+                c = _CARDINALS["fr"](int(m.group(1)))
+                if c.endswith("e"):
+                    c = c[:-1]           # quatre -> quatrième
+                elif c.endswith("cinq"):
+                    c += "u"             # cinquième
+                elif c.endswith("neuf"):
+                    c = c[:-1] + "v"     # neuvième
+                return c + "ième"
+
+            text = re.sub(r"\b(\d{1,2})e\b", _fr_ord, text)
         if base in ("de", "nl", "sv"):
             text = _teen_hundreds(base, card, text)
         return _INT_RE.sub(

@@ -398,3 +398,12 @@ def test_de_ordinal_dates():
         "Der einundzwanzigste Dezember."
     # non-date "N." stays a cardinal
     assert normalize("Kapitel 7. Ende.", "de") == "Kapitel sieben. Ende."
+
+
+def test_fr_ordinals():
+    from sonata_amd.text.normalize import normalize
+
+    assert normalize("Le 1er mai.", "fr") == "Le premier mai."
+    assert normalize("le 4e jour", "fr") == "le quatrième jour"
+    assert normalize("le 5e", "fr") == "le cinquième"
+    assert normalize("le 9e", "fr") == "le neuvième"
