@@ -252,7 +252,8 @@ class RuleG2P:
                 break
         else:
             if self.stress_default == "es-penult":
-                if word[-1] in "aeiouns" and len(starts) >= 2:
+                # -m behaves like -n (Portuguese imagem, ordem)
+                if word[-1] in "aeiounsm" and len(starts) >= 2:
                     idx = len(starts) - 2
                 else:
                     idx = len(starts) - 1
@@ -512,6 +513,9 @@ _ES_RULES = {
     "güe": "ɡwe", "güi": "ɡwi", "gue": "ɡe", "gui": "ɡi",
     "ge": "xe", "gi": "xi",
     "ñ": "ɲ", "j": "x", "v": "b", "z": "θ", "ce": "θe", "ci": "θi",
+    "ción": "θjˈon", "sión": "sjˈon",
+    # glide + stressed-vowel digraphs (también, después, acción)
+    "ié": "jˈe", "ió": "jˈo", "iá": "jˈa", "ué": "wˈe", "uá": "wˈa",
     "á": "ˈa", "é": "ˈe", "í": "ˈi", "ó": "ˈo", "ú": "ˈu", "ü": "w",
     "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
     "g": "ɡ", "h": "", "i": "i", "k": "k", "l": "l", "m": "m",
@@ -581,6 +585,8 @@ _IT_RULES = {
     "cia": "tʃa", "cio": "tʃo", "ciu": "tʃu", "ci": "tʃi", "ce": "tʃe",
     "gia": "dʒa", "gio": "dʒo", "giu": "dʒu", "gi": "dʒi", "ge": "dʒe",
     "zz": "tts", "qu": "kw",
+    "zione": "tsjˈone", "zioni": "tsjˈoni",
+    "nza": "ntsa", "nze": "ntse", "nzo": "ntso", "nzi": "ntsi",
     # accented vowels mark stress (word-final in native orthography)
     "à": "ˈa", "è": "ˈɛ", "é": "ˈe", "ì": "ˈi", "ò": "ˈɔ", "ù": "ˈu",
     "a": "a", "b": "b", "c": "k", "d": "d", "e": "e", "f": "f",
@@ -803,10 +809,30 @@ def _get_g2p(voice: str) -> RuleG2P:
                     stress_default="final")
         g._suffix_stress = []  # French stress is uniformly final
     elif base == "it":
-        g = RuleG2P(_IT_RULES, letters="a-zA-Zàèéìòù",
+        # sdrucciole (antepenult-stress) words are lexical in Italian —
+        # pin the frequent ones
+        _IT_LEX = {
+            "facile": "fˈatʃile", "difficile": "diffˈitʃile",
+            "utile": "ˈutile", "possibile": "possˈibile",
+            "albero": "ˈalbero", "giovane": "dʒˈovane",
+            "giovani": "dʒˈovani", "immagine": "immˈadʒine",
+            "origine": "orˈidʒine", "ordine": "ˈordine",
+            "zucchero": "dzˈukkero", "numero": "nˈumero",
+            "camera": "kˈamera", "musica": "mˈuzika",
+            "medico": "mˈɛdiko", "popolo": "pˈɔpolo",
+            "periodo": "perˈiodo", "telefono": "telˈɛfono",
+            "essere": "ˈɛssere", "credere": "krˈedere",
+            "prendere": "prˈɛndere", "vivere": "vˈivere",
+            "scrivere": "skrˈivere", "leggere": "lˈɛddʒere",
+            "vendere": "vˈendere", "perdere": "pˈɛrdere",
+        }
+        g = RuleG2P(_IT_RULES, _IT_LEX, letters="a-zA-Zàèéìòù",
                     stress_default="es-penult")
     elif base == "pt":
         def _pt_post(ipa: str) -> str:
+            # word-final -em is the nasal diphthong (imagem -> ẽi)
+            if ipa.endswith("em"):
+                ipa = ipa[:-2] + "ẽi"
             # BR final-vowel reduction: -o -> u, unstressed final -e -> i
             if ipa.endswith("o") and "ˈo" != ipa[-2:]:
                 ipa = ipa[:-1] + "u"

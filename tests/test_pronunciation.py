@@ -649,3 +649,19 @@ def test_de_st_sp_context():
     assert g.word_to_ipa("ruhig") == "ʁˈuːɪç"
     assert g.word_to_ipa("trotz") == "tʁˈɔts"
     assert g.word_to_ipa("dieses") == "dˈiːzəs"
+
+
+def test_es_it_pt_quality_final_batch():
+    """Spanish glide+accent digraphs, Italian -zione/-nza + sdrucciole
+    lexicon, Portuguese -em nasal diphthong with -m penult stress."""
+    es = _get_g2p("es")
+    assert es.word_to_ipa("información") == "infoɾmaθjˈon"
+    assert es.word_to_ipa("también") == "tambjˈen"
+    assert es.word_to_ipa("después") == "despwˈes"
+    it = _get_g2p("it")
+    assert it.word_to_ipa("informazione") == "informatsjˈone"
+    assert it.word_to_ipa("facile") == "fˈatʃile"
+    assert it.word_to_ipa("scienza") == "ʃˈentsa"
+    pt = _get_g2p("pt")
+    assert pt.word_to_ipa("imagem") == "imˈaʒẽi"
+    assert pt.word_to_ipa("jovem") == "ʒˈovẽi"
