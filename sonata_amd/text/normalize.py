@@ -573,10 +573,16 @@ _CURRENCY_WORDS = {
     "no": {"€": "euro", "$": "dollar", "kr": "kroner"},
     "da": {"€": "euro", "$": "dollar", "kr": "kroner"},
     "fi": {"€": "euroa", "$": "dollaria"},
-    "pl": {"€": "euro", "$": "dolarów", "zł": "złotych"},
-    "cs": {"€": "eur", "$": "dolarů", "Kč": "korun"},
-    "ru": {"€": "евро", "$": "долларов", "₽": "рублей"},
-    "uk": {"€": "євро", "$": "доларів", "₴": "гривень"},
+    # Slavic currencies agree in case with the number: (one, few, many)
+    "pl": {"€": "euro", "$": ("dolar", "dolary", "dolarów"),
+           "zł": ("złoty", "złote", "złotych")},
+    "cs": {"€": ("euro", "eura", "eur"),
+           "$": ("dolar", "dolary", "dolarů"),
+           "Kč": ("koruna", "koruny", "korun")},
+    "ru": {"€": "евро", "$": ("доллар", "доллара", "долларов"),
+           "₽": ("рубль", "рубля", "рублей")},
+    "uk": {"€": "євро", "$": ("долар", "долари", "доларів"),
+           "₴": ("гривня", "гривні", "гривень")},
     "tr": {"€": "avro", "$": "dolar", "₺": "lira"},
     "el": {"€": "ευρώ", "$": "δολάρια"},
     "ro": {"€": "euro", "$": "dolari"},
@@ -662,11 +668,26 @@ def normalize(text: str, language: str) -> str:
     if cw is not None:
         # €5 / 5€ / $5 / £5 -> "5 <unit>" before number expansion
         num = r"(\d[\d.,]*\d|\d)"  # cannot end in punctuation
+
+        def _unit(word, numstr: str) -> str:
+            if isinstance(word, tuple):  # Slavic case agreement
+                from .numbers2 import _ru_plural
+                try:
+                    n = int(re.sub(r"[.,]", "", numstr))
+                except ValueError:
+                    n = 5
+                return _ru_plural(n, *word)
+            return word
+
         for sym, word in cw.items():
             text = re.sub(
-                rf"{re.escape(sym)}\s?{num}", rf"\1 {word}", text)
+                rf"{re.escape(sym)}\s?{num}",
+                lambda m, _w=word: m.group(1) + " " + _unit(_w, m.group(1)),
+                text)
             text = re.sub(
-                rf"{num}\s?{re.escape(sym)}", rf"\1 {word}", text)
+                rf"{num}\s?{re.escape(sym)}",
+                lambda m, _w=word: m.group(1) + " " + _unit(_w, m.group(1)),
+                text)
     tw = _TIME_WORDS.get(base)
     if tw is not None:
         # 14:30 -> "14 <hour-word> 30" (espeak-style clock reading)

@@ -407,3 +407,13 @@ def test_fr_ordinals():
     assert normalize("le 4e jour", "fr") == "le quatrième jour"
     assert normalize("le 5e", "fr") == "le cinquième"
     assert normalize("le 9e", "fr") == "le neuvième"
+
+
+def test_slavic_currency_agreement():
+    from sonata_amd.text.normalize import normalize
+
+    assert normalize("1$", "ru") == "один доллар"
+    assert normalize("2$", "ru") == "два доллара"
+    assert normalize("5$", "ru") == "пять долларов"
+    assert normalize("21₽", "ru") == "двадцать один рубль"
+    assert normalize("5 zł", "pl") == "pięć złotych"
