@@ -633,6 +633,7 @@ _RU_RULES = {
 # Dutch
 # --------------------------------------------------------------------------- #
 _NL_RULES = {
+    "lijk": "lək",  # -lijk suffix is schwa (natuurlijk, eigenlijk)
     "sch": "sx", "ij": "ɛi", "ei": "ɛi", "ui": "œy",
     "ouw": "ʌu", "auw": "ʌu", "ieuw": "iu", "eeuw": "eːu",
     "ou": "ʌu", "au": "ʌu", "oe": "u", "eu": "ø", "ie": "i",
@@ -690,6 +691,7 @@ _PL_RULES = {
 # Turkish (highly regular orthography)
 # --------------------------------------------------------------------------- #
 _TR_RULES = {
+    "â": "aː", "î": "iː", "û": "uː",
     "ç": "tʃ", "ş": "ʃ", "ğ": "ː", "ı": "ɯ", "ö": "ø", "ü": "y",
     "c": "dʒ", "j": "ʒ", "y": "j", "v": "v",
     "a": "a", "b": "b", "d": "d", "e": "e", "f": "f", "g": "ɡ",
@@ -879,7 +881,16 @@ def _get_g2p(voice: str) -> RuleG2P:
 
         g._stress_rules_output = _ru_stress
     elif base == "nl":
-        g = RuleG2P(_NL_RULES, letters="a-zA-Z")
+        def _nl_pre(w: str) -> str:
+            # doubled consonants mark short vowels, not geminates
+            out = []
+            for ch in w:
+                if out and out[-1] == ch and ch not in "aeiou":
+                    continue
+                out.append(ch)
+            return "".join(out)
+
+        g = RuleG2P(_NL_RULES, letters="a-zA-Z", preprocess=_nl_pre)
         g.stress_skip_prefixes = ("ge", "be", "ver", "ont", "her")
 
         _NL_DEVOICE = {"b": "p", "d": "t", "ɣ": "x", "v": "f",
@@ -904,8 +915,11 @@ def _get_g2p(voice: str) -> RuleG2P:
     elif base == "pl":
         g = RuleG2P(_PL_RULES, letters="a-zA-Ząćęłńóśźż",
                     stress_default="penult")
+        # word-final ę denasalizes in standard speech (dziękuję)
+        g.postprocess = (lambda ipa: ipa[:-2] + "ɛ"
+                         if ipa.endswith("ɛ̃") else ipa)
     elif base == "tr":
-        g = RuleG2P(_TR_RULES, letters="a-zA-Zçğıöşü",
+        g = RuleG2P(_TR_RULES, letters="a-zA-Zçğıöşüâîû",
                     stress_default="final")
     elif base == "cs":
         g = RuleG2P(_CS_RULES, letters="a-zA-Začďéěíňóřšťúůýž")

@@ -482,7 +482,7 @@ def test_pl_softening_and_stress():
 
     g = _get_g2p("pl")
     cases = [
-        ("dziękuję", "dʑɛ̃kˈujɛ̃"), ("ciebie", "tɕˈɛbjɛ"),
+        ("dziękuję", "dʑɛ̃kˈujɛ"), ("ciebie", "tɕˈɛbjɛ"),
         ("siedem", "ɕˈɛdɛm"), ("zielony", "ʑɛlˈɔnɨ"),
         ("nie", "ɲˈɛ"), ("kobieta", "kɔbjˈɛta"),
         ("wiem", "vjˈɛm"), ("warszawa", "varʂˈava"),
@@ -560,7 +560,7 @@ def test_nl_quality_layer():
     cases = [
         ("goed", "ɣˈut"), ("vrouw", "vrˈʌu"), ("nieuw", "nˈiu"),
         ("meisje", "mˈɛisjə"), ("spreken", "sprˈɛkən"),
-        ("gezellig", "ɣɛzˈɛlləx"), ("hond", "ɦˈɔnt"),
+        ("gezellig", "ɣɛzˈɛləx"), ("hond", "ɦˈɔnt"),
         ("dag", "dˈɑx"),
     ]
     wrong = [(w, g.word_to_ipa(w), want) for w, want in cases
@@ -665,3 +665,12 @@ def test_es_it_pt_quality_final_batch():
     pt = _get_g2p("pt")
     assert pt.word_to_ipa("imagem") == "imˈaʒẽi"
     assert pt.word_to_ipa("jovem") == "ʒˈovẽi"
+
+
+def test_nl_pl_tr_final_batch():
+    """-lijk schwa + double collapse (nl), final-ę denasalization (pl),
+    circumflex vowels (tr)."""
+    assert _get_g2p("nl").word_to_ipa("natuurlijk") == "nˈɑtyrlək"
+    assert _get_g2p("nl").word_to_ipa("eigenlijk") == "ˈɛiɣɛnlək"
+    assert _get_g2p("pl").word_to_ipa("dziękuję") == "dʑɛ̃kˈujɛ"
+    assert _get_g2p("tr").word_to_ipa("imkân") == "imkˈaːn"
