@@ -988,6 +988,23 @@ def _get_g2p(voice: str) -> RuleG2P:
         if key2 in TABLES:
             g = RuleG2P(TABLES[key2], LEXICONS.get(key2),
                         letters=LETTERS[key2])
+            if key2 == "sv":
+                # Swedish vowel quantity is positional: long in open
+                # syllables (the table's default), short before a
+                # geminate/cluster; doubled consonants collapse;
+                # final unstressed -e is schwa
+                _sv_c = r"[^aeiouyøɛæɑɔʉœˈː]"
+
+                def _sv_post(ipa: str) -> str:
+                    ipa = re.sub(rf"ː({_sv_c})\1", r"\1", ipa)
+                    ipa = re.sub(rf"ː(?={_sv_c}{_sv_c})", "", ipa)
+                    ipa = re.sub(r"ː(?=ŋ)", "", ipa)
+                    ipa = re.sub(rf"({_sv_c})\1", r"\1", ipa)
+                    if ipa.endswith("eː") and not ipa.endswith("ˈeː"):
+                        ipa = ipa[:-2] + "ə"
+                    return ipa
+
+                g.postprocess = _sv_post
         elif key2 in TABLES2:
             g = RuleG2P(TABLES2[key2], LEXICONS.get(key2),
                         letters=LETTERS2[key2],

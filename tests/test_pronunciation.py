@@ -674,3 +674,13 @@ def test_nl_pl_tr_final_batch():
     assert _get_g2p("nl").word_to_ipa("eigenlijk") == "ˈɛiɣɛnlək"
     assert _get_g2p("pl").word_to_ipa("dziękuję") == "dʑɛ̃kˈujɛ"
     assert _get_g2p("tr").word_to_ipa("imkân") == "imkˈaːn"
+
+
+def test_sv_vowel_quantity():
+    """Swedish: long vowels in open syllables, short before clusters/
+    geminates; final unstressed -e is schwa."""
+    sv = _get_g2p("sv")
+    assert sv.word_to_ipa("hela") == "hˈeːla"       # open: long
+    assert sv.word_to_ipa("tack") == "tˈak"
+    assert sv.word_to_ipa("samhälle") == "sˈamhɛlə"
+    assert sv.word_to_ipa("utveckling") == "ˈʉtvekliŋ"
