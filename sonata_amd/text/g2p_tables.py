@@ -946,5 +946,26 @@ LEXICONS = {
         "тебе": "tɛbˈɛ", "себе": "sɛbˈɛ", "язик": "jɑzˈɪk",
         "завжди": "zˈɑʋʐdɪ", "тепер": "tɛpˈɛr", "тому": "tɔmˈu",
         "вода": "ʋɔdˈɑ", "земля": "zɛmlʲˈɑ", "зараз": "zˈɑrɑz",
+        # r2 final batch: non-initial stress the default misses
+        "україна": "ukrɑjˈinɑ", "розвиток": "rɔzʋˈɪtɔk",
+        "суспільство": "suspˈilʲstʋɔ", "можливість": "mɔʒlˈɪʋistʲ",
+        "здоров'я": "zdɔrˈɔʋjɑ", "звичайно": "zʋɪtʃˈɑjnɔ",
+        "важливий": "ʋɑʒlˈɪʋɪj", "писати": "pɪsˈɑtɪ",
+        "субота": "subˈɔtɑ", "університет": "unʲiʋɛrsɪtˈɛt",
+        "людина": "lʲudˈɪnɑ", "робота": "rɔbˈɔtɑ",
+        "питання": "pɪtˈɑnʲːɑ", "сьогодні": "sʲɔɦˈɔdnʲi",
+        "гарний": "ɦˈɑrnɪj", "спасибі": "spɑsˈɪbʲi",
+        "народ": "nɑrˈɔd", "країна": "krɑjˈinɑ",
+    },
+    "bg": {
+        # Bulgarian stress is lexical; pin frequent non-initial cases
+        "развитие": "razvˈitiɛ", "общество": "ɔbʃtˈɛstvɔ",
+        "възможност": "vɤzmˈɔʒnɔst", "различни": "razlˈitʃni",
+        "благодаря": "blaɡɔdarjˈa", "добре": "dɔbrˈɛ",
+        "човек": "tʃɔvˈɛk", "жена": "ʒɛnˈa", "вода": "vɔdˈa",
+        "глава": "ɡlavˈa", "ръка": "rɤkˈa", "земя": "zɛmjˈa",
+        "народ": "narˈɔd", "въпрос": "vɤprˈɔs", "отговор": "ˈɔtɡɔvɔr",
+        "работа": "rˈabɔta", "година": "ɡɔdˈina", "деня": "dɛnjˈa",
+        "страна": "stranˈa", "езика": "ɛzˈika", "език": "ɛzˈik",
     },
 }

@@ -257,6 +257,13 @@ class RuleG2P:
                     idx = len(starts) - 2
                 else:
                     idx = len(starts) - 1
+            elif self.stress_default == "ro-penult":
+                # Romanian: penult when vowel-final (incl. ă/â/î),
+                # final syllable otherwise (guvern, mulțumesc)
+                if word[-1] in "aeiouăâî" and len(starts) >= 2:
+                    idx = len(starts) - 2
+                else:
+                    idx = len(starts) - 1
             elif self.stress_default == "penult":
                 idx = max(len(starts) - 2, 0)
             elif self.stress_default == "antepenult":
@@ -986,8 +993,14 @@ def _get_g2p(voice: str) -> RuleG2P:
 
         key2 = ALIASES.get(base, base)
         if key2 in TABLES:
+            # Romanian stress behaves like the Spanish rule (penult
+            # when vowel-final, final otherwise); the other TABLES
+            # languages keep the first-syllable default (+ lexicons
+            # for the stress-lexical ones: uk, bg)
             g = RuleG2P(TABLES[key2], LEXICONS.get(key2),
-                        letters=LETTERS[key2])
+                        letters=LETTERS[key2],
+                        stress_default=("ro-penult" if key2 == "ro"
+                                        else "first"))
             if key2 == "sv":
                 # Swedish vowel quantity is positional: long in open
                 # syllables (the table's default), short before a

@@ -470,7 +470,7 @@ def test_uk_be_palatalization():
     uk = _get_g2p("uk")
     assert uk.word_to_ipa("дякую") == "dʲˈɑkuju"
     assert uk.word_to_ipa("привіт") == "prɪʋʲˈit"   # lexicon stress
-    assert uk.word_to_ipa("сьогодні") == "sʲˈɔɦɔdnʲi"
+    assert uk.word_to_ipa("сьогодні") == "sʲɔɦˈɔdnʲi"  # lexicon
     be = _get_g2p("be")
     assert be.word_to_ipa("дзякуй") == "dzʲˈakuj"
     assert be.word_to_ipa("дзень") == "dzʲˈɛnʲ"
@@ -684,3 +684,17 @@ def test_sv_vowel_quantity():
     assert sv.word_to_ipa("tack") == "tˈak"
     assert sv.word_to_ipa("samhälle") == "sˈamhɛlə"
     assert sv.word_to_ipa("utveckling") == "ˈʉtvekliŋ"
+
+
+def test_uk_bg_ro_stress_batch():
+    """uk/bg stress lexicons + Romanian vowel-final penult rule."""
+    uk = _get_g2p("uk")
+    assert uk.word_to_ipa("україна") == "ukrɑjˈinɑ"
+    assert uk.word_to_ipa("розвиток") == "rɔzʋˈɪtɔk"
+    bg = _get_g2p("bg")
+    assert bg.word_to_ipa("благодаря") == "blaɡɔdarjˈa"
+    assert bg.word_to_ipa("човек") == "tʃɔvˈɛk"
+    ro = _get_g2p("ro")
+    assert ro.word_to_ipa("guvern") == "ɡuvˈern"
+    assert ro.word_to_ipa("bună") == "bˈunə"
+    assert ro.word_to_ipa("societate") == "sotʃietˈate"
