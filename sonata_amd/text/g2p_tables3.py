@@ -48,7 +48,8 @@ def _tokenize_ipa(ipa: str):
     return toks
 
 
-def epenthesize(ipa: str, vowel: str) -> str:
+def epenthesize(ipa: str, vowel: str,
+                keep_final_cluster: bool = True) -> str:
     """Insert an epenthetic vowel between consecutive consonants.
 
     The word-final CC cluster is kept (Persian allows final clusters:
@@ -69,7 +70,8 @@ def epenthesize(ipa: str, vowel: str) -> str:
         nxt_c = nxt[0] not in _VOWEL_START
         if cur_c and nxt_c:
             final_pair = (i + 2 == n)
-            if not final_pair or n <= 2 or n_cons >= 4:
+            if (not final_pair or n <= 2 or n_cons >= 4
+                    or not keep_final_cluster):
                 out.append(vowel)
     return "".join(out)
 
@@ -107,8 +109,10 @@ FA_LEXICON = {
 
 
 def fa_postprocess(ipa: str) -> str:
-    # word-final ه is the vowel /e/ (خانه), not [h]
-    if ipa.endswith("h") and len(ipa) > 2:
+    # word-final ه after a CONSONANT is the vowel /e/ (خانه xɒːne);
+    # after a vowel it stays [h] (گاه ɡɒːh)
+    if (ipa.endswith("h") and len(ipa) > 2
+            and ipa[-2] not in _VOWEL_START and ipa[-2] != "ː"):
         ipa = ipa[:-1] + "e"
     return epenthesize(ipa, "æ")
 
@@ -190,21 +194,22 @@ HE_LEXICON = {
 
 
 def he_postprocess(ipa: str) -> str:
-    # glottal carriers vanish except word-initially before a vowel
-    ipa = epenthesize(ipa, "a")
-    ipa = ipa.replace("ʔ", "")
-    # word-final ה is silent (written vowel marker)
+    # word-final ה usually writes the vowel /a/ (מורה mora,
+    # משפחה mishpaxa) — convert BEFORE epenthesis so the final
+    # syllable is open; Hebrew mostly lacks final clusters, so
+    # epenthesis applies to the final pair too (sefer, not *sefr)
     if ipa.endswith("h") and len(ipa) > 2:
-        ipa = ipa[:-1]
-    return ipa
+        ipa = ipa[:-1] + "a"
+    ipa = epenthesize(ipa, "a", keep_final_cluster=False)
+    return ipa.replace("ʔ", "")
 
 
 def he_preprocess(w: str) -> str:
-    # medial י = /i/, medial ו = /o/ when flanked by consonants —
-    # approximate matres lectionis before the consonant rules run
+    # medial matres lectionis: ו = /o/, י = /i/ when flanked by
+    # consonants — approximated before the consonant rules run
     if len(w) > 2:
         core = w[1:-1]
-        core = core.replace("ו", "ֺ")  # private: map to vowel o
+        core = core.replace("ו", "ֺ").replace("י", "ִ")
         w = w[0] + core + w[-1]
     return w.replace("ֺ", "ֹ")
 

@@ -511,3 +511,16 @@ def test_indic_symbols_encodable():
                 if ch in ".,;:?!":
                     continue
                 assert ch in id_map, (lang, ch, hex(ord(ch)), sent)
+
+
+def test_fa_he_refinements():
+    """fa: final ه keeps /h/ after vowels (ɡɒːh) but is /e/ after
+    consonants (xɒːne); he: final ה = /a/, medial yod = /i/, no final
+    clusters (sefer not *sefr)."""
+    fa = _get_g2p("fa")
+    assert fa.word_to_ipa("خانه") == "xɒːnˈe"
+    assert fa.word_to_ipa("راه") == "rˈɒːh"
+    he = _get_g2p("he")
+    assert he.word_to_ipa("מורה") == "moʁˈa"
+    assert he.word_to_ipa("ספר") == "safˈaʁ"
+    assert he.word_to_ipa("מדינה") == "madinˈa"
