@@ -353,6 +353,55 @@ JA_KANJI = {
     "質問": "しつもん", "問題": "もんだい", "答え": "こたえ",
     "意味": "いみ", "説明": "せつめい", "最初": "さいしょ",
     "最後": "さいご", "次": "つぎ", "他": "ほか", "別": "べつ",
+    # second batch: common nouns/verbs with unambiguous readings
+    "天気": "てんき", "空気": "くうき", "気持ち": "きもち",
+    "気分": "きぶん", "病気": "びょうき", "病院": "びょういん",
+    "医者": "いしゃ", "薬": "くすり", "体操": "たいそう",
+    "運動": "うんどう", "散歩": "さんぽ", "買い物": "かいもの",
+    "荷物": "にもつ", "切符": "きっぷ", "地下鉄": "ちかてつ",
+    "飛行機": "ひこうき", "自転車": "じてんしゃ",
+    "自動車": "じどうしゃ", "新聞": "しんぶん", "雑誌": "ざっし",
+    "辞書": "じしょ", "図書館": "としょかん", "銀行": "ぎんこう",
+    "郵便局": "ゆうびんきょく", "公園": "こうえん",
+    "動物": "どうぶつ", "動物園": "どうぶつえん", "植物": "しょくぶつ",
+    "野菜": "やさい", "果物": "くだもの", "食事": "しょくじ",
+    "朝食": "ちょうしょく", "昼食": "ちゅうしょく",
+    "夕食": "ゆうしょく", "弁当": "べんとう", "牛乳": "ぎゅうにゅう",
+    "紅茶": "こうちゃ", "砂糖": "さとう", "塩": "しお",
+    "結婚": "けっこん", "約束": "やくそく", "予定": "よてい",
+    "準備": "じゅんび", "練習": "れんしゅう", "試験": "しけん",
+    "宿題": "しゅくだい", "授業": "じゅぎょう", "教室": "きょうしつ",
+    "教師": "きょうし", "生徒": "せいと", "留学生": "りゅうがくせい",
+    "外国": "がいこく", "外国人": "がいこくじん",
+    "旅館": "りょかん", "部長": "ぶちょう", "社長": "しゃちょう",
+    "会議": "かいぎ", "電気": "でんき", "冷蔵庫": "れいぞうこ",
+    "洗濯": "せんたく", "掃除": "そうじ", "台所": "だいどころ",
+    "風呂": "ふろ", "玄関": "げんかん", "庭": "にわ",
+    "建物": "たてもの", "住所": "じゅうしょ", "地図": "ちず",
+    "世紀": "せいき", "文化": "ぶんか", "文学": "ぶんがく",
+    "歴史": "れきし", "経済": "けいざい", "政治": "せいじ",
+    "社会": "しゃかい", "科学": "かがく", "数学": "すうがく",
+    "英語": "えいご", "中国": "ちゅうごく", "中国語": "ちゅうごくご",
+    "韓国": "かんこく", "韓国語": "かんこくご",
+    "銀": "ぎん", "鉄": "てつ", "紙": "かみ", "服": "ふく",
+    "靴": "くつ", "帽子": "ぼうし", "眼鏡": "めがね",
+    "時計": "とけい", "財布": "さいふ", "鍵": "かぎ",
+    "窓口": "まどぐち", "入口": "いりぐち", "出口": "でぐち",
+    "右": "みぎ", "左": "ひだり", "北": "きた", "南": "みなみ",
+    "東": "ひがし", "西": "にし", "近く": "ちかく", "遠く": "とおく",
+    "隣": "となり", "横": "よこ", "角": "かど", "橋": "はし",
+    "地震": "じしん", "台風": "たいふう", "火事": "かじ",
+    "事故": "じこ", "警察": "けいさつ", "消防": "しょうぼう",
+    "危険": "きけん", "安全": "あんぜん", "注意": "ちゅうい",
+    "質": "しつ", "量": "りょう", "形": "かたち", "色": "いろ",
+    "赤い": "あかい", "青い": "あおい", "白い": "しろい",
+    "黒い": "くろい", "明るい": "あかるい", "暗い": "くらい",
+    "重い": "おもい", "軽い": "かるい", "広い": "ひろい",
+    "狭い": "せまい", "深い": "ふかい", "浅い": "あさい",
+    "近い": "ちかい", "遠い": "とおい", "速い": "はやい",
+    "痛い": "いたい", "甘い": "あまい", "辛い": "からい",
+    "冷たい": "つめたい", "温かい": "あたたかい",
+    "涼しい": "すずしい", "暖かい": "あたたかい",
 }
 _JA_KANJI_MAX = max(len(k) for k in JA_KANJI)
 
