@@ -65,6 +65,10 @@ def ko_word_to_ipa(w: str) -> str:
             if (s > 0 and onset in _KO_VOICED
                     and _KO_FINALS[prev_fin] in ("", "l", "n", "m", "ŋ")):
                 onset = _KO_VOICED[onset]
+        medial = _KO_MEDIALS[med]
+        # ㅅ/ㅆ palatalize before i/j (시 = ʃi, 쉬 = ʃwi)
+        if onset == "s" and (medial.startswith(("i", "j", "wi"))):
+            onset = "ʃ"
         coda = _KO_FINALS[fin]
         if s + 1 < len(sylls):
             nxt_ini = sylls[s + 1][0]
@@ -72,7 +76,7 @@ def ko_word_to_ipa(w: str) -> str:
                 coda = ""  # moved to the next onset by liaison
             elif coda in _KO_NASALIZE and nxt_ini in (2, 6):  # ㄴ/ㅁ
                 coda = _KO_NASALIZE[coda]
-        out.append(onset + _KO_MEDIALS[med] + coda)
+        out.append(onset + medial + coda)
     return "".join(out)
 
 

@@ -141,6 +141,9 @@ def test_korean_decomposition_and_sandhi():
     assert ko.word_to_ipa("한국말") == "hanɡuŋmal"
     # coda neutralization: ㅅ final is [t]
     assert ko.word_to_ipa("옷") == "ot"
+    # ㅅ palatalizes before i/j
+    assert ko.word_to_ipa("시간") == "ʃiɡan"
+    assert ko.word_to_ipa("쉬다") == "ʃwida"
 
 
 def test_korean_medials():
