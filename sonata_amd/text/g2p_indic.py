@@ -192,8 +192,9 @@ class BrahmicG2P:
 # --------------------------------------------------------------------- #
 def _bn_engine(assamese: bool = False) -> BrahmicG2P:
     # Bengali: inherent /ɔ/; য = /dʒ/ (the য় nukta form stays /j/ via
-    # the shifted base table); anusvara ং is velar
-    over = {"য": "dʒ"}
+    # the shifted base table); no retroflex sibilant (ষ = /ʃ/);
+    # anusvara ং is velar
+    over = {"য": "dʒ", "ষ": "ʃ"}
     if assamese:
         over.update({"ৰ": "r", "ৱ": "w", "চ": "s", "ছ": "s"})
     return BrahmicG2P(0x80, "ɔ", final_del=True, cons_overrides=over,
@@ -282,8 +283,13 @@ def make_engine(lang: str) -> BrahmicG2P:
         return BrahmicG2P(0x380, "a", final_del=False,
                           final_anusvara="m")
     if lang == "ml":
-        return BrahmicG2P(0x400, "a", final_del=False,
-                          final_anusvara="m")
+        g = BrahmicG2P(0x400, "a", final_del=False,
+                       final_anusvara="m")
+        # chillu letters (U+0D7A-0D7F): consonants with NO inherent
+        # vowel in any position (സർവ = sarva)
+        g.extra_marks = {"ൺ": "ɳ", "ൻ": "n", "ർ": "r", "ൽ": "l",
+                         "ൾ": "ɭ", "ൿ": "k"}
+        return g
     raise KeyError(lang)
 
 

@@ -527,3 +527,14 @@ def test_fa_he_refinements():
     assert he.word_to_ipa("מורה") == "moʁˈa"
     assert he.word_to_ipa("ספר") == "safˈaʁ"
     assert he.word_to_ipa("מדינה") == "madinˈa"
+
+
+def test_ml_chillu_and_bn_sibilant():
+    """Malayalam chillu letters are vowel-less consonants in any
+    position; Bengali has no retroflex sibilant (ষ = ʃ)."""
+    ml = _get_g2p("ml")
+    assert ml.word_to_ipa("സർവകലാശാല") == "sarʋakalaːʃaːla"
+    assert ml.word_to_ipa("അവൾ") == "aʋaɭ"
+    bn = _get_g2p("bn")
+    assert bn.word_to_ipa("ভাষা") == "bʰaːʃaː"
+    assert bn.word_to_ipa("মানুষ") == "maːnuʃ"
