@@ -33,3 +33,13 @@ for wave in tts.synthesize("Hello again. How are you today?"):
 n = sum(len(c) for c in tts.synthesize_streamed(
     "Streaming synthesis delivers audio while the sentence still decodes."))
 print(f"streamed {n} PCM bytes")
+
+# multilingual: every supported language code works the same way
+# (full inventory in docs/LANGUAGES.md)
+for lang, text in [("hi", "नमस्ते दुनिया।"), ("ko", "안녕하세요 세계."),
+                   ("ja", "こんにちは、世界。")]:
+    p = create_random_voice(tempfile.mkdtemp(), f"ex_{lang}",
+                            quality="x_low", language=lang)
+    t = pysonata.Sonata.with_piper(pysonata.PiperModel(p))
+    print(lang, "->", pysonata.phonemize_text(text, language=lang))
+    t.synthesize_to_file(f"example_{lang}.wav", text)
