@@ -468,6 +468,8 @@ _EN_LEXICON = {
 
 # ordered longest-match English letter-to-sound rules
 _EN_RULES = {
+    "ation": "eɪʃən", "ution": "uʃən", "ition": "ɪʃən",
+    "otion": "oʊʃən",
     "tion": "ʃən", "sion": "ʒən", "ought": "ɔt", "ight": "aɪt",
     "tious": "ʃəs", "cious": "ʃəs", "ture": "tʃɚ", "sure": "ʒɚ",
     "augh": "ɔ", "ough": "ʌf", "eigh": "eɪ",
@@ -477,7 +479,8 @@ _EN_RULES = {
     "oi": "ɔɪ", "au": "ɔ", "aw": "ɔ", "ew": "u", "ue": "u",
     "ei": "eɪ", "ey": "eɪ", "ar": "ɑɹ", "er": "ɚ", "ir": "ɝ",
     "or": "ɔɹ", "ur": "ɝ", "ck": "k", "ch": "tʃ", "sh": "ʃ",
-    "th": "θ", "ph": "f", "wh": "w", "ng": "ŋ", "qu": "kw",
+    "th": "θ", "ph": "f", "wh": "w", "ng": "ŋ", "nk": "ŋk",
+    "qu": "kw",
     "gh": "ɡ", "kn": "n", "wr": "ɹ", "mb": "m", "ce": "s",
     "ci": "sɪ", "cy": "si", "ge": "dʒ", "gi": "dʒɪ", "gy": "dʒi",
     "a": "æ", "b": "b", "c": "k", "d": "d", "e": "ɛ", "f": "f",
