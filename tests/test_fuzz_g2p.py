@@ -57,3 +57,36 @@ def test_phonemizer_never_crashes(text, lang):
 def test_number_reading_never_crashes(n, lang):
     out = text_to_phonemes(f"x {n} y", lang)
     assert isinstance(out, list)
+
+
+def test_concurrent_phonemize_across_languages():
+    """The registry builds engines lazily; concurrent first-use from
+    many threads across many languages must be safe (espeak-ng, by
+    contrast, is a single global C state — SURVEY §5)."""
+    import threading
+
+    import sonata_amd.text.phonemizer as P
+
+    # force cold registry so threads race on construction
+    P._G2P_REGISTRY.clear()
+    langs = [l for l in _LANGS]
+    texts = ["Hello 42 world.", "नमस्ते 3 दुनिया।", "안녕 7!",
+             "مرحبا 9", "สวัสดี 5"]
+    errs = []
+
+    def worker(seed):
+        try:
+            for i in range(30):
+                lang = langs[(seed * 31 + i) % len(langs)]
+                for s in P.text_to_phonemes(texts[i % len(texts)], lang):
+                    phonemes_to_ids(s, _ID_MAP)
+        except Exception as e:  # pragma: no cover
+            errs.append((seed, repr(e)))
+
+    threads = [threading.Thread(target=worker, args=(k,))
+               for k in range(16)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=60)
+    assert not errs, errs
