@@ -112,6 +112,9 @@ class RuleG2P:
     # MUST stay off for other languages (French plural -s is silent —
     # "enfants" must not get an English /z/)
     english_inflections = False
+    # English-specific "4+ syllables -> stress the second" heuristic;
+    # fixed-initial-stress languages (hu/fi/cs/...) must NOT inherit it
+    long_word_second = False
 
     def word_to_ipa(self, word: str) -> str:
         w = word.lower()
@@ -270,7 +273,7 @@ class RuleG2P:
                 idx = max(len(starts) - 3, 0)
             elif self.stress_default == "final":
                 idx = len(starts) - 1
-            elif len(starts) >= 4:
+            elif self.long_word_second and len(starts) >= 4:
                 idx = 1
         pos = starts[min(idx, len(starts) - 1)]
         return ipa[:pos] + "ˈ" + ipa[pos:]
@@ -768,6 +771,7 @@ def _get_g2p(voice: str) -> RuleG2P:
                     unstressed=_EN_UNSTRESSED)
         g.spell_acronyms = True
         g.english_inflections = True
+        g.long_word_second = True
 
         def _en_fix(w: str, ipa: str) -> str:
             # orthography-aware ending fixes the context-free rules
@@ -1028,6 +1032,10 @@ def _get_g2p(voice: str) -> RuleG2P:
         else:
             raise PhonemizationError(
                 f"unsupported phonemizer language: {voice!r}")
+    if base != "en" and not hasattr(g, "_suffix_stress"):
+        # the orthographic suffix-stress table is English; languages
+        # without their own table (de sets one) must not inherit it
+        g._suffix_stress = []
     _G2P_REGISTRY[key] = g
     return g
 

@@ -698,3 +698,13 @@ def test_uk_bg_ro_stress_batch():
     assert ro.word_to_ipa("guvern") == "ɡuvˈern"
     assert ro.word_to_ipa("bună") == "bˈunə"
     assert ro.word_to_ipa("societate") == "sotʃietˈate"
+
+
+def test_fixed_initial_stress_not_poisoned_by_en_heuristics():
+    """The English long-word and suffix-stress heuristics must not
+    leak into fixed-initial-stress languages."""
+    assert _get_g2p("hu").word_to_ipa("magyarország").startswith("mˈ")
+    assert _get_g2p("fi").word_to_ipa("luonnollisesti").startswith("lˈ")
+    assert _get_g2p("cs").word_to_ipa("společnost").startswith("spˈ")
+    # English keeps both heuristics
+    assert _get_g2p("en").word_to_ipa("information") == "ɪnfɚmˈeɪʃən"
