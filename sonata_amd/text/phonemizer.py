@@ -540,6 +540,8 @@ _ES_RULES = {
 # Arabic (expects diacritized input; tashkeel pre-pass adds vowel marks)
 # --------------------------------------------------------------------------- #
 _AR_RULES = {
+    # haraka + mater lectionis = one long vowel (كِتَاب kitaːb)
+    "َا": "aː", "ِي": "iː", "ُو": "uː", "َى": "aː",
     # letters
     "ا": "aː", "ب": "b", "ت": "t", "ث": "θ", "ج": "dʒ", "ح": "ħ",
     "خ": "x", "د": "d", "ذ": "ð", "ر": "r", "ز": "z", "س": "s",
@@ -943,6 +945,17 @@ def _get_g2p(voice: str) -> RuleG2P:
             letters="؀-ۿ",
             stress=False,
         )
+
+        def _ar_post(ipa: str) -> str:
+            # definite article: hamzat-wasl is short (al-), and the lam
+            # assimilates into a following geminate sun letter
+            # (السَّلَام -> assalaːm)
+            if ipa.startswith("aːl"):
+                ipa = "al" + ipa[3:]
+            ipa = re.sub(r"^al(.)ː", r"a\1ː", ipa)
+            return ipa
+
+        g.postprocess = _ar_post
     elif base == "ja":
         # kana are an exact syllabary (gemination/long-vowel marks
         # handled); kanji needs a reading dictionary — dropped, and

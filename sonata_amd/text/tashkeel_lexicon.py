@@ -106,6 +106,10 @@ _WORDS: List[str] = [
     "فَاصِلَة", "أَحَدَ", "اِثْنَا",
     "ثَلَاثُمِائَة", "أَرْبَعُمِائَة", "خَمْسُمِائَة", "سِتُّمِائَة",
     "سَبْعُمِائَة", "ثَمَانِيمِائَة", "تِسْعُمِائَة",
+    # greetings / frequent verbs the OOV net was misguessing
+    "عَلَيْكُمْ", "أَتَكَلَّم", "تَتَكَلَّم", "يَتَكَلَّم",
+    "عَرَبِيَّة", "إِنْجِلِيزِيَّة", "رِسَالَة", "مَرْحَبًا",
+    "أَهْلًا", "سَهْلًا",
 ]
 
 # first occurrence wins: _WORDS is frequency-ordered, so for ambiguous

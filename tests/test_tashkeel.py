@@ -173,3 +173,15 @@ def test_number_words_diacritized():
     assert t.diacritize(num_to_words_ar(345)) == \
         "ثَلَاثُمِائَة وَخَمْسَة وَأَرْبَعُونَ"
     assert t.diacritize(num_to_words_ar(1000)) == "أَلْف"
+
+
+def test_ar_article_and_long_vowels():
+    """G2P-level Arabic: haraka+mater = one long vowel; the definite
+    article is short al- and assimilates into sun-letter geminates."""
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    ar = _get_g2p("ar")
+    assert ar.word_to_ipa("كِتَاب") == "kɪtaːb"
+    assert ar.word_to_ipa("كَبِير") == "kabiːr"
+    assert ar.word_to_ipa("السَّلَام") == "asːalaːm"
+    assert ar.word_to_ipa("الْوَلَد") == "alwalad"
