@@ -940,10 +940,19 @@ def _get_g2p(voice: str) -> RuleG2P:
     elif base == "cs":
         g = RuleG2P(_CS_RULES, letters="a-zA-Začďéěíňóřšťúůýž")
     elif base == "ar":
+        def _ar_pre(w: str) -> str:
+            # canonical combining order puts the vowel BEFORE shadda
+            # (fatha ccc=30 < shadda ccc=33); gemination must apply
+            # first, so swap to shadda+vowel
+            import unicodedata as _ud
+            w = _ud.normalize("NFC", w)
+            return re.sub("([ً-ِ])(ّ)", r"\2\1", w)
+
         g = RuleG2P(
             _AR_RULES,
             letters="؀-ۿ",
             stress=False,
+            preprocess=_ar_pre,
         )
 
         def _ar_post(ipa: str) -> str:
