@@ -563,8 +563,21 @@ TABLES = {
     "uk": UK_RULES, "hr": HR_RULES, "sk": SK_RULES, "id": ID_RULES,
     "sw": SW_RULES,
 }
-# sr (Serbian latin) shares the hr table; nb/nn map to no
-ALIASES = {"sr": "hr", "bs": "hr", "nb": "no", "nn": "no", "ms": "id"}
+# Serbian is digraphic: the hr Latin table plus the Cyrillic azbuka
+# in ONE rule set (codepoints don't collide), so both scripts work
+SR_CYR = {
+    "а": "a", "б": "b", "в": "ʋ", "г": "ɡ", "д": "d", "ђ": "dʑ",
+    "е": "e", "ж": "ʒ", "з": "z", "и": "i", "ј": "j", "к": "k",
+    "л": "l", "љ": "ʎ", "м": "m", "н": "n", "њ": "ɲ", "о": "o",
+    "п": "p", "р": "r", "с": "s", "т": "t", "ћ": "tɕ", "у": "u",
+    "ф": "f", "х": "x", "ц": "ts", "ч": "tʃ", "џ": "dʒ", "ш": "ʃ",
+}
+SR_RULES = {**HR_RULES, **SR_CYR}
+TABLES["sr"] = SR_RULES
+LETTERS["sr"] = "a-zA-ZčćđšžČĆĐŠŽа-шђјљњћџА-ШЂЈЉЊЋЏ"
+
+# bs shares the hr table; nb/nn map to no
+ALIASES = {"bs": "hr", "nb": "no", "nn": "no", "ms": "id"}
 
 
 # --------------------------------------------------------------------- #

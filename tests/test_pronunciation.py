@@ -708,3 +708,12 @@ def test_fixed_initial_stress_not_poisoned_by_en_heuristics():
     assert _get_g2p("cs").word_to_ipa("společnost").startswith("spˈ")
     # English keeps both heuristics
     assert _get_g2p("en").word_to_ipa("information") == "ɪnfɚmˈeɪʃən"
+
+
+def test_serbian_is_digraphic():
+    """Serbian works in BOTH scripts (espeak's sr reads Cyrillic; the
+    old alias to the Latin-only hr table silently dropped Cyrillic)."""
+    g = _get_g2p("sr")
+    assert g.word_to_ipa("земља") == "zˈemʎa"
+    assert g.word_to_ipa("zemlja") == "zˈemʎa"
+    assert g.word_to_ipa("Београд") == "bˈeoɡrad"
