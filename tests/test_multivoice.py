@@ -64,6 +64,71 @@ def test_tashkeel_applied_for_arabic(voices):
     assert len(sents) >= 1 and len(sents[0]) > 0
 
 
+_ALL_LANG_TEXTS = {
+    "en-us": "Hello world.", "en": "Hello world.",
+    "de": "Hallo Welt.", "es": "Hola mundo.",
+    "fr": "Bonjour le monde.", "it": "Ciao mondo.",
+    "pt": "Olá mundo.", "nl": "Hallo wereld.",
+    "pl": "Witaj świecie.", "ru": "Привет мир.",
+    "tr": "Merhaba dünya.", "cs": "Ahoj světe.",
+    "ar": "مرحبا بالعالم.",
+    # round-2 expansion languages (g2p_tables.py)
+    "sv": "Hej världen.", "no": "Hei verden.", "nb": "Hei verden.",
+    "nn": "Hei verda.", "da": "Hej verden.", "fi": "Hei maailma.",
+    "hu": "Helló világ.", "ro": "Salut lume.", "el": "Γεια σου κόσμε.",
+    "bg": "Здравей свят.", "uk": "Привіт світе.",
+    "hr": "Pozdrav svijete.", "sr": "Pozdrav svete.",
+    "sk": "Ahoj svet.", "id": "Halo dunia.", "ms": "Halo dunia.",
+    "sw": "Habari dunia.", "bs": "Pozdrav svijete.",
+    # second expansion batch
+    "eo": "Saluton mondo.", "ca": "Hola món.", "gl": "Ola mundo.",
+    "eu": "Kaixo mundua.", "az": "Salam dünya.",
+    "kk": "Сәлем әлем.", "ky": "Салам дүйнө.",
+    "uz": "Salom dunyo.", "mk": "Здраво свету.",
+    "be": "Прывітанне свет.", "sl": "Pozdravljen svet.",
+    "lt": "Labas pasauli.", "lv": "Sveika pasaule.",
+    "et": "Tere maailm.", "is": "Halló heimur.",
+    "sq": "Përshëndetje botë.", "hy": "Բարև աշխարհ.",
+    "ka": "გამარჯობა მსოფლიო.", "af": "Hallo wêreld.",
+    "cy": "Helo byd.", "mt": "Bongu dinja.",
+    "ht": "Bonjou monn.", "la": "Salve munde.",
+    "hi": "नमस्ते दुनिया.",
+    # third batch: Brahmic engine
+    "mr": "नमस्कार जग.", "ne": "नमस्ते संसार.",
+    "bn": "ওহে বিশ্ব.", "as": "নমস্কাৰ পৃথিৱী.",
+    "gu": "નમસ્તે દુનિયા.", "pa": "ਸਤ ਸ੍ਰੀ ਅਕਾਲ ਦੁਨਿਆ.",
+    "or": "ନମସ୍କାର ଜଗତ.", "ta": "வணக்கம் உலகம்.",
+    "te": "నమస్కారం ప్రపంచం.", "kn": "ನಮಸ್ಕಾರ ಜಗತ್ತು.",
+    "ml": "നമസ്കാരം ലോകം.", "si": "ආයුබෝවන් ලෝකය.",
+    # third batch: syllabic scripts + kana
+    "ko": "안녕하세요 세계.", "am": "ሰላም ለዓለም.",
+    "chr": "ᎣᏏᏲ ᎡᎶᎯ.", "ja": "こんにちは せかい.",
+    # third batch: rule tables
+    "fa": "سلام دنیا.", "ur": "ہیلو دنیا.",
+    "ug": "سالام دۇنيا.", "he": "שלום עולם.",
+    "vi": "Chào thế giới.", "mi": "Kia ora te ao.",
+    "haw": "Aloha honua.", "qu": "Allin p'unchay pacha.",
+    "gn": "Mba'éichapa arapy.", "nci": "Niltze cemanahuac.",
+    "om": "Akkam addunyaa.", "tn": "Dumela lefatshe.",
+    "pap": "Bon dia mundu.", "ia": "Salute mundo.",
+    "io": "Saluto mondo.", "lfn": "Saluta mundo.",
+    "jbo": "coi le munje", "tk": "Salam dünýä.",
+    "lb": "Moien Welt.", "kl": "Aluu silarsuaq.",
+    "ga": "Dia duit a dhomhain.", "grc": "χαῖρε κόσμε.",
+    "tt": "Сәлам дөнья.", "ba": "Сәләм донъя.",
+    "cv": "Салам тӗнче.",
+    "kok": "नमस्कार संसार.", "my": "မင်္ဂလာပါ ကမ္ဘာ.",
+    "th": "สวัสดีโลก.", "an": "Ola mundo.", "ku": "Silav cîhan.",
+    "gd": "Halò a shaoghail.", "quc": "Saqarik uleew.",
+    "sd": "سلام دنيا.", "nog": "Салам дуныя.",
+    "smj": "Buoris væráldda.",
+    "bpy": "আমার ঠার.", "shn": "မႂ်ႇသုင် ၵမ်ႇၽႃႇ.",
+    "qya": "Elen síla lúmenn omentielvo.",
+    "sjn": "Mae govannen, mellon nîn.",
+    "piqd": "tlhIngan Hol vIjatlh.",
+}
+
+
 def test_all_languages_end_to_end(tmp_path):
     """Every supported G2P language synthesizes audio through its own
     voice pack (text -> phonemes -> ids -> VITS -> waveform)."""
@@ -71,69 +136,7 @@ def test_all_languages_end_to_end(tmp_path):
     from sonata_amd.models.voice import load_voice
     from sonata_amd.text.phonemizer import available_languages
 
-    texts = {
-        "en-us": "Hello world.", "en": "Hello world.",
-        "de": "Hallo Welt.", "es": "Hola mundo.",
-        "fr": "Bonjour le monde.", "it": "Ciao mondo.",
-        "pt": "Olá mundo.", "nl": "Hallo wereld.",
-        "pl": "Witaj świecie.", "ru": "Привет мир.",
-        "tr": "Merhaba dünya.", "cs": "Ahoj světe.",
-        "ar": "مرحبا بالعالم.",
-        # round-2 expansion languages (g2p_tables.py)
-        "sv": "Hej världen.", "no": "Hei verden.", "nb": "Hei verden.",
-        "nn": "Hei verda.", "da": "Hej verden.", "fi": "Hei maailma.",
-        "hu": "Helló világ.", "ro": "Salut lume.", "el": "Γεια σου κόσμε.",
-        "bg": "Здравей свят.", "uk": "Привіт світе.",
-        "hr": "Pozdrav svijete.", "sr": "Pozdrav svete.",
-        "sk": "Ahoj svet.", "id": "Halo dunia.", "ms": "Halo dunia.",
-        "sw": "Habari dunia.", "bs": "Pozdrav svijete.",
-        # second expansion batch
-        "eo": "Saluton mondo.", "ca": "Hola món.", "gl": "Ola mundo.",
-        "eu": "Kaixo mundua.", "az": "Salam dünya.",
-        "kk": "Сәлем әлем.", "ky": "Салам дүйнө.",
-        "uz": "Salom dunyo.", "mk": "Здраво свету.",
-        "be": "Прывітанне свет.", "sl": "Pozdravljen svet.",
-        "lt": "Labas pasauli.", "lv": "Sveika pasaule.",
-        "et": "Tere maailm.", "is": "Halló heimur.",
-        "sq": "Përshëndetje botë.", "hy": "Բարև աշխարհ.",
-        "ka": "გამარჯობა მსოფლიო.", "af": "Hallo wêreld.",
-        "cy": "Helo byd.", "mt": "Bongu dinja.",
-        "ht": "Bonjou monn.", "la": "Salve munde.",
-        "hi": "नमस्ते दुनिया.",
-        # third batch: Brahmic engine
-        "mr": "नमस्कार जग.", "ne": "नमस्ते संसार.",
-        "bn": "ওহে বিশ্ব.", "as": "নমস্কাৰ পৃথিৱী.",
-        "gu": "નમસ્તે દુનિયા.", "pa": "ਸਤ ਸ੍ਰੀ ਅਕਾਲ ਦੁਨਿਆ.",
-        "or": "ନମସ୍କାର ଜଗତ.", "ta": "வணக்கம் உலகம்.",
-        "te": "నమస్కారం ప్రపంచం.", "kn": "ನಮಸ್ಕಾರ ಜಗತ್ತು.",
-        "ml": "നമസ്കാരം ലോകം.", "si": "ආයුබෝවන් ලෝකය.",
-        # third batch: syllabic scripts + kana
-        "ko": "안녕하세요 세계.", "am": "ሰላም ለዓለም.",
-        "chr": "ᎣᏏᏲ ᎡᎶᎯ.", "ja": "こんにちは せかい.",
-        # third batch: rule tables
-        "fa": "سلام دنیا.", "ur": "ہیلو دنیا.",
-        "ug": "سالام دۇنيا.", "he": "שלום עולם.",
-        "vi": "Chào thế giới.", "mi": "Kia ora te ao.",
-        "haw": "Aloha honua.", "qu": "Allin p'unchay pacha.",
-        "gn": "Mba'éichapa arapy.", "nci": "Niltze cemanahuac.",
-        "om": "Akkam addunyaa.", "tn": "Dumela lefatshe.",
-        "pap": "Bon dia mundu.", "ia": "Salute mundo.",
-        "io": "Saluto mondo.", "lfn": "Saluta mundo.",
-        "jbo": "coi le munje", "tk": "Salam dünýä.",
-        "lb": "Moien Welt.", "kl": "Aluu silarsuaq.",
-        "ga": "Dia duit a dhomhain.", "grc": "χαῖρε κόσμε.",
-        "tt": "Сәлам дөнья.", "ba": "Сәләм донъя.",
-        "cv": "Салам тӗнче.",
-        "kok": "नमस्कार संसार.", "my": "မင်္ဂလာပါ ကမ္ဘာ.",
-        "th": "สวัสดีโลก.", "an": "Ola mundo.", "ku": "Silav cîhan.",
-        "gd": "Halò a shaoghail.", "quc": "Saqarik uleew.",
-        "sd": "سلام دنيا.", "nog": "Салам дуныя.",
-        "smj": "Buoris væráldda.",
-        "bpy": "আমার ঠার.", "shn": "မႂ်ႇသုင် ၵမ်ႇၽႃႇ.",
-        "qya": "Elen síla lúmenn omentielvo.",
-        "sjn": "Mae govannen, mellon nîn.",
-        "piqd": "tlhIngan Hol vIjatlh.",
-    }
+    texts = _ALL_LANG_TEXTS
     for lang in available_languages():
         pack = create_random_voice(str(tmp_path), f"lang_{lang}",
                                    quality="x_low", language=lang)
@@ -166,3 +169,19 @@ def test_gpu_multilingual_synthesis(tmp_path):
         audio = v.speak_one_sentence(sents[0])
         assert len(audio.samples) > 1000, lang
         assert float(abs(audio.samples).max()) > 0, lang
+
+
+def test_no_word_silently_dropped():
+    """Every word of every language's sample must phonemize to a
+    non-empty string (guards against script-coverage holes like the
+    Cyrillic-Serbian alias bug)."""
+    import re as _re
+
+    from sonata_amd.text.phonemizer import _get_g2p
+
+    texts = _ALL_LANG_TEXTS
+    for lang, txt in texts.items():
+        g = _get_g2p(lang)
+        for w in _re.split(r"[\s.,!?।۔]+", txt):
+            if w:
+                assert g.word_to_ipa(w).strip(), (lang, w)
