@@ -716,6 +716,15 @@ def normalize(text: str, language: str) -> str:
             text = _DEC_COMMA_RE.sub(
                 lambda m: card(int(m.group(1))) + " " + dec + " "
                 + " ".join(digits[int(d)] for d in m.group(2)), text)
+        if base == "ja":
+            # counter readings differ from the bare kanji: 1月 is
+            # いちがつ (not つき), 3日/5分 approximated with the
+            # regular counters
+            text = re.sub(r"(\d{1,2})月", r"\1がつ ", text)
+            text = re.sub(r"(\d+)日", r"\1にち ", text)
+            text = re.sub(r"(\d+)分", r"\1ふん ", text)
+            text = re.sub(r"(\d+)年", r"\1ねん ", text)
+            text = re.sub(r"(\d+)円", r"\1えん ", text)
         if base == "de":
             text = _de_expand_ordinals(text)
         elif base == "fr":

@@ -417,3 +417,11 @@ def test_slavic_currency_agreement():
     assert normalize("5$", "ru") == "пять долларов"
     assert normalize("21₽", "ru") == "двадцать один рубль"
     assert normalize("5 zł", "pl") == "pięć złotych"
+
+
+def test_ja_date_counters():
+    from sonata_amd.text.normalize import normalize
+
+    out = normalize("1月 2023年 500円", "ja")
+    assert "いちがつ" in out and "ねん" in out and "えん" in out
+    assert "つき" not in out
