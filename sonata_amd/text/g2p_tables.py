@@ -2,9 +2,11 @@
 
 Languages with (near-)regular orthographies where an ordered
 longest-match rule table gives defensible pronunciations.  Together
-with phonemizer.py's original twelve this covers 25 languages — an
-honest fraction of the reference's ~100 espeak-ng dictionaries
-(documented in PARITY.md; quality corpora in tests/test_pronunciation.py).
+with phonemizer.py's originals, g2p_tables3.py, g2p_indic.py and
+g2p_scripts.py the build covers 111 language codes against the
+reference's ~105 espeak-ng dictionaries (documented in PARITY.md and
+docs/LANGUAGES.md; quality corpora in tests/test_pronunciation.py and
+tests/test_g2p_batch3.py).
 
 Conventions: IPA over the Piper symbol set; stress added by RuleG2P
 (first-syllable default unless noted; fixed-stress languages configure

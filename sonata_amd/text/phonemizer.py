@@ -9,11 +9,14 @@ splitting (:65-83).
 
 The reference shells into the espeak-ng C library (a patched fork).  That
 dependency does not exist here; this module is a fresh, self-contained
-rule-based G2P: a per-language ordered longest-match rule table plus an
-English exception lexicon.  It produces IPA over the same symbol set the
-Piper voices use, is deterministic, and is thread-safe (pure functions, no
-global C state — the reference's espeak is famously NOT thread-safe,
-SURVEY.md §5).
+G2P front: per-language ordered longest-match rule tables, exception
+lexicons, and script ENGINES (Brahmic abugidas in g2p_indic.py; Hangul/
+Ethiopic/Cherokee/Myanmar/Thai + kana in g2p_scripts.py / g2p_tables3.py)
+covering 111 language codes (docs/LANGUAGES.md).  It produces IPA over
+the same symbol set the Piper voices use, is deterministic, and is
+thread-safe (pure functions, no global C state — the reference's espeak
+is famously NOT thread-safe; see
+tests/test_fuzz_g2p.py::test_concurrent_phonemize_across_languages).
 """
 
 from __future__ import annotations
