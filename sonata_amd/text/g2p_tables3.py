@@ -402,6 +402,31 @@ JA_KANJI = {
     "痛い": "いたい", "甘い": "あまい", "辛い": "からい",
     "冷たい": "つめたい", "温かい": "あたたかい",
     "涼しい": "すずしい", "暖かい": "あたたかい",
+    # third batch
+    "言語": "げんご", "国語": "こくご", "単語": "たんご",
+    "文章": "ぶんしょう", "文字": "もじ", "漢字": "かんじ",
+    "発音": "はつおん", "翻訳": "ほんやく", "会話": "かいわ",
+    "空港": "くうこう", "港": "みなと", "島": "しま",
+    "番号": "ばんごう", "電子": "でんし", "情報": "じょうほう",
+    "技術": "ぎじゅつ", "機械": "きかい", "工場": "こうじょう",
+    "産業": "さんぎょう", "農業": "のうぎょう", "商業": "しょうぎょう",
+    "野球": "やきゅう", "映画館": "えいがかん", "美術館": "びじゅつかん",
+    "博物館": "はくぶつかん", "神社": "じんじゃ", "寺": "てら",
+    "城": "しろ", "村": "むら", "県": "けん", "区": "く",
+    "通り": "とおり", "交差点": "こうさてん", "信号": "しんごう",
+    "切手": "きって", "葉書": "はがき", "封筒": "ふうとう",
+    "新しい車": "あたらしいくるま", "運転": "うんてん",
+    "旅客": "りょかく", "乗客": "じょうきゃく", "駅員": "えきいん",
+    "店員": "てんいん", "銀行員": "ぎんこういん",
+    "公務員": "こうむいん", "会社員": "かいしゃいん",
+    "看護師": "かんごし", "記者": "きしゃ", "歌手": "かしゅ",
+    "選手": "せんしゅ", "俳優": "はいゆう", "作家": "さっか",
+    "画家": "がか", "写真家": "しゃしんか",
+    "兄": "あに", "姉": "あね", "弟": "おとうと", "妹": "いもうと",
+    "祖父": "そふ", "祖母": "そぼ", "両親": "りょうしん",
+    "夫": "おっと", "妻": "つま", "息子": "むすこ", "娘": "むすめ",
+    "赤ちゃん": "あかちゃん", "大人": "おとな", "老人": "ろうじん",
+    "青年": "せいねん", "少年": "しょうねん", "少女": "しょうじょ",
 }
 _JA_KANJI_MAX = max(len(k) for k in JA_KANJI)
 
