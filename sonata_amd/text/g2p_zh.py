@@ -1,0 +1,762 @@
+"""Chinese G2P: Mandarin (cmn, alias zh) and Cantonese (yue).
+
+Reference bar: espeak-ng's zh/zhy dictionaries
+(deps/dev/espeak-ng-data/{zh,zhy}_dict, reached through
+crates/text/espeak-phonemizer/src/lib.rs:65-156).  Hanzi is not a
+phonetic script, so — like espeak — readings come from a dictionary:
+a multi-character word dictionary (longest match first, which also
+disambiguates the common polyphones: 了 le/liǎo, 行 xíng/háng,
+银行 yín háng …) backed by a single-character dictionary for the
+frequency core of the writing system.  Unlisted hanzi are dropped,
+stated honestly in PARITY.md / docs/LANGUAGES.md.
+
+Phonology:
+- Mandarin syllables are stored as pinyin with tone digits
+  ("zhong1 guo2") and converted to IPA with Chao tone letters
+  (˥ ˧˥ ˨˩˦ ˥˩; neutral tone unmarked).  The converter applies the
+  standard sandhi: 3-3 -> 2-3, 不 bù -> bú before tone 4, 一 yī ->
+  yí before tone 4 / yì before tones 1-3.
+- Cantonese syllables are stored as jyutping with tone digits
+  ("gwong2 dung1") over the six-tone system (˥ ˧˥ ˧ ˨˩ ˩˧ ˨).
+
+Every IPA symbol emitted here is in ids.py's inventory (guarded by
+tests/test_g2p_zh.py::test_zh_symbols_encodable).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List
+
+# --------------------------------------------------------------------- #
+# Pinyin -> IPA
+# --------------------------------------------------------------------- #
+_PY_INITIALS = [
+    # longest first
+    ("zh", "ʈʂ"), ("ch", "ʈʂʰ"), ("sh", "ʂ"),
+    ("b", "p"), ("p", "pʰ"), ("m", "m"), ("f", "f"),
+    ("d", "t"), ("t", "tʰ"), ("n", "n"), ("l", "l"),
+    ("g", "k"), ("k", "kʰ"), ("h", "x"),
+    ("j", "tɕ"), ("q", "tɕʰ"), ("x", "ɕ"),
+    ("r", "ʐ"), ("z", "ts"), ("c", "tsʰ"), ("s", "s"),
+]
+
+# finals after an initial ("v" = ü); longest-key match
+_PY_FINALS = {
+    "a": "a", "o": "o", "e": "ɤ", "i": "i", "u": "u", "v": "y",
+    "ai": "ai", "ei": "ei", "ao": "au", "ou": "ou",
+    "an": "an", "en": "ən", "ang": "aŋ", "eng": "əŋ",
+    "ong": "ʊŋ", "er": "ɚ",
+    "ia": "ja", "ie": "jɛ", "iao": "jau", "iu": "jou",
+    "ian": "jɛn", "in": "in", "iang": "jaŋ", "ing": "iŋ",
+    "iong": "jʊŋ",
+    "ua": "wa", "uo": "wo", "uai": "wai", "ui": "wei",
+    "uan": "wan", "un": "wən", "uang": "waŋ", "ueng": "wəŋ",
+    "ve": "ɥɛ", "van": "ɥɛn", "vn": "yn",
+}
+
+# whole syllables with no initial (y-/w-/bare vowels)
+_PY_STANDALONE = {
+    "yi": "i", "ya": "ja", "ye": "jɛ", "yao": "jau", "you": "jou",
+    "yan": "jɛn", "yin": "in", "yang": "jaŋ", "ying": "iŋ",
+    "yong": "jʊŋ", "yo": "jo",
+    "yu": "y", "yue": "ɥɛ", "yuan": "ɥɛn", "yun": "yn",
+    "wu": "u", "wa": "wa", "wo": "wo", "wai": "wai", "wei": "wei",
+    "wan": "wan", "wen": "wən", "wang": "waŋ", "weng": "wəŋ",
+    "a": "a", "o": "o", "e": "ɤ", "ai": "ai", "ei": "ei",
+    "ao": "au", "ou": "ou", "an": "an", "en": "ən", "ang": "aŋ",
+    "eng": "əŋ", "er": "ɚ", "n": "n", "ng": "ŋ",
+}
+
+_CMN_TONES = {1: "˥", 2: "˧˥", 3: "˨˩˦", 4: "˥˩", 5: "", 0: ""}
+# apical vowel after the sibilant series (zi ci si zhi chi shi ri)
+_SIBILANTS = ("ts", "tsʰ", "s", "ʈʂ", "ʈʂʰ", "ʂ", "ʐ")
+# after j/q/x, written u IS ü
+_PALATALS = ("j", "q", "x")
+
+
+def pinyin_syllable_to_ipa(syl: str) -> str:
+    """One numbered pinyin syllable ("zhong1", "lv4", "er2") -> IPA
+    with a Chao tone contour.  Unknown syllables return ""."""
+    syl = syl.strip().lower().replace("ü", "v").replace("u:", "v")
+    tone = 5
+    if syl and syl[-1].isdigit():
+        tone = int(syl[-1])
+        syl = syl[:-1]
+    if not syl:
+        return ""
+    body = None
+    if syl in _PY_STANDALONE:
+        body = _PY_STANDALONE[syl]
+    else:
+        ini_py, ini_ipa = "", ""
+        for py, ipa in _PY_INITIALS:
+            if syl.startswith(py):
+                ini_py, ini_ipa = py, ipa
+                break
+        fin = syl[len(ini_py):]
+        if ini_py in _PALATALS and fin.startswith("u"):
+            fin = "v" + fin[1:]
+        if fin == "i" and ini_ipa in _SIBILANTS:
+            body = ini_ipa + "ɨ"
+        else:
+            fin_ipa = _PY_FINALS.get(fin)
+            if ini_py and fin_ipa is not None:
+                body = ini_ipa + fin_ipa
+    if body is None:
+        return ""
+    return body + _CMN_TONES.get(tone, "")
+
+
+def _cmn_sandhi(syls: List[str], hanzi: str) -> List[str]:
+    """Apply standard Mandarin tone sandhi to a word's numbered-pinyin
+    syllables.  `hanzi` aligns character-per-syllable when lengths
+    match (for the 不/一 rules)."""
+    tones = [int(s[-1]) if s and s[-1].isdigit() else 5 for s in syls]
+    bodies = [s[:-1] if s and s[-1].isdigit() else s for s in syls]
+    aligned = len(hanzi) == len(syls)
+    for i in range(len(syls) - 1):
+        nxt = tones[i + 1]
+        if aligned and hanzi[i] == "不" and nxt == 4:
+            tones[i] = 2
+        elif aligned and hanzi[i] == "一":
+            if nxt == 4:
+                tones[i] = 2
+            elif nxt in (1, 2, 3):
+                tones[i] = 4
+    # 3-3 -> 2-3, right-to-left so runs of 3s resolve (2 2 3)
+    for i in range(len(syls) - 2, -1, -1):
+        if tones[i] == 3 and tones[i + 1] == 3:
+            tones[i] = 2
+    return [b + str(t) for b, t in zip(bodies, tones)]
+
+
+# --------------------------------------------------------------------- #
+# Jyutping -> IPA (Cantonese)
+# --------------------------------------------------------------------- #
+_JP_INITIALS = [
+    ("gw", "kʷ"), ("kw", "kʷʰ"), ("ng", "ŋ"),
+    ("b", "p"), ("p", "pʰ"), ("m", "m"), ("f", "f"),
+    ("d", "t"), ("t", "tʰ"), ("n", "n"), ("l", "l"),
+    ("g", "k"), ("k", "kʰ"), ("h", "h"),
+    ("z", "ts"), ("c", "tsʰ"), ("s", "s"),
+    ("j", "j"), ("w", "w"),
+]
+
+# whole rimes (LSHK jyutping): long aa vs short a; ei/ou are close
+# diphthongs; i/u lower to ɪ/ʊ before velars
+_JP_RIMES = {
+    "aa": "aː", "aai": "aːi", "aau": "aːu", "aam": "aːm",
+    "aan": "aːn", "aang": "aːŋ", "aap": "aːp", "aat": "aːt",
+    "aak": "aːk",
+    "a": "ɐ", "ai": "ɐi", "au": "ɐu", "am": "ɐm", "an": "ɐn",
+    "ang": "ɐŋ", "ap": "ɐp", "at": "ɐt", "ak": "ɐk",
+    "e": "ɛː", "ei": "ei", "eu": "ɛːu", "em": "ɛːm", "eng": "ɛːŋ",
+    "ep": "ɛːp", "ek": "ɛːk",
+    "i": "iː", "iu": "iːu", "im": "iːm", "in": "iːn", "ing": "ɪŋ",
+    "ip": "iːp", "it": "iːt", "ik": "ɪk",
+    "o": "ɔː", "oi": "ɔːi", "ou": "ou", "on": "ɔːn", "ong": "ɔːŋ",
+    "ot": "ɔːt", "ok": "ɔːk", "om": "ɔːm",
+    "u": "uː", "ui": "uːi", "un": "uːn", "ung": "ʊŋ", "ut": "uːt",
+    "uk": "ʊk",
+    "oe": "œː", "oeng": "œːŋ", "oek": "œːk",
+    "eo": "ɵ", "eoi": "ɵy", "eon": "ɵn", "eot": "ɵt",
+    "yu": "yː", "yun": "yːn", "yut": "yːt",
+}
+
+_YUE_TONES = {1: "˥", 2: "˧˥", 3: "˧", 4: "˨˩", 5: "˩˧", 6: "˨"}
+
+
+def jyutping_syllable_to_ipa(syl: str) -> str:
+    syl = syl.strip().lower()
+    tone = 0
+    if syl and syl[-1].isdigit():
+        tone = int(syl[-1])
+        syl = syl[:-1]
+    if not syl:
+        return ""
+    if syl in ("m", "ng"):  # syllabic nasals (唔 m4, 五 ng5)
+        body = "m̩" if syl == "m" else "ŋ̩"
+        return (body.replace("̩", "")  # keep inventory small: plain nasal
+                + _YUE_TONES.get(tone, ""))
+    ini_py, ini_ipa = "", ""
+    for py, ipa in _JP_INITIALS:
+        if syl.startswith(py) and len(syl) > len(py):
+            ini_py, ini_ipa = py, ipa
+            break
+    rime = _JP_RIMES.get(syl[len(ini_py):])
+    if rime is None:
+        return ""
+    return ini_ipa + rime + _YUE_TONES.get(tone, "")
+
+
+# --------------------------------------------------------------------- #
+# Mandarin word dictionary (polyphone disambiguation + compounds).
+# Longest match wins over the single-character dictionary below.
+# --------------------------------------------------------------------- #
+CMN_WORDS: Dict[str, str] = {
+    # --- polyphone-bearing compounds ---
+    "银行": "yin2 hang2", "行业": "hang2 ye4", "行动": "xing2 dong4",
+    "自行车": "zi4 xing2 che1", "进行": "jin4 xing2",
+    "了解": "liao3 jie3", "受不了": "shou4 bu4 liao3",
+    "音乐": "yin1 yue4", "快乐": "kuai4 le4", "乐趣": "le4 qu4",
+    "长城": "chang2 cheng2", "长度": "chang2 du4",
+    "成长": "cheng2 zhang3", "长大": "zhang3 da4",
+    "校长": "xiao4 zhang3", "长江": "chang2 jiang1",
+    "重要": "zhong4 yao4", "重量": "zhong4 liang4",
+    "重新": "chong2 xin1", "重复": "chong2 fu4",
+    "还是": "hai2 shi4", "还有": "hai2 you3", "还没": "hai2 mei2",
+    "归还": "gui1 huan2", "还给": "huan2 gei3",
+    "都是": "dou1 shi4", "首都": "shou3 du1", "都市": "du1 shi4",
+    "得到": "de2 dao4", "觉得": "jue2 de5", "记得": "ji4 de5",
+    "得了": "de2 le5", "取得": "qu3 de2", "获得": "huo4 de2",
+    "睡觉": "shui4 jiao4", "感觉": "gan3 jue2", "觉得": "jue2 de5",
+    "发现": "fa1 xian4", "头发": "tou2 fa4", "发展": "fa1 zhan3",
+    "地方": "di4 fang1", "土地": "tu3 di4", "慢慢地": "man4 man4 de5",
+    "为什么": "wei4 shen2 me5", "因为": "yin1 wei4",
+    "为了": "wei4 le5", "认为": "ren4 wei2", "成为": "cheng2 wei2",
+    "作为": "zuo4 wei2", "行为": "xing2 wei2",
+    "便宜": "pian2 yi5", "方便": "fang1 bian4", "顺便": "shun4 bian4",
+    "干净": "gan1 jing4", "干活": "gan4 huo2", "干部": "gan4 bu4",
+    "教学": "jiao4 xue2", "教室": "jiao4 shi4", "教师": "jiao4 shi1",
+    "教育": "jiao4 yu4", "宗教": "zong1 jiao4", "请教": "qing3 jiao4",
+    "省会": "sheng3 hui4", "节省": "jie2 sheng3", "反省": "fan3 xing3",
+    "数学": "shu4 xue2", "数字": "shu4 zi4", "数量": "shu4 liang4",
+    "无数": "wu2 shu4", "数一数": "shu3 yi1 shu3",
+    "相信": "xiang1 xin4", "互相": "hu4 xiang1", "照相": "zhao4 xiang4",
+    "相片": "xiang4 pian4", "首相": "shou3 xiang4",
+    "朝鲜": "chao2 xian3", "朝着": "chao2 zhe5", "朝代": "chao2 dai4",
+    "朝阳": "zhao1 yang2",
+    "市场": "shi4 chang3", "现场": "xian4 chang3",
+    "一场": "yi1 chang3", "操场": "cao1 chang3",
+    "西藏": "xi1 zang4", "藏族": "zang4 zu2", "躲藏": "duo3 cang2",
+    "差不多": "cha4 bu4 duo1", "出差": "chu1 chai1",
+    "差别": "cha1 bie2", "差异": "cha1 yi4",
+    "传统": "chuan2 tong3", "传说": "chuan2 shuo1",
+    "传记": "zhuan4 ji4", "宣传": "xuan1 chuan2",
+    "处理": "chu3 li3", "处于": "chu3 yu2", "到处": "dao4 chu4",
+    "好处": "hao3 chu4", "办事处": "ban4 shi4 chu4",
+    "担心": "dan1 xin1", "负担": "fu4 dan1", "担子": "dan4 zi5",
+    "倒是": "dao4 shi4", "摔倒": "shuai1 dao3", "倒车": "dao4 che1",
+    "打倒": "da3 dao3", "倒水": "dao4 shui3",
+    "调查": "diao4 cha2", "调整": "tiao2 zheng3", "空调": "kong1 tiao2",
+    "声调": "sheng1 diao4", "调节": "tiao2 jie2",
+    "读书": "du2 shu1", "阅读": "yue4 du2",
+    "恶心": "e3 xin1", "可恶": "ke3 wu4", "恶劣": "e4 lie4",
+    "分钟": "fen1 zhong1", "部分": "bu4 fen5", "分数": "fen1 shu4",
+    "十分": "shi2 fen1", "身分": "shen1 fen4", "分别": "fen1 bie2",
+    "缝隙": "feng4 xi4", "缝纫": "feng2 ren4",
+    "提供": "ti2 gong1", "供应": "gong1 ying4", "供品": "gong4 pin3",
+    "冠军": "guan4 jun1", "皇冠": "huang2 guan1",
+    "爱好": "ai4 hao4", "好奇": "hao4 qi2", "好像": "hao3 xiang4",
+    "你好": "ni3 hao3", "好吃": "hao3 chi1",
+    "号码": "hao4 ma3", "记号": "ji4 hao5",
+    "和平": "he2 ping2", "暖和": "nuan3 huo5",
+    "中华": "zhong1 hua2", "华语": "hua2 yu3", "华山": "hua4 shan1",
+    "开会": "kai1 hui4", "会计": "kuai4 ji4", "机会": "ji1 hui4",
+    "会议": "hui4 yi4", "学会": "xue2 hui4",
+    "几乎": "ji1 hu1", "几个": "ji3 ge4", "几天": "ji3 tian1",
+    "茶几": "cha2 ji1",
+    "假期": "jia4 qi1", "假如": "jia3 ru2", "放假": "fang4 jia4",
+    "假装": "jia3 zhuang1", "请假": "qing3 jia4",
+    "时间": "shi2 jian1", "房间": "fang2 jian1", "中间": "zhong1 jian1",
+    "间接": "jian4 jie1", "空间": "kong1 jian1",
+    "将来": "jiang1 lai2", "将军": "jiang1 jun1", "麻将": "ma2 jiang4",
+    "角度": "jiao3 du4", "角色": "jue2 se4", "主角": "zhu3 jue2",
+    "三角": "san1 jiao3",
+    "结果": "jie2 guo3", "结婚": "jie2 hun1", "结实": "jie1 shi5",
+    "结束": "jie2 shu4", "团结": "tuan2 jie2",
+    "尽量": "jin3 liang4", "尽管": "jin3 guan3", "尽力": "jin4 li4",
+    "卷子": "juan4 zi5", "胶卷": "jiao1 juan3",
+    "空气": "kong1 qi4", "天空": "tian1 kong1", "空儿": "kong4 er2",
+    "有空": "you3 kong4", "空闲": "kong4 xian2",
+    "积累": "ji1 lei3", "累了": "lei4 le5", "劳累": "lao2 lei4",
+    "力量": "li4 liang4", "商量": "shang1 liang5",
+    "测量": "ce4 liang2", "大量": "da4 liang4", "质量": "zhi4 liang4",
+    "困难": "kun4 nan2", "难过": "nan2 guo4", "灾难": "zai1 nan4",
+    "难民": "nan4 min2", "难道": "nan2 dao4",
+    "宁静": "ning2 jing4", "宁可": "ning4 ke3", "宁愿": "ning4 yuan4",
+    "漂亮": "piao4 liang5", "漂流": "piao1 liu2",
+    "一切": "yi1 qie4", "切菜": "qie1 cai4", "亲切": "qin1 qie4",
+    "歌曲": "ge1 qu3", "弯曲": "wan1 qu1", "曲线": "qu1 xian4",
+    "散步": "san4 bu4", "散文": "san3 wen2", "分散": "fen1 san4",
+    "打扫": "da3 sao3", "扫把": "sao4 ba3",
+    "宿舍": "su4 she4", "舍不得": "she3 bu4 de5",
+    "盛饭": "cheng2 fan4", "盛大": "sheng4 da4", "茂盛": "mao4 sheng4",
+    "相似": "xiang1 si4", "似的": "shi4 de5", "类似": "lei4 si4",
+    "熟悉": "shu2 xi1", "成熟": "cheng2 shu2",
+    "收缩": "shou1 suo1",
+    "提高": "ti2 gao1", "提前": "ti2 qian2",
+    "挑选": "tiao1 xuan3", "挑战": "tiao3 zhan4",
+    "吐出": "tu3 chu1", "呕吐": "ou3 tu4",
+    "投降": "tou2 xiang2", "下降": "xia4 jiang4", "降落": "jiang4 luo4",
+    "高兴": "gao1 xing4", "兴趣": "xing4 qu4", "兴奋": "xing1 fen4",
+    "流血": "liu2 xue4", "血液": "xue4 ye4",
+    "重要": "zhong4 yao4", "要求": "yao1 qiu2", "需要": "xu1 yao4",
+    "应该": "ying1 gai1", "答应": "da1 ying5", "应用": "ying4 yong4",
+    "反应": "fan3 ying4", "适应": "shi4 ying4",
+    "下载": "xia4 zai3", "载重": "zai4 zhong4", "记载": "ji4 zai3",
+    "涨价": "zhang3 jia4", "高涨": "gao1 zhang3",
+    "只有": "zhi3 you3", "只是": "zhi3 shi4", "一只": "yi1 zhi1",
+    "只要": "zhi3 yao4", "船只": "chuan2 zhi1",
+    "种类": "zhong3 lei4", "种子": "zhong3 zi5", "种植": "zhong4 zhi2",
+    "各种": "ge4 zhong3", "种地": "zhong4 di4",
+    "转变": "zhuan3 bian4", "转身": "zhuan3 shen1",
+    "旋转": "xuan2 zhuan4", "转动": "zhuan4 dong4",
+    "钻石": "zuan4 shi2", "钻研": "zuan1 yan2",
+    "着急": "zhao2 ji2", "穿着": "chuan1 zhuo2", "着火": "zhao2 huo3",
+    "看着": "kan4 zhe5", "着手": "zhuo2 shou3",
+    "的确": "di2 que4", "目的": "mu4 di4",
+    "弹琴": "tan2 qin2", "子弹": "zi3 dan4", "弹性": "tan2 xing4",
+    "没有": "mei2 you3", "淹没": "yan1 mo4", "没收": "mo4 shou1",
+    "背包": "bei1 bao1", "背后": "bei4 hou4", "背景": "bei4 jing3",
+    "方面": "fang1 mian4", "里面": "li3 mian4", "外面": "wai4 mian4",
+    "上面": "shang4 mian4", "下面": "xia4 mian4", "前面": "qian2 mian4",
+    "后面": "hou4 mian4", "面条": "mian4 tiao2",
+    # --- common words (reading reinforcement / speed) ---
+    "中国": "zhong1 guo2", "中文": "zhong1 wen2",
+    "普通话": "pu3 tong1 hua4", "汉语": "han4 yu3",
+    "北京": "bei3 jing1", "上海": "shang4 hai3",
+    "广州": "guang3 zhou1", "香港": "xiang1 gang3",
+    "台湾": "tai2 wan1", "美国": "mei3 guo2", "英国": "ying1 guo2",
+    "法国": "fa3 guo2", "德国": "de2 guo2", "日本": "ri4 ben3",
+    "世界": "shi4 jie4", "今天": "jin1 tian1", "明天": "ming2 tian1",
+    "昨天": "zuo2 tian1", "现在": "xian4 zai4", "时候": "shi2 hou5",
+    "什么": "shen2 me5", "怎么": "zen3 me5", "这么": "zhe4 me5",
+    "那么": "na4 me5", "多少": "duo1 shao3", "这个": "zhe4 ge4",
+    "那个": "na4 ge4", "我们": "wo3 men5", "你们": "ni3 men5",
+    "他们": "ta1 men5", "她们": "ta1 men5", "朋友": "peng2 you5",
+    "先生": "xian1 sheng5", "小姐": "xiao3 jie3",
+    "谢谢": "xie4 xie5", "再见": "zai4 jian4",
+    "对不起": "dui4 bu4 qi3", "没关系": "mei2 guan1 xi5",
+    "学习": "xue2 xi2", "学生": "xue2 sheng5", "学校": "xue2 xiao4",
+    "老师": "lao3 shi1", "大学": "da4 xue2", "电脑": "dian4 nao3",
+    "电话": "dian4 hua4", "电视": "dian4 shi4", "手机": "shou3 ji1",
+    "东西": "dong1 xi5", "工作": "gong1 zuo4", "公司": "gong1 si1",
+    "问题": "wen4 ti2", "意思": "yi4 si5", "名字": "ming2 zi5",
+    "汽车": "qi4 che1", "火车": "huo3 che1", "飞机": "fei1 ji1",
+    "孩子": "hai2 zi5", "儿子": "er2 zi5", "女儿": "nv3 er2",
+    "妈妈": "ma1 ma5", "爸爸": "ba4 ba5", "家庭": "jia1 ting2",
+    "喜欢": "xi3 huan1", "知道": "zhi1 dao4", "认识": "ren4 shi5",
+    "开始": "kai1 shi3", "已经": "yi3 jing1", "可以": "ke3 yi3",
+    "可能": "ke3 neng2", "所以": "suo3 yi3", "但是": "dan4 shi4",
+    "如果": "ru2 guo3", "虽然": "sui1 ran2", "当然": "dang1 ran2",
+    "非常": "fei1 chang2",
+    "一起": "yi1 qi3", "一样": "yi1 yang4", "一点": "yi1 dian3",
+    "有点": "you3 dian3", "大家": "da4 jia1", "国家": "guo2 jia1",
+    "政府": "zheng4 fu3", "经济": "jing1 ji4", "社会": "she4 hui4",
+    "文化": "wen2 hua4", "历史": "li4 shi3", "科学": "ke1 xue2",
+    "技术": "ji4 shu4", "艺术": "yi4 shu4", "语言": "yu3 yan2",
+    "文字": "wen2 zi4", "新闻": "xin1 wen2", "消息": "xiao1 xi5",
+    "情况": "qing2 kuang4", "环境": "huan2 jing4",
+    "身体": "shen1 ti3", "健康": "jian4 kang1", "医院": "yi1 yuan4",
+    "医生": "yi1 sheng1", "时代": "shi2 dai4", "地球": "di4 qiu2",
+    "太阳": "tai4 yang2", "月亮": "yue4 liang5", "星星": "xing1 xing5",
+    "动物": "dong4 wu4", "植物": "zhi2 wu4",
+}
+
+# --------------------------------------------------------------------- #
+# Mandarin single-character readings (frequency core).  One reading per
+# character — the most common in running text; polyphones whose other
+# readings matter are disambiguated by CMN_WORDS above.
+# --------------------------------------------------------------------- #
+CMN_CHARS: Dict[str, str] = {
+    # top of the frequency list
+    "的": "de5", "一": "yi1", "是": "shi4", "不": "bu4", "了": "le5",
+    "人": "ren2", "我": "wo3", "在": "zai4", "有": "you3", "他": "ta1",
+    "这": "zhe4", "中": "zhong1", "大": "da4", "来": "lai2",
+    "上": "shang4", "国": "guo2", "个": "ge4", "到": "dao4",
+    "说": "shuo1", "们": "men5", "为": "wei4", "子": "zi3",
+    "和": "he2", "你": "ni3", "地": "di4", "出": "chu1", "道": "dao4",
+    "也": "ye3", "时": "shi2", "年": "nian2", "得": "de5",
+    "就": "jiu4", "那": "na4", "要": "yao4", "下": "xia4",
+    "以": "yi3", "生": "sheng1", "会": "hui4", "自": "zi4",
+    "着": "zhe5", "去": "qu4", "之": "zhi1", "过": "guo4",
+    "家": "jia1", "学": "xue2", "对": "dui4", "可": "ke3",
+    "她": "ta1", "里": "li3", "后": "hou4", "小": "xiao3",
+    "么": "me5", "心": "xin1", "多": "duo1", "天": "tian1",
+    "而": "er2", "能": "neng2", "好": "hao3", "都": "dou1",
+    "然": "ran2", "没": "mei2", "日": "ri4", "于": "yu2",
+    "起": "qi3", "还": "hai2", "发": "fa1", "成": "cheng2",
+    "事": "shi4", "只": "zhi3", "作": "zuo4", "当": "dang1",
+    "想": "xiang3", "看": "kan4", "文": "wen2", "无": "wu2",
+    "开": "kai1", "手": "shou3", "十": "shi2", "用": "yong4",
+    "主": "zhu3", "行": "xing2", "方": "fang1", "又": "you4",
+    "如": "ru2", "前": "qian2", "所": "suo3", "本": "ben3",
+    "见": "jian4", "经": "jing1", "头": "tou2", "面": "mian4",
+    "公": "gong1", "同": "tong2", "三": "san1", "已": "yi3",
+    "老": "lao3", "从": "cong2", "动": "dong4", "两": "liang3",
+    "长": "chang2", "知": "zhi1", "民": "min2", "样": "yang4",
+    "现": "xian4", "分": "fen1", "将": "jiang1", "外": "wai4",
+    "但": "dan4", "身": "shen1", "些": "xie1", "与": "yu3",
+    "高": "gao1", "意": "yi4", "进": "jin4", "把": "ba3",
+    "法": "fa3", "此": "ci3", "实": "shi2", "回": "hui2",
+    "二": "er4", "理": "li3", "美": "mei3", "点": "dian3",
+    "月": "yue4", "明": "ming2", "其": "qi2", "种": "zhong3",
+    "声": "sheng1", "全": "quan2", "工": "gong1", "己": "ji3",
+    "话": "hua4", "儿": "er2", "者": "zhe3", "向": "xiang4",
+    "情": "qing2", "部": "bu4", "正": "zheng4", "名": "ming2",
+    "定": "ding4", "女": "nv3", "问": "wen4", "力": "li4",
+    "机": "ji1", "给": "gei3", "等": "deng3", "几": "ji3",
+    "很": "hen3", "业": "ye4", "最": "zui4", "间": "jian1",
+    "新": "xin1", "什": "shen2", "打": "da3", "便": "bian4",
+    "位": "wei4", "因": "yin1", "重": "zhong4", "被": "bei4",
+    "走": "zou3", "电": "dian4", "四": "si4", "第": "di4",
+    "门": "men2", "相": "xiang1", "次": "ci4", "东": "dong1",
+    "政": "zheng4", "海": "hai3", "口": "kou3", "使": "shi3",
+    "教": "jiao4", "西": "xi1", "再": "zai4", "平": "ping2",
+    "真": "zhen1", "听": "ting1", "世": "shi4", "期": "qi1",
+    "才": "cai2", "放": "fang4",
+    # 200-400
+    "五": "wu3", "六": "liu4", "七": "qi1", "八": "ba1",
+    "九": "jiu3", "百": "bai3", "千": "qian1", "万": "wan4",
+    "亿": "yi4", "零": "ling2", "号": "hao4", "字": "zi4",
+    "水": "shui3", "火": "huo3", "山": "shan1", "石": "shi2",
+    "田": "tian2", "土": "tu3", "木": "mu4", "林": "lin2",
+    "花": "hua1", "草": "cao3", "树": "shu4", "鸟": "niao3",
+    "鱼": "yu2", "马": "ma3", "牛": "niu2", "羊": "yang2",
+    "狗": "gou3", "猫": "mao1", "猪": "zhu1", "鸡": "ji1",
+    "虫": "chong2", "风": "feng1", "云": "yun2", "雨": "yu3",
+    "雪": "xue3", "雷": "lei2", "冰": "bing1", "河": "he2",
+    "湖": "hu2", "江": "jiang1", "海洋": "hai3 yang2",
+    "岛": "dao3", "沙": "sha1", "光": "guang1", "色": "se4",
+    "红": "hong2", "黄": "huang2", "蓝": "lan2", "绿": "lv4",
+    "白": "bai2", "黑": "hei1", "灰": "hui1", "紫": "zi3",
+    "春": "chun1", "夏": "xia4", "秋": "qiu1", "冬": "dong1",
+    "早": "zao3", "晚": "wan3", "午": "wu3", "夜": "ye4",
+    "今": "jin1", "昨": "zuo2", "星": "xing1", "周": "zhou1",
+    "男": "nan2", "父": "fu4", "母": "mu3", "兄": "xiong1",
+    "弟": "di4", "姐": "jie3", "妹": "mei4", "哥": "ge1",
+    "爸": "ba4", "妈": "ma1", "叔": "shu1", "爷": "ye2",
+    "奶": "nai3", "孙": "sun1", "友": "you3", "师": "shi1",
+    "生活": "sheng1 huo2", "吃": "chi1", "喝": "he1", "穿": "chuan1",
+    "住": "zhu4", "睡": "shui4", "坐": "zuo4", "站": "zhan4",
+    "躺": "tang3", "跑": "pao3", "跳": "tiao4", "飞": "fei1",
+    "游": "you2", "爬": "pa2", "唱": "chang4", "歌": "ge1",
+    "跳舞": "tiao4 wu3", "画": "hua4", "写": "xie3", "读": "du2",
+    "书": "shu1", "笔": "bi3", "纸": "zhi3", "桌": "zhuo1",
+    "椅": "yi3", "床": "chuang2", "窗": "chuang1", "房": "fang2",
+    "屋": "wu1", "楼": "lou2", "城": "cheng2", "市": "shi4",
+    "村": "cun1", "乡": "xiang1", "县": "xian4", "省": "sheng3",
+    "区": "qu1", "街": "jie1", "路": "lu4", "桥": "qiao2",
+    "车": "che1", "船": "chuan2", "票": "piao4", "钱": "qian2",
+    "买": "mai3", "卖": "mai4", "店": "dian4", "货": "huo4",
+    "价": "jia4", "元": "yuan2", "角落": "jiao3 luo4",
+    "饭": "fan4", "菜": "cai4", "肉": "rou4", "蛋": "dan4",
+    "面包": "mian4 bao1", "米": "mi3", "油": "you2", "盐": "yan2",
+    "糖": "tang2", "茶": "cha2", "酒": "jiu3", "奶茶": "nai3 cha2",
+    "果": "guo3", "瓜": "gua1", "豆": "dou4", "汤": "tang1",
+    # 400-600
+    "眼": "yan3", "耳": "er3", "鼻": "bi2", "嘴": "zui3",
+    "牙": "ya2", "舌": "she2", "脸": "lian3", "脚": "jiao3",
+    "腿": "tui3", "臂": "bi4", "指": "zhi3", "血": "xue4",
+    "骨": "gu3", "皮": "pi2", "毛": "mao2", "汗": "han4",
+    "病": "bing4", "药": "yao4", "疼": "teng2", "痛": "tong4",
+    "死": "si3", "活": "huo2", "岁": "sui4", "命": "ming4",
+    "爱": "ai4", "恨": "hen4", "怕": "pa4", "怒": "nu4",
+    "哭": "ku1", "笑": "xiao4", "喜": "xi3", "悲": "bei1",
+    "忧": "you1", "愁": "chou2", "惊": "jing1", "吓": "xia4",
+    "思": "si1", "念": "nian4", "忘": "wang4", "记": "ji4",
+    "懂": "dong3", "信": "xin4", "疑": "yi2", "猜": "cai1",
+    "希": "xi1", "望": "wang4", "梦": "meng4", "醒": "xing3",
+    "讲": "jiang3", "谈": "tan2", "告": "gao4", "诉": "su4",
+    "请": "qing3", "谢": "xie4", "答": "da2", "应": "ying1",
+    "叫": "jiao4", "喊": "han3", "骂": "ma4", "吵": "chao3",
+    "闹": "nao4", "静": "jing4", "安": "an1", "危": "wei1",
+    "险": "xian3", "救": "jiu4", "帮": "bang1", "助": "zhu4",
+    "送": "song4", "接": "jie1", "迎": "ying2", "别": "bie2",
+    "离": "li2", "留": "liu2", "停": "ting2", "等待": "deng3 dai4",
+    "找": "zhao3", "丢": "diu1", "拿": "na2", "放下": "fang4 xia4",
+    "抱": "bao4", "推": "tui1", "拉": "la1", "提": "ti2",
+    "抬": "tai2", "扔": "reng1", "捡": "jian3", "挂": "gua4",
+    "摸": "mo1", "碰": "peng4", "敲": "qiao1", "按": "an4",
+    "洗": "xi3", "擦": "ca1", "扫": "sao3", "切": "qie1",
+    "煮": "zhu3", "烧": "shao1", "烤": "kao3", "炒": "chao3",
+    "蒸": "zheng1", "拌": "ban4", "倒": "dao4", "装": "zhuang1",
+    "包": "bao1", "盖": "gai4", "关": "guan1", "锁": "suo3",
+    "修": "xiu1", "建": "jian4", "造": "zao4", "拆": "chai1",
+    "种树": "zhong4 shu4", "收": "shou1", "割": "ge1",
+    "养": "yang3", "喂": "wei4", "骑": "qi2", "开车": "kai1 che1",
+    "坐车": "zuo4 che1", "上班": "shang4 ban1", "下班": "xia4 ban1",
+    # 600-800: grammar/function & common content
+    "呢": "ne5", "吗": "ma5", "吧": "ba5", "啊": "a5",
+    "呀": "ya5", "哦": "o5", "嗯": "en1", "哈": "ha1",
+    "唉": "ai1", "喂喂": "wei2 wei2",
+    "谁": "shei2", "哪": "na3", "怎": "zen3", "啥": "sha2",
+    "咱": "zan2", "您": "nin2", "它": "ta1", "每": "mei3",
+    "各": "ge4", "另": "ling4", "某": "mou3", "任": "ren4",
+    "即": "ji2", "既": "ji4", "且": "qie3", "或": "huo4",
+    "若": "ruo4", "虽": "sui1", "尽": "jin3", "却": "que4",
+    "仍": "reng2", "曾": "ceng2", "刚": "gang1", "总": "zong3",
+    "常": "chang2", "永": "yong3", "久": "jiu3", "快": "kuai4",
+    "慢": "man4", "先": "xian1", "后来": "hou4 lai2",
+    "初": "chu1", "末": "mo4", "终": "zhong1", "始": "shi3",
+    "近": "jin4", "远": "yuan3", "深": "shen1", "浅": "qian3",
+    "宽": "kuan1", "窄": "zhai3", "厚": "hou4", "薄": "bao2",
+    "粗": "cu1", "细": "xi4", "硬": "ying4", "软": "ruan3",
+    "轻": "qing1", "干": "gan1", "湿": "shi1",
+    "冷": "leng3", "热": "re4", "温": "wen1", "凉": "liang2",
+    "暖": "nuan3", "亮": "liang4", "暗": "an4", "清": "qing1",
+    "浊": "zhuo2", "净": "jing4", "脏": "zang1", "乱": "luan4",
+    "整": "zheng3", "齐": "qi2", "弯": "wan1", "直": "zhi2",
+    "斜": "xie2", "圆": "yuan2", "尖": "jian1", "平坦": "ping2 tan3",
+    "满": "man3", "空": "kong1", "忙": "mang2", "闲": "xian2",
+    "穷": "qiong2", "富": "fu4", "贵": "gui4", "贱": "jian4",
+    "强": "qiang2", "弱": "ruo4", "胖": "pang4", "瘦": "shou4",
+    "美丽": "mei3 li4", "丑": "chou3", "聪": "cong1", "笨": "ben4",
+    "勇": "yong3", "敢": "gan3", "坏": "huai4", "假": "jia3",
+    "错": "cuo4", "偏": "pian1", "巧": "qiao3", "妙": "miao4",
+    "奇": "qi2", "怪": "guai4", "特": "te4", "普": "pu3",
+    "通": "tong1", "遍": "bian4", "共": "gong4", "单": "dan1",
+    "双": "shuang1", "对面": "dui4 mian4", "半": "ban4",
+    "整个": "zheng3 ge4", "许": "xu3", "约": "yue1", "互": "hu4",
+    # 800-1000: society / abstract
+    "内": "nei4", "央": "yang1", "边": "bian1", "旁": "pang2",
+    "左": "zuo3", "右": "you4", "南": "nan2", "北": "bei3",
+    "东方": "dong1 fang1", "底": "di3", "顶": "ding3",
+    "根": "gen1", "枝": "zhi1", "叶": "ye4", "果实": "guo3 shi2",
+    "条": "tiao2", "块": "kuai4", "张": "zhang1", "片": "pian4",
+    "层": "ceng2", "排": "pai2", "队": "dui4", "组": "zu3",
+    "群": "qun2", "批": "pi1", "套": "tao4", "份": "fen4",
+    "件": "jian4", "台": "tai2", "架": "jia4", "座": "zuo4",
+    "栋": "dong4", "间隔": "jian4 ge2",
+    "军": "jun1", "兵": "bing1", "战": "zhan4", "争": "zheng1",
+    "胜": "sheng4", "败": "bai4", "攻": "gong1", "守": "shou3",
+    "敌": "di2", "枪": "qiang1", "炮": "pao4", "刀": "dao1",
+    "剑": "jian4", "箭": "jian4", "盾": "dun4",
+    "王": "wang2", "皇": "huang2", "帝": "di4", "官": "guan1",
+    "臣": "chen2", "将领": "jiang4 ling3", "权": "quan2",
+    "令": "ling4", "律": "lv4", "规": "gui1", "制": "zhi4",
+    "度": "du4", "策": "ce4", "选": "xuan3", "举": "ju3",
+    "投": "tou2", "治": "zhi4", "管": "guan3", "领": "ling3",
+    "导": "dao3", "指挥": "zhi3 hui1", "组织": "zu3 zhi1",
+    "团": "tuan2", "党": "dang3", "派": "pai4", "界": "jie4",
+    "层次": "ceng2 ci4", "级": "ji2", "等级": "deng3 ji2",
+    "贫": "pin2", "农": "nong2", "商": "shang1", "士": "shi4",
+    "工人": "gong1 ren2", "厂": "chang3", "矿": "kuang4",
+    "钢": "gang1", "铁": "tie3", "铜": "tong2", "金": "jin1",
+    "银": "yin2", "煤": "mei2", "汽": "qi4",
+    "布": "bu4", "丝": "si1", "棉": "mian2", "麻": "ma2",
+    "衣": "yi1", "裤": "ku4", "裙": "qun2", "鞋": "xie2",
+    "帽": "mao4", "袜": "wa4", "镜": "jing4", "表": "biao3",
+    "钟": "zhong1", "灯": "deng1", "伞": "san3", "袋": "dai4",
+    "盒": "he2", "瓶": "ping2", "杯": "bei1", "碗": "wan3",
+    "盘": "pan2", "筷": "kuai4", "勺": "shao2", "壶": "hu2",
+    "锅": "guo1", "炉": "lu2", "柜": "gui4", "箱": "xiang1",
+    "篮": "lan2", "网": "wang3", "绳": "sheng2", "线": "xian4",
+    "针": "zhen1", "剪": "jian3", "斧": "fu3", "锤": "chui2",
+    "钉": "ding1", "梯": "ti1", "轮": "lun2", "机器": "ji1 qi4",
+    # 1000+: verbs/abstracts rounding out running text
+    "变": "bian4", "化": "hua4", "增": "zeng1", "减": "jian3",
+    "加": "jia1", "除": "chu2", "乘": "cheng2", "算": "suan4",
+    "计": "ji4", "测": "ce4", "验": "yan4", "试": "shi4",
+    "研": "yan2", "究": "jiu1", "查": "cha2", "审": "shen3",
+    "观": "guan1", "察": "cha2", "视": "shi4", "望远": "wang4 yuan3",
+    "显": "xian3", "示": "shi4", "证": "zheng4", "据": "ju4",
+    "论": "lun4", "断": "duan4", "判": "pan4", "析": "xi1",
+    "较": "jiao4", "比": "bi3", "例": "li4", "率": "lv4",
+    "均": "jun1", "总共": "zong3 gong4", "值": "zhi2",
+    "质": "zhi4", "态": "tai4", "状": "zhuang4", "形": "xing2",
+    "式": "shi4", "型": "xing2", "类": "lei4", "项": "xiang4",
+    "系": "xi4", "统": "tong3", "构": "gou4", "素": "su4",
+    "质料": "zhi4 liao4", "料": "liao4", "源": "yuan2",
+    "能源": "neng2 yuan2", "核": "he2", "原": "yuan2",
+    "基": "ji1", "础": "chu3", "因素": "yin1 su4",
+    "效": "xiao4", "益": "yi4", "利": "li4", "害": "hai4",
+    "损": "sun3", "失": "shi1", "获": "huo4", "赢": "ying2",
+    "输": "shu1", "付": "fu4", "费": "fei4", "税": "shui4",
+    "账": "zhang4", "债": "zhai4", "租": "zu1", "借": "jie4",
+    "赔": "pei2", "赚": "zhuan4", "存": "cun2", "取": "qu3",
+    "换": "huan4", "交": "jiao1", "易": "yi4", "贸": "mao4",
+    "运": "yun4", "输送": "shu1 song4", "递": "di4", "寄": "ji4",
+    "邮": "you2", "航": "hang2", "港": "gang3", "站台": "zhan4 tai2",
+    "铺": "pu4", "厅": "ting1", "馆": "guan3", "院": "yuan4",
+    "室": "shi4", "堂": "tang2", "庙": "miao4", "塔": "ta3",
+    "宫": "gong1", "园": "yuan2", "场": "chang3", "所在": "suo3 zai4",
+    "址": "zhi3", "境": "jing4", "域": "yu4", "洲": "zhou1",
+    "陆": "lu4", "岸": "an4", "滩": "tan1", "谷": "gu3",
+    "坡": "po1", "峰": "feng1", "岭": "ling3", "洞": "dong4",
+    "泉": "quan2", "井": "jing3", "池": "chi2", "沟": "gou1",
+    "渠": "qu2", "坝": "ba4", "田野": "tian2 ye3",
+}
+
+_CMN_MAX = max(len(k) for k in list(CMN_WORDS) + ["一"])
+
+
+def _is_hanzi(ch: str) -> bool:
+    cp = ord(ch)
+    return (0x4E00 <= cp <= 0x9FFF or 0x3400 <= cp <= 0x4DBF
+            or cp in (0x3007,))  # 〇
+
+
+def cmn_word_to_ipa(w: str) -> str:
+    """Hanzi string -> Mandarin IPA.  Longest-match over CMN_WORDS,
+    then CMN_CHARS; unknown hanzi are dropped (needs a bigger reading
+    dictionary — honest, like ja kanji)."""
+    syls: List[str] = []
+    hanzi_for: List[str] = []
+    i, n = 0, len(w)
+    while i < n:
+        matched = False
+        for ln in range(min(_CMN_MAX, n - i), 1, -1):
+            seg = w[i:i + ln]
+            py = CMN_WORDS.get(seg)
+            if py:
+                parts = py.split()
+                syls.extend(parts)
+                hanzi_for.extend(seg if len(seg) == len(parts)
+                                 else ["·"] * len(parts))
+                i += ln
+                matched = True
+                break
+        if matched:
+            continue
+        ch = w[i]
+        py = CMN_WORDS.get(ch) or CMN_CHARS.get(ch)
+        if ch == "〇":
+            py = "ling2"
+        if py:
+            parts = py.split()
+            syls.extend(parts)
+            hanzi_for.extend([ch] * len(parts))
+        # unknown hanzi / non-hanzi: dropped
+        i += 1
+    syls = _cmn_sandhi(syls, "".join(hanzi_for))
+    return "".join(pinyin_syllable_to_ipa(s) for s in syls)
+
+
+# --------------------------------------------------------------------- #
+# Cantonese (yue) — traditional-script common characters with jyutping.
+# --------------------------------------------------------------------- #
+YUE_WORDS: Dict[str, str] = {
+    "香港": "hoeng1 gong2", "廣東": "gwong2 dung1",
+    "廣東話": "gwong2 dung1 waa2", "粵語": "jyut6 jyu5",
+    "普通話": "pou2 tung1 waa2", "中國": "zung1 gwok3",
+    "中文": "zung1 man4", "英文": "jing1 man4",
+    "你好": "nei5 hou2", "早晨": "zou2 san4", "多謝": "do1 ze6",
+    "唔該": "m4 goi1", "再見": "zoi3 gin3", "對唔住": "deoi3 m4 zyu6",
+    "而家": "ji4 gaa1", "今日": "gam1 jat6", "聽日": "ting1 jat6",
+    "琴日": "kam4 jat6", "依家": "ji1 gaa1",
+    "乜嘢": "mat1 je5", "點解": "dim2 gaai2", "點樣": "dim2 joeng2",
+    "邊個": "bin1 go3", "邊度": "bin1 dou6", "幾多": "gei2 do1",
+    "我哋": "ngo5 dei6", "你哋": "nei5 dei6", "佢哋": "keoi5 dei6",
+    "先生": "sin1 saang1", "小姐": "siu2 ze2", "朋友": "pang4 jau5",
+    "飲茶": "jam2 caa4", "食飯": "sik6 faan6", "飲水": "jam2 seoi2",
+    "返工": "faan1 gung1", "放工": "fong3 gung1",
+    "返學": "faan1 hok6", "放學": "fong3 hok6",
+    "鍾意": "zung1 ji3", "唔使": "m4 sai2", "唔好": "m4 hou2",
+    "好似": "hou2 ci5", "一齊": "jat1 cai4", "而且": "ji4 ce2",
+    "但係": "daan6 hai6", "因為": "jan1 wai6", "所以": "so2 ji5",
+    "如果": "jyu4 gwo2", "已經": "ji5 ging1", "仲有": "zung6 jau5",
+    "時間": "si4 gaan3", "地方": "dei6 fong1", "嘢食": "je5 sik6",
+    "巴士": "baa1 si2", "的士": "dik1 si2", "地鐵": "dei6 tit3",
+    "火車": "fo2 ce1", "飛機": "fei1 gei1", "電話": "din6 waa2",
+    "電腦": "din6 nou5", "電視": "din6 si6", "手機": "sau2 gei1",
+    "屋企": "uk1 kei2", "學校": "hok6 haau6", "老師": "lou5 si1",
+    "學生": "hok6 saang1", "醫生": "ji1 sang1", "醫院": "ji1 jyun2",
+    "警察": "ging2 caat3", "公司": "gung1 si1", "銀行": "ngan4 hong4",
+    "錢": "cin2", "問題": "man6 tai4", "意思": "ji3 si1",
+    "世界": "sai3 gaai3", "國家": "gwok3 gaa1",
+}
+
+YUE_CHARS: Dict[str, str] = {
+    "一": "jat1", "二": "ji6", "三": "saam1", "四": "sei3",
+    "五": "ng5", "六": "luk6", "七": "cat1", "八": "baat3",
+    "九": "gau2", "十": "sap6", "百": "baak3", "千": "cin1",
+    "萬": "maan6", "億": "jik1", "零": "ling4",
+    "嘅": "ge3", "係": "hai6", "唔": "m4", "咗": "zo2",
+    "喺": "hai2", "嗰": "go2", "呢": "ni1", "咁": "gam3",
+    "啲": "di1", "嘢": "je5", "冇": "mou5", "佢": "keoi5",
+    "哋": "dei6", "噉": "gam2", "啦": "laa1", "喇": "laa3",
+    "囉": "lo1", "㗎": "gaa3", "呀": "aa3", "咩": "me1",
+    "我": "ngo5", "你": "nei5", "人": "jan4", "大": "daai6",
+    "小": "siu2", "中": "zung1", "上": "soeng6", "下": "haa6",
+    "出": "ceot1", "入": "jap6", "來": "loi4", "去": "heoi3",
+    "返": "faan1", "行": "hang4", "走": "zau2", "企": "kei5",
+    "坐": "co5", "食": "sik6", "飲": "jam2", "講": "gong2",
+    "話": "waa6", "睇": "tai2", "聽": "teng1", "寫": "se2",
+    "讀": "duk6", "學": "hok6", "教": "gaau3", "買": "maai5",
+    "賣": "maai6", "畀": "bei2", "攞": "lo2", "搵": "wan2",
+    "做": "zou6", "整": "zing2", "開": "hoi1", "閂": "saan1",
+    "著": "zoek3", "住": "zyu6", "瞓": "fan3", "起": "hei2",
+    "想": "soeng2", "要": "jiu3", "可": "ho2", "以": "ji5",
+    "會": "wui5", "能": "nang4", "得": "dak1", "好": "hou2",
+    "靚": "leng3", "醜": "cau2", "快": "faai3", "慢": "maan6",
+    "新": "san1", "舊": "gau6", "多": "do1", "少": "siu2",
+    "長": "coeng4", "短": "dyun2", "高": "gou1", "矮": "ai2",
+    "肥": "fei4", "瘦": "sau3", "凍": "dung3", "熱": "jit6",
+    "暖": "nyun5", "乾": "gon1", "濕": "sap1", "平": "peng4",
+    "貴": "gwai3", "遠": "jyun5", "近": "kan5", "早": "zou2",
+    "晏": "aan3", "夜": "je6", "日": "jat6", "月": "jyut6",
+    "年": "nin4", "時": "si4", "分": "fan1", "秒": "miu5",
+    "水": "seoi2", "火": "fo2", "山": "saan1", "海": "hoi2",
+    "天": "tin1", "地": "dei6", "風": "fung1", "雨": "jyu5",
+    "雲": "wan4", "雪": "syut3", "花": "faa1", "草": "cou2",
+    "樹": "syu6", "石": "sek6", "金": "gam1", "銀": "ngan2",
+    "屋": "uk1", "門": "mun4", "窗": "coeng1", "房": "fong2",
+    "車": "ce1", "船": "syun4", "路": "lou6", "橋": "kiu4",
+    "街": "gaai1", "市": "si5", "城": "sing4", "國": "gwok3",
+    "家": "gaa1", "爸": "baa4", "媽": "maa1", "仔": "zai2",
+    "女": "neoi5", "男": "naam4", "哥": "go1", "姐": "ze2",
+    "妹": "mui6", "弟": "dai6", "公": "gung1", "婆": "po4",
+    "頭": "tau4", "手": "sau2", "腳": "goek3", "眼": "ngaan5",
+    "耳": "ji5", "口": "hau2", "鼻": "bei6", "心": "sam1",
+    "肚": "tou5", "面": "min6", "髮": "faat3", "牙": "ngaa4",
+    "飯": "faan6", "麵": "min6", "粥": "zuk1", "包": "baau1",
+    "蛋": "daan2", "肉": "juk6", "魚": "jyu2", "菜": "coi3",
+    "茶": "caa4", "奶": "naai5", "糖": "tong4", "鹽": "jim4",
+    "油": "jau4", "酒": "zau2", "生": "sang1", "死": "sei2",
+    "病": "beng6", "痛": "tung3", "攰": "gui6", "餓": "ngo6",
+    "飽": "baau2", "凍飲": "dung3 jam2",
+    "紅": "hung4", "黃": "wong4", "藍": "laam4", "綠": "luk6",
+    "白": "baak6", "黑": "hak1", "青": "ceng1", "紫": "zi2",
+    "東": "dung1", "南": "naam4", "西": "sai1", "北": "bak1",
+    "左": "zo2", "右": "jau6", "前": "cin4", "後": "hau6",
+    "內": "noi6", "外": "ngoi6", "邊": "bin1", "度": "dou6",
+    "點": "dim2", "樣": "joeng2", "個": "go3", "隻": "zek3",
+    "條": "tiu4", "張": "zoeng1", "間": "gaan1", "部": "bou6",
+    "架": "gaa3", "本": "bun2", "枝": "zi1", "杯": "bui1",
+    "碗": "wun2", "碟": "dip6", "樽": "zeon1", "袋": "doi2",
+    "同": "tung4", "埋": "maai4", "又": "jau6", "都": "dou1",
+    "仲": "zung6", "先": "sin1", "就": "zau6", "即": "zik1",
+    "真": "zan1", "假": "gaa2", "啱": "ngaam1", "錯": "co3",
+    "知": "zi1", "識": "sik1", "明": "ming4", "記": "gei3",
+    "忘": "mong4", "愛": "oi3", "惜": "sik1", "怕": "paa3",
+    "驚": "geng1", "喊": "haam3", "笑": "siu3", "嬲": "nau1",
+    "開心": "hoi1 sam1", "攪": "gaau2", "幫": "bong1",
+    "送": "sung3", "等": "dang2", "停": "ting4", "轉": "zyun3",
+    "過": "gwo3", "落": "lok6", "升": "sing1", "跌": "dit3",
+    "企起": "kei5 hei2", "慳": "haan1", "使": "sai2",
+    "舖": "pou3", "廁": "ci3", "廚": "cyu4", "檯": "toi2",
+    "櫈": "dang3", "牀": "cong4", "燈": "dang1", "波": "bo1",
+    "戲": "hei3", "相": "soeng2", "畫": "waa2", "書": "syu1",
+    "筆": "bat1", "紙": "zi2", "字": "zi6", "文": "man4",
+    "語": "jyu5", "音": "jam1", "聲": "seng1", "歌": "go1",
+}
+
+_YUE_MAX = max(len(k) for k in list(YUE_WORDS) + ["一"])
+
+
+def yue_word_to_ipa(w: str) -> str:
+    """Hanzi (traditional) -> Cantonese IPA via jyutping readings."""
+    out: List[str] = []
+    i, n = 0, len(w)
+    while i < n:
+        matched = False
+        for ln in range(min(_YUE_MAX, n - i), 1, -1):
+            seg = w[i:i + ln]
+            jp = YUE_WORDS.get(seg)
+            if jp:
+                out.extend(jp.split())
+                i += ln
+                matched = True
+                break
+        if matched:
+            continue
+        ch = w[i]
+        jp = YUE_WORDS.get(ch) or YUE_CHARS.get(ch)
+        if ch == "〇":
+            jp = "ling4"
+        if jp:
+            out.extend(jp.split())
+        i += 1
+    return "".join(jyutping_syllable_to_ipa(s) for s in out)

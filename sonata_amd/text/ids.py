@@ -39,6 +39,8 @@ _IPA_SYMBOLS = (
     "ɢɸχẽĩũ",
     # conlang batch (qya/sjn): voiceless w, combining ring (r̥)
     "ʍ̥",
+    # Chinese batch (cmn/yue): Chao tone letters + Cantonese ɵ
+    "˥˦˧˨˩ɵ",
 )
 
 

@@ -225,9 +225,17 @@ _DIGITS.update({
     "piqd": "pagh wa' cha' wej loS vagh jav Soch chorgh Hut".split(),
 })
 
+_DIGITS.update({
+    # Chinese batch: hanzi digit names (shared by simplified and
+    # traditional — 0-9 are the same characters), read by g2p_zh
+    "cmn": list("零一二三四五六七八九"),
+    "yue": list("零一二三四五六七八九"),
+})
+
 # orthography aliases share digit tables
 for _alias, _src in (("nb", "no"), ("nn", "no"), ("sr", "hr"),
-                     ("bs", "hr"), ("ms", "id")):
+                     ("bs", "hr"), ("ms", "id"), ("zh", "cmn"),
+                     ("hak", "yue")):
     _DIGITS[_alias] = _DIGITS[_src]
 
 _NUM_RE = re.compile(r"\d[\d,]*(?:\.\d+)?")
@@ -255,7 +263,9 @@ def _expand_number_en(tok: str) -> str:
 _DOTTED_ACRO_RE = re.compile(r"\b(?:[A-Z]\.){2,}")
 
 
-_TIME_RE = re.compile(r"\b(\d{1,2}):(\d{2})\b")
+# (?<!\d) instead of \b: CJK characters are word chars, so \b fails
+# to match between 是 and 14:30 in unspaced Chinese/Japanese text
+_TIME_RE = re.compile(r"(?<![\d:])(\d{1,2}):(\d{2})(?![\d:])")
 
 
 def _time_words(m: re.Match) -> str:
@@ -523,6 +533,9 @@ from .numbers3 import CARDINALS3, DECIMAL_WORDS3  # noqa: E402
 _CARDINALS.update(CARDINALS3)
 _DECIMAL_WORD.update(DECIMAL_WORDS3)
 _DOT_DECIMAL = dict(DOT_DECIMAL2)
+# Chinese uses comma grouping + dot decimals (12.5 -> 十二 点 五)
+_DOT_DECIMAL.update({"cmn": "点", "zh": "点", "yue": "點",
+                     "hak": "點"})
 _GROUP_COMMA_RE = re.compile(r"(?<=\d),(?=\d\d\d)")
 _DEC_DOT_RE = re.compile(r"(\d+)\.(\d+)")
 
@@ -592,7 +605,11 @@ _CURRENCY_WORDS = {
     "ja": {"¥": "えん", "円": "えん", "$": "ドル"},
     "ko": {"₩": "원", "$": "달러"},
     "ar": {"$": "دولار", "€": "يورو"},
+    "cmn": {"¥": "元", "$": "美元", "€": "欧元", "元": "元"},
+    "yue": {"$": "元", "¥": "元"},
 }
+_CURRENCY_WORDS["zh"] = _CURRENCY_WORDS["cmn"]
+_CURRENCY_WORDS["hak"] = _CURRENCY_WORDS["yue"]
 # the word between hours and minutes in clock times (14:30); only
 # languages where "H <word> MM" is a natural reading — Turkish (saat
 # precedes), Polish (ordinal hours) and Korean (native-numeral hours)
@@ -601,6 +618,7 @@ _TIME_WORDS = {
     "de": "Uhr", "fr": "heures", "es": "horas", "it": "e",
     "pt": "horas", "nl": "uur", "sv": "och", "ru": "часов",
     "uk": "годин", "fi": "ja", "ja": "じ",
+    "cmn": "点", "zh": "点", "yue": "點", "hak": "點",
 }
 
 
@@ -661,6 +679,9 @@ def normalize(text: str, language: str) -> str:
     base = language.lower().replace("_", "-").split("-")[0]
     if base == "en":
         return normalize_en(text)
+    if base in ("cmn", "zh", "yue", "hak"):
+        # Chinese percent is a PREFIX construction: 50% -> 百分之五十
+        text = _PERCENT_RE.sub(lambda m: "百分之" + m.group(1), text)
     pw = _PERCENT_WORDS.get(base)
     if pw is not None:
         text = _PERCENT_RE.sub(lambda m: m.group(1) + " " + pw, text)

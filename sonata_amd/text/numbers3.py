@@ -439,3 +439,85 @@ def num_to_words_hi(n: int) -> str:
 
 CARDINALS3["hi"] = num_to_words_hi
 DECIMAL_WORDS3["hi"] = "दशमलव"
+
+
+# --------------------------------------------------------------------- #
+# Chinese (cmn/yue): 万-based grouping with 零-gap insertion; the
+# output is hanzi, which the g2p_zh reading dictionaries then read
+# (so the same grammar serves Mandarin and Cantonese — yue gets the
+# traditional forms 萬/億/負).
+# --------------------------------------------------------------------- #
+_ZH_DIGITS_S = "零一二三四五六七八九"
+_ZH_DIGITS_T = "零一二三四五六七八九"  # digits are shared
+
+
+def _zh_under_10000(n: int, leading: bool) -> str:
+    out = []
+    need_zero = False
+    for div, name in ((1000, "千"), (100, "百"), (10, "十")):
+        d, n = divmod(n, div)
+        if d:
+            if need_zero:
+                out.append("零")
+                need_zero = False
+            out.append(_ZH_DIGITS_S[d] + name)
+        elif out:
+            need_zero = True
+    if n:
+        if need_zero:
+            out.append("零")
+        out.append(_ZH_DIGITS_S[n])
+    s = "".join(out)
+    # a leading 一十 reads 十 (10, 15 … but 110 keeps 一百一十);
+    # only in the most-significant group (20010 = 二万零一十)
+    if leading and s.startswith("一十"):
+        s = s[1:]
+    return s
+
+
+def _num_to_words_zh(n: int, trad: bool) -> str:
+    wan, yi, neg = ("萬", "億", "負") if trad else ("万", "亿", "负")
+    if n < 0:
+        return neg + _num_to_words_zh(-n, trad)
+    if n == 0:
+        return "零"
+    parts = []
+    groups = []  # (value, suffix) most-significant first
+    g, n = divmod(n, 10 ** 8)
+    if g:
+        groups.append((g, yi))
+    g, n = divmod(n, 10 ** 4)
+    if g:
+        groups.append((g, wan))
+    if n or not groups:
+        groups.append((n, ""))
+    prev_had_gap = False
+    for i, (val, suf) in enumerate(groups):
+        if val == 0:
+            prev_had_gap = True
+            continue
+        if i > 0 and (prev_had_gap or val < 1000):
+            # 100005 -> 十万零五 (gap between groups reads 零)
+            parts.append("零")
+        parts.append(_zh_under_10000(val, i == 0) + suf)
+        prev_had_gap = False
+    s = "".join(parts)
+    return s.strip("零") or "零"
+
+
+def num_to_words_cmn(n: int) -> str:
+    return _num_to_words_zh(n, trad=False)
+
+
+def num_to_words_yue(n: int) -> str:
+    return _num_to_words_zh(n, trad=True)
+
+
+CARDINALS3["cmn"] = num_to_words_cmn
+CARDINALS3["yue"] = num_to_words_yue
+CARDINALS3["zh"] = num_to_words_cmn
+CARDINALS3["hak"] = num_to_words_yue
+DECIMAL_WORDS3["cmn"] = "点"
+DECIMAL_WORDS3["zh"] = "点"
+DECIMAL_WORDS3["yue"] = "點"
+DECIMAL_WORDS3["hak"] = "點"

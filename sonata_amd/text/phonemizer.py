@@ -978,6 +978,18 @@ def _get_g2p(voice: str) -> RuleG2P:
         # unlisted kanji drop) + 々 iteration mark
         g = RuleG2P({}, letters="぀-ヿ一-鿿々", stress=False)
         g._apply_rules = ja_word_to_ipa
+    elif base in ("cmn", "zh", "yue", "hak"):
+        # Chinese (g2p_zh.py): hanzi -> dictionary readings (word dict
+        # disambiguates polyphones, then per-character frequency core)
+        # -> pinyin/jyutping -> IPA with Chao tone letters.  hak
+        # (Hakka) has no reading dictionary of its own yet and uses
+        # the Cantonese one as the nearest Yue-adjacent approximation,
+        # stated in docs/LANGUAGES.md.
+        from .g2p_zh import cmn_word_to_ipa, yue_word_to_ipa
+
+        g = RuleG2P({}, letters="一-鿿㐀-䶿〇", stress=False)
+        g._apply_rules = (cmn_word_to_ipa if base in ("cmn", "zh")
+                          else yue_word_to_ipa)
     elif base in _BATCH3:
         from .g2p_tables3 import (LETTERS3, LEXICONS3, POSTPROCESS3,
                                   PREPROCESS3, STRESS3, TABLES3)
@@ -1085,7 +1097,11 @@ def available_languages() -> List[str]:
             "nci", "om", "tn", "pap", "ia", "io", "lfn", "jbo", "tk",
             "lb", "kl", "ga", "grc", "tt", "ba", "cv",
             "an", "ku", "gd", "quc", "sd", "nog", "smj", "bpy", "shn",
-            "qya", "sjn", "piqd"]
+            "qya", "sjn", "piqd",
+            # Chinese batch (g2p_zh.py): Mandarin + Cantonese hanzi
+            # reading dictionaries (hak approximated via yue, see
+            # docs/LANGUAGES.md)
+            "cmn", "zh", "yue", "hak"]
 
 
 # script-native sentence/clause punctuation -> ASCII so the splitter

@@ -1,0 +1,127 @@
+"""Chinese G2P (g2p_zh.py): Mandarin + Cantonese reading dictionaries,
+pinyin/jyutping -> IPA, tone sandhi, number grammar.
+
+Reference bar: espeak-ng zh/zhy dictionaries
+(deps/dev/espeak-ng-data/{zh,zhy}_dict via
+crates/text/espeak-phonemizer/src/lib.rs:65-156).
+"""
+
+from sonata_amd.text.g2p_zh import (cmn_word_to_ipa, jyutping_syllable_to_ipa,
+                                    pinyin_syllable_to_ipa, yue_word_to_ipa)
+from sonata_amd.text.phonemizer import text_to_phonemes
+
+
+def test_pinyin_syllables():
+    assert pinyin_syllable_to_ipa("zhong1") == "ʈʂʊŋ˥"
+    assert pinyin_syllable_to_ipa("guo2") == "kwo˧˥"
+    assert pinyin_syllable_to_ipa("ni3") == "ni˨˩˦"
+    assert pinyin_syllable_to_ipa("shi4") == "ʂɨ˥˩"
+    assert pinyin_syllable_to_ipa("zi3") == "tsɨ˨˩˦"   # apical vowel
+    assert pinyin_syllable_to_ipa("lv4") == "ly˥˩"     # ü after l
+    assert pinyin_syllable_to_ipa("xu3") == "ɕy˨˩˦"    # j/q/x u = ü
+    assert pinyin_syllable_to_ipa("yuan2") == "ɥɛn˧˥"
+    assert pinyin_syllable_to_ipa("er2") == "ɚ˧˥"
+    assert pinyin_syllable_to_ipa("de5") == "tɤ"       # neutral: no tone
+    assert pinyin_syllable_to_ipa("qqq9") == ""        # junk -> empty
+
+
+def test_cmn_words_and_chars():
+    assert cmn_word_to_ipa("中国") == "ʈʂʊŋ˥kwo˧˥"
+    assert cmn_word_to_ipa("人") == "ʐən˧˥"
+    # unknown hanzi drop (honest) but known neighbours survive
+    out = cmn_word_to_ipa("人鬱人")
+    assert out == "ʐən˧˥ʐən˧˥"
+
+
+def test_cmn_polyphones():
+    # 行 háng in 银行 but xíng in 行动
+    assert cmn_word_to_ipa("银行").endswith("xaŋ˧˥")
+    assert cmn_word_to_ipa("行动").startswith("ɕiŋ˧˥")
+    # 乐 yuè in 音乐, lè in 快乐
+    assert cmn_word_to_ipa("音乐").endswith("ɥɛ˥˩")
+    assert cmn_word_to_ipa("快乐").endswith("lɤ˥˩")
+    # 重 chóng in 重新, zhòng in 重要
+    assert cmn_word_to_ipa("重新").startswith("ʈʂʰʊŋ˧˥")
+    assert cmn_word_to_ipa("重要").startswith("ʈʂʊŋ˥˩")
+
+
+def test_cmn_tone_sandhi():
+    # 3-3 -> 2-3: 你好 ni3 hao3 -> ni2 hao3
+    assert cmn_word_to_ipa("你好") == "ni˧˥xau˨˩˦"
+    # 不 + tone4 -> bu2: 不是
+    assert cmn_word_to_ipa("不是") == "pu˧˥ʂɨ˥˩"
+    # 不 + tone1 stays bu4: 不说
+    assert cmn_word_to_ipa("不说") == "pu˥˩ʂwo˥"
+    # 一 + tone4 -> yi2: 一个; 一 + tone1 -> yi4: 一天
+    assert cmn_word_to_ipa("一个") == "i˧˥kɤ˥˩"
+    assert cmn_word_to_ipa("一天") == "i˥˩tʰjɛn˥"
+
+
+def test_jyutping_syllables():
+    assert jyutping_syllable_to_ipa("gwong2") == "kʷɔːŋ˧˥"
+    assert jyutping_syllable_to_ipa("dung1") == "tʊŋ˥"
+    assert jyutping_syllable_to_ipa("sik6") == "sɪk˨"
+    assert jyutping_syllable_to_ipa("nei5") == "nei˩˧"
+    assert jyutping_syllable_to_ipa("m4") == "m˨˩"       # syllabic nasal
+    assert jyutping_syllable_to_ipa("ng5") == "ŋ˩˧"
+    assert jyutping_syllable_to_ipa("heoi3") == "hɵy˧"
+    assert jyutping_syllable_to_ipa("zyu6") == "tsyː˨"
+
+
+def test_yue_words():
+    assert yue_word_to_ipa("香港") == "hœːŋ˥kɔːŋ˧˥"
+    assert yue_word_to_ipa("唔該") == "m˨˩kɔːi˥"
+    assert yue_word_to_ipa("我哋") == "ŋɔː˩˧tei˨"
+
+
+def test_zh_numbers_and_normalize():
+    out = text_to_phonemes("我有3个苹果。", "cmn")[0]
+    assert "san˥" in out                       # 3 -> 三
+    out = text_to_phonemes("50%", "cmn")[0]
+    assert out.startswith("pai˨˩˦fən˥ʈʂɨ˥")   # 百分之五十 (prefix!)
+    out = text_to_phonemes("现在是14:30。", "cmn")[0]
+    assert "tjɛn˨˩˦" in out                    # 14 点 30
+    out = text_to_phonemes("这本书要¥25。", "zh")[0]
+    assert out.rstrip(".").endswith("ɥɛn˧˥")   # 25 元
+    out = text_to_phonemes("12.5", "cmn")[0]
+    assert "tjɛn˨˩˦" in out                    # 十二 点 五
+    # Cantonese gets traditional forms (萬/點)
+    out = text_to_phonemes("15000", "yue")[0]
+    assert "maːn˨" in out                      # 一萬五千
+
+
+def test_zh_sentences_end_to_end():
+    for lang, txt in (("cmn", "你好，世界！"), ("yue", "你好，世界！"),
+                      ("zh", "我们都是学生。"), ("hak", "食飯。")):
+        sents = text_to_phonemes(txt, lang)
+        assert sents and sents[0], (lang, txt)
+
+
+def test_zh_symbols_encodable():
+    """Every IPA char the Chinese engines can emit is in the voice
+    symbol table (tone letters were appended to ids.py)."""
+    from sonata_amd.text.g2p_zh import (CMN_CHARS, CMN_WORDS, YUE_CHARS,
+                                        YUE_WORDS)
+    from sonata_amd.text.ids import default_phoneme_id_map
+
+    idm = default_phoneme_id_map()
+    for d, fn in ((CMN_CHARS, pinyin_syllable_to_ipa),
+                  (CMN_WORDS, pinyin_syllable_to_ipa),
+                  (YUE_CHARS, jyutping_syllable_to_ipa),
+                  (YUE_WORDS, jyutping_syllable_to_ipa)):
+        for word, reading in d.items():
+            for syl in reading.split():
+                ipa = fn(syl)
+                assert ipa, (word, syl)  # every reading converts
+                for ch in ipa:
+                    assert ch in idm, (word, syl, ch)
+
+
+def test_zh_dictionary_size_floor():
+    """Coverage regression guard: the Mandarin reading dictionary
+    stays above its shipped size."""
+    from sonata_amd.text.g2p_zh import CMN_CHARS, CMN_WORDS, YUE_CHARS
+
+    assert len(CMN_CHARS) >= 700
+    assert len(CMN_WORDS) >= 300
+    assert len(YUE_CHARS) >= 250

@@ -126,6 +126,10 @@ _ALL_LANG_TEXTS = {
     "qya": "Elen síla lúmenn omentielvo.",
     "sjn": "Mae govannen, mellon nîn.",
     "piqd": "tlhIngan Hol vIjatlh.",
+    "cmn": "你好，世界！我们都是朋友。",
+    "zh": "今天天气很好。",
+    "yue": "你好，我哋今日去香港食飯。",
+    "hak": "多謝你。",
 }
 
 
