@@ -577,9 +577,210 @@ CMN_CHARS: Dict[str, str] = {
     "坡": "po1", "峰": "feng1", "岭": "ling3", "洞": "dong4",
     "泉": "quan2", "井": "jing3", "池": "chi2", "沟": "gou1",
     "渠": "qu2", "坝": "ba4", "田野": "tian2 ye3",
+    # batch 2: rounding out the top ~1500 of running text
+    "习": "xi2", "产": "chan3", "体": "ti3", "保": "bao3",
+    "入": "ru4", "准": "zhun3", "功": "gong1", "务": "wu4",
+    "医": "yi1", "受": "shou4", "司": "si1", "合": "he2",
+    "味": "wei4", "品": "pin3", "售": "shou4", "善": "shan4",
+    "图": "tu2", "圾": "ji1", "垃": "la1", "备": "bei4",
+    "复": "fu4", "太": "tai4", "奔": "ben1", "孩": "hai2",
+    "宣": "xuan1", "少": "shao3", "府": "fu3", "技": "ji4",
+    "护": "hu4", "捷": "jie2", "改": "gai3", "施": "shi1",
+    "晒": "shai4", "晨": "chen2", "智": "zhi4", "朋": "peng2",
+    "服": "fu2", "术": "shu4", "格": "ge2", "欢": "huan1",
+    "步": "bu4", "流": "liu2", "消": "xiao1", "澡": "zao3",
+    "炼": "lian4", "玩": "wan2", "环": "huan2", "班": "ban1",
+    "球": "qiu2", "畅": "chang4", "疗": "liao2", "科": "ke1",
+    "考": "kao3", "耍": "shua3", "聊": "liao2", "育": "yu4",
+    "节": "jie2", "芯": "xin1", "认": "ren4", "设": "she4",
+    "该": "gai1", "语": "yu3", "课": "ke4", "越": "yue4",
+    "转": "zhuan3", "速": "su4", "锻": "duan4", "阳": "yang2",
+    "需": "xu1", "青": "qing1", "非": "fei1", "音": "yin1",
+    "餐": "can1", "鲜": "xian1",
+    "握": "wo4", "言": "yan2", "论文": "lun4 wen2",
+    "报": "bao4", "告诉": "gao4 su5", "志": "zhi4", "愿": "yuan4",
+    "望着": "wang4 zhe5", "希望": "xi1 wang4", "感": "gan3",
+    "觉": "jue2", "意见": "yi4 jian4", "议": "yi4", "评": "ping2",
+    "批评": "pi1 ping2", "赞": "zan4", "支": "zhi1", "持": "chi2",
+    "反": "fan3", "对于": "dui4 yu2", "按照": "an4 zhao4",
+    "根据": "gen1 ju4", "由": "you2", "至": "zhi4", "于是": "yu2 shi4",
+    "并": "bing4", "及": "ji2", "以及": "yi3 ji2", "关于": "guan1 yu2",
+    "其中": "qi2 zhong1", "其他": "qi2 ta1", "其实": "qi2 shi2",
+    "甚": "shen4", "至于": "zhi4 yu2", "例如": "li4 ru2",
+    "比如": "bi3 ru2", "譬": "pi4", "似": "si4", "般": "ban1",
+    "像": "xiang4", "如同": "ru2 tong2", "仿": "fang3",
+    "佛": "fo2", "仿佛": "fang3 fu2",
+    "极": "ji2", "更": "geng4", "挺": "ting3", "稍": "shao1",
+    "略": "lve4", "颇": "po1", "相当": "xiang1 dang1",
+    "十分重": "shi2 fen1 zhong4",
+    "必": "bi4", "须": "xu1", "必须": "bi4 xu1", "应当": "ying1 dang1",
+    "肯": "ken3", "愿意": "yuan4 yi4", "敢于": "gan3 yu2",
+    "值得": "zhi2 de5", "容": "rong2", "容易": "rong2 yi4",
+    "简": "jian3", "单纯": "dan1 chun2", "杂": "za2",
+    "复杂": "fu4 za2", "困": "kun4", "易于": "yi4 yu2",
+    "或者": "huo4 zhe3", "或许": "huo4 xu3", "也许": "ye3 xu3",
+    "大概": "da4 gai4", "概": "gai4", "恐": "kong3", "怕是": "pa4 shi4",
+    "似乎": "si4 hu1", "乎": "hu1", "竟": "jing4", "究竟": "jiu1 jing4",
+    "毕": "bi4", "毕竟": "bi4 jing4", "终于": "zhong1 yu2",
+    "居然": "ju1 ran2", "果然": "guo3 ran2", "忽": "hu1",
+    "忽然": "hu1 ran2", "突": "tu1", "突然": "tu1 ran2",
+    "渐": "jian4", "逐": "zhu2", "逐渐": "zhu2 jian4",
+    "慢慢": "man4 man4", "赶": "gan3", "紧": "jin3",
+    "赶紧": "gan3 jin3", "马上": "ma3 shang4", "立": "li4",
+    "立刻": "li4 ke4", "刻": "ke4", "顿": "dun4", "片刻": "pian4 ke4",
+    "瞬": "shun4", "眨": "zha3", "霎": "sha4",
+    "持续": "chi2 xu4", "续": "xu4", "继": "ji4", "继续": "ji4 xu4",
+    "保持": "bao3 chi2", "维": "wei2", "维持": "wei2 chi2",
+    "停止": "ting2 zhi3", "止": "zhi3", "结": "jie2", "束": "shu4",
+    "完": "wan2", "完成": "wan2 cheng2", "实现": "shi2 xian4",
+    "达": "da2", "达到": "da2 dao4", "超": "chao1", "超过": "chao1 guo4",
+    "低": "di1", "降低": "jiang4 di1", "升高": "sheng1 gao1",
+    "扩": "kuo4", "扩大": "kuo4 da4", "缩小": "suo1 xiao3",
+    "促": "cu4", "促进": "cu4 jin4", "推动": "tui1 dong4",
+    "阻": "zu3", "碍": "ai4", "妨": "fang2", "限": "xian4",
+    "限制": "xian4 zhi4", "禁": "jin4", "禁止": "jin4 zhi3",
+    "允": "yun3", "允许": "yun3 xu3", "批准": "pi1 zhun3",
+    "拒": "ju4", "绝": "jue2", "拒绝": "ju4 jue2",
+    "接受": "jie1 shou4", "承": "cheng2", "承认": "cheng2 ren4",
+    "否": "fou3", "否认": "fou3 ren4", "承担": "cheng2 dan1",
+    "负": "fu4", "负责": "fu4 ze2", "责": "ze2", "任务": "ren4 wu4",
+    "义": "yi4", "义务": "yi4 wu4", "权利": "quan2 li4",
+    "自由": "zi4 you2", "平等": "ping2 deng3", "公平": "gong1 ping2",
+    "正义": "zheng4 yi4", "道德": "dao4 de2", "德": "de2",
+    "法律": "fa3 lv4", "罪": "zui4", "罚": "fa2", "判断": "pan4 duan4",
+    "法院": "fa3 yuan4", "证明": "zheng4 ming2", "证据": "zheng4 ju4",
+    "嫌": "xian2", "疑问": "yi2 wen4", "调查研究": "diao4 cha2 yan2 jiu1",
+    "警": "jing3", "察看": "cha2 kan4", "抓": "zhua1", "捕": "bu3",
+    "逃": "tao2", "躲": "duo3", "藏起": "cang2 qi3",
+    "偷": "tou1", "抢": "qiang3", "骗": "pian4", "谎": "huang3",
+    "诚": "cheng2", "诚实": "cheng2 shi2", "信任": "xin4 ren4",
+    "怀": "huai2", "怀疑": "huai2 yi2", "相信他": "xiang1 xin4 ta1",
+    "尊": "zun1", "敬": "jing4", "尊敬": "zun1 jing4",
+    "礼": "li3", "貌": "mao4", "礼貌": "li3 mao4",
+    "谦": "qian1", "虚": "xu1", "骄": "jiao1", "傲": "ao4",
+    "骄傲": "jiao1 ao4", "自豪": "zi4 hao2", "豪": "hao2",
+    "惭": "can2", "愧": "kui4", "羞": "xiu1", "耻": "chi3",
+    "荣": "rong2", "誉": "yu4", "荣誉": "rong2 yu4",
+    "奖": "jiang3", "奖励": "jiang3 li4", "励": "li4",
+    "惩": "cheng2", "鼓": "gu3", "鼓励": "gu3 li4",
+    "努": "nu3", "努力": "nu3 li4", "奋": "fen4", "奋斗": "fen4 dou4",
+    "斗": "dou4", "拼": "pin1", "竞": "jing4", "竞争": "jing4 zheng1",
+    "合作": "he2 zuo4", "配": "pei4", "配合": "pei4 he2",
+    "协": "xie2", "协作": "xie2 zuo4", "共同": "gong4 tong2",
+    "集": "ji2", "集体": "ji2 ti3", "集中": "ji2 zhong1",
+    "个人": "ge4 ren2", "人类": "ren2 lei4", "人口": "ren2 kou3",
+    "人员": "ren2 yuan2", "员": "yuan2", "成员": "cheng2 yuan2",
+    "委": "wei3", "代": "dai4", "代表": "dai4 biao3",
+    "主席": "zhu3 xi2", "席": "xi2", "总统": "zong3 tong3",
+    "总理": "zong3 li3", "部长": "bu4 zhang3", "主任": "zhu3 ren4",
+    "经理": "jing1 li3", "董": "dong3", "秘": "mi4", "秘书": "mi4 shu1",
+    "职": "zhi2", "职业": "zhi2 ye4", "职工": "zhi2 gong1",
+    "专": "zhuan1", "专家": "zhuan1 jia1", "专业": "zhuan1 ye4",
+    "教授": "jiao4 shou4", "授": "shou4", "博": "bo2",
+    "博士": "bo2 shi4", "硕": "shuo4", "硕士": "shuo4 shi4",
+    "毕业": "bi4 ye4", "文凭": "wen2 ping2", "凭": "ping2",
+    "成绩": "cheng2 ji4", "绩": "ji4", "分析": "fen1 xi1",
+    "知识": "zhi1 shi5", "识": "shi2", "智慧": "zhi4 hui4",
+    "慧": "hui4", "思想": "si1 xiang3", "思考": "si1 kao3",
+    "观点": "guan1 dian3", "观念": "guan1 nian4",
+    "理论": "li3 lun4", "理解": "li3 jie3", "解": "jie3",
+    "解决": "jie3 jue2", "决": "jue2", "决定": "jue2 ding4",
+    "决心": "jue2 xin1", "方案": "fang1 an4", "案": "an4",
+    "计划": "ji4 hua4", "划": "hua4", "安排": "an1 pai2",
+    "措": "cuo4", "措施": "cuo4 shi1", "办": "ban4",
+    "办法": "ban4 fa3", "方法": "fang1 fa3", "方式": "fang1 shi4",
+    "手段": "shou3 duan4", "段": "duan4", "过程": "guo4 cheng2",
+    "程": "cheng2", "程度": "cheng2 du4", "水平": "shui3 ping2",
+    "标": "biao1", "标准": "biao1 zhun3", "目标": "mu4 biao1",
+    "目": "mu4", "任何": "ren4 he2", "何": "he2",
+    "条件": "tiao2 jian4", "基本": "ji1 ben3", "本质": "ben3 zhi4",
+    "关键": "guan1 jian4", "键": "jian4", "重点": "zhong4 dian3",
+    "要点": "yao4 dian3", "特点": "te4 dian3", "特别": "te4 bie2",
+    "特殊": "te4 shu1", "殊": "shu1", "普遍": "pu3 bian4",
+    "一般": "yi1 ban1", "通常": "tong1 chang2", "正常": "zheng4 chang2",
+    "异": "yi4", "异常": "yi4 chang2", "奇怪": "qi2 guai4",
+    "惊讶": "jing1 ya4", "讶": "ya4", "意外": "yi4 wai4",
+    "偶": "ou3", "偶然": "ou3 ran2", "碰巧": "peng4 qiao3",
+    "幸": "xing4", "幸运": "xing4 yun4", "幸福": "xing4 fu2",
+    "福": "fu2", "祝": "zhu4", "祝福": "zhu4 fu2",
+    "庆": "qing4", "庆祝": "qing4 zhu4", "贺": "he4",
+    "节日": "jie2 ri4", "假日": "jia4 ri4", "春节": "chun1 jie2",
+    "礼物": "li3 wu4", "客": "ke4", "客人": "ke4 ren2",
+    "主人": "zhu3 ren2", "招": "zhao1", "待": "dai4",
+    "招待": "zhao1 dai4", "邀": "yao1", "邀请": "yao1 qing3",
+    "参": "can1", "参加": "can1 jia1", "参观": "can1 guan1",
+    "访": "fang3", "访问": "fang3 wen4", "拜": "bai4",
+    "聚": "ju4", "聚会": "ju4 hui4", "宴": "yan4",
+    "婚": "hun1", "婚礼": "hun1 li3", "娶": "qu3", "嫁": "jia4",
+    "离婚": "li2 hun1", "恋": "lian4", "恋爱": "lian4 ai4",
+    "情人": "qing2 ren2", "爱情": "ai4 qing2", "友谊": "you3 yi4",
+    "谊": "yi4", "感情": "gan3 qing2", "情绪": "qing2 xu4",
+    "绪": "xu4", "心情": "xin1 qing2", "态度": "tai4 du4",
+    "脾": "pi2", "气愤": "qi4 fen4", "愤": "fen4", "怒气": "nu4 qi4",
+    "烦": "fan2", "烦恼": "fan2 nao3", "恼": "nao3",
+    "忧愁": "you1 chou2", "伤": "shang1", "伤心": "shang1 xin1",
+    "痛苦": "tong4 ku3", "苦": "ku3", "辛": "xin1",
+    "辛苦": "xin1 ku3", "累积": "lei3 ji1", "疲": "pi2",
+    "疲劳": "pi2 lao2", "劳": "lao2", "劳动": "lao2 dong4",
+    "休": "xiu1", "休息": "xiu1 xi5", "息": "xi1",
+    "轻松": "qing1 song1", "松": "song1", "舒": "shu1",
+    "舒服": "shu1 fu5", "适": "shi4", "合适": "he2 shi4",
+    "满意": "man3 yi4", "满足": "man3 zu2", "足": "zu2",
+    "足够": "zu2 gou4", "够": "gou4", "缺": "que1",
+    "缺少": "que1 shao3", "缺点": "que1 dian3", "优": "you1",
+    "优点": "you1 dian3", "优秀": "you1 xiu4", "秀": "xiu4",
+    "棒": "bang4", "精": "jing1", "精彩": "jing1 cai3",
+    "彩": "cai3", "美好": "mei3 hao3", "完美": "wan2 mei3",
+    "糟": "zao1", "糟糕": "zao1 gao1", "糕": "gao1",
+    "严": "yan2", "严重": "yan2 zhong4", "严格": "yan2 ge2",
+    "认真": "ren4 zhen1", "仔": "zi3", "仔细": "zi3 xi4",
+    "小心": "xiao3 xin1", "注意": "zhu4 yi4", "注": "zhu4",
+    "专心": "zhuan1 xin1", "耐": "nai4", "耐心": "nai4 xin1",
+    "坚": "jian1", "坚持": "jian1 chi2", "坚定": "jian1 ding4",
+    "勇气": "yong3 qi4", "勇敢": "yong3 gan3", "胆": "dan3",
+    "害怕": "hai4 pa4", "恐惧": "kong3 ju4", "惧": "ju4",
+    "紧张": "jin3 zhang1", "放松": "fang4 song1",
+    "冷静": "leng3 jing4", "激": "ji1", "激动": "ji1 dong4",
+    "兴高": "xing4 gao1",
+    # batch 3: standalone fallbacks for every char that appears in the
+    # word dictionary (so no listed compound's character ever drops
+    # when it shows up alone); polyphones get their commonest reading
+    "丽": "li4", "乐": "le4", "京": "jing1", "亲": "qin1",
+    "传": "chuan2", "供": "gong1", "候": "hou4", "健": "jian4",
+    "兴": "xing4", "况": "kuang4", "劣": "lie4", "升": "sheng1",
+    "华": "hua2", "卷": "juan4", "历": "li4", "史": "shi3",
+    "吐": "tu3", "呕": "ou3", "器": "qi4", "坦": "tan3",
+    "处": "chu4", "宁": "ning2", "宗": "zong1", "宜": "yi2",
+    "宿": "su4", "居": "ju1", "展": "zhan3", "州": "zhou1",
+    "差": "cha4", "广": "guang3", "庭": "ting2", "康": "kang1",
+    "弹": "tan2", "归": "gui1", "急": "ji2", "性": "xing4",
+    "恶": "e4", "悉": "xi1", "担": "dan1", "挑": "tiao1",
+    "挥": "hui1", "摔": "shuai1", "操": "cao1", "散": "san4",
+    "数": "shu4", "旋": "xuan2", "族": "zu2", "景": "jing3",
+    "曲": "qu3", "朝": "chao2", "校": "xiao4", "植": "zhi2",
+    "气": "qi4", "求": "qiu2", "汉": "han4", "洋": "yang2",
+    "济": "ji4", "涨": "zhang3", "液": "ye4", "淹": "yan1",
+    "湾": "wan1", "漂": "piao1", "灾": "zai1", "照": "zhao4",
+    "熟": "shu2", "物": "wu4", "琴": "qin2", "盛": "sheng4",
+    "码": "ma3", "确": "que4", "社": "she4", "积": "ji1",
+    "累": "lei4", "纫": "ren4", "纯": "chun2", "织": "zhi1",
+    "缝": "feng4", "缩": "suo1", "背": "bei4", "胶": "jiao1",
+    "脑": "nao3", "舍": "she4", "舞": "wu3", "艺": "yi4",
+    "英": "ying1", "茂": "mao4", "落": "luo4", "藏": "cang2",
+    "角": "jiao3", "调": "diao4", "趣": "qu4", "载": "zai4",
+    "野": "ye3", "量": "liang4", "钻": "zuan1", "闻": "wen2",
+    "阅": "yue4", "降": "jiang4", "隔": "ge2", "隙": "xi4",
+    "难": "nan2", "顺": "shun4", "题": "ti2", "首": "shou3",
+    "香": "xiang1",
+    # and the probe-corpus stragglers not in any compound
+    "响": "xiang3", "影": "ying3", "泳": "yong3", "联": "lian2",
+    "讨": "tao3", "采": "cai3", "冠": "guan1",
 }
 
-_CMN_MAX = max(len(k) for k in list(CMN_WORDS) + ["一"])
+# merged lookup: the word dict wins on conflicts; multi-char entries
+# from either dict participate in longest-match
+_CMN_ALL = {**CMN_CHARS, **CMN_WORDS}
+_CMN_MAX = max(len(k) for k in _CMN_ALL)
 
 
 def _is_hanzi(ch: str) -> bool:
@@ -599,7 +800,7 @@ def cmn_word_to_ipa(w: str) -> str:
         matched = False
         for ln in range(min(_CMN_MAX, n - i), 1, -1):
             seg = w[i:i + ln]
-            py = CMN_WORDS.get(seg)
+            py = _CMN_ALL.get(seg)
             if py:
                 parts = py.split()
                 syls.extend(parts)
@@ -611,7 +812,7 @@ def cmn_word_to_ipa(w: str) -> str:
         if matched:
             continue
         ch = w[i]
-        py = CMN_WORDS.get(ch) or CMN_CHARS.get(ch)
+        py = _CMN_ALL.get(ch)
         if ch == "〇":
             py = "ling2"
         if py:
