@@ -12,7 +12,7 @@ dependency does not exist here; this module is a fresh, self-contained
 G2P front: per-language ordered longest-match rule tables, exception
 lexicons, and script ENGINES (Brahmic abugidas in g2p_indic.py; Hangul/
 Ethiopic/Cherokee/Myanmar/Thai + kana in g2p_scripts.py / g2p_tables3.py)
-covering 111 language codes (docs/LANGUAGES.md).  It produces IPA over
+covering 115 language codes (docs/LANGUAGES.md).  It produces IPA over
 the same symbol set the Piper voices use, is deterministic, and is
 thread-safe (pure functions, no global C state — the reference's espeak
 is famously NOT thread-safe; see
