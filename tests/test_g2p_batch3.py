@@ -538,3 +538,20 @@ def test_ml_chillu_and_bn_sibilant():
     bn = _get_g2p("bn")
     assert bn.word_to_ipa("ভাষা") == "bʰaːʃaː"
     assert bn.word_to_ipa("মানুষ") == "maːnuʃ"
+
+
+def test_ja_kanji_closure_and_on_fallback():
+    """Every kanji inside a listed compound has a standalone reading
+    (on-yomi fallback), so unlisted compounds approximate instead of
+    dropping (電力 = でん+りょく even without a dictionary entry)."""
+    from sonata_amd.text.g2p_tables3 import JA_KANJI, ja_word_to_ipa
+
+    def is_kanji(c):
+        return 0x4E00 <= ord(c) <= 0x9FFF
+
+    uncovered = {ch for k in JA_KANJI for ch in k
+                 if is_kanji(ch) and ch not in JA_KANJI}
+    assert not uncovered, uncovered
+    assert len(JA_KANJI) >= 900
+    assert ja_word_to_ipa("電力") == "denɾjoku"
+    assert ja_word_to_ipa("日本人") == "nihondʒin"
