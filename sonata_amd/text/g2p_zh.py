@@ -795,6 +795,7 @@ def cmn_word_to_ipa(w: str) -> str:
     dictionary — honest, like ja kanji)."""
     syls: List[str] = []
     hanzi_for: List[str] = []
+    solo: List[bool] = []   # True = single-char fallback match
     i, n = 0, len(w)
     while i < n:
         matched = False
@@ -806,6 +807,7 @@ def cmn_word_to_ipa(w: str) -> str:
                 syls.extend(parts)
                 hanzi_for.extend(seg if len(seg) == len(parts)
                                  else ["·"] * len(parts))
+                solo.extend([False] * len(parts))
                 i += ln
                 matched = True
                 break
@@ -819,10 +821,31 @@ def cmn_word_to_ipa(w: str) -> str:
             parts = py.split()
             syls.extend(parts)
             hanzi_for.extend([ch] * len(parts))
+            solo.extend([True] * len(parts))
         # unknown hanzi / non-hanzi: dropped
         i += 1
     syls = _cmn_sandhi(syls, "".join(hanzi_for))
-    return "".join(pinyin_syllable_to_ipa(s) for s in syls)
+    out: List[str] = []
+    for s, hz, alone in zip(syls, hanzi_for, solo):
+        ipa = pinyin_syllable_to_ipa(s)
+        if alone and hz == "儿" and s.startswith("er") and out:
+            # erhua: suffix-儿 rhotacizes the PRECEDING syllable
+            # (这儿 zhèr, 点儿 diǎnr) — strip its coda -n/-ŋ (and a
+            # closing -i after a/e), append ɚ inside its tone letters.
+            # Word-dict entries (儿子, 女儿) never reach here, and a
+            # token-initial 儿 stays ér.
+            prev = out.pop()
+            body = prev.rstrip("˥˦˧˨˩")
+            tone = prev[len(body):]
+            if body.endswith(("n", "ŋ")):
+                body = body[:-1]
+            elif (body.endswith("i") and len(body) > 1
+                    and body[-2] in "ae"):
+                body = body[:-1]
+            out.append(body + "ɚ" + tone)
+            continue
+        out.append(ipa)
+    return "".join(out)
 
 
 # --------------------------------------------------------------------- #

@@ -135,3 +135,13 @@ def test_cmn_word_chars_have_fallbacks():
     uncovered = {ch for k in _CMN_ALL for ch in k
                  if _is_hanzi(ch) and ch not in _CMN_ALL}
     assert not uncovered, uncovered
+
+
+def test_cmn_erhua():
+    """Suffix-儿 rhotacizes the preceding syllable; 儿-words from the
+    dictionary (女儿/儿子) keep their full ér syllable."""
+    assert cmn_word_to_ipa("这儿") == "ʈʂɤɚ˥˩"
+    assert cmn_word_to_ipa("点儿") == "tjɛɚ˨˩˦"   # -n coda dropped
+    assert cmn_word_to_ipa("花儿") == "xwaɚ˥"
+    assert cmn_word_to_ipa("女儿") == "ny˨˩˦ɚ˧˥"  # word entry: real ér
+    assert cmn_word_to_ipa("儿子").startswith("ɚ˧˥")
