@@ -775,6 +775,10 @@ CMN_CHARS: Dict[str, str] = {
     # and the probe-corpus stragglers not in any compound
     "响": "xiang3", "影": "ying3", "泳": "yong3", "联": "lian2",
     "讨": "tao3", "采": "cai3", "冠": "guan1",
+    # 儿-words where 儿 is the real syllable ér (protect from erhua)
+    "儿童": "er2 tong2", "婴儿": "ying1 er2", "幼儿": "you4 er2",
+    "幼儿园": "you4 er2 yuan2", "儿女": "er2 nv3",
+    "婴": "ying1", "幼": "you4", "童": "tong2",
 }
 
 # merged lookup: the word dict wins on conflicts; multi-char entries
@@ -955,9 +959,96 @@ YUE_CHARS: Dict[str, str] = {
     "戲": "hei3", "相": "soeng2", "畫": "waa2", "書": "syu1",
     "筆": "bat1", "紙": "zi2", "字": "zi6", "文": "man4",
     "語": "jyu5", "音": "jam1", "聲": "seng1", "歌": "go1",
+    # batch 2: more of the traditional-script frequency core
+    "係唔係": "hai6 m4 hai6", "乜": "mat1", "冧": "lam1",
+    "搞": "gaau2", "掂": "dim6", "喇喇": "laa4 laa4",
+    "嚟": "lai4", "咪": "mai5", "噃": "bo3", "囉囉": "lo4 lo4",
+    "呃": "ngaak1", "氹": "tam5", "攋": "laai6",
+    "事": "si6", "情": "cing4", "理": "lei5", "性": "sing3",
+    "法": "faat3", "律": "leot6", "規": "kwai1", "則": "zak1",
+    "制": "zai3", "政": "zing3", "府": "fu2", "黨": "dong2",
+    "選": "syun2", "舉": "geoi2", "投": "tau4", "票": "piu3",
+    "權": "kyun4", "利": "lei6", "義": "ji6", "務": "mou6",
+    "責": "zaak3", "任": "jam6", "管": "gun2", "領": "ling5",
+    "導": "dou6", "組": "zou2", "織": "zik1", "團": "tyun4",
+    "隊": "deoi2", "員": "jyun4", "長": "zoeng2", "主": "zyu2",
+    "席": "zik6", "總": "zung2", "統": "tung2", "經": "ging1",
+    "濟": "zai3", "貿": "mau6", "易": "ji6", "商": "soeng1",
+    "業": "jip6", "工": "gung1", "廠": "cong2", "產": "caan2",
+    "品": "ban2", "質": "zat1", "價": "gaa3", "值": "zik6",
+    "市場": "si5 coeng4", "股": "gu2", "資": "zi1", "本": "bun2",
+    "投資": "tau4 zi1", "利息": "lei6 sik1", "借": "ze3",
+    "還": "waan4", "賺": "zaan6", "蝕": "sit6", "慳錢": "haan1 cin2",
+    "貴價": "gwai3 gaa3", "平價": "peng4 gaa3",
+    "科": "fo1", "技": "gei6", "術": "seot6", "機": "gei1",
+    "器": "hei3", "設": "cit3", "計": "gai3", "程": "cing4",
+    "式": "sik1", "網": "mong5", "絡": "lok3", "線": "sin3",
+    "電子": "din6 zi2", "數": "sou3", "碼": "maa5",
+    "研": "jin4", "究": "gau3", "驗": "jim6", "測": "cak1",
+    "試": "si3", "題": "tai4", "答": "daap3", "案": "on3",
+    "教育": "gaau3 juk6", "課": "fo3", "堂": "tong4",
+    "班": "baan1", "級": "kap1", "考": "haau2", "卷": "gyun2",
+    "功課": "gung1 fo3", "練": "lin6", "習": "zaap6",
+    "圖": "tou4", "館": "gun2", "院": "jyun2", "場": "coeng4",
+    "公園": "gung1 jyun2", "酒店": "zau2 dim3", "餐廳": "caan1 teng1",
+    "超市": "ciu1 si5", "街市": "gaai1 si5", "商場": "soeng1 coeng4",
+    "公司仔": "gung1 si1 zai2", "寫字樓": "se2 zi6 lau4",
+    "飛": "fei1", "航": "hong4", "運": "wan6", "送": "sung3",
+    "搬": "bun1", "泊": "paak3", "揸": "zaa1", "踩": "caai2",
+    "站": "zaam6", "碼頭": "maa5 tau4", "機場": "gei1 coeng4",
+    "天氣": "tin1 hei3", "落雨": "lok6 jyu5", "打風": "daa2 fung1",
+    "凍冰冰": "dung3 bing1 bing1", "熱辣辣": "jit6 laat6 laat6",
+    "太陽": "taai3 joeng4", "月光": "jyut6 gwong1",
+    "星": "sing1", "空": "hung1", "雲": "wan4", "霧": "mou6",
+    "病": "beng6", "痛": "tung3", "藥": "joek6", "針": "zam1",
+    "傷": "soeng1", "燒": "siu1", "咳": "kat1", "攰": "gui6",
+    "瞓覺": "fan3 gaau3", "沖涼": "cung1 loeng4",
+    "洗手": "sai2 sau2", "刷牙": "caat3 ngaa4",
+    "著衫": "zoek3 saam1", "衫": "saam1", "褲": "fu3",
+    "裙": "kwan4", "鞋": "haai4", "襪": "mat6", "帽": "mou2",
+    "袋": "doi2", "遮": "ze1", "錶": "biu1", "戒指": "gaai3 zi2",
+    "飲食": "jam2 sik6", "早餐": "zou2 caan1",
+    "晏晝": "aan3 zau3", "晚飯": "maan5 faan6",
+    "味": "mei6", "甜": "tim4", "酸": "syun1", "苦": "fu2",
+    "辣": "laat6", "鹹": "haam4", "淡": "taam5", "香": "hoeng1",
+    "臭": "cau3", "新鮮": "san1 sin1",
+    "開心到": "hoi1 sam1 dou3", "傷心": "soeng1 sam1",
+    "擔心": "daam1 sam1", "放心": "fong3 sam1",
+    "細": "sai3", "細個": "sai3 go3", "大個": "daai6 go3",
+    "後生": "hau6 saang1", "老人家": "lou5 jan4 gaa1",
+    "亞": "aa3", "歐": "au1", "非": "fei1", "澳": "ou3",
+    "日本仔": "jat6 bun2 zai2", "韓國": "hon4 gwok3",
+    "台灣": "toi4 waan1", "澳門": "ou3 mun2",
+    "九龍": "gau2 lung4", "新界": "san1 gaai3",
+    # standalone fallbacks: every char in a compound key reads alone too
+    "且": "ce2", "世": "sai3", "今": "gam1", "似": "ci5",
+    "但": "daan6", "依": "ji1", "光": "gwong1", "再": "zoi3",
+    "冰": "bing1", "到": "dou3", "刷": "caat3", "功": "gung1",
+    "友": "jau5", "台": "toi4", "司": "si1", "問": "man6",
+    "因": "jan1", "園": "jyun4", "士": "si6", "太": "taai3",
+    "如": "jyu4", "子": "zi2", "察": "caat3", "對": "deoi3",
+    "已": "ji5", "巴": "baa1", "師": "si1", "幾": "gei2",
+    "店": "dim3", "廣": "gwong2", "廳": "teng1", "思": "si1",
+    "息": "sik1", "意": "ji3", "戒": "gaai3", "所": "so2",
+    "打": "daa2", "指": "zi2", "擔": "daam1", "放": "fong3",
+    "方": "fong1", "晚": "maan5", "晝": "zau3", "晨": "san4",
+    "普": "pou2", "有": "jau5", "朋": "pang4", "果": "gwo2",
+    "校": "haau6", "樓": "lau4", "氣": "hei3", "沖": "cung1",
+    "洗": "sai2", "涼": "loeng4", "港": "gong2", "灣": "waan1",
+    "為": "wai6", "琴": "kam4", "界": "gaai3", "的": "dik1",
+    "粵": "jyut6", "老": "lou5", "而": "ji4", "育": "juk6",
+    "腦": "nou5", "英": "jing1", "見": "gin3", "視": "si6",
+    "覺": "gok3", "解": "gaai2", "該": "goi1", "謝": "ze6",
+    "警": "ging2", "超": "ciu1", "通": "tung1", "醫": "ji1",
+    "鍾": "zung1", "鐵": "tit3", "陽": "joeng4", "電": "din6",
+    "韓": "hon4", "餐": "caan1", "鮮": "sin1", "齊": "cai4",
+    "龍": "lung4",
 }
 
-_YUE_MAX = max(len(k) for k in list(YUE_WORDS) + ["一"])
+# merged lookup (word dict wins); multi-char entries from either dict
+# participate in longest-match
+_YUE_ALL = {**YUE_CHARS, **YUE_WORDS}
+_YUE_MAX = max(len(k) for k in _YUE_ALL)
 
 
 def yue_word_to_ipa(w: str) -> str:
@@ -968,7 +1059,7 @@ def yue_word_to_ipa(w: str) -> str:
         matched = False
         for ln in range(min(_YUE_MAX, n - i), 1, -1):
             seg = w[i:i + ln]
-            jp = YUE_WORDS.get(seg)
+            jp = _YUE_ALL.get(seg)
             if jp:
                 out.extend(jp.split())
                 i += ln
@@ -977,7 +1068,7 @@ def yue_word_to_ipa(w: str) -> str:
         if matched:
             continue
         ch = w[i]
-        jp = YUE_WORDS.get(ch) or YUE_CHARS.get(ch)
+        jp = _YUE_ALL.get(ch)
         if ch == "〇":
             jp = "ling4"
         if jp:

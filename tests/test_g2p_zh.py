@@ -124,17 +124,18 @@ def test_zh_dictionary_size_floor():
 
     assert len(CMN_CHARS) >= 1100
     assert len(CMN_WORDS) >= 380
-    assert len(YUE_CHARS) >= 250
+    assert len(YUE_CHARS) >= 400
 
 
 def test_cmn_word_chars_have_fallbacks():
     """Every hanzi that appears in any dictionary key has a standalone
     reading too — no listed compound's character drops when alone."""
-    from sonata_amd.text.g2p_zh import _CMN_ALL, _is_hanzi
+    from sonata_amd.text.g2p_zh import _CMN_ALL, _YUE_ALL, _is_hanzi
 
-    uncovered = {ch for k in _CMN_ALL for ch in k
-                 if _is_hanzi(ch) and ch not in _CMN_ALL}
-    assert not uncovered, uncovered
+    for table in (_CMN_ALL, _YUE_ALL):
+        uncovered = {ch for k in table for ch in k
+                     if _is_hanzi(ch) and ch not in table}
+        assert not uncovered, uncovered
 
 
 def test_cmn_erhua():
