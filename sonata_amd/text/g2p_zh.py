@@ -781,6 +781,130 @@ CMN_CHARS: Dict[str, str] = {
     "婴": "ying1", "幼": "you4", "童": "tong2",
 }
 
+# Traditional -> simplified for the characters in the frequency core
+# that differ, so cmn reads traditional-script text too (Mandarin is
+# written in both; the reading is identical).  Only pairs where the
+# traditional form is NOT already a dictionary key matter.
+_T2S = str.maketrans({
+    "國": "国", "學": "学", "會": "会", "說": "说", "話": "话",
+    "語": "语", "漢": "汉", "時": "时", "間": "间", "東": "东",
+    "車": "车", "門": "门", "問": "问", "聞": "闻", "馬": "马",
+    "鳥": "鸟", "魚": "鱼", "龍": "龙", "風": "风", "雲": "云",
+    "電": "电", "點": "点", "鐘": "钟", "錢": "钱", "銀": "银",
+    "鐵": "铁", "鋼": "钢", "銅": "铜", "長": "长", "張": "张",
+    "開": "开", "關": "关", "買": "买", "賣": "卖", "貴": "贵",
+    "費": "费", "資": "资", "質": "质", "貨": "货", "員": "员",
+    "圓": "圆", "園": "园", "遠": "远", "運": "运", "還": "还",
+    "這": "这", "進": "进", "連": "连", "過": "过", "達": "达",
+    "遲": "迟", "邊": "边", "書": "书", "寫": "写", "讀": "读",
+    "課": "课", "試": "试", "誰": "谁", "請": "请", "謝": "谢",
+    "講": "讲", "記": "记", "計": "计", "認": "认", "識": "识",
+    "譯": "译", "讓": "让", "議": "议", "論": "论", "訴": "诉",
+    "評": "评", "詞": "词", "該": "该", "調": "调", "談": "谈",
+    "證": "证", "設": "设", "訪": "访", "許": "许", "護": "护",
+    "見": "见", "視": "视", "覺": "觉", "觀": "观", "規": "规",
+    "親": "亲", "頭": "头", "顏": "颜", "題": "题", "顧": "顾",
+    "頁": "页", "順": "顺", "須": "须", "領": "领", "飛": "飞",
+    "飯": "饭", "飲": "饮", "餐": "餐", "館": "馆", "養": "养",
+    "體": "体", "發": "发", "當": "当", "對": "对", "應": "应",
+    "幾": "几", "機": "机", "樹": "树", "樣": "样", "橋": "桥",
+    "權": "权", "樂": "乐", "標": "标", "歐": "欧", "歲": "岁",
+    "歷": "历", "歸": "归", "殘": "残", "氣": "气", "湯": "汤",
+    "溫": "温", "滿": "满", "漲": "涨", "濟": "济", "灣": "湾",
+    "燈": "灯", "營": "营", "爲": "为", "為": "为", "爺": "爷",
+    "狀": "状", "獨": "独", "現": "现", "環": "环", "產": "产",
+    "畫": "画", "異": "异", "當": "当", "發": "发", "百": "百",
+    "監": "监", "盡": "尽", "礎": "础", "禮": "礼", "萬": "万",
+    "億": "亿", "務": "务", "動": "动", "勞": "劳", "勢": "势",
+    "區": "区", "醫": "医", "協": "协", "單": "单", "賽": "赛",
+    "廠": "厂", "廣": "广", "慶": "庆", "應": "应", "廢": "废",
+    "彈": "弹", "強": "强", "後": "后", "從": "从", "復": "复",
+    "微": "微", "德": "德", "憶": "忆", "懂": "懂", "戰": "战",
+    "戲": "戏", "壓": "压", "廳": "厅", "臺": "台", "與": "与",
+    "興": "兴", "舊": "旧", "藝": "艺", "藥": "药", "蘇": "苏",
+    "蘭": "兰", "處": "处", "號": "号", "虧": "亏", "蟲": "虫",
+    "衆": "众", "眾": "众", "術": "术", "衛": "卫", "裝": "装",
+    "裏": "里", "裡": "里", "補": "补", "製": "制", "複": "复",
+    "節": "节", "筆": "笔", "簡": "简", "類": "类", "粗": "粗",
+    "納": "纳", "紅": "红", "級": "级", "紙": "纸", "組": "组",
+    "細": "细", "終": "终", "經": "经", "給": "给", "絕": "绝",
+    "統": "统", "繼": "继", "續": "续", "維": "维", "綠": "绿",
+    "網": "网", "練": "练", "線": "线", "縣": "县", "總": "总",
+    "織": "织", "繁": "繁", "紀": "纪", "約": "约", "結": "结",
+    "羅": "罗", "義": "义", "習": "习", "聯": "联", "聽": "听",
+    "聲": "声", "職": "职", "腦": "脑", "臉": "脸", "膚": "肤",
+    "臨": "临", "無": "无", "煙": "烟", "熱": "热", "愛": "爱",
+    "幹": "干", "乾": "干", "壞": "坏", "壘": "垒", "場": "场",
+    "塊": "块", "報": "报", "壽": "寿", "夢": "梦", "頂": "顶",
+    "項": "项", "預": "预", "頓": "顿", "顯": "显", "餘": "余",
+    "驗": "验", "驚": "惊", "骨": "骨",
+    "鬥": "斗", "鬧": "闹", "麥": "麦", "麵": "面", "黃": "黄",
+    "齊": "齐", "齒": "齿", "優": "优", "傳": "传", "傷": "伤",
+    "價": "价", "儀": "仪", "億": "亿", "們": "们", "個": "个",
+    "倆": "俩", "備": "备", "傢": "家", "兒": "儿", "內": "内",
+    "兩": "两", "冊": "册", "軍": "军", "農": "农", "凍": "冻",
+    "淨": "净", "準": "准", "涼": "凉", "減": "减", "湊": "凑",
+    "剛": "刚", "創": "创", "劃": "划", "別": "别", "劇": "剧",
+    "劉": "刘", "勝": "胜", "勤": "勤", "勵": "励", "勸": "劝",
+    "響": "响", "唐": "唐", "啓": "启", "啟": "启", "嚴": "严",
+    "壇": "坛", "壯": "壮", "聰": "聪", "聖": "圣", "堅": "坚",
+    "墳": "坟", "墻": "墙", "數": "数", "樓": "楼", "槍": "枪",
+    "條": "条", "極": "极", "構": "构", "槽": "槽", "檢": "检",
+    "業": "业", "榮": "荣", "實": "实", "寶": "宝", "審": "审",
+    "寬": "宽", "寫": "写", "導": "导", "將": "将", "專": "专",
+    "尋": "寻", "對": "对", "層": "层", "屬": "属", "島": "岛",
+    "峽": "峡", "帶": "带", "幣": "币", "師": "师", "帳": "帐",
+    "幫": "帮", "幾": "几", "庫": "库", "廟": "庙", "異": "异",
+    "彙": "汇", "徑": "径", "態": "态", "悶": "闷", "惡": "恶",
+    "憂": "忧", "慮": "虑", "懷": "怀", "憲": "宪", "戀": "恋",
+    "戶": "户", "擔": "担", "據": "据", "擇": "择", "擊": "击",
+    "掛": "挂", "採": "采", "換": "换", "揚": "扬", "搶": "抢",
+    "撐": "撑", "擴": "扩", "攝": "摄", "敗": "败", "敵": "敌",
+    "斷": "断", "舊": "旧", "昇": "升", "晝": "昼", "暈": "晕",
+    "暢": "畅", "曆": "历", "朵": "朵", "殺": "杀", "雜": "杂",
+    "權": "权", "測": "测", "滅": "灭", "滿": "满", "濕": "湿",
+    "濃": "浓", "潔": "洁", "淺": "浅", "滬": "沪", "漁": "渔",
+    "潤": "润", "澤": "泽", "濱": "滨", "烏": "乌", "無": "无",
+    "煩": "烦", "燒": "烧", "燦": "灿", "爐": "炉", "爭": "争",
+    "牆": "墙", "獎": "奖", "獲": "获", "玆": "兹", "環": "环",
+    "瑪": "玛", "瓊": "琼", "甕": "瓮", "疊": "叠", "療": "疗",
+    "瘋": "疯", "癢": "痒", "皺": "皱", "盜": "盗", "盤": "盘",
+    "盧": "卢", "眞": "真", "矚": "瞩", "確": "确", "碼": "码",
+    "磚": "砖", "礙": "碍", "祕": "秘", "禍": "祸", "禦": "御",
+    "禪": "禅", "禿": "秃", "稅": "税", "稱": "称", "穀": "谷",
+    "積": "积", "穩": "稳", "窮": "穷", "竊": "窃", "競": "竞",
+    "籃": "篮", "籌": "筹", "籍": "籍", "糧": "粮", "緊": "紧",
+    "緒": "绪", "緣": "缘", "縮": "缩", "缺": "缺", "罰": "罚",
+    "罵": "骂", "罷": "罢", "脫": "脱", "腸": "肠", "膽": "胆",
+    "艦": "舰", "芻": "刍", "華": "华", "萊": "莱", "萬": "万",
+    "落": "落", "葉": "叶", "蒙": "蒙", "蓋": "盖", "蔣": "蒋",
+    "薄": "薄", "藍": "蓝", "藏": "藏", "襯": "衬", "覽": "览",
+    "訓": "训", "訊": "讯", "託": "托", "詳": "详", "誇": "夸",
+    "誠": "诚", "誤": "误", "誼": "谊", "諷": "讽", "謀": "谋",
+    "謂": "谓", "謹": "谨", "譜": "谱", "警": "警", "貝": "贝",
+    "負": "负", "財": "财", "貢": "贡", "販": "贩", "責": "责",
+    "貪": "贪", "貧": "贫", "購": "购", "貯": "贮", "貸": "贷",
+    "貿": "贸", "賀": "贺", "賃": "赁", "賊": "贼", "賞": "赏",
+    "賠": "赔", "賢": "贤", "賺": "赚", "贈": "赠", "贊": "赞",
+    "趕": "赶", "趙": "赵", "躍": "跃", "輕": "轻", "輛": "辆",
+    "輝": "辉", "輪": "轮", "輸": "输", "轉": "转", "轟": "轰",
+    "辦": "办", "辭": "辞", "辯": "辩", "選": "选", "邏": "逻",
+    "鄉": "乡", "鄧": "邓", "鄭": "郑", "鄰": "邻", "釀": "酿",
+    "釋": "释", "鈴": "铃", "鉛": "铅", "銘": "铭", "鋒": "锋",
+    "鋪": "铺", "錄": "录", "錯": "错", "鍵": "键", "鎖": "锁",
+    "鎮": "镇", "鏡": "镜", "鑑": "鉴", "鑒": "鉴", "閃": "闪",
+    "閉": "闭", "閱": "阅", "闆": "板", "闊": "阔", "防": "防",
+    "陝": "陕", "陣": "阵", "除": "除", "陰": "阴", "陳": "陈",
+    "隊": "队", "階": "阶", "隨": "随", "險": "险", "隱": "隐",
+    "雙": "双", "雞": "鸡", "離": "离", "難": "难", "雲": "云",
+    "霧": "雾", "靈": "灵", "靜": "静", "響": "响", "頗": "颇",
+    "頻": "频", "額": "额", "願": "愿", "類": "类", "顯": "显",
+    "風": "风", "颱": "台", "餃": "饺", "餓": "饿", "餘": "余",
+    "驅": "驱", "驕": "骄", "骯": "肮", "鬆": "松", "鬍": "胡",
+    "魯": "鲁", "鮮": "鲜", "鴨": "鸭", "鴻": "鸿", "鵝": "鹅",
+    "鹹": "咸", "麗": "丽", "麼": "么", "齡": "龄",
+})
+
 # merged lookup: the word dict wins on conflicts; multi-char entries
 # from either dict participate in longest-match
 _CMN_ALL = {**CMN_CHARS, **CMN_WORDS}
@@ -797,6 +921,7 @@ def cmn_word_to_ipa(w: str) -> str:
     """Hanzi string -> Mandarin IPA.  Longest-match over CMN_WORDS,
     then CMN_CHARS; unknown hanzi are dropped (needs a bigger reading
     dictionary — honest, like ja kanji)."""
+    w = w.translate(_T2S)  # traditional text reads identically
     syls: List[str] = []
     hanzi_for: List[str] = []
     solo: List[bool] = []   # True = single-char fallback match

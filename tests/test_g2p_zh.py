@@ -146,3 +146,13 @@ def test_cmn_erhua():
     assert cmn_word_to_ipa("花儿") == "xwaɚ˥"
     assert cmn_word_to_ipa("女儿") == "ny˨˩˦ɚ˧˥"  # word entry: real ér
     assert cmn_word_to_ipa("儿子").startswith("ɚ˧˥")
+
+
+def test_cmn_reads_traditional_script():
+    """Mandarin is written in both scripts; the T2S pre-pass makes
+    traditional text read identically to simplified."""
+    assert (cmn_word_to_ipa("我們都是學生")
+            == cmn_word_to_ipa("我们都是学生"))
+    assert (cmn_word_to_ipa("中國經濟發展")
+            == cmn_word_to_ipa("中国经济发展"))
+    assert cmn_word_to_ipa("這兒") == cmn_word_to_ipa("这儿")  # erhua too

@@ -163,6 +163,8 @@ def test_gpu_multilingual_synthesis(tmp_path):
         "ta": "வணக்கம் உலகம்.",
         "ja": "こんにちは、日本語を話します。",
         "fa": "سلام دنیا، فارسی حرف می‌زنم.",
+        "cmn": "你好，世界！我说中文。",
+        "yue": "你好，我哋講廣東話。",
     }
     for lang, text in texts.items():
         pack = create_random_voice(str(tmp_path), f"g_{lang}",
