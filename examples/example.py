@@ -37,7 +37,8 @@ print(f"streamed {n} PCM bytes")
 # multilingual: every supported language code works the same way
 # (full inventory in docs/LANGUAGES.md)
 for lang, text in [("hi", "नमस्ते दुनिया।"), ("ko", "안녕하세요 세계."),
-                   ("ja", "こんにちは、世界。")]:
+                   ("ja", "こんにちは、世界。"), ("cmn", "你好，世界。"),
+                   ("yue", "你好，世界。")]:
     p = create_random_voice(tempfile.mkdtemp(), f"ex_{lang}",
                             quality="x_low", language=lang)
     t = pysonata.Sonata.with_piper(pysonata.PiperModel(p))
