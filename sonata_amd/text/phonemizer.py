@@ -1113,6 +1113,8 @@ _PUNCT_TRANS = str.maketrans({
     "؟": "?", "۔": ".", "،": ",", "؛": ";",
     "։": ".", "՞": "?",
     "。": ".", "、": ",", "！": "!", "？": "?",
+    # full-width forms (CJK texts use U+FF0C comma etc.)
+    "，": ",", "：": ":", "；": ";", "．": ".",
 })
 
 _LANG_SWITCH_RE = re.compile(r"\([a-z-]{2,10}\)")
