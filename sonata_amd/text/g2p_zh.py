@@ -779,6 +779,93 @@ CMN_CHARS: Dict[str, str] = {
     "儿童": "er2 tong2", "婴儿": "ying1 er2", "幼儿": "you4 er2",
     "幼儿园": "you4 er2 yuan2", "儿女": "er2 nv3",
     "婴": "ying1", "幼": "you4", "童": "tong2",
+    # polyphone batch 2
+    "说服": "shuo1 fu2", "游说": "you2 shui4",
+    "中奖": "zhong4 jiang3", "打中": "da3 zhong4",
+    "大夫": "dai4 fu5", "给予": "ji3 yu3",
+    "尽快": "jin3 kuai4", "尽头": "jin4 tou2",
+    "得去": "dei3 qu4", "非得": "fei1 dei3",
+    "磨坊": "mo4 fang2", "磨刀": "mo2 dao1", "磨": "mo2",
+    "坊": "fang1", "夫": "fu1", "予": "yu3",
+    "地上": "di4 shang4", "地下": "di4 xia4",
+    "好好地": "hao3 hao3 de5", "快快地": "kuai4 kuai4 de5",
+    "似地": "shi4 de5",
+    "薄荷": "bo4 he5", "荷花": "he2 hua1", "荷": "he2",
+    "单薄": "dan1 bo2", "薄弱": "bo2 ruo4",
+    "暴露": "bao4 lu4", "露面": "lou4 mian4", "露": "lu4",
+    "暴": "bao4", "系鞋带": "ji4 xie2 dai4", "关系": "guan1 xi5",
+    "系统": "xi4 tong3",
+    "塞车": "sai1 che1", "要塞": "yao4 sai4", "塞": "sai1",
+    "堵塞": "du3 se4",
+    "省长": "sheng3 zhang3", "厂长": "chang3 zhang3",
+    "家长": "jia1 zhang3", "市长": "shi4 zhang3",
+    "增长": "zeng1 zhang3", "长辈": "zhang3 bei4",
+    "辈": "bei4", "乐器": "yue4 qi4", "乐团": "yue4 tuan2",
+    "奏乐": "zou4 yue4", "奏": "zou4",
+    "处方": "chu3 fang1", "相处": "xiang1 chu3",
+    "发廊": "fa4 lang2", "理发": "li3 fa4", "廊": "lang2",
+    "假发": "jia3 fa4", "发型": "fa4 xing2",
+    "数落": "shu3 luo4", "数不清": "shu3 bu4 qing1",
+    "强迫": "qiang3 po4", "勉强": "mian3 qiang3",
+    "倔强": "jue2 jiang4", "迫": "po4", "勉": "mian3",
+    "应付": "ying4 fu4", "供不应求": "gong1 bu4 ying4 qiu2",
+    "称职": "chen4 zhi2", "对称": "dui4 chen4",
+    "称呼": "cheng1 hu5", "呼": "hu1",
+    "扒手": "pa2 shou3", "扒开": "ba1 kai1", "扒": "ba1",
+    "兴旺": "xing1 wang4", "旺": "wang4",
+    "泡茶": "pao4 cha2", "泡沫": "pao4 mo4", "泡": "pao4",
+    "沫": "mo4", "刹车": "sha1 che1", "古刹": "gu3 cha4",
+    "刹": "sha1", "藏书": "cang2 shu1", "宝藏": "bao3 zang4",
+    "宝": "bao3", "咽喉": "yan1 hou2", "哽咽": "geng3 ye4",
+    "咽": "yan4", "喉": "hou2",
+    "歌曲家": "ge1 qu3 jia1", "作曲": "zuo4 qu3",
+    "曲折": "qu1 zhe2", "折": "zhe2", "折本": "she2 ben3",
+    "打折": "da3 zhe2",
+    # common-word batch 3 (frequency reinforcement)
+    "时期": "shi2 qi1", "星期": "xing1 qi1", "期间": "qi1 jian1",
+    "星期天": "xing1 qi1 tian1", "星期日": "xing1 qi1 ri4",
+    "礼拜": "li3 bai4", "周末": "zhou1 mo4", "月份": "yue4 fen4",
+    "小时": "xiao3 shi2", "钟头": "zhong1 tou2",
+    "刚才": "gang1 cai2", "以后": "yi3 hou4", "之后": "zhi1 hou4",
+    "之前": "zhi1 qian2", "从前": "cong2 qian2",
+    "后天": "hou4 tian1", "前天": "qian2 tian1",
+    "早上": "zao3 shang4", "晚上": "wan3 shang4",
+    "中午": "zhong1 wu3", "下午": "xia4 wu3", "上午": "shang4 wu3",
+    "半夜": "ban4 ye4", "凌晨": "ling2 chen2", "凌": "ling2",
+    "白天": "bai2 tian1", "夜晚": "ye4 wan3",
+    "左右": "zuo3 you4", "上下": "shang4 xia4",
+    "前后": "qian2 hou4", "内外": "nei4 wai4",
+    "附近": "fu4 jin4", "附": "fu4", "周围": "zhou1 wei2",
+    "围": "wei2", "当中": "dang1 zhong1", "之间": "zhi1 jian1",
+    "对面儿": "dui4 mian4 er2",
+    "房子": "fang2 zi5", "房间里": "fang2 jian1 li3",
+    "屋子": "wu1 zi5", "院子": "yuan4 zi5", "桌子": "zhuo1 zi5",
+    "椅子": "yi3 zi5", "杯子": "bei1 zi5", "盘子": "pan2 zi5",
+    "瓶子": "ping2 zi5", "盒子": "he2 zi5", "袋子": "dai4 zi5",
+    "帽子": "mao4 zi5", "鞋子": "xie2 zi5", "袜子": "wa4 zi5",
+    "裤子": "ku4 zi5", "裙子": "qun2 zi5", "被子": "bei4 zi5",
+    "本子": "ben3 zi5", "票子": "piao4 zi5", "筷子": "kuai4 zi5",
+    "勺子": "shao2 zi5", "刀子": "dao1 zi5", "嗓子": "sang3 zi5",
+    "嗓": "sang3", "肚子": "du4 zi5", "肚": "du4",
+    "脑子": "nao3 zi5", "鼻子": "bi2 zi5", "脖子": "bo2 zi5",
+    "脖": "bo2", "样子": "yang4 zi5", "个子": "ge4 zi5",
+    "日子": "ri4 zi5", "村子": "cun1 zi5", "镇子": "zhen4 zi5",
+    "镇": "zhen4", "街上": "jie1 shang4", "路上": "lu4 shang4",
+    "网上": "wang3 shang4", "网络": "wang3 luo4",
+    "电子邮件": "dian4 zi3 you2 jian4", "邮件": "you2 jian4",
+    "软件": "ruan3 jian4", "硬件": "ying4 jian4",
+    "互联网": "hu4 lian2 wang3", "视频": "shi4 pin2",
+    "频": "pin2", "照片": "zhao4 pian4", "图片": "tu2 pian4",
+    "音频": "yin1 pin2", "语音": "yu3 yin1",
+    "声音": "sheng1 yin1", "噪音": "zao4 yin1", "噪": "zao4",
+    "音响": "yin1 xiang3", "喇叭": "la3 ba1", "喇": "la3",
+    "叭": "ba1", "麦克风": "mai4 ke4 feng1", "麦": "mai4",
+    "合成": "he2 cheng2", "系统地": "xi4 tong3 de5",
+    "程序": "cheng2 xu4", "序": "xu4", "代码": "dai4 ma3",
+    "键盘": "jian4 pan2", "屏幕": "ping2 mu4", "屏": "ping2",
+    "幕": "mu4", "鼠标": "shu3 biao1", "鼠": "shu3",
+    "倔": "jue2", "克": "ke4", "古": "gu3", "哽": "geng3",
+    "堵": "du3", "带": "dai4", "称": "cheng1", "络": "luo4",
 }
 
 # Traditional -> simplified for the characters in the frequency core
