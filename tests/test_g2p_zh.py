@@ -156,3 +156,26 @@ def test_cmn_reads_traditional_script():
     assert (cmn_word_to_ipa("中國經濟發展")
             == cmn_word_to_ipa("中国经济发展"))
     assert cmn_word_to_ipa("這兒") == cmn_word_to_ipa("这儿")  # erhua too
+
+
+def test_cmn_corpus_coverage():
+    """News/daily-life probe corpus reads with no dropped characters
+    (the frequency core covers running text; regression guard for
+    dictionary edits)."""
+    from sonata_amd.text.g2p_zh import _is_hanzi
+
+    corpus = (
+        "人工智能技术正在改变我们的生活方式。"
+        "语音合成系统可以把文字转换成自然流畅的声音。"
+        "科学家们经过多年研究，开发出了新型计算机芯片。"
+        "春天来了，公园里开满了鲜花，孩子们在草地上奔跑玩耍。"
+        "昨天晚上我和朋友一起去餐厅吃饭，味道非常好。"
+        "政府宣布将加大对教育和医疗的投入，提高人民生活水平。"
+        "火车站离机场不远，乘坐地铁大约需要三十分钟。"
+        "他每天早晨六点起床，先跑步锻炼身体，八点准时上班。"
+        "图书馆里非常安静，学生们认真地复习功课，准备期末考试。"
+        "环境保护越来越重要，我们应该节约用水用电，爱护地球家园。"
+    )
+    dropped = [c for c in corpus
+               if _is_hanzi(c) and not cmn_word_to_ipa(c)]
+    assert not dropped, "".join(dropped)
