@@ -866,6 +866,30 @@ CMN_CHARS: Dict[str, str] = {
     "幕": "mu4", "鼠标": "shu3 biao1", "鼠": "shu3",
     "倔": "jue2", "克": "ke4", "古": "gu3", "哽": "geng3",
     "堵": "du3", "带": "dai4", "称": "cheng1", "络": "luo4",
+    # batch 4: second probe corpus stragglers + neighbours
+    "临": "lin2", "众": "zhong4", "充": "chong1", "免": "mian3",
+    "压": "ya1", "巨": "ju4", "患": "huan4", "择": "ze2",
+    "掌": "zhang3", "故事": "gu4 shi4", "故": "gu4", "汇": "hui4",
+    "眠": "mian2", "端": "duan1", "繁": "fan2", "练": "lian4",
+    "致": "zhi4", "范": "fan4", "蔬": "shu1", "诈": "zha4",
+    "词": "ci2", "财": "cai2", "购": "gou4", "述": "shu4",
+    "避": "bi4", "防": "fang2", "陌": "mo4", "随": "sui2",
+    "睡眠": "shui4 mian2", "词汇": "ci2 hui4",
+    "避免": "bi4 mian3", "陌生": "mo4 sheng1",
+    "随着": "sui2 zhe5", "充分": "chong1 fen4",
+    "充足": "chong1 zu2", "压力": "ya1 li4",
+    "掌握": "zhang3 wo4", "防范": "fang2 fan4",
+    "零售": "ling2 shou4", "零": "ling2", "售货": "shou4 huo4",
+    "面临": "mian4 lin2", "观众": "guan1 zhong4",
+    "群众": "qun2 zhong4", "大众": "da4 zhong4",
+    "极端": "ji2 duan1", "端午": "duan1 wu3",
+    "频繁": "pin2 fan2", "繁荣": "fan2 rong2",
+    "叙述": "xu4 shu4", "叙": "xu4", "描述": "miao2 shu4",
+    "描": "miao2", "讲述": "jiang3 shu4",
+    "购物": "gou4 wu4", "采购": "cai3 gou4",
+    "财产": "cai2 chan3", "财富": "cai2 fu4",
+    "诈骗": "zha4 pian4", "骗子": "pian4 zi5",
+    "提醒": "ti2 xing3", "醒来": "xing3 lai2",
 }
 
 # Traditional -> simplified for the characters in the frequency core
