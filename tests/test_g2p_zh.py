@@ -179,3 +179,35 @@ def test_cmn_corpus_coverage():
     dropped = [c for c in corpus
                if _is_hanzi(c) and not cmn_word_to_ipa(c)]
     assert not dropped, "".join(dropped)
+
+
+def test_cmn_corpus_coverage_2():
+    """Harder probe (news/health/commerce register) also reads with no
+    dropped characters."""
+    from sonata_amd.text.g2p_zh import _is_hanzi
+
+    corpus = (
+        "根据最新统计数据，全球气候变化导致极端天气事件频繁发生。"
+        "医生建议患者多吃蔬菜水果，保持充足睡眠，避免过度劳累。"
+        "随着互联网技术的快速发展，越来越多的人选择在线购物，"
+        "传统零售行业面临巨大压力。"
+        "学习外语需要长期坚持，积累词汇，才能真正掌握。"
+        "警方提醒市民注意防范电信诈骗，保护好个人信息和财产安全。"
+    )
+    dropped = [c for c in corpus
+               if _is_hanzi(c) and not cmn_word_to_ipa(c)]
+    assert not dropped, "".join(dropped)
+
+
+def test_yue_corpus_coverage():
+    """Colloquial Cantonese probe reads with no dropped characters."""
+    from sonata_amd.text.g2p_zh import _is_hanzi
+
+    corpus = (
+        "我今日去香港買嘢食，天氣好熱，飲咗杯凍奶茶。"
+        "佢哋喺學校讀書，老師教中文同英文。"
+        "唔該畀張飛我，幾多錢呀？"
+    )
+    dropped = [c for c in corpus
+               if _is_hanzi(c) and not yue_word_to_ipa(c)]
+    assert not dropped, "".join(dropped)
