@@ -866,6 +866,11 @@ CMN_CHARS: Dict[str, str] = {
     "幕": "mu4", "鼠标": "shu3 biao1", "鼠": "shu3",
     "倔": "jue2", "克": "ke4", "古": "gu3", "哽": "geng3",
     "堵": "du3", "带": "dai4", "称": "cheng1", "络": "luo4",
+    "做": "zuo4", "印": "yin4", "川": "chuan1", "微": "wei1",
+    "忆": "yi4", "扑": "pu1", "材": "cai2", "诗": "shi1",
+    "赛": "sai4", "比赛": "bi3 sai4", "回忆": "hui2 yi4",
+    "微笑": "wei1 xiao4", "材料": "cai2 liao4",
+    "打印": "da3 yin4", "做饭": "zuo4 fan4",
     # batch 4: second probe corpus stragglers + neighbours
     "临": "lin2", "众": "zhong4", "充": "chong1", "免": "mian3",
     "压": "ya1", "巨": "ju4", "患": "huan4", "择": "ze2",
