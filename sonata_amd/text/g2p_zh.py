@@ -871,6 +871,32 @@ CMN_CHARS: Dict[str, str] = {
     "赛": "sai4", "比赛": "bi3 sai4", "回忆": "hui2 yi4",
     "微笑": "wei1 xiao4", "材料": "cai2 liao4",
     "打印": "da3 yin4", "做饭": "zuo4 fan4",
+    # batch 6: the hundred family surnames core + geography
+    "李": "li3", "赵": "zhao4", "陈": "chen2", "杨": "yang2",
+    "吴": "wu2", "徐": "xu2", "朱": "zhu1", "胡": "hu2",
+    "郭": "guo1", "罗": "luo2", "郑": "zheng4", "梁": "liang2",
+    "宋": "song4", "唐": "tang2", "韩": "han2", "冯": "feng2",
+    "曹": "cao2", "彭": "peng2", "萧": "xiao1", "蒋": "jiang3",
+    "沈": "shen3", "魏": "wei4", "孟": "meng4", "秦": "qin2",
+    "顾": "gu4", "侯": "hou2", "邵": "shao4", "孔": "kong3",
+    "邱": "qiu1", "戴": "dai4", "莫": "mo4", "苏": "su1",
+    "吕": "lv3", "丁": "ding1", "卢": "lu2", "傅": "fu4",
+    "姚": "yao2", "潘": "pan1", "杜": "du4", "余": "yu2",
+    "蔡": "cai4", "袁": "yuan2", "武": "wu3", "杭": "hang2",
+    "津": "jin1", "圳": "zhen4", "连": "lian2", "厦": "sha4",
+    "俄": "e2", "巴": "ba1", "泰": "tai4", "菲": "fei1",
+    "缅": "mian3", "柬": "jian3",
+    "重庆": "chong2 qing4", "厦门": "xia4 men2",
+    "俄罗斯": "e2 luo2 si1", "斯": "si1", "泰国": "tai4 guo2",
+    "印度": "yin4 du4", "巴西": "ba1 xi1", "越南": "yue4 nan2",
+    "缅甸": "mian3 dian4", "甸": "dian4",
+    "柬埔寨": "jian3 pu3 zhai4", "埔": "pu3", "寨": "zhai4",
+    "菲律宾": "fei1 lv4 bin1", "宾": "bin1",
+    "意大利": "yi4 da4 li4", "西班牙": "xi1 ban1 ya2",
+    "牙": "ya2", "加拿大": "jia1 na2 da4", "拿": "na2",
+    "澳大利亚": "ao4 da4 li4 ya4", "澳": "ao4", "亚": "ya4",
+    "欧洲": "ou1 zhou1", "欧": "ou1", "非洲": "fei1 zhou1",
+    "亚洲": "ya4 zhou1", "美洲": "mei3 zhou1",
     # batch 4: second probe corpus stragglers + neighbours
     "临": "lin2", "众": "zhong4", "充": "chong1", "免": "mian3",
     "压": "ya1", "巨": "ju4", "患": "huan4", "择": "ze2",
